@@ -1,0 +1,243 @@
+"""LLMEngine: the synchronous continuous-batching core.
+
+add_request() -> step() loop -> RequestOutputs. This is the native engine
+replacing the reference's vLLM delegation (vllm_model.py:109-111 start_engine)
+— SURVEY.md §7 stage 4.
+"""
+
+from __future__ import annotations
+
+import time
+import uuid
+from typing import Dict, List, Optional, Union
+
+import torch
+
+from kserve_amd import ops
+from kserve_amd.engine.config import EngineConfig
+from kserve_amd.engine.detokenizer import Detokenizer
+from kserve_amd.engine.model_runner import ModelRunner
+from kserve_amd.engine.request import Request, RequestOutput, RequestStatus
+from kserve_amd.engine.sampling_params import SamplingParams
+from kserve_amd.engine.scheduler import Scheduler
+from kserve_amd.logging import logger
+from kserve_amd.metrics import (
+    LLM_E2E_HIST,
+    LLM_GENERATION_TOKENS,
+    LLM_KV_USAGE,
+    LLM_PROMPT_TOKENS,
+    LLM_TTFT_HIST,
+)
+from kserve_amd.models.llama import LlamaForCausalLM
+from kserve_amd.parallel import comm
+
+
+class Sampler:
+    """Builds per-batch parameter tensors and invokes the sampling ops."""
+
+    def __init__(self, device: torch.device, seed: int = 0):
+        self.device = device
+        self.generator = None
+        if device.type == "cpu":
+            self.generator = torch.Generator()
+            self.generator.manual_seed(seed)
+        self._step = 0
+
+    def sample(self, logits: torch.Tensor, requests: List[Request]) -> List[int]:
+        self._step += 1
+        all_greedy = all(r.sampling_params.greedy for r in requests)
+        if all_greedy:
+            return ops.greedy_sample(logits).tolist()
+        temps = torch.tensor(
+            [max(r.sampling_params.temperature, 1e-5) for r in requests],
+            dtype=torch.float32,
+            device=logits.device,
+        )
+        top_p = torch.tensor(
+            [r.sampling_params.top_p for r in requests],
+            dtype=torch.float32,
+            device=logits.device,
+        )
+        top_k = torch.tensor(
+            [r.sampling_params.top_k for r in requests],
+            dtype=torch.int32,
+            device=logits.device,
+        )
+        seeds = torch.tensor(
+            [
+                (r.sampling_params.seed if r.sampling_params.seed is not None else hash(r.request_id) & 0x7FFFFFFF)
+                + self._step * 0x9E3779B1
+                for r in requests
+            ],
+            dtype=torch.int64,
+            device=logits.device,
+        )
+        out = ops.random_sample(
+            logits, temps, top_p, top_k, seeds=seeds, generator=self.generator
+        )
+        # greedy requests in a mixed batch: override with argmax
+        if any(r.sampling_params.greedy for r in requests):
+            greedy = ops.greedy_sample(logits)
+            mask = torch.tensor(
+                [r.sampling_params.greedy for r in requests],
+                dtype=torch.bool,
+                device=logits.device,
+            )
+            out = torch.where(mask, greedy, out)
+        return out.tolist()
+
+
+class LLMEngine:
+    def __init__(
+        self,
+        config: EngineConfig,
+        tokenizer=None,
+        model: Optional[LlamaForCausalLM] = None,
+    ):
+        self.config = config
+        self.tokenizer = tokenizer
+        self.detokenizer = Detokenizer(tokenizer)
+        st = comm.get_state()
+        self.device = torch.device(
+            st.device if st.device != "cpu" else config.device
+            if torch.cuda.is_available()
+            else "cpu"
+        )
+        t0 = time.monotonic()
+        if model is None:
+            model = self._build_model()
+        self.model = model
+        self.runner = ModelRunner(config, model, str(self.device))
+        num_blocks = self.runner.profile_and_allocate_kv()
+        self.scheduler = Scheduler(config.scheduler, config.cache, num_blocks)
+        self.sampler = Sampler(self.device, config.seed)
+        self.eos_token_id = config.eos_token_id
+        if tokenizer is not None and getattr(tokenizer, "eos_token_id", None) is not None:
+            self.eos_token_id = tokenizer.eos_token_id
+        if self.device.type == "cuda" and not config.enforce_eager:
+            self.runner.capture_decode_graphs()
+        logger.info(
+            "Engine ready in %.1fs: %s (%.2fB params, tp=%d, device=%s)",
+            time.monotonic() - t0,
+            config.model.model_name,
+            self.model.num_parameters() / 1e9,
+            comm.get_state().tp_size,
+            self.device,
+        )
+
+    def _build_model(self) -> LlamaForCausalLM:
+        cfg = self.config
+        with torch.device(self.device):
+            model = LlamaForCausalLM(cfg.model, device=str(self.device))
+        if cfg.model_path:
+            from kserve_amd.engine.weights import load_safetensors_weights
+
+            load_safetensors_weights(model, cfg.model_path)
+        else:
+            model.random_init(seed=cfg.seed)
+        return model
+
+    # -- request API ----------------------------------------------------------
+    def add_request(
+        self,
+        prompt: Union[str, List[int]],
+        sampling_params: Optional[SamplingParams] = None,
+        request_id: Optional[str] = None,
+    ) -> str:
+        request_id = request_id or str(uuid.uuid4())
+        sampling_params = sampling_params or SamplingParams()
+        if isinstance(prompt, str):
+            if self.tokenizer is None:
+                raise ValueError("String prompt requires a tokenizer")
+            prompt_token_ids = self.tokenizer.encode(prompt)
+        else:
+            prompt_token_ids = list(prompt)
+        req = Request(
+            request_id,
+            prompt_token_ids,
+            sampling_params,
+            eos_token_id=self.eos_token_id,
+        )
+        LLM_PROMPT_TOKENS.inc(len(prompt_token_ids))
+        self.scheduler.add_request(req)
+        return request_id
+
+    def abort_request(self, request_id: str):
+        self.scheduler.abort_request(request_id)
+
+    def has_unfinished(self) -> bool:
+        return self.scheduler.has_unfinished()
+
+    # -- step -----------------------------------------------------------------
+    def step(self) -> List[RequestOutput]:
+        batch = self.scheduler.schedule()
+        if not batch:
+            return []
+        if batch.is_prefill:
+            logits = self.runner.execute_prefill(batch, self.scheduler.block_manager)
+        else:
+            logits = self.runner.execute_decode(batch, self.scheduler.block_manager)
+        # advance computed-token counters for executed tokens
+        for req, n in zip(batch.requests, batch.num_scheduled_tokens):
+            req.num_computed_tokens += n
+        # chunked prefill: requests whose prompt isn't fully computed don't
+        # sample this step
+        sample_reqs = [
+            r for r in batch.requests if r.num_computed_tokens >= r.num_tokens
+        ]
+        if len(sample_reqs) < len(batch.requests):
+            idx = [
+                i
+                for i, r in enumerate(batch.requests)
+                if r.num_computed_tokens >= r.num_tokens
+            ]
+            logits = logits[idx]
+        outputs: List[RequestOutput] = []
+        finished: List[Request] = []
+        if sample_reqs:
+            tokens = self.sampler.sample(logits, sample_reqs)
+            LLM_GENERATION_TOKENS.inc(len(tokens))
+            now = time.monotonic()
+            for req, tok in zip(sample_reqs, tokens):
+                first = req.first_token_time is None
+                req.append_output_token(int(tok))
+                if first:
+                    LLM_TTFT_HIST.observe(now - req.arrival_time)
+                req.maybe_finish(self.config.scheduler.max_model_len)
+                delta = self.detokenizer.decode_new(req)
+                if not req.finished and req.sampling_params.stop:
+                    if self.detokenizer.check_stop_strings(req) is not None:
+                        req.status = RequestStatus.FINISHED_STOPPED
+                        req.finish_time = time.monotonic()
+                if req.finished:
+                    finished.append(req)
+                    LLM_E2E_HIST.observe(now - req.arrival_time)
+                outputs.append(
+                    RequestOutput(
+                        request_id=req.request_id,
+                        new_token_ids=[int(tok)],
+                        finished=req.finished,
+                        finish_reason=req.finish_reason,
+                        output_token_ids=list(req.output_token_ids),
+                        num_prompt_tokens=req.num_prompt_tokens,
+                        text_delta=delta,
+                    )
+                )
+        self.scheduler.finish_requests(finished)
+        LLM_KV_USAGE.set(self.scheduler.block_manager.usage)
+        return outputs
+
+    # -- convenience -----------------------------------------------------------
+    def generate(
+        self,
+        prompts: List[Union[str, List[int]]],
+        sampling_params: Optional[SamplingParams] = None,
+    ) -> Dict[str, RequestOutput]:
+        """Blocking batch generation (tests / offline use)."""
+        ids = [self.add_request(p, sampling_params) for p in prompts]
+        results: Dict[str, RequestOutput] = {}
+        while self.has_unfinished():
+            for out in self.step():
+                if out.finished:
+                    results[out.request_id] = out
+        return {rid: results[rid] for rid in ids if rid in results}

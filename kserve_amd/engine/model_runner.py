@@ -1,0 +1,281 @@
+"""Model runner: owns KV caches, prepares step inputs, executes forwards.
+
+MI355X notes:
+- KV layout [num_blocks, num_kv_heads, block_size, head_dim] keeps each
+  (block, head) tile contiguous (4 KB bf16 at 16x128) for LDS staging in the
+  decode kernel.
+- Decode steps are hipGraph-captured per batch-size bucket (static buffers)
+  to eliminate per-step launch overhead (SURVEY.md hard part #1).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from kserve_amd.engine.block_manager import BlockManager
+from kserve_amd.engine.config import EngineConfig
+from kserve_amd.engine.request import Request
+from kserve_amd.engine.scheduler import ScheduledBatch
+from kserve_amd.logging import logger
+from kserve_amd.models.llama import AttentionMetadata, LlamaForCausalLM
+
+_DEFAULT_GRAPH_BATCH_SIZES = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256]
+
+
+class ModelRunner:
+    def __init__(
+        self,
+        config: EngineConfig,
+        model: LlamaForCausalLM,
+        device: str,
+    ):
+        self.config = config
+        self.model = model
+        self.device = torch.device(device)
+        self.is_cuda = self.device.type == "cuda"
+        self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        self.num_gpu_blocks = 0
+        self.max_blocks_per_seq = (
+            config.scheduler.max_model_len + config.cache.block_size - 1
+        ) // config.cache.block_size
+        # hipGraph state
+        self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self._graph_buffers: Optional[Dict[str, torch.Tensor]] = None
+        self._graph_batch_sizes: List[int] = []
+
+    # -- KV cache -----------------------------------------------------------
+    def profile_and_allocate_kv(self) -> int:
+        m = self.config.model
+        tp = max(1, self._tp_size())
+        dtype = self.model.dtype
+        if self.is_cuda:
+            torch.cuda.synchronize()
+            free_bytes, _total = torch.cuda.mem_get_info(self.device)
+            # leave headroom for activations/graphs
+            budget = int(
+                free_bytes * self.config.cache.gpu_memory_utilization
+            ) - (2 << 30)
+            num_blocks = self.config.cache.num_gpu_blocks or max(
+                budget
+                // (
+                    m.kv_bytes_per_token_per_layer
+                    * self.config.cache.block_size
+                    * m.num_layers
+                    // tp
+                ),
+                16,
+            )
+        else:
+            num_blocks = self.config.cache.num_gpu_blocks or 512
+        self.allocate_kv_cache(num_blocks)
+        return num_blocks
+
+    def allocate_kv_cache(self, num_blocks: int):
+        m = self.config.model
+        kv_heads_local = max(1, m.num_kv_heads // max(1, self._tp_size()))
+        shape = (
+            num_blocks,
+            kv_heads_local,
+            self.config.cache.block_size,
+            m.head_dim,
+        )
+        dtype = self.model.dtype
+        self.kv_caches = [
+            (
+                torch.zeros(shape, dtype=dtype, device=self.device),
+                torch.zeros(shape, dtype=dtype, device=self.device),
+            )
+            for _ in range(m.num_layers)
+        ]
+        self.num_gpu_blocks = num_blocks
+        kv_gib = (
+            2 * num_blocks * kv_heads_local * self.config.cache.block_size
+            * m.head_dim * m.num_layers * dtype.itemsize
+        ) / (1 << 30)
+        logger.info(
+            "KV cache: %d blocks x %d tokens (%.1f GiB, layout [blocks, kv_heads, block, head_dim])",
+            num_blocks,
+            self.config.cache.block_size,
+            kv_gib,
+        )
+
+    def _tp_size(self) -> int:
+        from kserve_amd.parallel import comm
+
+        return comm.get_state().tp_size
+
+    # -- input prep -----------------------------------------------------------
+    def prepare_prefill(
+        self, batch: ScheduledBatch, block_manager: BlockManager
+    ):
+        tokens: List[int] = []
+        positions: List[int] = []
+        slot_mapping: List[int] = []
+        cu = [0]
+        max_seqlen = 0
+        last_token_idx: List[int] = []
+        for req, n in zip(batch.requests, batch.num_scheduled_tokens):
+            start = req.num_computed_tokens
+            end = start + n
+            ids = req.all_token_ids[start:end]
+            tokens.extend(ids)
+            positions.extend(range(start, end))
+            slot_mapping.extend(block_manager.slot_mapping(req, start, end))
+            cu.append(cu[-1] + n)
+            max_seqlen = max(max_seqlen, n)
+            last_token_idx.append(cu[-1] - 1)
+        dev = self.device
+        meta = AttentionMetadata(
+            is_prefill=True,
+            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int32, device=dev),
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            max_seqlen=max_seqlen,
+        )
+        input_ids = torch.tensor(tokens, dtype=torch.int64, device=dev)
+        pos = torch.tensor(positions, dtype=torch.int64, device=dev)
+        sel = torch.tensor(last_token_idx, dtype=torch.int64, device=dev)
+        return input_ids, pos, meta, sel
+
+    def prepare_decode(
+        self, batch: ScheduledBatch, block_manager: BlockManager
+    ):
+        tokens: List[int] = []
+        positions: List[int] = []
+        slot_mapping: List[int] = []
+        context_lens: List[int] = []
+        tables: List[List[int]] = []
+        max_blocks = 1
+        for req in batch.requests:
+            pos = req.num_computed_tokens
+            tokens.append(req.all_token_ids[pos])
+            positions.append(pos)
+            slot_mapping.extend(block_manager.slot_mapping(req, pos, pos + 1))
+            context_lens.append(pos + 1)
+            bt = block_manager.get_block_table(req)
+            tables.append(bt)
+            max_blocks = max(max_blocks, len(bt))
+        bt_tensor = torch.zeros(
+            (len(tables), max_blocks), dtype=torch.int32
+        )
+        for i, t in enumerate(tables):
+            bt_tensor[i, : len(t)] = torch.tensor(t, dtype=torch.int32)
+        dev = self.device
+        meta = AttentionMetadata(
+            is_prefill=False,
+            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int32, device=dev),
+            block_tables=bt_tensor.to(dev),
+            context_lens=torch.tensor(context_lens, dtype=torch.int32, device=dev),
+        )
+        input_ids = torch.tensor(tokens, dtype=torch.int64, device=dev)
+        pos = torch.tensor(positions, dtype=torch.int64, device=dev)
+        return input_ids, pos, meta
+
+    # -- execution ----------------------------------------------------------
+    @torch.no_grad()
+    def execute_prefill(self, batch, block_manager) -> torch.Tensor:
+        input_ids, positions, meta, sel = self.prepare_prefill(batch, block_manager)
+        hidden = self.model(input_ids, positions, self.kv_caches, meta)
+        return self.model.compute_logits(hidden[sel])
+
+    @torch.no_grad()
+    def execute_decode(self, batch, block_manager) -> torch.Tensor:
+        n = len(batch.requests)
+        if self._graphs and not self.config.enforce_eager:
+            bucket = self._graph_bucket(n)
+            if bucket is not None:
+                return self._run_graph(bucket, batch, block_manager)
+        input_ids, positions, meta = self.prepare_decode(batch, block_manager)
+        hidden = self.model(input_ids, positions, self.kv_caches, meta)
+        return self.model.compute_logits(hidden)
+
+    # -- hipGraph capture ------------------------------------------------------
+    def capture_decode_graphs(self, batch_sizes: Optional[List[int]] = None):
+        """Capture the decode forward into hipGraphs per batch-size bucket.
+
+        Static input buffers are sized for the largest bucket; smaller
+        batches replay the nearest bucket >= n with padded slots pointing at
+        a scratch block (block 0 is reserved as scratch by convention when
+        graphs are on — the allocator hands out blocks from the top).
+        """
+        if not self.is_cuda:
+            return
+        sizes = [
+            b
+            for b in (batch_sizes or _DEFAULT_GRAPH_BATCH_SIZES)
+            if b <= self.config.scheduler.max_num_seqs
+        ]
+        max_bs = max(sizes)
+        dev = self.device
+        mb = self.max_blocks_per_seq
+        self._graph_buffers = {
+            "input_ids": torch.zeros(max_bs, dtype=torch.int64, device=dev),
+            "positions": torch.zeros(max_bs, dtype=torch.int64, device=dev),
+            "slot_mapping": torch.zeros(max_bs, dtype=torch.int32, device=dev),
+            "block_tables": torch.zeros(
+                (max_bs, mb), dtype=torch.int32, device=dev
+            ),
+            "context_lens": torch.ones(max_bs, dtype=torch.int32, device=dev),
+            "logits": {},
+        }
+        torch.cuda.synchronize()
+        pool = None
+        for bs in sorted(sizes, reverse=True):
+            buf = self._graph_buffers
+            meta = AttentionMetadata(
+                is_prefill=False,
+                slot_mapping=buf["slot_mapping"][:bs],
+                block_tables=buf["block_tables"][:bs],
+                context_lens=buf["context_lens"][:bs],
+            )
+            # warmup run (allocator state, rccl comms)
+            hidden = self.model(
+                buf["input_ids"][:bs], buf["positions"][:bs], self.kv_caches, meta
+            )
+            logits = self.model.compute_logits(hidden)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool):
+                hidden = self.model(
+                    buf["input_ids"][:bs],
+                    buf["positions"][:bs],
+                    self.kv_caches,
+                    meta,
+                )
+                logits = self.model.compute_logits(hidden)
+            if pool is None:
+                pool = g.pool()
+            self._graphs[bs] = g
+            buf["logits"][bs] = logits
+        self._graph_batch_sizes = sorted(self._graphs.keys())
+        torch.cuda.synchronize()
+        logger.info("Captured decode hipGraphs for batch sizes %s", self._graph_batch_sizes)
+
+    def _graph_bucket(self, n: int) -> Optional[int]:
+        for b in self._graph_batch_sizes:
+            if b >= n:
+                return b
+        return None
+
+    def _run_graph(self, bucket: int, batch, block_manager) -> torch.Tensor:
+        n = len(batch.requests)
+        buf = self._graph_buffers
+        input_ids, positions, meta = self.prepare_decode(batch, block_manager)
+        buf["input_ids"][:n].copy_(input_ids)
+        buf["positions"][:n].copy_(positions)
+        buf["slot_mapping"][:n].copy_(meta.slot_mapping)
+        if n < bucket:
+            # padded rows: write scratch KV to slot of block 0, ctx len 1
+            buf["input_ids"][n:bucket].zero_()
+            buf["positions"][n:bucket].zero_()
+            buf["slot_mapping"][n:bucket].fill_(0)
+            buf["context_lens"][n:bucket].fill_(1)
+            buf["block_tables"][n:bucket].zero_()
+        nb = meta.block_tables.shape[1]
+        buf["block_tables"][:n, :nb].copy_(meta.block_tables)
+        if nb < buf["block_tables"].shape[1]:
+            buf["block_tables"][:n, nb:].zero_()
+        buf["context_lens"][:n].copy_(meta.context_lens)
+        self._graphs[bucket].replay()
+        return buf["logits"][bucket][:n]

@@ -1,0 +1,45 @@
+"""Per-request sampling parameters (engine-side mirror of the OpenAI fields).
+
+Semantics match vLLM's documented behavior (the engine the reference
+delegates to — vllm_model.py:248-271); implementation is fresh.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional, Union
+
+
+@dataclass
+class SamplingParams:
+    n: int = 1
+    temperature: float = 1.0
+    top_p: float = 1.0
+    top_k: int = -1
+    max_tokens: int = 16
+    min_tokens: int = 0
+    stop: List[str] = field(default_factory=list)
+    stop_token_ids: List[int] = field(default_factory=list)
+    ignore_eos: bool = False
+    seed: Optional[int] = None
+    logprobs: Optional[int] = None
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
+    echo: bool = False
+
+    def __post_init__(self):
+        if self.temperature < 0:
+            raise ValueError("temperature must be >= 0")
+        if not 0.0 < self.top_p <= 1.0:
+            raise ValueError("top_p must be in (0, 1]")
+        if self.top_k == 0 or self.top_k < -1:
+            raise ValueError("top_k must be -1 (disabled) or >= 1")
+        if self.max_tokens is not None and self.max_tokens < 1:
+            raise ValueError("max_tokens must be >= 1")
+        if self.n < 1:
+            raise ValueError("n must be >= 1")
+
+    @property
+    def greedy(self) -> bool:
+        return self.temperature == 0.0

@@ -1,0 +1,177 @@
+"""Continuous-batching scheduler.
+
+Fresh design (vLLM *semantics* per SURVEY.md §7 stage 4, no code studied):
+prefill-priority batching — each step schedules either a prefill batch
+(waiting requests, FIFO, bounded by token/seq budgets and KV headroom) or a
+decode batch (all running requests, one token each). Out-of-blocks on decode
+preempts the newest running request (recompute-style: its blocks are freed
+and it returns to the head of the waiting queue).
+
+This is the piece the reference's HF fallback lacks (request-serial,
+generative_model.py:372) and its vLLM path outsources.
+"""
+
+from __future__ import annotations
+
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Deque, List, Optional
+
+from kserve_amd.engine.block_manager import BlockManager
+from kserve_amd.engine.config import CacheConfig, SchedulerConfig
+from kserve_amd.engine.request import Request, RequestStatus
+from kserve_amd.metrics import LLM_NUM_RUNNING, LLM_NUM_WAITING, LLM_PREEMPTIONS
+
+
+@dataclass
+class ScheduledBatch:
+    is_prefill: bool
+    requests: List[Request] = field(default_factory=list)
+    # per-request number of tokens to run this step (prefill: prompt chunk;
+    # decode: 1)
+    num_scheduled_tokens: List[int] = field(default_factory=list)
+    preempted: List[Request] = field(default_factory=list)
+
+    @property
+    def total_tokens(self) -> int:
+        return sum(self.num_scheduled_tokens)
+
+    def __bool__(self):
+        return bool(self.requests)
+
+
+class Scheduler:
+    def __init__(
+        self,
+        scheduler_config: SchedulerConfig,
+        cache_config: CacheConfig,
+        num_gpu_blocks: int,
+    ):
+        self.config = scheduler_config
+        self.block_manager = BlockManager(num_gpu_blocks, cache_config.block_size)
+        self.waiting: Deque[Request] = deque()
+        self.running: List[Request] = []
+
+    # -- queue ops -----------------------------------------------------------
+    def add_request(self, request: Request) -> None:
+        if request.num_prompt_tokens > self.config.max_model_len:
+            request.status = RequestStatus.FINISHED_LENGTH
+            return
+        self.waiting.append(request)
+
+    def abort_request(self, request_id: str) -> Optional[Request]:
+        for q in (self.waiting, self.running):
+            for r in list(q):
+                if r.request_id == request_id:
+                    r.status = RequestStatus.FINISHED_ABORTED
+                    q.remove(r)
+                    self.block_manager.free(r)
+                    return r
+        return None
+
+    @property
+    def num_waiting(self) -> int:
+        return len(self.waiting)
+
+    @property
+    def num_running(self) -> int:
+        return len(self.running)
+
+    def has_unfinished(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    # -- scheduling ----------------------------------------------------------
+    def schedule(self) -> ScheduledBatch:
+        batch = self._schedule_prefill()
+        if not batch:
+            batch = self._schedule_decode()
+        LLM_NUM_RUNNING.set(len(self.running))
+        LLM_NUM_WAITING.set(len(self.waiting))
+        return batch
+
+    def _schedule_prefill(self) -> ScheduledBatch:
+        batch = ScheduledBatch(is_prefill=True)
+        token_budget = self.config.max_num_batched_tokens
+        while self.waiting:
+            req = self.waiting[0]
+            # num_tokens (not num_prompt_tokens): a preempted request
+            # re-prefills its generated tokens as context too
+            n_new = req.num_tokens - req.num_computed_tokens
+            if len(self.running) + len(batch.requests) + 1 > self.config.max_num_seqs:
+                break
+            if n_new > token_budget:
+                if self.config.enable_chunked_prefill and token_budget > 0 and batch.total_tokens == 0:
+                    n_new = token_budget
+                else:
+                    break
+            if req.block_table:
+                # resumed chunked prefill: blocks may partially exist
+                need = self.block_manager.blocks_needed(
+                    req.num_computed_tokens + n_new
+                ) - len(req.block_table)
+                if need > self.block_manager.num_free_blocks:
+                    break
+                for _ in range(need):
+                    req.block_table.append(self.block_manager._free.pop())
+            else:
+                if not self.block_manager.can_allocate(
+                    req, req.num_computed_tokens + n_new
+                ):
+                    break
+                self.block_manager.allocate(req, req.num_computed_tokens + n_new)
+            self.waiting.popleft()
+            req.status = RequestStatus.RUNNING
+            batch.requests.append(req)
+            batch.num_scheduled_tokens.append(n_new)
+            token_budget -= n_new
+        for r in batch.requests:
+            self.running.append(r)
+        return batch
+
+    def _schedule_decode(self) -> ScheduledBatch:
+        batch = ScheduledBatch(is_prefill=False)
+        if not self.running:
+            return batch
+        # ensure every running request can take one more token; preempt the
+        # newest-arrived requests when blocks run out (recompute preemption)
+        scheduled: List[Request] = list(self.running)
+
+        def blocks_needed_now(req: Request) -> int:
+            # blocks to extend capacity to num_tokens (the token being decoded)
+            have = len(self.block_manager.get_block_table(req))
+            need = self.block_manager.blocks_needed(req.num_tokens)
+            return max(0, need - have)
+
+        while scheduled:
+            total_needed = sum(blocks_needed_now(r) for r in scheduled)
+            if total_needed <= self.block_manager.num_free_blocks:
+                break
+            victim = scheduled.pop()  # newest in running order
+            self._preempt(victim)
+            batch.preempted.append(victim)
+        for req in scheduled:
+            self.block_manager.append_slot(req)
+            batch.requests.append(req)
+            batch.num_scheduled_tokens.append(1)
+        self.running = scheduled
+        return batch
+
+    def _preempt(self, req: Request) -> None:
+        """Recompute-style preemption: free blocks, re-prefill later.
+
+        Generated tokens stay in ``output_token_ids`` (accounting is
+        unchanged); ``num_computed_tokens=0`` makes the next prefill
+        reprocess ``all_token_ids``.
+        """
+        LLM_PREEMPTIONS.inc()
+        self.block_manager.free(req)
+        req.num_computed_tokens = 0
+        req.status = RequestStatus.PREEMPTED
+        self.waiting.appendleft(req)
+
+    # -- post-step ------------------------------------------------------------
+    def finish_requests(self, finished: List[Request]) -> None:
+        for req in finished:
+            if req in self.running:
+                self.running.remove(req)
+            self.block_manager.free(req)

@@ -1,0 +1,266 @@
+"""Llama-family decoder, MI355X-native.
+
+Fresh implementation of the architecture (replaces the reference's delegation
+to vLLM/HF — SURVEY.md §2.7 op inventory). Forward is built from:
+- kserve_amd.ops HIP kernels: fused residual+RMSNorm, RoPE (host-precomputed
+  cos/sin), flash prefill, paged decode attention, reshape_and_cache,
+  silu_and_mul
+- hipBLASLt GEMMs via F.linear inside the TP-sharded layers
+- RCCL all-reduce over xGMI at the o_proj / down_proj boundaries (2/layer)
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from kserve_amd import ops
+from kserve_amd.engine.config import ModelConfig
+from kserve_amd.ops.torch_ref import make_cos_sin_cache
+from kserve_amd.parallel import comm
+from kserve_amd.parallel.layers import (
+    ColumnParallelLinear,
+    MergedColumnParallelLinear,
+    QKVParallelLinear,
+    RowParallelLinear,
+    VocabParallelEmbedding,
+)
+
+
+@dataclass
+class AttentionMetadata:
+    """Per-step attention inputs prepared by the model runner."""
+
+    is_prefill: bool
+    slot_mapping: torch.Tensor  # [num_tokens] int32 — where new KV goes
+    # prefill
+    cu_seqlens: Optional[torch.Tensor] = None  # [num_seqs+1] int32
+    max_seqlen: int = 0
+    # decode
+    block_tables: Optional[torch.Tensor] = None  # [num_seqs, max_blocks] int32
+    context_lens: Optional[torch.Tensor] = None  # [num_seqs] int32
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, config: ModelConfig, dtype: torch.dtype):
+        super().__init__()
+        st = comm.get_state()
+        self.head_dim = config.head_dim
+        self.scale = 1.0 / math.sqrt(config.head_dim)
+        self.qkv_proj = QKVParallelLinear(
+            config.hidden_size,
+            config.head_dim,
+            config.num_heads,
+            config.num_kv_heads,
+            bias=config.attention_bias,
+            dtype=dtype,
+        )
+        self.o_proj = RowParallelLinear(
+            config.num_heads * config.head_dim,
+            config.hidden_size,
+            bias=False,
+            dtype=dtype,
+        )
+        self.num_heads_local = self.qkv_proj.heads_local
+        self.num_kv_heads_local = self.qkv_proj.kv_heads_local
+
+    def forward(
+        self,
+        hidden: torch.Tensor,  # [T, H]
+        positions: torch.Tensor,  # [T]
+        cos_sin_cache: torch.Tensor,
+        kv_cache: Tuple[torch.Tensor, torch.Tensor],
+        meta: AttentionMetadata,
+    ) -> torch.Tensor:
+        T = hidden.shape[0]
+        q, k, v = self.qkv_proj(hidden)
+        q = q.view(T, self.num_heads_local, self.head_dim)
+        k = k.view(T, self.num_kv_heads_local, self.head_dim)
+        v = v.view(T, self.num_kv_heads_local, self.head_dim)
+        q, k = ops.rotary_embedding(positions, q, k, cos_sin_cache)
+        k_cache, v_cache = kv_cache
+        if k_cache.numel() > 0:
+            ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+        if meta.is_prefill:
+            out = ops.flash_prefill_varlen(
+                q, k, v, meta.cu_seqlens, meta.max_seqlen, self.scale
+            )
+        else:
+            out = ops.paged_attention_decode(
+                q, k_cache, v_cache, meta.block_tables, meta.context_lens, self.scale
+            )
+        return self.o_proj(out.reshape(T, -1))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, config: ModelConfig, dtype: torch.dtype):
+        super().__init__()
+        self.gate_up_proj = MergedColumnParallelLinear(
+            config.hidden_size,
+            config.intermediate_size,
+            bias=config.mlp_bias,
+            dtype=dtype,
+        )
+        self.down_proj = RowParallelLinear(
+            config.intermediate_size, config.hidden_size, bias=False, dtype=dtype
+        )
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, config: ModelConfig, dtype: torch.dtype):
+        super().__init__()
+        self.self_attn = LlamaAttention(config, dtype)
+        self.mlp = LlamaMLP(config, dtype)
+        self.input_layernorm = nn.Parameter(
+            torch.empty(config.hidden_size, dtype=dtype), requires_grad=False
+        )
+        self.post_attention_layernorm = nn.Parameter(
+            torch.empty(config.hidden_size, dtype=dtype), requires_grad=False
+        )
+        self.eps = config.rms_norm_eps
+
+    def forward(self, hidden, residual, positions, cos_sin_cache, kv_cache, meta):
+        if residual is None:
+            residual = hidden
+            hidden = ops.rms_norm(hidden, self.input_layernorm, self.eps)
+        else:
+            hidden, residual = ops.fused_add_rms_norm(
+                hidden, residual, self.input_layernorm, self.eps
+            )
+        hidden = self.self_attn(hidden, positions, cos_sin_cache, kv_cache, meta)
+        hidden, residual = ops.fused_add_rms_norm(
+            hidden, residual, self.post_attention_layernorm, self.eps
+        )
+        hidden = self.mlp(hidden)
+        return hidden, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(
+        self,
+        config: ModelConfig,
+        dtype: Optional[torch.dtype] = None,
+        device: str = "cpu",
+    ):
+        super().__init__()
+        self.config = config
+        dtype = dtype or (
+            torch.bfloat16 if config.dtype == "bfloat16" else torch.float32
+        )
+        self.dtype = dtype
+        self.embed_tokens = VocabParallelEmbedding(
+            config.vocab_size, config.hidden_size, dtype=dtype
+        )
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(config, dtype) for _ in range(config.num_layers)]
+        )
+        self.norm = nn.Parameter(
+            torch.empty(config.hidden_size, dtype=dtype), requires_grad=False
+        )
+        self.lm_head = ColumnParallelLinear(
+            config.hidden_size, config.vocab_size, dtype=dtype, gather_output=True
+        )
+        self.register_buffer(
+            "cos_sin_cache",
+            make_cos_sin_cache(
+                config.head_dim,
+                config.max_position_embeddings,
+                config.rope_theta,
+                dtype=torch.float32,
+            ),
+            persistent=False,
+        )
+        self.eps = config.rms_norm_eps
+        self.to(device)
+
+    # -- init ---------------------------------------------------------------
+    @torch.no_grad()
+    def random_init(self, std: float = 0.02, seed: int = 0):
+        """Fast random init for the synthetic benchmark (BASELINE.json:
+        'random-init weights')."""
+        gen = torch.Generator(device=self.embed_tokens.weight.device)
+        gen.manual_seed(seed + comm.get_state().tp_rank)
+        for p in self.parameters():
+            if p.dim() >= 2:
+                p.normal_(0.0, std, generator=gen)
+            else:
+                p.fill_(1.0)
+        for buf_name in ():
+            pass
+        return self
+
+    # -- forward --------------------------------------------------------------
+    def forward(
+        self,
+        input_ids: torch.Tensor,  # [T]
+        positions: torch.Tensor,  # [T]
+        kv_caches: List[Tuple[torch.Tensor, torch.Tensor]],
+        meta: AttentionMetadata,
+    ) -> torch.Tensor:
+        hidden = self.embed_tokens(input_ids)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            hidden, residual = layer(
+                hidden, residual, positions, self.cos_sin_cache, kv_caches[i], meta
+            )
+        hidden, _ = ops.fused_add_rms_norm(hidden, residual, self.norm, self.eps)
+        return hidden
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        """hidden: [N, H] (already gathered to sampled positions)."""
+        return self.lm_head(hidden)
+
+    # -- HF weight loading ------------------------------------------------------
+    @torch.no_grad()
+    def load_hf_state_dict(self, tensors) -> None:
+        """Load HuggingFace-format llama weights.
+
+        ``tensors``: mapping name -> tensor (may be lazily materialized).
+        TP sharding applied per layer type.
+        """
+        def get(name):
+            t = tensors[name]
+            return t() if callable(t) else t
+
+        self.embed_tokens.load_shard(get("model.embed_tokens.weight"))
+        self.norm.data.copy_(get("model.norm.weight").to(self.dtype))
+        if self.config.tie_word_embeddings or "lm_head.weight" not in tensors:
+            self.lm_head.load_shard(get("model.embed_tokens.weight"))
+        else:
+            self.lm_head.load_shard(get("lm_head.weight"))
+        for i, layer in enumerate(self.layers):
+            p = f"model.layers.{i}."
+            layer.input_layernorm.data.copy_(
+                get(p + "input_layernorm.weight").to(self.dtype)
+            )
+            layer.post_attention_layernorm.data.copy_(
+                get(p + "post_attention_layernorm.weight").to(self.dtype)
+            )
+            qb = kb = vb = None
+            if self.config.attention_bias:
+                qb = get(p + "self_attn.q_proj.bias")
+                kb = get(p + "self_attn.k_proj.bias")
+                vb = get(p + "self_attn.v_proj.bias")
+            layer.self_attn.qkv_proj.load_shards(
+                get(p + "self_attn.q_proj.weight"),
+                get(p + "self_attn.k_proj.weight"),
+                get(p + "self_attn.v_proj.weight"),
+                qb,
+                kb,
+                vb,
+            )
+            layer.self_attn.o_proj.load_shard(get(p + "self_attn.o_proj.weight"))
+            layer.mlp.gate_up_proj.load_shards(
+                get(p + "mlp.gate_proj.weight"), get(p + "mlp.up_proj.weight")
+            )
+            layer.mlp.down_proj.load_shard(get(p + "mlp.down_proj.weight"))
+
+    def num_parameters(self) -> int:
+        return sum(p.numel() for p in self.parameters())
