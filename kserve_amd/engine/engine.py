@@ -221,6 +221,7 @@ class LLMEngine:
                         output_token_ids=list(req.output_token_ids),
                         num_prompt_tokens=req.num_prompt_tokens,
                         text_delta=delta,
+                        output_text=req.output_text,
                     )
                 )
         self.scheduler.finish_requests(finished)

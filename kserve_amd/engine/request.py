@@ -133,4 +133,5 @@ class RequestOutput:
     output_token_ids: List[int] = field(default_factory=list)
     num_prompt_tokens: int = 0
     text_delta: str = ""
+    output_text: str = ""
     logprobs: Optional[list] = None

@@ -1,0 +1,135 @@
+"""OpenAI endpoints end-to-end over the async engine (tiny CPU model)."""
+
+import json
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from kserve_amd.engine.config import (
+    CacheConfig,
+    EngineConfig,
+    ModelConfig,
+    SchedulerConfig,
+)
+from kserve_amd.model_repository import ModelRepository
+from kserve_amd.protocol.dataplane import DataPlane
+from kserve_amd.protocol.rest.openai.endpoints import register_openai_endpoints
+from kserve_amd.protocol.rest.server import create_app
+from kserve_amd.runtimes.llm_model import LLMModel
+
+
+@pytest.fixture(scope="module")
+def client():
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=128),
+        cache=CacheConfig(block_size=4, num_gpu_blocks=128),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=256, max_model_len=128
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    model = LLMModel("tiny", cfg)
+    repo = ModelRepository()
+    repo.update(model)
+    dataplane = DataPlane(repo)
+    app = create_app(dataplane)
+    register_openai_endpoints(app, dataplane, [model])
+
+    with TestClient(app) as c:
+        # start engine manually (TestClient doesn't run ModelServer lifecycle)
+        import asyncio
+
+        loop = asyncio.new_event_loop()
+        loop.run_until_complete(model.start_engine())
+        yield c
+        model.stop()
+
+
+def test_models_list(client):
+    r = client.get("/openai/v1/models")
+    assert r.status_code == 200
+    assert r.json()["data"][0]["id"] == "tiny"
+
+
+def test_completion_token_ids(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [1, 2, 3, 4],
+            "max_tokens": 5,
+            "temperature": 0.0,
+        },
+    )
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["object"] == "text_completion"
+    assert body["choices"][0]["finish_reason"] == "length"
+    assert body["usage"]["prompt_tokens"] == 4
+    assert body["usage"]["completion_tokens"] == 5
+    # no tokenizer -> text is space-separated token ids
+    assert len(body["choices"][0]["text"].split()) == 5
+
+
+def test_completion_stream(client):
+    with client.stream(
+        "POST",
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [5, 6, 7],
+            "max_tokens": 4,
+            "temperature": 0.0,
+            "stream": True,
+        },
+    ) as r:
+        assert r.status_code == 200
+        events = []
+        for line in r.iter_lines():
+            if line.startswith("data: "):
+                events.append(line[len("data: ") :])
+    assert events[-1] == "[DONE]"
+    chunks = [json.loads(e) for e in events[:-1]]
+    assert len(chunks) == 4
+    assert chunks[-1]["choices"][0]["finish_reason"] == "length"
+
+
+def test_chat_completion(client):
+    r = client.post(
+        "/openai/v1/chat/completions",
+        json={
+            "model": "tiny",
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 3,
+            "temperature": 0.0,
+        },
+    )
+    # no tokenizer -> chat template fallback requires tokenizer -> 400
+    assert r.status_code == 400
+
+
+def test_unknown_model(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={"model": "nope", "prompt": [1], "max_tokens": 1},
+    )
+    assert r.status_code == 404
+
+
+def test_bad_request(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={"model": "tiny", "max_tokens": 1},
+    )
+    assert r.status_code == 400
+
+
+def test_v1_alias(client):
+    r = client.post(
+        "/v1/completions",
+        json={"model": "tiny", "prompt": [1, 2], "max_tokens": 2, "temperature": 0.0},
+    )
+    assert r.status_code == 200
