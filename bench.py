@@ -1,0 +1,177 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark: Llama-3-8B decode throughput (output tok/s).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W` — for N>1
+launched under torch.distributed.run with one rank per GPU. Each rank runs an
+independent engine replica (data-parallel serving, the deployment shape of
+BASELINE.json config "Llama-3-8B /v1/completions at 1/2/4/8 MI355X"); the
+reported value is the WHOLE-JOB aggregate output tok/s.
+
+A "step" = one continuous-batching decode iteration over the full running
+batch. Synthetic random prompts, random-init bf16 weights (no network).
+TTFT p50 is measured on the prefill wave and reported alongside.
+"""
+
+import argparse
+import json
+import os
+import statistics
+import time
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=64)
+    parser.add_argument("--warmup", type=int, default=16)
+    parser.add_argument("--concurrency", type=int, default=128,
+                        help="concurrent sequences per GPU")
+    parser.add_argument("--prompt-len", type=int, default=512)
+    parser.add_argument("--model", type=str, default="llama-3-8b",
+                        choices=["llama-3-8b", "llama-3-70b", "tiny"])
+    parser.add_argument("--tp", type=int, default=1,
+                        help="tensor-parallel degree (ranks per engine)")
+    parser.add_argument("--kv-blocks", type=int, default=0,
+                        help="override KV pool size (0 = derive from HBM)")
+    parser.add_argument("--eager", action="store_true",
+                        help="disable hipGraph capture")
+    parser.add_argument("--temperature", type=float, default=0.0)
+    args = parser.parse_args()
+
+    import torch
+
+    from kserve_amd.engine.config import (
+        CacheConfig,
+        EngineConfig,
+        ModelConfig,
+        SchedulerConfig,
+    )
+    from kserve_amd.engine.engine import LLMEngine
+    from kserve_amd.engine.sampling_params import SamplingParams
+    from kserve_amd.parallel import comm
+
+    state = comm.init_distributed(tp_size=args.tp)
+    rank, world = state.rank, state.world_size
+    use_gpu = torch.cuda.is_available()
+    device = state.device if use_gpu else "cpu"
+
+    if args.model == "llama-3-8b":
+        mcfg = ModelConfig.llama3_8b()
+    elif args.model == "llama-3-70b":
+        mcfg = ModelConfig.llama3_70b()
+    else:
+        mcfg = ModelConfig(
+            vocab_size=1024, hidden_size=512, intermediate_size=1024,
+            num_layers=2, num_heads=4, num_kv_heads=2, head_dim=128,
+            max_position_embeddings=2048, model_name="tiny",
+        )
+
+    max_model_len = args.prompt_len + args.steps + args.warmup + 64
+    cfg = EngineConfig(
+        model=mcfg,
+        cache=CacheConfig(
+            block_size=16,
+            num_gpu_blocks=args.kv_blocks or None,
+            gpu_memory_utilization=0.9,
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=args.concurrency,
+            max_num_batched_tokens=16384,
+            max_model_len=max_model_len,
+        ),
+        device=device,
+        seed=1234 + state.dp_rank,
+        enforce_eager=args.eager or not use_gpu,
+        eos_token_id=-1,
+    )
+
+    engine = LLMEngine(cfg)
+
+    gen = torch.Generator().manual_seed(42 + state.dp_rank)
+    sp = SamplingParams(
+        temperature=args.temperature,
+        max_tokens=max_model_len,  # never finishes inside the run
+        ignore_eos=True,
+    )
+    t_submit = time.perf_counter()
+    for i in range(args.concurrency):
+        prompt = torch.randint(
+            0, mcfg.vocab_size, (args.prompt_len,), generator=gen
+        ).tolist()
+        engine.add_request(prompt, sp, request_id=f"bench-{rank}-{i}")
+
+    # ---- prefill wave (also measures TTFT) ----
+    ttfts = []
+    while engine.scheduler.num_waiting > 0 or any(
+        r.first_token_time is None for r in engine.scheduler.running
+    ):
+        engine.step()
+    for r in engine.scheduler.running:
+        ttfts.append(r.first_token_time - t_submit)
+    ttft_p50_ms = statistics.median(ttfts) * 1000 if ttfts else None
+    prefill_done = time.perf_counter()
+
+    # ---- warmup decode steps ----
+    for _ in range(args.warmup):
+        engine.step()
+
+    # ---- timed region ----
+    comm.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    tokens = 0
+    for _ in range(args.steps):
+        outs = engine.step()
+        tokens += len(outs)
+    comm.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+
+    # max elapsed over ranks; sum tokens over ranks
+    if world > 1:
+        import torch.distributed as dist
+
+        te = torch.tensor([elapsed], dtype=torch.float64)
+        tk = torch.tensor([float(tokens)], dtype=torch.float64)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        dist.all_reduce(tk, op=dist.ReduceOp.SUM)
+        elapsed = float(te[0])
+        tokens = int(tk[0])
+
+    if rank == 0:
+        value = tokens / elapsed
+        n_gpus = world if use_gpu else args.gpus
+        result = {
+            "metric": "output tok/s, Llama-3-8B /v1/completions decode",
+            "value": round(value, 1),
+            "unit": "tok/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bfloat16",
+            "data": "synthetic random prompts, random-init weights",
+            "ttft_p50_ms": round(ttft_p50_ms, 1) if ttft_p50_ms else None,
+            "config": {
+                "model": mcfg.model_name,
+                "global_batch": args.concurrency * (world // args.tp),
+                "seq_len": args.prompt_len,
+                "parallelism": f"dp{world // args.tp}"
+                + (f"xtp{args.tp}" if args.tp > 1 else ""),
+                "kv_block_size": 16,
+                "hipgraph": not cfg.enforce_eager,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    comm.destroy_distributed()
+
+
+if __name__ == "__main__":
+    main()

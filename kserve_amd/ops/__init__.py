@@ -153,7 +153,7 @@ def random_sample(
     top-k; top-p uses the torch sort path when < 1.0).
     """
     if logits.is_cuda:
-        needs_topp = bool((top_p < 1.0).any())
+        needs_topp = bool((top_p < 1.0).any()) or bool((top_k > 0).any())
         if not needs_topp and seeds is not None:
             out = torch.empty(
                 logits.shape[0], dtype=torch.int64, device=logits.device

@@ -1,0 +1,253 @@
+// Paged attention, decode (single query token per sequence).
+//
+// MI355X-first design (memory-bound KV read; CDNA guide Appendix B):
+//  - workgroup = (seq, kv_head), 4 waves; each wave owns one q head of the
+//    GQA group so the K/V tiles staged in LDS are read once per GROUP, not
+//    once per q head (4x less HBM traffic at group=4).
+//  - KV pages [block=16 tokens][D] staged via 16 B/lane vector loads into
+//    LDS padded to 272 B rows (4-way bank aliasing worst case, G4).
+//  - online softmax in registers; V-phase reads are bank-conflict-free
+//    (lane -> consecutive 4 B within a row).
+//  - split-context (flash-decode style) for small batches: partials
+//    (acc, m, l) to workspace, fused reduction kernel.
+//
+// Replaces the vLLM paged-attention path the reference delegates to
+// (SURVEY.md §2.7 row 1).
+#include "common.h"
+
+#define NEG_INF (-1e30f)
+
+namespace {
+
+constexpr int PAGE = 16;      // tokens per KV page
+constexpr int NWAVES = 4;     // waves per workgroup
+constexpr int LDS_STRIDE = 136;  // bf16 elems per padded row (128 + 8)
+
+// D: head dim (64 or 128). ACC = D/64 output dims per lane.
+template <int D>
+__global__ __launch_bounds__(256) void paged_attention_kernel(
+    short* __restrict__ out,            // [S, H, D] bf16
+    const short* __restrict__ q,        // [S, H, D]
+    const short* __restrict__ k_cache,  // [B, Hkv, PAGE, D]
+    const short* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [S, max_blocks]
+    const int* __restrict__ context_lens,  // [S]
+    const float scale,
+    const int num_kv_heads,
+    const int group,        // q heads per kv head
+    const int max_blocks,
+    // split-context: partial output when n_splits > 1
+    const int n_splits,
+    float* __restrict__ part_out,  // [S, H, n_splits, D]
+    float* __restrict__ part_ml    // [S, H, n_splits, 2] (m, l)
+) {
+  constexpr int ACC = D / 64;
+  const int kv_head = blockIdx.x;
+  const int seq = blockIdx.y;
+  const int split = blockIdx.z;
+  const int ctx = context_lens[seq];
+  if (ctx <= 0) return;
+  const int num_heads = num_kv_heads * group;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tok = lane >> 2;   // token within page handled by this lane
+  const int part = lane & 3;   // quarter of D handled in the QK phase
+
+  __shared__ short k_tile[PAGE * LDS_STRIDE];
+  __shared__ short v_tile[PAGE * LDS_STRIDE];
+
+  const int nblocks = (ctx + PAGE - 1) / PAGE;
+  // split-context range
+  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
+  const int blk_lo = split * blocks_per_split;
+  const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
+  if (blk_lo >= blk_hi) {
+    // record empty partial so the reducer can skip it
+    if (n_splits > 1 && threadIdx.x < (unsigned)group) {
+      const int h = kv_head * group + threadIdx.x;
+      float* ml = part_ml + (((long)seq * num_heads + h) * n_splits + split) * 2;
+      ml[0] = NEG_INF;
+      ml[1] = 0.f;
+    }
+    return;
+  }
+
+  // q fragment for this wave's head: dims [part*D/4, (part+1)*D/4) as f32
+  constexpr int QFRAG = D / 4;  // dims per part
+  float q_frag[QFRAG];
+  const int head = kv_head * group + wave;
+  const bool active = wave < group;
+  if (active) {
+    const short* qp = q + ((long)seq * num_heads + head) * D + part * QFRAG;
+#pragma unroll
+    for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
+  }
+
+  float m = NEG_INF;
+  float l = 0.f;
+  float acc[ACC];
+#pragma unroll
+  for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
+
+  const int dvec = D >> 3;  // short8 per row
+  for (int bi = blk_lo; bi < blk_hi; ++bi) {
+    const int block_id = block_tables[(long)seq * max_blocks + bi];
+    // ---- stage K and V page into LDS (all 256 threads) ----
+    const long src_base = (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    const short8_t* ks = reinterpret_cast<const short8_t*>(k_cache + src_base);
+    const short8_t* vs = reinterpret_cast<const short8_t*>(v_cache + src_base);
+    __syncthreads();
+    for (int i = threadIdx.x; i < PAGE * dvec; i += 256) {
+      const int r = i / dvec;
+      const int c = (i % dvec) * 8;
+      *reinterpret_cast<short8_t*>(&k_tile[r * LDS_STRIDE + c]) = ks[i];
+      *reinterpret_cast<short8_t*>(&v_tile[r * LDS_STRIDE + c]) = vs[i];
+    }
+    __syncthreads();
+    if (!active) continue;
+
+    // ---- QK: each lane computes a quarter-dot for its token ----
+    const short* krow = &k_tile[tok * LDS_STRIDE + part * QFRAG];
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < QFRAG; ++j) s += q_frag[j] * bf16_bits_to_float(krow[j]);
+    s = group_reduce_sum<4>(s);  // full dot in all 4 lanes of the token
+    const int gtok = bi * PAGE + tok;
+    s = (gtok < ctx) ? s * scale : NEG_INF;
+
+    // ---- online softmax update ----
+    const float tmax = wave_reduce_max(s);
+    if (tmax > NEG_INF) {
+      const float m_new = fmaxf(m, tmax);
+      const float rescale = (m > NEG_INF) ? __expf(m - m_new) : 0.f;
+      const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+      const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+      l = l * rescale + psum;
+#pragma unroll
+      for (int a = 0; a < ACC; ++a) acc[a] *= rescale;
+      m = m_new;
+
+      // ---- PV: lane owns ACC consecutive output dim pairs ----
+      // dims for ACC==2 (D=128): {2*lane, 2*lane+1}; ACC==1 (D=64): {lane}
+#pragma unroll 1
+      for (int t = 0; t < PAGE; ++t) {
+        const float pt = __shfl(p, t * 4, 64);
+        if (pt > 0.f) {
+          const short* vrow = &v_tile[t * LDS_STRIDE + lane * ACC];
+#pragma unroll
+          for (int a = 0; a < ACC; ++a)
+            acc[a] += pt * bf16_bits_to_float(vrow[a]);
+        }
+      }
+    }
+  }
+
+  if (!active) return;
+  if (n_splits == 1) {
+    const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
+    short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) op[a] = float_to_bf16_bits(acc[a] * inv_l);
+  } else {
+    float* po =
+        part_out + ((((long)seq * num_heads + head) * n_splits + split)) * D +
+        lane * ACC;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) po[a] = acc[a];
+    if (lane == 0) {
+      float* ml =
+          part_ml + (((long)seq * num_heads + head) * n_splits + split) * 2;
+      ml[0] = m;
+      ml[1] = l;
+    }
+  }
+}
+
+// Combine split-context partials: one wave per (seq, head); lane owns D/64
+// dims across all splits.
+template <int D>
+__global__ void paged_attention_reduce_kernel(
+    short* __restrict__ out,             // [S, H, D]
+    const float* __restrict__ part_out,  // [S, H, n_splits, D]
+    const float* __restrict__ part_ml,   // [S, H, n_splits, 2]
+    const long total_sh, const int n_splits) {
+  constexpr int ACC = D / 64;
+  const long sh = blockIdx.x * (long)(blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (sh >= total_sh) return;
+  const float* ml_base = part_ml + sh * n_splits * 2;
+  float m = NEG_INF;
+  for (int sp = 0; sp < n_splits; ++sp) m = fmaxf(m, ml_base[sp * 2]);
+  float l = 0.f;
+  float acc[ACC];
+#pragma unroll
+  for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
+  for (int sp = 0; sp < n_splits; ++sp) {
+    const float ms = ml_base[sp * 2];
+    const float ls = ml_base[sp * 2 + 1];
+    if (ls <= 0.f || ms <= NEG_INF) continue;
+    const float w = __expf(ms - m);
+    l += ls * w;
+    const float* po = part_out + (sh * n_splits + sp) * D + lane * ACC;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) acc[a] += w * po[a];
+  }
+  const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
+  short* op = out + sh * D + lane * ACC;
+#pragma unroll
+  for (int a = 0; a < ACC; ++a) op[a] = float_to_bf16_bits(acc[a] * inv_l);
+}
+
+}  // namespace
+
+extern "C" hipError_t ks_paged_attention_decode(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    const void* block_tables, const void* context_lens, float scale,
+    int num_seqs, int num_heads, int num_kv_heads, int head_dim,
+    int max_blocks, int block_size, int n_splits, void* part_out,
+    void* part_ml, hipStream_t stream) {
+  if (block_size != PAGE) return hipErrorInvalidValue;
+  const int group = num_heads / num_kv_heads;
+  // one wave per q head: groups > 4 are handled by the caller splitting the
+  // GQA group into 4-head subgroups (sharing a kv head across workgroups)
+  if (group > NWAVES) return hipErrorInvalidValue;
+  if (n_splits < 1) n_splits = 1;
+  dim3 grid(num_kv_heads, num_seqs, n_splits);
+  dim3 block(256);
+  if (head_dim == 128) {
+    hipLaunchKernelGGL((paged_attention_kernel<128>), grid, block, 0, stream,
+                       (short*)out, (const short*)q, (const short*)k_cache,
+                       (const short*)v_cache, (const int*)block_tables,
+                       (const int*)context_lens, scale, num_kv_heads, group,
+                       max_blocks, n_splits, (float*)part_out,
+                       (float*)part_ml);
+  } else if (head_dim == 64) {
+    hipLaunchKernelGGL((paged_attention_kernel<64>), grid, block, 0, stream,
+                       (short*)out, (const short*)q, (const short*)k_cache,
+                       (const short*)v_cache, (const int*)block_tables,
+                       (const int*)context_lens, scale, num_kv_heads, group,
+                       max_blocks, n_splits, (float*)part_out,
+                       (float*)part_ml);
+  } else {
+    return hipErrorInvalidValue;
+  }
+  HIP_CHECK_KERNEL();
+  if (n_splits > 1) {
+    const long sh = (long)num_seqs * num_heads;
+    int waves_per_block = 4;
+    dim3 rgrid((unsigned)((sh + waves_per_block - 1) / waves_per_block));
+    if (head_dim == 128) {
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(waves_per_block * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+    } else {
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<64>), rgrid,
+                         dim3(waves_per_block * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+    }
+    HIP_CHECK_KERNEL();
+  }
+  return hipSuccess;
+}

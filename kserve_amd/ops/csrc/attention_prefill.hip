@@ -1,0 +1,219 @@
+// Flash-style causal prefill attention (varlen, GQA) on MFMA.
+//
+// CDNA4 design (guide §5/§B): per workgroup 4 waves x 16 q-rows = 64 q rows
+// of one (seq, head); K/V staged in LDS in 32-token tiles; QK^T and PV on
+// v_mfma_f32_16x16x32_bf16; online softmax on the C-fragment rows; LDS
+// XOR-swizzled (G4) to kill the D=128 row-major bank conflict; V stored
+// transposed at stage time so the PV B-operand reads are contiguous.
+//
+// Replaces the vLLM flash prefill the reference delegates to
+// (SURVEY.md §2.7 row 1). Numerics oracle: ops/torch_ref.flash_prefill_varlen.
+#include "common.h"
+#include "mfma_layouts.h"
+
+#define NEG_INF (-1e30f)
+
+namespace {
+
+constexpr int KT = 32;        // kv tokens per tile
+constexpr int QW = 16;        // q rows per wave
+constexpr int NWAVES = 4;     // waves per workgroup (64 q rows)
+
+// swizzled LDS index helpers (short units)
+__device__ __forceinline__ int k_idx(int row, int col) {
+  // [32][128] shorts; XOR 8-short chunks with row&7
+  return row * 128 + (col ^ ((row & 7) << 3));
+}
+__device__ __forceinline__ int v_idx(int d, int tok) {
+  // transposed [128 dims][32 tokens] shorts; XOR 8-short chunks with d&3
+  return d * KT + (tok ^ ((d & 3) << 3));
+}
+
+template <int D>  // D == 128
+__global__ __launch_bounds__(256) void flash_prefill_kernel(
+    short* __restrict__ out,      // [T, Hq, D] bf16
+    const short* __restrict__ q,  // [T, Hq, D]
+    const short* __restrict__ k,  // [T, Hkv, D]
+    const short* __restrict__ v,  // [T, Hkv, D]
+    const int* __restrict__ cu_seqlens,  // [S+1]
+    const int Hq, const int Hkv, const float scale) {
+  const int head = blockIdx.x;
+  const int tile = blockIdx.y;
+  const int seq = blockIdx.z;
+  const int q_start = cu_seqlens[seq];
+  const int len = cu_seqlens[seq + 1] - q_start;
+  const int tile_base = tile * (NWAVES * QW);
+  if (tile_base >= len) return;
+  const int group = Hq / Hkv;
+  const int kv_head = head / group;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  __shared__ short k_tile[KT * D];        // 8 KB swizzled
+  __shared__ short v_t[D * KT];           // 8 KB transposed+swizzled
+  __shared__ short p_lds[NWAVES][QW * KT];  // 4 KB
+
+  const int wq0 = tile_base + wave * QW;  // first q row of this wave
+  const bool active = wq0 < len;
+
+  // ---- load Q fragments (A operand), rows clamped to len-1 ----
+  bf16x8_t a_q[D / 32];
+  {
+    const int qrow = min(wq0 + MFMA_RC_OF(lane), len - 1);
+    const short* qp = q + ((long)(q_start + qrow) * Hq + head) * D;
+#pragma unroll
+    for (int c = 0; c < D / 32; ++c) {
+      const int off = c * 32 + ((lane >> 4) << 3);
+      a_q[c] = *reinterpret_cast<const bf16x8_t*>(qp + off);
+    }
+  }
+
+  float m[4], l[4];
+  f32x4_t o_acc[D / 16];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m[r] = NEG_INF; l[r] = 0.f; }
+#pragma unroll
+  for (int c = 0; c < D / 16; ++c) o_acc[c] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+
+  // causal bound: last token needed by this workgroup
+  const int kv_limit = min(len, tile_base + NWAVES * QW);
+  const int ntiles = (kv_limit + KT - 1) / KT;
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int t0 = kt * KT;
+    __syncthreads();
+    // ---- stage K tile (swizzled) and V tile (transposed) ----
+    for (int i = threadIdx.x; i < KT * (D / 8); i += 256) {
+      const int r = i / (D / 8);          // token within tile
+      const int c8 = (i % (D / 8)) * 8;   // first dim of this short8
+      const int tok = t0 + r;
+      short8_t val;
+      if (tok < len) {
+        val = *reinterpret_cast<const short8_t*>(
+            k + ((long)(q_start + tok) * Hkv + kv_head) * D + c8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) val[j] = 0;
+      }
+      *reinterpret_cast<short8_t*>(&k_tile[k_idx(r, c8)]) = val;
+      short8_t vv;
+      if (tok < len) {
+        vv = *reinterpret_cast<const short8_t*>(
+            v + ((long)(q_start + tok) * Hkv + kv_head) * D + c8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vv[j] = 0;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v_t[v_idx(c8 + j, r)] = vv[j];
+    }
+    __syncthreads();
+    if (!active) continue;
+    // causal skip: this wave's rows are all below the tile's first token
+    if (t0 > wq0 + QW - 1) continue;
+
+    // ---- QK^T: S[16 q][32 t] as 2 sub-tiles of 16 tokens ----
+    f32x4_t s_frag[2] = {f32x4_t{0, 0, 0, 0}, f32x4_t{0, 0, 0, 0}};
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      const int tok_row = n * 16 + MFMA_RC_OF(lane);
+#pragma unroll
+      for (int c = 0; c < D / 32; ++c) {
+        const int col = c * 32 + ((lane >> 4) << 3);
+        bf16x8_t bk =
+            *reinterpret_cast<bf16x8_t*>(&k_tile[k_idx(tok_row, col)]);
+        s_frag[n] = mfma16x16x32(a_q[c], bk, s_frag[n]);
+      }
+    }
+    // ---- mask + online softmax ----
+    float p[2][4];
+    float rescale[4];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int qrow = wq0 + MFMA_C_ROW(lane, reg);
+      float sv[2];
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        const int tok = t0 + n * 16 + MFMA_C_COL(lane);
+        const bool valid = (tok <= qrow) && (qrow < len) && (tok < len);
+        sv[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
+      }
+      float rowmax = group_reduce_max<16>(fmaxf(sv[0], sv[1]));
+      const float m_new = fmaxf(m[reg], rowmax);
+      rescale[reg] = (m[reg] > NEG_INF && m_new > NEG_INF)
+                         ? __expf(m[reg] - m_new)
+                         : 0.f;
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+        p[n][reg] = (sv[n] > NEG_INF) ? __expf(sv[n] - m_new) : 0.f;
+      const float rowsum = group_reduce_sum<16>(p[0][reg] + p[1][reg]);
+      l[reg] = l[reg] * rescale[reg] + rowsum;
+      m[reg] = m_new;
+    }
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) o_acc[c][reg] *= rescale[reg];
+    }
+    // ---- P -> LDS (C layout -> A layout relayout) ----
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int prow = MFMA_C_ROW(lane, reg);
+        const int pcol = n * 16 + MFMA_C_COL(lane);
+        p_lds[wave][prow * KT + pcol] = float_to_bf16_bits(p[n][reg]);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    __builtin_amdgcn_sched_barrier(0);  // keep reads after the waitcnt (G#18)
+    // ---- PV: O[16 q][D] += P[16 q][32 t] @ V[32 t][D] ----
+    bf16x8_t pa = *reinterpret_cast<bf16x8_t*>(
+        &p_lds[wave][MFMA_RC_OF(lane) * KT + ((lane >> 4) << 3)]);
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c) {
+      const int d = c * 16 + MFMA_RC_OF(lane);
+      bf16x8_t bv =
+          *reinterpret_cast<bf16x8_t*>(&v_t[v_idx(d, (lane >> 4) << 3)]);
+      o_acc[c] = mfma16x16x32(pa, bv, o_acc[c]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    __builtin_amdgcn_sched_barrier(0);  // p_lds overwritten next iter
+  }
+
+  if (!active) return;
+  // ---- epilogue: normalize and store ----
+  float inv_l[4];
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg)
+    inv_l[reg] = (l[reg] > 0.f) ? 1.f / l[reg] : 0.f;
+#pragma unroll
+  for (int c = 0; c < D / 16; ++c) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int qrow = wq0 + MFMA_C_ROW(lane, reg);
+      if (qrow < len) {
+        out[((long)(q_start + qrow) * Hq + head) * D + c * 16 +
+            MFMA_C_COL(lane)] = float_to_bf16_bits(o_acc[c][reg] * inv_l[reg]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" hipError_t ks_flash_prefill_varlen(
+    void* out, const void* q, const void* k, const void* v,
+    const void* cu_seqlens, int num_seqs, int max_seqlen, int Hq, int Hkv,
+    int head_dim, float scale, hipStream_t stream) {
+  if (head_dim != 128) return hipErrorInvalidValue;
+  if (Hq % Hkv != 0) return hipErrorInvalidValue;
+  const int max_tiles = (max_seqlen + NWAVES * QW - 1) / (NWAVES * QW);
+  if (max_tiles == 0 || num_seqs == 0) return hipSuccess;
+  dim3 grid(Hq, max_tiles, num_seqs);
+  hipLaunchKernelGGL((flash_prefill_kernel<128>), grid, dim3(256), 0, stream,
+                     (short*)out, (const short*)q, (const short*)k,
+                     (const short*)v, (const int*)cu_seqlens, Hq, Hkv, scale);
+  HIP_CHECK_KERNEL();
+  return hipSuccess;
+}

@@ -1,0 +1,225 @@
+// Python bindings for the CDNA4 kernels (torch extension, ROCm-native:
+// c10::hip stream APIs, no CUDA-compat shims).
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include <hip/hip_runtime.h>
+
+// launchers from the .hip translation units
+extern "C" {
+hipError_t ks_rms_norm(void*, const void*, const void*, float, int, int,
+                       hipStream_t);
+hipError_t ks_fused_add_rms_norm(void*, void*, const void*, float, int, int,
+                                 hipStream_t);
+hipError_t ks_silu_and_mul(void*, const void*, int, int, hipStream_t);
+hipError_t ks_rotary_embedding(void*, void*, const void*, const void*, int,
+                               int, int, int, hipStream_t);
+hipError_t ks_reshape_and_cache(const void*, const void*, void*, void*,
+                                const void*, int, int, int, int, hipStream_t);
+hipError_t ks_paged_attention_decode(void*, const void*, const void*,
+                                     const void*, const void*, const void*,
+                                     float, int, int, int, int, int, int, int,
+                                     void*, void*, hipStream_t);
+hipError_t ks_flash_prefill_varlen(void*, const void*, const void*,
+                                   const void*, const void*, int, int, int,
+                                   int, int, float, hipStream_t);
+hipError_t ks_greedy_sample(void*, const void*, int, int, hipStream_t);
+hipError_t ks_gumbel_sample(void*, const void*, const void*, const void*,
+                            const void*, int, int, hipStream_t);
+hipError_t ks_mfma_probe(void*, const void*, const void*, hipStream_t);
+}
+
+namespace {
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_hip(hipError_t err, const char* op) {
+  TORCH_CHECK(err == hipSuccess, op, " failed: ", hipGetErrorString(err));
+}
+
+#define CHECK_BF16_CONTIG(t)                                       \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16"); \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");      \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
+void rms_norm(at::Tensor& out, at::Tensor& input, at::Tensor& weight,
+              double eps) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(input);
+  CHECK_BF16_CONTIG(weight);
+  int hidden = input.size(-1);
+  int rows = input.numel() / hidden;
+  check_hip(ks_rms_norm(out.data_ptr(), input.data_ptr(), weight.data_ptr(),
+                        (float)eps, rows, hidden, current_stream()),
+            "rms_norm");
+}
+
+void fused_add_rms_norm(at::Tensor& x, at::Tensor& residual, at::Tensor& weight,
+                        double eps) {
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(residual);
+  int hidden = x.size(-1);
+  int rows = x.numel() / hidden;
+  check_hip(
+      ks_fused_add_rms_norm(x.data_ptr(), residual.data_ptr(),
+                            weight.data_ptr(), (float)eps, rows, hidden,
+                            current_stream()),
+      "fused_add_rms_norm");
+}
+
+void silu_and_mul(at::Tensor& out, at::Tensor& input) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(input);
+  int d = out.size(-1);
+  TORCH_CHECK(input.size(-1) == 2 * d, "input last dim must be 2*out");
+  int rows = out.numel() / d;
+  check_hip(ks_silu_and_mul(out.data_ptr(), input.data_ptr(), rows, d,
+                            current_stream()),
+            "silu_and_mul");
+}
+
+void rotary_embedding(at::Tensor& positions, at::Tensor& q, at::Tensor& k,
+                      at::Tensor& cos_sin_cache) {
+  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_CONTIG(k);
+  TORCH_CHECK(positions.scalar_type() == at::kLong, "positions must be int64");
+  TORCH_CHECK(cos_sin_cache.scalar_type() == at::kFloat,
+              "cos_sin_cache must be fp32");
+  int T = q.size(0);
+  int Hq = q.size(1);
+  int Hk = k.size(1);
+  int D = q.size(2);
+  check_hip(ks_rotary_embedding(q.data_ptr(), k.data_ptr(),
+                                positions.data_ptr(), cos_sin_cache.data_ptr(),
+                                T, Hq, Hk, D, current_stream()),
+            "rotary_embedding");
+}
+
+void reshape_and_cache(at::Tensor& k, at::Tensor& v, at::Tensor& k_cache,
+                       at::Tensor& v_cache, at::Tensor& slot_mapping) {
+  CHECK_BF16_CONTIG(k);
+  CHECK_BF16_CONTIG(v);
+  CHECK_BF16_CONTIG(k_cache);
+  CHECK_BF16_CONTIG(v_cache);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kInt, "slot_mapping int32");
+  int T = k.size(0);
+  int Hkv = k.size(1);
+  int D = k.size(2);
+  int block_size = k_cache.size(2);
+  check_hip(ks_reshape_and_cache(k.data_ptr(), v.data_ptr(),
+                                 k_cache.data_ptr(), v_cache.data_ptr(),
+                                 slot_mapping.data_ptr(), T, Hkv, D,
+                                 block_size, current_stream()),
+            "reshape_and_cache");
+}
+
+void paged_attention_decode(at::Tensor& out, at::Tensor& q,
+                            at::Tensor& k_cache, at::Tensor& v_cache,
+                            at::Tensor& block_tables, at::Tensor& context_lens,
+                            double scale) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_CONTIG(k_cache);
+  CHECK_BF16_CONTIG(v_cache);
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt, "block_tables int32");
+  TORCH_CHECK(context_lens.scalar_type() == at::kInt, "context_lens int32");
+  int S = q.size(0);
+  int H = q.size(1);
+  int D = q.size(2);
+  int Hkv = k_cache.size(1);
+  int block_size = k_cache.size(2);
+  int max_blocks = block_tables.size(1);
+  // split-context heuristic: fill the 256 CUs (guide §1: need >>256 wgs)
+  int n_splits = 1;
+  long base_wgs = (long)S * Hkv;
+  if (base_wgs < 512) {
+    n_splits = (int)((512 + base_wgs - 1) / base_wgs);
+    if (n_splits > 16) n_splits = 16;
+    int max_split_blocks = (max_blocks + n_splits - 1) / n_splits;
+    if (max_split_blocks < 1) n_splits = 1;
+  }
+  at::Tensor part_out, part_ml;
+  void *po = nullptr, *pml = nullptr;
+  if (n_splits > 1) {
+    auto opts = at::TensorOptions().dtype(at::kFloat).device(q.device());
+    part_out = at::empty({S, H, n_splits, D}, opts);
+    part_ml = at::empty({S, H, n_splits, 2}, opts);
+    po = part_out.data_ptr();
+    pml = part_ml.data_ptr();
+  }
+  check_hip(ks_paged_attention_decode(
+                out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                v_cache.data_ptr(), block_tables.data_ptr(),
+                context_lens.data_ptr(), (float)scale, S, H, Hkv, D,
+                max_blocks, block_size, n_splits, po, pml, current_stream()),
+            "paged_attention_decode");
+}
+
+void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
+                          at::Tensor& v, at::Tensor& cu_seqlens,
+                          int64_t max_seqlen, double scale) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_CONTIG(k);
+  CHECK_BF16_CONTIG(v);
+  TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt, "cu_seqlens int32");
+  int num_seqs = cu_seqlens.size(0) - 1;
+  int Hq = q.size(1);
+  int Hkv = k.size(1);
+  int D = q.size(2);
+  check_hip(ks_flash_prefill_varlen(out.data_ptr(), q.data_ptr(), k.data_ptr(),
+                                    v.data_ptr(), cu_seqlens.data_ptr(),
+                                    num_seqs, (int)max_seqlen, Hq, Hkv, D,
+                                    (float)scale, current_stream()),
+            "flash_prefill_varlen");
+}
+
+void greedy_sample(at::Tensor& out, at::Tensor& logits) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(out.scalar_type() == at::kLong, "out int64");
+  check_hip(ks_greedy_sample(out.data_ptr(), logits.data_ptr(),
+                             logits.size(0), logits.size(1),
+                             current_stream()),
+            "greedy_sample");
+}
+
+void gumbel_sample(at::Tensor& out, at::Tensor& logits,
+                   at::Tensor& temperatures, at::Tensor& top_k,
+                   at::Tensor& seeds) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(out.scalar_type() == at::kLong, "out int64");
+  TORCH_CHECK(temperatures.scalar_type() == at::kFloat, "temps fp32");
+  TORCH_CHECK(seeds.scalar_type() == at::kLong, "seeds int64");
+  check_hip(ks_gumbel_sample(out.data_ptr(), logits.data_ptr(),
+                             temperatures.data_ptr(), top_k.data_ptr(),
+                             seeds.data_ptr(), logits.size(0), logits.size(1),
+                             current_stream()),
+            "gumbel_sample");
+}
+
+void mfma_probe(at::Tensor& c, at::Tensor& a, at::Tensor& b) {
+  check_hip(ks_mfma_probe(c.data_ptr(), a.data_ptr(), b.data_ptr(),
+                          current_stream()),
+            "mfma_probe");
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "RMSNorm (bf16, CDNA4)");
+  m.def("fused_add_rms_norm", &fused_add_rms_norm,
+        "fused residual-add + RMSNorm, in-place");
+  m.def("silu_and_mul", &silu_and_mul, "SwiGLU activation");
+  m.def("rotary_embedding", &rotary_embedding, "neox RoPE in-place");
+  m.def("reshape_and_cache", &reshape_and_cache, "KV page scatter");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "paged decode attention (GQA, split-context)");
+  m.def("flash_prefill_varlen", &flash_prefill_varlen,
+        "MFMA flash prefill (causal varlen GQA)");
+  m.def("greedy_sample", &greedy_sample, "argmax sampling");
+  m.def("gumbel_sample", &gumbel_sample, "Gumbel-max temperature sampling");
+  m.def("mfma_probe", &mfma_probe, "MFMA layout probe (tests)");
+}

@@ -1,0 +1,54 @@
+// Scatter new K/V rows into the paged cache.
+//
+// Cache layout [num_blocks, Hkv, block_size, D]: each (block, head) tile is a
+// contiguous block_size*D*2-byte region (4 KB at 16x128 bf16) — the unit the
+// decode kernel stages through LDS.
+#include "common.h"
+
+namespace {
+
+__global__ void reshape_and_cache_kernel(
+    const short* __restrict__ k,  // [T, Hkv, D]
+    const short* __restrict__ v,  // [T, Hkv, D]
+    short* __restrict__ k_cache,  // [B, Hkv, bs, D]
+    short* __restrict__ v_cache,
+    const int* __restrict__ slot_mapping,  // [T]
+    const int T, const int Hkv, const int D, const int block_size) {
+  const int t = blockIdx.x;
+  if (t >= T) return;
+  const int slot = slot_mapping[t];
+  if (slot < 0) return;  // padding slot
+  const int blk = slot / block_size;
+  const int off = slot % block_size;
+  const int nvec = (Hkv * D) >> 3;
+  const short8_t* k_src = reinterpret_cast<const short8_t*>(k) + (long)t * nvec;
+  const short8_t* v_src = reinterpret_cast<const short8_t*>(v) + (long)t * nvec;
+  const int dvec = D >> 3;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    const int h = i / dvec;
+    const int dv = i % dvec;
+    const long dst =
+        (((long)blk * Hkv + h) * block_size + off) * dvec + dv;
+    reinterpret_cast<short8_t*>(k_cache)[dst] = k_src[i];
+    reinterpret_cast<short8_t*>(v_cache)[dst] = v_src[i];
+  }
+}
+
+}  // namespace
+
+extern "C" hipError_t ks_reshape_and_cache(const void* k, const void* v,
+                                           void* k_cache, void* v_cache,
+                                           const void* slot_mapping, int T,
+                                           int Hkv, int D, int block_size,
+                                           hipStream_t stream) {
+  if (D % 8 != 0 || T == 0) return T == 0 ? hipSuccess : hipErrorInvalidValue;
+  int threads = (Hkv * D) >> 3;
+  if (threads > 256) threads = 256;
+  if (threads < 64) threads = 64;
+  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(T), dim3(threads), 0,
+                     stream, (const short*)k, (const short*)v, (short*)k_cache,
+                     (short*)v_cache, (const int*)slot_mapping, T, Hkv, D,
+                     block_size);
+  HIP_CHECK_KERNEL();
+  return hipSuccess;
+}

@@ -1,0 +1,222 @@
+"""GPU numerics: each CDNA4 HIP kernel vs the plain-PyTorch fp32 reference.
+
+All tests are @pytest.mark.gpu and run on a real MI355X via gpurun.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from kserve_amd.ops import torch_ref
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    from kserve_amd import ops
+
+    assert ops.has_native(), "native extension must be present on the GPU box"
+    torch.manual_seed(0)
+    return torch.device("cuda:0")
+
+
+def bf16_close(got, ref_f32, atol=2e-2, rtol=2e-2):
+    torch.testing.assert_close(
+        got.float().cpu(), ref_f32.float().cpu(), atol=atol, rtol=rtol
+    )
+
+
+class TestMfmaProbe:
+    def test_mfma_probe(self, dev):
+        """Verifies the A/B fragment layout assumptions in mfma_layouts.h."""
+        import kserve_amd_C
+
+        a = torch.randn(16, 32, dtype=torch.bfloat16, device=dev)
+        b = torch.randn(32, 16, dtype=torch.bfloat16, device=dev)
+        c = torch.zeros(16, 16, dtype=torch.float32, device=dev)
+        kserve_amd_C.mfma_probe(c, a.view(torch.int16), b.view(torch.int16))
+        torch.cuda.synchronize()
+        ref = a.float() @ b.float()
+        torch.testing.assert_close(c.cpu(), ref.cpu(), atol=1e-2, rtol=1e-2)
+
+
+class TestRMSNorm:
+    @pytest.mark.parametrize("rows,hidden", [(1, 4096), (64, 4096), (333, 8192), (7, 64)])
+    def test_rms_norm(self, dev, rows, hidden):
+        from kserve_amd import ops
+
+        x = torch.randn(rows, hidden, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(hidden, dtype=torch.bfloat16, device=dev)
+        got = ops.rms_norm(x, w, 1e-5)
+        ref = torch_ref.rms_norm(x.float().cpu(), w.float().cpu(), 1e-5)
+        bf16_close(got, ref)
+
+    def test_fused_add_rms_norm(self, dev):
+        from kserve_amd import ops
+
+        rows, hidden = 96, 4096
+        x = torch.randn(rows, hidden, dtype=torch.bfloat16, device=dev)
+        res = torch.randn(rows, hidden, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(hidden, dtype=torch.bfloat16, device=dev)
+        x_ref, res_ref = torch_ref.fused_add_rms_norm(
+            x.float().cpu(), res.float().cpu(), w.float().cpu(), 1e-5
+        )
+        got_x, got_res = ops.fused_add_rms_norm(x, res, w, 1e-5)
+        bf16_close(got_res, res_ref)
+        bf16_close(got_x, x_ref)
+
+
+class TestActivation:
+    @pytest.mark.parametrize("rows,d", [(16, 14336), (257, 128)])
+    def test_silu_and_mul(self, dev, rows, d):
+        from kserve_amd import ops
+
+        x = torch.randn(rows, 2 * d, dtype=torch.bfloat16, device=dev)
+        got = ops.silu_and_mul(x)
+        ref = torch_ref.silu_and_mul(x.float().cpu())
+        bf16_close(got, ref)
+
+
+class TestRope:
+    def test_rotary_embedding(self, dev):
+        from kserve_amd import ops
+
+        T, Hq, Hk, D = 77, 8, 2, 128
+        cache = torch_ref.make_cos_sin_cache(D, 512, 500000.0).to(dev)
+        q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=dev)
+        k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device=dev)
+        pos = torch.randint(0, 512, (T,), device=dev)
+        q_ref, k_ref = torch_ref.rotary_embedding(
+            pos.cpu(), q.float().cpu(), k.float().cpu(), cache.cpu()
+        )
+        got_q, got_k = ops.rotary_embedding(pos, q, k, cache)
+        bf16_close(got_q, q_ref)
+        bf16_close(got_k, k_ref)
+
+
+class TestKVCache:
+    def test_reshape_and_cache(self, dev):
+        from kserve_amd import ops
+
+        T, Hkv, D, bs, B = 50, 8, 128, 16, 32
+        k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev)
+        v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev)
+        kc = torch.zeros(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev)
+        vc = torch.zeros_like(kc)
+        slots = torch.randperm(B * bs, device=dev)[:T].to(torch.int32)
+        kc_ref = torch.zeros(B, Hkv, bs, D).float()
+        vc_ref = torch.zeros_like(kc_ref)
+        torch_ref.reshape_and_cache(
+            k.float().cpu(), v.float().cpu(), kc_ref, vc_ref, slots.cpu()
+        )
+        ops.reshape_and_cache(k, v, kc, vc, slots)
+        bf16_close(kc, kc_ref)
+        bf16_close(vc, vc_ref)
+
+
+class TestDecodeAttention:
+    @pytest.mark.parametrize(
+        "S,H,Hkv,ctx_max",
+        [
+            (4, 32, 8, 100),    # llama-8b shape, small batch (split-context)
+            (64, 32, 8, 500),   # moderate batch
+            (2, 8, 2, 17),      # ragged boundary
+            (8, 4, 4, 64),      # group=1 (MHA)
+        ],
+    )
+    def test_paged_decode(self, dev, S, H, Hkv, ctx_max):
+        from kserve_amd import ops
+
+        D, bs = 128, 16
+        torch.manual_seed(S)
+        ctx = torch.randint(1, ctx_max + 1, (S,), dtype=torch.int32)
+        max_blocks = int((int(ctx.max()) + bs - 1) // bs)
+        B = S * max_blocks + 1
+        kc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev)
+        vc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev)
+        # disjoint block tables
+        bt = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+            S, max_blocks
+        )
+        q = torch.randn(S, H, D, dtype=torch.bfloat16, device=dev)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.paged_attention_decode(
+            q, kc, vc, bt.to(dev), ctx.to(dev), scale
+        )
+        ref = torch_ref.paged_attention_decode(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt, ctx, scale
+        )
+        bf16_close(got, ref)
+
+
+class TestPrefillAttention:
+    @pytest.mark.parametrize(
+        "lens,H,Hkv",
+        [
+            ([128], 32, 8),
+            ([64, 200, 13], 32, 8),
+            ([1, 511], 8, 8),
+            ([65], 16, 4),
+        ],
+    )
+    def test_flash_prefill(self, dev, lens, H, Hkv):
+        from kserve_amd import ops
+
+        D = 128
+        torch.manual_seed(sum(lens))
+        total = sum(lens)
+        cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)), dtype=torch.int32)
+        q = torch.randn(total, H, D, dtype=torch.bfloat16, device=dev)
+        k = torch.randn(total, Hkv, D, dtype=torch.bfloat16, device=dev)
+        v = torch.randn(total, Hkv, D, dtype=torch.bfloat16, device=dev)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.flash_prefill_varlen(q, k, v, cu.to(dev), max(lens), scale)
+        ref = torch_ref.flash_prefill_varlen(
+            q.float().cpu(), k.float().cpu(), v.float().cpu(), cu, scale
+        )
+        bf16_close(got, ref, atol=3e-2, rtol=3e-2)
+
+
+class TestSampling:
+    def test_greedy(self, dev):
+        from kserve_amd import ops
+
+        logits = torch.randn(32, 128256, dtype=torch.bfloat16, device=dev)
+        got = ops.greedy_sample(logits)
+        ref = logits.float().argmax(dim=-1).cpu()
+        assert (got.cpu() == ref).all()
+
+    def test_gumbel_distribution(self, dev):
+        """Gumbel-max sampling matches softmax probabilities statistically."""
+        import kserve_amd_C
+
+        vocab = 8
+        probs = torch.tensor([0.4, 0.2, 0.15, 0.1, 0.05, 0.05, 0.03, 0.02])
+        logits_row = probs.log().to(torch.bfloat16)
+        n = 4096
+        logits = logits_row.repeat(n, 1).to(dev)
+        temps = torch.ones(n, dtype=torch.float32, device=dev)
+        top_k = torch.full((n,), -1, dtype=torch.int32, device=dev)
+        seeds = torch.arange(n, dtype=torch.int64, device=dev) * 7919
+        out = torch.empty(n, dtype=torch.int64, device=dev)
+        kserve_amd_C.gumbel_sample(out, logits, temps, top_k, seeds)
+        torch.cuda.synchronize()
+        counts = torch.bincount(out.cpu(), minlength=vocab).float() / n
+        assert torch.allclose(counts, probs, atol=0.05), counts
+
+    def test_gumbel_determinism(self, dev):
+        import kserve_amd_C
+
+        logits = torch.randn(16, 1000, dtype=torch.bfloat16, device=dev)
+        temps = torch.full((16,), 0.8, dtype=torch.float32, device=dev)
+        top_k = torch.full((16,), -1, dtype=torch.int32, device=dev)
+        seeds = torch.arange(16, dtype=torch.int64, device=dev)
+        out1 = torch.empty(16, dtype=torch.int64, device=dev)
+        out2 = torch.empty(16, dtype=torch.int64, device=dev)
+        kserve_amd_C.gumbel_sample(out1, logits, temps, top_k, seeds)
+        kserve_amd_C.gumbel_sample(out2, logits, temps, top_k, seeds)
+        torch.cuda.synchronize()
+        assert (out1 == out2).all()
