@@ -82,10 +82,13 @@ class ModelRunner:
             m.head_dim,
         )
         dtype = self.model.dtype
+        # empty, not zeros: context_lens bound all reads, and zeroing a
+        # 250 GB pool costs seconds of startup
+        alloc = torch.empty if self.is_cuda else torch.zeros
         self.kv_caches = [
             (
-                torch.zeros(shape, dtype=dtype, device=self.device),
-                torch.zeros(shape, dtype=dtype, device=self.device),
+                alloc(shape, dtype=dtype, device=self.device),
+                alloc(shape, dtype=dtype, device=self.device),
             )
             for _ in range(m.num_layers)
         ]

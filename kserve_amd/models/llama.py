@@ -78,10 +78,10 @@ class LlamaAttention(nn.Module):
     ) -> torch.Tensor:
         T = hidden.shape[0]
         q, k, v = self.qkv_proj(hidden)
-        # kernels take dense [T, heads, D] (the qkv split is strided)
-        q = q.contiguous().view(T, self.num_heads_local, self.head_dim)
-        k = k.contiguous().view(T, self.num_kv_heads_local, self.head_dim)
-        v = v.contiguous().view(T, self.num_kv_heads_local, self.head_dim)
+        # strided views into the fused qkv output; kernels take row strides
+        q = q.view(T, self.num_heads_local, self.head_dim)
+        k = k.view(T, self.num_kv_heads_local, self.head_dim)
+        v = v.view(T, self.num_kv_heads_local, self.head_dim)
         q, k = ops.rotary_embedding(positions, q, k, cos_sin_cache)
         k_cache, v_cache = kv_cache
         if k_cache.numel() > 0:

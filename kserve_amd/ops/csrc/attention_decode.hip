@@ -36,6 +36,7 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
     const int num_kv_heads,
     const int group,        // q heads per kv head
     const int max_blocks,
+    const long q_row_stride,  // elements between q token rows
     // split-context: partial output when n_splits > 1
     const int n_splits,
     float* __restrict__ part_out,  // [S, H, n_splits, D]
@@ -78,7 +79,7 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
   const int head = kv_head * group + wave;
   const bool active = wave < group;
   if (active) {
-    const short* qp = q + ((long)seq * num_heads + head) * D + part * QFRAG;
+    const short* qp = q + (long)seq * q_row_stride + (long)head * D + part * QFRAG;
 #pragma unroll
     for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
   }
@@ -204,8 +205,8 @@ extern "C" hipError_t ks_paged_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const void* block_tables, const void* context_lens, float scale,
     int num_seqs, int num_heads, int num_kv_heads, int head_dim,
-    int max_blocks, int block_size, int n_splits, void* part_out,
-    void* part_ml, hipStream_t stream) {
+    int max_blocks, int block_size, long q_row_stride, int n_splits,
+    void* part_out, void* part_ml, hipStream_t stream) {
   if (block_size != PAGE) return hipErrorInvalidValue;
   const int group = num_heads / num_kv_heads;
   // one wave per q head: groups > 4 are handled by the caller splitting the
@@ -219,14 +220,14 @@ extern "C" hipError_t ks_paged_attention_decode(
                        (short*)out, (const short*)q, (const short*)k_cache,
                        (const short*)v_cache, (const int*)block_tables,
                        (const int*)context_lens, scale, num_kv_heads, group,
-                       max_blocks, n_splits, (float*)part_out,
+                       max_blocks, q_row_stride, n_splits, (float*)part_out,
                        (float*)part_ml);
   } else if (head_dim == 64) {
     hipLaunchKernelGGL((paged_attention_kernel<64>), grid, block, 0, stream,
                        (short*)out, (const short*)q, (const short*)k_cache,
                        (const short*)v_cache, (const int*)block_tables,
                        (const int*)context_lens, scale, num_kv_heads, group,
-                       max_blocks, n_splits, (float*)part_out,
+                       max_blocks, q_row_stride, n_splits, (float*)part_out,
                        (float*)part_ml);
   } else {
     return hipErrorInvalidValue;

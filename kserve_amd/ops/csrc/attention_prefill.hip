@@ -36,7 +36,8 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
     const short* __restrict__ k,  // [T, Hkv, D]
     const short* __restrict__ v,  // [T, Hkv, D]
     const int* __restrict__ cu_seqlens,  // [S+1]
-    const int Hq, const int Hkv, const float scale) {
+    const int Hq, const int Hkv, const float scale, const long sq,
+    const long sk, const long sv) {
   const int head = blockIdx.x;
   const int tile = blockIdx.y;
   const int seq = blockIdx.z;
@@ -60,7 +61,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
   bf16x8_t a_q[D / 32];
   {
     const int qrow = min(wq0 + MFMA_RC_OF(lane), len - 1);
-    const short* qp = q + ((long)(q_start + qrow) * Hq + head) * D;
+    const short* qp = q + (long)(q_start + qrow) * sq + (long)head * D;
 #pragma unroll
     for (int c = 0; c < D / 32; ++c) {
       const int off = c * 32 + ((lane >> 4) << 3);
@@ -90,7 +91,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
       short8_t val;
       if (tok < len) {
         val = *reinterpret_cast<const short8_t*>(
-            k + ((long)(q_start + tok) * Hkv + kv_head) * D + c8);
+            k + (long)(q_start + tok) * sk + (long)kv_head * D + c8);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) val[j] = 0;
@@ -99,7 +100,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
       short8_t vv;
       if (tok < len) {
         vv = *reinterpret_cast<const short8_t*>(
-            v + ((long)(q_start + tok) * Hkv + kv_head) * D + c8);
+            v + (long)(q_start + tok) * sv + (long)kv_head * D + c8);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) vv[j] = 0;
@@ -205,7 +206,8 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 extern "C" hipError_t ks_flash_prefill_varlen(
     void* out, const void* q, const void* k, const void* v,
     const void* cu_seqlens, int num_seqs, int max_seqlen, int Hq, int Hkv,
-    int head_dim, float scale, hipStream_t stream) {
+    int head_dim, float scale, long sq, long sk, long sv,
+    hipStream_t stream) {
   if (head_dim != 128) return hipErrorInvalidValue;
   if (Hq % Hkv != 0) return hipErrorInvalidValue;
   const int max_tiles = (max_seqlen + NWAVES * QW - 1) / (NWAVES * QW);
@@ -213,7 +215,8 @@ extern "C" hipError_t ks_flash_prefill_varlen(
   dim3 grid(Hq, max_tiles, num_seqs);
   hipLaunchKernelGGL((flash_prefill_kernel<128>), grid, dim3(256), 0, stream,
                      (short*)out, (const short*)q, (const short*)k,
-                     (const short*)v, (const int*)cu_seqlens, Hq, Hkv, scale);
+                     (const short*)v, (const int*)cu_seqlens, Hq, Hkv, scale,
+                     sq, sk, sv);
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }

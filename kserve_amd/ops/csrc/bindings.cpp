@@ -14,16 +14,18 @@ hipError_t ks_fused_add_rms_norm(void*, void*, const void*, float, int, int,
                                  hipStream_t);
 hipError_t ks_silu_and_mul(void*, const void*, int, int, hipStream_t);
 hipError_t ks_rotary_embedding(void*, void*, const void*, const void*, int,
-                               int, int, int, hipStream_t);
+                               int, int, int, long, long, hipStream_t);
 hipError_t ks_reshape_and_cache(const void*, const void*, void*, void*,
-                                const void*, int, int, int, int, hipStream_t);
+                                const void*, int, int, int, int, long, long,
+                                hipStream_t);
 hipError_t ks_paged_attention_decode(void*, const void*, const void*,
                                      const void*, const void*, const void*,
-                                     float, int, int, int, int, int, int, int,
-                                     void*, void*, hipStream_t);
+                                     float, int, int, int, int, int, int,
+                                     long, int, void*, void*, hipStream_t);
 hipError_t ks_flash_prefill_varlen(void*, const void*, const void*,
                                    const void*, const void*, int, int, int,
-                                   int, int, float, hipStream_t);
+                                   int, int, float, long, long, long,
+                                   hipStream_t);
 hipError_t ks_greedy_sample(void*, const void*, int, int, hipStream_t);
 hipError_t ks_gumbel_sample(void*, const void*, const void*, const void*,
                             const void*, int, int, hipStream_t);
@@ -44,6 +46,15 @@ void check_hip(hipError_t err, const char* op) {
   TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16"); \
   TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");      \
   TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
+// [T, heads, D] where rows may be strided (qkv split views) but the
+// (head, dim) block of each row is dense
+#define CHECK_BF16_ROWS(t)                                              \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16");  \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                     \
+  TORCH_CHECK((t).dim() == 3 && (t).stride(2) == 1 &&                   \
+                  (t).stride(1) == (t).size(2),                         \
+              #t " inner dims must be dense")
 
 void rms_norm(at::Tensor& out, at::Tensor& input, at::Tensor& weight,
               double eps) {
@@ -83,8 +94,8 @@ void silu_and_mul(at::Tensor& out, at::Tensor& input) {
 
 void rotary_embedding(at::Tensor& positions, at::Tensor& q, at::Tensor& k,
                       at::Tensor& cos_sin_cache) {
-  CHECK_BF16_CONTIG(q);
-  CHECK_BF16_CONTIG(k);
+  CHECK_BF16_ROWS(q);
+  CHECK_BF16_ROWS(k);
   TORCH_CHECK(positions.scalar_type() == at::kLong, "positions must be int64");
   TORCH_CHECK(cos_sin_cache.scalar_type() == at::kFloat,
               "cos_sin_cache must be fp32");
@@ -94,14 +105,15 @@ void rotary_embedding(at::Tensor& positions, at::Tensor& q, at::Tensor& k,
   int D = q.size(2);
   check_hip(ks_rotary_embedding(q.data_ptr(), k.data_ptr(),
                                 positions.data_ptr(), cos_sin_cache.data_ptr(),
-                                T, Hq, Hk, D, current_stream()),
+                                T, Hq, Hk, D, (long)q.stride(0),
+                                (long)k.stride(0), current_stream()),
             "rotary_embedding");
 }
 
 void reshape_and_cache(at::Tensor& k, at::Tensor& v, at::Tensor& k_cache,
                        at::Tensor& v_cache, at::Tensor& slot_mapping) {
-  CHECK_BF16_CONTIG(k);
-  CHECK_BF16_CONTIG(v);
+  CHECK_BF16_ROWS(k);
+  CHECK_BF16_ROWS(v);
   CHECK_BF16_CONTIG(k_cache);
   CHECK_BF16_CONTIG(v_cache);
   TORCH_CHECK(slot_mapping.scalar_type() == at::kInt, "slot_mapping int32");
@@ -112,7 +124,8 @@ void reshape_and_cache(at::Tensor& k, at::Tensor& v, at::Tensor& k_cache,
   check_hip(ks_reshape_and_cache(k.data_ptr(), v.data_ptr(),
                                  k_cache.data_ptr(), v_cache.data_ptr(),
                                  slot_mapping.data_ptr(), T, Hkv, D,
-                                 block_size, current_stream()),
+                                 block_size, (long)k.stride(0),
+                                 (long)v.stride(0), current_stream()),
             "reshape_and_cache");
 }
 
@@ -121,7 +134,7 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
                             at::Tensor& block_tables, at::Tensor& context_lens,
                             double scale) {
   CHECK_BF16_CONTIG(out);
-  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_ROWS(q);
   CHECK_BF16_CONTIG(k_cache);
   CHECK_BF16_CONTIG(v_cache);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt, "block_tables int32");
@@ -154,7 +167,8 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
                 out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
                 v_cache.data_ptr(), block_tables.data_ptr(),
                 context_lens.data_ptr(), (float)scale, S, H, Hkv, D,
-                max_blocks, block_size, n_splits, po, pml, current_stream()),
+                max_blocks, block_size, (long)q.stride(0), n_splits, po, pml,
+                current_stream()),
             "paged_attention_decode");
 }
 
@@ -162,9 +176,9 @@ void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
                           at::Tensor& v, at::Tensor& cu_seqlens,
                           int64_t max_seqlen, double scale) {
   CHECK_BF16_CONTIG(out);
-  CHECK_BF16_CONTIG(q);
-  CHECK_BF16_CONTIG(k);
-  CHECK_BF16_CONTIG(v);
+  CHECK_BF16_ROWS(q);
+  CHECK_BF16_ROWS(k);
+  CHECK_BF16_ROWS(v);
   TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt, "cu_seqlens int32");
   int num_seqs = cu_seqlens.size(0) - 1;
   int Hq = q.size(1);
@@ -173,7 +187,9 @@ void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
   check_hip(ks_flash_prefill_varlen(out.data_ptr(), q.data_ptr(), k.data_ptr(),
                                     v.data_ptr(), cu_seqlens.data_ptr(),
                                     num_seqs, (int)max_seqlen, Hq, Hkv, D,
-                                    (float)scale, current_stream()),
+                                    (float)scale, (long)q.stride(0),
+                                    (long)k.stride(0), (long)v.stride(0),
+                                    current_stream()),
             "flash_prefill_varlen");
 }
 

@@ -13,7 +13,8 @@ __global__ void reshape_and_cache_kernel(
     short* __restrict__ k_cache,  // [B, Hkv, bs, D]
     short* __restrict__ v_cache,
     const int* __restrict__ slot_mapping,  // [T]
-    const int T, const int Hkv, const int D, const int block_size) {
+    const int T, const int Hkv, const int D, const int block_size,
+    const long sk, const long sv) {
   const int t = blockIdx.x;
   if (t >= T) return;
   const int slot = slot_mapping[t];
@@ -21,8 +22,8 @@ __global__ void reshape_and_cache_kernel(
   const int blk = slot / block_size;
   const int off = slot % block_size;
   const int nvec = (Hkv * D) >> 3;
-  const short8_t* k_src = reinterpret_cast<const short8_t*>(k) + (long)t * nvec;
-  const short8_t* v_src = reinterpret_cast<const short8_t*>(v) + (long)t * nvec;
+  const short8_t* k_src = reinterpret_cast<const short8_t*>(k + (long)t * sk);
+  const short8_t* v_src = reinterpret_cast<const short8_t*>(v + (long)t * sv);
   const int dvec = D >> 3;
   for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
     const int h = i / dvec;
@@ -40,6 +41,7 @@ extern "C" hipError_t ks_reshape_and_cache(const void* k, const void* v,
                                            void* k_cache, void* v_cache,
                                            const void* slot_mapping, int T,
                                            int Hkv, int D, int block_size,
+                                           long sk, long sv,
                                            hipStream_t stream) {
   if (D % 8 != 0 || T == 0) return T == 0 ? hipSuccess : hipErrorInvalidValue;
   int threads = (Hkv * D) >> 3;
@@ -48,7 +50,7 @@ extern "C" hipError_t ks_reshape_and_cache(const void* k, const void* v,
   hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(T), dim3(threads), 0,
                      stream, (const short*)k, (const short*)v, (short*)k_cache,
                      (short*)v_cache, (const int*)slot_mapping, T, Hkv, D,
-                     block_size);
+                     block_size, sk, sv);
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }

@@ -8,12 +8,12 @@ namespace {
 
 // One wave per (token, head). lane i < half handles dim pair (i, i+half).
 // head_dim <= 128 assumed per wave pass (loops for larger).
-__global__ void rope_kernel(short* __restrict__ q,  // [T, Hq, D]
-                            short* __restrict__ k,  // [T, Hk, D]
+__global__ void rope_kernel(short* __restrict__ q,  // [T, Hq, D] (row stride sq)
+                            short* __restrict__ k,  // [T, Hk, D] (row stride sk)
                             const long* __restrict__ positions,  // [T]
                             const float* __restrict__ cos_sin,   // [P, D]
                             const int T, const int Hq, const int Hk,
-                            const int D) {
+                            const int D, const long sq, const long sk) {
   const int half = D >> 1;
   const int waves_per_block = blockDim.x >> 6;
   const int gwave = blockIdx.x * waves_per_block + (threadIdx.x >> 6);
@@ -25,8 +25,8 @@ __global__ void rope_kernel(short* __restrict__ q,  // [T, Hq, D]
   const int h = gwave % total_heads;
   const long pos = positions[t];
   const float* cs = cos_sin + pos * D;
-  short* base = (h < Hq) ? q + ((long)t * Hq + h) * D
-                         : k + ((long)t * Hk + (h - Hq)) * D;
+  short* base = (h < Hq) ? q + (long)t * sq + (long)h * D
+                         : k + (long)t * sk + (long)(h - Hq) * D;
   for (int i = lane; i < half; i += 64) {
     float c = cs[i];
     float s = cs[half + i];
@@ -42,7 +42,8 @@ __global__ void rope_kernel(short* __restrict__ q,  // [T, Hq, D]
 extern "C" hipError_t ks_rotary_embedding(void* q, void* k,
                                           const void* positions,
                                           const void* cos_sin, int T, int Hq,
-                                          int Hk, int D, hipStream_t stream) {
+                                          int Hk, int D, long sq, long sk,
+                                          hipStream_t stream) {
   if (D % 2 != 0) return hipErrorInvalidValue;
   const long total_waves = (long)T * (Hq + Hk);
   if (total_waves == 0) return hipSuccess;
@@ -50,7 +51,7 @@ extern "C" hipError_t ks_rotary_embedding(void* q, void* k,
   int grid = (int)((total_waves + waves_per_block - 1) / waves_per_block);
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(256), 0, stream, (short*)q,
                      (short*)k, (const long*)positions, (const float*)cos_sin,
-                     T, Hq, Hk, D);
+                     T, Hq, Hk, D, sq, sk);
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }
