@@ -18,6 +18,15 @@ import os
 import statistics
 import time
 
+# hipBLASLt GEMM algo selection via TunableOp: load the committed gfx950
+# tuning table (profiles/tunableop_gfx950<ordinal>.csv) when present.
+_TUNE_BASE = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                          "profiles", "tunableop_gfx950.csv")
+if os.path.exists(_TUNE_BASE.replace(".csv", "0.csv")):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _TUNE_BASE)
+
 
 def main():
     parser = argparse.ArgumentParser()

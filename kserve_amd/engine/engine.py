@@ -225,6 +225,8 @@ class LLMEngine:
                     )
                 )
         self.scheduler.finish_requests(finished)
+        for req in finished:
+            self.runner.release_request(req.request_id)
         LLM_KV_USAGE.set(self.scheduler.block_manager.usage)
         return outputs
 

@@ -44,6 +44,10 @@ class ModelRunner:
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._graph_buffers: Optional[Dict[str, torch.Tensor]] = None
         self._graph_batch_sizes: List[int] = []
+        # pinned staging + per-request numpy block-table cache
+        self._pin: Optional[Dict[str, torch.Tensor]] = None
+        self._pin_np = None
+        self._bt_cache: Dict[str, object] = {}
 
     # -- KV cache -----------------------------------------------------------
     def profile_and_allocate_kv(self) -> int:
@@ -141,38 +145,83 @@ class ModelRunner:
         sel = torch.tensor(last_token_idx, dtype=torch.int64, device=dev)
         return input_ids, pos, meta, sel
 
+    def _ensure_decode_buffers(self, n: int, max_blocks: int):
+        """Pinned host staging buffers + numpy views (fast CPU fill, one
+        async H2D per tensor per step)."""
+        import numpy as np
+
+        need = self._pin is None or self._pin["input_ids"].shape[0] < n or self._pin[
+            "block_tables"
+        ].shape[1] < max_blocks
+        if need:
+            cap = max(n, self.config.scheduler.max_num_seqs)
+            mb = max(max_blocks, self.max_blocks_per_seq)
+            pin = self.is_cuda
+            self._pin = {
+                "input_ids": torch.empty(cap, dtype=torch.int64, pin_memory=pin),
+                "positions": torch.empty(cap, dtype=torch.int64, pin_memory=pin),
+                "slot_mapping": torch.empty(cap, dtype=torch.int32, pin_memory=pin),
+                "context_lens": torch.empty(cap, dtype=torch.int32, pin_memory=pin),
+                "block_tables": torch.zeros(
+                    (cap, mb), dtype=torch.int32, pin_memory=pin
+                ),
+            }
+            self._pin_np = {k: v.numpy() for k, v in self._pin.items()}
+        return self._pin, self._pin_np
+
+    def _fill_pinned(self, batch: ScheduledBatch, block_manager: BlockManager):
+        """Fill the pinned staging buffers for a decode batch; returns
+        (n, nb) = batch size and max block-table width used."""
+        import numpy as np
+
+        n = len(batch.requests)
+        bs = self.config.cache.block_size
+        max_blocks = 1
+        tables = []
+        for req in batch.requests:
+            bt = block_manager.get_block_table(req)
+            tables.append(bt)
+            if len(bt) > max_blocks:
+                max_blocks = len(bt)
+        pin, pnp = self._ensure_decode_buffers(n, max_blocks)
+        ii = pnp["input_ids"]
+        pp = pnp["positions"]
+        sm = pnp["slot_mapping"]
+        ctx = pnp["context_lens"]
+        btab = pnp["block_tables"]
+        for i, req in enumerate(batch.requests):
+            pos = req.num_computed_tokens
+            ii[i] = req.all_token_ids[pos]
+            pp[i] = pos
+            t = tables[i]
+            sm[i] = t[pos // bs] * bs + pos % bs
+            ctx[i] = pos + 1
+            # cached numpy row per request; refresh only on growth
+            cache = self._bt_cache.get(req.request_id)
+            if cache is None or cache.shape[0] != len(t):
+                cache = np.asarray(t, dtype=np.int32)
+                self._bt_cache[req.request_id] = cache
+            btab[i, : cache.shape[0]] = cache
+        return n, max_blocks
+
+    def release_request(self, request_id: str):
+        self._bt_cache.pop(request_id, None)
+
     def prepare_decode(
         self, batch: ScheduledBatch, block_manager: BlockManager
     ):
-        tokens: List[int] = []
-        positions: List[int] = []
-        slot_mapping: List[int] = []
-        context_lens: List[int] = []
-        tables: List[List[int]] = []
-        max_blocks = 1
-        for req in batch.requests:
-            pos = req.num_computed_tokens
-            tokens.append(req.all_token_ids[pos])
-            positions.append(pos)
-            slot_mapping.extend(block_manager.slot_mapping(req, pos, pos + 1))
-            context_lens.append(pos + 1)
-            bt = block_manager.get_block_table(req)
-            tables.append(bt)
-            max_blocks = max(max_blocks, len(bt))
-        bt_tensor = torch.zeros(
-            (len(tables), max_blocks), dtype=torch.int32
-        )
-        for i, t in enumerate(tables):
-            bt_tensor[i, : len(t)] = torch.tensor(t, dtype=torch.int32)
+        n, nb = self._fill_pinned(batch, block_manager)
+        pin = self._pin
         dev = self.device
         meta = AttentionMetadata(
             is_prefill=False,
-            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int32, device=dev),
-            block_tables=bt_tensor.to(dev),
-            context_lens=torch.tensor(context_lens, dtype=torch.int32, device=dev),
+            slot_mapping=pin["slot_mapping"][:n].to(dev, non_blocking=True),
+            block_tables=pin["block_tables"][:n, :nb].to(dev, non_blocking=True)
+            .contiguous(),
+            context_lens=pin["context_lens"][:n].to(dev, non_blocking=True),
         )
-        input_ids = torch.tensor(tokens, dtype=torch.int64, device=dev)
-        pos = torch.tensor(positions, dtype=torch.int64, device=dev)
+        input_ids = pin["input_ids"][:n].to(dev, non_blocking=True)
+        pos = pin["positions"][:n].to(dev, non_blocking=True)
         return input_ids, pos, meta
 
     # -- execution ----------------------------------------------------------
@@ -262,23 +311,28 @@ class ModelRunner:
         return None
 
     def _run_graph(self, bucket: int, batch, block_manager) -> torch.Tensor:
-        n = len(batch.requests)
-        buf = self._graph_buffers
-        input_ids, positions, meta = self.prepare_decode(batch, block_manager)
-        buf["input_ids"][:n].copy_(input_ids)
-        buf["positions"][:n].copy_(positions)
-        buf["slot_mapping"][:n].copy_(meta.slot_mapping)
+        n, nb = self._fill_pinned(batch, block_manager)
+        pin = self._pin
+        pnp = self._pin_np
         if n < bucket:
-            # padded rows: write scratch KV to slot of block 0, ctx len 1
-            buf["input_ids"][n:bucket].zero_()
-            buf["positions"][n:bucket].zero_()
-            buf["slot_mapping"][n:bucket].fill_(0)
-            buf["context_lens"][n:bucket].fill_(1)
-            buf["block_tables"][n:bucket].zero_()
-        nb = meta.block_tables.shape[1]
-        buf["block_tables"][:n, :nb].copy_(meta.block_tables)
-        if nb < buf["block_tables"].shape[1]:
-            buf["block_tables"][:n, nb:].zero_()
-        buf["context_lens"][:n].copy_(meta.context_lens)
+            # padded rows: write scratch KV to block 0 (reserved), ctx len 1
+            pnp["input_ids"][n:bucket] = 0
+            pnp["positions"][n:bucket] = 0
+            pnp["slot_mapping"][n:bucket] = 0
+            pnp["context_lens"][n:bucket] = 1
+            pnp["block_tables"][n:bucket, :nb] = 0
+        buf = self._graph_buffers
+        # single async H2D per tensor, pinned -> static graph buffers
+        buf["input_ids"][:bucket].copy_(pin["input_ids"][:bucket], non_blocking=True)
+        buf["positions"][:bucket].copy_(pin["positions"][:bucket], non_blocking=True)
+        buf["slot_mapping"][:bucket].copy_(
+            pin["slot_mapping"][:bucket], non_blocking=True
+        )
+        buf["context_lens"][:bucket].copy_(
+            pin["context_lens"][:bucket], non_blocking=True
+        )
+        buf["block_tables"][:bucket, :nb].copy_(
+            pin["block_tables"][:bucket, :nb], non_blocking=True
+        )
         self._graphs[bucket].replay()
         return buf["logits"][bucket][:n]
