@@ -21,7 +21,6 @@ namespace {
 
 constexpr int PAGE = 16;      // tokens per KV page
 constexpr int NWAVES = 4;     // waves per workgroup
-constexpr int LDS_STRIDE = 136;  // bf16 elems per padded row (128 + 8)
 
 // D: head dim (64 or 128). ACC = D/64 output dims per lane.
 template <int D>
@@ -53,9 +52,6 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
   const int wave = threadIdx.x >> 6;
   const int tok = lane >> 2;   // token within page handled by this lane
   const int part = lane & 3;   // quarter of D handled in the QK phase
-
-  __shared__ short k_tile[PAGE * LDS_STRIDE];
-  __shared__ short v_tile[PAGE * LDS_STRIDE];
 
   const int nblocks = (ctx + PAGE - 1) / PAGE;
   // split-context range
@@ -90,28 +86,28 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
 #pragma unroll
   for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
 
-  const int dvec = D >> 3;  // short8 per row
+  if (!active) return;
+  // Direct-global K/V reads, no LDS, no barriers: each page is read once per
+  // workgroup group-wise (4 waves share it through L1/L2); barrier-free
+  // iterations let the compiler keep many loads in flight across pages
+  // (CDNA guide common-mistake #7: don't stage what the cache covers).
+  const int bt_base = (int)((long)seq * max_blocks);
   for (int bi = blk_lo; bi < blk_hi; ++bi) {
-    const int block_id = block_tables[(long)seq * max_blocks + bi];
-    // ---- stage K and V page into LDS (all 256 threads) ----
-    const long src_base = (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
-    const short8_t* ks = reinterpret_cast<const short8_t*>(k_cache + src_base);
-    const short8_t* vs = reinterpret_cast<const short8_t*>(v_cache + src_base);
-    __syncthreads();
-    for (int i = threadIdx.x; i < PAGE * dvec; i += 256) {
-      const int r = i / dvec;
-      const int c = (i % dvec) * 8;
-      *reinterpret_cast<short8_t*>(&k_tile[r * LDS_STRIDE + c]) = ks[i];
-      *reinterpret_cast<short8_t*>(&v_tile[r * LDS_STRIDE + c]) = vs[i];
-    }
-    __syncthreads();
-    if (!active) continue;
-
-    // ---- QK: each lane computes a quarter-dot for its token ----
-    const short* krow = &k_tile[tok * LDS_STRIDE + part * QFRAG];
+    const int block_id = block_tables[bt_base + bi];
+    const short* page =
+        k_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    // ---- QK: lane computes a quarter-dot for its token (coalesced: the
+    // wave's 64 lanes cover the full 4 KB page) ----
+    const short8_t* kp =
+        reinterpret_cast<const short8_t*>(page + tok * D + part * QFRAG);
     float s = 0.f;
 #pragma unroll
-    for (int j = 0; j < QFRAG; ++j) s += q_frag[j] * bf16_bits_to_float(krow[j]);
+    for (int c = 0; c < QFRAG / 8; ++c) {
+      short8_t kv8 = kp[c];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        s += q_frag[c * 8 + j] * bf16_bits_to_float(kv8[j]);
+    }
     s = group_reduce_sum<4>(s);  // full dot in all 4 lanes of the token
     const int gtok = bi * PAGE + tok;
     s = (gtok < ctx) ? s * scale : NEG_INF;
@@ -128,17 +124,26 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
       for (int a = 0; a < ACC; ++a) acc[a] *= rescale;
       m = m_new;
 
-      // ---- PV: lane owns ACC consecutive output dim pairs ----
-      // dims for ACC==2 (D=128): {2*lane, 2*lane+1}; ACC==1 (D=64): {lane}
-#pragma unroll 1
+      // ---- PV: lane owns ACC consecutive output dims; the wave's reads of
+      // V row t are a single coalesced 256 B transaction ----
+      const short* vpage =
+          v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+#pragma unroll
       for (int t = 0; t < PAGE; ++t) {
         const float pt = __shfl(p, t * 4, 64);
-        if (pt > 0.f) {
-          const short* vrow = &v_tile[t * LDS_STRIDE + lane * ACC];
-#pragma unroll
-          for (int a = 0; a < ACC; ++a)
-            acc[a] += pt * bf16_bits_to_float(vrow[a]);
+        const short* vrow = vpage + t * D + lane * ACC;
+        float2_t vv;
+        if constexpr (ACC == 2) {
+          const unsigned int packed =
+              *reinterpret_cast<const unsigned int*>(vrow);
+          vv[0] = bf16_bits_to_float((short)(packed & 0xFFFF));
+          vv[1] = bf16_bits_to_float((short)(packed >> 16));
+        } else {
+          vv[0] = bf16_bits_to_float(vrow[0]);
+          vv[1] = 0.f;
         }
+#pragma unroll
+        for (int a = 0; a < ACC; ++a) acc[a] += pt * vv[a];
       }
     }
   }
