@@ -29,6 +29,7 @@ KERNEL_SOURCES = [
     "attention_decode.hip",
     "attention_prefill.hip",
     "sampling.hip",
+    "skinny_gemm.hip",
     "mfma_probe.hip",
 ]
 

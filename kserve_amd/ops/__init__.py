@@ -168,3 +168,27 @@ def random_sample(
     return torch_ref.random_sample(
         logits, temperatures, top_p, top_k, generator=generator
     )
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """GEMM dispatch: custom MFMA skinny-GEMM for decode-sized batches
+    (weight-streaming regime where hipBLASLt underperforms ~4x on gfx950 —
+    see tools/gemm_bench.py), hipBLASLt via F.linear otherwise."""
+    import torch.nn.functional as F
+
+    if (
+        _C is not None
+        and x.is_cuda
+        and bias is None
+        and x.dim() == 2
+        and 0 < x.shape[0] <= 256
+        and weight.shape[0] % 64 == 0
+        and weight.shape[1] % 64 == 0
+        and x.stride(1) == 1
+    ):
+        out = torch.empty(
+            (x.shape[0], weight.shape[0]), dtype=x.dtype, device=x.device
+        )
+        _C.skinny_gemm(out, x, weight)
+        return out
+    return F.linear(x, weight, bias)

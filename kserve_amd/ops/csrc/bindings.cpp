@@ -30,6 +30,8 @@ hipError_t ks_greedy_sample(void*, const void*, int, int, hipStream_t);
 hipError_t ks_gumbel_sample(void*, const void*, const void*, const void*,
                             const void*, int, int, hipStream_t);
 hipError_t ks_mfma_probe(void*, const void*, const void*, hipStream_t);
+hipError_t ks_skinny_gemm(void*, void*, const void*, const void*, int, int,
+                          int, long, hipStream_t);
 }
 
 namespace {
@@ -216,6 +218,30 @@ void gumbel_sample(at::Tensor& out, at::Tensor& logits,
             "gumbel_sample");
 }
 
+void skinny_gemm(at::Tensor& out, at::Tensor& x, at::Tensor& w) {
+  // out [N, M] bf16 = x [N, K] @ w [M, K]^T  (decode shapes, N <= 256)
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(w);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_cuda() &&
+                  x.stride(1) == 1,
+              "x must be bf16 row-dense");
+  int N = x.size(0);
+  int K = x.size(1);
+  int M = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "shape mismatch");
+  at::Tensor ws;
+  void* wsp = nullptr;
+  // split-K needed when the feature grid can't fill the chip
+  if ((M / 64) < 1024 && K / 2 >= 64) {
+    ws = at::empty({(long)N * M},
+                   at::TensorOptions().dtype(at::kFloat).device(x.device()));
+    wsp = ws.data_ptr();
+  }
+  check_hip(ks_skinny_gemm(out.data_ptr(), wsp, x.data_ptr(), w.data_ptr(),
+                           M, K, N, (long)x.stride(0), current_stream()),
+            "skinny_gemm");
+}
+
 void mfma_probe(at::Tensor& c, at::Tensor& a, at::Tensor& b) {
   check_hip(ks_mfma_probe(c.data_ptr(), a.data_ptr(), b.data_ptr(),
                           current_stream()),
@@ -237,5 +263,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA flash prefill (causal varlen GQA)");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");
   m.def("gumbel_sample", &gumbel_sample, "Gumbel-max temperature sampling");
+  m.def("skinny_gemm", &skinny_gemm, "decode GEMM (N<=256, MFMA streaming)");
   m.def("mfma_probe", &mfma_probe, "MFMA layout probe (tests)");
 }

@@ -15,6 +15,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from kserve_amd import ops
 from kserve_amd.parallel import comm
 
 
@@ -54,7 +55,7 @@ class ColumnParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight, self.bias)
+        y = ops.linear(x, self.weight, self.bias)
         if self.gather_output and self.tp_size > 1:
             y = comm.tp_all_gather(y, dim=-1)
         return y
@@ -108,7 +109,7 @@ class QKVParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor):
-        qkv = F.linear(x, self.weight, self.bias)
+        qkv = ops.linear(x, self.weight, self.bias)
         return qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
 
     def load_shards(
@@ -168,7 +169,7 @@ class RowParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight)
+        y = ops.linear(x, self.weight)
         if self.reduce_output:
             y = comm.tp_all_reduce(y)
         if self.bias is not None:
@@ -213,7 +214,7 @@ class MergedColumnParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.weight, self.bias)
+        return ops.linear(x, self.weight, self.bias)
 
     def load_shards(self, gate_weight: torch.Tensor, up_weight: torch.Tensor):
         r = comm.get_state().tp_rank

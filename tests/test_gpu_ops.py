@@ -220,3 +220,42 @@ class TestSampling:
         kserve_amd_C.gumbel_sample(out2, logits, temps, top_k, seeds)
         torch.cuda.synchronize()
         assert (out1 == out2).all()
+
+
+class TestSkinnyGemm:
+    @pytest.mark.parametrize(
+        "N,K,M",
+        [
+            (256, 4096, 6144),   # qkv
+            (256, 14336, 4096),  # down (split-K path)
+            (64, 4096, 4096),    # o_proj small batch
+            (1, 4096, 6144),     # single-token decode
+            (17, 512, 128),      # ragged N, tiny
+            (256, 4096, 128256), # lm_head (no split)
+        ],
+    )
+    def test_skinny_gemm_matches_matmul(self, dev, N, K, M):
+        import kserve_amd_C
+
+        torch.manual_seed(N + K)
+        x = torch.randn(N, K, dtype=torch.bfloat16, device=dev) / 8
+        w = torch.randn(M, K, dtype=torch.bfloat16, device=dev) / 8
+        out = torch.empty(N, M, dtype=torch.bfloat16, device=dev)
+        kserve_amd_C.skinny_gemm(out, x, w)
+        torch.cuda.synchronize()
+        ref = (x.float() @ w.float().t())
+        torch.testing.assert_close(
+            out.float(), ref, atol=0.05, rtol=0.05
+        )
+
+    def test_skinny_gemm_strided_x(self, dev):
+        import kserve_amd_C
+
+        x_full = torch.randn(64, 8192, dtype=torch.bfloat16, device=dev) / 8
+        x = x_full[:, :4096]  # strided rows
+        w = torch.randn(1024, 4096, dtype=torch.bfloat16, device=dev) / 8
+        out = torch.empty(64, 1024, dtype=torch.bfloat16, device=dev)
+        kserve_amd_C.skinny_gemm(out, x, w)
+        torch.cuda.synchronize()
+        ref = x.float() @ w.float().t()
+        torch.testing.assert_close(out.float(), ref, atol=0.05, rtol=0.05)

@@ -51,7 +51,21 @@ SHAPES = [
     ("gate_up_prefill", 16384, 4096, 28672),
 ]
 
-print(f"{'shape':18s} {'N':>6s} {'K':>6s} {'M':>7s} {'us':>9s} {'TB/s(w)':>8s} {'TFLOP/s':>8s}")
+import kserve_amd_C
+
+def bench_custom(x, w, iters):
+    out = torch.empty(x.shape[0], w.shape[0], dtype=torch.bfloat16, device=dev)
+    for _ in range(10):
+        kserve_amd_C.skinny_gemm(out, x, w)
+    torch.cuda.synchronize()
+    t0 = torch.cuda.Event(enable_timing=True); t1 = torch.cuda.Event(enable_timing=True)
+    t0.record()
+    for _ in range(iters):
+        kserve_amd_C.skinny_gemm(out, x, w)
+    t1.record(); torch.cuda.synchronize()
+    return t0.elapsed_time(t1) / iters * 1000
+
+print(f"{'shape':18s} {'N':>6s} {'K':>6s} {'M':>7s} {'us':>9s} {'TB/s(w)':>8s} {'TFLOP/s':>8s} {'cust_us':>8s} {'cust_TB/s':>9s}")
 for name, N, K, M in SHAPES:
     x = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
     w = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
@@ -69,4 +83,8 @@ for name, N, K, M in SHAPES:
     wbytes = M * K * 2
     tbs = wbytes / (us * 1e-6) / 1e12
     tf = 2 * N * K * M / (us * 1e-6) / 1e12
-    print(f"{name:18s} {N:6d} {K:6d} {M:7d} {us:9.1f} {tbs:8.2f} {tf:8.1f}")
+    cus, ctbs = float("nan"), float("nan")
+    if N <= 256 and M % 64 == 0 and K % 64 == 0:
+        cus = bench_custom(x, w, args.iters)
+        ctbs = wbytes / (cus * 1e-6) / 1e12
+    print(f"{name:18s} {N:6d} {K:6d} {M:7d} {us:9.1f} {tbs:8.2f} {tf:8.1f} {cus:8.1f} {ctbs:9.2f}")
