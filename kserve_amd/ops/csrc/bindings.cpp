@@ -231,9 +231,13 @@ void skinny_gemm(at::Tensor& out, at::Tensor& x, at::Tensor& w) {
   TORCH_CHECK(w.size(1) == K, "shape mismatch");
   at::Tensor ws;
   void* wsp = nullptr;
-  // split-K needed when the feature grid can't fill the chip
-  if ((M / 64) < 1024 && K / 2 >= 64) {
-    ws = at::empty({(long)N * M},
+  // mirror the kernel's K-split decision to size the partials workspace
+  int nsplit = 1;
+  int feat_wgs = M / 64;
+  while (feat_wgs * nsplit < 768 && nsplit < 8 && (K / (nsplit * 2)) >= 64)
+    nsplit *= 2;
+  if (nsplit > 1) {
+    ws = at::empty({(long)nsplit * N * M},
                    at::TensorOptions().dtype(at::kFloat).device(x.device()));
     wsp = ws.data_ptr();
   }
