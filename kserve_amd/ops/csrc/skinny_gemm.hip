@@ -42,16 +42,16 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   const int k_lo = blockIdx.y * k_per_split;
   const int k_hi = min(K, k_lo + k_per_split);
 
-  // LDS: X tile [NT*16][BK] then W tile [64][BK], both swizzled
+  // LDS: double-buffered {X tile [NT*16][BK] | W tile [64][BK]}, swizzled,
+  // + one 1 KB scratch slot for stage-count padding
   constexpr int X_CHUNKS = NT * 16 * (BK / 8);  // 16B chunks
   constexpr int W_CHUNKS = 64 * (BK / 8);
-  __shared__ short tile[(NT * 16 + 64) * BK];
-
-  // per-lane source row/chunk for each cooperative stage instruction
-  // (instr i covers chunks [i*64, i*64+64), lane -> chunk i*64+lane;
-  //  LDS dest is linear => source chunk is inverse-swizzled)
   constexpr int TOTAL_CHUNKS = X_CHUNKS + W_CHUNKS;
-  constexpr int NINSTR = TOTAL_CHUNKS / 64;
+  constexpr int NINSTR = TOTAL_CHUNKS / 64;          // wg-wide 1 KB instrs
+  constexpr int NI = (NINSTR + NWAVES - 1) / NWAVES;  // per-wave, padded
+  constexpr int TILE_SHORTS = (NT * 16 + 64) * BK;
+  __shared__ short tile[2][TILE_SHORTS];
+  __shared__ short scratch[512];  // dump target for padding instrs
 
   f32x4_t acc[NT];
 #pragma unroll
@@ -60,45 +60,73 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   const int m_local = lane & 15;
   const int kgrp = lane >> 4;
 
-  for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
-    __syncthreads();
-    // ---- cooperative stage: wave w issues instrs w, w+4, ... ----
-    // (source row/chunk recomputed inline: ~10 VALU per instr beats the
-    //  VGPR cost of precomputing NINSTR offsets at NT=16)
-    for (int i = wave; i < NINSTR; i += NWAVES) {
+  // stage buf with the BK-slice at k-offset k0; every wave issues exactly
+  // NI instructions so the counted s_waitcnt below is uniform (T4)
+  auto stage = [&](int buf, int k0) {
+#pragma unroll
+    for (int ii = 0; ii < NI; ++ii) {
+      const int i = wave + ii * NWAVES;
       const int g = i * 64 + lane;
       const short* src;
-      if (g < X_CHUNKS) {
-        int row = g >> 3;  // token row
-        const int kc_src = (g & 7) ^ (row & 7);
-        if (row >= N) row = N - 1;  // clamp padded token rows
-        src = x + (long)row * xs + (kc_src << 3) + k0;
-      } else {
-        const int gw = g - X_CHUNKS;
-        const int m = gw >> 3;
-        const int kc_src = (gw & 7) ^ (m & 7);
-        src = w + (long)(f_base + m) * K + (kc_src << 3) + k0;
+      short* dst;
+      if (i < NINSTR) {
+        dst = &tile[buf][i * 64 * 8];
+        if (g < X_CHUNKS) {
+          int row = g >> 3;  // token row
+          const int kc_src = (g & 7) ^ (row & 7);
+          if (row >= N) row = N - 1;  // clamp padded token rows
+          src = x + (long)row * xs + (kc_src << 3) + k0;
+        } else {
+          const int gw = g - X_CHUNKS;
+          const int m = gw >> 3;
+          const int kc_src = (gw & 7) ^ (m & 7);
+          src = w + (long)(f_base + m) * K + (kc_src << 3) + k0;
+        }
+      } else {  // padding instr keeps per-wave vmcnt uniform
+        dst = scratch;
+        src = x;
       }
       __builtin_amdgcn_global_load_lds(
           reinterpret_cast<const unsigned int*>(src),
-          reinterpret_cast<unsigned int*>(&tile[i * 64 * 8]), 16, 0, 0);
+          reinterpret_cast<unsigned int*>(dst), 16, 0, 0);
     }
-    asm volatile("s_waitcnt vmcnt(0)");
-    __syncthreads();
-    // ---- fragments + MFMA ----
+  };
+
+  // s_waitcnt imm waiting vmcnt==NI, ignore lgkm/exp (gfx9 encoding:
+  // vmcnt[3:0]|vmcnt[5:4]<<14, expcnt[6:4], lgkmcnt[11:8])
+  constexpr int WAIT_NI =
+      (0xF << 8) | (0x7 << 4) | (NI & 0xF) | (((NI >> 4) & 0x3) << 14);
+  constexpr int WAIT_0 = (0xF << 8) | (0x7 << 4);
+
+  // prologue: stage first slice into buf 0
+  int cur = 0;
+  stage(0, k_lo);
+  for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
+    const int k_next = k0 + BK;
+    if (k_next < k_hi) {
+      stage(cur ^ 1, k_next);
+      __builtin_amdgcn_s_waitcnt(WAIT_NI);  // buf[cur] landed; next in flight
+    } else {
+      __builtin_amdgcn_s_waitcnt(WAIT_0);
+    }
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_barrier();  // raw: no vmcnt drain (counted pipeline)
+    // ---- fragments + MFMA from buf[cur] ----
     const int wf = NT * 16 + wave * FEAT_PER_WAVE;  // W row base in tile
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const int kc = kk * 4 + kgrp;
       bf16x8_t b = *reinterpret_cast<bf16x8_t*>(
-          &tile[lds_row_idx(wf + m_local, kc)]);
+          &tile[cur][lds_row_idx(wf + m_local, kc)]);
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
         bf16x8_t a = *reinterpret_cast<bf16x8_t*>(
-            &tile[lds_row_idx(t * 16 + m_local, kc)]);
+            &tile[cur][lds_row_idx(t * 16 + m_local, kc)]);
         acc[t] = mfma16x16x32(a, b, acc[t]);
       }
     }
+    __builtin_amdgcn_s_barrier();  // readers done before buf[cur] is reused
+    cur ^= 1;
   }
 
   // ---- epilogue: C row = token, col = feature ----
