@@ -23,6 +23,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 KERNEL_SOURCES = [
     "rmsnorm.hip",
+    "layernorm.hip",
     "activation.hip",
     "rope.hip",
     "kv_cache.hip",

@@ -197,3 +197,40 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] =
         _C.skinny_gemm(out, x, weight)
         return out
     return F.linear(x, weight, bias)
+
+
+def layer_norm(x, weight, bias, eps: float):
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native("layer_norm").layer_norm(out, x, weight, bias, eps)
+        return out
+    return torch_ref.layer_norm(x, weight, bias, eps)
+
+
+def fused_add_layer_norm(x, residual, weight, bias, eps: float):
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native("fused_add_layer_norm").fused_add_layer_norm(
+            out, x, residual, weight, bias, eps
+        )
+        return out
+    return torch_ref.fused_add_layer_norm(x, residual, weight, bias, eps)
+
+
+def gelu(x):
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native("gelu").gelu(out, x)
+        return out
+    return torch_ref.gelu(x)
+
+
+def flash_attn_varlen(q, k, v, cu_seqlens, max_seqlen, scale, causal=True):
+    """Bidirectional-capable variant (encoder models)."""
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native("flash_prefill").flash_prefill_varlen(
+            out, q, k, v, cu_seqlens, int(max_seqlen), scale, causal
+        )
+        return out
+    return torch_ref.flash_prefill_varlen(q, k, v, cu_seqlens, scale, causal=causal)

@@ -20,16 +20,19 @@ constexpr int QW = 16;        // q rows per wave
 constexpr int NWAVES = 4;     // waves per workgroup (64 q rows)
 
 // swizzled LDS index helpers (short units)
+template <int D>
 __device__ __forceinline__ int k_idx(int row, int col) {
-  // [32][128] shorts; XOR 8-short chunks with row&7
-  return row * 128 + (col ^ ((row & 7) << 3));
+  // [32][D] shorts; XOR 8-short chunks with row&7 (D=64: modulo row width)
+  constexpr int MASK = (D / 8) - 1;
+  const int chunk = (col >> 3) ^ ((row & 7) & MASK);
+  return row * D + (chunk << 3) + (col & 7);
 }
 __device__ __forceinline__ int v_idx(int d, int tok) {
   // transposed [128 dims][32 tokens] shorts; XOR 8-short chunks with d&3
   return d * KT + (tok ^ ((d & 3) << 3));
 }
 
-template <int D>  // D == 128
+template <int D, bool CAUSAL>  // D == 128
 __global__ __launch_bounds__(256) void flash_prefill_kernel(
     short* __restrict__ out,      // [T, Hq, D] bf16
     const short* __restrict__ q,  // [T, Hq, D]
@@ -76,8 +79,8 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 #pragma unroll
   for (int c = 0; c < D / 16; ++c) o_acc[c] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
-  // causal bound: last token needed by this workgroup
-  const int kv_limit = min(len, tile_base + NWAVES * QW);
+  // causal bound: last token needed by this workgroup (bidirectional: all)
+  const int kv_limit = CAUSAL ? min(len, tile_base + NWAVES * QW) : len;
   const int ntiles = (kv_limit + KT - 1) / KT;
 
   for (int kt = 0; kt < ntiles; ++kt) {
@@ -96,7 +99,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) val[j] = 0;
       }
-      *reinterpret_cast<short8_t*>(&k_tile[k_idx(r, c8)]) = val;
+      *reinterpret_cast<short8_t*>(&k_tile[k_idx<D>(r, c8)]) = val;
       short8_t vv;
       if (tok < len) {
         vv = *reinterpret_cast<const short8_t*>(
@@ -111,7 +114,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
     __syncthreads();
     if (!active) continue;
     // causal skip: this wave's rows are all below the tile's first token
-    if (t0 > wq0 + QW - 1) continue;
+    if (CAUSAL && t0 > wq0 + QW - 1) continue;
 
     // ---- QK^T: S[16 q][32 t] as 2 sub-tiles of 16 tokens ----
     f32x4_t s_frag[2] = {f32x4_t{0, 0, 0, 0}, f32x4_t{0, 0, 0, 0}};
@@ -122,7 +125,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
       for (int c = 0; c < D / 32; ++c) {
         const int col = c * 32 + ((lane >> 4) << 3);
         bf16x8_t bk =
-            *reinterpret_cast<bf16x8_t*>(&k_tile[k_idx(tok_row, col)]);
+            *reinterpret_cast<bf16x8_t*>(&k_tile[k_idx<D>(tok_row, col)]);
         s_frag[n] = mfma16x16x32(a_q[c], bk, s_frag[n]);
       }
     }
@@ -136,7 +139,8 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 #pragma unroll
       for (int n = 0; n < 2; ++n) {
         const int tok = t0 + n * 16 + MFMA_C_COL(lane);
-        const bool valid = (tok <= qrow) && (qrow < len) && (tok < len);
+        const bool valid =
+            (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len);
         sv[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
       }
       float rowmax = group_reduce_max<16>(fmaxf(sv[0], sv[1]));
@@ -206,17 +210,26 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 extern "C" hipError_t ks_flash_prefill_varlen(
     void* out, const void* q, const void* k, const void* v,
     const void* cu_seqlens, int num_seqs, int max_seqlen, int Hq, int Hkv,
-    int head_dim, float scale, long sq, long sk, long sv,
+    int head_dim, float scale, long sq, long sk, long sv, int causal,
     hipStream_t stream) {
-  if (head_dim != 128) return hipErrorInvalidValue;
+  if (head_dim != 128 && head_dim != 64) return hipErrorInvalidValue;
   if (Hq % Hkv != 0) return hipErrorInvalidValue;
   const int max_tiles = (max_seqlen + NWAVES * QW - 1) / (NWAVES * QW);
   if (max_tiles == 0 || num_seqs == 0) return hipSuccess;
   dim3 grid(Hq, max_tiles, num_seqs);
-  hipLaunchKernelGGL((flash_prefill_kernel<128>), grid, dim3(256), 0, stream,
-                     (short*)out, (const short*)q, (const short*)k,
-                     (const short*)v, (const int*)cu_seqlens, Hq, Hkv, scale,
-                     sq, sk, sv);
+#define LAUNCH_FP(DD, CC)                                                    \
+  hipLaunchKernelGGL((flash_prefill_kernel<DD, CC>), grid, dim3(256), 0,     \
+                     stream, (short*)out, (const short*)q, (const short*)k,  \
+                     (const short*)v, (const int*)cu_seqlens, Hq, Hkv,       \
+                     scale, sq, sk, sv)
+  if (head_dim == 128) {
+    if (causal) LAUNCH_FP(128, true);
+    else LAUNCH_FP(128, false);
+  } else {
+    if (causal) LAUNCH_FP(64, true);
+    else LAUNCH_FP(64, false);
+  }
+#undef LAUNCH_FP
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }

@@ -24,8 +24,14 @@ hipError_t ks_paged_attention_decode(void*, const void*, const void*,
                                      long, int, void*, void*, hipStream_t);
 hipError_t ks_flash_prefill_varlen(void*, const void*, const void*,
                                    const void*, const void*, int, int, int,
-                                   int, int, float, long, long, long,
+                                   int, int, float, long, long, long, int,
                                    hipStream_t);
+hipError_t ks_layer_norm(void*, const void*, const void*, const void*, float,
+                         int, int, hipStream_t);
+hipError_t ks_fused_add_layer_norm(void*, const void*, const void*,
+                                   const void*, const void*, float, int, int,
+                                   hipStream_t);
+hipError_t ks_gelu(void*, const void*, long, hipStream_t);
 hipError_t ks_greedy_sample(void*, const void*, int, int, hipStream_t);
 hipError_t ks_gumbel_sample(void*, const void*, const void*, const void*,
                             const void*, int, int, hipStream_t);
@@ -176,7 +182,8 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
 
 void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
                           at::Tensor& v, at::Tensor& cu_seqlens,
-                          int64_t max_seqlen, double scale) {
+                          int64_t max_seqlen, double scale,
+                          bool causal = true) {
   CHECK_BF16_CONTIG(out);
   CHECK_BF16_ROWS(q);
   CHECK_BF16_ROWS(k);
@@ -191,8 +198,43 @@ void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
                                     num_seqs, (int)max_seqlen, Hq, Hkv, D,
                                     (float)scale, (long)q.stride(0),
                                     (long)k.stride(0), (long)v.stride(0),
-                                    current_stream()),
+                                    causal ? 1 : 0, current_stream()),
             "flash_prefill_varlen");
+}
+
+void layer_norm(at::Tensor& out, at::Tensor& input, at::Tensor& weight,
+                at::Tensor& bias, double eps) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(input);
+  int hidden = input.size(-1);
+  int rows = input.numel() / hidden;
+  check_hip(ks_layer_norm(out.data_ptr(), input.data_ptr(), weight.data_ptr(),
+                          bias.data_ptr(), (float)eps, rows, hidden,
+                          current_stream()),
+            "layer_norm");
+}
+
+void fused_add_layer_norm(at::Tensor& out, at::Tensor& input,
+                          at::Tensor& residual, at::Tensor& weight,
+                          at::Tensor& bias, double eps) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(input);
+  CHECK_BF16_CONTIG(residual);
+  int hidden = input.size(-1);
+  int rows = input.numel() / hidden;
+  check_hip(ks_fused_add_layer_norm(out.data_ptr(), input.data_ptr(),
+                                    residual.data_ptr(), weight.data_ptr(),
+                                    bias.data_ptr(), (float)eps, rows, hidden,
+                                    current_stream()),
+            "fused_add_layer_norm");
+}
+
+void gelu(at::Tensor& out, at::Tensor& input) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_CONTIG(input);
+  check_hip(ks_gelu(out.data_ptr(), input.data_ptr(), input.numel(),
+                    current_stream()),
+            "gelu");
 }
 
 void greedy_sample(at::Tensor& out, at::Tensor& logits) {
@@ -264,7 +306,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_decode", &paged_attention_decode,
         "paged decode attention (GQA, split-context)");
   m.def("flash_prefill_varlen", &flash_prefill_varlen,
-        "MFMA flash prefill (causal varlen GQA)");
+        "MFMA flash attention (varlen GQA, causal or bidirectional)",
+        pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k"),
+        pybind11::arg("v"), pybind11::arg("cu_seqlens"),
+        pybind11::arg("max_seqlen"), pybind11::arg("scale"),
+        pybind11::arg("causal") = true);
+  m.def("layer_norm", &layer_norm, "LayerNorm (bf16)");
+  m.def("fused_add_layer_norm", &fused_add_layer_norm,
+        "residual-add + LayerNorm");
+  m.def("gelu", &gelu, "erf GELU");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");
   m.def("gumbel_sample", &gumbel_sample, "Gumbel-max temperature sampling");
   m.def("skinny_gemm", &skinny_gemm, "decode GEMM (N<=256, MFMA streaming)");

@@ -229,3 +229,25 @@ def random_sample(
     probs = torch.softmax(sorted_logits, dim=-1)
     choice = torch.multinomial(probs, 1, generator=generator).squeeze(-1)
     return sorted_idx.gather(-1, choice.unsqueeze(-1)).squeeze(-1)
+
+
+def layer_norm(
+    x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor, eps: float
+) -> torch.Tensor:
+    return torch.nn.functional.layer_norm(
+        x.float(), (x.shape[-1],), weight.float(), bias.float(), eps
+    ).to(x.dtype)
+
+
+def fused_add_layer_norm(
+    x: torch.Tensor,
+    residual: torch.Tensor,
+    weight: torch.Tensor,
+    bias: torch.Tensor,
+    eps: float,
+) -> torch.Tensor:
+    return layer_norm((x.float() + residual.float()).to(x.dtype), weight, bias, eps)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    return torch.nn.functional.gelu(x.float()).to(x.dtype)

@@ -1,0 +1,113 @@
+"""HuggingFace-model server: backend selection between the native LLM engine
+(generative, Llama-family) and the native encoder (BERT-class).
+
+Reference parity: python/huggingfaceserver/__main__.py:60-323 (backend
+select: vllm-if-supported else HF; ours: native engine if decoder-only
+architecture, native encoder if BERT-class).
+Run: python -m kserve_amd.runtimes.huggingfaceserver --model_dir ... \
+     [--model_name m] [--task fill_mask] [--tensor-parallel-size N]
+"""
+
+from __future__ import annotations
+
+import json
+import os
+
+from kserve_amd.logging import configure_logging, logger
+
+GENERATIVE_ARCHITECTURES = (
+    "LlamaForCausalLM",
+    "MistralForCausalLM",
+    "Qwen2ForCausalLM",
+)
+ENCODER_SUFFIXES = (
+    "ForMaskedLM",
+    "ForSequenceClassification",
+    "ForTokenClassification",
+)
+
+
+def detect_backend(model_dir: str) -> str:
+    with open(os.path.join(model_dir, "config.json")) as f:
+        cfg = json.load(f)
+    archs = cfg.get("architectures") or []
+    for a in archs:
+        if a in GENERATIVE_ARCHITECTURES:
+            return "engine"
+        for sfx in ENCODER_SUFFIXES:
+            if a.endswith(sfx):
+                return "encoder"
+        if a.endswith("Model") and a.startswith(("Bert", "Roberta", "Distil")):
+            return "encoder"
+    model_type = cfg.get("model_type", "")
+    if model_type in ("llama", "mistral", "qwen2"):
+        return "engine"
+    return "encoder"
+
+
+def build_model(args):
+    backend = getattr(args, "backend", None) or detect_backend(args.model_dir)
+    logger.info("huggingfaceserver backend: %s", backend)
+    if backend == "engine":
+        import torch
+
+        from kserve_amd.engine.config import (
+            CacheConfig,
+            EngineConfig,
+            ModelConfig,
+            ParallelConfig,
+            SchedulerConfig,
+        )
+        from kserve_amd.parallel import comm
+        from kserve_amd.runtimes.llm_model import LLMModel
+
+        comm.init_distributed(tp_size=getattr(args, "tensor_parallel_size", 1))
+        from transformers import AutoTokenizer
+
+        tokenizer = AutoTokenizer.from_pretrained(args.model_dir)
+        cfg = EngineConfig(
+            model=ModelConfig.from_hf_config(
+                os.path.join(args.model_dir, "config.json")
+            ),
+            cache=CacheConfig(),
+            scheduler=SchedulerConfig(
+                max_num_seqs=getattr(args, "max_num_seqs", 256),
+                max_model_len=getattr(args, "max_model_len", 8192),
+            ),
+            parallel=ParallelConfig(
+                tensor_parallel_size=getattr(args, "tensor_parallel_size", 1)
+            ),
+            model_path=args.model_dir,
+            device="cuda" if torch.cuda.is_available() else "cpu",
+        )
+        return LLMModel(args.model_name, cfg, tokenizer=tokenizer)
+    from kserve_amd.runtimes.encoder_model import EncoderModel
+
+    model = EncoderModel(
+        args.model_name, args.model_dir, task=getattr(args, "task", None)
+    )
+    model.load()
+    return model
+
+
+def main(argv=None):
+    from kserve_amd.model_server import ModelServer, build_arg_parser
+
+    parser = build_arg_parser()
+    parser.add_argument("--backend", default=None, choices=["engine", "encoder"])
+    parser.add_argument("--task", default=None)
+    parser.add_argument("--tensor-parallel-size", dest="tensor_parallel_size", type=int, default=1)
+    parser.add_argument("--max_model_len", type=int, default=8192)
+    parser.add_argument("--max_num_seqs", type=int, default=256)
+    args = parser.parse_args(argv)
+    configure_logging()
+    model = build_model(args)
+    ModelServer(
+        http_port=args.http_port,
+        grpc_port=args.grpc_port,
+        enable_grpc=args.enable_grpc,
+    ).start([model])
+
+
+if __name__ == "__main__":
+    main()

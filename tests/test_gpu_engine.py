@@ -144,3 +144,36 @@ def test_preemption_on_gpu():
     assert len(out) == 6
     for o in out.values():
         assert len(o.output_token_ids) == 30
+
+
+def test_bert_fill_mask_gpu_matches_cpu():
+    """BERT encoder path on GPU bf16 (layer_norm/gelu/bidirectional flash
+    HIP kernels) vs CPU fp32 reference."""
+    import pytest
+
+    transformers = pytest.importorskip("transformers")
+    from kserve_amd.models.bert import BertConfig, BertForMaskedLM
+
+    cfg = transformers.BertConfig(
+        vocab_size=512,
+        hidden_size=128,
+        num_hidden_layers=2,
+        num_attention_heads=2,  # head_dim 64
+        intermediate_size=256,
+        max_position_embeddings=128,
+    )
+    torch.manual_seed(0)
+    hf = transformers.BertForMaskedLM(cfg).eval().float()
+    sd = dict(hf.state_dict())
+    ours_cpu = BertForMaskedLM(BertConfig.tiny(), dtype=torch.float32)
+    ours_cpu.load_hf_state_dict(sd)
+    ours_gpu = BertForMaskedLM(BertConfig.tiny(), dtype=torch.bfloat16, device="cuda")
+    ours_gpu.load_hf_state_dict(sd)
+    ours_gpu = ours_gpu.to("cuda")
+
+    ids = torch.randint(0, 512, (24,))
+    cu = torch.tensor([0, 10, 24], dtype=torch.int32)
+    ref = ours_cpu(ids, cu)
+    got = ours_gpu(ids.cuda(), cu.cuda())
+    agree = (got.float().cpu().argmax(-1) == ref.argmax(-1)).float().mean()
+    assert agree > 0.9, f"top-1 agreement {agree}"
