@@ -1,0 +1,237 @@
+"""Agent components: batcher, modelconfig watcher, payload logger."""
+
+import asyncio
+import json
+import os
+
+import httpx
+import pytest
+from fastapi.testclient import TestClient
+
+from kserve_amd.agent.batcher import Batcher, create_batcher_proxy_app
+from kserve_amd.agent.payload_logger import LogEntry, LogMode, PayloadLogger
+from kserve_amd.agent.watcher import ModelConfigWatcher, ModelSpec
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+class TestBatcher:
+    def test_flush_on_size(self):
+        async def main():
+            calls = []
+
+            async def predict(instances):
+                calls.append(list(instances))
+                return {"predictions": [i * 2 for i in instances]}
+
+            b = Batcher(predict, max_batch_size=4, max_latency_ms=60000)
+            results = await asyncio.gather(
+                b.predict([1, 2]), b.predict([3, 4])
+            )
+            assert len(calls) == 1  # one flush covered both
+            assert calls[0] == [1, 2, 3, 4]
+            assert results[0]["predictions"] == [2, 4]
+            assert results[1]["predictions"] == [6, 8]
+            assert results[0]["batchId"] == results[1]["batchId"]
+            assert results[0]["instanceCount"] == 2
+
+        run(main())
+
+    def test_flush_on_latency(self):
+        async def main():
+            async def predict(instances):
+                return {"predictions": [0 for _ in instances]}
+
+            b = Batcher(predict, max_batch_size=100, max_latency_ms=50)
+            t0 = asyncio.get_event_loop().time()
+            out = await b.predict([1])
+            dt = asyncio.get_event_loop().time() - t0
+            assert out["predictions"] == [0]
+            assert 0.03 < dt < 2.0
+
+        run(main())
+
+    def test_oversized_passthrough(self):
+        async def main():
+            calls = []
+
+            async def predict(instances):
+                calls.append(len(instances))
+                return {"predictions": [0] * len(instances)}
+
+            b = Batcher(predict, max_batch_size=2, max_latency_ms=60000)
+            out = await b.predict([1, 2, 3])
+            assert out["predictions"] == [0, 0, 0]
+            assert calls == [3]
+
+        run(main())
+
+    def test_prediction_count_mismatch(self):
+        async def main():
+            async def predict(instances):
+                return {"predictions": [0]}  # wrong count
+
+            b = Batcher(predict, max_batch_size=2, max_latency_ms=50)
+            with pytest.raises(RuntimeError):
+                await b.predict([1, 2])
+
+        run(main())
+
+    def test_proxy_app(self):
+        def backend(request: httpx.Request) -> httpx.Response:
+            body = json.loads(request.content)
+            return httpx.Response(
+                200, json={"predictions": [x + 1 for x in body["instances"]]}
+            )
+
+        app = create_batcher_proxy_app(
+            "http://backend",
+            "m",
+            max_batch_size=2,
+            max_latency_ms=30,
+            transport=httpx.MockTransport(backend),
+        )
+        with TestClient(app) as c:
+            r = c.post("/v1/models/m:predict", json={"instances": [5]})
+            assert r.status_code == 200
+            assert r.json()["predictions"] == [6]
+
+
+class TestWatcher:
+    def test_load_and_unload(self, tmp_path):
+        async def main():
+            cfg_dir = tmp_path / "configs"
+            cfg_dir.mkdir()
+            model_dir = tmp_path / "models"
+            model_dir.mkdir()
+            src = tmp_path / "src_model"
+            src.mkdir()
+            (src / "model.joblib").write_bytes(b"fake")
+
+            loaded, unloaded = [], []
+
+            async def on_load(name, local, spec):
+                loaded.append((name, local))
+
+            async def on_unload(name):
+                unloaded.append(name)
+
+            w = ModelConfigWatcher(
+                str(cfg_dir), str(model_dir), on_load, on_unload
+            )
+            cfg = [
+                {
+                    "modelName": "m1",
+                    "modelSpec": {"storageUri": str(src), "framework": "sklearn"},
+                }
+            ]
+            (cfg_dir / "modelconfig.json").write_text(json.dumps(cfg))
+            assert await w.sync_once()
+            assert loaded and loaded[0][0] == "m1"
+            assert os.path.exists(model_dir / "m1" / "model.joblib")
+
+            # remove the model -> unload
+            (cfg_dir / "modelconfig.json").write_text("[]")
+            os.utime(cfg_dir / "modelconfig.json", (1e9, 2e9))
+            assert await w.sync_once()
+            assert unloaded == ["m1"]
+            assert not os.path.exists(model_dir / "m1")
+
+        run(main())
+
+    def test_changed_uri_reloads(self, tmp_path):
+        async def main():
+            cfg_dir = tmp_path / "c"
+            cfg_dir.mkdir()
+            mdir = tmp_path / "m"
+            mdir.mkdir()
+            s1 = tmp_path / "s1"
+            s1.mkdir()
+            (s1 / "f").write_bytes(b"1")
+            s2 = tmp_path / "s2"
+            s2.mkdir()
+            (s2 / "f").write_bytes(b"2")
+            events = []
+
+            async def on_load(name, local, spec):
+                events.append(("load", spec.storage_uri))
+
+            async def on_unload(name):
+                events.append(("unload", name))
+
+            w = ModelConfigWatcher(str(cfg_dir), str(mdir), on_load, on_unload)
+            p = cfg_dir / "modelconfig.json"
+            p.write_text(json.dumps([{"modelName": "x", "modelSpec": {"storageUri": str(s1)}}]))
+            await w.sync_once()
+            p.write_text(json.dumps([{"modelName": "x", "modelSpec": {"storageUri": str(s2)}}]))
+            os.utime(p, (1e9, 2e9))
+            await w.sync_once()
+            assert events == [
+                ("load", str(s1)),
+                ("unload", "x"),
+                ("load", str(s2)),
+            ]
+
+        run(main())
+
+
+class TestPayloadLogger:
+    def test_cloudevent_http(self):
+        async def main():
+            received = []
+
+            def sink(request: httpx.Request) -> httpx.Response:
+                received.append((dict(request.headers), request.content))
+                return httpx.Response(200)
+
+            pl = PayloadLogger(
+                url="http://sink/",
+                transport=httpx.MockTransport(sink),
+            )
+            await pl.start()
+            await pl.log(
+                LogEntry(
+                    request_id="r1",
+                    event_type="org.kubeflow.serving.inference.request",
+                    model_name="m",
+                    payload=b'{"instances": [[1]]}',
+                )
+            )
+            await pl.stop()
+            assert len(received) == 1
+            hdrs, body = received[0]
+            assert hdrs["ce-specversion"] == "1.0"
+            assert hdrs["ce-type"].endswith("request")
+            assert hdrs["ce-requestid"] == "r1"
+            assert body == b'{"instances": [[1]]}'
+
+        run(main())
+
+    def test_file_store_and_csv(self, tmp_path):
+        async def main():
+            pl = PayloadLogger(store_path=str(tmp_path), marshaller="csv")
+            await pl.start()
+            await pl.log(
+                LogEntry(
+                    request_id="r2",
+                    event_type="org.kubeflow.serving.inference.response",
+                    model_name="m",
+                    payload=b'{"predictions": [[1, 2], [3, 4]]}',
+                )
+            )
+            await pl.stop()
+            path = tmp_path / "r2-response.csv"
+            assert path.exists()
+            assert path.read_text().strip().splitlines() == ["1,2", "3,4"]
+
+        run(main())
+
+    def test_mode_filter(self):
+        async def main():
+            pl = PayloadLogger(store_path="/tmp/unused", mode=LogMode.response)
+            assert not pl.should_log("org.kubeflow.serving.inference.request")
+            assert pl.should_log("org.kubeflow.serving.inference.response")
+
+        run(main())
