@@ -15,6 +15,9 @@ import time
 from dataclasses import dataclass, field
 from typing import Any, Awaitable, Callable, Dict, List, Optional
 
+import httpx
+from fastapi import FastAPI, Request, Response
+
 from kserve_amd.constants import DEFAULT_MAX_BATCH_SIZE, DEFAULT_MAX_LATENCY_MS
 from kserve_amd.logging import logger
 
@@ -118,9 +121,6 @@ def create_batcher_proxy_app(
 ):
     """Agent-style reverse proxy: batches ``POST /v1/models/{m}:predict``,
     passes everything else through (reference agent chain, main.go:429-449)."""
-    import httpx
-    from fastapi import FastAPI, Request, Response
-
     app = FastAPI()
     client = httpx.AsyncClient(base_url=backend_url, transport=transport, timeout=60)
 

@@ -71,9 +71,9 @@ class TestBatcher:
     def test_prediction_count_mismatch(self):
         async def main():
             async def predict(instances):
-                return {"predictions": [0]}  # wrong count
+                return {"predictions": [0, 0, 0]}  # wrong count
 
-            b = Batcher(predict, max_batch_size=2, max_latency_ms=50)
+            b = Batcher(predict, max_batch_size=4, max_latency_ms=50)
             with pytest.raises(RuntimeError):
                 await b.predict([1, 2])
 
