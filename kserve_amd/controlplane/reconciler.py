@@ -1,0 +1,273 @@
+"""InferenceService reconciler: render Kubernetes manifests (Standard /
+RawDeployment mode).
+
+Reference parity: pkg/controller/v1beta1/inferenceservice —
+Predictor.Reconcile / buildPodSpec (components/predictor.go:184-422),
+RawKubeReconciler (reconcilers/raw/raw_kube_reconciler.go:48-150),
+deployment/service reconcilers, HPA mapping
+(reconcilers/hpa/hpa_reconciler.go), modelconfig
+(pkg/modelconfig/configmap.go:37-161) — as pure functions emitting k8s
+manifest dicts, unit-testable without a cluster (the reference's envtest
+asserts translate to dict asserts).
+"""
+
+from __future__ import annotations
+
+import copy
+import json
+from typing import Dict, List, Optional
+
+from kserve_amd.controlplane.servingruntime import ServingRuntime, select_runtime
+from kserve_amd.controlplane.v1beta1 import (
+    InferenceService,
+    PredictorSpec,
+    default_inference_service,
+    validate_inference_service,
+)
+from kserve_amd.controlplane.webhook import mutate_pod
+
+# internal annotations carried from controller to webhook
+# (reference constants.go:158-171)
+ANN_STORAGE_URI = "internal.serving.kserve.io/storage-initializer-sourceuri"
+ANN_LOGGER = "internal.serving.kserve.io/logger"
+ANN_LOGGER_URL = "internal.serving.kserve.io/logger-sink-url"
+ANN_LOGGER_MODE = "internal.serving.kserve.io/logger-mode"
+ANN_BATCHER = "internal.serving.kserve.io/batcher"
+ANN_BATCHER_MAX_SIZE = "internal.serving.kserve.io/batcher-max-batchsize"
+ANN_BATCHER_MAX_LATENCY = "internal.serving.kserve.io/batcher-max-latency"
+ANN_AGENT = "internal.serving.kserve.io/agent"
+
+
+def component_annotations(spec: PredictorSpec) -> Dict[str, str]:
+    """reference components/component.go:40-92."""
+    ann: Dict[str, str] = {}
+    if spec.storage_uri:
+        ann[ANN_STORAGE_URI] = spec.storage_uri
+    if spec.logger is not None:
+        ann[ANN_LOGGER] = "true"
+        if spec.logger.url:
+            ann[ANN_LOGGER_URL] = spec.logger.url
+        ann[ANN_LOGGER_MODE] = spec.logger.mode
+        ann[ANN_AGENT] = "true"
+    if spec.batcher is not None:
+        ann[ANN_BATCHER] = "true"
+        ann[ANN_BATCHER_MAX_SIZE] = str(spec.batcher.max_batch_size)
+        ann[ANN_BATCHER_MAX_LATENCY] = str(spec.batcher.max_latency_ms)
+        ann[ANN_AGENT] = "true"
+    return ann
+
+
+def _render_placeholders(container: Dict, isvc: InferenceService) -> Dict:
+    """reference isvc utils ReplacePlaceholders ({{.Name}} templating)."""
+    raw = json.dumps(container)
+    raw = raw.replace("{{.Name}}", isvc.name).replace(
+        "{{.Namespace}}", isvc.namespace
+    )
+    return json.loads(raw)
+
+
+def predictor_service_name(isvc: InferenceService) -> str:
+    return f"{isvc.name}-predictor"
+
+
+def render_predictor_pod_spec(
+    isvc: InferenceService,
+    runtimes: List[ServingRuntime],
+    storage_init_image: str = "kserve-amd/storage-initializer:latest",
+    agent_image: str = "kserve-amd/agent:latest",
+) -> Dict:
+    """buildPodSpec + pod webhook mutation chain."""
+    p = isvc.spec.predictor
+    if p.containers:
+        container = copy.deepcopy(p.containers[0])
+        container.setdefault("name", "kserve-container")
+    else:
+        rt = select_runtime(
+            p.model.model_format.name,
+            p.model.protocol_version,
+            [],
+            runtimes,
+            explicit_runtime=p.model.runtime,
+            multinode=p.worker is not None,
+        )
+        container = _render_placeholders(copy.deepcopy(rt.container), isvc)
+        if p.model.args:
+            container.setdefault("args", []).extend(p.model.args)
+        if p.model.resources:
+            container["resources"] = p.model.resources
+        if p.model.image:
+            container["image"] = p.model.image
+    pod = {
+        "metadata": {
+            "labels": {
+                "serving.kserve.io/inferenceservice": isvc.name,
+                "component": "predictor",
+                **isvc.labels,
+            },
+            "annotations": {
+                **isvc.annotations,
+                **component_annotations(p),
+            },
+        },
+        "spec": {"containers": [container]},
+    }
+    return mutate_pod(
+        pod, storage_init_image=storage_init_image, agent_image=agent_image
+    )
+
+
+def render_deployment(
+    isvc: InferenceService,
+    runtimes: List[ServingRuntime],
+    canary: bool = False,
+) -> Dict:
+    """Deployment manifest (raw mode; reference deployment_reconciler.go)."""
+    p = isvc.spec.predictor
+    name = predictor_service_name(isvc) + ("-canary" if canary else "")
+    pod = render_predictor_pod_spec(isvc, runtimes)
+    pod["metadata"]["labels"]["app"] = name
+    return {
+        "apiVersion": "apps/v1",
+        "kind": "Deployment",
+        "metadata": {
+            "name": name,
+            "namespace": isvc.namespace,
+            "labels": pod["metadata"]["labels"],
+        },
+        "spec": {
+            "replicas": max(p.min_replicas, 1),
+            "selector": {"matchLabels": {"app": name}},
+            "template": pod,
+        },
+    }
+
+
+def render_service(isvc: InferenceService) -> Dict:
+    name = predictor_service_name(isvc)
+    return {
+        "apiVersion": "v1",
+        "kind": "Service",
+        "metadata": {"name": name, "namespace": isvc.namespace},
+        "spec": {
+            "selector": {"app": name},
+            "ports": [
+                {"name": "http", "port": 80, "targetPort": 8080},
+                {"name": "grpc", "port": 81, "targetPort": 8081},
+            ],
+        },
+    }
+
+
+def render_hpa(isvc: InferenceService) -> Optional[Dict]:
+    """HPA from ScaleMetric (reference hpa_reconciler.go)."""
+    p = isvc.spec.predictor
+    if not p.max_replicas or p.max_replicas <= p.min_replicas:
+        return None
+    metric = p.scale_metric or "cpu"
+    target = p.scale_target or 80
+    if metric in ("cpu", "memory"):
+        metrics = [
+            {
+                "type": "Resource",
+                "resource": {
+                    "name": metric,
+                    "target": {
+                        "type": "Utilization",
+                        "averageUtilization": target,
+                    },
+                },
+            }
+        ]
+    else:
+        metrics = [
+            {
+                "type": "Pods",
+                "pods": {
+                    "metric": {"name": metric},
+                    "target": {"type": "AverageValue", "averageValue": str(target)},
+                },
+            }
+        ]
+    return {
+        "apiVersion": "autoscaling/v2",
+        "kind": "HorizontalPodAutoscaler",
+        "metadata": {
+            "name": predictor_service_name(isvc),
+            "namespace": isvc.namespace,
+        },
+        "spec": {
+            "scaleTargetRef": {
+                "apiVersion": "apps/v1",
+                "kind": "Deployment",
+                "name": predictor_service_name(isvc),
+            },
+            "minReplicas": p.min_replicas,
+            "maxReplicas": p.max_replicas,
+            "metrics": metrics,
+        },
+    }
+
+
+def render_http_route(isvc: InferenceService, ingress_domain: str = "example.com") -> Dict:
+    """Gateway API HTTPRoute (reference httproute_reconciler.go)."""
+    host = f"{isvc.name}.{isvc.namespace}.{ingress_domain}"
+    return {
+        "apiVersion": "gateway.networking.k8s.io/v1",
+        "kind": "HTTPRoute",
+        "metadata": {"name": isvc.name, "namespace": isvc.namespace},
+        "spec": {
+            "hostnames": [host],
+            "rules": [
+                {
+                    "matches": [{"path": {"type": "PathPrefix", "value": "/"}}],
+                    "backendRefs": [
+                        {
+                            "name": predictor_service_name(isvc),
+                            "port": 80,
+                        }
+                    ],
+                }
+            ],
+        },
+    }
+
+
+def reconcile(
+    isvc: InferenceService, runtimes: List[ServingRuntime]
+) -> Dict[str, object]:
+    """Full reconcile: default -> validate -> manifests (+ canary pair when
+    canaryTrafficPercent set; reference reconcileCanaryDeployments
+    predictor.go:918)."""
+    default_inference_service(isvc)
+    validate_inference_service(isvc)
+    out: Dict[str, object] = {
+        "deployment": render_deployment(isvc, runtimes),
+        "service": render_service(isvc),
+        "httproute": render_http_route(isvc),
+    }
+    hpa = render_hpa(isvc)
+    if hpa:
+        out["hpa"] = hpa
+    pct = isvc.spec.predictor.canary_traffic_percent
+    if pct is not None and 0 < pct < 100:
+        out["canary_deployment"] = render_deployment(isvc, runtimes, canary=True)
+        out["traffic_split"] = {"stable": 100 - pct, "canary": pct}
+    return out
+
+
+def build_model_config(trained_models: List[Dict]) -> str:
+    """modelconfig ConfigMap payload for the agent (reference
+    pkg/modelconfig/configmap.go:37-161; MemoryStrategy shard 0)."""
+    entries = []
+    for tm in trained_models:
+        entries.append(
+            {
+                "modelName": tm["name"],
+                "modelSpec": {
+                    "storageUri": tm["storageUri"],
+                    "framework": tm.get("framework", ""),
+                    "memory": tm.get("memory", ""),
+                },
+            }
+        )
+    return json.dumps(entries, sort_keys=True)

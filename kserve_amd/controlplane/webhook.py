@@ -1,0 +1,141 @@
+"""Pod mutating webhook logic: sidecar/init-container injection.
+
+Reference parity: pkg/webhook/admission/pod — mutator chain
+(mutator.go:131-143): InjectStorageInitializer
+(storage_initializer_injector.go:729), InjectAgent (agent_injector.go:177),
+InjectBatcher (batcher_injector.go:79), InjectModelcar (:201) — as a pure
+function on pod manifests driven by the internal annotations.
+"""
+
+from __future__ import annotations
+
+import copy
+from typing import Dict
+
+from kserve_amd.constants import (
+    AGENT_PORT,
+    MODEL_MOUNT_PATH,
+)
+
+ANN_STORAGE_URI = "internal.serving.kserve.io/storage-initializer-sourceuri"
+ANN_LOGGER = "internal.serving.kserve.io/logger"
+ANN_LOGGER_URL = "internal.serving.kserve.io/logger-sink-url"
+ANN_LOGGER_MODE = "internal.serving.kserve.io/logger-mode"
+ANN_BATCHER = "internal.serving.kserve.io/batcher"
+ANN_BATCHER_MAX_SIZE = "internal.serving.kserve.io/batcher-max-batchsize"
+ANN_BATCHER_MAX_LATENCY = "internal.serving.kserve.io/batcher-max-latency"
+ANN_AGENT = "internal.serving.kserve.io/agent"
+
+
+def inject_storage_initializer(pod: Dict, image: str) -> Dict:
+    ann = pod["metadata"].get("annotations", {})
+    uri = ann.get(ANN_STORAGE_URI)
+    if not uri:
+        return pod
+    if uri.startswith(("oci://", "oci+native://")):
+        return inject_modelcar(pod, uri)
+    spec = pod["spec"]
+    volumes = spec.setdefault("volumes", [])
+    if not any(v["name"] == "kserve-provision-location" for v in volumes):
+        volumes.append({"name": "kserve-provision-location", "emptyDir": {}})
+    init = {
+        "name": "storage-initializer",
+        "image": image,
+        "args": [uri, MODEL_MOUNT_PATH],
+        "volumeMounts": [
+            {"name": "kserve-provision-location", "mountPath": MODEL_MOUNT_PATH}
+        ],
+    }
+    # pvc fast-path: mount the PVC directly (reference :44)
+    if uri.startswith("pvc://"):
+        pvc_name = uri[len("pvc://"):].split("/", 1)[0]
+        volumes.append(
+            {
+                "name": "kserve-pvc-source",
+                "persistentVolumeClaim": {"claimName": pvc_name},
+            }
+        )
+        init["volumeMounts"].append(
+            {"name": "kserve-pvc-source", "mountPath": "/mnt/pvc", "readOnly": True}
+        )
+    spec.setdefault("initContainers", []).insert(0, init)
+    for c in spec["containers"]:
+        if c.get("name") == "kserve-container":
+            c.setdefault("volumeMounts", []).append(
+                {
+                    "name": "kserve-provision-location",
+                    "mountPath": MODEL_MOUNT_PATH,
+                    "readOnly": True,
+                }
+            )
+    return pod
+
+
+def inject_modelcar(pod: Dict, uri: str) -> Dict:
+    """OCI model image as a sidecar sharing /mnt/models (reference
+    storage_initializer_injector.go:201-256)."""
+    image = uri.split("://", 1)[1]
+    spec = pod["spec"]
+    spec.setdefault("volumes", []).append(
+        {"name": "kserve-provision-location", "emptyDir": {}}
+    )
+    spec["containers"].append(
+        {
+            "name": "modelcar",
+            "image": image,
+            "command": ["sh", "-c", f"ln -sf /models/* {MODEL_MOUNT_PATH}/ && sleep infinity"],
+            "volumeMounts": [
+                {"name": "kserve-provision-location", "mountPath": MODEL_MOUNT_PATH}
+            ],
+        }
+    )
+    for c in spec["containers"]:
+        if c.get("name") == "kserve-container":
+            c.setdefault("volumeMounts", []).append(
+                {
+                    "name": "kserve-provision-location",
+                    "mountPath": MODEL_MOUNT_PATH,
+                    "readOnly": True,
+                }
+            )
+    return pod
+
+
+def inject_agent(pod: Dict, image: str) -> Dict:
+    """Agent sidecar proxy on :9081 for logger/batcher roles (reference
+    agent_injector.go:177; flag surface cmd/agent/main.go:55-99)."""
+    ann = pod["metadata"].get("annotations", {})
+    if ann.get(ANN_AGENT) != "true":
+        return pod
+    args = ["--port", str(AGENT_PORT), "--component-port", "8080"]
+    if ann.get(ANN_LOGGER) == "true":
+        args += ["--log-mode", ann.get(ANN_LOGGER_MODE, "all")]
+        if ann.get(ANN_LOGGER_URL):
+            args += ["--log-url", ann[ANN_LOGGER_URL]]
+    if ann.get(ANN_BATCHER) == "true":
+        args += [
+            "--enable-batcher",
+            "--max-batchsize", ann.get(ANN_BATCHER_MAX_SIZE, "32"),
+            "--max-latency", ann.get(ANN_BATCHER_MAX_LATENCY, "5000"),
+        ]
+    pod["spec"]["containers"].append(
+        {
+            "name": "agent",
+            "image": image,
+            "args": args,
+            "ports": [{"containerPort": AGENT_PORT, "name": "agent-port"}],
+        }
+    )
+    return pod
+
+
+def mutate_pod(
+    pod: Dict,
+    storage_init_image: str = "kserve-amd/storage-initializer:latest",
+    agent_image: str = "kserve-amd/agent:latest",
+) -> Dict:
+    """Ordered mutator chain (reference mutator.go:131-143)."""
+    pod = copy.deepcopy(pod)
+    pod = inject_storage_initializer(pod, storage_init_image)
+    pod = inject_agent(pod, agent_image)
+    return pod

@@ -1,0 +1,227 @@
+"""Control-plane rendering tests (the reference's envtest/table-driven
+webhook tests translated to manifest-dict asserts)."""
+
+import json
+
+import pytest
+
+from kserve_amd.controlplane.reconciler import (
+    build_model_config,
+    reconcile,
+    render_deployment,
+    render_hpa,
+)
+from kserve_amd.controlplane.servingruntime import (
+    ServingRuntime,
+    SupportedModelFormat,
+    default_cluster_runtimes,
+    select_runtime,
+)
+from kserve_amd.controlplane.v1beta1 import (
+    BatcherSpec,
+    FrameworkSpec,
+    InferenceService,
+    InferenceServiceSpec,
+    LoggerSpec,
+    ModelFormat,
+    PredictorModelSpec,
+    PredictorSpec,
+    ValidationError,
+    default_inference_service,
+    validate_inference_service,
+)
+from kserve_amd.controlplane.webhook import mutate_pod
+
+
+def make_isvc(**pred_kw):
+    return InferenceService(
+        name="iris",
+        namespace="ns1",
+        spec=InferenceServiceSpec(
+            predictor=PredictorSpec(
+                sklearn=FrameworkSpec(storage_uri="s3://bucket/iris"),
+                **pred_kw,
+            )
+        ),
+    )
+
+
+class TestValidation:
+    def test_valid(self):
+        isvc = make_isvc()
+        default_inference_service(isvc)
+        validate_inference_service(isvc)
+
+    def test_bad_name(self):
+        isvc = make_isvc()
+        isvc.name = "Iris_Bad"
+        with pytest.raises(ValidationError):
+            validate_inference_service(isvc)
+
+    def test_two_implementations(self):
+        isvc = make_isvc()
+        isvc.spec.predictor.xgboost = FrameworkSpec(storage_uri="s3://b/x")
+        with pytest.raises(ValidationError):
+            validate_inference_service(isvc)
+
+    def test_bad_scheme(self):
+        isvc = make_isvc()
+        isvc.spec.predictor.sklearn.storage_uri = "ftp://bucket/m"
+        default_inference_service(isvc)
+        with pytest.raises(ValidationError):
+            validate_inference_service(isvc)
+
+    def test_canary_bounds(self):
+        isvc = make_isvc(canary_traffic_percent=150)
+        default_inference_service(isvc)
+        with pytest.raises(ValidationError):
+            validate_inference_service(isvc)
+
+
+class TestRuntimeSelection:
+    def test_auto_select_by_format(self):
+        rts = default_cluster_runtimes()
+        rt = select_runtime("sklearn", "v1", [], rts)
+        assert rt.name == "kserve-amd-sklearnserver"
+
+    def test_priority_ordering(self):
+        low = ServingRuntime(
+            "low",
+            [SupportedModelFormat("sklearn", auto_select=True, priority=1)],
+            {"name": "kserve-container", "image": "low"},
+        )
+        high = ServingRuntime(
+            "high",
+            [SupportedModelFormat("sklearn", auto_select=True, priority=2)],
+            {"name": "kserve-container", "image": "high"},
+        )
+        assert select_runtime("sklearn", "v1", [], [low, high]).name == "high"
+
+    def test_namespace_before_cluster(self):
+        ns = ServingRuntime(
+            "ns-rt",
+            [SupportedModelFormat("sklearn", auto_select=True)],
+            {"name": "kserve-container", "image": "ns"},
+            cluster_scoped=False,
+        )
+        rt = select_runtime("sklearn", "v1", [ns], default_cluster_runtimes())
+        assert rt.name == "ns-rt"
+
+    def test_no_match(self):
+        with pytest.raises(LookupError):
+            select_runtime("unknownfmt", "v1", [], default_cluster_runtimes())
+
+    def test_explicit_runtime(self):
+        rts = default_cluster_runtimes()
+        rt = select_runtime("sklearn", "v1", [], rts, explicit_runtime="kserve-amd-xgbserver")
+        assert rt.name == "kserve-amd-xgbserver"
+
+
+class TestReconcile:
+    def test_full_reconcile(self):
+        isvc = make_isvc(min_replicas=2, max_replicas=5, scale_metric="cpu")
+        out = reconcile(isvc, default_cluster_runtimes())
+        dep = out["deployment"]
+        assert dep["metadata"]["name"] == "iris-predictor"
+        assert dep["spec"]["replicas"] == 2
+        tmpl = dep["spec"]["template"]
+        names = [c["name"] for c in tmpl["spec"]["containers"]]
+        assert "kserve-container" in names
+        # storage-initializer injected from annotation
+        inits = tmpl["spec"].get("initContainers", [])
+        assert inits and inits[0]["name"] == "storage-initializer"
+        assert inits[0]["args"] == ["s3://bucket/iris", "/mnt/models"]
+        # placeholder templating
+        kc = [c for c in tmpl["spec"]["containers"] if c["name"] == "kserve-container"][0]
+        assert "--model_name=iris" in kc["args"]
+        # service + route
+        assert out["service"]["spec"]["ports"][0]["targetPort"] == 8080
+        assert out["httproute"]["spec"]["hostnames"] == ["iris.ns1.example.com"]
+        # hpa
+        assert out["hpa"]["spec"]["maxReplicas"] == 5
+
+    def test_canary_pair(self):
+        isvc = make_isvc(canary_traffic_percent=20)
+        out = reconcile(isvc, default_cluster_runtimes())
+        assert out["traffic_split"] == {"stable": 80, "canary": 20}
+        assert out["canary_deployment"]["metadata"]["name"] == "iris-predictor-canary"
+
+    def test_custom_container_predictor(self):
+        isvc = InferenceService(
+            name="custom",
+            spec=InferenceServiceSpec(
+                predictor=PredictorSpec(
+                    containers=[{"image": "me/mymodel:1", "name": "kserve-container"}]
+                )
+            ),
+        )
+        out = reconcile(isvc, [])
+        kc = out["deployment"]["spec"]["template"]["spec"]["containers"][0]
+        assert kc["image"] == "me/mymodel:1"
+
+
+class TestWebhook:
+    def _pod(self, ann):
+        return {
+            "metadata": {"annotations": ann, "labels": {}},
+            "spec": {"containers": [{"name": "kserve-container", "image": "x"}]},
+        }
+
+    def test_agent_injection_with_logger_and_batcher(self):
+        pod = self._pod(
+            {
+                "internal.serving.kserve.io/agent": "true",
+                "internal.serving.kserve.io/logger": "true",
+                "internal.serving.kserve.io/logger-sink-url": "http://sink",
+                "internal.serving.kserve.io/batcher": "true",
+                "internal.serving.kserve.io/batcher-max-batchsize": "16",
+            }
+        )
+        out = mutate_pod(pod)
+        agent = [c for c in out["spec"]["containers"] if c["name"] == "agent"]
+        assert agent
+        args = agent[0]["args"]
+        assert "--log-url" in args and "http://sink" in args
+        assert "--enable-batcher" in args
+        assert "16" in args
+
+    def test_pvc_fast_path(self):
+        pod = self._pod(
+            {"internal.serving.kserve.io/storage-initializer-sourceuri": "pvc://my-pvc/models/a"}
+        )
+        out = mutate_pod(pod)
+        vols = {v["name"]: v for v in out["spec"]["volumes"]}
+        assert vols["kserve-pvc-source"]["persistentVolumeClaim"]["claimName"] == "my-pvc"
+
+    def test_modelcar(self):
+        pod = self._pod(
+            {"internal.serving.kserve.io/storage-initializer-sourceuri": "oci://reg/model:1"}
+        )
+        out = mutate_pod(pod)
+        names = [c["name"] for c in out["spec"]["containers"]]
+        assert "modelcar" in names
+        assert not out["spec"].get("initContainers")
+
+    def test_no_annotations_no_injection(self):
+        pod = self._pod({})
+        out = mutate_pod(pod)
+        assert len(out["spec"]["containers"]) == 1
+        assert not out["spec"].get("initContainers")
+
+
+class TestModelConfig:
+    def test_build(self):
+        payload = build_model_config(
+            [
+                {"name": "m1", "storageUri": "s3://b/m1", "framework": "sklearn"},
+                {"name": "m2", "storageUri": "s3://b/m2"},
+            ]
+        )
+        entries = json.loads(payload)
+        assert entries[0]["modelName"] == "m1"
+        assert entries[0]["modelSpec"]["storageUri"] == "s3://b/m1"
+        # round-trips through the agent watcher parser
+        from kserve_amd.agent.watcher import ModelConfigWatcher
+
+        parsed = ModelConfigWatcher.parse_config(payload)
+        assert set(parsed) == {"m1", "m2"}
