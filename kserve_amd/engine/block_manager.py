@@ -21,7 +21,13 @@ def cdiv(a: int, b: int) -> int:
 
 
 class BlockManager:
-    def __init__(self, num_blocks: int, block_size: int, watermark: float = 0.01):
+    def __init__(
+        self,
+        num_blocks: int,
+        block_size: int,
+        watermark: float = 0.01,
+        num_cpu_blocks: int = 0,
+    ):
         self.num_blocks = num_blocks
         self.block_size = block_size
         # watermark: headroom kept free so decodes can always append
@@ -30,6 +36,10 @@ class BlockManager:
         # (padded decode lanes write their KV there); never handed out.
         self._free: List[int] = list(range(1, num_blocks))
         self._tables: Dict[str, List[int]] = {}
+        # host-DRAM offload tier (pinned; hipMemcpyAsync side stream)
+        self.num_cpu_blocks = num_cpu_blocks
+        self._cpu_free: List[int] = list(range(num_cpu_blocks))
+        self._cpu_tables: Dict[str, List[int]] = {}
 
     # -- capacity ----------------------------------------------------------
     @property
@@ -99,3 +109,40 @@ class BlockManager:
             b = table[pos // self.block_size]
             out.append(b * self.block_size + pos % self.block_size)
         return out
+
+
+    # -- host-DRAM offload tier (north-star KV offload) ------
+    def can_swap_out(self, request: Request) -> bool:
+        table = self._tables.get(request.request_id)
+        return bool(table) and len(self._cpu_free) >= len(table)
+
+    def swap_out(self, request: Request) -> List[tuple]:
+        """Move the request's blocks to the CPU tier. Returns (gpu, cpu)
+        pairs; caller must COPY gpu->cpu before any kernel reuses the gpu
+        blocks (the runner orders this via a side-stream event)."""
+        table = self._tables.pop(request.request_id)
+        cpu_blocks = [self._cpu_free.pop() for _ in table]
+        self._cpu_tables[request.request_id] = cpu_blocks
+        pairs = list(zip(table, cpu_blocks))
+        self._free.extend(reversed(table))
+        request.block_table = []
+        return pairs
+
+    def can_swap_in(self, request: Request) -> bool:
+        cpu_table = self._cpu_tables.get(request.request_id)
+        return bool(cpu_table) and len(self._free) - len(cpu_table) >= self.watermark_blocks
+
+    def swap_in(self, request: Request) -> List[tuple]:
+        """Restore CPU-tier blocks to GPU. Returns (cpu, gpu) pairs."""
+        cpu_table = self._cpu_tables.pop(request.request_id)
+        gpu_blocks = [self._free.pop() for _ in cpu_table]
+        self._tables[request.request_id] = gpu_blocks
+        request.block_table = gpu_blocks
+        pairs = list(zip(cpu_table, gpu_blocks))
+        self._cpu_free.extend(reversed(cpu_table))
+        return pairs
+
+    def free_cpu(self, request: Request) -> None:
+        cpu_table = self._cpu_tables.pop(request.request_id, None)
+        if cpu_table:
+            self._cpu_free.extend(reversed(cpu_table))

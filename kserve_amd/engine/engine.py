@@ -109,7 +109,10 @@ class LLMEngine:
         self.model = model
         self.runner = ModelRunner(config, model, str(self.device))
         num_blocks = self.runner.profile_and_allocate_kv()
-        self.scheduler = Scheduler(config.scheduler, config.cache, num_blocks)
+        self.scheduler = Scheduler(
+            config.scheduler, config.cache, num_blocks,
+            num_cpu_blocks=self.runner.num_cpu_blocks,
+        )
         self.sampler = Sampler(self.device, config.seed)
         self.eos_token_id = config.eos_token_id
         if tokenizer is not None and getattr(tokenizer, "eos_token_id", None) is not None:
@@ -171,6 +174,11 @@ class LLMEngine:
     # -- step -----------------------------------------------------------------
     def step(self) -> List[RequestOutput]:
         batch = self.scheduler.schedule()
+        # KV tier copies ordered before this batch's kernels
+        if batch.swap_out:
+            self.runner.swap_blocks(batch.swap_out, to_gpu=False)
+        if batch.swap_in:
+            self.runner.swap_blocks(batch.swap_in, to_gpu=True)
         if not batch:
             return []
         if batch.is_prefill:

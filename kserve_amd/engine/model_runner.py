@@ -48,6 +48,12 @@ class ModelRunner:
         self._pin: Optional[Dict[str, torch.Tensor]] = None
         self._pin_np = None
         self._bt_cache: Dict[str, object] = {}
+        # host-DRAM KV offload tier (pinned) + dedicated copy stream
+        self.cpu_kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        self.num_cpu_blocks = 0
+        self._swap_stream = (
+            torch.cuda.Stream() if self.is_cuda else None
+        )
 
     # -- KV cache -----------------------------------------------------------
     def profile_and_allocate_kv(self) -> int:
@@ -97,6 +103,7 @@ class ModelRunner:
             for _ in range(m.num_layers)
         ]
         self.num_gpu_blocks = num_blocks
+        self._allocate_cpu_kv()
         kv_gib = (
             2 * num_blocks * kv_heads_local * self.config.cache.block_size
             * m.head_dim * m.num_layers * dtype.itemsize
@@ -107,6 +114,66 @@ class ModelRunner:
             self.config.cache.block_size,
             kv_gib,
         )
+
+    def _allocate_cpu_kv(self):
+        """Pinned host-DRAM KV tier (reference KVCacheOffloadingSpec
+        cpu_bytes_to_use -> vLLM OffloadingConnector; ours is native:
+        hipMemcpyAsync block copies on a side stream)."""
+        m = self.config.model
+        cache = self.config.cache
+        kv_heads_local = max(1, m.num_kv_heads // max(1, self._tp_size()))
+        per_block = (
+            2 * kv_heads_local * cache.block_size * m.head_dim
+            * m.num_layers * self.model.dtype.itemsize
+        )
+        n = cache.num_cpu_blocks
+        if not n and cache.cpu_offload_bytes:
+            n = int(cache.cpu_offload_bytes // per_block)
+        if not n:
+            return
+        shape = (n, kv_heads_local, cache.block_size, m.head_dim)
+        pin = self.is_cuda
+        self.cpu_kv_caches = [
+            (
+                torch.empty(shape, dtype=self.model.dtype, pin_memory=pin),
+                torch.empty(shape, dtype=self.model.dtype, pin_memory=pin),
+            )
+            for _ in range(m.num_layers)
+        ]
+        self.num_cpu_blocks = n
+        logger.info(
+            "KV host-offload tier: %d pinned blocks (%.2f GiB)",
+            n,
+            n * per_block / (1 << 30),
+        )
+
+    def swap_blocks(self, pairs, to_gpu: bool):
+        """Copy KV pages between tiers on the side stream; the compute
+        stream waits on the copies before the next kernel touches them."""
+        if not pairs:
+            return
+        def do_copies():
+            for layer, (gk, gv) in enumerate(self.kv_caches):
+                ck, cv = self.cpu_kv_caches[layer]
+                for a, b in pairs:
+                    if to_gpu:  # (cpu, gpu)
+                        gk[b].copy_(ck[a], non_blocking=True)
+                        gv[b].copy_(cv[a], non_blocking=True)
+                    else:       # (gpu, cpu)
+                        ck[b].copy_(gk[a], non_blocking=True)
+                        cv[b].copy_(gv[a], non_blocking=True)
+        if self.is_cuda:
+            ev_before = torch.cuda.Event()
+            ev_before.record(torch.cuda.current_stream())
+            with torch.cuda.stream(self._swap_stream):
+                # copies must not race kernels still using the pages
+                self._swap_stream.wait_event(ev_before)
+                do_copies()
+                ev = torch.cuda.Event()
+                ev.record(self._swap_stream)
+            torch.cuda.current_stream().wait_event(ev)
+        else:
+            do_copies()
 
     def _tp_size(self) -> int:
         from kserve_amd.parallel import comm

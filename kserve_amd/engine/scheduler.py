@@ -31,6 +31,10 @@ class ScheduledBatch:
     # decode: 1)
     num_scheduled_tokens: List[int] = field(default_factory=list)
     preempted: List[Request] = field(default_factory=list)
+    # KV offload tier: (gpu_block, cpu_block) / (cpu_block, gpu_block) pairs
+    # the runner must copy before this batch executes
+    swap_out: List[tuple] = field(default_factory=list)
+    swap_in: List[tuple] = field(default_factory=list)
 
     @property
     def total_tokens(self) -> int:
@@ -46,11 +50,17 @@ class Scheduler:
         scheduler_config: SchedulerConfig,
         cache_config: CacheConfig,
         num_gpu_blocks: int,
+        num_cpu_blocks: int = 0,
     ):
         self.config = scheduler_config
-        self.block_manager = BlockManager(num_gpu_blocks, cache_config.block_size)
+        self.block_manager = BlockManager(
+            num_gpu_blocks, cache_config.block_size,
+            num_cpu_blocks=num_cpu_blocks or cache_config.num_cpu_blocks,
+        )
         self.waiting: Deque[Request] = deque()
         self.running: List[Request] = []
+        # requests whose KV lives in the host-DRAM tier
+        self.swapped: List[Request] = []
 
     # -- queue ops -----------------------------------------------------------
     def add_request(self, request: Request) -> None:
@@ -60,12 +70,13 @@ class Scheduler:
         self.waiting.append(request)
 
     def abort_request(self, request_id: str) -> Optional[Request]:
-        for q in (self.waiting, self.running):
+        for q in (self.waiting, self.running, self.swapped):
             for r in list(q):
                 if r.request_id == request_id:
                     r.status = RequestStatus.FINISHED_ABORTED
                     q.remove(r)
                     self.block_manager.free(r)
+                    self.block_manager.free_cpu(r)
                     return r
         return None
 
@@ -78,7 +89,7 @@ class Scheduler:
         return len(self.running)
 
     def has_unfinished(self) -> bool:
-        return bool(self.waiting or self.running)
+        return bool(self.waiting or self.running or self.swapped)
 
     # -- scheduling ----------------------------------------------------------
     def schedule(self) -> ScheduledBatch:
@@ -130,6 +141,17 @@ class Scheduler:
 
     def _schedule_decode(self) -> ScheduledBatch:
         batch = ScheduledBatch(is_prefill=False)
+        # resume swapped requests first (KV restored from the host tier,
+        # no recompute) while capacity allows
+        while (
+            self.swapped
+            and len(self.running) < self.config.max_num_seqs
+            and self.block_manager.can_swap_in(self.swapped[0])
+        ):
+            req = self.swapped.pop(0)
+            batch.swap_in.extend(self.block_manager.swap_in(req))
+            req.status = RequestStatus.RUNNING
+            self.running.append(req)
         if not self.running:
             return batch
         # ensure every running request can take one more token; preempt the
@@ -147,7 +169,10 @@ class Scheduler:
             if total_needed <= self.block_manager.num_free_blocks:
                 break
             victim = scheduled.pop()  # newest in running order
-            self._preempt(victim)
+            if self.block_manager.can_swap_out(victim):
+                batch.swap_out.extend(self._preempt_swap(victim))
+            else:
+                self._preempt(victim)
             batch.preempted.append(victim)
         for req in scheduled:
             self.block_manager.append_slot(req)
@@ -169,9 +194,19 @@ class Scheduler:
         req.status = RequestStatus.PREEMPTED
         self.waiting.appendleft(req)
 
+    def _preempt_swap(self, req: Request):
+        """Offload preemption: KV pages move to pinned host DRAM; the request
+        resumes later via swap_in with no recompute."""
+        LLM_PREEMPTIONS.inc()
+        pairs = self.block_manager.swap_out(req)
+        req.status = RequestStatus.PREEMPTED
+        self.swapped.append(req)
+        return pairs
+
     # -- post-step ------------------------------------------------------------
     def finish_requests(self, finished: List[Request]) -> None:
         for req in finished:
             if req in self.running:
                 self.running.remove(req)
             self.block_manager.free(req)
+            self.block_manager.free_cpu(req)

@@ -177,3 +177,60 @@ def test_bert_fill_mask_gpu_matches_cpu():
     got = ours_gpu(ids.cuda(), cu.cuda())
     agree = (got.float().cpu().argmax(-1) == ref.argmax(-1)).float().mean()
     assert agree > 0.9, f"top-1 agreement {agree}"
+
+
+def test_kv_offload_swap_on_gpu():
+    """Swap-based preemption over pinned host DRAM (hipMemcpyAsync side
+    stream) preserves greedy outputs on the GPU path."""
+    from kserve_amd.engine.config import (
+        CacheConfig,
+        EngineConfig,
+        ModelConfig,
+        SchedulerConfig,
+    )
+    from kserve_amd.engine.engine import LLMEngine
+    from kserve_amd.engine.sampling_params import SamplingParams
+
+    def cfg(num_blocks, cpu_blocks):
+        return EngineConfig(
+            model=ModelConfig(
+                vocab_size=2048,
+                hidden_size=512,
+                intermediate_size=1024,
+                num_layers=4,
+                num_heads=4,
+                num_kv_heads=2,
+                head_dim=128,
+                max_position_embeddings=1024,
+            ),
+            cache=CacheConfig(
+                block_size=16, num_gpu_blocks=num_blocks, num_cpu_blocks=cpu_blocks
+            ),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=2048, max_model_len=512
+            ),
+            device="cuda",
+            seed=0,
+            eos_token_id=-1,
+            enforce_eager=True,
+        )
+
+    prompts = [[i + 1, i + 2, i + 3, i + 4] for i in range(3)]
+    sp = SamplingParams(temperature=0.0, max_tokens=60)
+    big = LLMEngine(cfg(256, 0))
+    ref = [o.output_token_ids for o in big.generate(prompts, sp).values()]
+    del big
+    torch.cuda.empty_cache()
+    small = LLMEngine(cfg(10, 64))
+    swaps = {"n": 0}
+    orig = small.scheduler.schedule
+
+    def counting():
+        b = orig()
+        swaps["n"] += len(b.swap_out)
+        return b
+
+    small.scheduler.schedule = counting
+    got = [o.output_token_ids for o in small.generate(prompts, sp).values()]
+    assert got == ref
+    assert swaps["n"] > 0
