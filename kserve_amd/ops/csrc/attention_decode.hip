@@ -23,7 +23,8 @@ constexpr int PAGE = 16;      // tokens per KV page
 constexpr int NWAVES = 4;     // waves per workgroup
 
 // D: head dim (64 or 128). ACC = D/64 output dims per lane.
-template <int D>
+// HPW: q heads per wave (GQA group = NWAVES*HPW covered per workgroup).
+template <int D, int HPW>
 __global__ __launch_bounds__(256) void paged_attention_kernel(
     short* __restrict__ out,            // [S, H, D] bf16
     const short* __restrict__ q,        // [S, H, D]
@@ -69,24 +70,37 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
     return;
   }
 
-  // q fragment for this wave's head: dims [part*D/4, (part+1)*D/4) as f32
+  // q fragments for this wave's heads: dims [part*D/4, (part+1)*D/4) f32
   constexpr int QFRAG = D / 4;  // dims per part
-  float q_frag[QFRAG];
-  const int head = kv_head * group + wave;
-  const bool active = wave < group;
-  if (active) {
-    const short* qp = q + (long)seq * q_row_stride + (long)head * D + part * QFRAG;
+  float q_frag[HPW][QFRAG];
+  int heads[HPW];
+  bool hact[HPW];
+  bool any_active = false;
 #pragma unroll
-    for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
+  for (int h = 0; h < HPW; ++h) {
+    const int head = kv_head * group + wave + h * NWAVES;
+    heads[h] = head;
+    hact[h] = (wave + h * NWAVES) < group;
+    any_active |= hact[h];
+    if (hact[h]) {
+      const short* qp =
+          q + (long)seq * q_row_stride + (long)head * D + part * QFRAG;
+#pragma unroll
+      for (int j = 0; j < QFRAG; ++j) q_frag[h][j] = bf16_bits_to_float(qp[j]);
+    }
   }
 
-  float m = NEG_INF;
-  float l = 0.f;
-  float acc[ACC];
+  float m[HPW], l[HPW];
+  float acc[HPW][ACC];
 #pragma unroll
-  for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
+  for (int h = 0; h < HPW; ++h) {
+    m[h] = NEG_INF;
+    l[h] = 0.f;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) acc[h][a] = 0.f;
+  }
 
-  if (!active) return;
+  if (!any_active) return;
   // Direct-global K/V reads, no LDS, no barriers: each page is read once per
   // workgroup group-wise (4 waves share it through L1/L2); barrier-free
   // iterations let the compiler keep many loads in flight across pages
@@ -97,74 +111,91 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
     const short* page =
         k_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
     // ---- QK: lane computes a quarter-dot for its token (coalesced: the
-    // wave's 64 lanes cover the full 4 KB page) ----
+    // wave's 64 lanes cover the full 4 KB page); K is read ONCE for all
+    // HPW heads of this wave ----
     const short8_t* kp =
         reinterpret_cast<const short8_t*>(page + tok * D + part * QFRAG);
-    float s = 0.f;
+    float kbuf[QFRAG];
 #pragma unroll
     for (int c = 0; c < QFRAG / 8; ++c) {
       short8_t kv8 = kp[c];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        s += q_frag[c * 8 + j] * bf16_bits_to_float(kv8[j]);
+        kbuf[c * 8 + j] = bf16_bits_to_float(kv8[j]);
     }
-    s = group_reduce_sum<4>(s);  // full dot in all 4 lanes of the token
     const int gtok = bi * PAGE + tok;
-    s = (gtok < ctx) ? s * scale : NEG_INF;
-
-    // ---- online softmax update ----
-    const float tmax = wave_reduce_max(s);
-    if (tmax > NEG_INF) {
-      const float m_new = fmaxf(m, tmax);
-      const float rescale = (m > NEG_INF) ? __expf(m - m_new) : 0.f;
-      const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
-      const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
-      l = l * rescale + psum;
+    const bool tok_valid = gtok < ctx;
+    const short* vpage =
+        v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    float2_t vbuf[PAGE];
 #pragma unroll
-      for (int a = 0; a < ACC; ++a) acc[a] *= rescale;
-      m = m_new;
-
-      // ---- PV: lane owns ACC consecutive output dims; the wave's reads of
-      // V row t are a single coalesced 256 B transaction ----
-      const short* vpage =
-          v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    for (int t = 0; t < PAGE; ++t) {
+      const short* vrow = vpage + t * D + lane * ACC;
+      if constexpr (ACC == 2) {
+        const unsigned int packed =
+            *reinterpret_cast<const unsigned int*>(vrow);
+        vbuf[t][0] = bf16_bits_to_float((short)(packed & 0xFFFF));
+        vbuf[t][1] = bf16_bits_to_float((short)(packed >> 16));
+      } else {
+        vbuf[t][0] = bf16_bits_to_float(vrow[0]);
+        vbuf[t][1] = 0.f;
+      }
+    }
 #pragma unroll
-      for (int t = 0; t < PAGE; ++t) {
-        const float pt = __shfl(p, t * 4, 64);
-        const short* vrow = vpage + t * D + lane * ACC;
-        float2_t vv;
-        if constexpr (ACC == 2) {
-          const unsigned int packed =
-              *reinterpret_cast<const unsigned int*>(vrow);
-          vv[0] = bf16_bits_to_float((short)(packed & 0xFFFF));
-          vv[1] = bf16_bits_to_float((short)(packed >> 16));
-        } else {
-          vv[0] = bf16_bits_to_float(vrow[0]);
-          vv[1] = 0.f;
+    for (int h = 0; h < HPW; ++h) {
+      if (!hact[h]) continue;
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < QFRAG; ++j) s += q_frag[h][j] * kbuf[j];
+      s = group_reduce_sum<4>(s);  // full dot in all 4 lanes of the token
+      s = tok_valid ? s * scale : NEG_INF;
+
+      // ---- online softmax update ----
+      const float tmax = wave_reduce_max(s);
+      if (tmax > NEG_INF) {
+        const float m_new = fmaxf(m[h], tmax);
+        const float rescale = (m[h] > NEG_INF) ? __expf(m[h] - m_new) : 0.f;
+        const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+        const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+        l[h] = l[h] * rescale + psum;
+#pragma unroll
+        for (int a = 0; a < ACC; ++a) acc[h][a] *= rescale;
+        m[h] = m_new;
+
+        // ---- PV: lane owns ACC consecutive output dims ----
+#pragma unroll
+        for (int t = 0; t < PAGE; ++t) {
+          const float pt = __shfl(p, t * 4, 64);
+#pragma unroll
+          for (int a = 0; a < ACC; ++a) acc[h][a] += pt * vbuf[t][a];
         }
-#pragma unroll
-        for (int a = 0; a < ACC; ++a) acc[a] += pt * vv[a];
       }
     }
   }
 
-  if (!active) return;
-  if (n_splits == 1) {
-    const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
-    short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
 #pragma unroll
-    for (int a = 0; a < ACC; ++a) op[a] = float_to_bf16_bits(acc[a] * inv_l);
-  } else {
-    float* po =
-        part_out + ((((long)seq * num_heads + head) * n_splits + split)) * D +
-        lane * ACC;
+  for (int h = 0; h < HPW; ++h) {
+    if (!hact[h]) continue;
+    const int head = heads[h];
+    if (n_splits == 1) {
+      const float inv_l = (l[h] > 0.f) ? 1.f / l[h] : 0.f;
+      short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
 #pragma unroll
-    for (int a = 0; a < ACC; ++a) po[a] = acc[a];
-    if (lane == 0) {
-      float* ml =
-          part_ml + (((long)seq * num_heads + head) * n_splits + split) * 2;
-      ml[0] = m;
-      ml[1] = l;
+      for (int a = 0; a < ACC; ++a)
+        op[a] = float_to_bf16_bits(acc[h][a] * inv_l);
+    } else {
+      float* po =
+          part_out +
+          ((((long)seq * num_heads + head) * n_splits + split)) * D +
+          lane * ACC;
+#pragma unroll
+      for (int a = 0; a < ACC; ++a) po[a] = acc[h][a];
+      if (lane == 0) {
+        float* ml =
+            part_ml + (((long)seq * num_heads + head) * n_splits + split) * 2;
+        ml[0] = m[h];
+        ml[1] = l[h];
+      }
     }
   }
 }
@@ -214,29 +245,28 @@ extern "C" hipError_t ks_paged_attention_decode(
     void* part_out, void* part_ml, hipStream_t stream) {
   if (block_size != PAGE) return hipErrorInvalidValue;
   const int group = num_heads / num_kv_heads;
-  // one wave per q head: groups > 4 are handled by the caller splitting the
-  // GQA group into 4-head subgroups (sharing a kv head across workgroups)
-  if (group > NWAVES) return hipErrorInvalidValue;
+  const int hpw = (group + NWAVES - 1) / NWAVES;  // q heads per wave
+  if (hpw > 2) return hipErrorInvalidValue;  // groups up to 8 supported
   if (n_splits < 1) n_splits = 1;
   dim3 grid(num_kv_heads, num_seqs, n_splits);
   dim3 block(256);
+#define LAUNCH_PA(DD, HH)                                                  \
+  hipLaunchKernelGGL((paged_attention_kernel<DD, HH>), grid, block, 0,     \
+                     stream, (short*)out, (const short*)q,                 \
+                     (const short*)k_cache, (const short*)v_cache,         \
+                     (const int*)block_tables, (const int*)context_lens,   \
+                     scale, num_kv_heads, group, max_blocks, q_row_stride, \
+                     n_splits, (float*)part_out, (float*)part_ml)
   if (head_dim == 128) {
-    hipLaunchKernelGGL((paged_attention_kernel<128>), grid, block, 0, stream,
-                       (short*)out, (const short*)q, (const short*)k_cache,
-                       (const short*)v_cache, (const int*)block_tables,
-                       (const int*)context_lens, scale, num_kv_heads, group,
-                       max_blocks, q_row_stride, n_splits, (float*)part_out,
-                       (float*)part_ml);
+    if (hpw == 1) LAUNCH_PA(128, 1);
+    else LAUNCH_PA(128, 2);
   } else if (head_dim == 64) {
-    hipLaunchKernelGGL((paged_attention_kernel<64>), grid, block, 0, stream,
-                       (short*)out, (const short*)q, (const short*)k_cache,
-                       (const short*)v_cache, (const int*)block_tables,
-                       (const int*)context_lens, scale, num_kv_heads, group,
-                       max_blocks, q_row_stride, n_splits, (float*)part_out,
-                       (float*)part_ml);
+    if (hpw == 1) LAUNCH_PA(64, 1);
+    else LAUNCH_PA(64, 2);
   } else {
     return hipErrorInvalidValue;
   }
+#undef LAUNCH_PA
   HIP_CHECK_KERNEL();
   if (n_splits > 1) {
     const long sh = (long)num_seqs * num_heads;

@@ -125,6 +125,8 @@ class TestDecodeAttention:
             (64, 32, 8, 500),   # moderate batch
             (2, 8, 2, 17),      # ragged boundary
             (8, 4, 4, 64),      # group=1 (MHA)
+            (4, 8, 1, 100),     # group=8 (llama-70b TP=8 shape)
+            (16, 64, 8, 200),   # full 70b head config
         ],
     )
     def test_paged_decode(self, dev, S, H, Hkv, ctx_max):
