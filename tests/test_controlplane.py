@@ -225,3 +225,90 @@ class TestModelConfig:
 
         parsed = ModelConfigWatcher.parse_config(payload)
         assert set(parsed) == {"m1", "m2"}
+
+
+class TestLLMInferenceService:
+    def _llm(self, **wk):
+        from kserve_amd.controlplane.llmisvc import (
+            LLMInferenceService,
+            LLMInferenceServiceSpec,
+            LLMModelSpec,
+            ParallelismSpec,
+            WorkloadSpec,
+        )
+
+        return LLMInferenceService(
+            name="llama",
+            namespace="prod",
+            spec=LLMInferenceServiceSpec(
+                model=LLMModelSpec(uri="hf://meta-llama/Llama-3-8B", name="llama-3-8b"),
+                workload=WorkloadSpec(
+                    parallelism=ParallelismSpec(**wk.pop("parallelism", {})), **wk
+                ),
+            ),
+        )
+
+    def test_single_node_tp8(self):
+        from kserve_amd.controlplane.llmisvc import reconcile_llm
+
+        llm = self._llm(parallelism={"tensor": 8})
+        out = reconcile_llm(llm)
+        dep = out["decode"]
+        assert dep["kind"] == "Deployment"
+        c = dep["spec"]["template"]["spec"]["containers"][0]
+        assert "--tensor-parallel-size=8" in c["args"]
+        assert c["resources"]["limits"]["amd.com/gpu"] == "8"
+        env = {e["name"]: e["value"] for e in c["env"]}
+        assert env["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
+
+    def test_multi_node_lws(self):
+        from kserve_amd.controlplane.llmisvc import reconcile_llm
+
+        llm = self._llm(parallelism={"tensor": 8, "pipeline": 2})
+        out = reconcile_llm(llm)
+        lws = out["decode"]
+        assert lws["kind"] == "LeaderWorkerSet"
+        assert lws["spec"]["leaderWorkerTemplate"]["size"] == 2
+
+    def test_prefill_disagg_and_scheduler(self):
+        from kserve_amd.controlplane.llmisvc import (
+            SchedulerSpec,
+            WorkloadSpec,
+            reconcile_llm,
+        )
+
+        llm = self._llm(parallelism={"tensor": 4})
+        llm.spec.prefill = WorkloadSpec()
+        llm.spec.scheduler = SchedulerSpec()
+        out = reconcile_llm(llm)
+        assert out["prefill"]["metadata"]["name"] == "llama-prefill"
+        assert out["scheduler"]["metadata"]["name"] == "llama-epp"
+
+    def test_kv_offload_args(self):
+        from kserve_amd.controlplane.llmisvc import (
+            KVCacheOffloadingSpec,
+            reconcile_llm,
+        )
+
+        llm = self._llm()
+        llm.spec.workload.kv_cache_offloading = KVCacheOffloadingSpec(
+            cpu_bytes_to_use=64 << 30,
+            filesystem_tiers=[{"emptyDir": {}}],
+        )
+        out = reconcile_llm(llm)
+        c = out["decode"]["spec"]["template"]["spec"]["containers"][0]
+        assert f"--kv-offload-bytes={64 << 30}" in c["args"]
+        assert out["decode"]["spec"]["template"]["spec"]["volumes"]
+
+    def test_tracing_env(self):
+        from kserve_amd.controlplane.llmisvc import TracingSpec, reconcile_llm
+
+        llm = self._llm()
+        llm.spec.tracing = TracingSpec(enabled=True, otlp_endpoint="http://otel:4317")
+        out = reconcile_llm(llm)
+        env = {
+            e["name"]: e["value"]
+            for e in out["decode"]["spec"]["template"]["spec"]["containers"][0]["env"]
+        }
+        assert env["OTEL_EXPORTER_OTLP_ENDPOINT"] == "http://otel:4317"
+        assert env["OTEL_TRACES_SAMPLER_ARG"] == "0.05"
