@@ -226,7 +226,11 @@ class LLMEngine:
                         new_token_ids=[int(tok)],
                         finished=req.finished,
                         finish_reason=req.finish_reason,
-                        output_token_ids=list(req.output_token_ids),
+                        # full-list copy only when the request completes
+                        # (per-step copies are O(len) x batch)
+                        output_token_ids=(
+                            list(req.output_token_ids) if req.finished else req.output_token_ids
+                        ),
                         num_prompt_tokens=req.num_prompt_tokens,
                         text_delta=delta,
                         output_text=req.output_text,
