@@ -115,6 +115,52 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
     // HPW heads of this wave ----
     const short8_t* kp =
         reinterpret_cast<const short8_t*>(page + tok * D + part * QFRAG);
+    const int gtok = bi * PAGE + tok;
+    const bool tok_valid = gtok < ctx;
+    const short* vpage =
+        v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+
+    if constexpr (HPW == 1) {
+      // fast path: dot folded into the loads, V read inline during PV —
+      // lowest VGPR pressure (highest occupancy; this loop is latency-bound)
+      float s = 0.f;
+#pragma unroll
+      for (int c = 0; c < QFRAG / 8; ++c) {
+        short8_t kv8 = kp[c];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          s += q_frag[0][c * 8 + j] * bf16_bits_to_float(kv8[j]);
+      }
+      s = group_reduce_sum<4>(s);
+      s = tok_valid ? s * scale : NEG_INF;
+      const float tmax = wave_reduce_max(s);
+      if (tmax > NEG_INF) {
+        const float m_new = fmaxf(m[0], tmax);
+        const float rescale = (m[0] > NEG_INF) ? __expf(m[0] - m_new) : 0.f;
+        const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+        const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+        l[0] = l[0] * rescale + psum;
+#pragma unroll
+        for (int a = 0; a < ACC; ++a) acc[0][a] *= rescale;
+        m[0] = m_new;
+#pragma unroll
+        for (int t = 0; t < PAGE; ++t) {
+          const float pt = __shfl(p, t * 4, 64);
+          const short* vrow = vpage + t * D + lane * ACC;
+          if constexpr (ACC == 2) {
+            const unsigned int packed =
+                *reinterpret_cast<const unsigned int*>(vrow);
+            acc[0][0] += pt * bf16_bits_to_float((short)(packed & 0xFFFF));
+            acc[0][1] += pt * bf16_bits_to_float((short)(packed >> 16));
+          } else {
+            acc[0][0] += pt * bf16_bits_to_float(vrow[0]);
+          }
+        }
+      }
+      continue;
+    }
+
+    // HPW >= 2: K/V buffered once per page and reused across heads
     float kbuf[QFRAG];
 #pragma unroll
     for (int c = 0; c < QFRAG / 8; ++c) {
@@ -123,10 +169,6 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
       for (int j = 0; j < 8; ++j)
         kbuf[c * 8 + j] = bf16_bits_to_float(kv8[j]);
     }
-    const int gtok = bi * PAGE + tok;
-    const bool tok_valid = gtok < ctx;
-    const short* vpage =
-        v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
     float2_t vbuf[PAGE];
 #pragma unroll
     for (int t = 0; t < PAGE; ++t) {
