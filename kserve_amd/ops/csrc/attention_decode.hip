@@ -121,40 +121,50 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
         v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
 
     if constexpr (HPW == 1) {
-      // fast path: dot folded into the loads, V read inline during PV —
-      // lowest VGPR pressure (highest occupancy; this loop is latency-bound)
+      // fast path: both the K fragments and the V row words for the page
+      // are ISSUED as one burst (8 KB/wave in flight) before any use, and
+      // the softmax update is branchless (NEG_INF is finite, so the
+      // fully-masked-page case degenerates to a no-op: rescale=exp(0)=1,
+      // p=0) — nothing blocks the scheduler from overlapping pages.
+      short8_t kreg[QFRAG / 8];
+#pragma unroll
+      for (int c = 0; c < QFRAG / 8; ++c) kreg[c] = kp[c];
+      unsigned int vreg[PAGE];
+#pragma unroll
+      for (int t = 0; t < PAGE; ++t) {
+        if constexpr (ACC == 2) {
+          vreg[t] = *reinterpret_cast<const unsigned int*>(
+              vpage + t * D + lane * ACC);
+        } else {
+          vreg[t] = (unsigned short)*(vpage + t * D + lane);
+        }
+      }
       float s = 0.f;
 #pragma unroll
       for (int c = 0; c < QFRAG / 8; ++c) {
-        short8_t kv8 = kp[c];
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          s += q_frag[0][c * 8 + j] * bf16_bits_to_float(kv8[j]);
+          s += q_frag[0][c * 8 + j] * bf16_bits_to_float(kreg[c][j]);
       }
       s = group_reduce_sum<4>(s);
       s = tok_valid ? s * scale : NEG_INF;
       const float tmax = wave_reduce_max(s);
-      if (tmax > NEG_INF) {
-        const float m_new = fmaxf(m[0], tmax);
-        const float rescale = (m[0] > NEG_INF) ? __expf(m[0] - m_new) : 0.f;
-        const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
-        const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
-        l[0] = l[0] * rescale + psum;
+      const float m_new = fmaxf(m[0], tmax);
+      const float rescale = __expf(m[0] - m_new);
+      const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+      const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+      l[0] = l[0] * rescale + psum;
 #pragma unroll
-        for (int a = 0; a < ACC; ++a) acc[0][a] *= rescale;
-        m[0] = m_new;
+      for (int a = 0; a < ACC; ++a) acc[0][a] *= rescale;
+      m[0] = m_new;
 #pragma unroll
-        for (int t = 0; t < PAGE; ++t) {
-          const float pt = __shfl(p, t * 4, 64);
-          const short* vrow = vpage + t * D + lane * ACC;
-          if constexpr (ACC == 2) {
-            const unsigned int packed =
-                *reinterpret_cast<const unsigned int*>(vrow);
-            acc[0][0] += pt * bf16_bits_to_float((short)(packed & 0xFFFF));
-            acc[0][1] += pt * bf16_bits_to_float((short)(packed >> 16));
-          } else {
-            acc[0][0] += pt * bf16_bits_to_float(vrow[0]);
-          }
+      for (int t = 0; t < PAGE; ++t) {
+        const float pt = __shfl(p, t * 4, 64);
+        if constexpr (ACC == 2) {
+          acc[0][0] += pt * bf16_bits_to_float((short)(vreg[t] & 0xFFFF));
+          acc[0][1] += pt * bf16_bits_to_float((short)(vreg[t] >> 16));
+        } else {
+          acc[0][0] += pt * bf16_bits_to_float((short)vreg[t]);
         }
       }
       continue;
