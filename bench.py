@@ -45,6 +45,8 @@ def main():
     parser.add_argument("--eager", action="store_true",
                         help="disable hipGraph capture")
     parser.add_argument("--temperature", type=float, default=0.0)
+    parser.add_argument("--profile-cpu", action="store_true",
+                        help="cProfile the timed loop and print hot functions")
     args = parser.parse_args()
 
     import torch
@@ -130,9 +132,17 @@ def main():
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     tokens = 0
+    prof = None
+    if args.profile_cpu:
+        import cProfile
+
+        prof = cProfile.Profile()
+        prof.enable()
     for _ in range(args.steps):
         outs = engine.step()
         tokens += len(outs)
+    if prof is not None:
+        prof.disable()
     comm.barrier()
     if use_gpu:
         torch.cuda.synchronize()
@@ -149,6 +159,13 @@ def main():
         dist.all_reduce(tk, op=dist.ReduceOp.SUM)
         elapsed = float(te[0])
         tokens = int(tk[0])
+
+    if prof is not None and rank == 0:
+        import pstats
+        import sys
+
+        stats = pstats.Stats(prof, stream=sys.stderr)
+        stats.sort_stats("cumulative").print_stats(25)
 
     if rank == 0:
         value = tokens / elapsed
