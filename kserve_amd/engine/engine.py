@@ -190,16 +190,19 @@ class LLMEngine:
             req.num_computed_tokens += n
         # chunked prefill: requests whose prompt isn't fully computed don't
         # sample this step
-        sample_reqs = [
-            r for r in batch.requests if r.num_computed_tokens >= r.num_tokens
-        ]
-        if len(sample_reqs) < len(batch.requests):
-            idx = [
-                i
-                for i, r in enumerate(batch.requests)
-                if r.num_computed_tokens >= r.num_tokens
+        if batch.is_prefill:
+            sample_reqs = [
+                r for r in batch.requests if r.num_computed_tokens >= r.num_tokens
             ]
-            logits = logits[idx]
+            if len(sample_reqs) < len(batch.requests):
+                idx = [
+                    i
+                    for i, r in enumerate(batch.requests)
+                    if r.num_computed_tokens >= r.num_tokens
+                ]
+                logits = logits[idx]
+        else:
+            sample_reqs = batch.requests
         outputs: List[RequestOutput] = []
         finished: List[Request] = []
         if sample_reqs:
@@ -216,6 +219,7 @@ class LLMEngine:
                 if not req.finished and req.sampling_params.stop:
                     if self.detokenizer.check_stop_strings(req) is not None:
                         req.status = RequestStatus.FINISHED_STOPPED
+                        req.is_finished = True
                         req.finish_time = time.monotonic()
                 if req.finished:
                     finished.append(req)

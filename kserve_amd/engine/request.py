@@ -56,6 +56,7 @@ class Request:
         self.status = RequestStatus.WAITING
         self.arrival_time = arrival_time if arrival_time is not None else time.monotonic()
         self.eos_token_id = eos_token_id
+        self.is_finished = False  # cached; see maybe_finish / set_status
         # paging state
         self.block_table: List[int] = []
         # number of prompt tokens whose KV is already computed (chunked prefill)
@@ -87,7 +88,7 @@ class Request:
 
     @property
     def finished(self) -> bool:
-        return self.status.finished
+        return self.is_finished
 
     @property
     def finish_reason(self) -> Optional[str]:
@@ -105,21 +106,27 @@ class Request:
     def maybe_finish(self, max_model_len: int) -> bool:
         """Apply stop conditions after a new output token. Returns finished."""
         sp = self.sampling_params
-        last = self.output_token_ids[-1] if self.output_token_ids else None
-        n_out = self.num_output_tokens
+        out = self.output_token_ids
+        last = out[-1] if out else None
+        n_out = len(out)
+        status = None
         if n_out >= sp.min_tokens and last is not None:
             if not sp.ignore_eos and self.eos_token_id is not None and last == self.eos_token_id:
-                self.status = RequestStatus.FINISHED_STOPPED
-            elif last in sp.stop_token_ids:
-                self.status = RequestStatus.FINISHED_STOPPED
-        if not self.finished:
+                status = RequestStatus.FINISHED_STOPPED
+            elif sp.stop_token_ids and last in sp.stop_token_ids:
+                status = RequestStatus.FINISHED_STOPPED
+        if status is None:
             if sp.max_tokens is not None and n_out >= sp.max_tokens:
-                self.status = RequestStatus.FINISHED_LENGTH
-            elif self.num_tokens >= max_model_len:
-                self.status = RequestStatus.FINISHED_LENGTH
-        if self.finished and self.finish_time is None:
-            self.finish_time = time.monotonic()
-        return self.finished
+                status = RequestStatus.FINISHED_LENGTH
+            elif len(self.prompt_token_ids) + n_out >= max_model_len:
+                status = RequestStatus.FINISHED_LENGTH
+        if status is not None:
+            self.status = status
+            self.is_finished = True
+            if self.finish_time is None:
+                self.finish_time = time.monotonic()
+            return True
+        return False
 
 
 @dataclass

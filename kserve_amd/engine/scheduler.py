@@ -66,6 +66,7 @@ class Scheduler:
     def add_request(self, request: Request) -> None:
         if request.num_prompt_tokens > self.config.max_model_len:
             request.status = RequestStatus.FINISHED_LENGTH
+            request.is_finished = True
             return
         self.waiting.append(request)
 
@@ -74,6 +75,7 @@ class Scheduler:
             for r in list(q):
                 if r.request_id == request_id:
                     r.status = RequestStatus.FINISHED_ABORTED
+                    r.is_finished = True
                     q.remove(r)
                     self.block_manager.free(r)
                     self.block_manager.free_cpu(r)
@@ -165,6 +167,10 @@ class Scheduler:
             return max(0, need - have)
 
         while scheduled:
+            # fast path: each decode step needs at most one new block per
+            # request, so enough free blocks means no preemption check
+            if self.block_manager.num_free_blocks >= len(scheduled):
+                break
             total_needed = sum(blocks_needed_now(r) for r in scheduled)
             if total_needed <= self.block_manager.num_free_blocks:
                 break
