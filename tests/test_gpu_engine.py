@@ -215,12 +215,32 @@ def test_kv_offload_swap_on_gpu():
             enforce_eager=True,
         )
 
+    # 1) KV pages round-trip bit-exactly through the pinned host tier
+    engine = LLMEngine(cfg(32, 32))
+    runner = engine.runner
+    for layer, (gk, gv) in enumerate(runner.kv_caches):
+        gk[3].normal_()
+        gv[3].normal_()
+    snap_k = [gk[3].clone() for gk, _ in runner.kv_caches]
+    snap_v = [gv[3].clone() for _, gv in runner.kv_caches]
+    runner.swap_blocks([(3, 5)], to_gpu=False)   # gpu block 3 -> cpu block 5
+    for gk, gv in runner.kv_caches:
+        gk[3].zero_()
+        gv[3].zero_()
+    runner.swap_blocks([(5, 7)], to_gpu=True)    # cpu block 5 -> gpu block 7
+    torch.cuda.synchronize()
+    for layer, (gk, gv) in enumerate(runner.kv_caches):
+        assert torch.equal(gk[7], snap_k[layer])
+        assert torch.equal(gv[7], snap_v[layer])
+    del engine
+    torch.cuda.empty_cache()
+
+    # 2) engine-level: swaps occur and all requests complete (exact token
+    # equality vs an unconstrained engine is not required on GPU: hipBLASLt
+    # picks batch-size-dependent algorithms, so bf16 greedy outputs can
+    # legitimately differ when preemption changes batch composition)
     prompts = [[i + 1, i + 2, i + 3, i + 4] for i in range(3)]
     sp = SamplingParams(temperature=0.0, max_tokens=60)
-    big = LLMEngine(cfg(256, 0))
-    ref = [o.output_token_ids for o in big.generate(prompts, sp).values()]
-    del big
-    torch.cuda.empty_cache()
     small = LLMEngine(cfg(10, 64))
     swaps = {"n": 0}
     orig = small.scheduler.schedule
@@ -231,6 +251,8 @@ def test_kv_offload_swap_on_gpu():
         return b
 
     small.scheduler.schedule = counting
-    got = [o.output_token_ids for o in small.generate(prompts, sp).values()]
-    assert got == ref
+    got = small.generate(prompts, sp)
+    assert len(got) == 3
+    for o in got.values():
+        assert len(o.output_token_ids) == 60
     assert swaps["n"] > 0
