@@ -136,3 +136,69 @@ def test_tp2_parallel_linears():
 def test_tp2_llama_matches_hf():
     pytest.importorskip("transformers")
     _run_workers(_tp2_llama_worker, port=29613)
+
+
+def _tp2_engine_worker(rank, world, port, q):
+    try:
+        _set_env(rank, world, port)
+        import torch
+
+        from kserve_amd.engine.config import (
+            CacheConfig,
+            EngineConfig,
+            ModelConfig,
+            SchedulerConfig,
+        )
+        from kserve_amd.engine.engine import LLMEngine
+        from kserve_amd.engine.sampling_params import SamplingParams
+        from kserve_amd.parallel import comm
+
+        comm.init_distributed(tp_size=2, backend="gloo")
+        cfg = EngineConfig(
+            model=ModelConfig.tiny(vocab_size=128),
+            cache=CacheConfig(block_size=4, num_gpu_blocks=64),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=128, max_model_len=64
+            ),
+            device="cpu",
+            seed=7,
+            eos_token_id=-1,
+        )
+        engine = LLMEngine(cfg)
+        out = engine.generate(
+            [[1, 2, 3], [9, 8, 7, 6]],
+            SamplingParams(temperature=0.0, max_tokens=6),
+        )
+        toks = [o.output_token_ids for o in out.values()]
+        comm.destroy_distributed()
+        q.put((rank, ("ok", toks)))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, (f"FAIL: {e}\n{traceback.format_exc()}", None)))
+
+
+@pytest.mark.timeout(300)
+def test_tp2_engine_generate_consistent():
+    """TP=2 engine must produce identical greedy tokens on both ranks.
+
+    (Random-init weights are seeded per-rank via tp_rank offsets in
+    random_init — ranks hold different SHARDS of one logical model, so the
+    all-reduced logits and therefore tokens must agree across ranks.)"""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_tp2_engine_worker, args=(r, 2, 29617, q))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, (status, toks) = q.get(timeout=240)
+        assert status == "ok", f"rank {rank}: {status}"
+        results[rank] = toks
+    for p in procs:
+        p.join(timeout=60)
+    assert results[0] == results[1]
+    assert all(len(t) == 6 for t in results[0])
