@@ -252,6 +252,179 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Wave-split variant: the 4 waves partition the PAGES (each page is read by
+// exactly one wave — no cross-wave L1 re-reads), every wave computes ALL
+// GROUP heads for its pages, and the four per-wave online-softmax partials
+// are merged through LDS at the end. Q is kept as bf16 fragments
+// (GROUP x 16 VGPR) and converted in the dot.
+// ---------------------------------------------------------------------------
+template <int D, int GROUP>
+__global__ __launch_bounds__(256) void paged_attention_ws_kernel(
+    short* __restrict__ out, const short* __restrict__ q,
+    const short* __restrict__ k_cache, const short* __restrict__ v_cache,
+    const int* __restrict__ block_tables, const int* __restrict__ context_lens,
+    const float scale, const int num_kv_heads, const int max_blocks,
+    const long q_row_stride, const int n_splits,
+    float* __restrict__ part_out, float* __restrict__ part_ml) {
+  constexpr int ACC = D / 64;
+  constexpr int QFRAG = D / 4;
+  const int kv_head = blockIdx.x;
+  const int seq = blockIdx.y;
+  const int split = blockIdx.z;
+  const int ctx = context_lens[seq];
+  if (ctx <= 0) return;
+  const int num_heads = num_kv_heads * GROUP;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tok = lane >> 2;
+  const int part = lane & 3;
+
+  const int nblocks = (ctx + PAGE - 1) / PAGE;
+  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
+  const int blk_lo = split * blocks_per_split;
+  const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
+
+  __shared__ float lds_ml[NWAVES][GROUP][2];
+  __shared__ float lds_acc[NWAVES][GROUP][64 * ACC];
+
+  if (blk_lo >= blk_hi) {
+    if (n_splits > 1 && threadIdx.x < (unsigned)GROUP) {
+      const int h = kv_head * GROUP + threadIdx.x;
+      float* ml = part_ml + (((long)seq * num_heads + h) * n_splits + split) * 2;
+      ml[0] = NEG_INF;
+      ml[1] = 0.f;
+    }
+    return;
+  }
+
+  // q rows stay in L1 (re-read per page) instead of registers: GROUP x 16
+  // VGPRs of persistent fragments would cap occupancy at 1-2 waves/SIMD
+  const short8_t* qrows[GROUP];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    qrows[g] = reinterpret_cast<const short8_t*>(
+        q + (long)seq * q_row_stride + (long)(kv_head * GROUP + g) * D +
+        part * QFRAG);
+  }
+
+  float m[GROUP], l[GROUP], acc[GROUP][ACC];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    m[g] = NEG_INF;
+    l[g] = 0.f;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) acc[g][a] = 0.f;
+  }
+
+  const int bt_base = (int)((long)seq * max_blocks);
+  for (int bi = blk_lo + wave; bi < blk_hi; bi += NWAVES) {
+    const int block_id = block_tables[bt_base + bi];
+    const long pbase = (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    const short8_t* kp =
+        reinterpret_cast<const short8_t*>(k_cache + pbase + tok * D + part * QFRAG);
+    short8_t kreg[QFRAG / 8];
+#pragma unroll
+    for (int c = 0; c < QFRAG / 8; ++c) kreg[c] = kp[c];
+    const short* vpage = v_cache + pbase;
+    unsigned int vreg[PAGE];
+#pragma unroll
+    for (int t = 0; t < PAGE; ++t) {
+      if constexpr (ACC == 2) {
+        vreg[t] =
+            *reinterpret_cast<const unsigned int*>(vpage + t * D + lane * ACC);
+      } else {
+        vreg[t] = (unsigned short)*(vpage + t * D + lane);
+      }
+    }
+    const int gtok = bi * PAGE + tok;
+    const bool tok_valid = gtok < ctx;
+#pragma unroll 1
+    for (int g = 0; g < GROUP; ++g) {
+      float s = 0.f;
+#pragma unroll
+      for (int c = 0; c < QFRAG / 8; ++c) {
+        const short8_t qv = qrows[g][c];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          s += bf16_bits_to_float(qv[j]) * bf16_bits_to_float(kreg[c][j]);
+      }
+      s = group_reduce_sum<4>(s);
+      s = tok_valid ? s * scale : NEG_INF;
+      const float tmax = wave_reduce_max(s);
+      const float m_new = fmaxf(m[g], tmax);
+      const float rescale = __expf(m[g] - m_new);
+      const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+      const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+      l[g] = l[g] * rescale + psum;
+#pragma unroll
+      for (int a = 0; a < ACC; ++a) acc[g][a] *= rescale;
+      m[g] = m_new;
+#pragma unroll
+      for (int t = 0; t < PAGE; ++t) {
+        const float pt = __shfl(p, t * 4, 64);
+        if constexpr (ACC == 2) {
+          acc[g][0] += pt * bf16_bits_to_float((short)(vreg[t] & 0xFFFF));
+          acc[g][1] += pt * bf16_bits_to_float((short)(vreg[t] >> 16));
+        } else {
+          acc[g][0] += pt * bf16_bits_to_float((short)vreg[t]);
+        }
+      }
+    }
+  }
+
+  // ---- cross-wave merge via LDS ----
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    if (lane == 0) {
+      lds_ml[wave][g][0] = m[g];
+      lds_ml[wave][g][1] = l[g];
+    }
+#pragma unroll
+    for (int a = 0; a < ACC; ++a)
+      lds_acc[wave][g][lane * ACC + a] = acc[g][a];
+  }
+  __syncthreads();
+  // wave w merges heads {w, w+4, ...}
+  for (int g = wave; g < GROUP; g += NWAVES) {
+    float m_all = NEG_INF;
+#pragma unroll
+    for (int w = 0; w < NWAVES; ++w) m_all = fmaxf(m_all, lds_ml[w][g][0]);
+    float l_all = 0.f;
+    float a_all[ACC];
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) a_all[a] = 0.f;
+#pragma unroll
+    for (int w = 0; w < NWAVES; ++w) {
+      const float wgt = __expf(lds_ml[w][g][0] - m_all);
+      l_all += wgt * lds_ml[w][g][1];
+#pragma unroll
+      for (int a = 0; a < ACC; ++a)
+        a_all[a] += wgt * lds_acc[w][g][lane * ACC + a];
+    }
+    const int head = kv_head * GROUP + g;
+    if (n_splits == 1) {
+      const float inv_l = (l_all > 0.f) ? 1.f / l_all : 0.f;
+      short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
+#pragma unroll
+      for (int a = 0; a < ACC; ++a)
+        op[a] = float_to_bf16_bits(a_all[a] * inv_l);
+    } else {
+      float* po = part_out +
+                  (((long)seq * num_heads + head) * n_splits + split) * D +
+                  lane * ACC;
+#pragma unroll
+      for (int a = 0; a < ACC; ++a) po[a] = a_all[a];
+      if (lane == 0) {
+        float* ml =
+            part_ml + (((long)seq * num_heads + head) * n_splits + split) * 2;
+        ml[0] = m_all;
+        ml[1] = l_all;
+      }
+    }
+  }
+}
+
 // Combine split-context partials: one wave per (seq, head); lane owns D/64
 // dims across all splits.
 template <int D>
@@ -302,6 +475,40 @@ extern "C" hipError_t ks_paged_attention_decode(
   if (n_splits < 1) n_splits = 1;
   dim3 grid(num_kv_heads, num_seqs, n_splits);
   dim3 block(256);
+
+  // wave-split variant (each page read by one wave; measured A/B via env)
+  static const bool use_ws = [] {
+    const char* e = getenv("KS_ATTN_WS");
+    return e == nullptr || e[0] != '0';
+  }();
+  const bool pow2_group =
+      group == 1 || group == 2 || group == 4 || group == 8;
+  if (use_ws && pow2_group && head_dim == 128) {
+#define LAUNCH_WS(GG)                                                       \
+  hipLaunchKernelGGL((paged_attention_ws_kernel<128, GG>), grid, block, 0,  \
+                     stream, (short*)out, (const short*)q,                  \
+                     (const short*)k_cache, (const short*)v_cache,          \
+                     (const int*)block_tables, (const int*)context_lens,    \
+                     scale, num_kv_heads, max_blocks, q_row_stride,         \
+                     n_splits, (float*)part_out, (float*)part_ml)
+    if (group == 1) LAUNCH_WS(1);
+    else if (group == 2) LAUNCH_WS(2);
+    else if (group == 4) LAUNCH_WS(4);
+    else LAUNCH_WS(8);
+#undef LAUNCH_WS
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      const long sh = (long)num_seqs * num_heads;
+      int wpb = 4;
+      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(wpb * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return hipSuccess;
+  }
 #define LAUNCH_PA(DD, HH)                                                  \
   hipLaunchKernelGGL((paged_attention_kernel<DD, HH>), grid, block, 0,     \
                      stream, (short*)out, (const short*)q,                 \
