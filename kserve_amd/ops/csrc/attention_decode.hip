@@ -476,10 +476,13 @@ extern "C" hipError_t ks_paged_attention_decode(
   dim3 grid(num_kv_heads, num_seqs, n_splits);
   dim3 block(256);
 
-  // wave-split variant (each page read by one wave; measured A/B via env)
+  // wave-split variant (each page read by one wave). Measured SLOWER than
+  // the head-split kernel on MI355X (profiles/attn_decode_bench.md):
+  // the per-head serial chains outweigh the 4x L1-reuse saving. Kept as an
+  // opt-in ablation (KS_ATTN_WS=1).
   static const bool use_ws = [] {
     const char* e = getenv("KS_ATTN_WS");
-    return e == nullptr || e[0] != '0';
+    return e != nullptr && e[0] == '1';
   }();
   const bool pow2_group =
       group == 1 || group == 2 || group == 4 || group == 8;
