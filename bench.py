@@ -171,7 +171,7 @@ def main():
         value = tokens / elapsed
         n_gpus = world if use_gpu else args.gpus
         result = {
-            "metric": "output tok/s, Llama-3-8B /v1/completions decode",
+            "metric": f"output tok/s, {mcfg.model_name} /v1/completions decode",
             "value": round(value, 1),
             "unit": "tok/s",
             "n_gpus": n_gpus,
