@@ -103,6 +103,9 @@ def create_app(
 
     register_v1_endpoints(app, dataplane)
     register_v2_endpoints(app, dataplane, model_repository_extension)
+    from kserve_amd.protocol.rest.timeseries import register_timeseries_endpoints
+
+    register_timeseries_endpoints(app, dataplane)
     install_exception_handlers(app)
     return app
 

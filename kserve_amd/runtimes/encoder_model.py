@@ -221,3 +221,46 @@ class EncoderModel(Model):
             o.set_data_from_numpy(arr, binary_data=False)
             return InferResponse(v2.id, self.name, [o])
         return {"predictions": predictions}
+
+
+class OpenAIEmbeddingAdapter:
+    """Exposes an EncoderModel on the OpenAI /v1/embeddings route
+    (reference: OpenAIEncoderModel, openai_model.py)."""
+
+    def __init__(self, encoder: "EncoderModel"):
+        self.encoder = encoder
+        self.name = encoder.name
+        self.ready = True
+        self.engine = False
+
+    async def create_embedding(self, request, raw_request=None, context=None):
+        from kserve_amd.protocol.rest.openai.types import (
+            Embedding,
+            EmbeddingObject,
+            UsageInfo,
+        )
+
+        texts = request.input if isinstance(request.input, list) else [request.input]
+        texts = [str(t) for t in texts]
+        payload = self.encoder.preprocess({"instances": texts})
+        result = self.encoder.predict(payload)
+        post = self.encoder.postprocess(result)
+        vectors = post["predictions"]
+        data = [
+            EmbeddingObject(index=i, embedding=v) for i, v in enumerate(vectors)
+        ]
+        total = sum(len(ids) for ids in result["ids_list"])
+        return Embedding(
+            data=data,
+            model=self.name,
+            usage=UsageInfo(prompt_tokens=total, total_tokens=total),
+        )
+
+    async def create_completion(self, *a, **k):
+        raise NotImplementedError("encoder models do not generate")
+
+    async def create_chat_completion(self, *a, **k):
+        raise NotImplementedError("encoder models do not generate")
+
+    async def create_rerank(self, *a, **k):
+        raise NotImplementedError

@@ -133,7 +133,16 @@ class LLMModel(OpenAIModel):
             ids = self._encode_prompt(p)
             prompt_tokens += len(ids)
             for j in range(sp.n):
-                out = await self.async_engine.generate_full(ids, sp)
+                sp_j = sp
+                if sp.n > 1:
+                    import dataclasses
+
+                    sp_j = dataclasses.replace(
+                        sp,
+                        n=1,
+                        seed=(sp.seed + j) if sp.seed is not None else None,
+                    )
+                out = await self.async_engine.generate_full(ids, sp_j)
                 completion_tokens += len(out.output_token_ids)
                 text = (
                     out.output_text

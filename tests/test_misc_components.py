@@ -1,0 +1,171 @@
+"""Timeseries protocol, multiprocess REST, LocalModelCache, OpenAI
+embeddings adapter."""
+
+import asyncio
+import json
+import os
+
+import httpx
+import pytest
+from fastapi.testclient import TestClient
+
+from kserve_amd.model import Model
+from kserve_amd.model_repository import ModelRepository
+from kserve_amd.protocol.dataplane import DataPlane
+from kserve_amd.protocol.rest.server import create_app
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+class ForecastModel(Model):
+    def __init__(self):
+        super().__init__("ts")
+        self.ready = True
+
+    async def forecast(self, inputs, horizon, quantiles=None, parameters=None):
+        out = []
+        for series in inputs:
+            vals = series["values"]
+            mean = sum(vals) / len(vals)
+            out.append({"forecast": [mean] * horizon})
+        return out
+
+
+class TestTimeSeries:
+    def test_forecast(self):
+        repo = ModelRepository()
+        repo.update(ForecastModel())
+        client = TestClient(create_app(DataPlane(repo)))
+        r = client.post(
+            "/timeseries/v1/forecast",
+            json={
+                "model": "ts",
+                "inputs": [{"timestamps": [1, 2], "values": [2.0, 4.0]}],
+                "horizon": 3,
+            },
+        )
+        assert r.status_code == 200
+        assert r.json()["outputs"][0]["forecast"] == [3.0, 3.0, 3.0]
+
+    def test_unsupported_model(self):
+        repo = ModelRepository()
+
+        class Plain(Model):
+            def __init__(self):
+                super().__init__("p")
+                self.ready = True
+
+        repo.update(Plain())
+        client = TestClient(create_app(DataPlane(repo)))
+        r = client.post(
+            "/timeseries/v1/forecast",
+            json={"model": "p", "inputs": [{"values": [1]}], "horizon": 1},
+        )
+        assert r.status_code == 400
+
+
+class TestMultiprocessREST:
+    def test_workers_share_socket(self):
+        import requests
+
+        from kserve_amd.protocol.rest.multiprocess import RESTServerMultiProcess
+
+        def app_factory():
+            repo = ModelRepository()
+
+            class Echo(Model):
+                def __init__(self):
+                    super().__init__("e")
+                    self.ready = True
+
+                def predict(self, payload, headers=None):
+                    return {"predictions": payload["instances"], "pid": os.getpid()}
+
+            repo.update(Echo())
+            return create_app(DataPlane(repo))
+
+        srv = RESTServerMultiProcess(app_factory, http_port=0, workers=2)
+        srv.start()
+        try:
+            import time
+
+            deadline = time.time() + 15
+            ok = False
+            while time.time() < deadline:
+                try:
+                    r = requests.post(
+                        f"http://127.0.0.1:{srv.port}/v1/models/e:predict",
+                        json={"instances": [1, 2]},
+                        timeout=2,
+                    )
+                    if r.status_code == 200:
+                        ok = True
+                        break
+                except Exception:
+                    time.sleep(0.3)
+            assert ok
+            assert r.json()["predictions"] == [1, 2]
+        finally:
+            srv.stop()
+
+
+class TestLocalModelCache:
+    def test_cache_and_evict(self, tmp_path):
+        from kserve_amd.controlplane.localmodel import (
+            LocalModelCacheSpec,
+            LocalModelNodeAgent,
+            mount_for_isvc,
+        )
+
+        src = tmp_path / "src"
+        src.mkdir()
+        (src / "weights.bin").write_bytes(b"w" * 128)
+        agent = LocalModelNodeAgent(str(tmp_path / "cache"), node_group="gpu")
+        spec = LocalModelCacheSpec(
+            name="llama", source_model_uri=str(src), node_groups=["gpu"]
+        )
+        agent.apply(spec)
+        status = run(agent.reconcile_once())
+        assert status["llama"].state == "Ready"
+        assert agent.is_cached("llama")
+        assert mount_for_isvc(agent, str(src)) == agent.model_path("llama")
+        # other node group: not applied
+        agent2 = LocalModelNodeAgent(str(tmp_path / "cache2"), node_group="cpu")
+        agent2.apply(spec)
+        assert "llama" not in agent2.desired
+        # eviction
+        agent.delete("llama")
+        run(agent.reconcile_once())
+        assert not agent.is_cached("llama")
+
+
+class TestOpenAIEmbeddingsAdapter:
+    def test_adapter_shape(self):
+        from kserve_amd.protocol.rest.openai.types import EmbeddingRequest
+        from kserve_amd.runtimes.encoder_model import OpenAIEmbeddingAdapter
+
+        class FakeEncoder:
+            name = "emb"
+
+            def preprocess(self, payload, headers=None):
+                return {"texts": payload["instances"], "_v2": None}
+
+            def predict(self, payload, headers=None):
+                return {
+                    "output": None,
+                    "ids_list": [[1, 2, 3]] * len(payload["texts"]),
+                    "cu": None,
+                    "_v2": None,
+                }
+
+            def postprocess(self, result, headers=None):
+                return {"predictions": [[0.1, 0.2]] * len(result["ids_list"])}
+
+        adapter = OpenAIEmbeddingAdapter(FakeEncoder())
+        req = EmbeddingRequest(model="emb", input=["a", "b"])
+        out = run(adapter.create_embedding(req))
+        assert len(out.data) == 2
+        assert out.data[0].embedding == [0.1, 0.2]
+        assert out.usage.prompt_tokens == 6
