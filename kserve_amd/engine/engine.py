@@ -207,6 +207,30 @@ class LLMEngine:
         finished: List[Request] = []
         if sample_reqs:
             tokens = self.sampler.sample(logits, sample_reqs)
+            # top-N logprobs for the (rare) requests that ask for them
+            lp_idx = [
+                i for i, r in enumerate(sample_reqs)
+                if r.sampling_params.logprobs is not None
+            ]
+            if lp_idx:
+                n_top = max(
+                    sample_reqs[i].sampling_params.logprobs or 1 for i in lp_idx
+                )
+                n_top = max(n_top, 1)
+                sub = torch.log_softmax(logits[lp_idx].float(), dim=-1)
+                topv, topi = sub.topk(n_top, dim=-1)
+                topv = topv.cpu()
+                topi = topi.cpu()
+                for row, i in enumerate(lp_idx):
+                    req = sample_reqs[i]
+                    entry = {
+                        int(topi[row][c]): float(topv[row][c])
+                        for c in range(n_top)
+                    }
+                    tok = int(tokens[i])
+                    if tok not in entry:
+                        entry[tok] = float(sub[row][tok])
+                    req.logprobs.append(entry)
             LLM_GENERATION_TOKENS.inc(len(tokens))
             now = time.monotonic()
             for req, tok in zip(sample_reqs, tokens):
@@ -238,6 +262,7 @@ class LLMEngine:
                         num_prompt_tokens=req.num_prompt_tokens,
                         text_delta=delta,
                         output_text=req.output_text,
+                        logprobs=req.logprobs if req.sampling_params.logprobs is not None else None,
                     )
                 )
         self.scheduler.finish_requests(finished)

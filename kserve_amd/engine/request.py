@@ -68,6 +68,8 @@ class Request:
         self.prefix_offset = 0
         self.read_offset = 0
         self.output_text = ""
+        # per-step {token_id: logprob} maps when sampling_params.logprobs
+        self.logprobs: List[dict] = []
 
     # -- lengths -----------------------------------------------------------
     @property

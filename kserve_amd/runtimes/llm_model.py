@@ -26,6 +26,7 @@ from kserve_amd.protocol.rest.openai.types import (
     ChatCompletionRequest,
     Completion,
     CompletionChoice,
+    CompletionLogprobs,
     CompletionRequest,
     UsageInfo,
 )
@@ -37,8 +38,12 @@ def _to_sampling_params(
 ) -> SamplingParams:
     max_tokens = getattr(req, "max_completion_tokens", None) or req.max_tokens
     stop = req.stop if isinstance(req.stop, list) else ([req.stop] if req.stop else [])
+    logprobs = getattr(req, "logprobs", None)
+    if isinstance(logprobs, bool):
+        logprobs = 1 if logprobs else None
     return SamplingParams(
         n=req.n or 1,
+        logprobs=logprobs,
         temperature=req.temperature if req.temperature is not None else 1.0,
         top_p=req.top_p if req.top_p is not None else 1.0,
         top_k=req.top_k if req.top_k is not None else -1,
@@ -149,10 +154,30 @@ class LLMModel(OpenAIModel):
                     if getattr(out, "output_text", None)
                     else self._decode(out.output_token_ids)
                 )
+                if request.echo:
+                    text = self._decode(ids) + text
+                lp = None
+                if out.logprobs:
+                    toks = [
+                        self._decode([t]) for t in out.output_token_ids
+                    ]
+                    lp = CompletionLogprobs(
+                        tokens=toks,
+                        token_logprobs=[
+                            step.get(t)
+                            for t, step in zip(out.output_token_ids, out.logprobs)
+                        ],
+                        top_logprobs=[
+                            {self._decode([k]): v for k, v in step.items()}
+                            for step in out.logprobs
+                        ],
+                        text_offset=[],
+                    )
                 choices.append(
                     CompletionChoice(
                         index=i * sp.n + j,
                         text=text,
+                        logprobs=lp,
                         finish_reason=out.finish_reason or "stop",
                     )
                 )

@@ -133,3 +133,42 @@ def test_v1_alias(client):
         json={"model": "tiny", "prompt": [1, 2], "max_tokens": 2, "temperature": 0.0},
     )
     assert r.status_code == 200
+
+
+def test_completion_logprobs_and_echo(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [1, 2, 3],
+            "max_tokens": 3,
+            "temperature": 0.0,
+            "logprobs": 2,
+            "echo": True,
+        },
+    )
+    assert r.status_code == 200, r.text
+    choice = r.json()["choices"][0]
+    lp = choice["logprobs"]
+    assert lp is not None
+    assert len(lp["token_logprobs"]) == 3
+    assert all(v is not None and v <= 0 for v in lp["token_logprobs"])
+    assert len(lp["top_logprobs"][0]) >= 2
+    # echo: prompt token ids prefixed (no tokenizer -> id text)
+    assert choice["text"].startswith("1 2 3")
+
+
+def test_completion_n_choices(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [4, 5],
+            "max_tokens": 2,
+            "n": 3,
+            "temperature": 0.8,
+            "seed": 1,
+        },
+    )
+    assert r.status_code == 200, r.text
+    assert len(r.json()["choices"]) == 3
