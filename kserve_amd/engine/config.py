@@ -156,6 +156,10 @@ class SchedulerConfig:
     max_model_len: int = 8192
     # chunked prefill: cap on prompt tokens scheduled per step
     enable_chunked_prefill: bool = False
+    # multi-step decode: consecutive greedy decode iterations run as pure
+    # hipGraph replays (sampled token fed back on-GPU); stop conditions are
+    # applied after the window. 1 disables.
+    multi_step: int = 8
 
 
 @dataclass

@@ -184,6 +184,18 @@ class LLMEngine:
         if batch.is_prefill:
             logits = self.runner.execute_prefill(batch, self.scheduler.block_manager)
         else:
+            # multi-step window: all-greedy decode with no pending scheduling
+            # events runs as back-to-back hipGraph replays
+            k = 1
+            if self.config.scheduler.multi_step > 1 and all(
+                r.sampling_params.greedy and r.sampling_params.logprobs is None
+                for r in batch.requests
+            ):
+                k = self.scheduler.reserve_decode_window(
+                    batch, self.config.scheduler.multi_step
+                )
+            if k > 1:
+                return self._run_decode_window(batch, k)
             logits = self.runner.execute_decode(batch, self.scheduler.block_manager)
         # advance computed-token counters for executed tokens
         for req, n in zip(batch.requests, batch.num_scheduled_tokens):
@@ -265,6 +277,54 @@ class LLMEngine:
                         logprobs=req.logprobs if req.sampling_params.logprobs is not None else None,
                     )
                 )
+        self.scheduler.finish_requests(finished)
+        for req in finished:
+            self.runner.release_request(req.request_id)
+        LLM_KV_USAGE.set(self.scheduler.block_manager.usage)
+        return outputs
+
+    def _run_decode_window(self, batch, k: int) -> List[RequestOutput]:
+        tokens_k = self.runner.multi_step_decode(
+            batch, self.scheduler.block_manager, k
+        )
+        outputs: List[RequestOutput] = []
+        finished: List[Request] = []
+        now = time.monotonic()
+        max_len = self.config.scheduler.max_model_len
+        for step in range(k):
+            row = tokens_k[step]
+            for i, req in enumerate(batch.requests):
+                if req.is_finished:
+                    continue  # stopped mid-window; later tokens discarded
+                req.num_computed_tokens += 1
+                req.append_output_token(int(row[i]))
+                req.maybe_finish(max_len)
+                delta = self.detokenizer.decode_new(req)
+                if not req.is_finished and req.sampling_params.stop:
+                    if self.detokenizer.check_stop_strings(req) is not None:
+                        req.status = RequestStatus.FINISHED_STOPPED
+                        req.is_finished = True
+                        req.finish_time = now
+                if req.is_finished:
+                    finished.append(req)
+                    LLM_E2E_HIST.observe(now - req.arrival_time)
+                outputs.append(
+                    RequestOutput(
+                        request_id=req.request_id,
+                        new_token_ids=[int(row[i])],
+                        finished=req.is_finished,
+                        finish_reason=req.finish_reason,
+                        output_token_ids=(
+                            list(req.output_token_ids)
+                            if req.is_finished
+                            else req.output_token_ids
+                        ),
+                        num_prompt_tokens=req.num_prompt_tokens,
+                        text_delta=delta,
+                        output_text=req.output_text,
+                    )
+                )
+        LLM_GENERATION_TOKENS.inc(len(outputs))
         self.scheduler.finish_requests(finished)
         for req in finished:
             self.runner.release_request(req.request_id)

@@ -331,42 +331,56 @@ class ModelRunner:
         self._graph_buffers = {
             "input_ids": torch.zeros(max_bs, dtype=torch.int64, device=dev),
             "positions": torch.zeros(max_bs, dtype=torch.int64, device=dev),
-            "slot_mapping": torch.zeros(max_bs, dtype=torch.int32, device=dev),
             "block_tables": torch.zeros(
                 (max_bs, mb), dtype=torch.int32, device=dev
             ),
             "context_lens": torch.ones(max_bs, dtype=torch.int32, device=dev),
             "logits": {},
+            "sampled": {},
         }
         torch.cuda.synchronize()
         pool = None
+        from kserve_amd import ops as _ops
+
+        bsz = self.config.cache.block_size
         for bs in sorted(sizes, reverse=True):
             buf = self._graph_buffers
-            meta = AttentionMetadata(
-                is_prefill=False,
-                slot_mapping=buf["slot_mapping"][:bs],
-                block_tables=buf["block_tables"][:bs],
-                context_lens=buf["context_lens"][:bs],
-            )
+
+            def body(bs=bs, buf=buf):
+                # slot mapping derived IN-GRAPH from (block_tables, positions)
+                # so multi-step windows advance with zero host involvement
+                pos = buf["positions"][:bs]
+                bt = buf["block_tables"][:bs]
+                blk = torch.gather(bt, 1, (pos // bsz).unsqueeze(1)).squeeze(1)
+                slot = (blk * bsz + (pos % bsz).to(torch.int32)).to(torch.int32)
+                meta = AttentionMetadata(
+                    is_prefill=False,
+                    slot_mapping=slot,
+                    block_tables=bt,
+                    context_lens=buf["context_lens"][:bs],
+                )
+                hidden = self.model(
+                    buf["input_ids"][:bs], pos, self.kv_caches, meta
+                )
+                logits = self.model.compute_logits(hidden)
+                # greedy sample in-graph; feed back as the next step's input
+                sampled = _ops.greedy_sample(logits)
+                buf["input_ids"][:bs].copy_(sampled)
+                buf["positions"][:bs].add_(1)
+                buf["context_lens"][:bs].add_(1)
+                return logits, sampled
+
             # warmup run (allocator state, rccl comms)
-            hidden = self.model(
-                buf["input_ids"][:bs], buf["positions"][:bs], self.kv_caches, meta
-            )
-            logits = self.model.compute_logits(hidden)
+            body()
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, pool=pool):
-                hidden = self.model(
-                    buf["input_ids"][:bs],
-                    buf["positions"][:bs],
-                    self.kv_caches,
-                    meta,
-                )
-                logits = self.model.compute_logits(hidden)
+                logits, sampled = body()
             if pool is None:
                 pool = g.pool()
             self._graphs[bs] = g
             buf["logits"][bs] = logits
+            buf["sampled"][bs] = sampled
         self._graph_batch_sizes = sorted(self._graphs.keys())
         torch.cuda.synchronize()
         logger.info("Captured decode hipGraphs for batch sizes %s", self._graph_batch_sizes)
@@ -385,16 +399,13 @@ class ModelRunner:
             # padded rows: write scratch KV to block 0 (reserved), ctx len 1
             pnp["input_ids"][n:bucket] = 0
             pnp["positions"][n:bucket] = 0
-            pnp["slot_mapping"][n:bucket] = 0
             pnp["context_lens"][n:bucket] = 1
             pnp["block_tables"][n:bucket, :nb] = 0
         buf = self._graph_buffers
         # single async H2D per tensor, pinned -> static graph buffers
+        # (slot mapping is derived in-graph from block_tables + positions)
         buf["input_ids"][:bucket].copy_(pin["input_ids"][:bucket], non_blocking=True)
         buf["positions"][:bucket].copy_(pin["positions"][:bucket], non_blocking=True)
-        buf["slot_mapping"][:bucket].copy_(
-            pin["slot_mapping"][:bucket], non_blocking=True
-        )
         buf["context_lens"][:bucket].copy_(
             pin["context_lens"][:bucket], non_blocking=True
         )
@@ -403,3 +414,71 @@ class ModelRunner:
         )
         self._graphs[bucket].replay()
         return buf["logits"][bucket][:n]
+
+    def _stage_graph_inputs(self, bucket: int, batch, block_manager) -> int:
+        n, nb = self._fill_pinned(batch, block_manager)
+        pin = self._pin
+        pnp = self._pin_np
+        if n < bucket:
+            pnp["input_ids"][n:bucket] = 0
+            pnp["positions"][n:bucket] = 0
+            pnp["context_lens"][n:bucket] = 1
+            pnp["block_tables"][n:bucket, :nb] = 0
+        buf = self._graph_buffers
+        buf["input_ids"][:bucket].copy_(pin["input_ids"][:bucket], non_blocking=True)
+        buf["positions"][:bucket].copy_(pin["positions"][:bucket], non_blocking=True)
+        buf["context_lens"][:bucket].copy_(
+            pin["context_lens"][:bucket], non_blocking=True
+        )
+        buf["block_tables"][:bucket, :nb].copy_(
+            pin["block_tables"][:bucket, :nb], non_blocking=True
+        )
+        return n
+
+    @torch.no_grad()
+    def multi_step_decode(self, batch, block_manager, k_steps: int) -> torch.Tensor:
+        """Run ``k_steps`` greedy decode iterations with the sampled token
+        fed back on-GPU (graph path: pure replays, zero host work in the
+        loop). Caller must have reserved KV capacity for the whole window.
+        Returns host int64 [k_steps, n]."""
+        n = len(batch.requests)
+        bucket = None
+        if self.is_cuda and self._graphs and not self.config.enforce_eager:
+            bucket = self._graph_bucket(n)
+        if bucket is not None:
+            self._stage_graph_inputs(bucket, batch, block_manager)
+            sampled = self._graph_buffers["sampled"][bucket]
+            out_host = torch.empty(
+                (k_steps, n), dtype=torch.int64, pin_memory=True
+            )
+            g = self._graphs[bucket]
+            for k in range(k_steps):
+                g.replay()
+                out_host[k].copy_(sampled[:n], non_blocking=True)
+            torch.cuda.synchronize()
+            return out_host
+        # eager fallback (CPU tests / enforce_eager): same semantics
+        from kserve_amd import ops as _ops
+
+        input_ids, positions, meta = self.prepare_decode(batch, block_manager)
+        bsz = self.config.cache.block_size
+        bt = meta.block_tables
+        ctx = meta.context_lens
+        out = torch.empty((k_steps, n), dtype=torch.int64)
+        for k in range(k_steps):
+            blk = torch.gather(bt, 1, (positions // bsz).unsqueeze(1)).squeeze(1)
+            slot = (blk * bsz + (positions % bsz).to(torch.int32)).to(torch.int32)
+            meta_k = AttentionMetadata(
+                is_prefill=False,
+                slot_mapping=slot,
+                block_tables=bt,
+                context_lens=ctx,
+            )
+            hidden = self.model(input_ids, positions, self.kv_caches, meta_k)
+            logits = self.model.compute_logits(hidden)
+            sampled = _ops.greedy_sample(logits)
+            out[k] = sampled.cpu()
+            input_ids = sampled.to(input_ids.device)
+            positions = positions + 1
+            ctx = ctx + 1
+        return out

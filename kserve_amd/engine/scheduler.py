@@ -216,3 +216,44 @@ class Scheduler:
                 self.running.remove(req)
             self.block_manager.free(req)
             self.block_manager.free_cpu(req)
+
+    # -- multi-step decode windows -------------------------------------------
+    def reserve_decode_window(self, batch: ScheduledBatch, max_k: int) -> int:
+        """Largest k such that all scheduled requests can decode k steps with
+        no scheduling events (no prefill waiting, no swaps, no preemption,
+        no per-request finish before step k except stop-token/EOS which is
+        handled post-hoc). Reserves the KV blocks for the window."""
+        if max_k <= 1 or self.waiting or self.swapped or batch.preempted:
+            return 1
+        reqs = batch.requests
+        if not reqs:
+            return 1
+        k = max_k
+        for r in reqs:
+            sp = r.sampling_params
+            if sp.max_tokens is not None:
+                k = min(k, sp.max_tokens - r.num_output_tokens)
+            k = min(k, self.config.max_model_len - r.num_tokens)
+        if k <= 1:
+            return 1
+        bm = self.block_manager
+        bs = bm.block_size
+        while k > 1:
+            needed = 0
+            for r in reqs:
+                have = len(bm.get_block_table(r)) * bs
+                want = r.num_tokens + k - 1  # max position + 1 in the window
+                if want > have:
+                    needed += (want - have + bs - 1) // bs
+            if needed <= bm.num_free_blocks - bm.watermark_blocks:
+                break
+            k -= max(1, k // 2)
+        if k <= 1:
+            return 1
+        for r in reqs:
+            want = r.num_tokens + k - 1
+            table = bm._tables[r.request_id]
+            while len(table) * bs < want:
+                table.append(bm._free.pop())
+            r.block_table = table
+        return k
