@@ -77,7 +77,10 @@ def main():
             max_position_embeddings=2048, model_name="tiny",
         )
 
-    max_model_len = args.prompt_len + args.steps + args.warmup + 64
+    # each timed "step" is one scheduling iteration = up to `multi_step`
+    # decode iterations (hipGraph window); budget the KV/model length for it
+    window = 8
+    max_model_len = args.prompt_len + (args.steps + args.warmup) * window + 64
     cfg = EngineConfig(
         model=mcfg,
         cache=CacheConfig(
