@@ -143,7 +143,7 @@ def main():
         prof.enable()
     for _ in range(args.steps):
         outs = engine.step()
-        tokens += len(outs)
+        tokens += sum(len(o.new_token_ids) for o in outs)
     if prof is not None:
         prof.disable()
     comm.barrier()

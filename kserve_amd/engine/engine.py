@@ -287,44 +287,60 @@ class LLMEngine:
         tokens_k = self.runner.multi_step_decode(
             batch, self.scheduler.block_manager, k
         )
+        # ONE RequestOutput per request per window (streaming consumers get
+        # k-token chunks); per-token python only for stop conditions
+        cols = tokens_k.t().tolist()  # [n][k]
         outputs: List[RequestOutput] = []
         finished: List[Request] = []
         now = time.monotonic()
         max_len = self.config.scheduler.max_model_len
-        for step in range(k):
-            row = tokens_k[step]
-            for i, req in enumerate(batch.requests):
-                if req.is_finished:
-                    continue  # stopped mid-window; later tokens discarded
-                req.num_computed_tokens += 1
-                req.append_output_token(int(row[i]))
-                req.maybe_finish(max_len)
-                delta = self.detokenizer.decode_new(req)
-                if not req.is_finished and req.sampling_params.stop:
-                    if self.detokenizer.check_stop_strings(req) is not None:
-                        req.status = RequestStatus.FINISHED_STOPPED
-                        req.is_finished = True
-                        req.finish_time = now
-                if req.is_finished:
-                    finished.append(req)
-                    LLM_E2E_HIST.observe(now - req.arrival_time)
-                outputs.append(
-                    RequestOutput(
-                        request_id=req.request_id,
-                        new_token_ids=[int(row[i])],
-                        finished=req.is_finished,
-                        finish_reason=req.finish_reason,
-                        output_token_ids=(
-                            list(req.output_token_ids)
-                            if req.is_finished
-                            else req.output_token_ids
-                        ),
-                        num_prompt_tokens=req.num_prompt_tokens,
-                        text_delta=delta,
-                        output_text=req.output_text,
-                    )
+        n_tokens = 0
+        for i, req in enumerate(batch.requests):
+            col = cols[i]
+            sp = req.sampling_params
+            if req.first_token_time is None:
+                req.first_token_time = now
+                LLM_TTFT_HIST.observe(now - req.arrival_time)
+            stop_at = k
+            if not sp.ignore_eos or sp.stop_token_ids:
+                stops = set(sp.stop_token_ids)
+                if not sp.ignore_eos and req.eos_token_id is not None:
+                    stops.add(req.eos_token_id)
+                for j, t in enumerate(col):
+                    if t in stops and j + 1 >= sp.min_tokens - req.num_output_tokens:
+                        stop_at = j + 1
+                        break
+            col = col[:stop_at]
+            req.output_token_ids.extend(col)
+            req.num_computed_tokens += stop_at
+            req.maybe_finish(max_len)
+            delta = self.detokenizer.decode_new(req)
+            if not req.is_finished and sp.stop:
+                if self.detokenizer.check_stop_strings(req) is not None:
+                    req.status = RequestStatus.FINISHED_STOPPED
+                    req.is_finished = True
+                    req.finish_time = now
+            if req.is_finished:
+                finished.append(req)
+                LLM_E2E_HIST.observe(now - req.arrival_time)
+            n_tokens += len(col)
+            outputs.append(
+                RequestOutput(
+                    request_id=req.request_id,
+                    new_token_ids=col,
+                    finished=req.is_finished,
+                    finish_reason=req.finish_reason,
+                    output_token_ids=(
+                        list(req.output_token_ids)
+                        if req.is_finished
+                        else req.output_token_ids
+                    ),
+                    num_prompt_tokens=req.num_prompt_tokens,
+                    text_delta=delta,
+                    output_text=req.output_text,
                 )
-        LLM_GENERATION_TOKENS.inc(len(outputs))
+            )
+        LLM_GENERATION_TOKENS.inc(n_tokens)
         self.scheduler.finish_requests(finished)
         for req in finished:
             self.runner.release_request(req.request_id)

@@ -44,6 +44,7 @@ class ModelRunner:
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._graph_buffers: Optional[Dict[str, torch.Tensor]] = None
         self._graph_batch_sizes: List[int] = []
+        self._window_host: Optional[torch.Tensor] = None
         # pinned staging + per-request numpy block-table cache
         self._pin: Optional[Dict[str, torch.Tensor]] = None
         self._pin_np = None
@@ -448,9 +449,17 @@ class ModelRunner:
         if bucket is not None:
             self._stage_graph_inputs(bucket, batch, block_manager)
             sampled = self._graph_buffers["sampled"][bucket]
-            out_host = torch.empty(
-                (k_steps, n), dtype=torch.int64, pin_memory=True
-            )
+            if (
+                self._window_host is None
+                or self._window_host.shape[0] < k_steps
+                or self._window_host.shape[1] < n
+            ):
+                self._window_host = torch.empty(
+                    (max(k_steps, 16), self.config.scheduler.max_num_seqs),
+                    dtype=torch.int64,
+                    pin_memory=True,
+                )
+            out_host = self._window_host[:k_steps, :n]
             g = self._graphs[bucket]
             for k in range(k_steps):
                 g.replay()

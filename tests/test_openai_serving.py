@@ -93,7 +93,11 @@ def test_completion_stream(client):
                 events.append(line[len("data: ") :])
     assert events[-1] == "[DONE]"
     chunks = [json.loads(e) for e in events[:-1]]
-    assert len(chunks) == 4
+    # multi-step windows may deliver several tokens per SSE chunk
+    total_tokens = sum(
+        len(c["choices"][0]["text"].split()) for c in chunks
+    )
+    assert total_tokens == 4
     assert chunks[-1]["choices"][0]["finish_reason"] == "length"
 
 
