@@ -202,7 +202,9 @@ class LLMModel(OpenAIModel):
         ids = self._encode_prompt(prompt)
         async for out in self.async_engine.generate(ids, sp):
             text = out.text_delta or (
-                "" if self.tokenizer is not None else f"{out.new_token_ids[0]} "
+                ""
+                if self.tokenizer is not None
+                else "".join(f"{t} " for t in out.new_token_ids)
             )
             yield Completion(
                 model=self.name,
@@ -249,7 +251,11 @@ class LLMModel(OpenAIModel):
             delta = ChatCompletionChunkDelta(
                 role="assistant" if first else None,
                 content=out.text_delta
-                or ("" if self.tokenizer is not None else f"{out.new_token_ids[0]} "),
+                or (
+                    ""
+                    if self.tokenizer is not None
+                    else "".join(f"{t} " for t in out.new_token_ids)
+                ),
             )
             first = False
             yield ChatCompletionChunk(
