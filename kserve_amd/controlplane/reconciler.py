@@ -240,6 +240,11 @@ def reconcile(
     predictor.go:918)."""
     default_inference_service(isvc)
     validate_inference_service(isvc)
+    if isvc.deployment_mode == "Serverless":
+        return {
+            "knative_service": render_knative_service(isvc, runtimes),
+            "httproute": render_http_route(isvc),
+        }
     out: Dict[str, object] = {
         "deployment": render_deployment(isvc, runtimes),
         "service": render_service(isvc),
@@ -271,3 +276,46 @@ def build_model_config(trained_models: List[Dict]) -> str:
             }
         )
     return json.dumps(entries, sort_keys=True)
+
+
+def render_knative_service(isvc: InferenceService, runtimes: List[ServingRuntime]) -> Dict:
+    """Knative Service manifest (Serverless mode; reference
+    ksvc_reconciler.go:159 with canaryTrafficPercent traffic split)."""
+    p = isvc.spec.predictor
+    pod = render_predictor_pod_spec(isvc, runtimes)
+    ann = {
+        "autoscaling.knative.dev/min-scale": str(p.min_replicas),
+    }
+    if p.max_replicas:
+        ann["autoscaling.knative.dev/max-scale"] = str(p.max_replicas)
+    if p.scale_target:
+        ann["autoscaling.knative.dev/target"] = str(p.scale_target)
+    traffic = [{"latestRevision": True, "percent": 100}]
+    pct = p.canary_traffic_percent
+    if pct is not None and 0 <= pct < 100:
+        traffic = [
+            {"latestRevision": True, "percent": pct},
+            {
+                "latestRevision": False,
+                "revisionName": f"{predictor_service_name(isvc)}-prev",
+                "percent": 100 - pct,
+            },
+        ]
+    return {
+        "apiVersion": "serving.knative.dev/v1",
+        "kind": "Service",
+        "metadata": {
+            "name": predictor_service_name(isvc),
+            "namespace": isvc.namespace,
+        },
+        "spec": {
+            "template": {
+                "metadata": {
+                    "labels": pod["metadata"]["labels"],
+                    "annotations": {**pod["metadata"]["annotations"], **ann},
+                },
+                "spec": pod["spec"],
+            },
+            "traffic": traffic,
+        },
+    }

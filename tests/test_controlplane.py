@@ -312,3 +312,20 @@ class TestLLMInferenceService:
         }
         assert env["OTEL_EXPORTER_OTLP_ENDPOINT"] == "http://otel:4317"
         assert env["OTEL_TRACES_SAMPLER_ARG"] == "0.05"
+
+
+class TestKnativeMode:
+    def test_serverless_rendering(self):
+        isvc = make_isvc(min_replicas=1, max_replicas=4, canary_traffic_percent=10)
+        isvc.annotations["serving.kserve.io/deploymentMode"] = "Serverless"
+        from kserve_amd.controlplane.reconciler import reconcile
+        from kserve_amd.controlplane.servingruntime import default_cluster_runtimes
+
+        out = reconcile(isvc, default_cluster_runtimes())
+        ksvc = out["knative_service"]
+        assert ksvc["kind"] == "Service"
+        ann = ksvc["spec"]["template"]["metadata"]["annotations"]
+        assert ann["autoscaling.knative.dev/max-scale"] == "4"
+        traffic = ksvc["spec"]["traffic"]
+        assert traffic[0]["percent"] == 10 and traffic[1]["percent"] == 90
+        assert "deployment" not in out

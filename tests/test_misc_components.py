@@ -169,3 +169,29 @@ class TestOpenAIEmbeddingsAdapter:
         assert len(out.data) == 2
         assert out.data[0].embedding == [0.1, 0.2]
         assert out.usage.prompt_tokens == 6
+
+
+class TestExplainer:
+    def test_occlusion_explainer_end_to_end(self):
+        import httpx
+
+        from kserve_amd.model import PredictorConfig
+        from kserve_amd.runtimes.explainer import ExplainerModel
+
+        # fake predictor: prediction = 3*x0 + 1*x1 (feature 0 dominates)
+        def predictor(request: httpx.Request) -> httpx.Response:
+            body = json.loads(request.content)
+            preds = [3 * r[0] + r[1] for r in body["instances"]]
+            return httpx.Response(200, json={"predictions": preds})
+
+        model = ExplainerModel(
+            "exp", PredictorConfig(predictor_host="pred:80")
+        )
+        model._http_client = httpx.AsyncClient(
+            transport=httpx.MockTransport(predictor)
+        )
+        out = run(model.explain({"instances": [[2.0, 5.0]]}))
+        imps = out["explanations"]["importances"][0]
+        assert imps[0] > imps[1]  # x0 matters more
+        assert imps[0] == pytest.approx(6.0)
+        assert imps[1] == pytest.approx(5.0)
