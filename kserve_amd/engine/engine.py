@@ -43,8 +43,46 @@ class Sampler:
             self.generator.manual_seed(seed)
         self._step = 0
 
+    @staticmethod
+    def _apply_penalties(logits: torch.Tensor, requests: List[Request]) -> torch.Tensor:
+        """Presence/frequency/repetition penalties (OpenAI semantics) for the
+        requests that use them; no-op rows untouched."""
+        rows = [
+            i
+            for i, r in enumerate(requests)
+            if r.sampling_params.presence_penalty != 0.0
+            or r.sampling_params.frequency_penalty != 0.0
+            or r.sampling_params.repetition_penalty != 1.0
+        ]
+        if not rows:
+            return logits
+        logits = logits.clone()
+        for i in rows:
+            r = requests[i]
+            sp = r.sampling_params
+            seen: dict = {}
+            for t in r.output_token_ids:
+                seen[t] = seen.get(t, 0) + 1
+            if not seen:
+                continue
+            idx = torch.tensor(list(seen.keys()), device=logits.device)
+            cnt = torch.tensor(
+                list(seen.values()), device=logits.device, dtype=logits.dtype
+            )
+            row = logits[i].float()
+            vals = row[idx]
+            if sp.repetition_penalty != 1.0:
+                vals = torch.where(
+                    vals > 0, vals / sp.repetition_penalty, vals * sp.repetition_penalty
+                )
+            vals = vals - sp.presence_penalty - sp.frequency_penalty * cnt.float()
+            row[idx] = vals
+            logits[i] = row.to(logits.dtype)
+        return logits
+
     def sample(self, logits: torch.Tensor, requests: List[Request]) -> List[int]:
         self._step += 1
+        logits = self._apply_penalties(logits, requests)
         all_greedy = all(r.sampling_params.greedy for r in requests)
         if all_greedy:
             return ops.greedy_sample(logits).tolist()
@@ -188,7 +226,11 @@ class LLMEngine:
             # events runs as back-to-back hipGraph replays
             k = 1
             if self.config.scheduler.multi_step > 1 and all(
-                r.sampling_params.greedy and r.sampling_params.logprobs is None
+                r.sampling_params.greedy
+                and r.sampling_params.logprobs is None
+                and r.sampling_params.presence_penalty == 0.0
+                and r.sampling_params.frequency_penalty == 0.0
+                and r.sampling_params.repetition_penalty == 1.0
                 for r in batch.requests
             ):
                 k = self.scheduler.reserve_decode_window(

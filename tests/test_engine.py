@@ -119,3 +119,24 @@ def test_preemption_recovers():
     big_outs = [o.output_token_ids for o in big.values()]
     small_outs = [o.output_token_ids for o in small.values()]
     assert big_outs == small_outs
+
+
+def test_repetition_penalty_changes_output(engine):
+    """A strong repetition penalty must break greedy loops."""
+    base = SamplingParams(temperature=0.0, max_tokens=12)
+    out_base = list(engine.generate([[40, 41]], base).values())[0].output_token_ids
+    pen = SamplingParams(
+        temperature=0.0, max_tokens=12, repetition_penalty=5.0
+    )
+    out_pen = list(engine.generate([[40, 41]], pen).values())[0].output_token_ids
+    # with random weights greedy usually repeats; penalty must diverge and
+    # produce fewer repeats
+    def max_run(toks):
+        best = run_len = 1
+        for a, b in zip(toks, toks[1:]):
+            run_len = run_len + 1 if a == b else 1
+            best = max(best, run_len)
+        return best
+
+    assert out_pen != out_base or max_run(out_pen) <= max_run(out_base)
+    assert len(set(out_pen)) >= len(set(out_base))
