@@ -253,6 +253,148 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// 2-page unrolled variant of the HPW=1 fast path: both pages' K+V bursts
+// (16 KB/wave) are issued before either page's compute, doubling the
+// bytes-in-flight per wave of the latency-bound loop.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(256) void paged_attention_u2_kernel(
+    short* __restrict__ out, const short* __restrict__ q,
+    const short* __restrict__ k_cache, const short* __restrict__ v_cache,
+    const int* __restrict__ block_tables, const int* __restrict__ context_lens,
+    const float scale, const int num_kv_heads, const int group,
+    const int max_blocks, const long q_row_stride, const int n_splits,
+    float* __restrict__ part_out, float* __restrict__ part_ml) {
+  constexpr int ACC = D / 64;
+  constexpr int QFRAG = D / 4;
+  const int kv_head = blockIdx.x;
+  const int seq = blockIdx.y;
+  const int split = blockIdx.z;
+  const int ctx = context_lens[seq];
+  if (ctx <= 0) return;
+  const int num_heads = num_kv_heads * group;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tok = lane >> 2;
+  const int part = lane & 3;
+
+  const int nblocks = (ctx + PAGE - 1) / PAGE;
+  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
+  const int blk_lo = split * blocks_per_split;
+  const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
+  if (blk_lo >= blk_hi) {
+    if (n_splits > 1 && threadIdx.x < (unsigned)group) {
+      const int h = kv_head * group + threadIdx.x;
+      float* ml = part_ml + (((long)seq * num_heads + h) * n_splits + split) * 2;
+      ml[0] = NEG_INF;
+      ml[1] = 0.f;
+    }
+    return;
+  }
+  const int head = kv_head * group + wave;
+  if (wave >= group) return;
+
+  float q_frag[QFRAG];
+  {
+    const short* qp =
+        q + (long)seq * q_row_stride + (long)head * D + part * QFRAG;
+#pragma unroll
+    for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
+  }
+  float m = NEG_INF, l = 0.f;
+  float acc[ACC];
+#pragma unroll
+  for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
+
+  const int bt_base = (int)((long)seq * max_blocks);
+
+  auto load_page = [&](int bi, short8_t* kreg, unsigned int* vreg) {
+    const int block_id = block_tables[bt_base + bi];
+    const long pbase = (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    const short8_t* kp =
+        reinterpret_cast<const short8_t*>(k_cache + pbase + tok * D + part * QFRAG);
+#pragma unroll
+    for (int c = 0; c < QFRAG / 8; ++c) kreg[c] = kp[c];
+    const short* vpage = v_cache + pbase;
+#pragma unroll
+    for (int t = 0; t < PAGE; ++t) {
+      if constexpr (ACC == 2) {
+        vreg[t] =
+            *reinterpret_cast<const unsigned int*>(vpage + t * D + lane * ACC);
+      } else {
+        vreg[t] = (unsigned short)*(vpage + t * D + lane);
+      }
+    }
+  };
+
+  auto compute_page = [&](int bi, const short8_t* kreg,
+                          const unsigned int* vreg) {
+    float s = 0.f;
+#pragma unroll
+    for (int c = 0; c < QFRAG / 8; ++c) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        s += q_frag[c * 8 + j] * bf16_bits_to_float(kreg[c][j]);
+    }
+    s = group_reduce_sum<4>(s);
+    const int gtok = bi * PAGE + tok;
+    s = (gtok < ctx) ? s * scale : NEG_INF;
+    const float tmax = wave_reduce_max(s);
+    const float m_new = fmaxf(m, tmax);
+    const float rescale = __expf(m - m_new);
+    const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+    const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+    l = l * rescale + psum;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) acc[a] *= rescale;
+    m = m_new;
+#pragma unroll
+    for (int t = 0; t < PAGE; ++t) {
+      const float pt = __shfl(p, t * 4, 64);
+      if constexpr (ACC == 2) {
+        acc[0] += pt * bf16_bits_to_float((short)(vreg[t] & 0xFFFF));
+        acc[1] += pt * bf16_bits_to_float((short)(vreg[t] >> 16));
+      } else {
+        acc[0] += pt * bf16_bits_to_float((short)vreg[t]);
+      }
+    }
+  };
+
+  short8_t kA[QFRAG / 8], kB[QFRAG / 8];
+  unsigned int vA[PAGE], vB[PAGE];
+  int bi = blk_lo;
+  for (; bi + 1 < blk_hi; bi += 2) {
+    load_page(bi, kA, vA);
+    load_page(bi + 1, kB, vB);
+    compute_page(bi, kA, vA);
+    compute_page(bi + 1, kB, vB);
+  }
+  if (bi < blk_hi) {
+    load_page(bi, kA, vA);
+    compute_page(bi, kA, vA);
+  }
+
+  if (n_splits == 1) {
+    const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
+    short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) op[a] = float_to_bf16_bits(acc[a] * inv_l);
+  } else {
+    float* po = part_out +
+                (((long)seq * num_heads + head) * n_splits + split) * D +
+                lane * ACC;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) po[a] = acc[a];
+    if (lane == 0) {
+      float* ml =
+          part_ml + (((long)seq * num_heads + head) * n_splits + split) * 2;
+      ml[0] = m;
+      ml[1] = l;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Wave-split variant: the 4 waves partition the PAGES (each page is read by
 // exactly one wave — no cross-wave L1 re-reads), every wave computes ALL
 // GROUP heads for its pages, and the four per-wave online-softmax partials
@@ -499,6 +641,32 @@ extern "C" hipError_t ks_paged_attention_decode(
     else if (group == 4) LAUNCH_WS(4);
     else LAUNCH_WS(8);
 #undef LAUNCH_WS
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      const long sh = (long)num_seqs * num_heads;
+      int wpb = 4;
+      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(wpb * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return hipSuccess;
+  }
+
+  // 2-page unrolled fast path (A/B: KS_ATTN_U2=0 disables)
+  static const bool use_u2 = [] {
+    const char* e = getenv("KS_ATTN_U2");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (use_u2 && hpw == 1 && head_dim == 128) {
+    hipLaunchKernelGGL((paged_attention_u2_kernel<128>), grid, block, 0,
+                       stream, (short*)out, (const short*)q,
+                       (const short*)k_cache, (const short*)v_cache,
+                       (const int*)block_tables, (const int*)context_lens,
+                       scale, num_kv_heads, group, max_blocks, q_row_stride,
+                       n_splits, (float*)part_out, (float*)part_ml);
     HIP_CHECK_KERNEL();
     if (n_splits > 1) {
       const long sh = (long)num_seqs * num_heads;
