@@ -76,11 +76,12 @@ __global__ __launch_bounds__(256) void paged_attention_kernel(
   int heads[HPW];
   bool hact[HPW];
   bool any_active = false;
+  const int nw = blockDim.x >> 6;  // active waves (4, or 2 for L1-relief)
 #pragma unroll
   for (int h = 0; h < HPW; ++h) {
-    const int head = kv_head * group + wave + h * NWAVES;
+    const int head = kv_head * group + wave + h * nw;
     heads[h] = head;
-    hact[h] = (wave + h * NWAVES) < group;
+    hact[h] = (wave + h * nw) < group;
     any_active |= hact[h];
     if (hact[h]) {
       const short* qp =
@@ -655,10 +656,37 @@ extern "C" hipError_t ks_paged_attention_decode(
     return hipSuccess;
   }
 
+  // 2-wave x 2-head mode (A/B: KS_ATTN_H2=1): halves the per-CU L1
+  // traffic from duplicated page reads at the cost of HPW=2 registers
+  static const bool use_h2 = [] {
+    const char* e = getenv("KS_ATTN_H2");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_h2 && group == 4 && head_dim == 128) {
+    hipLaunchKernelGGL((paged_attention_kernel<128, 2>), grid, dim3(128), 0,
+                       stream, (short*)out, (const short*)q,
+                       (const short*)k_cache, (const short*)v_cache,
+                       (const int*)block_tables, (const int*)context_lens,
+                       scale, num_kv_heads, group, max_blocks, q_row_stride,
+                       n_splits, (float*)part_out, (float*)part_ml);
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      const long sh = (long)num_seqs * num_heads;
+      int wpb = 4;
+      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(wpb * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return hipSuccess;
+  }
+
   // 2-page unrolled fast path (A/B: KS_ATTN_U2=0 disables)
-  static const bool use_u2 = [] {
+  static const bool use_u2 = [] {  // measured slower; opt-in ablation
     const char* e = getenv("KS_ATTN_U2");
-    return e == nullptr || e[0] != '0';
+    return e != nullptr && e[0] == '1';
   }();
   if (use_u2 && hpw == 1 && head_dim == 128) {
     hipLaunchKernelGGL((paged_attention_u2_kernel<128>), grid, block, 0,
