@@ -279,8 +279,9 @@ void skinny_gemm(at::Tensor& out, at::Tensor& x, at::Tensor& w) {
   while (feat_wgs * nsplit < 768 && nsplit < 8 && (K / (nsplit * 2)) >= 64)
     nsplit *= 2;
   if (nsplit > 1) {
+    // bf16 partials (each a full-precision-accumulated K/nsplit dot)
     ws = at::empty({(long)nsplit * N * M},
-                   at::TensorOptions().dtype(at::kFloat).device(x.device()));
+                   at::TensorOptions().dtype(at::kBFloat16).device(x.device()));
     wsp = ws.data_ptr();
   }
   check_hip(ks_skinny_gemm(out.data_ptr(), wsp, x.data_ptr(), w.data_ptr(),

@@ -31,7 +31,7 @@ __device__ __forceinline__ int lds_row_idx(int row, int kc) {
 template <int NT, bool SPLIT>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     short* __restrict__ out,      // [N][M] bf16 (!SPLIT)
-    float* __restrict__ out_ws,   // [nsplit][N][M] f32 (SPLIT)
+    short* __restrict__ out_ws,   // [nsplit][N][M] bf16 partials (SPLIT)
     const short* __restrict__ x,  // [N][K] row stride xs
     const short* __restrict__ w,  // [M][K]
     const int M, const int K, const int N, const long xs,
@@ -138,7 +138,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
       const int tokn = t * 16 + MFMA_C_ROW(lane, reg);
       if (tokn < N) {
         if constexpr (SPLIT) {
-          out_ws[((long)blockIdx.y * N + tokn) * M + feat] = acc[t][reg];
+          // bf16 partials: each is a full K/nsplit-length dot in f32 first,
+          // so the rounding is one bf16 quantization per partial
+          out_ws[((long)blockIdx.y * N + tokn) * M + feat] =
+              float_to_bf16_bits(acc[t][reg]);
         } else {
           out[(long)tokn * M + feat] = float_to_bf16_bits(acc[t][reg]);
         }
@@ -147,20 +150,21 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
 }
 
-// sum partials over splits and cast to bf16
+// sum bf16 partials (f32 accumulate) and store bf16
 __global__ void reduce_ws_kernel(short* __restrict__ out,
-                                 const float* __restrict__ ws, const long nm,
+                                 const short* __restrict__ ws, const long nm,
                                  const int nsplit) {
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < nm;
        i += (long)gridDim.x * blockDim.x) {
     float acc = 0.f;
-    for (int s = 0; s < nsplit; ++s) acc += ws[s * nm + i];
+    for (int s = 0; s < nsplit; ++s)
+      acc += bf16_bits_to_float(ws[s * nm + i]);
     out[i] = float_to_bf16_bits(acc);
   }
 }
 
 template <int NT>
-void launch_nt(short* out, float* ws, const short* x, const short* w, int M,
+void launch_nt(short* out, short* ws, const short* x, const short* w, int M,
                int K, int N, long xs, int nsplit, int k_per_split,
                hipStream_t stream) {
   dim3 grid(M / (NWAVES * FEAT_PER_WAVE), nsplit);
@@ -188,7 +192,7 @@ extern "C" hipError_t ks_skinny_gemm(void* out, void* workspace,
   if (nsplit > 1 && workspace == nullptr) return hipErrorInvalidValue;
   int k_per_split = ((K / nsplit + BK - 1) / BK) * BK;
   const int NT = (N + 15) / 16;
-  float* ws = (float*)workspace;
+  short* ws = (short*)workspace;
   const short* xs = (const short*)x;
   const short* wp = (const short*)w;
 #define CASE(nt)                                                            \
