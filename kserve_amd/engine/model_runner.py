@@ -185,31 +185,43 @@ class ModelRunner:
     def prepare_prefill(
         self, batch: ScheduledBatch, block_manager: BlockManager
     ):
-        tokens: List[int] = []
-        positions: List[int] = []
-        slot_mapping: List[int] = []
+        import numpy as np
+
+        total = sum(batch.num_scheduled_tokens)
+        tokens = np.empty(total, dtype=np.int64)
+        positions = np.empty(total, dtype=np.int64)
+        slots = np.empty(total, dtype=np.int32)
         cu = [0]
         max_seqlen = 0
         last_token_idx: List[int] = []
+        off = 0
+        bs = self.config.cache.block_size
         for req, n in zip(batch.requests, batch.num_scheduled_tokens):
             start = req.num_computed_tokens
             end = start + n
-            ids = req.all_token_ids[start:end]
-            tokens.extend(ids)
-            positions.extend(range(start, end))
-            slot_mapping.extend(block_manager.slot_mapping(req, start, end))
+            tokens[off : off + n] = req.all_token_ids[start:end]
+            positions[off : off + n] = np.arange(start, end, dtype=np.int64)
+            # vectorized slot mapping from the block table
+            bt = np.asarray(
+                block_manager.get_block_table(req), dtype=np.int32
+            )
+            pos_range = np.arange(start, end, dtype=np.int64)
+            slots[off : off + n] = bt[pos_range // bs] * bs + (
+                pos_range % bs
+            ).astype(np.int32)
+            off += n
             cu.append(cu[-1] + n)
             max_seqlen = max(max_seqlen, n)
             last_token_idx.append(cu[-1] - 1)
         dev = self.device
         meta = AttentionMetadata(
             is_prefill=True,
-            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int32, device=dev),
+            slot_mapping=torch.from_numpy(slots).to(dev, non_blocking=True),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
             max_seqlen=max_seqlen,
         )
-        input_ids = torch.tensor(tokens, dtype=torch.int64, device=dev)
-        pos = torch.tensor(positions, dtype=torch.int64, device=dev)
+        input_ids = torch.from_numpy(tokens).to(dev, non_blocking=True)
+        pos = torch.from_numpy(positions).to(dev, non_blocking=True)
         sel = torch.tensor(last_token_idx, dtype=torch.int64, device=dev)
         return input_ids, pos, meta, sel
 
