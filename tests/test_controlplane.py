@@ -329,3 +329,48 @@ class TestKnativeMode:
         traffic = ksvc["spec"]["traffic"]
         assert traffic[0]["percent"] == 10 and traffic[1]["percent"] == 90
         assert "deployment" not in out
+
+
+class TestKServeClient:
+    def test_crud_with_injected_apply(self):
+        from kserve_amd.client import KServeClient
+
+        applied = []
+        client = KServeClient(
+            apply_fn=lambda m: applied.append((m["kind"], m["metadata"]["name"])),
+            delete_fn=lambda k, n, ns: applied.append(("DEL", n)),
+        )
+        isvc = make_isvc()
+        manifests = client.create(isvc)
+        assert "deployment" in manifests
+        kinds = [k for k, _ in applied]
+        assert "Deployment" in kinds and "Service" in kinds
+        assert client.get("iris", "ns1") is not None
+        assert client.wait_isvc_ready("iris", "ns1", timeout_seconds=1)
+        client.delete("iris", "ns1")
+        assert client.get("iris", "ns1") is None
+        assert ("DEL", "iris-predictor") in applied
+
+    def test_llm_crud(self):
+        from kserve_amd.client import KServeClient
+        from kserve_amd.controlplane.llmisvc import (
+            LLMInferenceService,
+            LLMInferenceServiceSpec,
+            LLMModelSpec,
+            ParallelismSpec,
+            WorkloadSpec,
+        )
+
+        client = KServeClient(apply_fn=lambda m: None)
+        llm = LLMInferenceService(
+            name="llm1",
+            spec=LLMInferenceServiceSpec(
+                model=LLMModelSpec(uri="hf://x", name="m"),
+                workload=WorkloadSpec(parallelism=ParallelismSpec(tensor=2)),
+            ),
+        )
+        out = client.create_llm(llm)
+        assert out["decode"]["kind"] == "Deployment"
+        assert client.get_llm("llm1") is not None
+        client.delete_llm("llm1")
+        assert client.get_llm("llm1") is None
