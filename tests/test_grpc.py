@@ -112,3 +112,28 @@ def test_grpc_end_to_end(dataplane):
         start_task.cancel()
 
     _run(main())
+
+
+def test_grpc_client_roundtrip(dataplane):
+    """InferenceGRPCClient against the in-process server."""
+    async def main():
+        from kserve_amd.inference_client import InferenceGRPCClient
+        from kserve_amd.protocol.grpc.server import GRPCServer
+        from kserve_amd.protocol.infer_type import InferInput, InferRequest
+
+        server = GRPCServer(dataplane, port=0)
+        task = asyncio.create_task(server.start())
+        await asyncio.sleep(0.2)
+        client = InferenceGRPCClient(f"127.0.0.1:{server.bound_port}")
+        assert await client.is_server_ready()
+        assert await client.is_model_ready("echo")
+        x = np.arange(6, dtype=np.float32).reshape(2, 3)
+        inp = InferInput("x", [2, 3], "FP32")
+        inp.set_data_from_numpy(x, binary_data=True)
+        resp = await client.infer(InferRequest("echo", [inp]))
+        np.testing.assert_array_equal(resp.outputs[0].as_numpy(), x * 2)
+        await client.close()
+        await server.stop(grace=0.1)
+        task.cancel()
+
+    _run(main())

@@ -37,6 +37,7 @@ def client():
     dataplane = DataPlane(repo)
     app = create_app(dataplane)
     register_openai_endpoints(app, dataplane, [model])
+    app.state.llm_model = model
 
     with TestClient(app) as c:
         # start engine manually (TestClient doesn't run ModelServer lifecycle)
@@ -176,3 +177,35 @@ def test_completion_n_choices(client):
     )
     assert r.status_code == 200, r.text
     assert len(r.json()["choices"]) == 3
+
+
+def test_stream_disconnect_aborts(client):
+    """Closing the SSE stream mid-generation aborts the engine request
+    (AsyncLLMEngine submits an abort sentinel on generator close)."""
+    import time
+
+    from kserve_amd.runtimes.llm_model import LLMModel
+
+    model = client.app.state.llm_model
+    assert isinstance(model, LLMModel)
+    engine = model.async_engine.engine
+
+    with client.stream(
+        "POST",
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [9, 9, 9],
+            "max_tokens": 100,
+            "temperature": 0.0,
+            "stream": True,
+        },
+    ) as r:
+        for i, line in enumerate(r.iter_lines()):
+            if i >= 2:
+                break  # client disconnects early
+
+    deadline = time.time() + 10
+    while time.time() < deadline and engine.has_unfinished():
+        time.sleep(0.1)
+    assert not engine.has_unfinished(), "aborted request still scheduled"
