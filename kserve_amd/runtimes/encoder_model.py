@@ -262,5 +262,43 @@ class OpenAIEmbeddingAdapter:
     async def create_chat_completion(self, *a, **k):
         raise NotImplementedError("encoder models do not generate")
 
-    async def create_rerank(self, *a, **k):
-        raise NotImplementedError
+    async def create_rerank(self, request, raw_request=None, context=None):
+        from kserve_amd.protocol.rest.openai.types import (
+            Rerank,
+            RerankResult,
+            UsageInfo,
+        )
+
+        texts = [request.query] + list(request.documents)
+        payload = self.encoder.preprocess({"instances": texts})
+        result = self.encoder.predict(payload)
+        post = self.encoder.postprocess(result)
+        vectors = post["predictions"]
+        scores = cosine_rerank(vectors[0], vectors[1:])
+        order = sorted(range(len(scores)), key=lambda i: -scores[i])
+        if request.top_n:
+            order = order[: request.top_n]
+        results = [
+            RerankResult(
+                index=i,
+                relevance_score=scores[i],
+                document={"text": request.documents[i]}
+                if request.return_documents
+                else None,
+            )
+            for i in order
+        ]
+        total = sum(len(ids) for ids in result["ids_list"])
+        return Rerank(results=results, usage=UsageInfo(prompt_tokens=total, total_tokens=total))
+
+
+def cosine_rerank(query_vec, doc_vecs):
+    import numpy as np
+
+    q = np.asarray(query_vec, dtype=np.float64)
+    qn = q / (np.linalg.norm(q) + 1e-9)
+    scores = []
+    for d in doc_vecs:
+        dv = np.asarray(d, dtype=np.float64)
+        scores.append(float(qn @ (dv / (np.linalg.norm(dv) + 1e-9))))
+    return scores

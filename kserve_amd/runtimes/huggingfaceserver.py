@@ -69,7 +69,9 @@ def build_model(args):
             model=ModelConfig.from_hf_config(
                 os.path.join(args.model_dir, "config.json")
             ),
-            cache=CacheConfig(),
+            cache=CacheConfig(
+                cpu_offload_bytes=getattr(args, "kv_offload_bytes", 0)
+            ),
             scheduler=SchedulerConfig(
                 max_num_seqs=getattr(args, "max_num_seqs", 256),
                 max_model_len=getattr(args, "max_model_len", 8192),
@@ -99,6 +101,16 @@ def main(argv=None):
     parser.add_argument("--tensor-parallel-size", dest="tensor_parallel_size", type=int, default=1)
     parser.add_argument("--max_model_len", type=int, default=8192)
     parser.add_argument("--max_num_seqs", type=int, default=256)
+    # data-parallel contract flags (reference preset
+    # config-llm-worker-data-parallel.yaml:188-199)
+    parser.add_argument("--data-parallel-size", dest="data_parallel_size", type=int, default=1)
+    parser.add_argument("--data-parallel-size-local", dest="data_parallel_size_local", type=int, default=None)
+    parser.add_argument("--data-parallel-rpc-port", dest="data_parallel_rpc_port", type=int, default=5555)
+    parser.add_argument("--data-parallel-address", dest="data_parallel_address", default="127.0.0.1")
+    parser.add_argument("--data-parallel-start-rank", dest="data_parallel_start_rank", type=int, default=0)
+    parser.add_argument("--enable-expert-parallel", dest="expert_parallel", action="store_true")
+    # host-DRAM KV offload tier (LLMInferenceService KVCacheOffloadingSpec)
+    parser.add_argument("--kv-offload-bytes", dest="kv_offload_bytes", type=int, default=0)
     args = parser.parse_args(argv)
     configure_logging()
     model = build_model(args)

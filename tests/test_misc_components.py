@@ -195,3 +195,37 @@ class TestExplainer:
         assert imps[0] > imps[1]  # x0 matters more
         assert imps[0] == pytest.approx(6.0)
         assert imps[1] == pytest.approx(5.0)
+
+
+class TestRerankAdapter:
+    def test_cosine_rerank(self):
+        from kserve_amd.protocol.rest.openai.types import RerankRequest
+        from kserve_amd.runtimes.encoder_model import OpenAIEmbeddingAdapter
+
+        class FakeEncoder:
+            name = "emb"
+
+            def preprocess(self, payload, headers=None):
+                return {"texts": payload["instances"], "_v2": None}
+
+            def predict(self, payload, headers=None):
+                vecs = {
+                    "q": [1.0, 0.0],
+                    "близко": [0.9, 0.1],
+                    "far": [0.0, 1.0],
+                }
+                return {
+                    "output": [vecs.get(t, [0.5, 0.5]) for t in payload["texts"]],
+                    "ids_list": [[1]] * len(payload["texts"]),
+                    "cu": None,
+                    "_v2": None,
+                }
+
+            def postprocess(self, result, headers=None):
+                return {"predictions": result["output"]}
+
+        adapter = OpenAIEmbeddingAdapter(FakeEncoder())
+        req = RerankRequest(model="emb", query="q", documents=["far", "близко"])
+        out = run(adapter.create_rerank(req))
+        assert out.results[0].index == 1  # близко ranks first
+        assert out.results[0].relevance_score > out.results[1].relevance_score
