@@ -156,8 +156,10 @@ def main():
     if world > 1:
         import torch.distributed as dist
 
-        te = torch.tensor([elapsed], dtype=torch.float64)
-        tk = torch.tensor([float(tokens)], dtype=torch.float64)
+        # NCCL/RCCL requires device tensors for collectives
+        red_dev = device if use_gpu else "cpu"
+        te = torch.tensor([elapsed], dtype=torch.float64, device=red_dev)
+        tk = torch.tensor([float(tokens)], dtype=torch.float64, device=red_dev)
         dist.all_reduce(te, op=dist.ReduceOp.MAX)
         dist.all_reduce(tk, op=dist.ReduceOp.SUM)
         elapsed = float(te[0])
