@@ -207,6 +207,211 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 
 }  // namespace
 
+// ---------------------------------------------------------------------------
+// v2: 32 q-rows per wave (128/wg) x 64-token KV tiles — 4x the MFMA work per
+// barrier of v1 (guide ladder: bigger tiles amortize staging + softmax).
+// D=128 only; v1 stays for D=64 and as the KS_PREFILL_V1 ablation.
+// ---------------------------------------------------------------------------
+namespace {
+
+constexpr int KT2 = 64;   // kv tokens per tile
+constexpr int QW2 = 32;   // q rows per wave
+constexpr int PSTR = 72;  // p_lds row stride (64 + 8 shorts: 2-way banks)
+
+template <int D, bool CAUSAL>  // D == 128
+__global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
+    short* __restrict__ out, const short* __restrict__ q,
+    const short* __restrict__ k, const short* __restrict__ v,
+    const int* __restrict__ cu_seqlens, const int Hq, const int Hkv,
+    const float scale, const long sq, const long sk, const long sv) {
+  const int head = blockIdx.x;
+  const int tile = blockIdx.y;
+  const int seq = blockIdx.z;
+  const int q_start = cu_seqlens[seq];
+  const int len = cu_seqlens[seq + 1] - q_start;
+  const int tile_base = tile * (NWAVES * QW2);
+  if (tile_base >= len) return;
+  const int group = Hq / Hkv;
+  const int kv_head = head / group;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  __shared__ short k_tile[KT2 * D];            // 16 KB swizzled
+  __shared__ short v_t[D * KT2];               // 16 KB transposed+swizzled
+  __shared__ short p_lds[NWAVES][QW2 * PSTR];  // 18 KB
+
+  const int wq0 = tile_base + wave * QW2;
+  const bool active = wq0 < len;
+
+  // Q fragments: 2 row-tiles x 4 k-chunks (A operand)
+  bf16x8_t a_q[2][D / 32];
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt) {
+    const int qrow = min(wq0 + rt * 16 + MFMA_RC_OF(lane), len - 1);
+    const short* qp = q + (long)(q_start + qrow) * sq + (long)head * D;
+#pragma unroll
+    for (int c = 0; c < D / 32; ++c) {
+      a_q[rt][c] = *reinterpret_cast<const bf16x8_t*>(
+          qp + c * 32 + ((lane >> 4) << 3));
+    }
+  }
+
+  float m[2][4], l[2][4];
+  f32x4_t o_acc[2][D / 16];
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m[rt][r] = NEG_INF;
+      l[rt][r] = 0.f;
+    }
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c)
+      o_acc[rt][c] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int kv_limit = CAUSAL ? min(len, tile_base + NWAVES * QW2) : len;
+  const int ntiles = (kv_limit + KT2 - 1) / KT2;
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int t0 = kt * KT2;
+    __syncthreads();
+    for (int i = threadIdx.x; i < KT2 * (D / 8); i += 256) {
+      const int r = i / (D / 8);
+      const int c8 = (i % (D / 8)) * 8;
+      const int tok = t0 + r;
+      short8_t val, vv;
+      if (tok < len) {
+        val = *reinterpret_cast<const short8_t*>(
+            k + (long)(q_start + tok) * sk + (long)kv_head * D + c8);
+        vv = *reinterpret_cast<const short8_t*>(
+            v + (long)(q_start + tok) * sv + (long)kv_head * D + c8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          val[j] = 0;
+          vv[j] = 0;
+        }
+      }
+      *reinterpret_cast<short8_t*>(&k_tile[k_idx<D>(r, c8)]) = val;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v_t[(c8 + j) * KT2 + (r ^ (((c8 + j) & 3) << 3))] = vv[j];
+    }
+    __syncthreads();
+    if (!active) continue;
+    if (CAUSAL && t0 > wq0 + QW2 - 1) continue;
+
+    // ---- per row-tile: QK^T over 4 token sub-tiles, online softmax ----
+#pragma unroll
+    for (int rt = 0; rt < 2; ++rt) {
+      const int rbase = wq0 + rt * 16;
+      if (CAUSAL && t0 > rbase + 15) continue;  // fully-masked row-tile
+      f32x4_t s_frag[KT2 / 16];
+#pragma unroll
+      for (int n = 0; n < KT2 / 16; ++n) {
+        s_frag[n] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+        const int tok_row = n * 16 + MFMA_RC_OF(lane);
+#pragma unroll
+        for (int c = 0; c < D / 32; ++c) {
+          const int col = c * 32 + ((lane >> 4) << 3);
+          bf16x8_t bk =
+              *reinterpret_cast<bf16x8_t*>(&k_tile[k_idx<D>(tok_row, col)]);
+          s_frag[n] = mfma16x16x32(a_q[rt][c], bk, s_frag[n]);
+        }
+      }
+      float p[KT2 / 16][4];
+      float rescale[4];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int qrow = rbase + MFMA_C_ROW(lane, reg);
+        float rowmax = NEG_INF;
+        float sv_[KT2 / 16];
+#pragma unroll
+        for (int n = 0; n < KT2 / 16; ++n) {
+          const int tok = t0 + n * 16 + MFMA_C_COL(lane);
+          const bool valid =
+              (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len);
+          sv_[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
+          rowmax = fmaxf(rowmax, sv_[n]);
+        }
+        rowmax = group_reduce_max<16>(rowmax);
+        const float m_new = fmaxf(m[rt][reg], rowmax);
+        rescale[reg] = (m[rt][reg] > NEG_INF && m_new > NEG_INF)
+                           ? __expf(m[rt][reg] - m_new)
+                           : 0.f;
+        float psum = 0.f;
+#pragma unroll
+        for (int n = 0; n < KT2 / 16; ++n) {
+          p[n][reg] = (sv_[n] > NEG_INF) ? __expf(sv_[n] - m_new) : 0.f;
+          psum += p[n][reg];
+        }
+        psum = group_reduce_sum<16>(psum);
+        l[rt][reg] = l[rt][reg] * rescale[reg] + psum;
+        m[rt][reg] = m_new;
+      }
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) o_acc[rt][c][reg] *= rescale[reg];
+      }
+#pragma unroll
+      for (int n = 0; n < KT2 / 16; ++n) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          p_lds[wave][(rt * 16 + MFMA_C_ROW(lane, reg)) * PSTR + n * 16 +
+                      MFMA_C_COL(lane)] = float_to_bf16_bits(p[n][reg]);
+        }
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    __builtin_amdgcn_sched_barrier(0);
+    // ---- PV: O[32 q][D] += P[32 q][64 t] @ V[64 t][D] ----
+#pragma unroll
+    for (int rt = 0; rt < 2; ++rt) {
+      if (CAUSAL && t0 > wq0 + rt * 16 + 15) continue;
+#pragma unroll
+      for (int kk = 0; kk < KT2 / 32; ++kk) {
+        bf16x8_t pa = *reinterpret_cast<bf16x8_t*>(
+            &p_lds[wave][(rt * 16 + MFMA_RC_OF(lane)) * PSTR + kk * 32 +
+                         ((lane >> 4) << 3)]);
+#pragma unroll
+        for (int c = 0; c < D / 16; ++c) {
+          const int d = c * 16 + MFMA_RC_OF(lane);
+          const int tok0 = kk * 32 + ((lane >> 4) << 3);
+          bf16x8_t bv = *reinterpret_cast<bf16x8_t*>(
+              &v_t[d * KT2 + (tok0 ^ ((d & 3) << 3))]);
+          o_acc[rt][c] = mfma16x16x32(pa, bv, o_acc[rt][c]);
+        }
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    __builtin_amdgcn_sched_barrier(0);
+  }
+
+  if (!active) return;
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt) {
+    float inv_l[4];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg)
+      inv_l[reg] = (l[rt][reg] > 0.f) ? 1.f / l[rt][reg] : 0.f;
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int qrow = wq0 + rt * 16 + MFMA_C_ROW(lane, reg);
+        if (qrow < len) {
+          out[((long)(q_start + qrow) * Hq + head) * D + c * 16 +
+              MFMA_C_COL(lane)] =
+              float_to_bf16_bits(o_acc[rt][c][reg] * inv_l[reg]);
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
 extern "C" hipError_t ks_flash_prefill_varlen(
     void* out, const void* q, const void* k, const void* v,
     const void* cu_seqlens, int num_seqs, int max_seqlen, int Hq, int Hkv,
@@ -217,6 +422,28 @@ extern "C" hipError_t ks_flash_prefill_varlen(
   const int max_tiles = (max_seqlen + NWAVES * QW - 1) / (NWAVES * QW);
   if (max_tiles == 0 || num_seqs == 0) return hipSuccess;
   dim3 grid(Hq, max_tiles, num_seqs);
+  // v2 (32 q-rows/wave, 64-token tiles) for D=128; KS_PREFILL_V1=1 reverts
+  static const bool use_v1 = [] {
+    const char* e = getenv("KS_PREFILL_V1");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (head_dim == 128 && !use_v1) {
+    const int max_tiles2 = (max_seqlen + NWAVES * QW2 - 1) / (NWAVES * QW2);
+    dim3 grid2(Hq, max_tiles2, num_seqs);
+    if (causal) {
+      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, true>), grid2,
+                         dim3(256), 0, stream, (short*)out, (const short*)q,
+                         (const short*)k, (const short*)v,
+                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv);
+    } else {
+      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, false>), grid2,
+                         dim3(256), 0, stream, (short*)out, (const short*)q,
+                         (const short*)k, (const short*)v,
+                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv);
+    }
+    HIP_CHECK_KERNEL();
+    return hipSuccess;
+  }
 #define LAUNCH_FP(DD, CC)                                                    \
   hipLaunchKernelGGL((flash_prefill_kernel<DD, CC>), grid, dim3(256), 0,     \
                      stream, (short*)out, (const short*)q, (const short*)k,  \
