@@ -33,7 +33,7 @@ def main():
     parser.add_argument("--gpus", type=int, default=1)
     parser.add_argument("--steps", type=int, default=64)
     parser.add_argument("--warmup", type=int, default=16)
-    parser.add_argument("--concurrency", type=int, default=768,
+    parser.add_argument("--concurrency", type=int, default=1536,
                         help="concurrent sequences per GPU")
     parser.add_argument("--prompt-len", type=int, default=512)
     parser.add_argument("--model", type=str, default="llama-3-8b",

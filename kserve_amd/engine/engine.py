@@ -255,6 +255,11 @@ class LLMEngine:
                     if r.num_computed_tokens >= r.num_tokens
                 ]
                 logits = logits[idx]
+                # partial chunks go back to the head of the waiting queue so
+                # the next step continues their prefill (not decode)
+                for r in reversed(batch.requests):
+                    if r.num_computed_tokens < r.num_tokens:
+                        self.scheduler.requeue_partial_prefill(r)
         else:
             sample_reqs = batch.requests
         outputs: List[RequestOutput] = []

@@ -220,6 +220,26 @@ class ModelRunner:
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
             max_seqlen=max_seqlen,
         )
+        # chunked-prefill resume: any request with context already in pages
+        # must attend to it — switch to the paged-context attention path
+        if any(r.num_computed_tokens > 0 for r in batch.requests) and (
+            self.kv_caches and self.kv_caches[0][0].numel() > 0
+        ):
+            nb_per = [
+                -(-(r.num_computed_tokens + n) // bs)
+                for r, n in zip(batch.requests, batch.num_scheduled_tokens)
+            ]
+            max_nb = max(nb_per)
+            btab = np.zeros((len(batch.requests), max_nb), dtype=np.int32)
+            ctx = np.empty(len(batch.requests), dtype=np.int32)
+            for i, (req, n) in enumerate(
+                zip(batch.requests, batch.num_scheduled_tokens)
+            ):
+                t = block_manager.get_block_table(req)
+                btab[i, : nb_per[i]] = t[: nb_per[i]]
+                ctx[i] = req.num_computed_tokens + n
+            meta.block_tables = torch.from_numpy(btab).to(dev, non_blocking=True)
+            meta.context_lens = torch.from_numpy(ctx).to(dev, non_blocking=True)
         input_ids = torch.from_numpy(tokens).to(dev, non_blocking=True)
         pos = torch.from_numpy(positions).to(dev, non_blocking=True)
         sel = torch.tensor(last_token_idx, dtype=torch.int64, device=dev)

@@ -200,6 +200,14 @@ class Scheduler:
         req.status = RequestStatus.PREEMPTED
         self.waiting.appendleft(req)
 
+    def requeue_partial_prefill(self, req: Request) -> None:
+        """A chunked prefill finished its chunk but not the prompt: return it
+        to the head of the waiting queue (KV blocks kept) so the next step
+        schedules the following chunk instead of decoding it."""
+        self.running.remove(req)
+        req.status = RequestStatus.WAITING
+        self.waiting.appendleft(req)
+
     def _preempt_swap(self, req: Request):
         """Offload preemption: KV pages move to pinned host DRAM; the request
         resumes later via swap_in with no recompute."""

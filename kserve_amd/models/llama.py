@@ -40,7 +40,8 @@ class AttentionMetadata:
     # prefill
     cu_seqlens: Optional[torch.Tensor] = None  # [num_seqs+1] int32
     max_seqlen: int = 0
-    # decode
+    # decode — and chunked prefill, where block_tables/context_lens describe
+    # the paged context the chunk attends to (context_lens includes the chunk)
     block_tables: Optional[torch.Tensor] = None  # [num_seqs, max_blocks] int32
     context_lens: Optional[torch.Tensor] = None  # [num_seqs] int32
 
@@ -87,9 +88,17 @@ class LlamaAttention(nn.Module):
         if k_cache.numel() > 0:
             ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
         if meta.is_prefill:
-            out = ops.flash_prefill_varlen(
-                q, k, v, meta.cu_seqlens, meta.max_seqlen, self.scale
-            )
+            if meta.context_lens is not None:
+                # chunked prefill: attend to the whole paged context (past
+                # chunks' KV + this chunk's, written to pages above)
+                out = ops.context_attention_varlen(
+                    q, k_cache, v_cache, meta.block_tables, meta.cu_seqlens,
+                    meta.context_lens, meta.max_seqlen, self.scale
+                )
+            else:
+                out = ops.flash_prefill_varlen(
+                    q, k, v, meta.cu_seqlens, meta.max_seqlen, self.scale
+                )
         else:
             out = ops.paged_attention_decode(
                 q, k_cache, v_cache, meta.block_tables, meta.context_lens, self.scale

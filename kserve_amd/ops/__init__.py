@@ -128,6 +128,31 @@ def flash_prefill_varlen(
     return torch_ref.flash_prefill_varlen(q, k, v, cu_seqlens, scale, causal=True)
 
 
+def context_attention_varlen(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    cu_seqlens_q: torch.Tensor,
+    context_lens: torch.Tensor,
+    max_q_len: int,
+    scale: float,
+) -> torch.Tensor:
+    """Prefill attention against the paged cache (chunked prefill): the
+    chunk's KV must already be written to the cache; queries attend causally
+    to the whole per-seq context."""
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native("context_prefill").context_prefill_varlen(
+            out, q, k_cache, v_cache, block_tables, context_lens,
+            cu_seqlens_q, int(max_q_len), scale
+        )
+        return out
+    return torch_ref.context_attention_varlen(
+        q, k_cache, v_cache, block_tables, cu_seqlens_q, context_lens, scale
+    )
+
+
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     if logits.is_cuda:
         out = torch.empty(

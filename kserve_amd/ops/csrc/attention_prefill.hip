@@ -410,7 +410,237 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Paged-context prefill (chunked prefill): the chunk's q rows attend causally
+// to the WHOLE context — previous chunks' KV plus this chunk's — read from
+// the paged cache (the chunk's KV was written by reshape_and_cache first).
+// Same MFMA/LDS structure as v2; staging gathers 64-token tiles as 4
+// contiguous 16xD pages via the block table. Reference parity: the chunked
+// prefill path of the vLLM backend the reference delegates to.
+// Numerics oracle: ops/torch_ref.context_attention_varlen.
+// ---------------------------------------------------------------------------
+template <int D>  // D == 128
+__global__ __launch_bounds__(256) void context_prefill_kernel(
+    short* __restrict__ out,        // [Tq, Hq, D] bf16
+    const short* __restrict__ q,    // [Tq, Hq, D] (row stride sq)
+    const short* __restrict__ k_cache,  // [NB, Hkv, 16, D]
+    const short* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [S, max_nb]
+    const int* __restrict__ ctx_lens,      // [S] total context incl. chunk
+    const int* __restrict__ cu_seqlens_q,  // [S+1]
+    const int Hq, const int Hkv, const int max_nb, const float scale,
+    const long sq) {
+  const int head = blockIdx.x;
+  const int tile = blockIdx.y;
+  const int seq = blockIdx.z;
+  const int q_start = cu_seqlens_q[seq];
+  const int q_len = cu_seqlens_q[seq + 1] - q_start;
+  const int ctx = ctx_lens[seq];
+  const int start_pos = ctx - q_len;  // absolute position of chunk row 0
+  const int tile_base = tile * (NWAVES * QW2);
+  if (tile_base >= q_len) return;
+  const int group = Hq / Hkv;
+  const int kv_head = head / group;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int* bt = block_tables + (long)seq * max_nb;
+
+  __shared__ short k_tile[KT2 * D];
+  __shared__ short v_t[D * KT2];
+  __shared__ short p_lds[NWAVES][QW2 * PSTR];
+
+  const int wq0 = tile_base + wave * QW2;
+  const bool active = wq0 < q_len;
+
+  bf16x8_t a_q[2][D / 32];
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt) {
+    const int qrow = min(wq0 + rt * 16 + MFMA_RC_OF(lane), q_len - 1);
+    const short* qp = q + (long)(q_start + qrow) * sq + (long)head * D;
+#pragma unroll
+    for (int c = 0; c < D / 32; ++c) {
+      a_q[rt][c] = *reinterpret_cast<const bf16x8_t*>(
+          qp + c * 32 + ((lane >> 4) << 3));
+    }
+  }
+
+  float m[2][4], l[2][4];
+  f32x4_t o_acc[2][D / 16];
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m[rt][r] = NEG_INF;
+      l[rt][r] = 0.f;
+    }
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c)
+      o_acc[rt][c] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  }
+
+  // causal bound in ABSOLUTE positions: last key this workgroup's rows see
+  const int kv_limit =
+      min(ctx, start_pos + tile_base + NWAVES * QW2);
+  const int ntiles = (kv_limit + KT2 - 1) / KT2;
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int t0 = kt * KT2;
+    __syncthreads();
+    // stage 64 context tokens = 4 pages of 16, gathered via the block table
+    for (int i = threadIdx.x; i < KT2 * (D / 8); i += 256) {
+      const int r = i / (D / 8);
+      const int c8 = (i % (D / 8)) * 8;
+      const int tok = t0 + r;
+      short8_t val, vv;
+      if (tok < ctx) {
+        const long base =
+            (((long)bt[tok >> 4] * Hkv + kv_head) * 16 + (tok & 15)) * D + c8;
+        val = *reinterpret_cast<const short8_t*>(k_cache + base);
+        vv = *reinterpret_cast<const short8_t*>(v_cache + base);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          val[j] = 0;
+          vv[j] = 0;
+        }
+      }
+      *reinterpret_cast<short8_t*>(&k_tile[k_idx<D>(r, c8)]) = val;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v_t[(c8 + j) * KT2 + (r ^ (((c8 + j) & 3) << 3))] = vv[j];
+    }
+    __syncthreads();
+    if (!active) continue;
+    // wave-level causal skip: first key of tile past this wave's last q pos
+    if (t0 > start_pos + wq0 + QW2 - 1) continue;
+
+#pragma unroll
+    for (int rt = 0; rt < 2; ++rt) {
+      const int rbase = wq0 + rt * 16;
+      if (t0 > start_pos + rbase + 15) continue;
+      f32x4_t s_frag[KT2 / 16];
+#pragma unroll
+      for (int n = 0; n < KT2 / 16; ++n) {
+        s_frag[n] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+        const int tok_row = n * 16 + MFMA_RC_OF(lane);
+#pragma unroll
+        for (int c = 0; c < D / 32; ++c) {
+          const int col = c * 32 + ((lane >> 4) << 3);
+          bf16x8_t bk =
+              *reinterpret_cast<bf16x8_t*>(&k_tile[k_idx<D>(tok_row, col)]);
+          s_frag[n] = mfma16x16x32(a_q[rt][c], bk, s_frag[n]);
+        }
+      }
+      float p[KT2 / 16][4];
+      float rescale[4];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int qrow = rbase + MFMA_C_ROW(lane, reg);
+        const int qpos = start_pos + qrow;
+        float rowmax = NEG_INF;
+        float sv_[KT2 / 16];
+#pragma unroll
+        for (int n = 0; n < KT2 / 16; ++n) {
+          const int tok = t0 + n * 16 + MFMA_C_COL(lane);
+          const bool valid = (tok <= qpos) && (qrow < q_len) && (tok < ctx);
+          sv_[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
+          rowmax = fmaxf(rowmax, sv_[n]);
+        }
+        rowmax = group_reduce_max<16>(rowmax);
+        const float m_new = fmaxf(m[rt][reg], rowmax);
+        rescale[reg] = (m[rt][reg] > NEG_INF && m_new > NEG_INF)
+                           ? __expf(m[rt][reg] - m_new)
+                           : 0.f;
+        float psum = 0.f;
+#pragma unroll
+        for (int n = 0; n < KT2 / 16; ++n) {
+          p[n][reg] = (sv_[n] > NEG_INF) ? __expf(sv_[n] - m_new) : 0.f;
+          psum += p[n][reg];
+        }
+        psum = group_reduce_sum<16>(psum);
+        l[rt][reg] = l[rt][reg] * rescale[reg] + psum;
+        m[rt][reg] = m_new;
+      }
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) o_acc[rt][c][reg] *= rescale[reg];
+      }
+#pragma unroll
+      for (int n = 0; n < KT2 / 16; ++n) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          p_lds[wave][(rt * 16 + MFMA_C_ROW(lane, reg)) * PSTR + n * 16 +
+                      MFMA_C_COL(lane)] = float_to_bf16_bits(p[n][reg]);
+        }
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+    for (int rt = 0; rt < 2; ++rt) {
+      if (t0 > start_pos + wq0 + rt * 16 + 15) continue;
+#pragma unroll
+      for (int kk = 0; kk < KT2 / 32; ++kk) {
+        bf16x8_t pa = *reinterpret_cast<bf16x8_t*>(
+            &p_lds[wave][(rt * 16 + MFMA_RC_OF(lane)) * PSTR + kk * 32 +
+                         ((lane >> 4) << 3)]);
+#pragma unroll
+        for (int c = 0; c < D / 16; ++c) {
+          const int d = c * 16 + MFMA_RC_OF(lane);
+          const int tok0 = kk * 32 + ((lane >> 4) << 3);
+          bf16x8_t bv = *reinterpret_cast<bf16x8_t*>(
+              &v_t[d * KT2 + (tok0 ^ ((d & 3) << 3))]);
+          o_acc[rt][c] = mfma16x16x32(pa, bv, o_acc[rt][c]);
+        }
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+    __builtin_amdgcn_sched_barrier(0);
+  }
+
+  if (!active) return;
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt) {
+    float inv_l[4];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg)
+      inv_l[reg] = (l[rt][reg] > 0.f) ? 1.f / l[rt][reg] : 0.f;
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int qrow = wq0 + rt * 16 + MFMA_C_ROW(lane, reg);
+        if (qrow < q_len) {
+          out[((long)(q_start + qrow) * Hq + head) * D + c * 16 +
+              MFMA_C_COL(lane)] =
+              float_to_bf16_bits(o_acc[rt][c][reg] * inv_l[reg]);
+        }
+      }
+    }
+  }
+}
+
 }  // namespace
+
+extern "C" hipError_t ks_context_prefill_varlen(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    const void* block_tables, const void* ctx_lens, const void* cu_seqlens_q,
+    int num_seqs, int max_q_len, int Hq, int Hkv, int head_dim, int max_nb,
+    float scale, long sq, hipStream_t stream) {
+  if (head_dim != 128) return hipErrorInvalidValue;
+  if (Hq % Hkv != 0) return hipErrorInvalidValue;
+  const int max_tiles = (max_q_len + NWAVES * QW2 - 1) / (NWAVES * QW2);
+  if (max_tiles == 0 || num_seqs == 0) return hipSuccess;
+  dim3 grid(Hq, max_tiles, num_seqs);
+  hipLaunchKernelGGL((context_prefill_kernel<128>), grid, dim3(256), 0,
+                     stream, (short*)out, (const short*)q,
+                     (const short*)k_cache, (const short*)v_cache,
+                     (const int*)block_tables, (const int*)ctx_lens,
+                     (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq);
+  HIP_CHECK_KERNEL();
+  return hipSuccess;
+}
 
 extern "C" hipError_t ks_flash_prefill_varlen(
     void* out, const void* q, const void* k, const void* v,

@@ -26,6 +26,10 @@ hipError_t ks_flash_prefill_varlen(void*, const void*, const void*,
                                    const void*, const void*, int, int, int,
                                    int, int, float, long, long, long, int,
                                    hipStream_t);
+hipError_t ks_context_prefill_varlen(void*, const void*, const void*,
+                                     const void*, const void*, const void*,
+                                     const void*, int, int, int, int, int,
+                                     int, float, long, hipStream_t);
 hipError_t ks_layer_norm(void*, const void*, const void*, const void*, float,
                          int, int, hipStream_t);
 hipError_t ks_fused_add_layer_norm(void*, const void*, const void*,
@@ -202,6 +206,33 @@ void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
             "flash_prefill_varlen");
 }
 
+void context_prefill_varlen(at::Tensor& out, at::Tensor& q,
+                            at::Tensor& k_cache, at::Tensor& v_cache,
+                            at::Tensor& block_tables, at::Tensor& ctx_lens,
+                            at::Tensor& cu_seqlens_q, int64_t max_q_len,
+                            double scale) {
+  CHECK_BF16_CONTIG(out);
+  CHECK_BF16_ROWS(q);
+  CHECK_BF16_CONTIG(k_cache);
+  CHECK_BF16_CONTIG(v_cache);
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt, "block_tables int32");
+  TORCH_CHECK(block_tables.is_contiguous(), "block_tables contiguous");
+  TORCH_CHECK(ctx_lens.scalar_type() == at::kInt, "ctx_lens int32");
+  TORCH_CHECK(cu_seqlens_q.scalar_type() == at::kInt, "cu_seqlens_q int32");
+  TORCH_CHECK(k_cache.size(2) == 16, "block_size must be 16");
+  int num_seqs = cu_seqlens_q.size(0) - 1;
+  int Hq = q.size(1);
+  int Hkv = k_cache.size(1);
+  int D = q.size(2);
+  check_hip(ks_context_prefill_varlen(
+                out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                v_cache.data_ptr(), block_tables.data_ptr(),
+                ctx_lens.data_ptr(), cu_seqlens_q.data_ptr(), num_seqs,
+                (int)max_q_len, Hq, Hkv, D, (int)block_tables.size(1),
+                (float)scale, (long)q.stride(0), current_stream()),
+            "context_prefill_varlen");
+}
+
 void layer_norm(at::Tensor& out, at::Tensor& input, at::Tensor& weight,
                 at::Tensor& bias, double eps) {
   CHECK_BF16_CONTIG(out);
@@ -312,6 +343,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("v"), pybind11::arg("cu_seqlens"),
         pybind11::arg("max_seqlen"), pybind11::arg("scale"),
         pybind11::arg("causal") = true);
+  m.def("context_prefill_varlen", &context_prefill_varlen,
+        "MFMA prefill attention against the paged KV cache (chunked prefill)");
   m.def("layer_norm", &layer_norm, "LayerNorm (bf16)");
   m.def("fused_add_layer_norm", &fused_add_layer_norm,
         "residual-add + LayerNorm");

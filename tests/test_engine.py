@@ -140,3 +140,35 @@ def test_repetition_penalty_changes_output(engine):
 
     assert out_pen != out_base or max_run(out_pen) <= max_run(out_base)
     assert len(set(out_pen)) >= len(set(out_base))
+
+
+def test_chunked_prefill_matches_full():
+    """Chunked prefill (paged-context attention) must reproduce full-prefill
+    greedy outputs exactly: chunk boundaries change kernel tiling, not math."""
+    torch.manual_seed(0)
+    full = make_engine()
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=128),
+        cache=CacheConfig(block_size=4, num_gpu_blocks=128),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8,
+            max_num_batched_tokens=8,  # prompts below are longer -> chunks
+            max_model_len=128,
+            enable_chunked_prefill=True,
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    chunked = LLMEngine(cfg)
+    sp = SamplingParams(temperature=0.0, max_tokens=6)
+    prompts = [
+        list(range(1, 20)),          # 19 tokens -> 3 chunks
+        list(range(30, 39)),         # 9 tokens -> 2 chunks
+        [3, 1, 4],                   # fits one chunk
+    ]
+    a = full.generate(prompts, sp)
+    b = chunked.generate(prompts, sp)
+    assert [o.output_token_ids for o in a.values()] == [
+        o.output_token_ids for o in b.values()
+    ]
