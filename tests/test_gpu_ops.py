@@ -182,6 +182,47 @@ class TestPrefillAttention:
         bf16_close(got, ref, atol=3e-2, rtol=3e-2)
 
 
+    @pytest.mark.parametrize(
+        "qlens,ctx_extra,H,Hkv",
+        [
+            ([64], [128], 32, 8),       # single resumed chunk
+            ([128, 32, 5], [0, 64, 200], 32, 8),  # mixed fresh+resumed
+            ([1, 300], [500, 17], 8, 8),          # MHA, long context
+            ([256], [0], 16, 4),                  # fresh chunk (ctx==qlen)
+        ],
+    )
+    def test_context_prefill(self, dev, qlens, ctx_extra, H, Hkv):
+        """Chunked-prefill paged-context attention vs fp32 torch ref."""
+        from kserve_amd import ops
+
+        D, bs = 128, 16
+        torch.manual_seed(sum(qlens))
+        S = len(qlens)
+        ctx = [q_ + e for q_, e in zip(qlens, ctx_extra)]
+        max_blocks = max((c + bs - 1) // bs for c in ctx)
+        B = S * max_blocks + 1
+        kc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev)
+        vc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev)
+        bt = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+            S, max_blocks
+        )
+        total_q = sum(qlens)
+        cu = torch.tensor(
+            [0] + list(torch.tensor(qlens).cumsum(0)), dtype=torch.int32
+        )
+        ctx_t = torch.tensor(ctx, dtype=torch.int32)
+        q = torch.randn(total_q, H, D, dtype=torch.bfloat16, device=dev)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.context_attention_varlen(
+            q, kc, vc, bt.to(dev), cu.to(dev), ctx_t.to(dev), max(qlens), scale
+        )
+        ref = torch_ref.context_attention_varlen(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt, cu,
+            ctx_t, scale
+        )
+        bf16_close(got, ref, atol=3e-2, rtol=3e-2)
+
+
 class TestSampling:
     def test_greedy(self, dev):
         from kserve_amd import ops
