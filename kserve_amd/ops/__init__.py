@@ -173,20 +173,25 @@ def random_sample(
 ) -> torch.Tensor:
     """Temperature/top-k/top-p sampling.
 
-    GPU: Gumbel-max HIP kernel over masked logits (exact for top-k/top-p via
-    sort-free rejection is future work; the kernel handles temperature +
-    top-k; top-p uses the torch sort path when < 1.0).
+    GPU: one fused HIP kernel. Plain temperature sampling is Gumbel-max;
+    top-k/top-p use a bf16 radix-histogram select (exact k-th value / mass
+    threshold, no sort) followed by Gumbel-max over the surviving set.
     """
-    if logits.is_cuda:
-        needs_topp = bool((top_p < 1.0).any()) or bool((top_k > 0).any())
-        if not needs_topp and seeds is not None:
-            out = torch.empty(
-                logits.shape[0], dtype=torch.int64, device=logits.device
+    if logits.is_cuda and seeds is not None:
+        out = torch.empty(
+            logits.shape[0], dtype=torch.int64, device=logits.device
+        )
+        needs_trunc = bool((top_p < 1.0).any()) or bool((top_k > 0).any())
+        if needs_trunc:
+            _native("random_sample").topk_topp_sample(
+                out, logits, temperatures, top_p, top_k.to(torch.int32), seeds
             )
+        else:
             _native("random_sample").gumbel_sample(
-                out, logits, temperatures, top_k, seeds
+                out, logits, temperatures, top_k.to(torch.int32), seeds
             )
-            return out
+        return out
+    if logits.is_cuda:
         return torch_ref.random_sample(
             logits, temperatures, top_p, top_k, generator=generator
         )

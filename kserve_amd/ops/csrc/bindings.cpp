@@ -39,6 +39,9 @@ hipError_t ks_gelu(void*, const void*, long, hipStream_t);
 hipError_t ks_greedy_sample(void*, const void*, int, int, hipStream_t);
 hipError_t ks_gumbel_sample(void*, const void*, const void*, const void*,
                             const void*, int, int, hipStream_t);
+hipError_t ks_topk_topp_sample(void*, const void*, const void*, const void*,
+                               const void*, const void*, int, int,
+                               hipStream_t);
 hipError_t ks_mfma_probe(void*, const void*, const void*, hipStream_t);
 hipError_t ks_skinny_gemm(void*, void*, const void*, const void*, int, int,
                           int, long, hipStream_t);
@@ -291,6 +294,23 @@ void gumbel_sample(at::Tensor& out, at::Tensor& logits,
             "gumbel_sample");
 }
 
+void topk_topp_sample(at::Tensor& out, at::Tensor& logits,
+                      at::Tensor& temperatures, at::Tensor& top_p,
+                      at::Tensor& top_k, at::Tensor& seeds) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(out.scalar_type() == at::kLong, "out int64");
+  TORCH_CHECK(temperatures.scalar_type() == at::kFloat, "temps fp32");
+  TORCH_CHECK(top_p.scalar_type() == at::kFloat, "top_p fp32");
+  TORCH_CHECK(top_k.scalar_type() == at::kInt, "top_k int32");
+  TORCH_CHECK(seeds.scalar_type() == at::kLong, "seeds int64");
+  check_hip(ks_topk_topp_sample(out.data_ptr(), logits.data_ptr(),
+                                temperatures.data_ptr(), top_p.data_ptr(),
+                                top_k.data_ptr(), seeds.data_ptr(),
+                                logits.size(0), logits.size(1),
+                                current_stream()),
+            "topk_topp_sample");
+}
+
 void skinny_gemm(at::Tensor& out, at::Tensor& x, at::Tensor& w) {
   // out [N, M] bf16 = x [N, K] @ w [M, K]^T  (decode shapes, N <= 256)
   CHECK_BF16_CONTIG(out);
@@ -351,6 +371,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu", &gelu, "erf GELU");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");
   m.def("gumbel_sample", &gumbel_sample, "Gumbel-max temperature sampling");
+  m.def("topk_topp_sample", &topk_topp_sample,
+        "fused top-k/top-p + Gumbel-max sampler (radix-histogram select)");
   m.def("skinny_gemm", &skinny_gemm, "decode GEMM (N<=256, MFMA streaming)");
   m.def("mfma_probe", &mfma_probe, "MFMA layout probe (tests)");
 }

@@ -296,4 +296,17 @@ hipError_t ks_gumbel_sample(void* out, const void* logits,
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }
+
+hipError_t ks_topk_topp_sample(void* out, const void* logits,
+                               const void* temperatures, const void* top_p,
+                               const void* top_k, const void* seeds,
+                               int num_seqs, int vocab, hipStream_t stream) {
+  if (num_seqs == 0) return hipSuccess;
+  hipLaunchKernelGGL(topk_topp_sample_kernel, dim3(num_seqs), dim3(256), 0,
+                     stream, (long*)out, (const short*)logits,
+                     (const float*)temperatures, (const float*)top_p,
+                     (const int*)top_k, (const long*)seeds, vocab);
+  HIP_CHECK_KERNEL();
+  return hipSuccess;
+}
 }

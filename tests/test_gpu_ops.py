@@ -265,6 +265,85 @@ class TestSampling:
         assert (out1 == out2).all()
 
 
+    def test_topk_support_and_distribution(self, dev):
+        """top-k sampler: support is EXACTLY the top-k set; frequencies match
+        the renormalized softmax."""
+        import kserve_amd_C
+
+        vocab = 1024
+        torch.manual_seed(3)
+        logits_row = torch.randn(vocab, dtype=torch.bfloat16)
+        k = 8
+        topk_idx = set(logits_row.float().topk(k).indices.tolist())
+        n = 8192
+        logits = logits_row.repeat(n, 1).to(dev)
+        temps = torch.ones(n, dtype=torch.float32, device=dev)
+        top_p = torch.ones(n, dtype=torch.float32, device=dev)
+        top_k = torch.full((n,), k, dtype=torch.int32, device=dev)
+        seeds = torch.arange(n, dtype=torch.int64, device=dev) * 104729
+        out = torch.empty(n, dtype=torch.int64, device=dev)
+        kserve_amd_C.topk_topp_sample(out, logits, temps, top_p, top_k, seeds)
+        torch.cuda.synchronize()
+        got = out.cpu()
+        assert set(got.tolist()) <= topk_idx, set(got.tolist()) - topk_idx
+        # renormalized probabilities over the top-k set
+        idx = torch.tensor(sorted(topk_idx))
+        p_ref = torch.softmax(logits_row.float()[idx], dim=-1)
+        counts = torch.zeros(len(idx))
+        for j, i in enumerate(idx.tolist()):
+            counts[j] = (got == i).sum()
+        assert torch.allclose(counts / n, p_ref, atol=0.04), (counts / n, p_ref)
+
+    def test_topp_support(self, dev):
+        """top-p sampler: support equals the sorted-cumsum top-p set."""
+        import kserve_amd_C
+
+        vocab = 4096
+        torch.manual_seed(7)
+        logits_row = (torch.randn(vocab) * 3).to(torch.bfloat16)
+        p = 0.85
+        # reference top-p set: sorted desc, include until cumsum >= p
+        pr = torch.softmax(logits_row.float(), dim=-1)
+        vals, order = pr.sort(descending=True)
+        keep = int((vals.cumsum(0) < p).sum()) + 1
+        allowed = set(order[:keep].tolist())
+        n = 8192
+        logits = logits_row.repeat(n, 1).to(dev)
+        temps = torch.ones(n, dtype=torch.float32, device=dev)
+        top_p = torch.full((n,), p, dtype=torch.float32, device=dev)
+        top_k = torch.full((n,), -1, dtype=torch.int32, device=dev)
+        seeds = torch.arange(n, dtype=torch.int64, device=dev) * 7919 + 13
+        out = torch.empty(n, dtype=torch.int64, device=dev)
+        kserve_amd_C.topk_topp_sample(out, logits, temps, top_p, top_k, seeds)
+        torch.cuda.synchronize()
+        got = set(out.cpu().tolist())
+        # bf16 ties at the boundary value may widen the set by equal-valued
+        # tokens; anything sampled outside must have the boundary value
+        boundary = logits_row[order[keep - 1]]
+        extra = got - allowed
+        for e in extra:
+            assert logits_row[e] == boundary, (e, float(logits_row[e]))
+
+    def test_topk_topp_determinism(self, dev):
+        import kserve_amd_C
+
+        torch.manual_seed(11)
+        logits = torch.randn(32, 128256, dtype=torch.bfloat16, device=dev)
+        temps = torch.full((32,), 0.7, dtype=torch.float32, device=dev)
+        # top_p left at 1.0: the mass histogram uses float atomics, whose
+        # ordering can flip an exact-boundary token between runs; the top-k
+        # threshold is integer-exact and must be bit-deterministic
+        top_p = torch.ones(32, dtype=torch.float32, device=dev)
+        top_k = torch.full((32,), 50, dtype=torch.int32, device=dev)
+        seeds = torch.arange(32, dtype=torch.int64, device=dev)
+        out1 = torch.empty(32, dtype=torch.int64, device=dev)
+        out2 = torch.empty(32, dtype=torch.int64, device=dev)
+        kserve_amd_C.topk_topp_sample(out1, logits, temps, top_p, top_k, seeds)
+        kserve_amd_C.topk_topp_sample(out2, logits, temps, top_p, top_k, seeds)
+        torch.cuda.synchronize()
+        assert (out1 == out2).all()
+
+
 class TestSkinnyGemm:
     @pytest.mark.parametrize(
         "N,K,M",
