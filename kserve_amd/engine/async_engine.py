@@ -25,9 +25,10 @@ _STOP = object()
 
 
 class AsyncLLMEngine:
-    def __init__(self, config: EngineConfig, tokenizer=None):
+    def __init__(self, config: EngineConfig, tokenizer=None, lora_modules=None):
         self.config = config
         self.tokenizer = tokenizer
+        self.lora_modules = dict(lora_modules or {})  # name -> adapter path
         self.engine: Optional[LLMEngine] = None
         self._submit_q: "queue.Queue" = queue.Queue()
         self._streams: Dict[str, asyncio.Queue] = {}
@@ -51,6 +52,8 @@ class AsyncLLMEngine:
 
     def _build(self):
         self.engine = LLMEngine(self.config, tokenizer=self.tokenizer)
+        for name, path in self.lora_modules.items():
+            self.engine.register_lora(name, path)
 
     def stop(self):
         self._stopping = True

@@ -152,6 +152,7 @@ class LLMEngine:
             num_cpu_blocks=self.runner.num_cpu_blocks,
         )
         self.sampler = Sampler(self.device, config.seed)
+        self.lora_manager = None
         self.eos_token_id = config.eos_token_id
         if tokenizer is not None and getattr(tokenizer, "eos_token_id", None) is not None:
             self.eos_token_id = tokenizer.eos_token_id
@@ -199,9 +200,29 @@ class LLMEngine:
             sampling_params,
             eos_token_id=self.eos_token_id,
         )
+        if sampling_params.lora_name:
+            if self.lora_manager is None:
+                raise ValueError("no LoRA adapters registered")
+            req.lora_id = self.lora_manager.lookup(sampling_params.lora_name)
         LLM_PROMPT_TOKENS.inc(len(prompt_token_ids))
         self.scheduler.add_request(req)
         return request_id
+
+    def register_lora(self, name: str, path: str) -> int:
+        """Load a PEFT adapter and make it addressable by name (reference:
+        --lora-modules name=path registering adapter model names)."""
+        if self.lora_manager is None:
+            from kserve_amd.engine.lora import LoRAManager
+
+            st = comm.get_state()
+            self.lora_manager = LoRAManager(
+                str(self.device),
+                self.model.dtype,
+                tp_rank=st.tp_rank,
+                tp_size=st.tp_size,
+            )
+            self.runner.lora_manager = self.lora_manager
+        return self.lora_manager.register(name, path)
 
     def abort_request(self, request_id: str):
         self.scheduler.abort_request(request_id)
@@ -231,6 +252,7 @@ class LLMEngine:
                 and r.sampling_params.presence_penalty == 0.0
                 and r.sampling_params.frequency_penalty == 0.0
                 and r.sampling_params.repetition_penalty == 1.0
+                and r.lora_id == 0
                 for r in batch.requests
             ):
                 k = self.scheduler.reserve_decode_window(

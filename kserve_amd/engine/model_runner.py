@@ -55,6 +55,8 @@ class ModelRunner:
         self._swap_stream = (
             torch.cuda.Stream() if self.is_cuda else None
         )
+        # multi-LoRA (set by LLMEngine.register_lora)
+        self.lora_manager = None
 
     # -- KV cache -----------------------------------------------------------
     def profile_and_allocate_kv(self) -> int:
@@ -324,21 +326,52 @@ class ModelRunner:
         pos = pin["positions"][:n].to(dev, non_blocking=True)
         return input_ids, pos, meta
 
+    def _lora_meta(self, batch: ScheduledBatch, per_request_rows: bool):
+        """Contiguous (adapter_id, start_row, end_row) segments for the batch.
+        Prefill rows span each request's scheduled tokens; decode rows are one
+        per request."""
+        from kserve_amd.engine.lora import LoRABatchMeta
+
+        segments = []
+        row = 0
+        cur_id, cur_start = None, 0
+        widths = (
+            [1] * len(batch.requests)
+            if per_request_rows
+            else batch.num_scheduled_tokens
+        )
+        for req, w in zip(batch.requests, widths):
+            if req.lora_id != cur_id:
+                if cur_id:
+                    segments.append((cur_id, cur_start, row))
+                cur_id, cur_start = req.lora_id, row
+            row += w
+        if cur_id:
+            segments.append((cur_id, cur_start, row))
+        return LoRABatchMeta(segments, self.lora_manager)
+
     # -- execution ----------------------------------------------------------
     @torch.no_grad()
     def execute_prefill(self, batch, block_manager) -> torch.Tensor:
         input_ids, positions, meta, sel = self.prepare_prefill(batch, block_manager)
+        if self.lora_manager is not None and any(
+            r.lora_id for r in batch.requests
+        ):
+            meta.lora = self._lora_meta(batch, per_request_rows=False)
         hidden = self.model(input_ids, positions, self.kv_caches, meta)
         return self.model.compute_logits(hidden[sel])
 
     @torch.no_grad()
     def execute_decode(self, batch, block_manager) -> torch.Tensor:
         n = len(batch.requests)
-        if self._graphs and not self.config.enforce_eager:
+        has_lora = any(r.lora_id for r in batch.requests)
+        if self._graphs and not self.config.enforce_eager and not has_lora:
             bucket = self._graph_bucket(n)
             if bucket is not None:
                 return self._run_graph(bucket, batch, block_manager)
         input_ids, positions, meta = self.prepare_decode(batch, block_manager)
+        if has_lora:
+            meta.lora = self._lora_meta(batch, per_request_rows=True)
         hidden = self.model(input_ids, positions, self.kv_caches, meta)
         return self.model.compute_logits(hidden)
 

@@ -61,6 +61,8 @@ class Request:
         self.block_table: List[int] = []
         # number of prompt tokens whose KV is already computed (chunked prefill)
         self.num_computed_tokens = 0
+        # multi-LoRA adapter id (0 = base model)
+        self.lora_id = 0
         # metrics
         self.first_token_time: Optional[float] = None
         self.finish_time: Optional[float] = None

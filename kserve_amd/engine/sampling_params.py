@@ -27,6 +27,8 @@ class SamplingParams:
     frequency_penalty: float = 0.0
     repetition_penalty: float = 1.0
     echo: bool = False
+    # multi-LoRA: adapter name registered with the engine (None = base)
+    lora_name: Optional[str] = None
 
     def __post_init__(self):
         if self.temperature < 0:

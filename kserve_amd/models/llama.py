@@ -44,11 +44,14 @@ class AttentionMetadata:
     # the paged context the chunk attends to (context_lens includes the chunk)
     block_tables: Optional[torch.Tensor] = None  # [num_seqs, max_blocks] int32
     context_lens: Optional[torch.Tensor] = None  # [num_seqs] int32
+    # multi-LoRA: per-batch adapter segments (engine/lora.LoRABatchMeta)
+    lora: Optional[object] = None
 
 
 class LlamaAttention(nn.Module):
-    def __init__(self, config: ModelConfig, dtype: torch.dtype):
+    def __init__(self, config: ModelConfig, dtype: torch.dtype, layer_idx: int = 0):
         super().__init__()
+        self.layer_idx = layer_idx
         st = comm.get_state()
         self.head_dim = config.head_dim
         self.scale = 1.0 / math.sqrt(config.head_dim)
@@ -79,6 +82,10 @@ class LlamaAttention(nn.Module):
     ) -> torch.Tensor:
         T = hidden.shape[0]
         q, k, v = self.qkv_proj(hidden)
+        if meta.lora is not None:
+            meta.lora.apply(self.layer_idx, "q_proj", hidden, q)
+            meta.lora.apply(self.layer_idx, "k_proj", hidden, k)
+            meta.lora.apply(self.layer_idx, "v_proj", hidden, v)
         # strided views into the fused qkv output; kernels take row strides
         q = q.view(T, self.num_heads_local, self.head_dim)
         k = k.view(T, self.num_kv_heads_local, self.head_dim)
@@ -103,12 +110,19 @@ class LlamaAttention(nn.Module):
             out = ops.paged_attention_decode(
                 q, k_cache, v_cache, meta.block_tables, meta.context_lens, self.scale
             )
-        return self.o_proj(out.reshape(T, -1))
+        o_in = out.reshape(T, -1)
+        o_delta = (
+            meta.lora.delta_for(self.layer_idx, "o_proj", o_in)
+            if meta.lora is not None
+            else None
+        )
+        return self.o_proj(o_in, o_delta)
 
 
 class LlamaMLP(nn.Module):
-    def __init__(self, config: ModelConfig, dtype: torch.dtype):
+    def __init__(self, config: ModelConfig, dtype: torch.dtype, layer_idx: int = 0):
         super().__init__()
+        self.layer_idx = layer_idx
         self.gate_up_proj = MergedColumnParallelLinear(
             config.hidden_size,
             config.intermediate_size,
@@ -119,15 +133,27 @@ class LlamaMLP(nn.Module):
             config.intermediate_size, config.hidden_size, bias=False, dtype=dtype
         )
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
+    def forward(self, x: torch.Tensor, meta=None) -> torch.Tensor:
+        gate_up = self.gate_up_proj(x)
+        lora = meta.lora if meta is not None else None
+        if lora is not None:
+            half = gate_up.shape[-1] // 2
+            lora.apply(self.layer_idx, "gate_proj", x, gate_up, 0)
+            lora.apply(self.layer_idx, "up_proj", x, gate_up, half)
+        act = ops.silu_and_mul(gate_up)
+        d_delta = (
+            lora.delta_for(self.layer_idx, "down_proj", act)
+            if lora is not None
+            else None
+        )
+        return self.down_proj(act, d_delta)
 
 
 class LlamaDecoderLayer(nn.Module):
-    def __init__(self, config: ModelConfig, dtype: torch.dtype):
+    def __init__(self, config: ModelConfig, dtype: torch.dtype, layer_idx: int = 0):
         super().__init__()
-        self.self_attn = LlamaAttention(config, dtype)
-        self.mlp = LlamaMLP(config, dtype)
+        self.self_attn = LlamaAttention(config, dtype, layer_idx)
+        self.mlp = LlamaMLP(config, dtype, layer_idx)
         self.input_layernorm = nn.Parameter(
             torch.empty(config.hidden_size, dtype=dtype), requires_grad=False
         )
@@ -148,7 +174,7 @@ class LlamaDecoderLayer(nn.Module):
         hidden, residual = ops.fused_add_rms_norm(
             hidden, residual, self.post_attention_layernorm, self.eps
         )
-        hidden = self.mlp(hidden)
+        hidden = self.mlp(hidden, meta)
         return hidden, residual
 
 
@@ -169,7 +195,7 @@ class LlamaForCausalLM(nn.Module):
             config.vocab_size, config.hidden_size, dtype=dtype
         )
         self.layers = nn.ModuleList(
-            [LlamaDecoderLayer(config, dtype) for _ in range(config.num_layers)]
+            [LlamaDecoderLayer(config, dtype, i) for i in range(config.num_layers)]
         )
         self.norm = nn.Parameter(
             torch.empty(config.hidden_size, dtype=dtype), requires_grad=False

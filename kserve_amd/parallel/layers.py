@@ -168,8 +168,14 @@ class RowParallelLinear(nn.Module):
             else None
         )
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(
+        self, x: torch.Tensor, delta: Optional[torch.Tensor] = None
+    ) -> torch.Tensor:
         y = ops.linear(x, self.weight)
+        if delta is not None:
+            # per-rank partial (e.g. LoRA with A column-sharded): must be
+            # summed by the same all-reduce as the main GEMM
+            y = y + delta
         if self.reduce_output:
             y = comm.tp_all_reduce(y)
         if self.bias is not None:

@@ -41,7 +41,11 @@ def create_error_response(message: str, status_code: int = 400) -> JSONResponse:
 class OpenAIEndpoints:
     def __init__(self, dataplane: DataPlane, models: List[OpenAIModel]):
         self.dataplane = dataplane
-        self._models: Dict[str, OpenAIModel] = {m.name: m for m in models}
+        # a model may serve several ids (base name + LoRA adapter names)
+        self._models: Dict[str, OpenAIModel] = {}
+        for m in models:
+            for name in getattr(m, "served_names", [m.name]):
+                self._models[name] = m
 
     def _get_model(self, name: str) -> OpenAIModel:
         model = self._models.get(name)

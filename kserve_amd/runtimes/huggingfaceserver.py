@@ -82,7 +82,16 @@ def build_model(args):
             model_path=args.model_dir,
             device="cuda" if torch.cuda.is_available() else "cpu",
         )
-        return LLMModel(args.model_name, cfg, tokenizer=tokenizer)
+        lora_modules = {}
+        if getattr(args, "enable_lora", False) or getattr(args, "lora_modules", None):
+            for spec in getattr(args, "lora_modules", []) or []:
+                name, _, path = spec.partition("=")
+                if not path:
+                    raise ValueError(f"--lora-modules expects name=path, got {spec!r}")
+                lora_modules[name] = path
+        return LLMModel(
+            args.model_name, cfg, tokenizer=tokenizer, lora_modules=lora_modules
+        )
     from kserve_amd.runtimes.encoder_model import EncoderModel
 
     model = EncoderModel(
@@ -111,6 +120,12 @@ def main(argv=None):
     parser.add_argument("--enable-expert-parallel", dest="expert_parallel", action="store_true")
     # host-DRAM KV offload tier (LLMInferenceService KVCacheOffloadingSpec)
     parser.add_argument("--kv-offload-bytes", dest="kv_offload_bytes", type=int, default=0)
+    # LoRA adapter serving (reference: --enable-lora --lora-modules name=path)
+    parser.add_argument("--enable-lora", dest="enable_lora", action="store_true")
+    parser.add_argument(
+        "--lora-modules", dest="lora_modules", nargs="*", default=[],
+        help="name=path pairs; each name becomes a served model id",
+    )
     args = parser.parse_args(argv)
     configure_logging()
     model = build_model(args)

@@ -66,11 +66,25 @@ class LLMModel(OpenAIModel):
         name: str,
         engine_config: EngineConfig,
         tokenizer=None,
+        lora_modules=None,
     ):
         super().__init__(name)
         self.engine = True  # ModelServer awaits start_engine()
         self.tokenizer = tokenizer
-        self.async_engine = AsyncLLMEngine(engine_config, tokenizer=tokenizer)
+        # name -> adapter dir; each name is served as its own model id
+        # (reference registers --lora-modules names with the model server)
+        self.lora_modules = dict(lora_modules or {})
+        self.async_engine = AsyncLLMEngine(
+            engine_config, tokenizer=tokenizer, lora_modules=self.lora_modules
+        )
+
+    @property
+    def served_names(self):
+        # base + registered LoRA adapter names
+        return [self.name, *self.lora_modules]
+
+    def _lora_for(self, requested_model):
+        return requested_model if requested_model in self.lora_modules else None
 
     async def start_engine(self):
         await self.async_engine.start()
@@ -127,6 +141,7 @@ class LLMModel(OpenAIModel):
         else:
             raise InvalidInput("Empty prompt")
         sp = _to_sampling_params(request)
+        sp.lora_name = self._lora_for(request.model)
         if request.stream:
             if len(prompt_list) != 1:
                 raise InvalidInput("Streaming supports a single prompt")
@@ -223,6 +238,7 @@ class LLMModel(OpenAIModel):
     ):
         ids = self._chat_to_prompt(request)
         sp = _to_sampling_params(request, max_tokens_default=256)
+        sp.lora_name = self._lora_for(request.model)
         if request.stream:
             return self._stream_chat(ids, sp, request)
         out = await self.async_engine.generate_full(ids, sp)
