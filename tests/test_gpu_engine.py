@@ -295,3 +295,26 @@ def test_chunked_prefill_on_gpu():
         # bf16 GEMM rounding differs with row-batch shape; the first token of
         # each completion is far from any tie for random weights and must match
         assert ta[0] == tb[0], (ta, tb)
+
+
+def test_lora_on_gpu(tmp_path):
+    """LoRA requests run the eager decode path on GPU and match the
+    merged-weight engine's first token."""
+    from tests.test_lora import make_adapter_dir
+    from kserve_amd.engine.engine import LLMEngine
+    from kserve_amd.engine.sampling_params import SamplingParams
+
+    cfg = _cfg(enforce_eager=True)
+    torch.manual_seed(0)
+    engine = LLMEngine(cfg)
+    path, _ = make_adapter_dir(tmp_path, cfg.model)
+    engine.register_lora("adapt", path)
+    prompts = [[1, 2, 3, 4, 5]]
+    base = list(engine.generate(prompts, SamplingParams(temperature=0.0, max_tokens=6)).values())[0]
+    lora = list(
+        engine.generate(
+            prompts, SamplingParams(temperature=0.0, max_tokens=6, lora_name="adapt")
+        ).values()
+    )[0]
+    assert len(lora.output_token_ids) == 6
+    assert base.output_token_ids != lora.output_token_ids
