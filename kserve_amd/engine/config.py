@@ -160,6 +160,13 @@ class SchedulerConfig:
     # hipGraph replays (sampled token fed back on-GPU); stop conditions are
     # applied after the window. 1 disables.
     multi_step: int = 8
+    # speculative decoding (prompt-lookup/n-gram): max draft tokens verified
+    # per step through the paged-context prefill path. 0 disables. When >0 it
+    # replaces multi-step windows for greedy batches (exact same outputs —
+    # rejected drafts are corrected by the verify forward).
+    speculative_ngram: int = 0
+    speculative_ngram_min: int = 2
+    speculative_ngram_max: int = 3
 
 
 @dataclass

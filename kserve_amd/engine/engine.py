@@ -23,6 +23,7 @@ from kserve_amd.engine.scheduler import Scheduler
 from kserve_amd.logging import logger
 from kserve_amd.metrics import (
     LLM_E2E_HIST,
+    LLM_SPEC_ACCEPTED,
     LLM_GENERATION_TOKENS,
     LLM_KV_USAGE,
     LLM_PROMPT_TOKENS,
@@ -243,6 +244,32 @@ class LLMEngine:
         if batch.is_prefill:
             logits = self.runner.execute_prefill(batch, self.scheduler.block_manager)
         else:
+            sched = self.config.scheduler
+            pure_greedy = all(
+                r.sampling_params.greedy
+                and r.sampling_params.logprobs is None
+                and r.sampling_params.presence_penalty == 0.0
+                and r.sampling_params.frequency_penalty == 0.0
+                and r.sampling_params.repetition_penalty == 1.0
+                and r.lora_id == 0
+                for r in batch.requests
+            )
+            # speculative decoding (prompt-lookup): draft from the request's
+            # own context, verify k+1 positions in ONE forward through the
+            # paged-context prefill path; exact greedy outputs by construction
+            if sched.speculative_ngram > 0 and pure_greedy:
+                k_cap = (
+                    self.scheduler.reserve_decode_window(
+                        batch, sched.speculative_ngram + 1
+                    )
+                    - 1
+                )
+                if k_cap > 0:
+                    drafts = [
+                        self._draft_ngram(r, k_cap) for r in batch.requests
+                    ]
+                    if any(drafts):
+                        return self._run_spec_decode(batch, drafts)
             # multi-step window: all-greedy decode with no pending scheduling
             # events runs as back-to-back hipGraph replays
             k = 1
@@ -350,6 +377,99 @@ class LLMEngine:
         for req in finished:
             self.runner.release_request(req.request_id)
         LLM_KV_USAGE.set(self.scheduler.block_manager.usage)
+        return outputs
+
+    def _draft_ngram(self, req: Request, k: int) -> List[int]:
+        """Prompt-lookup drafting: if the last n-gram (n in [min,max]) occurred
+        earlier in the sequence, propose the tokens that followed it."""
+        sched = self.config.scheduler
+        toks = req.all_token_ids
+        L = len(toks)
+        for n in range(sched.speculative_ngram_max, sched.speculative_ngram_min - 1, -1):
+            if L <= n:
+                continue
+            suffix = toks[-n:]
+            # most recent earlier occurrence wins
+            for i in range(L - n - 1, -1, -1):
+                if toks[i : i + n] == suffix:
+                    cont = toks[i + n : i + n + k]
+                    if cont:
+                        return list(cont)
+                    break
+        return []
+
+    def _run_spec_decode(self, batch, drafts) -> List[RequestOutput]:
+        logits, cu = self.runner.execute_verify(
+            batch, drafts, self.scheduler.block_manager
+        )
+        sampled = ops.greedy_sample(logits).tolist()
+        outputs: List[RequestOutput] = []
+        finished: List[Request] = []
+        now = time.monotonic()
+        max_len = self.config.scheduler.max_model_len
+        n_tokens = 0
+        n_accepted = 0
+        for i, req in enumerate(batch.requests):
+            row = cu[i]
+            d = drafts[i]
+            # emit the model's token at each verified position while the
+            # draft agrees; the first disagreement is the model's correction
+            emitted: List[int] = []
+            for j in range(len(d) + 1):
+                t = sampled[row + j]
+                emitted.append(t)
+                if j < len(d) and t == d[j]:
+                    continue
+                break
+            n_accepted += len(emitted) - 1
+            sp = req.sampling_params
+            if req.first_token_time is None:
+                req.first_token_time = now
+                LLM_TTFT_HIST.observe(now - req.arrival_time)
+            stop_at = len(emitted)
+            if not sp.ignore_eos or sp.stop_token_ids:
+                stops = set(sp.stop_token_ids)
+                if not sp.ignore_eos and req.eos_token_id is not None:
+                    stops.add(req.eos_token_id)
+                for j, t in enumerate(emitted):
+                    if t in stops and j + 1 >= sp.min_tokens - req.num_output_tokens:
+                        stop_at = j + 1
+                        break
+            emitted = emitted[:stop_at]
+            req.output_token_ids.extend(emitted)
+            req.num_computed_tokens += len(emitted)
+            req.maybe_finish(max_len)
+            delta = self.detokenizer.decode_new(req)
+            if not req.is_finished and sp.stop:
+                if self.detokenizer.check_stop_strings(req) is not None:
+                    req.status = RequestStatus.FINISHED_STOPPED
+                    req.is_finished = True
+                    req.finish_time = now
+            if req.is_finished:
+                finished.append(req)
+                LLM_E2E_HIST.observe(now - req.arrival_time)
+            n_tokens += len(emitted)
+            outputs.append(
+                RequestOutput(
+                    request_id=req.request_id,
+                    new_token_ids=emitted,
+                    finished=req.is_finished,
+                    finish_reason=req.finish_reason,
+                    output_token_ids=(
+                        list(req.output_token_ids)
+                        if req.is_finished
+                        else req.output_token_ids
+                    ),
+                    num_prompt_tokens=req.num_prompt_tokens,
+                    text_delta=delta,
+                    output_text=req.output_text,
+                )
+            )
+        LLM_GENERATION_TOKENS.inc(n_tokens)
+        LLM_SPEC_ACCEPTED.inc(n_accepted)
+        self.scheduler.finish_requests(finished)
+        for req in finished:
+            self.runner.release_request(req.request_id)
         return outputs
 
     def _run_decode_window(self, batch, k: int) -> List[RequestOutput]:

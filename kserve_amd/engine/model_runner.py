@@ -375,6 +375,58 @@ class ModelRunner:
         hidden = self.model(input_ids, positions, self.kv_caches, meta)
         return self.model.compute_logits(hidden)
 
+    @torch.no_grad()
+    def execute_verify(self, batch, drafts, block_manager):
+        """Speculative-decode verify forward: per request feed
+        [last_token, draft_1..draft_k] as a chunk attending to the paged
+        context (same math as chunked prefill), returning logits for EVERY
+        row plus the per-request row offsets (cu)."""
+        import numpy as np
+
+        widths = [1 + len(d) for d in drafts]
+        total = sum(widths)
+        bs = self.config.cache.block_size
+        tokens = np.empty(total, dtype=np.int64)
+        positions = np.empty(total, dtype=np.int64)
+        slots = np.empty(total, dtype=np.int32)
+        cu = [0]
+        nb_per = []
+        for req, d, w in zip(batch.requests, drafts, widths):
+            pos0 = req.num_computed_tokens
+            nb_per.append(-(-(pos0 + w) // bs))
+        max_nb = max(nb_per)
+        btab = np.zeros((len(batch.requests), max_nb), dtype=np.int32)
+        ctx = np.empty(len(batch.requests), dtype=np.int32)
+        off = 0
+        for i, (req, d, w) in enumerate(zip(batch.requests, drafts, widths)):
+            pos0 = req.num_computed_tokens
+            tokens[off] = req.all_token_ids[pos0]
+            tokens[off + 1 : off + w] = d
+            pr = np.arange(pos0, pos0 + w, dtype=np.int64)
+            positions[off : off + w] = pr
+            t = block_manager.get_block_table(req)
+            bt = np.asarray(t, dtype=np.int32)
+            slots[off : off + w] = bt[pr // bs] * bs + (pr % bs).astype(np.int32)
+            btab[i, : nb_per[i]] = bt[: nb_per[i]]
+            ctx[i] = pos0 + w
+            off += w
+            cu.append(off)
+        dev = self.device
+        from kserve_amd.models.llama import AttentionMetadata
+
+        meta = AttentionMetadata(
+            is_prefill=True,
+            slot_mapping=torch.from_numpy(slots).to(dev, non_blocking=True),
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            max_seqlen=max(widths),
+            block_tables=torch.from_numpy(btab).to(dev, non_blocking=True),
+            context_lens=torch.from_numpy(ctx).to(dev, non_blocking=True),
+        )
+        input_ids = torch.from_numpy(tokens).to(dev, non_blocking=True)
+        pos = torch.from_numpy(positions).to(dev, non_blocking=True)
+        hidden = self.model(input_ids, pos, self.kv_caches, meta)
+        return self.model.compute_logits(hidden), cu
+
     # -- hipGraph capture ------------------------------------------------------
     def capture_decode_graphs(self, batch_sizes: Optional[List[int]] = None):
         """Capture the decode forward into hipGraphs per batch-size bucket.

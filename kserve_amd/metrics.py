@@ -26,6 +26,10 @@ LLM_GENERATION_TOKENS = Counter(
     "llm_generation_tokens_total", "output tokens generated"
 )
 LLM_PROMPT_TOKENS = Counter("llm_prompt_tokens_total", "prompt tokens processed")
+LLM_SPEC_ACCEPTED = Counter(
+    "llm_spec_decode_accepted_tokens_total",
+    "draft tokens accepted by speculative decoding",
+)
 LLM_TTFT_HIST = Histogram(
     "llm_time_to_first_token_seconds",
     "time to first token",

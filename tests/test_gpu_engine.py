@@ -318,3 +318,31 @@ def test_lora_on_gpu(tmp_path):
     )[0]
     assert len(lora.output_token_ids) == 6
     assert base.output_token_ids != lora.output_token_ids
+
+
+def test_spec_decode_on_gpu():
+    """Spec decode verify runs the paged-context kernel on GPU; greedy
+    outputs must match the plain hipGraph decode engine exactly (both paths
+    sample argmax from the same bf16 logits)."""
+    from kserve_amd.engine.config import SchedulerConfig
+    from kserve_amd.engine.engine import LLMEngine
+    from kserve_amd.engine.sampling_params import SamplingParams
+
+    prompts = [[1, 2, 3, 4, 1, 2, 3, 4, 1, 2], [7, 8, 9]]
+    sp = SamplingParams(temperature=0.0, max_tokens=20)
+
+    torch.manual_seed(0)
+    plain = LLMEngine(_cfg(enforce_eager=True))
+    a = [o.output_token_ids for o in plain.generate(prompts, sp).values()]
+    del plain
+    torch.cuda.empty_cache()
+
+    cfg = _cfg(enforce_eager=True)
+    cfg.scheduler = SchedulerConfig(
+        max_num_seqs=8, max_num_batched_tokens=2048, max_model_len=512,
+        speculative_ngram=4,
+    )
+    torch.manual_seed(0)
+    spec = LLMEngine(cfg)
+    b = [o.output_token_ids for o in spec.generate(prompts, sp).values()]
+    assert a == b
