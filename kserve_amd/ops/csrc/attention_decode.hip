@@ -24,8 +24,10 @@ constexpr int NWAVES = 4;     // waves per workgroup
 
 // D: head dim (64 or 128). ACC = D/64 output dims per lane.
 // HPW: q heads per wave (GQA group = NWAVES*HPW covered per workgroup).
-template <int D, int HPW>
-__global__ __launch_bounds__(256) void paged_attention_kernel(
+// OCC: min waves/EU hint (occupancy ablation KS_ATTN_OCC; 1 = compiler's
+// choice, 127 VGPRs -> 4 waves/SIMD on gfx950)
+template <int D, int HPW, int OCC = 1>
+__global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     short* __restrict__ out,            // [S, H, D] bf16
     const short* __restrict__ q,        // [S, H, D]
     const short* __restrict__ k_cache,  // [B, Hkv, PAGE, D]
@@ -695,6 +697,41 @@ extern "C" hipError_t ks_paged_attention_decode(
                        (const int*)block_tables, (const int*)context_lens,
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
                        n_splits, (float*)part_out, (float*)part_ml);
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      const long sh = (long)num_seqs * num_heads;
+      int wpb = 4;
+      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(wpb * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return hipSuccess;
+  }
+  // occupancy ablation: KS_ATTN_OCC=5/6 forces a tighter VGPR budget so
+  // more waves are resident to hide the scattered-page load latency
+  static const int occ = [] {
+    const char* e = getenv("KS_ATTN_OCC");
+    return e ? atoi(e) : 0;
+  }();
+  if (occ >= 5 && hpw == 1 && head_dim == 128) {
+    if (occ >= 6) {
+      hipLaunchKernelGGL((paged_attention_kernel<128, 1, 6>), grid, block, 0,
+                         stream, (short*)out, (const short*)q,
+                         (const short*)k_cache, (const short*)v_cache,
+                         (const int*)block_tables, (const int*)context_lens,
+                         scale, num_kv_heads, group, max_blocks, q_row_stride,
+                         n_splits, (float*)part_out, (float*)part_ml);
+    } else {
+      hipLaunchKernelGGL((paged_attention_kernel<128, 1, 5>), grid, block, 0,
+                         stream, (short*)out, (const short*)q,
+                         (const short*)k_cache, (const short*)v_cache,
+                         (const int*)block_tables, (const int*)context_lens,
+                         scale, num_kv_heads, group, max_blocks, q_row_stride,
+                         n_splits, (float*)part_out, (float*)part_ml);
+    }
     HIP_CHECK_KERNEL();
     if (n_splits > 1) {
       const long sh = (long)num_seqs * num_heads;
