@@ -45,6 +45,10 @@ def main():
     parser.add_argument("--eager", action="store_true",
                         help="disable hipGraph capture")
     parser.add_argument("--temperature", type=float, default=0.0)
+    parser.add_argument("--kv-cache-dtype", type=str, default="auto",
+                        choices=["auto", "fp8"],
+                        help="opt-in fp8 E4M3 KV cache (NOT the headline "
+                             "config; bf16 compute is unchanged)")
     parser.add_argument("--profile-cpu", action="store_true",
                         help="cProfile the timed loop and print hot functions")
     args = parser.parse_args()
@@ -87,6 +91,7 @@ def main():
             block_size=16,
             num_gpu_blocks=args.kv_blocks or None,
             gpu_memory_utilization=0.9,
+            kv_cache_dtype=args.kv_cache_dtype,
         ),
         scheduler=SchedulerConfig(
             max_num_seqs=args.concurrency,

@@ -128,6 +128,17 @@ class CacheConfig:
     # host-DRAM offload tier (pinned memory, hipMemcpyAsync side stream)
     num_cpu_blocks: int = 0
     cpu_offload_bytes: int = 0
+    # "auto" stores KV in the model dtype (bf16). "fp8" stores OCP E4M3
+    # bytes (scale 1.0): half the KV bandwidth/capacity; attention math
+    # stays fp32/bf16 after in-register conversion. GQA group <= 4, D=128.
+    kv_cache_dtype: str = "auto"
+
+    def cache_torch_dtype(self, model_dtype):
+        if self.kv_cache_dtype in ("fp8", "fp8_e4m3"):
+            import torch
+
+            return torch.float8_e4m3fn
+        return model_dtype
 
     def derive_num_gpu_blocks(
         self, model: ModelConfig, tp_size: int = 1, weight_bytes: Optional[int] = None,

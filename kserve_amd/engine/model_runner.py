@@ -70,14 +70,17 @@ class ModelRunner:
             budget = int(
                 free_bytes * self.config.cache.gpu_memory_utilization
             ) - (2 << 30)
+            kv_itemsize = self.config.cache.cache_torch_dtype(dtype).itemsize
+            # kv_bytes_per_token_per_layer assumes bf16 (2 B); rescale
+            per_block_bytes = (
+                m.kv_bytes_per_token_per_layer
+                * kv_itemsize
+                * self.config.cache.block_size
+                * m.num_layers
+                // (tp * 2)
+            )
             num_blocks = self.config.cache.num_gpu_blocks or max(
-                budget
-                // (
-                    m.kv_bytes_per_token_per_layer
-                    * self.config.cache.block_size
-                    * m.num_layers
-                    // tp
-                ),
+                budget // per_block_bytes,
                 16,
             )
         else:
@@ -94,7 +97,7 @@ class ModelRunner:
             self.config.cache.block_size,
             m.head_dim,
         )
-        dtype = self.model.dtype
+        dtype = self.config.cache.cache_torch_dtype(self.model.dtype)
         # empty, not zeros: context_lens bound all reads, and zeroing a
         # 250 GB pool costs seconds of startup
         alloc = torch.empty if self.is_cuda else torch.zeros
@@ -111,6 +114,8 @@ class ModelRunner:
             2 * num_blocks * kv_heads_local * self.config.cache.block_size
             * m.head_dim * m.num_layers * dtype.itemsize
         ) / (1 << 30)
+        if dtype.itemsize == 1:
+            logger.info("KV cache dtype: fp8_e4m3 (scale 1.0)")
         logger.info(
             "KV cache: %d blocks x %d tokens (%.1f GiB, layout [blocks, kv_heads, block, head_dim])",
             num_blocks,
@@ -125,9 +130,10 @@ class ModelRunner:
         m = self.config.model
         cache = self.config.cache
         kv_heads_local = max(1, m.num_kv_heads // max(1, self._tp_size()))
+        kv_dtype = cache.cache_torch_dtype(self.model.dtype)
         per_block = (
             2 * kv_heads_local * cache.block_size * m.head_dim
-            * m.num_layers * self.model.dtype.itemsize
+            * m.num_layers * kv_dtype.itemsize
         )
         n = cache.num_cpu_blocks
         if not n and cache.cpu_offload_bytes:
@@ -138,8 +144,8 @@ class ModelRunner:
         pin = self.is_cuda
         self.cpu_kv_caches = [
             (
-                torch.empty(shape, dtype=self.model.dtype, pin_memory=pin),
-                torch.empty(shape, dtype=self.model.dtype, pin_memory=pin),
+                torch.empty(shape, dtype=kv_dtype, pin_memory=pin),
+                torch.empty(shape, dtype=kv_dtype, pin_memory=pin),
             )
             for _ in range(m.num_layers)
         ]

@@ -255,6 +255,139 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// fp8 (OCP E4M3) KV-cache decode attention: the HPW=1 fast path with byte
+// pages — half the KV bytes of bf16, converted pairwise in-register with
+// v_cvt_pk_f32_fp8. D=128 (ACC=2), GQA group <= 4. Split-context shares the
+// bf16 reduce kernel (partials are fp32 either way).
+// ---------------------------------------------------------------------------
+typedef unsigned int uint4_t __attribute__((ext_vector_type(4)));
+
+template <int D>  // D == 128
+__global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
+    short* __restrict__ out, const short* __restrict__ q,
+    const unsigned char* __restrict__ k_cache,
+    const unsigned char* __restrict__ v_cache,
+    const int* __restrict__ block_tables, const int* __restrict__ context_lens,
+    const float scale, const int num_kv_heads, const int group,
+    const int max_blocks, const long q_row_stride, const int n_splits,
+    float* __restrict__ part_out, float* __restrict__ part_ml) {
+  constexpr int ACC = D / 64;
+  constexpr int QFRAG = D / 4;
+  const int kv_head = blockIdx.x;
+  const int seq = blockIdx.y;
+  const int split = blockIdx.z;
+  const int ctx = context_lens[seq];
+  if (ctx <= 0) return;
+  const int num_heads = num_kv_heads * group;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tok = lane >> 2;
+  const int part = lane & 3;
+
+  const int nblocks = (ctx + PAGE - 1) / PAGE;
+  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
+  const int blk_lo = split * blocks_per_split;
+  const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
+  const int head = kv_head * group + wave;
+  const bool active = wave < group;
+  if (blk_lo >= blk_hi) {
+    if (n_splits > 1 && threadIdx.x < (unsigned)group) {
+      const int h = kv_head * group + threadIdx.x;
+      float* ml = part_ml + (((long)seq * num_heads + h) * n_splits + split) * 2;
+      ml[0] = NEG_INF;
+      ml[1] = 0.f;
+    }
+    return;
+  }
+  if (!active) return;
+
+  float q_frag[QFRAG];
+  {
+    const short* qp =
+        q + (long)seq * q_row_stride + (long)head * D + part * QFRAG;
+#pragma unroll
+    for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
+  }
+  float m = NEG_INF, l = 0.f;
+  float acc[ACC];
+#pragma unroll
+  for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
+
+  const int bt_base = (int)((long)seq * max_blocks);
+  for (int bi = blk_lo; bi < blk_hi; ++bi) {
+    const int block_id = block_tables[bt_base + bi];
+    const unsigned char* page =
+        k_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    const unsigned char* vpage =
+        v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+    const int gtok = bi * PAGE + tok;
+    const bool tok_valid = gtok < ctx;
+    // burst-issue: K quarter-row (32 B) + V dims (2 B x 16 tokens)
+    uint4_t kreg[QFRAG / 16];
+#pragma unroll
+    for (int c = 0; c < QFRAG / 16; ++c)
+      kreg[c] = reinterpret_cast<const uint4_t*>(
+          page + tok * D + part * QFRAG)[c];
+    unsigned short vreg[PAGE];
+#pragma unroll
+    for (int t = 0; t < PAGE; ++t)
+      vreg[t] = *reinterpret_cast<const unsigned short*>(
+          vpage + t * D + lane * ACC);
+    float s = 0.f;
+#pragma unroll
+    for (int c = 0; c < QFRAG / 16; ++c) {
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        const int ui = (int)kreg[c][w];
+        const float2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(ui, false);
+        const float2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(ui, true);
+        const int j = c * 16 + w * 4;
+        s += q_frag[j] * lo[0] + q_frag[j + 1] * lo[1] +
+             q_frag[j + 2] * hi[0] + q_frag[j + 3] * hi[1];
+      }
+    }
+    s = group_reduce_sum<4>(s);
+    s = tok_valid ? s * scale : NEG_INF;
+    const float tmax = wave_reduce_max(s);
+    const float m_new = fmaxf(m, tmax);
+    const float rescale = __expf(m - m_new);
+    const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+    const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+    l = l * rescale + psum;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) acc[a] *= rescale;
+    m = m_new;
+#pragma unroll
+    for (int t = 0; t < PAGE; ++t) {
+      const float pt = __shfl(p, t * 4, 64);
+      const float2_t v2 = __builtin_amdgcn_cvt_pk_f32_fp8((int)vreg[t], false);
+      acc[0] += pt * v2[0];
+      acc[1] += pt * v2[1];
+    }
+  }
+
+  if (n_splits == 1) {
+    const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
+    short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) op[a] = float_to_bf16_bits(acc[a] * inv_l);
+  } else {
+    float* po = part_out +
+                ((((long)seq * num_heads + head) * n_splits + split)) * D +
+                lane * ACC;
+#pragma unroll
+    for (int a = 0; a < ACC; ++a) po[a] = acc[a];
+    if (lane == 0) {
+      float* ml =
+          part_ml + (((long)seq * num_heads + head) * n_splits + split) * 2;
+      ml[0] = m;
+      ml[1] = l;
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // 2-page unrolled variant of the HPW=1 fast path: both pages' K+V bursts
 // (16 KB/wave) are issued before either page's compute, doubling the
@@ -606,6 +739,38 @@ __global__ void paged_attention_reduce_kernel(
 }
 
 }  // namespace
+
+extern "C" hipError_t ks_paged_attention_decode_fp8(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    const void* block_tables, const void* context_lens, float scale,
+    int num_seqs, int num_heads, int num_kv_heads, int head_dim,
+    int max_blocks, int block_size, long q_row_stride, int n_splits,
+    void* part_out, void* part_ml, hipStream_t stream) {
+  if (block_size != PAGE || head_dim != 128) return hipErrorInvalidValue;
+  const int group = num_heads / num_kv_heads;
+  if (group > NWAVES) return hipErrorInvalidValue;  // fp8: GQA group <= 4
+  if (n_splits < 1) n_splits = 1;
+  dim3 grid(num_kv_heads, num_seqs, n_splits);
+  hipLaunchKernelGGL((paged_attention_fp8_kernel<128>), grid, dim3(256), 0,
+                     stream, (short*)out, (const short*)q,
+                     (const unsigned char*)k_cache,
+                     (const unsigned char*)v_cache,
+                     (const int*)block_tables, (const int*)context_lens,
+                     scale, num_kv_heads, group, max_blocks, q_row_stride,
+                     n_splits, (float*)part_out, (float*)part_ml);
+  HIP_CHECK_KERNEL();
+  if (n_splits > 1) {
+    const long sh = (long)num_seqs * num_heads;
+    int wpb = 4;
+    dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+    hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                       dim3(wpb * 64), 0, stream, (short*)out,
+                       (const float*)part_out, (const float*)part_ml, sh,
+                       n_splits);
+    HIP_CHECK_KERNEL();
+  }
+  return hipSuccess;
+}
 
 extern "C" hipError_t ks_paged_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,

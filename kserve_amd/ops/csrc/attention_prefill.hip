@@ -419,12 +419,14 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
 // prefill path of the vLLM backend the reference delegates to.
 // Numerics oracle: ops/torch_ref.context_attention_varlen.
 // ---------------------------------------------------------------------------
-template <int D>  // D == 128
+// CT: cache element type — short (bf16) or unsigned char (fp8 E4M3,
+// converted to bf16 during LDS staging; MFMA math stays bf16)
+template <int D, typename CT = short>  // D == 128
 __global__ __launch_bounds__(256) void context_prefill_kernel(
     short* __restrict__ out,        // [Tq, Hq, D] bf16
     const short* __restrict__ q,    // [Tq, Hq, D] (row stride sq)
-    const short* __restrict__ k_cache,  // [NB, Hkv, 16, D]
-    const short* __restrict__ v_cache,
+    const CT* __restrict__ k_cache,  // [NB, Hkv, 16, D]
+    const CT* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [S, max_nb]
     const int* __restrict__ ctx_lens,      // [S] total context incl. chunk
     const int* __restrict__ cu_seqlens_q,  // [S+1]
@@ -495,8 +497,37 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
       if (tok < ctx) {
         const long base =
             (((long)bt[tok >> 4] * Hkv + kv_head) * 16 + (tok & 15)) * D + c8;
-        val = *reinterpret_cast<const short8_t*>(k_cache + base);
-        vv = *reinterpret_cast<const short8_t*>(v_cache + base);
+        if constexpr (sizeof(CT) == 2) {
+          val = *reinterpret_cast<const short8_t*>(k_cache + base);
+          vv = *reinterpret_cast<const short8_t*>(v_cache + base);
+        } else {
+          // fp8: 8 bytes -> 8 bf16 via two packed converts per dword
+          const unsigned int k2[2] = {
+              reinterpret_cast<const unsigned int*>(k_cache + base)[0],
+              reinterpret_cast<const unsigned int*>(k_cache + base)[1]};
+          const unsigned int v2[2] = {
+              reinterpret_cast<const unsigned int*>(v_cache + base)[0],
+              reinterpret_cast<const unsigned int*>(v_cache + base)[1]};
+#pragma unroll
+          for (int d = 0; d < 2; ++d) {
+            const float2_t klo =
+                __builtin_amdgcn_cvt_pk_f32_fp8((int)k2[d], false);
+            const float2_t khi =
+                __builtin_amdgcn_cvt_pk_f32_fp8((int)k2[d], true);
+            const float2_t vlo =
+                __builtin_amdgcn_cvt_pk_f32_fp8((int)v2[d], false);
+            const float2_t vhi =
+                __builtin_amdgcn_cvt_pk_f32_fp8((int)v2[d], true);
+            val[d * 4 + 0] = float_to_bf16_bits(klo[0]);
+            val[d * 4 + 1] = float_to_bf16_bits(klo[1]);
+            val[d * 4 + 2] = float_to_bf16_bits(khi[0]);
+            val[d * 4 + 3] = float_to_bf16_bits(khi[1]);
+            vv[d * 4 + 0] = float_to_bf16_bits(vlo[0]);
+            vv[d * 4 + 1] = float_to_bf16_bits(vlo[1]);
+            vv[d * 4 + 2] = float_to_bf16_bits(vhi[0]);
+            vv[d * 4 + 3] = float_to_bf16_bits(vhi[1]);
+          }
+        }
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -627,17 +658,26 @@ extern "C" hipError_t ks_context_prefill_varlen(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const void* block_tables, const void* ctx_lens, const void* cu_seqlens_q,
     int num_seqs, int max_q_len, int Hq, int Hkv, int head_dim, int max_nb,
-    float scale, long sq, hipStream_t stream) {
+    float scale, long sq, int fp8_cache, hipStream_t stream) {
   if (head_dim != 128) return hipErrorInvalidValue;
   if (Hq % Hkv != 0) return hipErrorInvalidValue;
   const int max_tiles = (max_q_len + NWAVES * QW2 - 1) / (NWAVES * QW2);
   if (max_tiles == 0 || num_seqs == 0) return hipSuccess;
   dim3 grid(Hq, max_tiles, num_seqs);
-  hipLaunchKernelGGL((context_prefill_kernel<128>), grid, dim3(256), 0,
-                     stream, (short*)out, (const short*)q,
-                     (const short*)k_cache, (const short*)v_cache,
-                     (const int*)block_tables, (const int*)ctx_lens,
-                     (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq);
+  if (fp8_cache) {
+    hipLaunchKernelGGL((context_prefill_kernel<128, unsigned char>), grid,
+                       dim3(256), 0, stream, (short*)out, (const short*)q,
+                       (const unsigned char*)k_cache,
+                       (const unsigned char*)v_cache,
+                       (const int*)block_tables, (const int*)ctx_lens,
+                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq);
+  } else {
+    hipLaunchKernelGGL((context_prefill_kernel<128, short>), grid, dim3(256),
+                       0, stream, (short*)out, (const short*)q,
+                       (const short*)k_cache, (const short*)v_cache,
+                       (const int*)block_tables, (const int*)ctx_lens,
+                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq);
+  }
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }

@@ -29,7 +29,15 @@ hipError_t ks_flash_prefill_varlen(void*, const void*, const void*,
 hipError_t ks_context_prefill_varlen(void*, const void*, const void*,
                                      const void*, const void*, const void*,
                                      const void*, int, int, int, int, int,
-                                     int, float, long, hipStream_t);
+                                     int, float, long, int, hipStream_t);
+hipError_t ks_paged_attention_decode_fp8(void*, const void*, const void*,
+                                         const void*, const void*,
+                                         const void*, float, int, int, int,
+                                         int, int, int, long, int, void*,
+                                         void*, hipStream_t);
+hipError_t ks_reshape_and_cache_fp8(const void*, const void*, void*, void*,
+                                    const void*, int, int, int, int, long,
+                                    long, hipStream_t);
 hipError_t ks_layer_norm(void*, const void*, const void*, const void*, float,
                          int, int, hipStream_t);
 hipError_t ks_fused_add_layer_norm(void*, const void*, const void*,
@@ -64,6 +72,13 @@ void check_hip(hipError_t err, const char* op) {
 
 // [T, heads, D] where rows may be strided (qkv split views) but the
 // (head, dim) block of each row is dense
+#define CHECK_KV_CACHE(t)                                               \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16 ||                       \
+                  (t).scalar_type() == at::kFloat8_e4m3fn,                \
+              #t " must be bf16 or fp8_e4m3");                            \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");             \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
 #define CHECK_BF16_ROWS(t)                                              \
   TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16");  \
   TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                     \
@@ -129,13 +144,22 @@ void reshape_and_cache(at::Tensor& k, at::Tensor& v, at::Tensor& k_cache,
                        at::Tensor& v_cache, at::Tensor& slot_mapping) {
   CHECK_BF16_ROWS(k);
   CHECK_BF16_ROWS(v);
-  CHECK_BF16_CONTIG(k_cache);
-  CHECK_BF16_CONTIG(v_cache);
+  CHECK_KV_CACHE(k_cache);
+  CHECK_KV_CACHE(v_cache);
   TORCH_CHECK(slot_mapping.scalar_type() == at::kInt, "slot_mapping int32");
   int T = k.size(0);
   int Hkv = k.size(1);
   int D = k.size(2);
   int block_size = k_cache.size(2);
+  if (k_cache.scalar_type() == at::kFloat8_e4m3fn) {
+    check_hip(ks_reshape_and_cache_fp8(k.data_ptr(), v.data_ptr(),
+                                       k_cache.data_ptr(), v_cache.data_ptr(),
+                                       slot_mapping.data_ptr(), T, Hkv, D,
+                                       block_size, (long)k.stride(0),
+                                       (long)v.stride(0), current_stream()),
+              "reshape_and_cache_fp8");
+    return;
+  }
   check_hip(ks_reshape_and_cache(k.data_ptr(), v.data_ptr(),
                                  k_cache.data_ptr(), v_cache.data_ptr(),
                                  slot_mapping.data_ptr(), T, Hkv, D,
@@ -150,10 +174,11 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
                             double scale) {
   CHECK_BF16_CONTIG(out);
   CHECK_BF16_ROWS(q);
-  CHECK_BF16_CONTIG(k_cache);
-  CHECK_BF16_CONTIG(v_cache);
+  CHECK_KV_CACHE(k_cache);
+  CHECK_KV_CACHE(v_cache);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt, "block_tables int32");
   TORCH_CHECK(context_lens.scalar_type() == at::kInt, "context_lens int32");
+  const bool fp8 = k_cache.scalar_type() == at::kFloat8_e4m3fn;
   int S = q.size(0);
   int H = q.size(1);
   int D = q.size(2);
@@ -177,6 +202,18 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
     part_ml = at::empty({S, H, n_splits, 2}, opts);
     po = part_out.data_ptr();
     pml = part_ml.data_ptr();
+  }
+  if (fp8) {
+    TORCH_CHECK(H / Hkv <= 4 && D == 128,
+                "fp8 KV decode supports D=128 and GQA group <= 4");
+    check_hip(ks_paged_attention_decode_fp8(
+                  out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                  v_cache.data_ptr(), block_tables.data_ptr(),
+                  context_lens.data_ptr(), (float)scale, S, H, Hkv, D,
+                  max_blocks, block_size, (long)q.stride(0), n_splits, po,
+                  pml, current_stream()),
+              "paged_attention_decode_fp8");
+    return;
   }
   check_hip(ks_paged_attention_decode(
                 out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
@@ -216,8 +253,8 @@ void context_prefill_varlen(at::Tensor& out, at::Tensor& q,
                             double scale) {
   CHECK_BF16_CONTIG(out);
   CHECK_BF16_ROWS(q);
-  CHECK_BF16_CONTIG(k_cache);
-  CHECK_BF16_CONTIG(v_cache);
+  CHECK_KV_CACHE(k_cache);
+  CHECK_KV_CACHE(v_cache);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt, "block_tables int32");
   TORCH_CHECK(block_tables.is_contiguous(), "block_tables contiguous");
   TORCH_CHECK(ctx_lens.scalar_type() == at::kInt, "ctx_lens int32");
@@ -227,12 +264,13 @@ void context_prefill_varlen(at::Tensor& out, at::Tensor& q,
   int Hq = q.size(1);
   int Hkv = k_cache.size(1);
   int D = q.size(2);
+  int fp8 = k_cache.scalar_type() == at::kFloat8_e4m3fn ? 1 : 0;
   check_hip(ks_context_prefill_varlen(
                 out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
                 v_cache.data_ptr(), block_tables.data_ptr(),
                 ctx_lens.data_ptr(), cu_seqlens_q.data_ptr(), num_seqs,
                 (int)max_q_len, Hq, Hkv, D, (int)block_tables.size(1),
-                (float)scale, (long)q.stride(0), current_stream()),
+                (float)scale, (long)q.stride(0), fp8, current_stream()),
             "context_prefill_varlen");
 }
 

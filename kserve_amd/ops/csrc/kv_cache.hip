@@ -35,7 +35,69 @@ __global__ void reshape_and_cache_kernel(
   }
 }
 
+// fp8 (OCP E4M3) cache variant: bf16 rows are converted pairwise with
+// v_cvt_pk_fp8_f32 at store time (scale 1.0; E4M3 range +-448 covers
+// RMSNorm'd projections). Halves KV bytes -> ~2x decode-attention bandwidth
+// for engines opting into kv_cache_dtype="fp8".
+__global__ void reshape_and_cache_fp8_kernel(
+    const short* __restrict__ k, const short* __restrict__ v,
+    unsigned char* __restrict__ k_cache, unsigned char* __restrict__ v_cache,
+    const int* __restrict__ slot_mapping, const int T, const int Hkv,
+    const int D, const int block_size, const long sk, const long sv) {
+  const int t = blockIdx.x;
+  if (t >= T) return;
+  const int slot = slot_mapping[t];
+  if (slot < 0) return;
+  const int blk = slot / block_size;
+  const int off = slot % block_size;
+  const int nvec = (Hkv * D) >> 3;  // 8 values -> 8 bytes out
+  const short8_t* k_src = reinterpret_cast<const short8_t*>(k + (long)t * sk);
+  const short8_t* v_src = reinterpret_cast<const short8_t*>(v + (long)t * sv);
+  const int dvec = D >> 3;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    const int h = i / dvec;
+    const int dv = i % dvec;
+    const long dst = (((long)blk * Hkv + h) * block_size + off) * dvec + dv;
+    short8_t kv = k_src[i];
+    short8_t vv = v_src[i];
+    // the builtin's word-select must be a literal; unroll by hand
+#define CVT2(dst_, a, b, w) \
+    dst_ = __builtin_amdgcn_cvt_pk_fp8_f32( \
+        bf16_bits_to_float(a), bf16_bits_to_float(b), dst_, w)
+    unsigned int kpk0 = 0, kpk1 = 0, vpk0 = 0, vpk1 = 0;
+    CVT2(kpk0, kv[0], kv[1], false);
+    CVT2(kpk0, kv[2], kv[3], true);
+    CVT2(kpk1, kv[4], kv[5], false);
+    CVT2(kpk1, kv[6], kv[7], true);
+    CVT2(vpk0, vv[0], vv[1], false);
+    CVT2(vpk0, vv[2], vv[3], true);
+    CVT2(vpk1, vv[4], vv[5], false);
+    CVT2(vpk1, vv[6], vv[7], true);
+#undef CVT2
+    reinterpret_cast<unsigned int*>(k_cache)[dst * 2] = kpk0;
+    reinterpret_cast<unsigned int*>(k_cache)[dst * 2 + 1] = kpk1;
+    reinterpret_cast<unsigned int*>(v_cache)[dst * 2] = vpk0;
+    reinterpret_cast<unsigned int*>(v_cache)[dst * 2 + 1] = vpk1;
+  }
+}
+
 }  // namespace
+
+extern "C" hipError_t ks_reshape_and_cache_fp8(
+    const void* k, const void* v, void* k_cache, void* v_cache,
+    const void* slot_mapping, int T, int Hkv, int D, int block_size, long sk,
+    long sv, hipStream_t stream) {
+  if (D % 8 != 0 || T == 0) return T == 0 ? hipSuccess : hipErrorInvalidValue;
+  int threads = (Hkv * D) >> 3;
+  if (threads > 256) threads = 256;
+  if (threads < 64) threads = 64;
+  hipLaunchKernelGGL(reshape_and_cache_fp8_kernel, dim3(T), dim3(threads), 0,
+                     stream, (const short*)k, (const short*)v,
+                     (unsigned char*)k_cache, (unsigned char*)v_cache,
+                     (const int*)slot_mapping, T, Hkv, D, block_size, sk, sv);
+  HIP_CHECK_KERNEL();
+  return hipSuccess;
+}
 
 extern "C" hipError_t ks_reshape_and_cache(const void* k, const void* v,
                                            void* k_cache, void* v_cache,

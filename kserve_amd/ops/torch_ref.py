@@ -88,8 +88,9 @@ def reshape_and_cache(
     blk = slots // block_size
     off = slots % block_size
     # advanced indexing: k_cache[blk[i], h, off[i], :] = k[i, h, :]
-    k_cache[blk, :, off] = k
-    v_cache[blk, :, off] = v
+    # (fp8 caches need an explicit cast: index_put refuses mixed dtypes)
+    k_cache[blk, :, off] = k.to(k_cache.dtype)
+    v_cache[blk, :, off] = v.to(v_cache.dtype)
 
 
 def paged_attention_decode(

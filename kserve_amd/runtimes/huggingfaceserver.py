@@ -70,7 +70,8 @@ def build_model(args):
                 os.path.join(args.model_dir, "config.json")
             ),
             cache=CacheConfig(
-                cpu_offload_bytes=getattr(args, "kv_offload_bytes", 0)
+                cpu_offload_bytes=getattr(args, "kv_offload_bytes", 0),
+                kv_cache_dtype=getattr(args, "kv_cache_dtype", "auto"),
             ),
             scheduler=SchedulerConfig(
                 max_num_seqs=getattr(args, "max_num_seqs", 256),
@@ -120,6 +121,8 @@ def main(argv=None):
     parser.add_argument("--enable-expert-parallel", dest="expert_parallel", action="store_true")
     # host-DRAM KV offload tier (LLMInferenceService KVCacheOffloadingSpec)
     parser.add_argument("--kv-offload-bytes", dest="kv_offload_bytes", type=int, default=0)
+    parser.add_argument("--kv-cache-dtype", dest="kv_cache_dtype", default="auto",
+                        choices=["auto", "fp8"])
     # LoRA adapter serving (reference: --enable-lora --lora-modules name=path)
     parser.add_argument("--enable-lora", dest="enable_lora", action="store_true")
     parser.add_argument(

@@ -172,3 +172,28 @@ def test_chunked_prefill_matches_full():
     assert [o.output_token_ids for o in a.values()] == [
         o.output_token_ids for o in b.values()
     ]
+
+
+def test_fp8_kv_cache_cpu():
+    """fp8 E4M3 KV cache (opt-in): engine runs and its greedy logits stay
+    close to the bf16-KV engine's (quantization noise only)."""
+    torch.manual_seed(0)
+    base = make_engine()
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=128),
+        cache=CacheConfig(block_size=4, num_gpu_blocks=128, kv_cache_dtype="fp8"),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=256, max_model_len=128
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    fp8 = LLMEngine(cfg)
+    assert fp8.runner.kv_caches[0][0].dtype == torch.float8_e4m3fn
+    sp = SamplingParams(temperature=0.0, max_tokens=8)
+    a = list(base.generate([[1, 2, 3, 4, 5]], sp).values())[0]
+    b = list(fp8.generate([[1, 2, 3, 4, 5]], sp).values())[0]
+    assert len(b.output_token_ids) == 8
+    # prefill (first token) ignores the cache entirely -> identical
+    assert a.output_token_ids[0] == b.output_token_ids[0]

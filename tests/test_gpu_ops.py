@@ -223,6 +223,89 @@ class TestPrefillAttention:
         bf16_close(got, ref, atol=3e-2, rtol=3e-2)
 
 
+    def test_paged_decode_fp8(self, dev):
+        """fp8 E4M3 KV cache decode vs fp32 ref reading the SAME quantized
+        cache (so only the kernel's conversion/accumulation is under test)."""
+        from kserve_amd import ops
+
+        S, H, Hkv, ctx_max, D, bs = 32, 32, 8, 300, 128, 16
+        torch.manual_seed(99)
+        ctx = torch.randint(1, ctx_max + 1, (S,), dtype=torch.int32)
+        max_blocks = int((int(ctx.max()) + bs - 1) // bs)
+        B = S * max_blocks + 1
+        kc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev).to(
+            torch.float8_e4m3fn
+        )
+        vc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev).to(
+            torch.float8_e4m3fn
+        )
+        bt = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+            S, max_blocks
+        )
+        q = torch.randn(S, H, D, dtype=torch.bfloat16, device=dev)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.paged_attention_decode(q, kc, vc, bt.to(dev), ctx.to(dev), scale)
+        ref = torch_ref.paged_attention_decode(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt, ctx, scale
+        )
+        bf16_close(got, ref)
+
+    def test_reshape_and_cache_fp8(self, dev):
+        from kserve_amd import ops
+
+        T, Hkv, D, bs, B = 50, 8, 128, 16, 32
+        torch.manual_seed(1)
+        k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev)
+        v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev)
+        kc = torch.zeros(B, Hkv, bs, D, dtype=torch.float8_e4m3fn, device=dev)
+        vc = torch.zeros_like(kc)
+        slots = torch.randperm(B * bs, device=dev)[:T].to(torch.int32)
+        ops.reshape_and_cache(k, v, kc, vc, slots)
+        torch.cuda.synchronize()
+        kc_ref = torch.zeros(B, Hkv, bs, D, dtype=torch.float8_e4m3fn)
+        vc_ref = torch.zeros_like(kc_ref)
+        torch_ref.reshape_and_cache(k.cpu(), v.cpu(), kc_ref, vc_ref, slots.cpu())
+        # both sides quantize bf16 -> E4M3 round-to-nearest: bit comparable
+        torch.testing.assert_close(
+            kc.float().cpu(), kc_ref.float(), atol=0.08, rtol=0.08
+        )
+        torch.testing.assert_close(
+            vc.float().cpu(), vc_ref.float(), atol=0.08, rtol=0.08
+        )
+
+    def test_context_prefill_fp8(self, dev):
+        from kserve_amd import ops
+
+        D, bs = 128, 16
+        torch.manual_seed(5)
+        qlens, ctx_extra, H, Hkv = [48, 7], [100, 30], 32, 8
+        S = len(qlens)
+        ctx = [a + b for a, b in zip(qlens, ctx_extra)]
+        max_blocks = max((c + bs - 1) // bs for c in ctx)
+        B = S * max_blocks + 1
+        kc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev).to(
+            torch.float8_e4m3fn
+        )
+        vc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev).to(
+            torch.float8_e4m3fn
+        )
+        bt = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+            S, max_blocks
+        )
+        cu = torch.tensor([0] + list(torch.tensor(qlens).cumsum(0)), dtype=torch.int32)
+        ctx_t = torch.tensor(ctx, dtype=torch.int32)
+        q = torch.randn(sum(qlens), H, D, dtype=torch.bfloat16, device=dev)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.context_attention_varlen(
+            q, kc, vc, bt.to(dev), cu.to(dev), ctx_t.to(dev), max(qlens), scale
+        )
+        ref = torch_ref.context_attention_varlen(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt, cu,
+            ctx_t, scale
+        )
+        bf16_close(got, ref, atol=3e-2, rtol=3e-2)
+
+
 class TestSampling:
     def test_greedy(self, dev):
         from kserve_amd import ops
