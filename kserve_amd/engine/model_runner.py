@@ -21,7 +21,7 @@ from kserve_amd.engine.scheduler import ScheduledBatch
 from kserve_amd.logging import logger
 from kserve_amd.models.llama import AttentionMetadata, LlamaForCausalLM
 
-_DEFAULT_GRAPH_BATCH_SIZES = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 384, 512, 768, 1024]
+_DEFAULT_GRAPH_BATCH_SIZES = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 384, 512, 768, 1024, 1280, 1536, 2048]
 
 
 class ModelRunner:
