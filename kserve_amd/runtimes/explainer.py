@@ -85,3 +85,83 @@ class ExplainerModel(Model):
             return result["predictions"]
 
         return await self.explainer.explain(np.asarray(instances, dtype=float), predict_fn)
+
+
+class SquareAttackExplainer(Explainer):
+    """Native black-box square attack (the algorithm behind the reference's
+    artexplainer: random square perturbations, accepted when they push the
+    prediction away from the true label — artserver/model.py:74-96). No ART
+    dependency; works against any predictor returning class labels or
+    per-class scores."""
+
+    def __init__(
+        self,
+        nb_classes: int,
+        max_iter: int = 100,
+        eps: float = 0.3,
+        seed: int = 0,
+    ):
+        self.nb_classes = int(nb_classes)
+        self.max_iter = int(max_iter)
+        self.eps = float(eps)
+        self.seed = seed
+
+    @staticmethod
+    def _to_scores(pred, nb_classes) -> np.ndarray:
+        arr = np.asarray(pred, dtype=float)
+        if arr.ndim == 0 or (arr.ndim == 1 and arr.size == 1):
+            one_hot = np.zeros(nb_classes)
+            one_hot[int(arr)] = 1.0
+            return one_hot
+        return arr.reshape(-1)
+
+    async def explain(self, instances: np.ndarray, predict_fn) -> Dict:
+        # payload convention (reference artserver): [image, label]
+        x = np.asarray(instances[0], dtype=float)
+        label = int(np.asarray(instances[1]).reshape(-1)[0])
+        rng = np.random.default_rng(self.seed)
+        flat = x.reshape(-1)
+        n = flat.size
+        side = int(np.sqrt(n)) if int(np.sqrt(n)) ** 2 == n else None
+
+        async def score(v: np.ndarray) -> np.ndarray:
+            p = await predict_fn([v.reshape(x.shape).tolist()])
+            return self._to_scores(p[0], self.nb_classes)
+
+        orig_scores = await score(flat)
+        orig_pred = int(np.argmax(orig_scores))
+        best = flat.copy()
+        best_margin = orig_scores[label] - np.max(
+            np.delete(orig_scores, label)
+        )
+        # square-attack loop: progressively smaller random squares of +-eps
+        for it in range(self.max_iter):
+            frac = max(0.05, 0.5 * (1 - it / self.max_iter))
+            cand = best.copy()
+            if side is not None:
+                h = max(1, int(side * frac))
+                r0 = rng.integers(0, side - h + 1)
+                c0 = rng.integers(0, side - h + 1)
+                sq = (slice(r0, r0 + h), slice(c0, c0 + h))
+                img = cand.reshape(side, side)
+                img[sq] = img[sq] + rng.choice([-self.eps, self.eps])
+            else:
+                w = max(1, int(n * frac))
+                i0 = rng.integers(0, n - w + 1)
+                cand[i0 : i0 + w] += rng.choice([-self.eps, self.eps])
+            s = await score(cand)
+            margin = s[label] - np.max(np.delete(s, label))
+            if margin < best_margin:
+                best, best_margin = cand, margin
+                if margin < 0:
+                    break
+        adv_scores = await score(best)
+        adv = best.reshape(x.shape)
+        return {
+            "explanations": {
+                "adversarial_example": [adv.tolist()],
+                "L2 error": float(np.linalg.norm((adv - x).reshape(-1))),
+                "adversarial_prediction": int(np.argmax(adv_scores)),
+                "prediction": orig_pred,
+            }
+        }
