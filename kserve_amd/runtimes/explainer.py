@@ -84,7 +84,12 @@ class ExplainerModel(Model):
             result = await self._forward_predict({"instances": batch}, headers)
             return result["predictions"]
 
-        return await self.explainer.explain(np.asarray(instances, dtype=float), predict_fn)
+        try:
+            arr = np.asarray(instances, dtype=float)
+        except (TypeError, ValueError):
+            # ragged payloads (e.g. the ART convention [image, label])
+            arr = instances
+        return await self.explainer.explain(arr, predict_fn)
 
 
 class SquareAttackExplainer(Explainer):

@@ -197,6 +197,76 @@ class TestExplainer:
         assert imps[1] == pytest.approx(5.0)
 
 
+    def test_square_attack_explainer_flips_label(self):
+        import httpx
+
+        from kserve_amd.model import PredictorConfig
+        from kserve_amd.runtimes.explainer import (
+            ExplainerModel,
+            SquareAttackExplainer,
+        )
+
+        # black-box scorer: class-1 score rises with sum(pixels); decision
+        # boundary at 8 (soft scores give the square search its signal)
+        def predictor(request: httpx.Request) -> httpx.Response:
+            body = json.loads(request.content)
+            preds = []
+            for inst in body["instances"]:
+                s = sum(sum(row) for row in inst)
+                preds.append([8.0 - s, s - 8.0])
+            return httpx.Response(200, json={"predictions": preds})
+
+        model = ExplainerModel(
+            "art",
+            PredictorConfig(predictor_host="pred:80"),
+            explainer=SquareAttackExplainer(nb_classes=2, max_iter=60, eps=1.0),
+        )
+        model._http_client = httpx.AsyncClient(
+            transport=httpx.MockTransport(predictor)
+        )
+        image = [[1.2, 1.2, 1.2], [1.2, 1.2, 1.2], [1.2, 1.2, 1.2]]  # sum 10.8
+        out = run(model.explain({"instances": [image, 1]}))
+        exp = out["explanations"]
+        assert exp["prediction"] == 1
+        assert exp["adversarial_prediction"] == 0  # attack flipped the label
+        assert exp["L2 error"] > 0
+
+    def test_aif_fairness_metrics(self):
+        from kserve_amd.runtimes.aifserver import AIFFairnessModel
+
+        model = AIFFairnessModel(
+            "aif",
+            feature_names=["age", "income"],
+            label_names=["approved"],
+            favorable_label=1.0,
+            unfavorable_label=0.0,
+            privileged_groups=[{"age": 1.0}],
+            unprivileged_groups=[{"age": 0.0}],
+        )
+        # privileged (age=1): 3/4 favorable; unprivileged (age=0): 1/4
+        instances = [
+            [1, 10], [1, 20], [1, 30], [1, 40],
+            [0, 10], [0, 20], [0, 30], [0, 40],
+        ]
+        outputs = [1, 1, 1, 0, 1, 0, 0, 0]
+        out = run(model.explain({"instances": instances, "outputs": outputs}))
+        m = out["metrics"]
+        assert m["num_instances"] == 8
+        assert m["num_positives"] == 4
+        assert m["num_negatives"] == 4
+        assert m["base_rate"] == pytest.approx(0.5)
+        assert m["statistical_parity_difference"] == pytest.approx(0.25 - 0.75)
+        assert m["disparate_impact"] == pytest.approx(0.25 / 0.75)
+        assert 0.0 <= m["consistency"][0] <= 1.0
+
+    def test_autogluon_gated(self):
+        from kserve_amd.runtimes.autogluonserver import AutoGluonModel
+
+        model = AutoGluonModel("ag", "/nonexistent")
+        with pytest.raises(RuntimeError, match="autogluon"):
+            model.load()
+
+
 class TestRerankAdapter:
     def test_cosine_rerank(self):
         from kserve_amd.protocol.rest.openai.types import RerankRequest
