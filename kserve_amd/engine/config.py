@@ -128,6 +128,10 @@ class CacheConfig:
     # host-DRAM offload tier (pinned memory, hipMemcpyAsync side stream)
     num_cpu_blocks: int = 0
     cpu_offload_bytes: int = 0
+    # automatic prefix caching: content-addressed full prompt blocks are
+    # shared across requests; cache-hit prompts only compute their suffix
+    # through the paged-context prefill kernel
+    enable_prefix_caching: bool = False
     # "auto" stores KV in the model dtype (bf16). "fp8" stores OCP E4M3
     # bytes (scale 1.0): half the KV bandwidth/capacity; attention math
     # stays fp32/bf16 after in-register conversion. GQA group <= 4, D=128.

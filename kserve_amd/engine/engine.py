@@ -291,6 +291,9 @@ class LLMEngine:
         # advance computed-token counters for executed tokens
         for req, n in zip(batch.requests, batch.num_scheduled_tokens):
             req.num_computed_tokens += n
+        if batch.is_prefill and self.scheduler.block_manager.enable_prefix_caching:
+            for req in batch.requests:
+                self.scheduler.block_manager.register_computed_blocks(req)
         # chunked prefill: requests whose prompt isn't fully computed don't
         # sample this step
         if batch.is_prefill:

@@ -56,6 +56,7 @@ class Scheduler:
         self.block_manager = BlockManager(
             num_gpu_blocks, cache_config.block_size,
             num_cpu_blocks=num_cpu_blocks or cache_config.num_cpu_blocks,
+            enable_prefix_caching=cache_config.enable_prefix_caching,
         )
         self.waiting: Deque[Request] = deque()
         self.running: List[Request] = []
@@ -124,14 +125,19 @@ class Scheduler:
                 ) - len(req.block_table)
                 if need > self.block_manager.num_free_blocks:
                     break
-                for _ in range(need):
-                    req.block_table.append(self.block_manager._free.pop())
+                req.block_table.extend(self.block_manager.take_blocks(need))
             else:
                 if not self.block_manager.can_allocate(
                     req, req.num_computed_tokens + n_new
                 ):
                     break
                 self.block_manager.allocate(req, req.num_computed_tokens + n_new)
+                cached = getattr(req, "num_cached_tokens", 0)
+                if cached > req.num_computed_tokens:
+                    # prefix-cache hit: KV for the prefix already resident;
+                    # only the suffix runs (paged-context prefill path)
+                    req.num_computed_tokens = cached
+                    n_new = req.num_tokens - cached
             self.waiting.popleft()
             req.status = RequestStatus.RUNNING
             batch.requests.append(req)
@@ -262,6 +268,6 @@ class Scheduler:
             want = r.num_tokens + k - 1
             table = bm._tables[r.request_id]
             while len(table) * bs < want:
-                table.append(bm._free.pop())
+                table.extend(bm.take_blocks(1))
             r.block_table = table
         return k
