@@ -319,3 +319,111 @@ def render_knative_service(isvc: InferenceService, runtimes: List[ServingRuntime
             "traffic": traffic,
         },
     }
+
+
+# ---- InferenceGraph controller (reference v1alpha1/inferencegraph) ---------
+
+GRAPH_ROUTER_IMAGE = "kserve-amd/graph-router:latest"
+
+
+def render_graph_deployment(
+    name: str,
+    namespace: str,
+    graph_spec: Dict,
+    router_image: str = GRAPH_ROUTER_IMAGE,
+    min_replicas: int = 1,
+    max_replicas: int = 1,
+) -> Dict:
+    """Deployment running the graph router with --graph-json <spec>
+    (reference raw_ig.go:50-104 createInferenceGraphPodSpec)."""
+    import json as _json
+
+    labels = {"serving.kserve.io/inferencegraph": name}
+    return {
+        "apiVersion": "apps/v1",
+        "kind": "Deployment",
+        "metadata": {"name": name, "namespace": namespace, "labels": labels},
+        "spec": {
+            "replicas": min_replicas,
+            "selector": {"matchLabels": labels},
+            "template": {
+                "metadata": {"labels": labels},
+                "spec": {
+                    "containers": [
+                        {
+                            "name": "kserve-router",
+                            "image": router_image,
+                            "args": [
+                                "--graph-json",
+                                _json.dumps(graph_spec, separators=(",", ":")),
+                            ],
+                            "ports": [{"containerPort": 8080}],
+                            "readinessProbe": {
+                                "httpGet": {"path": "/readyz", "port": 8080}
+                            },
+                        }
+                    ],
+                    "automountServiceAccountToken": False,
+                },
+            },
+        },
+    }
+
+
+def render_graph_service(name: str, namespace: str) -> Dict:
+    labels = {"serving.kserve.io/inferencegraph": name}
+    return {
+        "apiVersion": "v1",
+        "kind": "Service",
+        "metadata": {"name": name, "namespace": namespace, "labels": labels},
+        "spec": {
+            "selector": labels,
+            "ports": [{"name": "http", "port": 80, "targetPort": 8080}],
+        },
+    }
+
+
+def reconcile_graph(
+    name: str,
+    namespace: str,
+    graph_spec: Dict,
+    min_replicas: int = 1,
+    max_replicas: int = 1,
+) -> Dict[str, Dict]:
+    """Desired state for an InferenceGraph CR (raw-deployment mode):
+    Deployment + Service (+ HPA when max_replicas > min_replicas)."""
+    out = {
+        "deployment": render_graph_deployment(
+            name, namespace, graph_spec, min_replicas=min_replicas,
+            max_replicas=max_replicas,
+        ),
+        "service": render_graph_service(name, namespace),
+    }
+    if max_replicas > min_replicas:
+        out["hpa"] = {
+            "apiVersion": "autoscaling/v2",
+            "kind": "HorizontalPodAutoscaler",
+            "metadata": {"name": name, "namespace": namespace},
+            "spec": {
+                "scaleTargetRef": {
+                    "apiVersion": "apps/v1",
+                    "kind": "Deployment",
+                    "name": name,
+                },
+                "minReplicas": min_replicas,
+                "maxReplicas": max_replicas,
+                "metrics": [
+                    {
+                        "type": "Resource",
+                        "resource": {
+                            "name": "cpu",
+                            "target": {
+                                "type": "Utilization",
+                                "averageUtilization": 80,
+                            },
+                        },
+                    }
+                ],
+            },
+        }
+    return out

@@ -374,3 +374,39 @@ class TestKServeClient:
         assert client.get_llm("llm1") is not None
         client.delete_llm("llm1")
         assert client.get_llm("llm1") is None
+
+
+class TestInferenceGraphController:
+    def test_reconcile_graph_renders_router_contract(self):
+        """The rendered Deployment launches the router with --graph-json,
+        round-trippable into our InferenceGraphSpec (reference
+        raw_ig.go createInferenceGraphPodSpec)."""
+        import json
+
+        from kserve_amd.controlplane.reconciler import reconcile_graph
+        from kserve_amd.graph.types import InferenceGraphSpec
+
+        spec = {
+            "nodes": {
+                "root": {
+                    "routerType": "Sequence",
+                    "steps": [
+                        {"name": "a", "serviceUrl": "http://a/v1/models/a:predict"},
+                        {"name": "b", "serviceUrl": "http://b/v1/models/b:predict"},
+                    ],
+                }
+            }
+        }
+        out = reconcile_graph("my-graph", "prod", spec, min_replicas=1, max_replicas=3)
+        dep = out["deployment"]
+        args = dep["spec"]["template"]["spec"]["containers"][0]["args"]
+        assert args[0] == "--graph-json"
+        parsed = InferenceGraphSpec.from_dict(json.loads(args[1]))
+        assert [s.step_name for s in parsed.nodes["root"].steps] == ["a", "b"]
+        svc = out["service"]
+        assert svc["spec"]["selector"] == {
+            "serving.kserve.io/inferencegraph": "my-graph"
+        }
+        hpa = out["hpa"]
+        assert hpa["spec"]["maxReplicas"] == 3
+        assert hpa["spec"]["scaleTargetRef"]["name"] == "my-graph"
