@@ -299,3 +299,60 @@ class TestRerankAdapter:
         out = run(adapter.create_rerank(req))
         assert out.results[0].index == 1  # близко ranks first
         assert out.results[0].relevance_score > out.results[1].relevance_score
+
+
+class TestExampleSamples:
+    """The examples/ directory mirrors the reference's custom_model /
+    custom_transformer samples; keep them importable and serving-correct."""
+
+    def test_custom_model_sample_end_to_end(self):
+        import sys
+
+        sys.path.insert(0, "examples")
+        try:
+            from custom_model import CustomModel
+        finally:
+            sys.path.pop(0)
+        from fastapi.testclient import TestClient
+
+        from kserve_amd.model_repository import ModelRepository
+        from kserve_amd.protocol.dataplane import DataPlane
+        from kserve_amd.protocol.rest.server import create_app
+
+        model = CustomModel("custom-model")
+        repo = ModelRepository()
+        repo.update(model)
+        app = create_app(DataPlane(repo))
+        with TestClient(app) as c:
+            r = c.post(
+                "/v1/models/custom-model:predict",
+                json={"instances": [[0.5] * 784, [0.1] * 784]},
+            )
+            assert r.status_code == 200
+            assert len(r.json()["predictions"]) == 2
+
+    def test_custom_transformer_sample(self):
+        import sys
+
+        import httpx
+
+        sys.path.insert(0, "examples")
+        try:
+            from custom_transformer import ImageTransformer
+        finally:
+            sys.path.pop(0)
+
+        def predictor(request: httpx.Request) -> httpx.Response:
+            body = json.loads(request.content)
+            # echo back the max of each (normalized) row
+            return httpx.Response(
+                200,
+                json={"predictions": [max(r) for r in body["instances"]]},
+            )
+
+        t = ImageTransformer("t", predictor_host="pred:80")
+        t._http_client = httpx.AsyncClient(transport=httpx.MockTransport(predictor))
+        pre = t.preprocess({"instances": [[0, 128, 255]]})
+        assert pre["instances"][0][2] == 1.0
+        out = run(t.predict(pre))
+        assert out["predictions"] == [1.0]
