@@ -209,3 +209,101 @@ def test_stream_disconnect_aborts(client):
     while time.time() < deadline and engine.has_unfinished():
         time.sleep(0.1)
     assert not engine.has_unfinished(), "aborted request still scheduled"
+
+
+class TestOpenAIProxyModel:
+    """Proxy model forwards completions to an upstream OpenAI server
+    (reference openai_proxy_model.py), with hooks and SSE passthrough."""
+
+    def _upstream_app(self):
+        """In-process upstream: the tiny LLM engine behind OpenAI routes."""
+        from kserve_amd.model_repository import ModelRepository
+        from kserve_amd.protocol.dataplane import DataPlane
+        from kserve_amd.protocol.rest.server import create_app
+        from kserve_amd.runtimes.llm_model import LLMModel
+
+        torch.manual_seed(0)
+        cfg = EngineConfig(
+            model=ModelConfig.tiny(vocab_size=128),
+            cache=CacheConfig(block_size=4, num_gpu_blocks=128),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+            ),
+            device="cpu",
+            eos_token_id=-1,
+        )
+        model = LLMModel("up", cfg)
+        repo = ModelRepository()
+        repo.update(model)
+        dp = DataPlane(repo)
+        app = create_app(dp)
+        register_openai_endpoints(app, dp, [model])
+        return app, model
+
+    def test_proxy_completion_and_stream(self):
+        import asyncio
+
+        import httpx
+
+        from kserve_amd.protocol.rest.openai.proxy_model import OpenAIProxyModel
+
+        up_app, up_model = self._upstream_app()
+
+        class CountingProxy(OpenAIProxyModel):
+            pre = post = 0
+
+            async def preprocess_completion_request(self, body):
+                CountingProxy.pre += 1
+                body["model"] = "up"  # route to the upstream model id
+                return body
+
+            async def postprocess_completion(self, body):
+                CountingProxy.post += 1
+                body["proxied"] = True
+                return body
+
+        client = httpx.AsyncClient(
+            transport=httpx.ASGITransport(app=up_app), base_url="http://up"
+        )
+        proxy = CountingProxy("front", "http://up", http_client=client)
+        from kserve_amd.model_repository import ModelRepository
+        from kserve_amd.protocol.dataplane import DataPlane
+        from kserve_amd.protocol.rest.server import create_app
+
+        repo = ModelRepository()
+        repo.update(proxy)
+        dp = DataPlane(repo)
+        front = create_app(dp)
+        register_openai_endpoints(front, dp, [proxy])
+
+        with TestClient(front) as c:
+            asyncio.new_event_loop().run_until_complete(up_model.start_engine())
+            r = c.post(
+                "/openai/v1/completions",
+                json={"model": "front", "prompt": [1, 2, 3], "max_tokens": 4,
+                      "temperature": 0.0},
+            )
+            assert r.status_code == 200, r.text
+            out = r.json()
+            assert out["proxied"] is True
+            assert out["usage"]["completion_tokens"] == 4
+            assert CountingProxy.pre == 1 and CountingProxy.post == 1
+
+            # streaming passthrough
+            r = c.post(
+                "/openai/v1/completions",
+                json={"model": "front", "prompt": [1, 2, 3], "max_tokens": 4,
+                      "temperature": 0.0, "stream": True},
+            )
+            assert r.status_code == 200
+            assert r.headers["content-type"].startswith("text/event-stream")
+            chunks = [
+                json.loads(line[len("data: "):])
+                for line in r.text.splitlines()
+                if line.startswith("data: ") and "[DONE]" not in line
+            ]
+            total = sum(
+                len(ch["choices"][0]["text"] or "") for ch in chunks if ch.get("choices")
+            )
+            assert total > 0
+            up_model.stop()
