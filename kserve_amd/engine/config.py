@@ -111,7 +111,10 @@ class ModelConfig:
             rope_theta=cfg.get("rope_theta", 10000.0),
             max_position_embeddings=cfg.get("max_position_embeddings", 8192),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
-            attention_bias=cfg.get("attention_bias", False),
+            # Qwen2 has qkv bias implicitly (no attention_bias key in HF)
+            attention_bias=cfg.get(
+                "attention_bias", cfg.get("model_type") == "qwen2"
+            ),
             mlp_bias=cfg.get("mlp_bias", False),
             model_name=cfg.get("_name_or_path", os.path.dirname(config_path) or "model"),
         )

@@ -141,3 +141,93 @@ class TestPagedDecodeConsistency:
             seq.append(tok)
 
         assert engine_tokens == oracle_tokens
+
+
+class TestMistralQwen2VsTransformers:
+    """The huggingfaceserver advertises Mistral and Qwen2 (llama-family
+    variants: Mistral = llama weights/rope; Qwen2 adds qkv bias). Verify
+    logits against HF transformers fp32."""
+
+    def _compare(self, hf_model, cfg):
+        ours = LlamaForCausalLM(cfg, dtype=torch.float32, device="cpu")
+        ours.load_hf_state_dict(dict(hf_model.state_dict()))
+        token_ids = list(torch.randint(0, cfg.vocab_size, (19,)).tolist())
+        with torch.no_grad():
+            hf_logits = hf_model(
+                torch.tensor([token_ids]), use_cache=False
+            ).logits[0]
+        our_logits = full_forward_logits(ours, token_ids)
+        torch.testing.assert_close(our_logits, hf_logits, rtol=2e-4, atol=2e-4)
+
+    def test_mistral_logits(self):
+        transformers = pytest.importorskip("transformers")
+        from kserve_amd.engine.config import ModelConfig
+
+        torch.manual_seed(11)
+        hf_cfg = transformers.MistralConfig(
+            vocab_size=256,
+            hidden_size=256,
+            intermediate_size=512,
+            num_hidden_layers=2,
+            num_attention_heads=4,
+            num_key_value_heads=2,
+            head_dim=64,
+            rms_norm_eps=1e-5,
+            rope_theta=10000.0,
+            max_position_embeddings=512,
+            sliding_window=None,
+            tie_word_embeddings=False,
+        )
+        hf = transformers.MistralForCausalLM(hf_cfg).eval().float()
+        cfg = ModelConfig(
+            vocab_size=256, hidden_size=256, intermediate_size=512,
+            num_layers=2, num_heads=4, num_kv_heads=2, head_dim=64,
+            rms_norm_eps=1e-5, rope_theta=10000.0,
+            max_position_embeddings=512, model_name="mistral-tiny",
+        )
+        self._compare(hf, cfg)
+
+    def test_qwen2_logits(self):
+        transformers = pytest.importorskip("transformers")
+        from kserve_amd.engine.config import ModelConfig
+
+        torch.manual_seed(13)
+        hf_cfg = transformers.Qwen2Config(
+            vocab_size=256,
+            hidden_size=256,
+            intermediate_size=512,
+            num_hidden_layers=2,
+            num_attention_heads=4,
+            num_key_value_heads=2,
+            rms_norm_eps=1e-5,
+            rope_theta=10000.0,
+            max_position_embeddings=512,
+            tie_word_embeddings=False,
+        )
+        hf = transformers.Qwen2ForCausalLM(hf_cfg).eval().float()
+        cfg = ModelConfig(
+            vocab_size=256, hidden_size=256, intermediate_size=512,
+            num_layers=2, num_heads=4, num_kv_heads=2, head_dim=64,
+            rms_norm_eps=1e-5, rope_theta=10000.0,
+            max_position_embeddings=512, attention_bias=True,
+            model_name="qwen2-tiny",
+        )
+        self._compare(hf, cfg)
+
+    def test_qwen2_config_json_implies_bias(self):
+        import json as _json
+        import tempfile
+
+        from kserve_amd.engine.config import ModelConfig
+
+        with tempfile.NamedTemporaryFile("w", suffix=".json", delete=False) as f:
+            _json.dump(
+                {
+                    "model_type": "qwen2", "vocab_size": 8, "hidden_size": 8,
+                    "intermediate_size": 16, "num_hidden_layers": 1,
+                    "num_attention_heads": 2,
+                },
+                f,
+            )
+            path = f.name
+        assert ModelConfig.from_hf_config(path).attention_bias is True
