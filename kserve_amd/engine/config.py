@@ -41,6 +41,9 @@ class ModelConfig:
     # attention bias (Qwen2-style) / mlp bias
     attention_bias: bool = False
     mlp_bias: bool = False
+    # mixture-of-experts (Mixtral): 0 experts = dense MLP
+    num_local_experts: int = 0
+    num_experts_per_tok: int = 2
 
     def __post_init__(self):
         if self.head_dim is None:
@@ -115,6 +118,8 @@ class ModelConfig:
             attention_bias=cfg.get(
                 "attention_bias", cfg.get("model_type") == "qwen2"
             ),
+            num_local_experts=cfg.get("num_local_experts", 0),
+            num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             mlp_bias=cfg.get("mlp_bias", False),
             model_name=cfg.get("_name_or_path", os.path.dirname(config_path) or "model"),
         )

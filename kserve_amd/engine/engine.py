@@ -157,7 +157,13 @@ class LLMEngine:
         self.eos_token_id = config.eos_token_id
         if tokenizer is not None and getattr(tokenizer, "eos_token_id", None) is not None:
             self.eos_token_id = tokenizer.eos_token_id
-        if self.device.type == "cuda" and not config.enforce_eager:
+        if (
+            self.device.type == "cuda"
+            and not config.enforce_eager
+            and config.model.num_local_experts == 0
+        ):
+            # MoE routing (data-dependent expert buckets) is not yet
+            # hipGraph-capturable; MoE decodes run eager
             self.runner.capture_decode_graphs()
         logger.info(
             "Engine ready in %.1fs: %s (%.2fB params, tp=%d, device=%s)",

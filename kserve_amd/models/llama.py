@@ -149,11 +149,78 @@ class LlamaMLP(nn.Module):
         return self.down_proj(act, d_delta)
 
 
+class MixtralMoE(nn.Module):
+    """Sparse mixture-of-experts MLP block (Mixtral architecture).
+
+    Reference parity: the Mixtral family the reference's vLLM backend
+    serves. MI355X-native v1: token-grouped expert GEMMs via hipBLASLt —
+    tokens are bucketed per expert, each expert runs the dense SwiGLU path
+    on its bucket, outputs are weighted scatter-adds. Under TP each
+    expert's weights are column/row-sharded like the dense MLP and ONE
+    all-reduce covers the whole block.
+    """
+
+    def __init__(self, config: ModelConfig, dtype: torch.dtype, layer_idx: int = 0):
+        super().__init__()
+        self.layer_idx = layer_idx
+        self.num_experts = config.num_local_experts
+        self.top_k = config.num_experts_per_tok
+        self.gate = nn.Parameter(
+            torch.empty(self.num_experts, config.hidden_size, dtype=dtype),
+            requires_grad=False,
+        )
+        self.gate_up = nn.ModuleList(
+            [
+                MergedColumnParallelLinear(
+                    config.hidden_size, config.intermediate_size, dtype=dtype
+                )
+                for _ in range(self.num_experts)
+            ]
+        )
+        self.down = nn.ModuleList(
+            [
+                RowParallelLinear(
+                    config.intermediate_size,
+                    config.hidden_size,
+                    dtype=dtype,
+                    reduce_output=False,  # one all-reduce for the whole block
+                )
+                for _ in range(self.num_experts)
+            ]
+        )
+
+    def forward(self, x: torch.Tensor, meta=None) -> torch.Tensor:
+        T = x.shape[0]
+        router_logits = torch.nn.functional.linear(x.float(), self.gate.float())
+        probs = torch.softmax(router_logits, dim=-1)
+        topw, topi = probs.topk(self.top_k, dim=-1)
+        topw = topw / topw.sum(dim=-1, keepdim=True)  # mixtral renorm
+        out = torch.zeros_like(x)
+        flat_i = topi.flatten()
+        flat_w = topw.flatten()
+        token_idx = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
+        for e in range(self.num_experts):
+            sel = (flat_i == e).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            rows = token_idx[sel]
+            xe = x.index_select(0, rows)
+            ye = self.down[e](ops.silu_and_mul(self.gate_up[e](xe)))
+            out.index_add_(0, rows, ye * flat_w[sel].unsqueeze(1).to(ye.dtype))
+        if comm.get_state().tp_size > 1:
+            out = comm.tp_all_reduce(out)
+        return out
+
+
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, config: ModelConfig, dtype: torch.dtype, layer_idx: int = 0):
         super().__init__()
         self.self_attn = LlamaAttention(config, dtype, layer_idx)
-        self.mlp = LlamaMLP(config, dtype, layer_idx)
+        self.mlp = (
+            MixtralMoE(config, dtype, layer_idx)
+            if config.num_local_experts > 0
+            else LlamaMLP(config, dtype, layer_idx)
+        )
         self.input_layernorm = nn.Parameter(
             torch.empty(config.hidden_size, dtype=dtype), requires_grad=False
         )
@@ -293,10 +360,36 @@ class LlamaForCausalLM(nn.Module):
                 vb,
             )
             layer.self_attn.o_proj.load_shard(get(p + "self_attn.o_proj.weight"))
-            layer.mlp.gate_up_proj.load_shards(
-                get(p + "mlp.gate_proj.weight"), get(p + "mlp.up_proj.weight")
-            )
-            layer.mlp.down_proj.load_shard(get(p + "mlp.down_proj.weight"))
+            if isinstance(layer.mlp, MixtralMoE):
+                moe = layer.mlp
+                if (p + "block_sparse_moe.gate.weight") in tensors:
+                    # classic Mixtral checkpoint: experts.j.w1/w3/w2
+                    moe.gate.data.copy_(
+                        get(p + "block_sparse_moe.gate.weight").to(moe.gate.dtype)
+                    )
+                    for j in range(moe.num_experts):
+                        ep = p + f"block_sparse_moe.experts.{j}."
+                        moe.gate_up[j].load_shards(
+                            get(ep + "w1.weight"), get(ep + "w3.weight")
+                        )
+                        moe.down[j].load_shard(get(ep + "w2.weight"))
+                else:
+                    # fused layout (transformers >= 4.5x): mlp.gate.weight,
+                    # mlp.experts.gate_up_proj [E, 2I, H], .down_proj [E, H, I]
+                    moe.gate.data.copy_(
+                        get(p + "mlp.gate.weight").to(moe.gate.dtype)
+                    )
+                    gu = get(p + "mlp.experts.gate_up_proj")
+                    dn = get(p + "mlp.experts.down_proj")
+                    inter = gu.shape[1] // 2
+                    for j in range(moe.num_experts):
+                        moe.gate_up[j].load_shards(gu[j][:inter], gu[j][inter:])
+                        moe.down[j].load_shard(dn[j])
+            else:
+                layer.mlp.gate_up_proj.load_shards(
+                    get(p + "mlp.gate_proj.weight"), get(p + "mlp.up_proj.weight")
+                )
+                layer.mlp.down_proj.load_shard(get(p + "mlp.down_proj.weight"))
 
     def num_parameters(self) -> int:
         return sum(p.numel() for p in self.parameters())

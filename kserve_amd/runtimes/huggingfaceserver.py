@@ -18,6 +18,7 @@ from kserve_amd.logging import configure_logging, logger
 GENERATIVE_ARCHITECTURES = (
     "LlamaForCausalLM",
     "MistralForCausalLM",
+    "MixtralForCausalLM",
     "Qwen2ForCausalLM",
 )
 ENCODER_SUFFIXES = (
@@ -40,7 +41,7 @@ def detect_backend(model_dir: str) -> str:
         if a.endswith("Model") and a.startswith(("Bert", "Roberta", "Distil")):
             return "encoder"
     model_type = cfg.get("model_type", "")
-    if model_type in ("llama", "mistral", "qwen2"):
+    if model_type in ("llama", "mistral", "mixtral", "qwen2"):
         return "engine"
     return "encoder"
 

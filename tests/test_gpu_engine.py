@@ -379,3 +379,42 @@ def test_prefix_cache_on_gpu():
     for ta, tb0, tb1 in zip(a, b0, b1):
         assert ta[0] == tb0[0] == tb1[0]
         assert len(tb1) == 8
+
+
+def test_mixtral_moe_on_gpu():
+    """Sparse-MoE engine path on GPU (eager: MoE routing is not yet
+    graph-captured); batched greedy equals single greedy."""
+    from kserve_amd.engine.config import (
+        CacheConfig,
+        EngineConfig,
+        ModelConfig,
+        SchedulerConfig,
+    )
+    from kserve_amd.engine.engine import LLMEngine
+    from kserve_amd.engine.sampling_params import SamplingParams
+
+    torch.manual_seed(3)
+    cfg = EngineConfig(
+        model=ModelConfig(
+            vocab_size=2048, hidden_size=256, intermediate_size=512,
+            num_layers=2, num_heads=2, num_kv_heads=1, head_dim=128,
+            max_position_embeddings=512, num_local_experts=4,
+            num_experts_per_tok=2, model_name="moe-gpu-tiny",
+        ),
+        cache=CacheConfig(block_size=16, num_gpu_blocks=64),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4, max_num_batched_tokens=1024, max_model_len=256
+        ),
+        device="cuda",
+        seed=0,
+        eos_token_id=-1,
+    )
+    engine = LLMEngine(cfg)
+    sp = SamplingParams(temperature=0.0, max_tokens=10)
+    out = engine.generate([[1, 2, 3], [9, 8, 7, 6]], sp)
+    assert all(len(o.output_token_ids) == 10 for o in out.values())
+    single = engine.generate([[1, 2, 3]], sp)
+    assert (
+        list(single.values())[0].output_token_ids
+        == list(out.values())[0].output_token_ids
+    )
