@@ -136,3 +136,42 @@ class TestStorage:
         rc = main([iris_model_dir, dest])
         assert rc == 0
         assert os.path.exists(os.path.join(dest, "model.joblib"))
+
+
+class TestPredictiveServer:
+    """Unified predictive runtime (reference python/predictiveserver):
+    framework dispatch + auto-detect + multi-model repository."""
+
+    def test_sklearn_dispatch_and_autodetect(self, iris_model_dir):
+        from kserve_amd.runtimes.predictiveserver import (
+            PredictiveServerModel,
+            detect_framework,
+        )
+
+        assert detect_framework(iris_model_dir) == "sklearn"
+        m = PredictiveServerModel("iris", iris_model_dir)  # auto-detected
+        assert m.framework == "sklearn"
+        m.load()
+        out = m.predict({"instances": [[5.1, 3.5, 1.4, 0.2]]})
+        assert out["predictions"][0] in (0, 1, 2)
+
+    def test_unsupported_framework_rejected(self, tmp_path):
+        from kserve_amd.runtimes.predictiveserver import PredictiveServerModel
+
+        with pytest.raises(ValueError, match="Unsupported framework"):
+            PredictiveServerModel("m", str(tmp_path), framework="tensorflow")
+
+    def test_multi_model_repository(self, iris_model_dir, tmp_path):
+        import shutil
+
+        from kserve_amd.runtimes.predictiveserver import (
+            PredictiveServerModelRepository,
+        )
+
+        models_dir = tmp_path / "models"
+        for name in ("iris-a", "iris-b"):
+            shutil.copytree(iris_model_dir, models_dir / name)
+        repo = PredictiveServerModelRepository(str(models_dir))
+        assert repo.get_model("iris-a") is not None
+        assert repo.get_model("iris-b") is not None
+        assert repo.is_model_ready("iris-a")
