@@ -26,7 +26,10 @@ constexpr int NWAVES = 4;     // waves per workgroup
 // HPW: q heads per wave (GQA group = NWAVES*HPW covered per workgroup).
 // OCC: min waves/EU hint (occupancy ablation KS_ATTN_OCC; 1 = compiler's
 // choice, 127 VGPRs -> 4 waves/SIMD on gfx950)
-template <int D, int HPW, int OCC = 1>
+// PB: broadcast the softmax probabilities through LDS (1 write + 4 b128
+// reads) instead of 16 sequential ds_bpermute shuffles per page
+// (ablation KS_ATTN_PB — probing whether the DS pipe bounds the loop)
+template <int D, int HPW, int OCC = 1, bool PB = false>
 __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     short* __restrict__ out,            // [S, H, D] bf16
     const short* __restrict__ q,        // [S, H, D]
@@ -55,6 +58,7 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
   const int wave = threadIdx.x >> 6;
   const int tok = lane >> 2;   // token within page handled by this lane
   const int part = lane & 3;   // quarter of D handled in the QK phase
+  __shared__ float p_bc[NWAVES][PAGE];  // PB variant: per-wave p broadcast
 
   const int nblocks = (ctx + PAGE - 1) / PAGE;
   // split-context range
@@ -160,14 +164,35 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 #pragma unroll
       for (int a = 0; a < ACC; ++a) acc[0][a] *= rescale;
       m[0] = m_new;
+      if constexpr (PB) {
+        // one LDS write per token group, then 4 x b128 reads give every
+        // lane all 16 p values (same-wave, no barrier; lgkm wait only)
+        if (part == 0) p_bc[wave][tok] = p;
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        float4_t pv[PAGE / 4];
 #pragma unroll
-      for (int t = 0; t < PAGE; ++t) {
-        const float pt = __shfl(p, t * 4, 64);
-        if constexpr (ACC == 2) {
-          acc[0][0] += pt * bf16_bits_to_float((short)(vreg[t] & 0xFFFF));
-          acc[0][1] += pt * bf16_bits_to_float((short)(vreg[t] >> 16));
-        } else {
-          acc[0][0] += pt * bf16_bits_to_float((short)vreg[t]);
+        for (int c = 0; c < PAGE / 4; ++c)
+          pv[c] = reinterpret_cast<const float4_t*>(p_bc[wave])[c];
+#pragma unroll
+        for (int t = 0; t < PAGE; ++t) {
+          const float pt = pv[t / 4][t % 4];
+          if constexpr (ACC == 2) {
+            acc[0][0] += pt * bf16_bits_to_float((short)(vreg[t] & 0xFFFF));
+            acc[0][1] += pt * bf16_bits_to_float((short)(vreg[t] >> 16));
+          } else {
+            acc[0][0] += pt * bf16_bits_to_float((short)vreg[t]);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int t = 0; t < PAGE; ++t) {
+          const float pt = __shfl(p, t * 4, 64);
+          if constexpr (ACC == 2) {
+            acc[0][0] += pt * bf16_bits_to_float((short)(vreg[t] & 0xFFFF));
+            acc[0][1] += pt * bf16_bits_to_float((short)(vreg[t] >> 16));
+          } else {
+            acc[0][0] += pt * bf16_bits_to_float((short)vreg[t]);
+          }
         }
       }
       continue;
@@ -881,6 +906,30 @@ extern "C" hipError_t ks_paged_attention_decode(
     const char* e = getenv("KS_ATTN_OCC");
     return e ? atoi(e) : 0;
   }();
+  static const bool use_pb = [] {
+    const char* e = getenv("KS_ATTN_PB");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_pb && hpw == 1 && head_dim == 128) {
+    hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true>), grid, block,
+                       0, stream, (short*)out, (const short*)q,
+                       (const short*)k_cache, (const short*)v_cache,
+                       (const int*)block_tables, (const int*)context_lens,
+                       scale, num_kv_heads, group, max_blocks, q_row_stride,
+                       n_splits, (float*)part_out, (float*)part_ml);
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      const long sh = (long)num_seqs * num_heads;
+      int wpb = 4;
+      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(wpb * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return hipSuccess;
+  }
   if (occ >= 5 && hpw == 1 && head_dim == 128) {
     if (occ >= 6) {
       hipLaunchKernelGGL((paged_attention_kernel<128, 1, 6>), grid, block, 0,
