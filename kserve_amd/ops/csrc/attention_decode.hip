@@ -155,11 +155,22 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
       }
       s = group_reduce_sum<4>(s);
       s = tok_valid ? s * scale : NEG_INF;
-      const float tmax = wave_reduce_max(s);
+      // s/p are uniform within each 4-lane token group: reducing across
+      // the 16 groups needs only the xor offsets {4,8,16,32} (4 DS ops,
+      // not the 6 of a full wave reduce)
+      float tmax = s;
+#pragma unroll
+      for (int off = 4; off < 64; off <<= 1)
+        tmax = fmaxf(tmax, __shfl_xor(tmax, off, 64));
       const float m_new = fmaxf(m[0], tmax);
       const float rescale = __expf(m[0] - m_new);
       const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
-      const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+      // xor offsets {4,8,16,32} never mix the low 2 lane bits, so each
+      // lane sums exactly one lane per token group: no overcount
+      float psum = p;
+#pragma unroll
+      for (int off = 4; off < 64; off <<= 1)
+        psum += __shfl_xor(psum, off, 64);
       l[0] = l[0] * rescale + psum;
 #pragma unroll
       for (int a = 0; a < ACC; ++a) acc[0][a] *= rescale;
