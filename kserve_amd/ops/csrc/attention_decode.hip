@@ -241,22 +241,36 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
       s = group_reduce_sum<4>(s);  // full dot in all 4 lanes of the token
       s = tok_valid ? s * scale : NEG_INF;
 
-      // ---- online softmax update ----
-      const float tmax = wave_reduce_max(s);
+      // ---- online softmax update (4-step reduces: s/p are uniform
+      // within each 4-lane token group) ----
+      float tmax = s;
+#pragma unroll
+      for (int off = 4; off < 64; off <<= 1)
+        tmax = fmaxf(tmax, __shfl_xor(tmax, off, 64));
       if (tmax > NEG_INF) {
         const float m_new = fmaxf(m[h], tmax);
         const float rescale = (m[h] > NEG_INF) ? __expf(m[h] - m_new) : 0.f;
         const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
-        const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
+        float psum = p;
+#pragma unroll
+        for (int off = 4; off < 64; off <<= 1)
+          psum += __shfl_xor(psum, off, 64);
         l[h] = l[h] * rescale + psum;
 #pragma unroll
         for (int a = 0; a < ACC; ++a) acc[h][a] *= rescale;
         m[h] = m_new;
 
-        // ---- PV: lane owns ACC consecutive output dims ----
+        // ---- PV via LDS p-broadcast (1 write + 4 b128 reads instead of
+        // 16 ds_bpermute) ----
+        if (part == 0) p_bc[wave][tok] = p;
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        float4_t pv4[PAGE / 4];
+#pragma unroll
+        for (int c = 0; c < PAGE / 4; ++c)
+          pv4[c] = reinterpret_cast<const float4_t*>(p_bc[wave])[c];
 #pragma unroll
         for (int t = 0; t < PAGE; ++t) {
-          const float pt = __shfl(p, t * 4, 64);
+          const float pt = pv4[t / 4][t % 4];
 #pragma unroll
           for (int a = 0; a < ACC; ++a) acc[h][a] += pt * vbuf[t][a];
         }
