@@ -29,7 +29,11 @@ constexpr int NWAVES = 4;     // waves per workgroup
 // PB: broadcast the softmax probabilities through LDS (1 write + 4 b128
 // reads) instead of 16 sequential ds_bpermute shuffles per page
 // (ablation KS_ATTN_PB — probing whether the DS pipe bounds the loop)
-template <int D, int HPW, int OCC = 1, bool PB = false>
+// D2: QK dot via v_dot2_f32_bf16 on packed bf16 pairs (halves the dot
+// VALU ops and the q-fragment registers; ablation KS_ATTN_D2)
+typedef __bf16 bf16x2v_t __attribute__((ext_vector_type(2)));
+
+template <int D, int HPW, int OCC = 1, bool PB = false, bool D2 = false>
 __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     short* __restrict__ out,            // [S, H, D] bf16
     const short* __restrict__ q,        // [S, H, D]
@@ -78,7 +82,8 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 
   // q fragments for this wave's heads: dims [part*D/4, (part+1)*D/4) f32
   constexpr int QFRAG = D / 4;  // dims per part
-  float q_frag[HPW][QFRAG];
+  float q_frag[HPW][D2 ? 1 : QFRAG];
+  bf16x2v_t q_pk[HPW][D2 ? QFRAG / 2 : 1];  // packed-bf16 variant (D2)
   int heads[HPW];
   bool hact[HPW];
   bool any_active = false;
@@ -92,8 +97,15 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     if (hact[h]) {
       const short* qp =
           q + (long)seq * q_row_stride + (long)head * D + part * QFRAG;
+      if constexpr (D2) {
+        const bf16x2v_t* qp2 = reinterpret_cast<const bf16x2v_t*>(qp);
 #pragma unroll
-      for (int j = 0; j < QFRAG; ++j) q_frag[h][j] = bf16_bits_to_float(qp[j]);
+        for (int j = 0; j < QFRAG / 2; ++j) q_pk[h][j] = qp2[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < QFRAG; ++j)
+          q_frag[h][j] = bf16_bits_to_float(qp[j]);
+      }
     }
   }
 
@@ -147,11 +159,22 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
         }
       }
       float s = 0.f;
+      if constexpr (D2) {
 #pragma unroll
-      for (int c = 0; c < QFRAG / 8; ++c) {
+        for (int c = 0; c < QFRAG / 8; ++c) {
+          const bf16x2v_t* k2 = reinterpret_cast<const bf16x2v_t*>(&kreg[c]);
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          s += q_frag[0][c * 8 + j] * bf16_bits_to_float(kreg[c][j]);
+          for (int j = 0; j < 4; ++j)
+            s = __builtin_amdgcn_fdot2_f32_bf16(
+                k2[j], q_pk[0][c * 4 + j], s, false);
+        }
+      } else {
+#pragma unroll
+        for (int c = 0; c < QFRAG / 8; ++c) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            s += q_frag[0][c * 8 + j] * bf16_bits_to_float(kreg[c][j]);
+        }
       }
       s = group_reduce_sum<4>(s);
       s = tok_valid ? s * scale : NEG_INF;
@@ -938,6 +961,30 @@ extern "C" hipError_t ks_paged_attention_decode(
     const char* e = getenv("KS_ATTN_PB");
     return e == nullptr || e[0] != '0';
   }();
+  static const bool use_d2 = [] {
+    const char* e = getenv("KS_ATTN_D2");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_d2 && hpw == 1 && head_dim == 128) {
+    hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true, true>), grid,
+                       block, 0, stream, (short*)out, (const short*)q,
+                       (const short*)k_cache, (const short*)v_cache,
+                       (const int*)block_tables, (const int*)context_lens,
+                       scale, num_kv_heads, group, max_blocks, q_row_stride,
+                       n_splits, (float*)part_out, (float*)part_ml);
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      const long sh = (long)num_seqs * num_heads;
+      int wpb = 4;
+      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(wpb * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return hipSuccess;
+  }
   if (use_pb && hpw == 1 && head_dim == 128) {
     hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true>), grid, block,
                        0, stream, (short*)out, (const short*)q,
