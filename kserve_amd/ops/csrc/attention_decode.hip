@@ -961,10 +961,16 @@ extern "C" hipError_t ks_paged_attention_decode(
     const char* e = getenv("KS_ATTN_PB");
     return e == nullptr || e[0] != '0';
   }();
-  static const bool use_d2 = [] {
+  // packed-bf16 dot wins in the latency regime (+10% at S=64, +5% at
+  // S=8 split-context) and is noise-level at large batch where the PV
+  // phase dominates; dispatch it for small batches. KS_ATTN_D2=1 forces
+  // it everywhere, =0 disables.
+  static const int d2_mode = [] {
     const char* e = getenv("KS_ATTN_D2");
-    return e != nullptr && e[0] == '1';
+    return e ? (e[0] == '1' ? 1 : 0) : -1;  // -1 = auto
   }();
+  const bool use_d2 =
+      d2_mode == 1 || (d2_mode == -1 && num_seqs <= 128);
   if (use_d2 && hpw == 1 && head_dim == 128) {
     hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true, true>), grid,
                        block, 0, stream, (short*)out, (const short*)q,
