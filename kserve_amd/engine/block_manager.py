@@ -116,6 +116,20 @@ class BlockManager:
         else:
             self._free.append(blk)
 
+    def query_cached_prefix(self, request: Request) -> int:
+        """Tokens of the request's prompt currently resident in the prefix
+        cache (pure lookup, no refcounting) — lets the scheduler size the
+        chunk before allocate() claims the blocks."""
+        if not self.enable_prefix_caching:
+            return 0
+        hashes = self._prompt_block_hashes(request, request.num_prompt_tokens)
+        n = 0
+        for h in hashes:
+            if h not in self._hash_to_block:
+                break
+            n += 1
+        return n * self.block_size
+
     def register_computed_blocks(self, request: Request) -> None:
         """Publish the request's fully-computed full prompt blocks into the
         prefix cache (called by the engine AFTER their KV exists)."""

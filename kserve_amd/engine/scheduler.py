@@ -108,9 +108,15 @@ class Scheduler:
         token_budget = self.config.max_num_batched_tokens
         while self.waiting:
             req = self.waiting[0]
+            # prefix cache: size the chunk from the cached-prefix boundary
+            # so budget/allocation cover exactly the tokens that will run
+            cached_hint = 0
+            if not req.block_table and req.num_computed_tokens == 0:
+                cached_hint = self.block_manager.query_cached_prefix(req)
+            base_computed = max(req.num_computed_tokens, cached_hint)
             # num_tokens (not num_prompt_tokens): a preempted request
             # re-prefills its generated tokens as context too
-            n_new = req.num_tokens - req.num_computed_tokens
+            n_new = req.num_tokens - base_computed
             if len(self.running) + len(batch.requests) + 1 > self.config.max_num_seqs:
                 break
             if n_new > token_budget:
@@ -128,16 +134,22 @@ class Scheduler:
                 req.block_table.extend(self.block_manager.take_blocks(need))
             else:
                 if not self.block_manager.can_allocate(
-                    req, req.num_computed_tokens + n_new
+                    req, base_computed + n_new
                 ):
                     break
-                self.block_manager.allocate(req, req.num_computed_tokens + n_new)
+                self.block_manager.allocate(req, base_computed + n_new)
                 cached = getattr(req, "num_cached_tokens", 0)
                 if cached > req.num_computed_tokens:
                     # prefix-cache hit: KV for the prefix already resident;
-                    # only the suffix runs (paged-context prefill path)
+                    # only the suffix runs (paged-context prefill path).
+                    # cached == cached_hint (same cache state as the query
+                    # above), so the allocation already covers the chunk;
+                    # the capacity clamp is a safety net only.
                     req.num_computed_tokens = cached
-                    n_new = req.num_tokens - cached
+                    capacity = (
+                        len(req.block_table) * self.block_manager.block_size
+                    )
+                    n_new = min(req.num_tokens - cached, capacity - cached)
             self.waiting.popleft()
             req.status = RequestStatus.RUNNING
             batch.requests.append(req)

@@ -125,3 +125,33 @@ def test_engine_prefix_cache_with_preemption():
     tiny = make_engine(True, blocks=28)  # pressure: eviction + preemption
     b = [o.output_token_ids for o in tiny.generate(prompts, sp).values()]
     assert a == b
+
+
+def test_prefix_cache_with_chunked_prefill():
+    """Both features on: cached prefix + chunked suffix compute."""
+    torch.manual_seed(0)
+    plain = make_engine(False)
+    sp = SamplingParams(temperature=0.0, max_tokens=6)
+    shared = list(range(1, 41))  # 40 tokens
+    prompts = [shared + [60 + i] for i in range(3)]
+    a = [o.output_token_ids for o in plain.generate(prompts, sp).values()]
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=128),
+        cache=CacheConfig(
+            block_size=4, num_gpu_blocks=128, enable_prefix_caching=True
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8,
+            max_num_batched_tokens=16,  # forces chunking of the 41-token prompts
+            max_model_len=128,
+            enable_chunked_prefill=True,
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    both = LLMEngine(cfg)
+    b0 = [o.output_token_ids for o in both.generate(prompts, sp).values()]
+    b1 = [o.output_token_ids for o in both.generate(prompts, sp).values()]
+    assert a == b0 == b1
+    assert both.scheduler.block_manager.cache_hit_tokens > 0
