@@ -33,7 +33,12 @@ constexpr int NWAVES = 4;     // waves per workgroup
 // VALU ops and the q-fragment registers; ablation KS_ATTN_D2)
 typedef __bf16 bf16x2v_t __attribute__((ext_vector_type(2)));
 
-template <int D, int HPW, int OCC = 1, bool PB = false, bool D2 = false>
+// V4: wide V loads — each lane reads 16 B (8 dims) of one token row, 2
+// tokens per lane per page (4 dwordx4 instead of 16 dword loads); the
+// per-dim accumulator holds a 2-token partial and the cross-token sum is
+// deferred to a single epilogue reduction (ablation KS_ATTN_V4)
+template <int D, int HPW, int OCC = 1, bool PB = false, bool D2 = false,
+          bool V4 = false>
 __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     short* __restrict__ out,            // [S, H, D] bf16
     const short* __restrict__ q,        // [S, H, D]
@@ -111,6 +116,7 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 
   float m[HPW], l[HPW];
   float acc[HPW][ACC];
+  float acc8[V4 ? 8 : 1];  // V4: 8-dim 2-token partial accumulator
 #pragma unroll
   for (int h = 0; h < HPW; ++h) {
     m[h] = NEG_INF;
@@ -118,6 +124,8 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 #pragma unroll
     for (int a = 0; a < ACC; ++a) acc[h][a] = 0.f;
   }
+#pragma unroll
+  for (int a = 0; a < (V4 ? 8 : 1); ++a) acc8[a] = 0.f;
 
   if (!any_active) return;
   // Direct-global K/V reads, no LDS, no barriers: each page is read once per
@@ -149,13 +157,25 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 #pragma unroll
       for (int c = 0; c < QFRAG / 8; ++c) kreg[c] = kp[c];
       unsigned int vreg[PAGE];
+      short8_t vwide[V4 ? 4 : 1];  // V4: 8 dims (16 B) of one token row/load
+      if constexpr (V4) {
+        // 16 lanes span a 128-dim row; lane covers dims [8*(lane%16), +8)
+        // of tokens lane/16, 4+lane/16, 8+lane/16, 12+lane/16
+        const int vdim = (lane & 15) * 8;
+        const int vtok = lane >> 4;
 #pragma unroll
-      for (int t = 0; t < PAGE; ++t) {
-        if constexpr (ACC == 2) {
-          vreg[t] = *reinterpret_cast<const unsigned int*>(
-              vpage + t * D + lane * ACC);
-        } else {
-          vreg[t] = (unsigned short)*(vpage + t * D + lane);
+        for (int q4 = 0; q4 < 4; ++q4)
+          vwide[q4] = *reinterpret_cast<const short8_t*>(
+              vpage + (q4 * 4 + vtok) * D + vdim);
+      } else {
+#pragma unroll
+        for (int t = 0; t < PAGE; ++t) {
+          if constexpr (ACC == 2) {
+            vreg[t] = *reinterpret_cast<const unsigned int*>(
+                vpage + t * D + lane * ACC);
+          } else {
+            vreg[t] = (unsigned short)*(vpage + t * D + lane);
+          }
         }
       }
       float s = 0.f;
@@ -198,7 +218,23 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 #pragma unroll
       for (int a = 0; a < ACC; ++a) acc[0][a] *= rescale;
       m[0] = m_new;
-      if constexpr (PB) {
+      if constexpr (V4) {
+        // V4 PV: lane multiplies its two token rows by their p (read from
+        // the broadcast) and accumulates 8-dim partials; rescale applies
+        // to partials exactly like the full sum
+        if (part == 0) p_bc[wave][tok] = p;
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        const int vtok = lane >> 4;
+#pragma unroll
+        for (int a = 0; a < 8; ++a) acc8[a] *= rescale;
+#pragma unroll
+        for (int q4 = 0; q4 < 4; ++q4) {
+          const float pt = p_bc[wave][q4 * 4 + vtok];
+#pragma unroll
+          for (int a = 0; a < 8; ++a)
+            acc8[a] += pt * bf16_bits_to_float(vwide[q4][a]);
+        }
+      } else if constexpr (PB) {
         // one LDS write per token group, then 4 x b128 reads give every
         // lane all 16 p values (same-wave, no barrier; lgkm wait only)
         if (part == 0) p_bc[wave][tok] = p;
@@ -305,6 +341,41 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
   for (int h = 0; h < HPW; ++h) {
     if (!hact[h]) continue;
     const int head = heads[h];
+    if constexpr (V4) {
+      // cross token-slice reduction (deferred from the per-page PV): lanes
+      // sharing lane%16 hold 4-token partials of the same 8 dims
+#pragma unroll
+      for (int a = 0; a < 8; ++a) {
+        acc8[a] += __shfl_xor(acc8[a], 16, 64);
+        acc8[a] += __shfl_xor(acc8[a], 32, 64);
+      }
+      const int vdim = (lane & 15) * 8;
+      if (n_splits == 1) {
+        const float inv_l = (l[h] > 0.f) ? 1.f / l[h] : 0.f;
+        if (lane < 16) {
+          short* op = out + ((long)seq * num_heads + head) * D + vdim;
+#pragma unroll
+          for (int a = 0; a < 8; ++a)
+            op[a] = float_to_bf16_bits(acc8[a] * inv_l);
+        }
+      } else {
+        if (lane < 16) {
+          float* po =
+              part_out +
+              ((((long)seq * num_heads + head) * n_splits + split)) * D +
+              vdim;
+#pragma unroll
+          for (int a = 0; a < 8; ++a) po[a] = acc8[a];
+        }
+        if (lane == 0) {
+          float* ml = part_ml +
+                      (((long)seq * num_heads + head) * n_splits + split) * 2;
+          ml[0] = m[h];
+          ml[1] = l[h];
+        }
+      }
+      continue;
+    }
     if (n_splits == 1) {
       const float inv_l = (l[h] > 0.f) ? 1.f / l[h] : 0.f;
       short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
@@ -965,6 +1036,30 @@ extern "C" hipError_t ks_paged_attention_decode(
   // S=8 split-context) and is noise-level at large batch where the PV
   // phase dominates; dispatch it for small batches. KS_ATTN_D2=1 forces
   // it everywhere, =0 disables.
+  static const bool use_v4 = [] {
+    const char* e = getenv("KS_ATTN_V4");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_v4 && hpw == 1 && head_dim == 128) {
+    hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true, false, true>),
+                       grid, block, 0, stream, (short*)out, (const short*)q,
+                       (const short*)k_cache, (const short*)v_cache,
+                       (const int*)block_tables, (const int*)context_lens,
+                       scale, num_kv_heads, group, max_blocks, q_row_stride,
+                       n_splits, (float*)part_out, (float*)part_ml);
+    HIP_CHECK_KERNEL();
+    if (n_splits > 1) {
+      const long sh = (long)num_seqs * num_heads;
+      int wpb = 4;
+      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                         dim3(wpb * 64), 0, stream, (short*)out,
+                         (const float*)part_out, (const float*)part_ml, sh,
+                         n_splits);
+      HIP_CHECK_KERNEL();
+    }
+    return hipSuccess;
+  }
   static const int d2_mode = [] {
     const char* e = getenv("KS_ATTN_D2");
     return e ? (e[0] == '1' ? 1 : 0) : -1;  // -1 = auto
