@@ -1036,10 +1036,16 @@ extern "C" hipError_t ks_paged_attention_decode(
   // S=8 split-context) and is noise-level at large batch where the PV
   // phase dominates; dispatch it for small batches. KS_ATTN_D2=1 forces
   // it everywhere, =0 disables.
-  static const bool use_v4 = [] {
+  // wide-V-load variant wins the throughput regime (+9% at S=256/512,
+  // +23% at long context) but drops occupancy (110 VGPR) and loses ~8%
+  // in the latency regime — auto-dispatch by batch size.
+  // KS_ATTN_V4=1 forces it, =0 disables.
+  static const int v4_mode = [] {
     const char* e = getenv("KS_ATTN_V4");
-    return e != nullptr && e[0] == '1';
+    return e ? (e[0] == '1' ? 1 : 0) : -1;  // -1 = auto
   }();
+  const bool use_v4 =
+      v4_mode == 1 || (v4_mode == -1 && num_seqs > 128);
   if (use_v4 && hpw == 1 && head_dim == 128) {
     hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true, false, true>),
                        grid, block, 0, stream, (short*)out, (const short*)q,
