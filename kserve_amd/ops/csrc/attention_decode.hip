@@ -447,6 +447,7 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
   }
   if (!active) return;
 
+  __shared__ float p_bc8[4][PAGE];
   float q_frag[QFRAG];
   {
     const short* qp =
@@ -455,9 +456,14 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
     for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
   }
   float m = NEG_INF, l = 0.f;
-  float acc[ACC];
+  // wide-V scheme (same as the bf16 V4 path): 8 lanes span a 128-B fp8
+  // row; lane covers dims [16*(lane%8), +16) of tokens lane/8 and
+  // 8+lane/8; accumulator keeps 2-token partials, summed in the epilogue
+  float acc[16];
 #pragma unroll
-  for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
+  for (int a = 0; a < 16; ++a) acc[a] = 0.f;
+  const int vdim = (lane & 7) * 16;
+  const int vtok8 = lane >> 3;
 
   const int bt_base = (int)((long)seq * max_blocks);
   for (int bi = blk_lo; bi < blk_hi; ++bi) {
@@ -468,17 +474,17 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
         v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
     const int gtok = bi * PAGE + tok;
     const bool tok_valid = gtok < ctx;
-    // burst-issue: K quarter-row (32 B) + V dims (2 B x 16 tokens)
+    // burst-issue: K quarter-row (32 B) + V 16-B row chunks (2 tokens)
     uint4_t kreg[QFRAG / 16];
 #pragma unroll
     for (int c = 0; c < QFRAG / 16; ++c)
       kreg[c] = reinterpret_cast<const uint4_t*>(
           page + tok * D + part * QFRAG)[c];
-    unsigned short vreg[PAGE];
+    uint4_t vwide[2];
 #pragma unroll
-    for (int t = 0; t < PAGE; ++t)
-      vreg[t] = *reinterpret_cast<const unsigned short*>(
-          vpage + t * D + lane * ACC);
+    for (int q8 = 0; q8 < 2; ++q8)
+      vwide[q8] = *reinterpret_cast<const uint4_t*>(
+          vpage + (q8 * 8 + vtok8) * D + vdim);
     float s = 0.f;
 #pragma unroll
     for (int c = 0; c < QFRAG / 16; ++c) {
@@ -501,28 +507,49 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
     const float psum = wave_reduce_sum(part == 0 ? p : 0.f);
     l = l * rescale + psum;
 #pragma unroll
-    for (int a = 0; a < ACC; ++a) acc[a] *= rescale;
+    for (int a = 0; a < 16; ++a) acc[a] *= rescale;
     m = m_new;
+    if (part == 0) p_bc8[wave][tok] = p;
+    asm volatile("s_waitcnt lgkmcnt(0)");
 #pragma unroll
-    for (int t = 0; t < PAGE; ++t) {
-      const float pt = __shfl(p, t * 4, 64);
-      const float2_t v2 = __builtin_amdgcn_cvt_pk_f32_fp8((int)vreg[t], false);
-      acc[0] += pt * v2[0];
-      acc[1] += pt * v2[1];
+    for (int q8 = 0; q8 < 2; ++q8) {
+      const float pt = p_bc8[wave][q8 * 8 + vtok8];
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        const int ui = (int)vwide[q8][w];
+        const float2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(ui, false);
+        const float2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(ui, true);
+        acc[w * 4 + 0] += pt * lo[0];
+        acc[w * 4 + 1] += pt * lo[1];
+        acc[w * 4 + 2] += pt * hi[0];
+        acc[w * 4 + 3] += pt * hi[1];
+      }
     }
   }
 
+  // sum the 2-token partials: lanes sharing lane%8 (xor 8,16,32)
+#pragma unroll
+  for (int a = 0; a < 16; ++a) {
+#pragma unroll
+    for (int off = 8; off < 64; off <<= 1)
+      acc[a] += __shfl_xor(acc[a], off, 64);
+  }
   if (n_splits == 1) {
     const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
-    short* op = out + ((long)seq * num_heads + head) * D + lane * ACC;
+    if (lane < 8) {
+      short* op = out + ((long)seq * num_heads + head) * D + vdim;
 #pragma unroll
-    for (int a = 0; a < ACC; ++a) op[a] = float_to_bf16_bits(acc[a] * inv_l);
+      for (int a = 0; a < 16; ++a)
+        op[a] = float_to_bf16_bits(acc[a] * inv_l);
+    }
   } else {
-    float* po = part_out +
-                ((((long)seq * num_heads + head) * n_splits + split)) * D +
-                lane * ACC;
+    if (lane < 8) {
+      float* po = part_out +
+                  ((((long)seq * num_heads + head) * n_splits + split)) * D +
+                  vdim;
 #pragma unroll
-    for (int a = 0; a < ACC; ++a) po[a] = acc[a];
+      for (int a = 0; a < 16; ++a) po[a] = acc[a];
+    }
     if (lane == 0) {
       float* ml =
           part_ml + (((long)seq * num_heads + head) * n_splits + split) * 2;
@@ -574,6 +601,7 @@ __global__ __launch_bounds__(256) void paged_attention_u2_kernel(
   const int head = kv_head * group + wave;
   if (wave >= group) return;
 
+  __shared__ float p_bc8[4][PAGE];
   float q_frag[QFRAG];
   {
     const short* qp =
@@ -582,9 +610,14 @@ __global__ __launch_bounds__(256) void paged_attention_u2_kernel(
     for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
   }
   float m = NEG_INF, l = 0.f;
-  float acc[ACC];
+  // wide-V scheme (same as the bf16 V4 path): 8 lanes span a 128-B fp8
+  // row; lane covers dims [16*(lane%8), +16) of tokens lane/8 and
+  // 8+lane/8; accumulator keeps 2-token partials, summed in the epilogue
+  float acc[16];
 #pragma unroll
-  for (int a = 0; a < ACC; ++a) acc[a] = 0.f;
+  for (int a = 0; a < 16; ++a) acc[a] = 0.f;
+  const int vdim = (lane & 7) * 16;
+  const int vtok8 = lane >> 3;
 
   const int bt_base = (int)((long)seq * max_blocks);
 
