@@ -1079,13 +1079,27 @@ extern "C" hipError_t ks_paged_attention_decode(
   }();
   const bool use_v4 =
       v4_mode == 1 || (v4_mode == -1 && num_seqs > 128);
+  static const bool v4_d2 = [] {  // A/B: packed-bf16 dot inside V4
+    const char* e = getenv("KS_ATTN_V4D2");
+    return e != nullptr && e[0] == '1';
+  }();
   if (use_v4 && hpw == 1 && head_dim == 128) {
-    hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true, false, true>),
-                       grid, block, 0, stream, (short*)out, (const short*)q,
-                       (const short*)k_cache, (const short*)v_cache,
-                       (const int*)block_tables, (const int*)context_lens,
-                       scale, num_kv_heads, group, max_blocks, q_row_stride,
-                       n_splits, (float*)part_out, (float*)part_ml);
+    if (v4_d2) {
+      hipLaunchKernelGGL(
+          (paged_attention_kernel<128, 1, 1, true, true, true>), grid, block,
+          0, stream, (short*)out, (const short*)q, (const short*)k_cache,
+          (const short*)v_cache, (const int*)block_tables,
+          (const int*)context_lens, scale, num_kv_heads, group, max_blocks,
+          q_row_stride, n_splits, (float*)part_out, (float*)part_ml);
+    } else {
+      hipLaunchKernelGGL(
+          (paged_attention_kernel<128, 1, 1, true, false, true>), grid,
+          block, 0, stream, (short*)out, (const short*)q,
+          (const short*)k_cache, (const short*)v_cache,
+          (const int*)block_tables, (const int*)context_lens, scale,
+          num_kv_heads, group, max_blocks, q_row_stride, n_splits,
+          (float*)part_out, (float*)part_ml);
+    }
     HIP_CHECK_KERNEL();
     if (n_splits > 1) {
       const long sh = (long)num_seqs * num_heads;
