@@ -161,6 +161,7 @@ class LLMEngine:
             self.device.type == "cuda"
             and not config.enforce_eager
             and config.model.num_local_experts == 0
+            and comm.get_state().pp_size == 1
         ):
             # MoE routing (data-dependent expert buckets) is not yet
             # hipGraph-capturable; MoE decodes run eager
@@ -263,7 +264,8 @@ class LLMEngine:
             # speculative decoding (prompt-lookup): draft from the request's
             # own context, verify k+1 positions in ONE forward through the
             # paged-context prefill path; exact greedy outputs by construction
-            if sched.speculative_ngram > 0 and pure_greedy:
+            pp = comm.get_state().pp_size
+            if sched.speculative_ngram > 0 and pure_greedy and pp == 1:
                 k_cap = (
                     self.scheduler.reserve_decode_window(
                         batch, sched.speculative_ngram + 1
@@ -279,7 +281,7 @@ class LLMEngine:
             # multi-step window: all-greedy decode with no pending scheduling
             # events runs as back-to-back hipGraph replays
             k = 1
-            if self.config.scheduler.multi_step > 1 and all(
+            if self.config.scheduler.multi_step > 1 and comm.get_state().pp_size == 1 and all(
                 r.sampling_params.greedy
                 and r.sampling_params.logprobs is None
                 and r.sampling_params.presence_penalty == 0.0
@@ -323,13 +325,13 @@ class LLMEngine:
         outputs: List[RequestOutput] = []
         finished: List[Request] = []
         if sample_reqs:
-            tokens = self.sampler.sample(logits, sample_reqs)
+            tokens = self._sample_tokens(logits, sample_reqs)
             # top-N logprobs for the (rare) requests that ask for them
             lp_idx = [
                 i for i, r in enumerate(sample_reqs)
                 if r.sampling_params.logprobs is not None
             ]
-            if lp_idx:
+            if lp_idx and logits is not None:
                 n_top = max(
                     sample_reqs[i].sampling_params.logprobs or 1 for i in lp_idx
                 )
@@ -387,6 +389,22 @@ class LLMEngine:
             self.runner.release_request(req.request_id)
         LLM_KV_USAGE.set(self.scheduler.block_manager.usage)
         return outputs
+
+    def _sample_tokens(self, logits, sample_reqs) -> List[int]:
+        """Sample on the last pipeline stage and broadcast the token ids so
+        every stage's scheduler advances in lockstep (pp_size=1: direct)."""
+        st = comm.get_state()
+        if st.pp_size == 1:
+            return self.sampler.sample(logits, sample_reqs)
+        if st.is_last_pp:
+            tokens = self.sampler.sample(logits, sample_reqs)
+            t = torch.tensor(tokens, dtype=torch.int64, device=self.device)
+        else:
+            t = torch.empty(
+                len(sample_reqs), dtype=torch.int64, device=self.device
+            )
+        comm.pp_broadcast_from_last(t)
+        return t.tolist()
 
     def _draft_ngram(self, req: Request, k: int) -> List[int]:
         """Prompt-lookup drafting: if the last n-gram (n in [min,max]) occurred

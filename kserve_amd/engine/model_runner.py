@@ -76,7 +76,7 @@ class ModelRunner:
                 m.kv_bytes_per_token_per_layer
                 * kv_itemsize
                 * self.config.cache.block_size
-                * m.num_layers
+                * self._num_local_layers()
                 // (tp * 2)
             )
             num_blocks = self.config.cache.num_gpu_blocks or max(
@@ -87,6 +87,9 @@ class ModelRunner:
             num_blocks = self.config.cache.num_gpu_blocks or 512
         self.allocate_kv_cache(num_blocks)
         return num_blocks
+
+    def _num_local_layers(self) -> int:
+        return getattr(self.model, "num_local_layers", self.config.model.num_layers)
 
     def allocate_kv_cache(self, num_blocks: int):
         m = self.config.model
@@ -106,13 +109,13 @@ class ModelRunner:
                 alloc(shape, dtype=dtype, device=self.device),
                 alloc(shape, dtype=dtype, device=self.device),
             )
-            for _ in range(m.num_layers)
+            for _ in range(self._num_local_layers())
         ]
         self.num_gpu_blocks = num_blocks
         self._allocate_cpu_kv()
         kv_gib = (
             2 * num_blocks * kv_heads_local * self.config.cache.block_size
-            * m.head_dim * m.num_layers * dtype.itemsize
+            * m.head_dim * self._num_local_layers() * dtype.itemsize
         ) / (1 << 30)
         if dtype.itemsize == 1:
             logger.info("KV cache dtype: fp8_e4m3 (scale 1.0)")
@@ -133,7 +136,7 @@ class ModelRunner:
         kv_dtype = cache.cache_torch_dtype(self.model.dtype)
         per_block = (
             2 * kv_heads_local * cache.block_size * m.head_dim
-            * m.num_layers * kv_dtype.itemsize
+            * self._num_local_layers() * kv_dtype.itemsize
         )
         n = cache.num_cpu_blocks
         if not n and cache.cpu_offload_bytes:
@@ -147,7 +150,7 @@ class ModelRunner:
                 torch.empty(shape, dtype=kv_dtype, pin_memory=pin),
                 torch.empty(shape, dtype=kv_dtype, pin_memory=pin),
             )
-            for _ in range(m.num_layers)
+            for _ in range(self._num_local_layers())
         ]
         self.num_cpu_blocks = n
         logger.info(

@@ -258,17 +258,42 @@ class LlamaForCausalLM(nn.Module):
             torch.bfloat16 if config.dtype == "bfloat16" else torch.float32
         )
         self.dtype = dtype
-        self.embed_tokens = VocabParallelEmbedding(
-            config.vocab_size, config.hidden_size, dtype=dtype
+        # pipeline parallelism: this rank holds layers [pp_first, pp_last);
+        # embeddings live on the first stage, norm + lm_head on the last
+        st = comm.get_state()
+        per = -(-config.num_layers // st.pp_size)
+        self.pp_first = st.pp_rank * per
+        self.pp_last = min(config.num_layers, self.pp_first + per)
+        self.is_first_stage = st.is_first_pp
+        self.is_last_stage = st.is_last_pp
+        self.embed_tokens = (
+            VocabParallelEmbedding(
+                config.vocab_size, config.hidden_size, dtype=dtype
+            )
+            if self.is_first_stage
+            else None
         )
         self.layers = nn.ModuleList(
-            [LlamaDecoderLayer(config, dtype, i) for i in range(config.num_layers)]
+            [
+                LlamaDecoderLayer(config, dtype, i)
+                for i in range(self.pp_first, self.pp_last)
+            ]
         )
-        self.norm = nn.Parameter(
-            torch.empty(config.hidden_size, dtype=dtype), requires_grad=False
+        self.num_local_layers = len(self.layers)
+        self.norm = (
+            nn.Parameter(
+                torch.empty(config.hidden_size, dtype=dtype), requires_grad=False
+            )
+            if self.is_last_stage
+            else None
         )
-        self.lm_head = ColumnParallelLinear(
-            config.hidden_size, config.vocab_size, dtype=dtype, gather_output=True
+        self.lm_head = (
+            ColumnParallelLinear(
+                config.hidden_size, config.vocab_size, dtype=dtype,
+                gather_output=True,
+            )
+            if self.is_last_stage
+            else None
         )
         self.register_buffer(
             "cos_sin_cache",
@@ -288,7 +313,8 @@ class LlamaForCausalLM(nn.Module):
     def random_init(self, std: float = 0.02, seed: int = 0):
         """Fast random init for the synthetic benchmark (BASELINE.json:
         'random-init weights')."""
-        gen = torch.Generator(device=self.embed_tokens.weight.device)
+        dev = next(self.parameters()).device
+        gen = torch.Generator(device=dev)
         gen.manual_seed(seed + comm.get_state().tp_rank)
         for p in self.parameters():
             if p.dim() >= 2:
@@ -307,17 +333,33 @@ class LlamaForCausalLM(nn.Module):
         kv_caches: List[Tuple[torch.Tensor, torch.Tensor]],
         meta: AttentionMetadata,
     ) -> torch.Tensor:
-        hidden = self.embed_tokens(input_ids)
-        residual = None
+        st = comm.get_state()
+        if self.is_first_stage:
+            hidden = self.embed_tokens(input_ids)
+            residual = None
+        else:
+            # synchronous pipeline: activations + residual arrive from the
+            # previous stage over RCCL p2p (xGMI)
+            T = input_ids.shape[0]
+            H = self.config.hidden_size
+            hidden = comm.pp_recv((T, H), self.dtype, input_ids.device)
+            residual = comm.pp_recv((T, H), self.dtype, input_ids.device)
         for i, layer in enumerate(self.layers):
             hidden, residual = layer(
                 hidden, residual, positions, self.cos_sin_cache, kv_caches[i], meta
             )
+        if not self.is_last_stage:
+            comm.pp_send(hidden)
+            comm.pp_send(residual)
+            return hidden
         hidden, _ = ops.fused_add_rms_norm(hidden, residual, self.norm, self.eps)
         return hidden
 
-    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        """hidden: [N, H] (already gathered to sampled positions)."""
+    def compute_logits(self, hidden: torch.Tensor):
+        """hidden: [N, H] (already gathered to sampled positions).
+        None on non-final pipeline stages."""
+        if self.lm_head is None:
+            return None
         return self.lm_head(hidden)
 
     # -- HF weight loading ------------------------------------------------------
@@ -332,14 +374,17 @@ class LlamaForCausalLM(nn.Module):
             t = tensors[name]
             return t() if callable(t) else t
 
-        self.embed_tokens.load_shard(get("model.embed_tokens.weight"))
-        self.norm.data.copy_(get("model.norm.weight").to(self.dtype))
-        if self.config.tie_word_embeddings or "lm_head.weight" not in tensors:
-            self.lm_head.load_shard(get("model.embed_tokens.weight"))
-        else:
-            self.lm_head.load_shard(get("lm_head.weight"))
+        if self.embed_tokens is not None:
+            self.embed_tokens.load_shard(get("model.embed_tokens.weight"))
+        if self.norm is not None:
+            self.norm.data.copy_(get("model.norm.weight").to(self.dtype))
+        if self.lm_head is not None:
+            if self.config.tie_word_embeddings or "lm_head.weight" not in tensors:
+                self.lm_head.load_shard(get("model.embed_tokens.weight"))
+            else:
+                self.lm_head.load_shard(get("lm_head.weight"))
         for i, layer in enumerate(self.layers):
-            p = f"model.layers.{i}."
+            p = f"model.layers.{self.pp_first + i}."
             layer.input_layernorm.data.copy_(
                 get(p + "input_layernorm.weight").to(self.dtype)
             )

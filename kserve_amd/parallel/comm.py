@@ -30,8 +30,21 @@ class ParallelState:
     tp_size: int = 1
     dp_rank: int = 0
     dp_size: int = 1
+    pp_rank: int = 0
+    pp_size: int = 1
     tp_group: Optional[object] = None
+    pp_group: Optional[object] = None
+    pp_prev: int = -1  # global rank of the previous pipeline stage
+    pp_next: int = -1
     device: str = "cpu"
+
+    @property
+    def is_first_pp(self) -> bool:
+        return self.pp_rank == 0
+
+    @property
+    def is_last_pp(self) -> bool:
+        return self.pp_rank == self.pp_size - 1
 
 
 _STATE = ParallelState()
@@ -49,9 +62,11 @@ def init_distributed(
     tp_size: Optional[int] = None,
     backend: Optional[str] = None,
     timeout_s: int = 600,
+    pp_size: int = 1,
 ) -> ParallelState:
-    """Initialize from torchrun env vars. World splits into DP replicas of
-    TP groups: ranks [i*tp, (i+1)*tp) form TP group i."""
+    """Initialize from torchrun env vars. Rank layout (slow to fast):
+    (dp, pp, tp) — world splits into DP replicas of PP pipelines of TP
+    groups. rank = ((dp*pp_size + pp)*tp_size + tp)."""
     global _STATE
     rank = int(os.environ.get("RANK", "0"))
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -76,23 +91,42 @@ def init_distributed(
             timeout=datetime.timedelta(seconds=timeout_s),
         )
 
-    tp_size = tp_size or world_size
-    assert world_size % tp_size == 0, "world_size must be divisible by tp_size"
-    dp_size = world_size // tp_size
+    tp_size = tp_size or (world_size // pp_size)
+    assert world_size % (tp_size * pp_size) == 0, (
+        "world_size must be divisible by tp_size*pp_size"
+    )
+    dp_size = world_size // (tp_size * pp_size)
     tp_rank = rank % tp_size
-    dp_rank = rank // tp_size
+    pp_rank = (rank // tp_size) % pp_size
+    dp_rank = rank // (tp_size * pp_size)
 
     tp_group = None
+    pp_group = None
+    pp_prev = pp_next = -1
     if world_size > 1:
         if tp_size == world_size:
             tp_group = dist.group.WORLD
         else:
             # build all TP subgroups (every rank must call new_group)
-            for i in range(dp_size):
-                ranks = list(range(i * tp_size, (i + 1) * tp_size))
-                g = dist.new_group(ranks)
-                if dp_rank == i:
-                    tp_group = g
+            for d in range(dp_size):
+                for s in range(pp_size):
+                    base = (d * pp_size + s) * tp_size
+                    g = dist.new_group(list(range(base, base + tp_size)))
+                    if dp_rank == d and pp_rank == s:
+                        tp_group = g
+        if pp_size > 1:
+            for d in range(dp_size):
+                for t in range(tp_size):
+                    ranks = [
+                        (d * pp_size + s) * tp_size + t for s in range(pp_size)
+                    ]
+                    g = dist.new_group(ranks)
+                    if dp_rank == d and tp_rank == t:
+                        pp_group = g
+            if pp_rank > 0:
+                pp_prev = (dp_rank * pp_size + pp_rank - 1) * tp_size + tp_rank
+            if pp_rank < pp_size - 1:
+                pp_next = (dp_rank * pp_size + pp_rank + 1) * tp_size + tp_rank
 
     _STATE = ParallelState(
         rank=rank,
@@ -102,7 +136,12 @@ def init_distributed(
         tp_size=tp_size,
         dp_rank=dp_rank,
         dp_size=dp_size,
+        pp_rank=pp_rank,
+        pp_size=pp_size,
         tp_group=tp_group,
+        pp_group=pp_group,
+        pp_prev=pp_prev,
+        pp_next=pp_next,
         device=device,
     )
     return _STATE
@@ -133,3 +172,28 @@ def tp_all_gather(t: torch.Tensor, dim: int = -1) -> torch.Tensor:
 def barrier():
     if dist.is_initialized():
         dist.barrier()
+
+
+# ---- pipeline-parallel point-to-point (RCCL send/recv over xGMI) ----------
+
+def pp_send(t: torch.Tensor) -> None:
+    dist.send(t.contiguous(), dst=_STATE.pp_next)
+
+
+def pp_recv(shape, dtype, device) -> torch.Tensor:
+    t = torch.empty(shape, dtype=dtype, device=device)
+    dist.recv(t, src=_STATE.pp_prev)
+    return t
+
+
+def pp_broadcast_from_last(t: torch.Tensor) -> torch.Tensor:
+    """Broadcast a tensor from the LAST pipeline stage to all stages of
+    this (dp, tp) pipeline (sampled tokens travel back to the schedulers)."""
+    if _STATE.pp_size == 1:
+        return t
+    last_global = (
+        (_STATE.dp_rank * _STATE.pp_size + _STATE.pp_size - 1) * _STATE.tp_size
+        + _STATE.tp_rank
+    )
+    dist.broadcast(t, src=last_global, group=_STATE.pp_group)
+    return t

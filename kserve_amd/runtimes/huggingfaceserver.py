@@ -62,7 +62,10 @@ def build_model(args):
         from kserve_amd.parallel import comm
         from kserve_amd.runtimes.llm_model import LLMModel
 
-        comm.init_distributed(tp_size=getattr(args, "tensor_parallel_size", 1))
+        comm.init_distributed(
+            tp_size=getattr(args, "tensor_parallel_size", 1),
+            pp_size=getattr(args, "pipeline_parallel_size", 1),
+        )
         from transformers import AutoTokenizer
 
         tokenizer = AutoTokenizer.from_pretrained(args.model_dir)
@@ -79,7 +82,8 @@ def build_model(args):
                 max_model_len=getattr(args, "max_model_len", 8192),
             ),
             parallel=ParallelConfig(
-                tensor_parallel_size=getattr(args, "tensor_parallel_size", 1)
+                tensor_parallel_size=getattr(args, "tensor_parallel_size", 1),
+                pipeline_parallel_size=getattr(args, "pipeline_parallel_size", 1),
             ),
             model_path=args.model_dir,
             device="cuda" if torch.cuda.is_available() else "cpu",
@@ -110,6 +114,7 @@ def main(argv=None):
     parser.add_argument("--backend", default=None, choices=["engine", "encoder"])
     parser.add_argument("--task", default=None)
     parser.add_argument("--tensor-parallel-size", dest="tensor_parallel_size", type=int, default=1)
+    parser.add_argument("--pipeline-parallel-size", dest="pipeline_parallel_size", type=int, default=1)
     parser.add_argument("--max_model_len", type=int, default=8192)
     parser.add_argument("--max_num_seqs", type=int, default=256)
     # data-parallel contract flags (reference preset
