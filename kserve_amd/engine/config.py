@@ -44,6 +44,9 @@ class ModelConfig:
     # mixture-of-experts (Mixtral): 0 experts = dense MLP
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
+    # EP: partition experts across the parallel group (full-width weights)
+    # instead of TP-sharding every expert
+    expert_parallel: bool = False
 
     def __post_init__(self):
         if self.head_dim is None:

@@ -160,33 +160,76 @@ class MixtralMoE(nn.Module):
     all-reduce covers the whole block.
     """
 
-    def __init__(self, config: ModelConfig, dtype: torch.dtype, layer_idx: int = 0):
+    def __init__(
+        self,
+        config: ModelConfig,
+        dtype: torch.dtype,
+        layer_idx: int = 0,
+        expert_parallel: bool = False,
+    ):
         super().__init__()
+        st = comm.get_state()
         self.layer_idx = layer_idx
         self.num_experts = config.num_local_experts
         self.top_k = config.num_experts_per_tok
+        # expert parallelism: experts are PARTITIONED across the group
+        # (full-width weights per expert) instead of every expert being
+        # TP-sharded; the per-block all-reduce sums the partial outputs
+        # either way (reference --enable-expert-parallel / ParallelismSpec
+        # .Expert — vLLM EP there, RCCL all-reduce here).
+        self.expert_parallel = expert_parallel and st.tp_size > 1
+        if self.expert_parallel:
+            assert self.num_experts % st.tp_size == 0, (
+                "num_local_experts must divide the group size for EP"
+            )
+            self.experts_per_rank = self.num_experts // st.tp_size
+            self.expert_lo = st.tp_rank * self.experts_per_rank
+        else:
+            self.experts_per_rank = self.num_experts
+            self.expert_lo = 0
         self.gate = nn.Parameter(
             torch.empty(self.num_experts, config.hidden_size, dtype=dtype),
             requires_grad=False,
         )
-        self.gate_up = nn.ModuleList(
-            [
-                MergedColumnParallelLinear(
-                    config.hidden_size, config.intermediate_size, dtype=dtype
-                )
-                for _ in range(self.num_experts)
-            ]
-        )
-        self.down = nn.ModuleList(
-            [
-                RowParallelLinear(
-                    config.intermediate_size,
-                    config.hidden_size,
-                    dtype=dtype,
-                    reduce_output=False,  # one all-reduce for the whole block
-                )
-                for _ in range(self.num_experts)
-            ]
+        if self.expert_parallel:
+            H, inter = config.hidden_size, config.intermediate_size
+            self.ep_gate_up = nn.Parameter(
+                torch.empty(self.experts_per_rank, 2 * inter, H, dtype=dtype),
+                requires_grad=False,
+            )
+            self.ep_down = nn.Parameter(
+                torch.empty(self.experts_per_rank, H, inter, dtype=dtype),
+                requires_grad=False,
+            )
+            self.gate_up = self.down = None
+        else:
+            self.gate_up = nn.ModuleList(
+                [
+                    MergedColumnParallelLinear(
+                        config.hidden_size, config.intermediate_size, dtype=dtype
+                    )
+                    for _ in range(self.num_experts)
+                ]
+            )
+            self.down = nn.ModuleList(
+                [
+                    RowParallelLinear(
+                        config.intermediate_size,
+                        config.hidden_size,
+                        dtype=dtype,
+                        reduce_output=False,  # one all-reduce for the block
+                    )
+                    for _ in range(self.num_experts)
+                ]
+            )
+
+    def _expert_fwd(self, local_e: int, xe: torch.Tensor) -> torch.Tensor:
+        if self.expert_parallel:
+            gu = torch.nn.functional.linear(xe, self.ep_gate_up[local_e])
+            act = ops.silu_and_mul(gu)
+            return torch.nn.functional.linear(act, self.ep_down[local_e])
+        return self.down[local_e](
+            ops.silu_and_mul(self.gate_up[local_e](xe))
         )
 
     def forward(self, x: torch.Tensor, meta=None) -> torch.Tensor:
@@ -199,13 +242,14 @@ class MixtralMoE(nn.Module):
         flat_i = topi.flatten()
         flat_w = topw.flatten()
         token_idx = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
-        for e in range(self.num_experts):
+        for le in range(self.experts_per_rank):
+            e = self.expert_lo + le
             sel = (flat_i == e).nonzero(as_tuple=True)[0]
             if sel.numel() == 0:
                 continue
             rows = token_idx[sel]
             xe = x.index_select(0, rows)
-            ye = self.down[e](ops.silu_and_mul(self.gate_up[e](xe)))
+            ye = self._expert_fwd(le, xe)
             out.index_add_(0, rows, ye * flat_w[sel].unsqueeze(1).to(ye.dtype))
         if comm.get_state().tp_size > 1:
             out = comm.tp_all_reduce(out)
@@ -217,7 +261,10 @@ class LlamaDecoderLayer(nn.Module):
         super().__init__()
         self.self_attn = LlamaAttention(config, dtype, layer_idx)
         self.mlp = (
-            MixtralMoE(config, dtype, layer_idx)
+            MixtralMoE(
+                config, dtype, layer_idx,
+                expert_parallel=getattr(config, "expert_parallel", False),
+            )
             if config.num_local_experts > 0
             else LlamaMLP(config, dtype, layer_idx)
         )
@@ -407,17 +454,27 @@ class LlamaForCausalLM(nn.Module):
             layer.self_attn.o_proj.load_shard(get(p + "self_attn.o_proj.weight"))
             if isinstance(layer.mlp, MixtralMoE):
                 moe = layer.mlp
+                owned = range(
+                    moe.expert_lo, moe.expert_lo + moe.experts_per_rank
+                )
                 if (p + "block_sparse_moe.gate.weight") in tensors:
                     # classic Mixtral checkpoint: experts.j.w1/w3/w2
                     moe.gate.data.copy_(
                         get(p + "block_sparse_moe.gate.weight").to(moe.gate.dtype)
                     )
-                    for j in range(moe.num_experts):
+                    for le, j in enumerate(owned):
                         ep = p + f"block_sparse_moe.experts.{j}."
-                        moe.gate_up[j].load_shards(
-                            get(ep + "w1.weight"), get(ep + "w3.weight")
-                        )
-                        moe.down[j].load_shard(get(ep + "w2.weight"))
+                        w1 = get(ep + "w1.weight")
+                        w3 = get(ep + "w3.weight")
+                        w2 = get(ep + "w2.weight")
+                        if moe.expert_parallel:
+                            moe.ep_gate_up.data[le].copy_(
+                                torch.cat([w1, w3], dim=0).to(moe.gate.dtype)
+                            )
+                            moe.ep_down.data[le].copy_(w2.to(moe.gate.dtype))
+                        else:
+                            moe.gate_up[le].load_shards(w1, w3)
+                            moe.down[le].load_shard(w2)
                 else:
                     # fused layout (transformers >= 4.5x): mlp.gate.weight,
                     # mlp.experts.gate_up_proj [E, 2I, H], .down_proj [E, H, I]
@@ -427,9 +484,17 @@ class LlamaForCausalLM(nn.Module):
                     gu = get(p + "mlp.experts.gate_up_proj")
                     dn = get(p + "mlp.experts.down_proj")
                     inter = gu.shape[1] // 2
-                    for j in range(moe.num_experts):
-                        moe.gate_up[j].load_shards(gu[j][:inter], gu[j][inter:])
-                        moe.down[j].load_shard(dn[j])
+                    for le, j in enumerate(owned):
+                        if moe.expert_parallel:
+                            moe.ep_gate_up.data[le].copy_(
+                                gu[j].to(moe.gate.dtype)
+                            )
+                            moe.ep_down.data[le].copy_(dn[j].to(moe.gate.dtype))
+                        else:
+                            moe.gate_up[le].load_shards(
+                                gu[j][:inter], gu[j][inter:]
+                            )
+                            moe.down[le].load_shard(dn[j])
             else:
                 layer.mlp.gate_up_proj.load_shards(
                     get(p + "mlp.gate_proj.weight"), get(p + "mlp.up_proj.weight")

@@ -69,10 +69,13 @@ def build_model(args):
         from transformers import AutoTokenizer
 
         tokenizer = AutoTokenizer.from_pretrained(args.model_dir)
+        model_cfg = ModelConfig.from_hf_config(
+            os.path.join(args.model_dir, "config.json")
+        )
+        if getattr(args, "expert_parallel", False):
+            model_cfg.expert_parallel = True
         cfg = EngineConfig(
-            model=ModelConfig.from_hf_config(
-                os.path.join(args.model_dir, "config.json")
-            ),
+            model=model_cfg,
             cache=CacheConfig(
                 cpu_offload_bytes=getattr(args, "kv_offload_bytes", 0),
                 kv_cache_dtype=getattr(args, "kv_cache_dtype", "auto"),

@@ -192,3 +192,71 @@ def test_pp2_engine_matches_single_process():
     # every stage's lockstep scheduler must report the same tokens, equal
     # to the single-process engine
     assert toks[0] == toks[1] == ref
+
+
+def _ep2_worker(rank, world, port, q):
+    """Expert-parallel Mixtral (tp group=2, 2 experts/rank) vs HF."""
+    try:
+        _set_env(rank, world, port)
+        import transformers
+
+        from kserve_amd.engine.config import ModelConfig
+        from kserve_amd.models.llama import AttentionMetadata, LlamaForCausalLM
+        from kserve_amd.parallel import comm
+
+        comm.init_distributed(tp_size=2, backend="gloo")
+        torch.manual_seed(21)
+        hf_cfg = transformers.MixtralConfig(
+            vocab_size=128,
+            hidden_size=64,
+            intermediate_size=128,
+            num_hidden_layers=2,
+            num_attention_heads=4,
+            num_key_value_heads=2,
+            num_local_experts=4,
+            num_experts_per_tok=2,
+            rms_norm_eps=1e-5,
+            rope_theta=10000.0,
+            sliding_window=None,
+            tie_word_embeddings=False,
+        )
+        hf = transformers.MixtralForCausalLM(hf_cfg).eval().float()
+        cfg = ModelConfig(
+            vocab_size=128, hidden_size=64, intermediate_size=128,
+            num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+            rms_norm_eps=1e-5, rope_theta=10000.0,
+            num_local_experts=4, num_experts_per_tok=2,
+            expert_parallel=True, model_name="mixtral-ep",
+        )
+        model = LlamaForCausalLM(cfg, dtype=torch.float32, device="cpu")
+        model.load_hf_state_dict(dict(hf.state_dict()))
+        moe = model.layers[0].mlp
+        assert moe.expert_parallel and moe.experts_per_rank == 2
+        assert moe.expert_lo == rank * 2
+        token_ids = list(range(9))
+        T = len(token_ids)
+        meta = AttentionMetadata(
+            is_prefill=True,
+            slot_mapping=torch.zeros(T, dtype=torch.int32),
+            cu_seqlens=torch.tensor([0, T], dtype=torch.int32),
+            max_seqlen=T,
+        )
+        caches = [(torch.empty(0), torch.empty(0))] * model.num_local_layers
+        hidden = model(torch.tensor(token_ids), torch.arange(T), caches, meta)
+        logits = model.compute_logits(hidden)
+        with torch.no_grad():
+            ref = hf(torch.tensor([token_ids]), use_cache=False).logits[0]
+        torch.testing.assert_close(logits, ref, atol=5e-4, rtol=5e-4)
+        comm.destroy_distributed()
+        q.put((rank, ("ok", None)))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, (f"FAIL: {e}\n{traceback.format_exc()}", None)))
+
+
+@pytest.mark.timeout(300)
+def test_ep2_mixtral_matches_hf():
+    """Expert parallelism: experts partitioned across ranks, outputs summed
+    by the block all-reduce (reference --enable-expert-parallel)."""
+    _run(_ep2_worker, 29627)
