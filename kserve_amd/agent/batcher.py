@@ -133,6 +133,34 @@ def create_batcher_proxy_app(
 
     batcher = Batcher(call_backend, max_batch_size, max_latency_ms)
     app.state.batcher = batcher
+    # Drainer (outermost handler of the reference agent chain,
+    # cmd/agent/main.go:429-449): readiness passes through to the user
+    # container until drain starts; on drain, in-flight requests finish
+    # while the probe reports not-ready so the endpoint is removed first.
+    app.state.draining = False
+    app.state.inflight = 0
+
+    @app.get("/readyz")
+    async def readyz():
+        if app.state.draining:
+            return Response(content="draining", status_code=503)
+        try:
+            r = await client.get("/")
+            return Response(status_code=200 if r.status_code < 500 else 503)
+        except Exception:
+            return Response(content="backend unreachable", status_code=503)
+
+    @app.post("/drain")
+    async def drain():
+        app.state.draining = True
+        import asyncio as _aio
+
+        # wait for in-flight work (bounded) before reporting drained
+        for _ in range(300):
+            if app.state.inflight == 0:
+                break
+            await _aio.sleep(0.1)
+        return {"draining": True, "inflight": app.state.inflight}
 
     @app.post("/v1/models/{name}:predict")
     async def predict(name: str, request: Request):
@@ -143,7 +171,11 @@ def create_batcher_proxy_app(
                 content=json.dumps({"error": "instances must be a list"}),
                 status_code=400,
             )
-        result = await batcher.predict(instances)
+        app.state.inflight += 1
+        try:
+            result = await batcher.predict(instances)
+        finally:
+            app.state.inflight -= 1
         return result
 
     @app.api_route(

@@ -235,3 +235,34 @@ class TestPayloadLogger:
             assert pl.should_log("org.kubeflow.serving.inference.response")
 
         run(main())
+
+
+class TestAgentDrainer:
+    def test_readyz_and_drain(self):
+        import httpx
+        from fastapi.testclient import TestClient
+
+        from kserve_amd.agent.batcher import create_batcher_proxy_app
+
+        def backend(request: httpx.Request) -> httpx.Response:
+            if request.url.path == "/":
+                return httpx.Response(200, json={"status": "alive"})
+            body = json.loads(request.content)
+            return httpx.Response(
+                200, json={"predictions": [x for x in body["instances"]]}
+            )
+
+        app = create_batcher_proxy_app(
+            "http://backend", "m",
+            transport=httpx.MockTransport(backend),
+        )
+        with TestClient(app) as c:
+            assert c.get("/readyz").status_code == 200
+            r = c.post(
+                "/v1/models/m:predict", json={"instances": [[1, 2]]}
+            )
+            assert r.status_code == 200
+            # drain: probe flips to 503, in-flight drained
+            d = c.post("/drain")
+            assert d.json()["draining"] is True
+            assert c.get("/readyz").status_code == 503
