@@ -118,3 +118,34 @@ def test_spec_with_preemption():
     tiny = LLMEngine(cfg)
     b = [o.output_token_ids for o in tiny.generate(PROMPTS, sp).values()]
     assert a == b
+
+
+def test_all_features_combined_matches_plain():
+    """Spec decode + prefix caching + chunked prefill + tiny KV pool
+    (preemption) all on together must still match plain greedy outputs."""
+    torch.manual_seed(0)
+    plain = make_engine(spec=0)
+    sp = SamplingParams(temperature=0.0, max_tokens=16)
+    shared = [5, 6, 7, 8] * 4  # 16-token shared, repetitive (spec accepts)
+    prompts = [shared + [40 + i] for i in range(4)] + [PROMPTS[0]]
+    a = [o.output_token_ids for o in plain.generate(prompts, sp).values()]
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=128),
+        cache=CacheConfig(
+            block_size=4, num_gpu_blocks=96, enable_prefix_caching=True
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8,
+            max_num_batched_tokens=12,  # forces chunking
+            max_model_len=256,
+            speculative_ngram=4,
+            enable_chunked_prefill=True,
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    full = LLMEngine(cfg)
+    b0 = [o.output_token_ids for o in full.generate(prompts, sp).values()]
+    b1 = [o.output_token_ids for o in full.generate(prompts, sp).values()]
+    assert a == b0 == b1
