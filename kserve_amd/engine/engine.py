@@ -314,7 +314,8 @@ class LLMEngine:
                     for i, r in enumerate(batch.requests)
                     if r.num_computed_tokens >= r.num_tokens
                 ]
-                logits = logits[idx]
+                if logits is not None:  # None on non-final pipeline stages
+                    logits = logits[idx]
                 # partial chunks go back to the head of the waiting queue so
                 # the next step continues their prefill (not decode)
                 for r in reversed(batch.requests):

@@ -80,7 +80,13 @@ def _reference_outputs():
     engine = _build_engine()
     sp = SamplingParams(temperature=0.0, max_tokens=8)
     out = engine.generate([[1, 2, 3, 4, 5], [9, 8, 7]], sp)
-    return [o.output_token_ids for o in out.values()]
+    toks = [o.output_token_ids for o in out.values()]
+    # mirror the workers' second (chunked-prefill) run
+    engine.config.scheduler.enable_chunked_prefill = True
+    engine.scheduler.config.max_num_batched_tokens = 4
+    out2 = engine.generate([list(range(1, 12))], sp)
+    toks.append(list(out2.values())[0].output_token_ids)
+    return toks
 
 
 def _pp2_engine_worker(rank, world, port, q):
@@ -101,6 +107,12 @@ def _pp2_engine_worker(rank, world, port, q):
         sp = SamplingParams(temperature=0.0, max_tokens=8)
         out = engine.generate([[1, 2, 3, 4, 5], [9, 8, 7]], sp)
         toks = [o.output_token_ids for o in out.values()]
+        # chunked prefill under PP: partial chunks sample nothing on any
+        # stage; logits slicing must tolerate None on non-final stages
+        engine.config.scheduler.enable_chunked_prefill = True
+        engine.scheduler.config.max_num_batched_tokens = 4
+        out2 = engine.generate([list(range(1, 12))], sp)
+        toks.append(list(out2.values())[0].output_token_ids)
         comm.destroy_distributed()
         q.put((rank, ("ok", toks)))
     except Exception as e:  # pragma: no cover
