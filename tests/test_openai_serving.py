@@ -307,3 +307,38 @@ class TestOpenAIProxyModel:
             )
             assert total > 0
             up_model.stop()
+
+
+def test_stream_disconnect_aborts_request(client):
+    """Closing a stream mid-generation must abort the engine request
+    (reference vLLM with_cancellation semantics)."""
+    import time
+
+    model = client.app.state.llm_model
+    eng = model.async_engine
+
+    with client.stream(
+        "POST",
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [1, 2, 3],
+            "max_tokens": 64,
+            "temperature": 0.0,
+            "stream": True,
+        },
+    ) as r:
+        # read a couple of SSE lines then drop the connection
+        it = r.iter_lines()
+        got = 0
+        for line in it:
+            if line.startswith("data: "):
+                got += 1
+            if got >= 2:
+                break
+    # the abort sentinel is processed by the engine thread between steps
+    for _ in range(100):
+        if eng.engine is not None and not eng.engine.has_unfinished():
+            break
+        time.sleep(0.05)
+    assert not eng.engine.has_unfinished(), "request not aborted after disconnect"
