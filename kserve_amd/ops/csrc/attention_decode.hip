@@ -917,6 +917,33 @@ __global__ void paged_attention_reduce_kernel(
 
 }  // namespace
 
+
+namespace {
+// shared split-context reduction launch (used by every kernel variant)
+inline hipError_t launch_split_reduce(void* out, void* part_out,
+                                      void* part_ml, int num_seqs,
+                                      int num_heads, int head_dim,
+                                      int n_splits, hipStream_t stream) {
+  if (n_splits <= 1) return hipSuccess;
+  const long sh = (long)num_seqs * num_heads;
+  const int wpb = 4;
+  dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
+  if (head_dim == 128) {
+    hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
+                       dim3(wpb * 64), 0, stream, (short*)out,
+                       (const float*)part_out, (const float*)part_ml, sh,
+                       n_splits);
+  } else {
+    hipLaunchKernelGGL((paged_attention_reduce_kernel<64>), rgrid,
+                       dim3(wpb * 64), 0, stream, (short*)out,
+                       (const float*)part_out, (const float*)part_ml, sh,
+                       n_splits);
+  }
+  HIP_CHECK_KERNEL();
+  return hipSuccess;
+}
+}  // namespace
+
 extern "C" hipError_t ks_paged_attention_decode_fp8(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const void* block_tables, const void* context_lens, float scale,
@@ -936,17 +963,8 @@ extern "C" hipError_t ks_paged_attention_decode_fp8(
                      scale, num_kv_heads, group, max_blocks, q_row_stride,
                      n_splits, (float*)part_out, (float*)part_ml);
   HIP_CHECK_KERNEL();
-  if (n_splits > 1) {
-    const long sh = (long)num_seqs * num_heads;
-    int wpb = 4;
-    dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-    hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                       dim3(wpb * 64), 0, stream, (short*)out,
-                       (const float*)part_out, (const float*)part_ml, sh,
-                       n_splits);
-    HIP_CHECK_KERNEL();
-  }
-  return hipSuccess;
+  return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                             head_dim, n_splits, stream);
 }
 
 extern "C" hipError_t ks_paged_attention_decode(
@@ -987,17 +1005,8 @@ extern "C" hipError_t ks_paged_attention_decode(
     else LAUNCH_WS(8);
 #undef LAUNCH_WS
     HIP_CHECK_KERNEL();
-    if (n_splits > 1) {
-      const long sh = (long)num_seqs * num_heads;
-      int wpb = 4;
-      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(wpb * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-      HIP_CHECK_KERNEL();
-    }
-    return hipSuccess;
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
   }
 
   // 2-wave x 2-head mode (A/B: KS_ATTN_H2=1): halves the per-CU L1
@@ -1014,17 +1023,8 @@ extern "C" hipError_t ks_paged_attention_decode(
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
                        n_splits, (float*)part_out, (float*)part_ml);
     HIP_CHECK_KERNEL();
-    if (n_splits > 1) {
-      const long sh = (long)num_seqs * num_heads;
-      int wpb = 4;
-      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(wpb * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-      HIP_CHECK_KERNEL();
-    }
-    return hipSuccess;
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
   }
 
   // 2-page unrolled fast path (A/B: KS_ATTN_U2=0 disables)
@@ -1040,17 +1040,8 @@ extern "C" hipError_t ks_paged_attention_decode(
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
                        n_splits, (float*)part_out, (float*)part_ml);
     HIP_CHECK_KERNEL();
-    if (n_splits > 1) {
-      const long sh = (long)num_seqs * num_heads;
-      int wpb = 4;
-      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(wpb * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-      HIP_CHECK_KERNEL();
-    }
-    return hipSuccess;
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
   }
   // occupancy ablation: KS_ATTN_OCC=5/6 forces a tighter VGPR budget so
   // more waves are resident to hide the scattered-page load latency
@@ -1101,17 +1092,8 @@ extern "C" hipError_t ks_paged_attention_decode(
           (float*)part_out, (float*)part_ml);
     }
     HIP_CHECK_KERNEL();
-    if (n_splits > 1) {
-      const long sh = (long)num_seqs * num_heads;
-      int wpb = 4;
-      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(wpb * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-      HIP_CHECK_KERNEL();
-    }
-    return hipSuccess;
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
   }
   static const int d2_mode = [] {
     const char* e = getenv("KS_ATTN_D2");
@@ -1127,17 +1109,8 @@ extern "C" hipError_t ks_paged_attention_decode(
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
                        n_splits, (float*)part_out, (float*)part_ml);
     HIP_CHECK_KERNEL();
-    if (n_splits > 1) {
-      const long sh = (long)num_seqs * num_heads;
-      int wpb = 4;
-      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(wpb * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-      HIP_CHECK_KERNEL();
-    }
-    return hipSuccess;
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
   }
   if (use_pb && hpw == 1 && head_dim == 128) {
     hipLaunchKernelGGL((paged_attention_kernel<128, 1, 1, true>), grid, block,
@@ -1147,17 +1120,8 @@ extern "C" hipError_t ks_paged_attention_decode(
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
                        n_splits, (float*)part_out, (float*)part_ml);
     HIP_CHECK_KERNEL();
-    if (n_splits > 1) {
-      const long sh = (long)num_seqs * num_heads;
-      int wpb = 4;
-      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(wpb * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-      HIP_CHECK_KERNEL();
-    }
-    return hipSuccess;
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
   }
   if (occ >= 5 && hpw == 1 && head_dim == 128) {
     if (occ >= 6) {
@@ -1176,17 +1140,8 @@ extern "C" hipError_t ks_paged_attention_decode(
                          n_splits, (float*)part_out, (float*)part_ml);
     }
     HIP_CHECK_KERNEL();
-    if (n_splits > 1) {
-      const long sh = (long)num_seqs * num_heads;
-      int wpb = 4;
-      dim3 rgrid((unsigned)((sh + wpb - 1) / wpb));
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(wpb * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-      HIP_CHECK_KERNEL();
-    }
-    return hipSuccess;
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
   }
 #define LAUNCH_PA(DD, HH)                                                  \
   hipLaunchKernelGGL((paged_attention_kernel<DD, HH>), grid, block, 0,     \
@@ -1206,22 +1161,6 @@ extern "C" hipError_t ks_paged_attention_decode(
   }
 #undef LAUNCH_PA
   HIP_CHECK_KERNEL();
-  if (n_splits > 1) {
-    const long sh = (long)num_seqs * num_heads;
-    int waves_per_block = 4;
-    dim3 rgrid((unsigned)((sh + waves_per_block - 1) / waves_per_block));
-    if (head_dim == 128) {
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<128>), rgrid,
-                         dim3(waves_per_block * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-    } else {
-      hipLaunchKernelGGL((paged_attention_reduce_kernel<64>), rgrid,
-                         dim3(waves_per_block * 64), 0, stream, (short*)out,
-                         (const float*)part_out, (const float*)part_ml, sh,
-                         n_splits);
-    }
-    HIP_CHECK_KERNEL();
-  }
-  return hipSuccess;
+  return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                             head_dim, n_splits, stream);
 }
