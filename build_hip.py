@@ -31,6 +31,7 @@ KERNEL_SOURCES = [
     "attention_prefill.hip",
     "sampling.hip",
     "skinny_gemm.hip",
+    "gemm8.hip",
     "mfma_probe.hip",
 ]
 

@@ -53,6 +53,8 @@ hipError_t ks_topk_topp_sample(void*, const void*, const void*, const void*,
 hipError_t ks_mfma_probe(void*, const void*, const void*, hipStream_t);
 hipError_t ks_skinny_gemm(void*, void*, const void*, const void*, int, int,
                           int, long, hipStream_t);
+hipError_t ks_gemm8(void*, const void*, const void*, int, int, int, int,
+                    hipStream_t);
 }
 
 namespace {
@@ -378,6 +380,19 @@ void skinny_gemm(at::Tensor& out, at::Tensor& x, at::Tensor& w) {
             "skinny_gemm");
 }
 
+void gemm8(at::Tensor& d, at::Tensor& a, at::Tensor& w, bool swizzle) {
+  // EXPERIMENTAL (see gemm8.hip header): not used by ops.linear dispatch
+  CHECK_BF16_CONTIG(d);
+  CHECK_BF16_CONTIG(a);
+  CHECK_BF16_CONTIG(w);
+  int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && d.size(0) == M && d.size(1) == N,
+              "shape mismatch");
+  check_hip(ks_gemm8(d.data_ptr(), a.data_ptr(), w.data_ptr(), M, N, K,
+                     swizzle ? 1 : 0, current_stream()),
+            "gemm8");
+}
+
 void mfma_probe(at::Tensor& c, at::Tensor& a, at::Tensor& b) {
   check_hip(ks_mfma_probe(c.data_ptr(), a.data_ptr(), b.data_ptr(),
                           current_stream()),
@@ -412,5 +427,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_topp_sample", &topk_topp_sample,
         "fused top-k/top-p + Gumbel-max sampler (radix-histogram select)");
   m.def("skinny_gemm", &skinny_gemm, "decode GEMM (N<=256, MFMA streaming)");
+  m.def("gemm8", &gemm8,
+        "EXPERIMENTAL 8-phase 256x256 MFMA GEMM (round-2 candidate)");
   m.def("mfma_probe", &mfma_probe, "MFMA layout probe (tests)");
 }
