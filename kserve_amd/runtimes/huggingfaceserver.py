@@ -74,6 +74,12 @@ def build_model(args):
         )
         if getattr(args, "expert_parallel", False):
             model_cfg.expert_parallel = True
+        if model_cfg.sliding_window:
+            logger.info(
+                "model declares sliding_window=%d; this engine attends the "
+                "full context (a superset of windowed attention)",
+                model_cfg.sliding_window,
+            )
         cfg = EngineConfig(
             model=model_cfg,
             cache=CacheConfig(
