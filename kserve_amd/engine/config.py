@@ -47,6 +47,8 @@ class ModelConfig:
     # EP: partition experts across the parallel group (full-width weights)
     # instead of TP-sharding every expert
     expert_parallel: bool = False
+    # HF rope_scaling dict (Llama-3.1 'llama3' type supported)
+    rope_scaling: Optional[dict] = None
     # Mistral-style sliding-window size from the HF config. This engine
     # runs FULL attention regardless (strictly more context than the
     # windowed reference behavior); recorded so servers can warn when a
@@ -129,6 +131,7 @@ class ModelConfig:
             num_local_experts=cfg.get("num_local_experts", 0),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             sliding_window=cfg.get("sliding_window"),
+            rope_scaling=cfg.get("rope_scaling"),
             mlp_bias=cfg.get("mlp_bias", False),
             model_name=cfg.get("_name_or_path", os.path.dirname(config_path) or "model"),
         )

@@ -349,6 +349,7 @@ class LlamaForCausalLM(nn.Module):
                 config.max_position_embeddings,
                 config.rope_theta,
                 dtype=torch.float32,
+                rope_scaling=getattr(config, "rope_scaling", None),
             ),
             persistent=False,
         )

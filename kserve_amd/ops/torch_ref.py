@@ -37,18 +37,48 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(gate.float()) * up.float()).to(x.dtype)
 
 
+def _llama3_scale_inv_freq(inv_freq, rope_scaling: dict):
+    """Llama-3.1 rope scaling (HF 'rope_type: llama3'): low-frequency
+    components are slowed by `factor`, high-frequency ones kept, with a
+    smooth interpolation between the two wavelength thresholds."""
+    import math
+
+    factor = rope_scaling.get("factor", 8.0)
+    lo = rope_scaling.get("low_freq_factor", 1.0)
+    hi = rope_scaling.get("high_freq_factor", 4.0)
+    orig = rope_scaling.get("original_max_position_embeddings", 8192)
+    low_wl = orig / lo
+    high_wl = orig / hi
+    out = inv_freq.clone()
+    for i, f in enumerate(inv_freq.tolist()):
+        wl = 2 * math.pi / f
+        if wl < high_wl:
+            continue  # high frequency: unscaled
+        if wl > low_wl:
+            out[i] = f / factor  # low frequency: fully scaled
+        else:
+            smooth = (orig / wl - lo) / (hi - lo)
+            out[i] = (1 - smooth) * f / factor + smooth * f
+    return out
+
+
 def make_cos_sin_cache(
     head_dim: int,
     max_positions: int,
     theta: float = 10000.0,
     dtype: torch.dtype = torch.float32,
     device="cpu",
+    rope_scaling: Optional[dict] = None,
 ) -> torch.Tensor:
     """[max_positions, head_dim] with cos in the first half, sin in the second
-    (host-precomputed per CDNA guide Appendix B: no on-device trig)."""
+    (host-precomputed per CDNA guide Appendix B: no on-device trig).
+    rope_scaling: HF dict; 'llama3' rope_type applies 3.1-style frequency
+    scaling."""
     inv_freq = 1.0 / (
         theta ** (torch.arange(0, head_dim, 2, dtype=torch.float64) / head_dim)
     )
+    if rope_scaling and rope_scaling.get("rope_type", rope_scaling.get("type")) == "llama3":
+        inv_freq = _llama3_scale_inv_freq(inv_freq, rope_scaling)
     t = torch.arange(max_positions, dtype=torch.float64)
     freqs = torch.outer(t, inv_freq)  # [P, head_dim/2]
     cache = torch.cat([freqs.cos(), freqs.sin()], dim=-1).to(dtype).to(device)

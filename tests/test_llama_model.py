@@ -305,3 +305,49 @@ class TestMixtralVsTransformers:
         single = engine.generate([[1, 2, 3]], sp)
         batched_first = [o for o in out.values()][0].output_token_ids
         assert list(single.values())[0].output_token_ids == batched_first
+
+
+class TestLlama31RopeScaling:
+    def test_rope_scaled_logits_match_hf(self):
+        """Llama-3.1 'llama3' rope scaling vs HF transformers fp32."""
+        transformers = pytest.importorskip("transformers")
+        from kserve_amd.engine.config import ModelConfig
+
+        scaling = {
+            "rope_type": "llama3",
+            "factor": 8.0,
+            "low_freq_factor": 1.0,
+            "high_freq_factor": 4.0,
+            "original_max_position_embeddings": 64,
+        }
+        torch.manual_seed(23)
+        hf_cfg = transformers.LlamaConfig(
+            vocab_size=256,
+            hidden_size=128,
+            intermediate_size=256,
+            num_hidden_layers=2,
+            num_attention_heads=2,
+            num_key_value_heads=1,
+            rms_norm_eps=1e-5,
+            rope_theta=500000.0,
+            max_position_embeddings=256,
+            rope_scaling=dict(scaling),
+            tie_word_embeddings=False,
+        )
+        hf = transformers.LlamaForCausalLM(hf_cfg).eval().float()
+        cfg = ModelConfig(
+            vocab_size=256, hidden_size=128, intermediate_size=256,
+            num_layers=2, num_heads=2, num_kv_heads=1, head_dim=64,
+            rms_norm_eps=1e-5, rope_theta=500000.0,
+            max_position_embeddings=256, rope_scaling=dict(scaling),
+            model_name="llama31-tiny",
+        )
+        ours = LlamaForCausalLM(cfg, dtype=torch.float32, device="cpu")
+        ours.load_hf_state_dict(dict(hf.state_dict()))
+        # positions beyond original_max_position_embeddings exercise the
+        # scaled low-frequency components
+        token_ids = list(torch.randint(0, 256, (100,)).tolist())
+        with torch.no_grad():
+            hf_logits = hf(torch.tensor([token_ids]), use_cache=False).logits[0]
+        our_logits = full_forward_logits(ours, token_ids)
+        torch.testing.assert_close(our_logits, hf_logits, rtol=3e-4, atol=3e-4)
