@@ -77,8 +77,12 @@ def make_cos_sin_cache(
     inv_freq = 1.0 / (
         theta ** (torch.arange(0, head_dim, 2, dtype=torch.float64) / head_dim)
     )
-    if rope_scaling and rope_scaling.get("rope_type", rope_scaling.get("type")) == "llama3":
-        inv_freq = _llama3_scale_inv_freq(inv_freq, rope_scaling)
+    if rope_scaling:
+        rtype = rope_scaling.get("rope_type", rope_scaling.get("type"))
+        if rtype == "llama3":
+            inv_freq = _llama3_scale_inv_freq(inv_freq, rope_scaling)
+        elif rtype == "linear":
+            inv_freq = inv_freq / rope_scaling.get("factor", 1.0)
     t = torch.arange(max_positions, dtype=torch.float64)
     freqs = torch.outer(t, inv_freq)  # [P, head_dim/2]
     cache = torch.cat([freqs.cos(), freqs.sin()], dim=-1).to(dtype).to(device)

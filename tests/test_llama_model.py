@@ -351,3 +351,33 @@ class TestLlama31RopeScaling:
             hf_logits = hf(torch.tensor([token_ids]), use_cache=False).logits[0]
         our_logits = full_forward_logits(ours, token_ids)
         torch.testing.assert_close(our_logits, hf_logits, rtol=3e-4, atol=3e-4)
+
+    def test_linear_rope_scaling_matches_hf(self):
+        transformers = pytest.importorskip("transformers")
+        from kserve_amd.engine.config import ModelConfig
+
+        scaling = {"rope_type": "linear", "factor": 4.0}
+        torch.manual_seed(29)
+        hf_cfg = transformers.LlamaConfig(
+            vocab_size=128, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=1, num_attention_heads=2,
+            num_key_value_heads=1, rope_theta=10000.0,
+            max_position_embeddings=128, rope_scaling=dict(scaling),
+            tie_word_embeddings=False,
+        )
+        hf = transformers.LlamaForCausalLM(hf_cfg).eval().float()
+        cfg = ModelConfig(
+            vocab_size=128, hidden_size=64, intermediate_size=128,
+            num_layers=1, num_heads=2, num_kv_heads=1, head_dim=32,
+            rope_theta=10000.0, max_position_embeddings=128,
+            rope_scaling=dict(scaling), model_name="lin-tiny",
+        )
+        ours = LlamaForCausalLM(cfg, dtype=torch.float32, device="cpu")
+        ours.load_hf_state_dict(dict(hf.state_dict()))
+        token_ids = list(torch.randint(0, 128, (40,)).tolist())
+        with torch.no_grad():
+            hf_logits = hf(torch.tensor([token_ids]), use_cache=False).logits[0]
+        torch.testing.assert_close(
+            full_forward_logits(ours, token_ids), hf_logits,
+            rtol=3e-4, atol=3e-4,
+        )
