@@ -361,7 +361,7 @@ class TestLlama31RopeScaling:
         hf_cfg = transformers.LlamaConfig(
             vocab_size=128, hidden_size=64, intermediate_size=128,
             num_hidden_layers=1, num_attention_heads=2,
-            num_key_value_heads=1, rope_theta=10000.0,
+            num_key_value_heads=1, rope_theta=10000.0, rms_norm_eps=1e-5,
             max_position_embeddings=128, rope_scaling=dict(scaling),
             tie_word_embeddings=False,
         )
