@@ -197,3 +197,61 @@ def test_fp8_kv_cache_cpu():
     assert len(b.output_token_ids) == 8
     # prefill (first token) ignores the cache entirely -> identical
     assert a.output_token_ids[0] == b.output_token_ids[0]
+
+
+@pytest.mark.timeout(180)
+def test_engine_fuzz_mixed_workload():
+    """Mini-soak: 150 requests with randomized lengths/sampling params under
+    a small KV pool (preemption active). Everything must finish, with the
+    right lengths, and the pool must drain back to empty."""
+    import random
+
+    rng = random.Random(1234)
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=128),
+        cache=CacheConfig(
+            block_size=4, num_gpu_blocks=64, enable_prefix_caching=True
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=6,
+            max_num_batched_tokens=64,
+            max_model_len=96,
+            enable_chunked_prefill=True,
+            speculative_ngram=3,
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    engine = LLMEngine(cfg)
+    expected = {}
+    shared = [7, 8, 9, 10, 11, 12, 13, 14]
+    for i in range(150):
+        plen = rng.randint(1, 40)
+        prompt = (shared if rng.random() < 0.4 else []) + [
+            rng.randint(0, 127) for _ in range(plen)
+        ]
+        prompt = prompt[:40]
+        max_toks = rng.randint(1, 12)
+        sp = SamplingParams(
+            temperature=rng.choice([0.0, 0.0, 0.8]),
+            top_k=rng.choice([-1, 5]),
+            top_p=rng.choice([1.0, 0.9]),
+            max_tokens=max_toks,
+            seed=i,
+        )
+        rid = engine.add_request(prompt, sp, request_id=f"f{i}")
+        expected[rid] = max_toks
+    done = {}
+    steps = 0
+    while engine.has_unfinished():
+        for out in engine.step():
+            if out.finished:
+                done[out.request_id] = len(out.output_token_ids)
+        steps += 1
+        assert steps < 5000, "engine wedged"
+    assert len(done) == 150
+    for rid, n in done.items():
+        assert n == expected[rid], (rid, n, expected[rid])
+    bm = engine.scheduler.block_manager
+    assert bm.num_free_blocks == bm.num_blocks - 1  # all blocks returned
