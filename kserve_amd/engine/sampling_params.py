@@ -29,6 +29,8 @@ class SamplingParams:
     echo: bool = False
     # multi-LoRA: adapter name registered with the engine (None = base)
     lora_name: Optional[str] = None
+    # scheduling priority (lower value runs first; same-priority FIFO)
+    priority: int = 0
 
     def __post_init__(self):
         if self.temperature < 0:

@@ -58,18 +58,36 @@ class Scheduler:
             num_cpu_blocks=num_cpu_blocks or cache_config.num_cpu_blocks,
             enable_prefix_caching=cache_config.enable_prefix_caching,
         )
-        self.waiting: Deque[Request] = deque()
+        # waiting queue ordered by (priority, arrival seq): lower priority
+        # value first, FIFO within a priority class. A preempted or
+        # partially-prefilled request re-enters at its ORIGINAL seq, so it
+        # returns to the head of its class.
+        self.waiting: List[Request] = []
+        self._arrival_seq = 0
         self.running: List[Request] = []
         # requests whose KV lives in the host-DRAM tier
         self.swapped: List[Request] = []
 
     # -- queue ops -----------------------------------------------------------
+    def _queue_key(self, request: Request):
+        return (request.sampling_params.priority, request.arrival_seq)
+
+    def _enqueue(self, request: Request) -> None:
+        key = self._queue_key(request)
+        for i, r in enumerate(self.waiting):
+            if self._queue_key(r) > key:
+                self.waiting.insert(i, request)
+                return
+        self.waiting.append(request)
+
     def add_request(self, request: Request) -> None:
         if request.num_prompt_tokens > self.config.max_model_len:
             request.status = RequestStatus.FINISHED_LENGTH
             request.is_finished = True
             return
-        self.waiting.append(request)
+        request.arrival_seq = self._arrival_seq
+        self._arrival_seq += 1
+        self._enqueue(request)
 
     def abort_request(self, request_id: str) -> Optional[Request]:
         for q in (self.waiting, self.running, self.swapped):
@@ -150,7 +168,7 @@ class Scheduler:
                         len(req.block_table) * self.block_manager.block_size
                     )
                     n_new = min(req.num_tokens - cached, capacity - cached)
-            self.waiting.popleft()
+            self.waiting.pop(0)
             req.status = RequestStatus.RUNNING
             batch.requests.append(req)
             batch.num_scheduled_tokens.append(n_new)
@@ -216,7 +234,7 @@ class Scheduler:
         self.block_manager.free(req)
         req.num_computed_tokens = 0
         req.status = RequestStatus.PREEMPTED
-        self.waiting.appendleft(req)
+        self._enqueue(req)
 
     def requeue_partial_prefill(self, req: Request) -> None:
         """A chunked prefill finished its chunk but not the prompt: return it
@@ -224,7 +242,7 @@ class Scheduler:
         schedules the following chunk instead of decoding it."""
         self.running.remove(req)
         req.status = RequestStatus.WAITING
-        self.waiting.appendleft(req)
+        self._enqueue(req)
 
     def _preempt_swap(self, req: Request):
         """Offload preemption: KV pages move to pinned host DRAM; the request

@@ -49,6 +49,8 @@ class CompletionRequest(BaseModel):
     top_k: Optional[int] = -1
     min_tokens: Optional[int] = 0
     ignore_eos: Optional[bool] = False
+    # scheduling priority (vLLM extension: lower value runs first)
+    priority: Optional[int] = 0
     user: Optional[str] = None
 
 

@@ -142,6 +142,7 @@ class LLMModel(OpenAIModel):
             raise InvalidInput("Empty prompt")
         sp = _to_sampling_params(request)
         sp.lora_name = self._lora_for(request.model)
+        sp.priority = getattr(request, "priority", 0) or 0
         if request.stream:
             if len(prompt_list) != 1:
                 raise InvalidInput("Streaming supports a single prompt")

@@ -125,3 +125,41 @@ def test_preempted_request_refills_all_tokens():
     batch = s._schedule_prefill()
     assert batch.requests == [r]
     assert batch.num_scheduled_tokens == [10]
+
+
+def test_priority_scheduling():
+    """Lower priority value runs first; FIFO within a class; a preempted
+    high-priority request returns to the head of its class."""
+    from kserve_amd.engine.config import CacheConfig, SchedulerConfig
+    from kserve_amd.engine.request import Request
+    from kserve_amd.engine.sampling_params import SamplingParams
+    from kserve_amd.engine.scheduler import Scheduler
+
+    sched = Scheduler(
+        SchedulerConfig(max_num_seqs=1, max_num_batched_tokens=64,
+                        max_model_len=64),
+        CacheConfig(block_size=4),
+        num_gpu_blocks=64,
+    )
+
+    def add(rid, prio):
+        r = Request(rid, [1, 2, 3],
+                    SamplingParams(max_tokens=4, priority=prio),
+                    eos_token_id=-1)
+        sched.add_request(r)
+        return r
+
+    add("low-a", 5)
+    add("normal", 0)
+    add("low-b", 5)
+    urgent = add("urgent", -1)
+    order = [r.request_id for r in sched.waiting]
+    assert order == ["urgent", "normal", "low-a", "low-b"]
+
+    # max_num_seqs=1: only the urgent request is scheduled
+    batch = sched.schedule()
+    assert [r.request_id for r in batch.requests] == ["urgent"]
+    # preempt it: it must come back ahead of everything in its class
+    sched._preempt(urgent)
+    sched.running.remove(urgent)
+    assert sched.waiting[0].request_id == "urgent"
