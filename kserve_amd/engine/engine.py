@@ -111,8 +111,16 @@ class Sampler:
             dtype=torch.int64,
             device=logits.device,
         )
+        min_p = None
+        if any(r.sampling_params.min_p > 0 for r in requests):
+            min_p = torch.tensor(
+                [r.sampling_params.min_p for r in requests],
+                dtype=torch.float32,
+                device=logits.device,
+            )
         out = ops.random_sample(
-            logits, temps, top_p, top_k, seeds=seeds, generator=self.generator
+            logits, temps, top_p, top_k, seeds=seeds,
+            generator=self.generator, min_p=min_p,
         )
         # greedy requests in a mixed batch: override with argmax
         if any(r.sampling_params.greedy for r in requests):

@@ -16,6 +16,8 @@ class SamplingParams:
     temperature: float = 1.0
     top_p: float = 1.0
     top_k: int = -1
+    # keep tokens with prob >= min_p * max_prob (0 disables)
+    min_p: float = 0.0
     max_tokens: int = 16
     min_tokens: int = 0
     stop: List[str] = field(default_factory=list)
@@ -39,6 +41,8 @@ class SamplingParams:
             raise ValueError("top_p must be in (0, 1]")
         if self.top_k == 0 or self.top_k < -1:
             raise ValueError("top_k must be -1 (disabled) or >= 1")
+        if not 0.0 <= self.min_p <= 1.0:
+            raise ValueError("min_p must be in [0, 1]")
         if self.max_tokens is not None and self.max_tokens < 1:
             raise ValueError("max_tokens must be >= 1")
         if self.n < 1:

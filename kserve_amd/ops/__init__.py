@@ -170,6 +170,7 @@ def random_sample(
     top_k: torch.Tensor,
     seeds: Optional[torch.Tensor] = None,
     generator: Optional[torch.Generator] = None,
+    min_p: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Temperature/top-k/top-p sampling.
 
@@ -177,7 +178,8 @@ def random_sample(
     top-k/top-p use a bf16 radix-histogram select (exact k-th value / mass
     threshold, no sort) followed by Gumbel-max over the surviving set.
     """
-    if logits.is_cuda and seeds is not None:
+    has_min_p = min_p is not None and bool((min_p > 0).any())
+    if logits.is_cuda and seeds is not None and not has_min_p:
         out = torch.empty(
             logits.shape[0], dtype=torch.int64, device=logits.device
         )
@@ -191,12 +193,9 @@ def random_sample(
                 out, logits, temperatures, top_k.to(torch.int32), seeds
             )
         return out
-    if logits.is_cuda:
-        return torch_ref.random_sample(
-            logits, temperatures, top_p, top_k, generator=generator
-        )
+    # min-p (or CPU): fp32 sort path
     return torch_ref.random_sample(
-        logits, temperatures, top_p, top_k, generator=generator
+        logits, temperatures, top_p, top_k, generator=generator, min_p=min_p
     )
 
 

@@ -245,8 +245,9 @@ def random_sample(
     top_p: torch.Tensor,  # [num_seqs]
     top_k: torch.Tensor,  # [num_seqs] (-1 disables)
     generator: Optional[torch.Generator] = None,
+    min_p: Optional[torch.Tensor] = None,  # [num_seqs] (0 disables)
 ) -> torch.Tensor:
-    """Temperature + top-k + top-p sampling, fp32 reference."""
+    """Temperature + top-k + top-p + min-p sampling, fp32 reference."""
     lf = logits.float()
     temps = temperatures.clamp_min(1e-5).unsqueeze(-1)
     lf = lf / temps
@@ -260,6 +261,9 @@ def random_sample(
     probs_sorted = torch.softmax(sorted_logits, dim=-1)
     cumprobs = probs_sorted.cumsum(dim=-1)
     mask |= (cumprobs - probs_sorted) > top_p.unsqueeze(-1)
+    if min_p is not None:
+        # min-p: keep tokens with prob >= min_p * p_max
+        mask |= probs_sorted < (min_p.unsqueeze(-1) * probs_sorted[:, :1])
     sorted_logits = sorted_logits.masked_fill(mask, float("-inf"))
     probs = torch.softmax(sorted_logits, dim=-1)
     choice = torch.multinomial(probs, 1, generator=generator).squeeze(-1)

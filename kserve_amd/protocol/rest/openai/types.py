@@ -47,6 +47,7 @@ class CompletionRequest(BaseModel):
     temperature: Optional[float] = 1.0
     top_p: Optional[float] = 1.0
     top_k: Optional[int] = -1
+    min_p: Optional[float] = 0.0
     min_tokens: Optional[int] = 0
     ignore_eos: Optional[bool] = False
     # scheduling priority (vLLM extension: lower value runs first)
@@ -112,6 +113,7 @@ class ChatCompletionRequest(BaseModel):
     temperature: Optional[float] = 1.0
     top_p: Optional[float] = 1.0
     top_k: Optional[int] = -1
+    min_p: Optional[float] = 0.0
     min_tokens: Optional[int] = 0
     ignore_eos: Optional[bool] = False
     user: Optional[str] = None

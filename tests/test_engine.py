@@ -255,3 +255,20 @@ def test_engine_fuzz_mixed_workload():
         assert n == expected[rid], (rid, n, expected[rid])
     bm = engine.scheduler.block_manager
     assert bm.num_free_blocks == bm.num_blocks - 1  # all blocks returned
+
+
+def test_min_p_restricts_support(engine):
+    """min_p close to 1 forces near-greedy sampling; support must collapse
+    to the argmax token across seeds."""
+    greedy = engine.generate(
+        [[1, 2, 3]], SamplingParams(temperature=0.0, max_tokens=1)
+    )
+    g_tok = list(greedy.values())[0].output_token_ids[0]
+    for seed in range(5):
+        out = engine.generate(
+            [[1, 2, 3]],
+            SamplingParams(
+                temperature=1.0, min_p=0.999, max_tokens=1, seed=seed
+            ),
+        )
+        assert list(out.values())[0].output_token_ids[0] == g_tok
