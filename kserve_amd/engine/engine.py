@@ -54,6 +54,7 @@ class Sampler:
             if r.sampling_params.presence_penalty != 0.0
             or r.sampling_params.frequency_penalty != 0.0
             or r.sampling_params.repetition_penalty != 1.0
+            or r.sampling_params.logit_bias
         ]
         if not rows:
             return logits
@@ -61,6 +62,21 @@ class Sampler:
         for i in rows:
             r = requests[i]
             sp = r.sampling_params
+            if sp.logit_bias:
+                ids = torch.tensor(
+                    [t for t in sp.logit_bias if 0 <= t < logits.shape[-1]],
+                    device=logits.device,
+                    dtype=torch.long,
+                )
+                if ids.numel():
+                    vals = torch.tensor(
+                        [sp.logit_bias[int(t)] for t in ids],
+                        device=logits.device,
+                        dtype=torch.float32,
+                    )
+                    logits[i, ids] = (
+                        logits[i, ids].float() + vals
+                    ).to(logits.dtype)
             seen: dict = {}
             for t in r.output_token_ids:
                 seen[t] = seen.get(t, 0) + 1

@@ -7,7 +7,7 @@ delegates to — vllm_model.py:248-271); implementation is fresh.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional, Union
+from typing import Dict, List, Optional, Union
 
 
 @dataclass
@@ -18,6 +18,8 @@ class SamplingParams:
     top_k: int = -1
     # keep tokens with prob >= min_p * max_prob (0 disables)
     min_p: float = 0.0
+    # OpenAI logit_bias: token-id -> additive bias (-100..100)
+    logit_bias: Optional[Dict[int, float]] = None
     max_tokens: int = 16
     min_tokens: int = 0
     stop: List[str] = field(default_factory=list)

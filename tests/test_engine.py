@@ -272,3 +272,23 @@ def test_min_p_restricts_support(engine):
             ),
         )
         assert list(out.values())[0].output_token_ids[0] == g_tok
+
+
+def test_logit_bias(engine):
+    """+100 bias forces a token; -100 bans the greedy choice."""
+    base = engine.generate(
+        [[4, 5, 6]], SamplingParams(temperature=0.0, max_tokens=1)
+    )
+    g_tok = list(base.values())[0].output_token_ids[0]
+    forced = engine.generate(
+        [[4, 5, 6]],
+        SamplingParams(temperature=0.0, max_tokens=1, logit_bias={77: 100.0}),
+    )
+    assert list(forced.values())[0].output_token_ids[0] == 77
+    banned = engine.generate(
+        [[4, 5, 6]],
+        SamplingParams(
+            temperature=0.0, max_tokens=1, logit_bias={g_tok: -100.0}
+        ),
+    )
+    assert list(banned.values())[0].output_token_ids[0] != g_tok
