@@ -222,12 +222,14 @@ class LLMModel(OpenAIModel):
         self, prompt, sp: SamplingParams, request: CompletionRequest
     ) -> AsyncIterator[Completion]:
         ids = self._encode_prompt(prompt)
+        completion_tokens = 0
         async for out in self.async_engine.generate(ids, sp):
             text = out.text_delta or (
                 ""
                 if self.tokenizer is not None
                 else "".join(f"{t} " for t in out.new_token_ids)
             )
+            completion_tokens += len(out.new_token_ids)
             yield Completion(
                 model=self.name,
                 choices=[
@@ -237,6 +239,18 @@ class LLMModel(OpenAIModel):
                         finish_reason=out.finish_reason if out.finished else None,
                     )
                 ],
+            )
+        # OpenAI stream_options.include_usage: a final usage-only chunk
+        opts = getattr(request, "stream_options", None) or {}
+        if opts.get("include_usage"):
+            yield Completion(
+                model=self.name,
+                choices=[],
+                usage=UsageInfo(
+                    prompt_tokens=len(ids),
+                    completion_tokens=completion_tokens,
+                    total_tokens=len(ids) + completion_tokens,
+                ),
             )
 
     # -- chat ---------------------------------------------------------------

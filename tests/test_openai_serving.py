@@ -342,3 +342,28 @@ def test_stream_disconnect_aborts_request(client):
             break
         time.sleep(0.05)
     assert not eng.engine.has_unfinished(), "request not aborted after disconnect"
+
+
+def test_stream_include_usage(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [1, 2, 3],
+            "max_tokens": 4,
+            "temperature": 0.0,
+            "stream": True,
+            "stream_options": {"include_usage": True},
+        },
+    )
+    assert r.status_code == 200
+    chunks = [
+        json.loads(line[len("data: "):])
+        for line in r.text.splitlines()
+        if line.startswith("data: ")
+    ]
+    usage_chunks = [c for c in chunks if c.get("usage")]
+    assert len(usage_chunks) == 1
+    u = usage_chunks[-1]["usage"]
+    assert u["completion_tokens"] == 4 and u["prompt_tokens"] == 3
+    assert chunks[-1].get("usage")  # usage chunk is last
