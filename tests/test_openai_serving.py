@@ -360,7 +360,7 @@ def test_stream_include_usage(client):
     chunks = [
         json.loads(line[len("data: "):])
         for line in r.text.splitlines()
-        if line.startswith("data: ")
+        if line.startswith("data: ") and "[DONE]" not in line
     ]
     usage_chunks = [c for c in chunks if c.get("usage")]
     assert len(usage_chunks) == 1
