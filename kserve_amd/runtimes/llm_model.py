@@ -284,7 +284,9 @@ class LLMModel(OpenAIModel):
         self, ids: List[int], sp: SamplingParams, request: ChatCompletionRequest
     ) -> AsyncIterator[ChatCompletionChunk]:
         first = True
+        completion_tokens = 0
         async for out in self.async_engine.generate(ids, sp):
+            completion_tokens += len(out.new_token_ids)
             delta = ChatCompletionChunkDelta(
                 role="assistant" if first else None,
                 content=out.text_delta
@@ -304,4 +306,15 @@ class LLMModel(OpenAIModel):
                         finish_reason=out.finish_reason if out.finished else None,
                     )
                 ],
+            )
+        opts = getattr(request, "stream_options", None) or {}
+        if opts.get("include_usage"):
+            yield ChatCompletionChunk(
+                model=self.name,
+                choices=[],
+                usage=UsageInfo(
+                    prompt_tokens=len(ids),
+                    completion_tokens=completion_tokens,
+                    total_tokens=len(ids) + completion_tokens,
+                ),
             )
