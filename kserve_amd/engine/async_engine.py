@@ -88,7 +88,17 @@ class AsyncLLMEngine:
                         if prompt is None:  # abort sentinel
                             self.engine.abort_request(rid)
                         else:
-                            self.engine.add_request(prompt, sp, request_id=rid)
+                            try:
+                                self.engine.add_request(
+                                    prompt, sp, request_id=rid
+                                )
+                            except ValueError as e:
+                                # invalid request (e.g. over-long prompt):
+                                # surface to ITS stream, keep the loop alive
+                                entry = self._streams.pop(rid, None)
+                                if entry is not None:
+                                    q, loop = entry
+                                    loop.call_soon_threadsafe(q.put_nowait, e)
                         block = False
                 except queue.Empty:
                     pass
@@ -127,6 +137,8 @@ class AsyncLLMEngine:
         try:
             while True:
                 item = await q.get()
+                if isinstance(item, ValueError):
+                    raise item  # per-request validation error
                 if isinstance(item, BaseException):
                     raise EngineDead(str(item))
                 yield item

@@ -226,6 +226,12 @@ class LLMEngine:
             prompt_token_ids = self.tokenizer.encode(prompt)
         else:
             prompt_token_ids = list(prompt)
+        if len(prompt_token_ids) > self.config.scheduler.max_model_len:
+            # fail fast: a silently-dropped request would hang its caller
+            raise ValueError(
+                f"prompt length {len(prompt_token_ids)} exceeds "
+                f"max_model_len {self.config.scheduler.max_model_len}"
+            )
         req = Request(
             request_id,
             prompt_token_ids,

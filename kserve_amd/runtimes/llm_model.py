@@ -169,7 +169,10 @@ class LLMModel(OpenAIModel):
                         n=1,
                         seed=(sp.seed + j) if sp.seed is not None else None,
                     )
-                out = await self.async_engine.generate_full(ids, sp_j)
+                try:
+                    out = await self.async_engine.generate_full(ids, sp_j)
+                except ValueError as e:
+                    raise InvalidInput(str(e)) from e
                 completion_tokens += len(out.output_token_ids)
                 text = (
                     out.output_text
@@ -262,7 +265,10 @@ class LLMModel(OpenAIModel):
         sp.lora_name = self._lora_for(request.model)
         if request.stream:
             return self._stream_chat(ids, sp, request)
-        out = await self.async_engine.generate_full(ids, sp)
+        try:
+            out = await self.async_engine.generate_full(ids, sp)
+        except ValueError as e:
+            raise InvalidInput(str(e)) from e
         text = out.output_text if out.output_text else self._decode(out.output_token_ids)
         return ChatCompletion(
             model=self.name,

@@ -367,3 +367,13 @@ def test_stream_include_usage(client):
     u = usage_chunks[-1]["usage"]
     assert u["completion_tokens"] == 4 and u["prompt_tokens"] == 3
     assert chunks[-1].get("usage")  # usage chunk is last
+
+
+def test_overlong_prompt_is_400_not_hang(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={"model": "tiny", "prompt": list(range(100)) * 3,
+              "max_tokens": 2, "temperature": 0.0},
+        )
+    assert r.status_code == 400
+    assert "max_model_len" in r.text
