@@ -87,6 +87,26 @@ class AsyncLLMEngine:
                         rid, prompt, sp = item
                         if prompt is None:  # abort sentinel
                             self.engine.abort_request(rid)
+                            # if anyone is still consuming this stream,
+                            # terminate it instead of leaving it hanging
+                            entry = self._streams.pop(rid, None)
+                            if entry is not None:
+                                from kserve_amd.engine.request import (
+                                    RequestOutput,
+                                )
+
+                                q, loop = entry
+                                loop.call_soon_threadsafe(
+                                    q.put_nowait,
+                                    RequestOutput(
+                                        request_id=rid,
+                                        new_token_ids=[],
+                                        finished=True,
+                                        finish_reason="abort",
+                                        output_token_ids=[],
+                                        num_prompt_tokens=0,
+                                    ),
+                                )
                         else:
                             try:
                                 self.engine.add_request(
@@ -149,6 +169,11 @@ class AsyncLLMEngine:
                 # client disconnected mid-stream: abort in engine thread
                 self._streams.pop(rid, None)
                 self._submit_abort(rid)
+
+    async def abort(self, rid: str) -> None:
+        """Abort a request by id; any consumer of its stream receives a
+        final finished output with reason "abort"."""
+        self._submit_abort(rid)
 
     def _submit_abort(self, rid: str):
         if self.engine is not None:

@@ -377,3 +377,31 @@ def test_overlong_prompt_is_400_not_hang(client):
         )
     assert r.status_code == 400
     assert "max_model_len" in r.text
+
+
+def test_external_abort_terminates_stream(client):
+    """AsyncLLMEngine.abort() must end a consumer's stream with an abort
+    output instead of hanging it."""
+    import asyncio
+
+    model = client.app.state.llm_model
+    eng = model.async_engine
+
+    async def run():
+        from kserve_amd.engine.sampling_params import SamplingParams
+
+        gen = eng.generate(
+            [1, 2, 3], SamplingParams(temperature=0.0, max_tokens=4096),
+            request_id="abort-me",
+        )
+        first = await asyncio.wait_for(gen.__anext__(), timeout=30)
+        assert not first.finished
+        await eng.abort("abort-me")
+        # the stream must terminate promptly
+        while True:
+            out = await asyncio.wait_for(gen.__anext__(), timeout=30)
+            if out.finished:
+                assert out.finish_reason == "abort"
+                break
+
+    asyncio.new_event_loop().run_until_complete(run())
