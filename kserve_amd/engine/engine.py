@@ -242,9 +242,27 @@ class LLMEngine:
             if self.lora_manager is None:
                 raise ValueError("no LoRA adapters registered")
             req.lora_id = self.lora_manager.lookup(sampling_params.lora_name)
+        if sampling_params.response_format in ("json", "json_object"):
+            from kserve_amd.engine.guided import JsonMachine
+
+            req.guided_machine = JsonMachine(object_only=True)
         LLM_PROMPT_TOKENS.inc(len(prompt_token_ids))
         self.scheduler.add_request(req)
         return request_id
+
+    @property
+    def guided_json(self):
+        """Lazily-built guided-JSON token-mask processor (engine/guided.py)."""
+        if getattr(self, "_guided_json", None) is None:
+            from kserve_amd.engine.guided import GuidedJsonProcessor
+
+            eos = self.eos_token_id
+            if eos is not None and eos < 0:  # -1 = EOS disabled
+                eos = None
+            self._guided_json = GuidedJsonProcessor.from_tokenizer(
+                self.tokenizer, self.config.model.vocab_size, eos
+            )
+        return self._guided_json
 
     def register_lora(self, name: str, path: str) -> int:
         """Load a PEFT adapter and make it addressable by name (reference:
@@ -289,13 +307,15 @@ class LLMEngine:
                 and r.sampling_params.frequency_penalty == 0.0
                 and r.sampling_params.repetition_penalty == 1.0
                 and r.lora_id == 0
+                and r.guided_machine is None
                 for r in batch.requests
             )
             # speculative decoding (prompt-lookup): draft from the request's
             # own context, verify k+1 positions in ONE forward through the
             # paged-context prefill path; exact greedy outputs by construction
             pp = comm.get_state().pp_size
-            if sched.speculative_ngram > 0 and pure_greedy and pp == 1:
+            no_guided = all(r.guided_machine is None for r in batch.requests)
+            if sched.speculative_ngram > 0 and pure_greedy and pp == 1 and no_guided:
                 k_cap = (
                     self.scheduler.reserve_decode_window(
                         batch, sched.speculative_ngram + 1
@@ -318,6 +338,7 @@ class LLMEngine:
                 and r.sampling_params.frequency_penalty == 0.0
                 and r.sampling_params.repetition_penalty == 1.0
                 and r.lora_id == 0
+                and r.guided_machine is None
                 for r in batch.requests
             ):
                 k = self.scheduler.reserve_decode_window(
@@ -356,7 +377,32 @@ class LLMEngine:
         outputs: List[RequestOutput] = []
         finished: List[Request] = []
         if sample_reqs:
+            guided_rows = [
+                i for i, r in enumerate(sample_reqs)
+                if r.guided_machine is not None
+            ]
+            if guided_rows and logits is not None:
+                logits = logits.clone()
+                neg = float("-inf")
+                for i in guided_rows:
+                    allowed = self.guided_json.allowed_tokens(
+                        sample_reqs[i].guided_machine
+                    )
+                    row = torch.full_like(logits[i], neg)
+                    if allowed:
+                        idx = torch.tensor(allowed, device=logits.device)
+                        row[idx] = logits[i, idx]
+                    logits[i] = row
             tokens = self._sample_tokens(logits, sample_reqs)
+            if guided_rows:
+                for i in guided_rows:
+                    ok = self.guided_json.advance(
+                        sample_reqs[i].guided_machine, int(tokens[i])
+                    )
+                    if not ok:  # no legal token existed (masked-out row)
+                        r = sample_reqs[i]
+                        r.status = RequestStatus.FINISHED_STOPPED
+                        r.is_finished = True
             # top-N logprobs for the (rare) requests that ask for them
             lp_idx = [
                 i for i, r in enumerate(sample_reqs)

@@ -63,6 +63,8 @@ class Request:
         self.num_computed_tokens = 0
         # multi-LoRA adapter id (0 = base model)
         self.lora_id = 0
+        # guided decoding (engine/guided.py): per-request JSON machine
+        self.guided_machine = None
         # scheduler arrival sequence (set by Scheduler.add_request)
         self.arrival_seq = 0
         # prefix caching: tokens whose KV was found resident at allocate

@@ -20,6 +20,8 @@ class SamplingParams:
     min_p: float = 0.0
     # OpenAI logit_bias: token-id -> additive bias (-100..100)
     logit_bias: Optional[Dict[int, float]] = None
+    # guided decoding: "json_object" constrains output to valid JSON
+    response_format: Optional[str] = None
     max_tokens: int = 16
     min_tokens: int = 0
     stop: List[str] = field(default_factory=list)

@@ -48,6 +48,8 @@ class CompletionRequest(BaseModel):
     top_p: Optional[float] = 1.0
     top_k: Optional[int] = -1
     min_p: Optional[float] = 0.0
+    # {"type": "json_object"} constrains output to valid JSON
+    response_format: Optional[Dict[str, Any]] = None
     min_tokens: Optional[int] = 0
     ignore_eos: Optional[bool] = False
     # scheduling priority (vLLM extension: lower value runs first)
@@ -110,6 +112,7 @@ class ChatCompletionRequest(BaseModel):
     stop: Optional[Union[str, List[str]]] = None
     stream: Optional[bool] = False
     stream_options: Optional[Dict[str, Any]] = None
+    response_format: Optional[Dict[str, Any]] = None
     temperature: Optional[float] = 1.0
     top_p: Optional[float] = 1.0
     top_k: Optional[int] = -1

@@ -48,6 +48,11 @@ def _to_sampling_params(
         top_p=req.top_p if req.top_p is not None else 1.0,
         top_k=req.top_k if req.top_k is not None else -1,
         min_p=getattr(req, "min_p", 0.0) or 0.0,
+        response_format=(
+            (req.response_format or {}).get("type")
+            if getattr(req, "response_format", None)
+            else None
+        ),
         logit_bias=(
             {int(k): float(v) for k, v in req.logit_bias.items()}
             if getattr(req, "logit_bias", None)
