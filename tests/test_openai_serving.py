@@ -405,3 +405,40 @@ def test_external_abort_terminates_stream(client):
                 break
 
     asyncio.new_event_loop().run_until_complete(run())
+
+
+def test_response_format_json_object(client):
+    r = client.post(
+        "/openai/v1/completions",
+        json={
+            "model": "tiny",
+            "prompt": [1, 2, 3],
+            "max_tokens": 40,
+            "temperature": 0.8,
+            "seed": 3,
+            "response_format": {"type": "json_object"},
+        },
+    )
+    assert r.status_code == 200, r.text
+    # tokenizer-less fallback streams token ids as text; re-check the raw
+    # ids through the engine's own notion of validity instead
+    from kserve_amd.engine.guided import JsonMachine
+
+    model = client.app.state.llm_model
+    # round-trip through a direct engine call for byte-level verification
+    import asyncio
+
+    from kserve_amd.engine.sampling_params import SamplingParams
+
+    async def run():
+        out = await model.async_engine.generate_full(
+            [1, 2, 3],
+            SamplingParams(
+                temperature=0.8, seed=3, max_tokens=40,
+                response_format="json_object",
+            ),
+        )
+        return out
+
+    out = asyncio.new_event_loop().run_until_complete(run())
+    assert JsonMachine().accepts(bytes(out.output_token_ids))
