@@ -242,7 +242,13 @@ class LLMEngine:
             if self.lora_manager is None:
                 raise ValueError("no LoRA adapters registered")
             req.lora_id = self.lora_manager.lookup(sampling_params.lora_name)
-        if sampling_params.response_format in ("json", "json_object"):
+        if sampling_params.guided_choice:
+            from kserve_amd.engine.guided import ChoiceMachine
+
+            req.guided_machine = ChoiceMachine(
+                [c.encode("utf-8") for c in sampling_params.guided_choice]
+            )
+        elif sampling_params.response_format in ("json", "json_object"):
             from kserve_amd.engine.guided import JsonMachine
 
             req.guided_machine = JsonMachine(object_only=True)
@@ -394,15 +400,16 @@ class LLMEngine:
                         row[idx] = logits[i, idx]
                     logits[i] = row
             tokens = self._sample_tokens(logits, sample_reqs)
+            guided_exhausted = set()
             if guided_rows:
                 for i in guided_rows:
                     ok = self.guided_json.advance(
                         sample_reqs[i].guided_machine, int(tokens[i])
                     )
-                    if not ok:  # no legal token existed (masked-out row)
-                        r = sample_reqs[i]
-                        r.status = RequestStatus.FINISHED_STOPPED
-                        r.is_finished = True
+                    if not ok:
+                        # the machine has no legal continuation (row fully
+                        # masked): stop WITHOUT emitting the bogus token
+                        guided_exhausted.add(sample_reqs[i].request_id)
             # top-N logprobs for the (rare) requests that ask for them
             lp_idx = [
                 i for i, r in enumerate(sample_reqs)
@@ -431,7 +438,12 @@ class LLMEngine:
             now = time.monotonic()
             for req, tok in zip(sample_reqs, tokens):
                 first = req.first_token_time is None
-                req.append_output_token(int(tok))
+                if req.request_id in guided_exhausted:
+                    req.status = RequestStatus.FINISHED_STOPPED
+                    req.is_finished = True
+                    req.finish_time = now
+                else:
+                    req.append_output_token(int(tok))
                 if first:
                     LLM_TTFT_HIST.observe(now - req.arrival_time)
                 req.maybe_finish(self.config.scheduler.max_model_len)

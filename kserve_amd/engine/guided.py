@@ -277,3 +277,56 @@ class GuidedJsonProcessor:
 
     def advance(self, machine: JsonMachine, token_id: int) -> bool:
         return machine.accepts(self.token_bytes[token_id])
+
+
+class ChoiceMachine:
+    """Constrain output to one of a fixed set of byte strings (vLLM
+    guided_choice). State = the set of candidate indices still live plus
+    the byte position; legal bytes are the candidates' next bytes."""
+
+    __slots__ = ("choices", "live", "pos", "done")
+
+    def __init__(self, choices: Sequence[bytes]):
+        self.choices = [bytes(c) for c in choices]
+        self.live = list(range(len(self.choices)))
+        self.pos = 0
+        self.done = False
+
+    def clone(self) -> "ChoiceMachine":
+        m = ChoiceMachine.__new__(ChoiceMachine)
+        m.choices = self.choices
+        m.live = list(self.live)
+        m.pos = self.pos
+        m.done = self.done
+        return m
+
+    def signature(self) -> Tuple:
+        return ("choice", tuple(self.live), self.pos, self.done)
+
+    @property
+    def complete(self) -> bool:
+        return self.done
+
+    def advance(self, b: int) -> bool:
+        if self.done:
+            return False
+        nxt = [
+            i for i in self.live
+            if self.pos < len(self.choices[i]) and self.choices[i][self.pos] == b
+        ]
+        if not nxt:
+            return False
+        self.live = nxt
+        self.pos += 1
+        if any(len(self.choices[i]) == self.pos for i in self.live):
+            # a candidate is fully matched; prefer finishing exactly when
+            # no longer candidate shares the prefix
+            if all(len(self.choices[i]) == self.pos for i in self.live):
+                self.done = True
+        return True
+
+    def accepts(self, data: bytes) -> bool:
+        for b in data:
+            if not self.advance(b):
+                return False
+        return True

@@ -22,6 +22,8 @@ class SamplingParams:
     logit_bias: Optional[Dict[int, float]] = None
     # guided decoding: "json_object" constrains output to valid JSON
     response_format: Optional[str] = None
+    # guided decoding: constrain output to one of these strings
+    guided_choice: Optional[List[str]] = None
     max_tokens: int = 16
     min_tokens: int = 0
     stop: List[str] = field(default_factory=list)

@@ -265,6 +265,12 @@ def random_sample(
         # min-p: keep tokens with prob >= min_p * p_max
         mask |= probs_sorted < (min_p.unsqueeze(-1) * probs_sorted[:, :1])
     sorted_logits = sorted_logits.masked_fill(mask, float("-inf"))
+    # degenerate rows (everything masked, e.g. an exhausted guided-decoding
+    # machine with EOS disabled): pick rank 0 deterministically instead of
+    # feeding NaNs to multinomial — callers discard the token
+    dead = torch.isinf(sorted_logits).all(dim=-1)
+    if bool(dead.any()):
+        sorted_logits[dead, 0] = 0.0
     probs = torch.softmax(sorted_logits, dim=-1)
     choice = torch.multinomial(probs, 1, generator=generator).squeeze(-1)
     return sorted_idx.gather(-1, choice.unsqueeze(-1)).squeeze(-1)

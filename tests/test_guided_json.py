@@ -105,3 +105,28 @@ def test_engine_guided_completes_to_valid_json():
             jsonlib.loads(data.decode("utf-8", errors="strict"))
     # at least one of six seeds should finish a small object
     assert completed >= 1, "no seed completed a JSON object"
+
+
+def test_guided_choice_outputs_one_of_choices():
+    engine = make_engine()
+    choices = ["yes", "no", "maybe"]
+    seen = set()
+    for seed in range(8):
+        sp = SamplingParams(
+            temperature=1.0, seed=seed, max_tokens=10,
+            guided_choice=choices,
+        )
+        o = list(engine.generate([[seed + 1, 2]], sp).values())[0]
+        text = bytes(o.output_token_ids).decode()
+        assert text in choices, text
+        seen.add(text)
+    assert len(seen) >= 2, "sampling should reach multiple choices"
+
+
+def test_guided_choice_shared_prefix():
+    engine = make_engine()
+    sp = SamplingParams(
+        temperature=0.0, max_tokens=10, guided_choice=["app", "apple"]
+    )
+    o = list(engine.generate([[1]], sp).values())[0]
+    assert bytes(o.output_token_ids).decode() in ("app", "apple")
