@@ -320,8 +320,7 @@ class LLMEngine:
             # own context, verify k+1 positions in ONE forward through the
             # paged-context prefill path; exact greedy outputs by construction
             pp = comm.get_state().pp_size
-            no_guided = all(r.guided_machine is None for r in batch.requests)
-            if sched.speculative_ngram > 0 and pure_greedy and pp == 1 and no_guided:
+            if sched.speculative_ngram > 0 and pure_greedy and pp == 1:
                 k_cap = (
                     self.scheduler.reserve_decode_window(
                         batch, sched.speculative_ngram + 1
@@ -337,18 +336,9 @@ class LLMEngine:
             # multi-step window: all-greedy decode with no pending scheduling
             # events runs as back-to-back hipGraph replays
             k = 1
-            if self.config.scheduler.multi_step > 1 and comm.get_state().pp_size == 1 and all(
-                r.sampling_params.greedy
-                and r.sampling_params.logprobs is None
-                and r.sampling_params.presence_penalty == 0.0
-                and r.sampling_params.frequency_penalty == 0.0
-                and r.sampling_params.repetition_penalty == 1.0
-                and r.lora_id == 0
-                and r.guided_machine is None
-                for r in batch.requests
-            ):
+            if sched.multi_step > 1 and pp == 1 and pure_greedy:
                 k = self.scheduler.reserve_decode_window(
-                    batch, self.config.scheduler.multi_step
+                    batch, sched.multi_step
                 )
             if k > 1:
                 return self._run_decode_window(batch, k)
