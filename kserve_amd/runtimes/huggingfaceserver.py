@@ -43,6 +43,10 @@ def detect_backend(model_dir: str) -> str:
     model_type = cfg.get("model_type", "")
     if model_type in ("llama", "mistral", "mixtral", "qwen2"):
         return "engine"
+    if any(a.endswith(("ForCausalLM", "LMHeadModel")) for a in archs):
+        # decoder the native engine doesn't implement: HF generate fallback
+        # (reference generative_model.py behavior — request-serial)
+        return "hf"
     return "encoder"
 
 
@@ -107,6 +111,12 @@ def build_model(args):
         return LLMModel(
             args.model_name, cfg, tokenizer=tokenizer, lora_modules=lora_modules
         )
+    if backend == "hf":
+        from kserve_amd.runtimes.hf_generative import HFGenerativeModel
+
+        model = HFGenerativeModel(args.model_name, model_dir=args.model_dir)
+        model.load()
+        return model
     from kserve_amd.runtimes.encoder_model import EncoderModel
 
     model = EncoderModel(
@@ -120,7 +130,7 @@ def main(argv=None):
     from kserve_amd.model_server import ModelServer, build_arg_parser
 
     parser = build_arg_parser()
-    parser.add_argument("--backend", default=None, choices=["engine", "encoder"])
+    parser.add_argument("--backend", default=None, choices=["engine", "encoder", "hf"])
     parser.add_argument("--task", default=None)
     parser.add_argument("--tensor-parallel-size", dest="tensor_parallel_size", type=int, default=1)
     parser.add_argument("--pipeline-parallel-size", dest="pipeline_parallel_size", type=int, default=1)

@@ -175,3 +175,68 @@ class TestPredictiveServer:
         assert repo.get_model("iris-a") is not None
         assert repo.get_model("iris-b") is not None
         assert repo.is_model_ready("iris-a")
+
+
+class TestHFGenerativeFallback:
+    """Request-serial transformers.generate fallback (reference
+    generative_model.py) for decoder architectures the native engine does
+    not implement."""
+
+    @pytest.fixture(scope="class")
+    def hf_model(self):
+        transformers = pytest.importorskip("transformers")
+        import torch
+
+        from kserve_amd.runtimes.hf_generative import HFGenerativeModel
+
+        torch.manual_seed(0)
+        cfg = transformers.GPT2Config(
+            vocab_size=128, n_positions=128, n_embd=32, n_layer=1, n_head=2
+        )
+        gpt2 = transformers.GPT2LMHeadModel(cfg).eval()
+        return HFGenerativeModel("gpt2-tiny", model=gpt2, tokenizer=None)
+
+    def test_completion(self, hf_model):
+        import asyncio
+
+        from kserve_amd.protocol.rest.openai.types import CompletionRequest
+
+        req = CompletionRequest(
+            model="gpt2-tiny", prompt=[1, 2, 3], max_tokens=5, temperature=0.0
+        )
+        out = asyncio.new_event_loop().run_until_complete(
+            hf_model.create_completion(req)
+        )
+        assert out.usage.completion_tokens == 5
+        assert out.choices[0].text.strip()
+
+    def test_stream(self, hf_model):
+        import asyncio
+
+        from kserve_amd.protocol.rest.openai.types import CompletionRequest
+
+        req = CompletionRequest(
+            model="gpt2-tiny", prompt=[1, 2, 3], max_tokens=4,
+            temperature=0.0, stream=True,
+        )
+
+        async def run():
+            gen = await hf_model.create_completion(req)
+            return [c async for c in gen]
+
+        chunks = asyncio.new_event_loop().run_until_complete(run())
+        assert len(chunks) == 4
+
+    def test_backend_detection_routes_unknown_decoder(self, tmp_path):
+        import json as _json
+
+        from kserve_amd.runtimes.huggingfaceserver import detect_backend
+
+        d = tmp_path / "m"
+        d.mkdir()
+        (d / "config.json").write_text(
+            _json.dumps(
+                {"architectures": ["GPT2LMHeadModel"], "model_type": "gpt2"}
+            )
+        )
+        assert detect_backend(str(d)) == "hf"
