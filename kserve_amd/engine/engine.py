@@ -449,7 +449,11 @@ class LLMEngine:
                 outputs.append(
                     RequestOutput(
                         request_id=req.request_id,
-                        new_token_ids=[int(tok)],
+                        new_token_ids=(
+                            []
+                            if req.request_id in guided_exhausted
+                            else [int(tok)]
+                        ),
                         finished=req.finished,
                         finish_reason=req.finish_reason,
                         # full-list copy only when the request completes
