@@ -381,3 +381,43 @@ class TestLlama31RopeScaling:
             full_forward_logits(ours, token_ids), hf_logits,
             rtol=3e-4, atol=3e-4,
         )
+
+    def test_mixtral_chunked_prefill_matches_full(self):
+        """MoE + chunked prefill (paged-context attention) equivalence."""
+        from kserve_amd.engine.config import (
+            CacheConfig,
+            EngineConfig,
+            ModelConfig,
+            SchedulerConfig,
+        )
+        from kserve_amd.engine.engine import LLMEngine
+        from kserve_amd.engine.sampling_params import SamplingParams
+
+        def cfg(chunked):
+            return EngineConfig(
+                model=ModelConfig(
+                    vocab_size=128, hidden_size=64, intermediate_size=128,
+                    num_layers=2, num_heads=2, num_kv_heads=1, head_dim=32,
+                    max_position_embeddings=256, num_local_experts=4,
+                    num_experts_per_tok=2, model_name="moe-chunk",
+                ),
+                cache=CacheConfig(block_size=4, num_gpu_blocks=64),
+                scheduler=SchedulerConfig(
+                    max_num_seqs=4,
+                    max_num_batched_tokens=8 if chunked else 128,
+                    max_model_len=128,
+                    enable_chunked_prefill=chunked,
+                ),
+                device="cpu",
+                eos_token_id=-1,
+            )
+
+        sp = SamplingParams(temperature=0.0, max_tokens=6)
+        prompts = [list(range(1, 20)), [7, 8, 9]]
+        torch.manual_seed(3)
+        a = [o.output_token_ids
+             for o in LLMEngine(cfg(False)).generate(prompts, sp).values()]
+        torch.manual_seed(3)
+        b = [o.output_token_ids
+             for o in LLMEngine(cfg(True)).generate(prompts, sp).values()]
+        assert a == b
