@@ -87,7 +87,9 @@ class BlockManager:
         max_full = min(limit_tokens, n_prompt - 1) // bs
         toks = request.prompt_token_ids
         hashes: List[int] = []
-        h = 0
+        # seed with the LoRA adapter id: adapters change the K/V projections,
+        # so identical tokens under different adapters must NOT share blocks
+        h = hash(("lora", getattr(request, "lora_id", 0)))
         for i in range(max_full):
             h = hash((h, tuple(toks[i * bs : (i + 1) * bs])))
             hashes.append(h)

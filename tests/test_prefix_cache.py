@@ -155,3 +155,33 @@ def test_prefix_cache_with_chunked_prefill():
     b1 = [o.output_token_ids for o in both.generate(prompts, sp).values()]
     assert a == b0 == b1
     assert both.scheduler.block_manager.cache_hit_tokens > 0
+
+
+def test_prefix_cache_isolates_lora_adapters(tmp_path):
+    """LoRA changes K/V projections: identical prompts under different
+    adapters must not share cached blocks (the hash chain is seeded with
+    the adapter id)."""
+    from tests.test_lora import make_adapter_dir
+
+    torch.manual_seed(0)
+    cached = make_engine(True)
+    path, _ = make_adapter_dir(tmp_path, cached.config.model)
+    cached.register_lora("adapt", path)
+    torch.manual_seed(0)
+    plain = make_engine(False)
+    plain.register_lora("adapt", path)
+
+    shared = list(range(1, 21))
+    sp_base = SamplingParams(temperature=0.0, max_tokens=6)
+    sp_lora = SamplingParams(temperature=0.0, max_tokens=6, lora_name="adapt")
+
+    # populate the cache with BASE-model KV for the prompt
+    a_base = list(cached.generate([shared], sp_base).values())[0]
+    # the LoRA request must NOT hit those blocks
+    a_lora = list(cached.generate([shared], sp_lora).values())[0]
+    ref_lora = list(plain.generate([shared], sp_lora).values())[0]
+    assert a_lora.output_token_ids == ref_lora.output_token_ids
+    assert a_lora.output_token_ids != a_base.output_token_ids
+    # same-adapter reuse still works
+    b_lora = list(cached.generate([shared], sp_lora).values())[0]
+    assert b_lora.output_token_ids == ref_lora.output_token_ids
