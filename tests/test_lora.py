@@ -213,3 +213,32 @@ def test_lora_served_as_model_name(tmp_path):
         assert lora.status_code == 200, lora.text
         assert base.json()["choices"][0]["text"] != lora.json()["choices"][0]["text"]
         model.stop()
+
+
+def test_lora_with_chunked_prefill(tmp_path):
+    """LoRA deltas must apply identically when the prompt is chunked
+    (paged-context attention path)."""
+    torch.manual_seed(0)
+    full = make_engine()
+    path, _ = make_adapter_dir(tmp_path, full.config.model)
+    full.register_lora("adapt", path)
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=128),
+        cache=CacheConfig(block_size=4, num_gpu_blocks=128),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4,
+            max_num_batched_tokens=8,  # chunks the 19-token prompt
+            max_model_len=128,
+            enable_chunked_prefill=True,
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    chunked = LLMEngine(cfg)
+    chunked.register_lora("adapt", path)
+    sp = SamplingParams(temperature=0.0, max_tokens=6, lora_name="adapt")
+    prompts = [list(range(1, 20))]
+    a = [o.output_token_ids for o in full.generate(prompts, sp).values()]
+    b = [o.output_token_ids for o in chunked.generate(prompts, sp).values()]
+    assert a == b
