@@ -31,8 +31,11 @@ if os.path.exists(_TUNE_BASE.replace(".csv", "0.csv")):
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=64)
-    parser.add_argument("--warmup", type=int, default=16)
+    # defaults sized so a flagless run finishes in minutes AND its KV
+    # demand (concurrency * (prompt + 8*(steps+warmup) + 64) tokens) fits
+    # the pool with no preemption churn
+    parser.add_argument("--steps", type=int, default=16)
+    parser.add_argument("--warmup", type=int, default=4)
     parser.add_argument("--concurrency", type=int, default=1536,
                         help="concurrent sequences per GPU")
     parser.add_argument("--prompt-len", type=int, default=512)
