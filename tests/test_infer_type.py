@@ -163,3 +163,33 @@ class TestDataFrame:
         df = InferRequest("m", [i1, i2]).as_dataframe()
         assert list(df.columns) == ["a", "b"]
         assert df["b"].tolist() == ["x", "y"]
+
+
+class TestBytesEdgeCases:
+    def test_empty_and_unicode_bytes_roundtrip(self):
+        """BYTES tensors with empty and multibyte elements survive the
+        4-byte-LE length-prefixed binary codec."""
+        import numpy as np
+
+        from kserve_amd.protocol.infer_type import InferInput, InferRequest
+
+        vals = np.array(["", "héllo", "日本語", "x" * 300], dtype=object)
+        inp = InferInput(name="s", shape=[4], datatype="BYTES")
+        inp.set_data_from_numpy(vals, binary_data=True)
+        req = InferRequest(model_name="m", infer_inputs=[inp])
+        body, json_len = req.to_rest()
+        assert isinstance(body, bytes)
+        back = InferRequest.from_bytes(body, json_len, "m")
+        out = back.inputs[0].as_numpy()
+        assert [
+            v.decode("utf-8") if isinstance(v, bytes) else v for v in out
+        ] == list(vals)
+
+    def test_zero_element_tensor(self):
+        import numpy as np
+
+        from kserve_amd.protocol.infer_type import InferInput
+
+        inp = InferInput(name="e", shape=[0], datatype="FP32")
+        inp.set_data_from_numpy(np.zeros((0,), dtype=np.float32), binary_data=True)
+        assert inp.parameters["binary_data_size"] == 0
