@@ -163,3 +163,33 @@ def test_priority_scheduling():
     sched._preempt(urgent)
     sched.running.remove(urgent)
     assert sched.waiting[0].request_id == "urgent"
+
+
+def test_decode_not_starved_when_seats_full():
+    """Prefill priority is bounded by max_num_seqs: once the running set is
+    full, waiting prefills cannot block decode progress."""
+    from kserve_amd.engine.config import CacheConfig, SchedulerConfig
+    from kserve_amd.engine.request import Request
+    from kserve_amd.engine.sampling_params import SamplingParams
+    from kserve_amd.engine.scheduler import Scheduler
+
+    sched = Scheduler(
+        SchedulerConfig(max_num_seqs=2, max_num_batched_tokens=64,
+                        max_model_len=64),
+        CacheConfig(block_size=4),
+        num_gpu_blocks=64,
+    )
+    for i in range(4):
+        sched.add_request(
+            Request(f"r{i}", [1, 2, 3], SamplingParams(max_tokens=8),
+                    eos_token_id=-1)
+        )
+    b1 = sched.schedule()
+    assert b1.is_prefill and len(b1.requests) == 2  # seats now full
+    for r in b1.requests:
+        r.num_computed_tokens = r.num_tokens  # prompt done
+        r.append_output_token(5)
+    b2 = sched.schedule()
+    assert not b2.is_prefill, "decode must proceed while prefills wait"
+    assert len(b2.requests) == 2
+    assert sched.num_waiting == 2
