@@ -172,11 +172,12 @@ def random_sample(
     generator: Optional[torch.Generator] = None,
     min_p: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
-    """Temperature/top-k/top-p sampling.
+    """Temperature/top-k/top-p/min-p sampling.
 
     GPU: one fused HIP kernel. Plain temperature sampling is Gumbel-max;
     top-k/top-p use a bf16 radix-histogram select (exact k-th value / mass
     threshold, no sort) followed by Gumbel-max over the surviving set.
+    min_p (rare) routes to the fp32 sort path.
     """
     has_min_p = min_p is not None and bool((min_p > 0).any())
     if logits.is_cuda and seeds is not None and not has_min_p:
