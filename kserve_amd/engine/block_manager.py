@@ -11,6 +11,8 @@ Replaces the paged-KV machinery the reference delegates to vLLM
 
 from __future__ import annotations
 
+import hashlib
+
 from collections import OrderedDict
 from typing import Dict, List, Optional, Tuple
 
@@ -38,6 +40,13 @@ class BlockManager:
         # (padded decode lanes write their KV there); never handed out.
         self._free: List[int] = list(range(1, num_blocks))
         self._tables: Dict[str, List[int]] = {}
+        # monotonic per-request table identity: a new value every time the
+        # request's block ids are (re)assigned (allocate / swap_in). The
+        # runner's pinned block-table cache compares this, so a preempted or
+        # swapped request with a same-LENGTH but different-CONTENT table is
+        # never served stale block ids.
+        self._seq = 0
+        self._table_seq: Dict[str, int] = {}
         # host-DRAM offload tier (pinned; hipMemcpyAsync side stream)
         self.num_cpu_blocks = num_cpu_blocks
         self._cpu_free: List[int] = list(range(num_cpu_blocks))
@@ -47,8 +56,8 @@ class BlockManager:
         # for the uncached suffix) --
         self.enable_prefix_caching = enable_prefix_caching
         self._refcount: Dict[int, int] = {}
-        self._hash_to_block: Dict[int, int] = {}   # READY (computed) blocks
-        self._block_to_hash: Dict[int, int] = {}
+        self._hash_to_block: Dict[bytes, int] = {}  # READY (computed) blocks
+        self._block_to_hash: Dict[int, bytes] = {}
         # refcount-0 cached blocks, insertion order = LRU
         self._evictable: "OrderedDict[int, None]" = OrderedDict()
         self.cache_hit_tokens = 0
@@ -78,7 +87,7 @@ class BlockManager:
         return self.num_free_blocks >= 1
 
     # -- prefix-cache internals ---------------------------------------------
-    def _prompt_block_hashes(self, request: Request, limit_tokens: int) -> List[int]:
+    def _prompt_block_hashes(self, request: Request, limit_tokens: int) -> List[bytes]:
         """Chained hashes of the FULL prompt blocks within limit_tokens,
         capped so the final prompt token is always computed fresh (its
         logits are needed for the first sampled token)."""
@@ -86,12 +95,21 @@ class BlockManager:
         n_prompt = request.num_prompt_tokens
         max_full = min(limit_tokens, n_prompt - 1) // bs
         toks = request.prompt_token_ids
-        hashes: List[int] = []
-        # seed with the LoRA adapter id: adapters change the K/V projections,
-        # so identical tokens under different adapters must NOT share blocks
-        h = hash(("lora", getattr(request, "lora_id", 0)))
+        hashes: List[bytes] = []
+        # chained sha256 over packed token ids: collision-proof against both
+        # accidental and adversarially crafted prompts (a 64-bit hash()
+        # collision would alias different content to the same KV block and
+        # leak context across requests). Seeded with the LoRA adapter id:
+        # adapters change the K/V projections, so identical tokens under
+        # different adapters must NOT share blocks.
+        h = hashlib.sha256(
+            b"lora:%d" % getattr(request, "lora_id", 0)
+        ).digest()
         for i in range(max_full):
-            h = hash((h, tuple(toks[i * bs : (i + 1) * bs])))
+            chunk = toks[i * bs : (i + 1) * bs]
+            h = hashlib.sha256(
+                h + b"".join(t.to_bytes(8, "little") for t in chunk)
+            ).digest()
             hashes.append(h)
         return hashes
 
@@ -189,9 +207,16 @@ class BlockManager:
             self._refcount[blk] = 1
             blocks.append(blk)
         self._tables[request.request_id] = blocks
+        self._seq += 1
+        self._table_seq[request.request_id] = self._seq
         request.block_table = blocks
         request.num_cached_tokens = cached_tokens
         return blocks
+
+    def table_seq(self, request_id: str) -> int:
+        """Identity of the request's current block table (changes whenever
+        the block ids are reassigned, not merely appended to)."""
+        return self._table_seq.get(request_id, -1)
 
     def take_blocks(self, n: int) -> List[int]:
         """Pop n fresh blocks (refcount 1), evicting cached blocks if the
@@ -221,6 +246,7 @@ class BlockManager:
 
     def free(self, request: Request) -> None:
         table = self._tables.pop(request.request_id, None)
+        self._table_seq.pop(request.request_id, None)
         if table:
             for blk in reversed(table):
                 self._release_block(blk)
@@ -253,6 +279,7 @@ class BlockManager:
         pairs; caller must COPY gpu->cpu before any kernel reuses the gpu
         blocks (the runner orders this via a side-stream event)."""
         table = self._tables.pop(request.request_id)
+        self._table_seq.pop(request.request_id, None)
         cpu_blocks = [self._cpu_free.pop() for _ in table]
         self._cpu_tables[request.request_id] = cpu_blocks
         pairs = list(zip(table, cpu_blocks))
@@ -279,6 +306,8 @@ class BlockManager:
             self._refcount[blk] = 1
             gpu_blocks.append(blk)
         self._tables[request.request_id] = gpu_blocks
+        self._seq += 1
+        self._table_seq[request.request_id] = self._seq
         request.block_table = gpu_blocks
         pairs = list(zip(cpu_table, gpu_blocks))
         self._cpu_free.extend(reversed(cpu_table))

@@ -307,12 +307,15 @@ class ModelRunner:
             t = tables[i]
             sm[i] = t[pos // bs] * bs + pos % bs
             ctx[i] = pos + 1
-            # cached numpy row per request; refresh only on growth
+            # cached numpy row per request; refreshed on growth AND on table
+            # identity change (swap_in / recompute preemption reassign block
+            # ids at the same length — block_manager.table_seq catches those)
+            seq = block_manager.table_seq(req.request_id)
             cache = self._bt_cache.get(req.request_id)
-            if cache is None or cache.shape[0] != len(t):
-                cache = np.asarray(t, dtype=np.int32)
+            if cache is None or cache[0] != seq or cache[1].shape[0] != len(t):
+                cache = (seq, np.asarray(t, dtype=np.int32))
                 self._bt_cache[req.request_id] = cache
-            btab[i, : cache.shape[0]] = cache
+            btab[i, : cache[1].shape[0]] = cache[1]
         return n, max_blocks
 
     def release_request(self, request_id: str):

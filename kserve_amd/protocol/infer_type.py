@@ -520,6 +520,13 @@ class InferResponse:
             if bsz is None:
                 continue
             bsz = int(bsz)
+            if off + bsz > len(body):
+                # mirror the request-side truncation check: fail loudly here
+                # instead of with an unrelated reshape error downstream
+                raise InvalidInput(
+                    f"Truncated binary output {out.name}: need {bsz} bytes "
+                    f"at offset {off}, body has {len(body)}"
+                )
             out.set_raw_data(body[off : off + bsz])
             off += bsz
         return resp

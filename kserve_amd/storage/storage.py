@@ -116,10 +116,20 @@ class Storage:
     def _maybe_unpack(path: str, out_dir: str) -> None:
         if path.endswith((".tar.gz", ".tgz")):
             with tarfile.open(path) as t:
-                t.extractall(out_dir)
+                # filter="data" blocks tar-slip path traversal (absolute
+                # paths, "..", symlink escapes) from untrusted archives
+                # fetched over http(s) — mirrors the reference's
+                # kserve_storage extraction hardening.
+                t.extractall(out_dir, filter="data")
             os.remove(path)
         elif path.endswith(".zip"):
             with zipfile.ZipFile(path) as z:
+                for member in z.namelist():
+                    dest = os.path.realpath(os.path.join(out_dir, member))
+                    if not dest.startswith(os.path.realpath(out_dir) + os.sep):
+                        raise ValueError(
+                            f"zip member escapes extraction dir: {member!r}"
+                        )
                 z.extractall(out_dir)
             os.remove(path)
 
