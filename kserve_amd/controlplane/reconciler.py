@@ -38,10 +38,11 @@ ANN_BATCHER_MAX_LATENCY = "internal.serving.kserve.io/batcher-max-latency"
 ANN_AGENT = "internal.serving.kserve.io/agent"
 
 
-def component_annotations(spec: PredictorSpec) -> Dict[str, str]:
-    """reference components/component.go:40-92."""
+def component_annotations(spec) -> Dict[str, str]:
+    """reference components/component.go:40-92 (any component spec:
+    predictor has storage_uri, transformer/explainer usually do not)."""
     ann: Dict[str, str] = {}
-    if spec.storage_uri:
+    if getattr(spec, "storage_uri", None):
         ann[ANN_STORAGE_URI] = spec.storage_uri
     if spec.logger is not None:
         ann[ANN_LOGGER] = "true"
@@ -120,11 +121,16 @@ def render_deployment(
     isvc: InferenceService,
     runtimes: List[ServingRuntime],
     canary: bool = False,
+    storage_init_image: str = "kserve-amd/storage-initializer:latest",
+    agent_image: str = "kserve-amd/agent:latest",
 ) -> Dict:
     """Deployment manifest (raw mode; reference deployment_reconciler.go)."""
     p = isvc.spec.predictor
     name = predictor_service_name(isvc) + ("-canary" if canary else "")
-    pod = render_predictor_pod_spec(isvc, runtimes)
+    pod = render_predictor_pod_spec(
+        isvc, runtimes, storage_init_image=storage_init_image,
+        agent_image=agent_image,
+    )
     pod["metadata"]["labels"]["app"] = name
     return {
         "apiVersion": "apps/v1",
@@ -427,3 +433,202 @@ def reconcile_graph(
             },
         }
     return out
+
+
+# ---- Transformer / Explainer component deployments -------------------------
+# (reference components/transformer.go, explainer.go; --predictor_host arg
+# injection per v1beta1/transformer_custom.go:99-105, explainer_custom.go:80-84)
+
+def component_service_name(isvc: InferenceService, component: str) -> str:
+    return f"{isvc.name}-{component}"
+
+
+def render_component_deployment(
+    isvc: InferenceService,
+    component: str,  # "transformer" | "explainer"
+    storage_init_image: str = "kserve-amd/storage-initializer:latest",
+    agent_image: str = "kserve-amd/agent:latest",
+) -> Dict:
+    """Deployment for a transformer or explainer component. The container
+    comes from the component spec; --model_name and --predictor_host are
+    injected so the data plane forwards predict/explain calls to the
+    predictor service (reference TransformerSpec.GetContainer)."""
+    spec = getattr(isvc.spec, component)
+    if spec is None:
+        raise ValueError(f"isvc has no {component} spec")
+    containers = getattr(spec, "containers", None) or []
+    if containers:
+        container = copy.deepcopy(containers[0])
+    elif component == "explainer" and getattr(spec, "art", None) is not None:
+        container = {
+            "image": "kserve-amd/artexplainer:latest",
+            "command": ["python", "-m", "kserve_amd.runtimes.aifserver"],
+        }
+    else:
+        raise ValueError(f"{component} must specify a container")
+    container.setdefault("name", "kserve-container")
+    args = container.setdefault("args", [])
+    predictor_host = (
+        f"{predictor_service_name(isvc)}.{isvc.namespace}"
+    )
+    if not any(a.startswith("--predictor_host") for a in args):
+        args.extend(["--predictor_host", predictor_host])
+    if not any(a.startswith("--model_name") for a in args):
+        args.extend(["--model_name", isvc.name])
+    name = component_service_name(isvc, component)
+    pod = {
+        "metadata": {
+            "labels": {
+                "serving.kserve.io/inferenceservice": isvc.name,
+                "component": component,
+                "app": name,
+                **isvc.labels,
+            },
+            "annotations": {
+                **isvc.annotations,
+                **component_annotations(spec),
+            },
+        },
+        "spec": {"containers": [container]},
+    }
+    pod = mutate_pod(
+        pod, storage_init_image=storage_init_image, agent_image=agent_image
+    )
+    return {
+        "apiVersion": "apps/v1",
+        "kind": "Deployment",
+        "metadata": {
+            "name": name,
+            "namespace": isvc.namespace,
+            "labels": pod["metadata"]["labels"],
+        },
+        "spec": {
+            "replicas": max(getattr(spec, "min_replicas", 1) or 1, 1),
+            "selector": {"matchLabels": {"app": name}},
+            "template": pod,
+        },
+    }
+
+
+def render_component_service(isvc: InferenceService, component: str) -> Dict:
+    name = component_service_name(isvc, component)
+    return {
+        "apiVersion": "v1",
+        "kind": "Service",
+        "metadata": {"name": name, "namespace": isvc.namespace},
+        "spec": {
+            "selector": {"app": name},
+            "ports": [
+                {"name": "http", "port": 80, "targetPort": 8080},
+                {"name": "grpc", "port": 81, "targetPort": 8081},
+            ],
+        },
+    }
+
+
+def desired_state(
+    isvc: InferenceService,
+    runtimes: List[ServingRuntime],
+    config=None,
+) -> List[Dict]:
+    """Complete desired manifest list for the LIVE controller
+    (isvc_controller.py): predictor (+canary) + transformer + explainer
+    deployments/services, autoscaler (HPA or KEDA per AutoscalerConfig /
+    serving.kserve.io/autoscalerClass annotation), and the ingress backend
+    chosen by IngressConfig (Istio VS / HTTPRoute / k8s Ingress).
+
+    reference: the per-component fan-out of controller.go:281-305 plus
+    factory.go:CreateIngressReconciler."""
+    from kserve_amd.controlplane.configmap import InferenceServicesConfig
+    from kserve_amd.controlplane.ingress import (
+        render_keda_scaled_object,
+        select_ingress,
+    )
+
+    cfg = config or InferenceServicesConfig()
+    default_inference_service(isvc)
+    validate_inference_service(isvc)
+    p = isvc.spec.predictor
+    manifests: List[Dict] = []
+
+    mode = isvc.annotations.get(
+        "serving.kserve.io/deploymentMode", cfg.deploy.default_deployment_mode
+    )
+    has_transformer = isvc.spec.transformer is not None
+    has_explainer = isvc.spec.explainer is not None
+    traffic_split = None
+
+    if mode == "Serverless":
+        manifests.append(render_knative_service(isvc, runtimes))
+    else:
+        manifests.append(
+            render_deployment(
+                isvc, runtimes,
+                storage_init_image=cfg.storage_initializer.image,
+                agent_image=cfg.agent.image,
+            )
+        )
+        manifests.append(render_service(isvc))
+        pct = p.canary_traffic_percent
+        if pct is not None and 0 < pct < 100:
+            manifests.append(
+                render_deployment(
+                    isvc, runtimes, canary=True,
+                    storage_init_image=cfg.storage_initializer.image,
+                    agent_image=cfg.agent.image,
+                )
+            )
+            canary_svc = render_service(isvc)
+            canary_svc["metadata"]["name"] += "-canary"
+            canary_svc["spec"]["selector"] = {
+                "app": predictor_service_name(isvc) + "-canary"
+            }
+            manifests.append(canary_svc)
+            traffic_split = {"stable": 100 - pct, "canary": pct}
+        # autoscaler: keda class -> ScaledObject, else HPA
+        autoscaler_class = isvc.annotations.get(
+            "serving.kserve.io/autoscalerClass", cfg.autoscaler.autoscaler_class
+        )
+        if p.max_replicas and p.max_replicas > p.min_replicas:
+            if autoscaler_class == "keda":
+                manifests.append(
+                    render_keda_scaled_object(
+                        predictor_service_name(isvc),
+                        isvc.namespace,
+                        predictor_service_name(isvc),
+                        p.min_replicas,
+                        p.max_replicas,
+                        p.scale_metric or "cpu",
+                        p.scale_target or 80,
+                    )
+                )
+            else:
+                hpa = render_hpa(isvc)
+                if hpa:
+                    manifests.append(hpa)
+
+    for component in ("transformer", "explainer"):
+        if getattr(isvc.spec, component) is None:
+            continue
+        manifests.append(
+            render_component_deployment(
+                isvc,
+                component,
+                storage_init_image=cfg.storage_initializer.image,
+                agent_image=cfg.agent.image,
+            )
+        )
+        manifests.append(render_component_service(isvc, component))
+
+    ingress = select_ingress(
+        isvc.name,
+        isvc.namespace,
+        cfg.ingress,
+        mode,
+        has_transformer=has_transformer,
+        has_explainer=has_explainer,
+        traffic_split=traffic_split,
+    )
+    if ingress is not None:
+        manifests.append(ingress)
+    return manifests
