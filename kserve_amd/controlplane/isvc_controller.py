@@ -96,6 +96,8 @@ def _ext_fields(dst, src: Dict) -> None:
             max_batch_size=b.get("maxBatchSize", 32),
             max_latency_ms=b.get("maxLatency", 5000),
         )
+    if src.get("serviceAccountName"):
+        dst.service_account_name = src["serviceAccountName"]
 
 
 def _framework_spec(src: Dict) -> FrameworkSpec:
@@ -306,9 +308,25 @@ class InferenceServiceController:
             self._update_status(obj, status)
             return None  # invalid spec: wait for the user to fix it
 
+        # credentials: ServiceAccount secrets -> storage-initializer env
+        # (reference CreateSecretVolumeAndEnv via the pod webhook)
+        sa_name = isvc.spec.predictor.service_account_name
+        cred = None
+        if sa_name:
+            from kserve_amd.controlplane.credentials import (
+                CredentialsBuilder,
+                inject_credentials,
+            )
+
+            cred = CredentialsBuilder(self.server, namespace).for_service_account(
+                sa_name
+            )
+
         # apply all desired manifests with owner references
         applied_keys = set()
         for m in manifests:
+            if cred and m.get("kind") == "Deployment":
+                inject_credentials(m["spec"]["template"], *cred)
             m["metadata"].setdefault("namespace", namespace)
             m["metadata"].setdefault("labels", {})[
                 "serving.kserve.io/inferenceservice"

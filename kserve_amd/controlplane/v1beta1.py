@@ -90,6 +90,7 @@ class ComponentExtensionSpec:
     timeout_seconds: Optional[int] = None
     logger: Optional[LoggerSpec] = None
     batcher: Optional[BatcherSpec] = None
+    service_account_name: Optional[str] = None
     labels: Dict[str, str] = field(default_factory=dict)
     annotations: Dict[str, str] = field(default_factory=dict)
 
