@@ -147,27 +147,38 @@ class S3Client:
         self.http = session or requests.Session()
 
     def _request(self, method: str, path: str,
-                 query: List[Tuple[str, str]]) -> requests.Response:
+                 query: List[Tuple[str, str]],
+                 body: Optional[bytes] = None) -> requests.Response:
         parsed = urlparse(self.endpoint)
         host = parsed.netloc
         headers = {}
         if not self.anonymous:
+            payload_hash = (
+                hashlib.sha256(body).hexdigest() if body is not None
+                else EMPTY_SHA256
+            )
             headers = sigv4_headers(
                 method, host, path, query, self.region,
                 self.access_key, self.secret_key,
+                payload_hash=payload_hash,
             )
         qs = "&".join(
             f"{quote(k, safe='~')}={quote(v, safe='~')}" for k, v in sorted(query)
         )
         url = f"{self.endpoint}{quote(path, safe='/')}" + (f"?{qs}" if qs else "")
         r = self.http.request(
-            method, url, headers=headers, verify=self.verify_ssl, stream=True
+            method, url, headers=headers, data=body,
+            verify=self.verify_ssl, stream=True,
         )
         if r.status_code >= 400:
             raise RuntimeError(
                 f"S3 {method} {path} failed: {r.status_code} {r.text[:300]}"
             )
         return r
+
+    def put_object(self, bucket: str, key: str, body: bytes) -> None:
+        """PutObject (used by the payload-logger blob sink)."""
+        self._request("PUT", f"/{bucket}/{key}", [], body=body)
 
     def list_objects(self, bucket: str, prefix: str) -> List[str]:
         """ListObjectsV2 with continuation."""

@@ -266,3 +266,85 @@ class TestAgentDrainer:
             d = c.post("/drain")
             assert d.json()["draining"] is True
             assert c.get("/readyz").status_code == 503
+
+
+class TestLoggerStoreAndBatching:
+    """Parquet marshaller + blob-store sinks + batch strategies (reference
+    pkg/logger/store.go:64-104, marshaller_parquet.go, batch_*.go)."""
+
+    def _entry(self, rid="r1", payload=b'{"instances": [[1, 2]]}'):
+        from kserve_amd.agent.payload_logger import LogEntry
+
+        return LogEntry(
+            request_id=rid,
+            event_type="org.kubeflow.serving.inference.request",
+            model_name="m",
+            payload=payload,
+        )
+
+    def test_parquet_marshaller_roundtrip(self):
+        import io
+
+        import pyarrow.parquet as pq
+
+        from kserve_amd.agent.payload_logger import Marshaller
+
+        m = Marshaller("parquet")
+        data = m.marshal_batch([self._entry("a"), self._entry("b")])
+        table = pq.read_table(io.BytesIO(data))
+        assert table.num_rows == 2
+        assert table.column("request_id").to_pylist() == ["a", "b"]
+        assert "instances" in table.column("payload").to_pylist()[0]
+
+    def test_file_blob_store(self, tmp_path):
+        from kserve_amd.agent.payload_logger import BlobStore
+
+        store = BlobStore(f"file://{tmp_path}/logs")
+        store.put("x.json", b"{}")
+        assert (tmp_path / "logs" / "x.json").read_bytes() == b"{}"
+
+    def test_s3_blob_store_uploads(self):
+        from kserve_amd.agent.payload_logger import BlobStore
+
+        uploads = {}
+
+        class FakeS3:
+            def put_object(self, bucket, key, body):
+                uploads[(bucket, key)] = body
+
+        store = BlobStore("s3://logs-bucket/payloads", s3_client=FakeS3())
+        store.put("a.parquet", b"PAR1")
+        assert uploads == {("logs-bucket", "payloads/a.parquet"): b"PAR1"}
+
+    def test_size_batch_flushes_at_threshold(self, tmp_path):
+        import asyncio
+
+        from kserve_amd.agent.payload_logger import (
+            BatchStrategy,
+            BlobStore,
+            PayloadLogger,
+        )
+
+        logger = PayloadLogger(
+            marshaller="parquet",
+            store=BlobStore(f"file://{tmp_path}/batched"),
+            batch=BatchStrategy("size", size=3),
+        )
+
+        async def run():
+            await logger.start()
+            for i in range(7):
+                await logger.log(self._entry(f"r{i}"))
+            await asyncio.sleep(0.2)
+            await logger.flush()  # drain the 7th entry
+            await logger.stop()
+
+        asyncio.run(run())
+        import glob
+
+        files = sorted(glob.glob(str(tmp_path / "batched" / "*.parquet")))
+        assert len(files) == 3  # 3 + 3 + 1(flush)
+        import pyarrow.parquet as pq
+
+        total = sum(pq.read_table(f).num_rows for f in files)
+        assert total == 7
