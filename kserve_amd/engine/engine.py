@@ -306,15 +306,21 @@ class LLMEngine:
             logits = self.runner.execute_prefill(batch, self.scheduler.block_manager)
         else:
             sched = self.config.scheduler
-            pure_greedy = all(
-                r.sampling_params.greedy
-                and r.sampling_params.logprobs is None
+            # windows need zero host work per token: anything that must see
+            # the logits or token history on the host each step disqualifies
+            window_ok = all(
+                r.sampling_params.logprobs is None
                 and r.sampling_params.presence_penalty == 0.0
                 and r.sampling_params.frequency_penalty == 0.0
                 and r.sampling_params.repetition_penalty == 1.0
+                and not r.sampling_params.logit_bias
+                and r.sampling_params.min_p == 0.0
                 and r.lora_id == 0
                 and r.guided_machine is None
                 for r in batch.requests
+            )
+            pure_greedy = window_ok and all(
+                r.sampling_params.greedy for r in batch.requests
             )
             # speculative decoding (prompt-lookup): draft from the request's
             # own context, verify k+1 positions in ONE forward through the
@@ -333,15 +339,18 @@ class LLMEngine:
                     ]
                     if any(drafts):
                         return self._run_spec_decode(batch, drafts)
-            # multi-step window: all-greedy decode with no pending scheduling
-            # events runs as back-to-back hipGraph replays
+            # multi-step window: decode with no pending scheduling events
+            # runs as back-to-back hipGraph replays. temperature>0 batches
+            # use the sampled-graph variant (in-graph fused sampler).
             k = 1
-            if sched.multi_step > 1 and pp == 1 and pure_greedy:
+            if sched.multi_step > 1 and pp == 1 and window_ok:
                 k = self.scheduler.reserve_decode_window(
                     batch, sched.multi_step
                 )
             if k > 1:
-                return self._run_decode_window(batch, k)
+                return self._run_decode_window(
+                    batch, k, sampled=not pure_greedy
+                )
             logits = self.runner.execute_decode(batch, self.scheduler.block_manager)
         # advance computed-token counters for executed tokens
         for req, n in zip(batch.requests, batch.num_scheduled_tokens):
@@ -582,9 +591,11 @@ class LLMEngine:
             self.runner.release_request(req.request_id)
         return outputs
 
-    def _run_decode_window(self, batch, k: int) -> List[RequestOutput]:
+    def _run_decode_window(
+        self, batch, k: int, sampled: bool = False
+    ) -> List[RequestOutput]:
         tokens_k = self.runner.multi_step_decode(
-            batch, self.scheduler.block_manager, k
+            batch, self.scheduler.block_manager, k, sampled=sampled
         )
         # ONE RequestOutput per request per window (streaming consumers get
         # k-token chunks); per-token python only for stop conditions

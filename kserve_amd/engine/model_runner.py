@@ -42,9 +42,18 @@ class ModelRunner:
         ) // config.cache.block_size
         # hipGraph state
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        # sampled-decode variants: same buckets, sampler inside the graph
+        # (captured lazily on first temperature>0 window; shares the pool)
+        self._sampled_graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self._graph_pool = None
         self._graph_buffers: Optional[Dict[str, torch.Tensor]] = None
         self._graph_batch_sizes: List[int] = []
         self._window_host: Optional[torch.Tensor] = None
+        self._sample_pin: Optional[Dict[str, torch.Tensor]] = None
+        # monotonically advancing RNG step for windowed sampling (seeds are
+        # base + step*PRIME; the graph advances seeds in-replay, this mirrors
+        # the count on the host between windows)
+        self._rng_step = 0
         # pinned staging + per-request numpy block-table cache
         self._pin: Optional[Dict[str, torch.Tensor]] = None
         self._pin_np = None
@@ -465,8 +474,15 @@ class ModelRunner:
                 (max_bs, mb), dtype=torch.int32, device=dev
             ),
             "context_lens": torch.ones(max_bs, dtype=torch.int32, device=dev),
+            # per-request sampling params (sampled-window variants)
+            "temps": torch.ones(max_bs, dtype=torch.float32, device=dev),
+            "top_p": torch.ones(max_bs, dtype=torch.float32, device=dev),
+            "top_k": torch.zeros(max_bs, dtype=torch.int32, device=dev),
+            "seeds": torch.zeros(max_bs, dtype=torch.int64, device=dev),
+            "greedy_mask": torch.zeros(max_bs, dtype=torch.bool, device=dev),
             "logits": {},
             "sampled": {},
+            "sampled_rand": {},
         }
         torch.cuda.synchronize()
         pool = None
@@ -511,9 +527,65 @@ class ModelRunner:
             self._graphs[bs] = g
             buf["logits"][bs] = logits
             buf["sampled"][bs] = sampled
+        self._graph_pool = pool
         self._graph_batch_sizes = sorted(self._graphs.keys())
         torch.cuda.synchronize()
         logger.info("Captured decode hipGraphs for batch sizes %s", self._graph_batch_sizes)
+
+    _SEED_PRIME = 0x9E3779B1
+
+    def _ensure_sampled_graph(self, bs: int) -> torch.cuda.CUDAGraph:
+        """Capture (once, lazily) the sampled-decode variant for a bucket:
+        the fused temperature/top-k/top-p Gumbel sampler runs IN-GRAPH, the
+        seed buffer advances in-replay, and greedy rows are overridden by
+        argmax — so multi-step windows work for temperature>0 batches
+        (round-1 limitation: windows were greedy-only)."""
+        g = self._sampled_graphs.get(bs)
+        if g is not None:
+            return g
+        from kserve_amd import ops as _ops
+
+        buf = self._graph_buffers
+        bsz = self.config.cache.block_size
+
+        def body():
+            pos = buf["positions"][:bs]
+            bt = buf["block_tables"][:bs]
+            blk = torch.gather(bt, 1, (pos // bsz).unsqueeze(1)).squeeze(1)
+            slot = (blk * bsz + (pos % bsz).to(torch.int32)).to(torch.int32)
+            meta = AttentionMetadata(
+                is_prefill=False,
+                slot_mapping=slot,
+                block_tables=bt,
+                context_lens=buf["context_lens"][:bs],
+            )
+            hidden = self.model(
+                buf["input_ids"][:bs], pos, self.kv_caches, meta
+            )
+            logits = self.model.compute_logits(hidden)
+            greedy = _ops.greedy_sample(logits)
+            rand = torch.empty(bs, dtype=torch.int64, device=logits.device)
+            _ops.topk_topp_sample_into(
+                rand, logits, buf["temps"][:bs], buf["top_p"][:bs],
+                buf["top_k"][:bs], buf["seeds"][:bs],
+            )
+            sampled = torch.where(buf["greedy_mask"][:bs], greedy, rand)
+            buf["input_ids"][:bs].copy_(sampled)
+            buf["positions"][:bs].add_(1)
+            buf["context_lens"][:bs].add_(1)
+            # fresh Gumbel noise next replay, zero host involvement
+            buf["seeds"][:bs].add_(self._SEED_PRIME)
+            return sampled
+
+        body()
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g, pool=self._graph_pool):
+            sampled = body()
+        self._sampled_graphs[bs] = g
+        buf["sampled_rand"][bs] = sampled
+        logger.info("Captured sampled-decode hipGraph for batch size %d", bs)
+        return g
 
     def _graph_bucket(self, n: int) -> Optional[int]:
         for b in self._graph_batch_sizes:
@@ -565,19 +637,74 @@ class ModelRunner:
         )
         return n
 
+    def _stage_sampling_params(self, bucket: int, requests) -> None:
+        """Stage per-request sampling params into the static graph buffers
+        (pinned → device, one async copy per tensor)."""
+        import numpy as np
+
+        if self._sample_pin is None or self._sample_pin["temps"].shape[0] < bucket:
+            max_bs = self._graph_buffers["temps"].shape[0]
+            self._sample_pin = {
+                "temps": torch.ones(max_bs, dtype=torch.float32, pin_memory=True),
+                "top_p": torch.ones(max_bs, dtype=torch.float32, pin_memory=True),
+                "top_k": torch.zeros(max_bs, dtype=torch.int32, pin_memory=True),
+                "seeds": torch.zeros(max_bs, dtype=torch.int64, pin_memory=True),
+                "greedy_mask": torch.zeros(max_bs, dtype=torch.bool, pin_memory=True),
+            }
+        pin = self._sample_pin
+        tn = pin["temps"].numpy()
+        pn = pin["top_p"].numpy()
+        kn = pin["top_k"].numpy()
+        sn = pin["seeds"].numpy()
+        gn = pin["greedy_mask"].numpy()
+        step_off = self._rng_step * self._SEED_PRIME
+        for i, r in enumerate(requests):
+            sp = r.sampling_params
+            g = sp.greedy
+            gn[i] = g
+            tn[i] = max(sp.temperature, 1e-5)
+            pn[i] = sp.top_p
+            kn[i] = sp.top_k
+            base = (
+                sp.seed
+                if sp.seed is not None
+                else hash(r.request_id) & 0x7FFFFFFF
+            )
+            sn[i] = (base + step_off) & 0x7FFFFFFFFFFFFFFF
+        n = len(requests)
+        if n < bucket:
+            # padded rows: greedy (cheap, deterministic)
+            gn[n:bucket] = True
+            tn[n:bucket] = 1.0
+            pn[n:bucket] = 1.0
+            kn[n:bucket] = 0
+            sn[n:bucket] = 1
+        buf = self._graph_buffers
+        for key in ("temps", "top_p", "top_k", "seeds", "greedy_mask"):
+            buf[key][:bucket].copy_(pin[key][:bucket], non_blocking=True)
+
     @torch.no_grad()
-    def multi_step_decode(self, batch, block_manager, k_steps: int) -> torch.Tensor:
-        """Run ``k_steps`` greedy decode iterations with the sampled token
-        fed back on-GPU (graph path: pure replays, zero host work in the
-        loop). Caller must have reserved KV capacity for the whole window.
-        Returns host int64 [k_steps, n]."""
+    def multi_step_decode(
+        self, batch, block_manager, k_steps: int, sampled: bool = False
+    ) -> torch.Tensor:
+        """Run ``k_steps`` decode iterations with the sampled token fed back
+        on-GPU (graph path: pure replays, zero host work in the loop).
+        ``sampled=True`` uses the in-graph temperature/top-k/top-p variant;
+        default is the greedy graph. Caller must have reserved KV capacity
+        for the whole window. Returns host int64 [k_steps, n]."""
         n = len(batch.requests)
         bucket = None
         if self.is_cuda and self._graphs and not self.config.enforce_eager:
             bucket = self._graph_bucket(n)
         if bucket is not None:
+            if sampled:
+                g = self._ensure_sampled_graph(bucket)
+                self._stage_sampling_params(bucket, batch.requests)
+                out_dev = self._graph_buffers["sampled_rand"][bucket]
+            else:
+                g = self._graphs[bucket]
+                out_dev = self._graph_buffers["sampled"][bucket]
             self._stage_graph_inputs(bucket, batch, block_manager)
-            sampled = self._graph_buffers["sampled"][bucket]
             if (
                 self._window_host is None
                 or self._window_host.shape[0] < k_steps
@@ -589,11 +716,12 @@ class ModelRunner:
                     pin_memory=True,
                 )
             out_host = self._window_host[:k_steps, :n]
-            g = self._graphs[bucket]
             for k in range(k_steps):
                 g.replay()
-                out_host[k].copy_(sampled[:n], non_blocking=True)
+                out_host[k].copy_(out_dev[:n], non_blocking=True)
             torch.cuda.synchronize()
+            if sampled:
+                self._rng_step += k_steps
             return out_host
         # eager fallback (CPU tests / enforce_eager): same semantics
         from kserve_amd import ops as _ops
@@ -603,6 +731,36 @@ class ModelRunner:
         bt = meta.block_tables
         ctx = meta.context_lens
         out = torch.empty((k_steps, n), dtype=torch.int64)
+        dev = input_ids.device
+        if sampled:
+            reqs = batch.requests
+            temps = torch.tensor(
+                [max(r.sampling_params.temperature, 1e-5) for r in reqs],
+                dtype=torch.float32, device=dev,
+            )
+            top_p = torch.tensor(
+                [r.sampling_params.top_p for r in reqs],
+                dtype=torch.float32, device=dev,
+            )
+            top_k = torch.tensor(
+                [r.sampling_params.top_k for r in reqs],
+                dtype=torch.int32, device=dev,
+            )
+            gmask = torch.tensor(
+                [r.sampling_params.greedy for r in reqs],
+                dtype=torch.bool, device=dev,
+            )
+            gen = torch.Generator()
+            gen.manual_seed(self.config.seed + self._rng_step)
+            seed_base = torch.tensor(
+                [
+                    (r.sampling_params.seed
+                     if r.sampling_params.seed is not None
+                     else hash(r.request_id) & 0x7FFFFFFF)
+                    for r in reqs
+                ],
+                dtype=torch.int64, device=dev,
+            )
         for k in range(k_steps):
             blk = torch.gather(bt, 1, (positions // bsz).unsqueeze(1)).squeeze(1)
             slot = (blk * bsz + (positions % bsz).to(torch.int32)).to(torch.int32)
@@ -614,9 +772,19 @@ class ModelRunner:
             )
             hidden = self.model(input_ids, positions, self.kv_caches, meta_k)
             logits = self.model.compute_logits(hidden)
-            sampled = _ops.greedy_sample(logits)
-            out[k] = sampled.cpu()
-            input_ids = sampled.to(input_ids.device)
+            if sampled:
+                seeds = seed_base + (self._rng_step + k) * self._SEED_PRIME
+                rand = _ops.random_sample(
+                    logits, temps, top_p, top_k, seeds=seeds, generator=gen
+                )
+                greedy = _ops.greedy_sample(logits)
+                tok = torch.where(gmask, greedy, rand)
+            else:
+                tok = _ops.greedy_sample(logits)
+            out[k] = tok.cpu()
+            input_ids = tok.to(dev)
             positions = positions + 1
             ctx = ctx + 1
+        if sampled:
+            self._rng_step += k_steps
         return out

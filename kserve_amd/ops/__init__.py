@@ -264,3 +264,22 @@ def flash_attn_varlen(q, k, v, cu_seqlens, max_seqlen, scale, causal=True):
         )
         return out
     return torch_ref.flash_prefill_varlen(q, k, v, cu_seqlens, scale, causal=causal)
+
+
+def topk_topp_sample_into(
+    out: torch.Tensor,
+    logits: torch.Tensor,
+    temperatures: torch.Tensor,
+    top_p: torch.Tensor,
+    top_k: torch.Tensor,
+    seeds: torch.Tensor,
+) -> torch.Tensor:
+    """Destination-passing form of the fused sampler for hipGraph capture
+    (static out buffer, per-row device seeds; see model_runner
+    _ensure_sampled_graph). GPU-only."""
+    _native("random_sample").topk_topp_sample(
+        out, logits, temperatures, top_p,
+        top_k if top_k.dtype == torch.int32 else top_k.to(torch.int32),
+        seeds,
+    )
+    return out

@@ -292,3 +292,81 @@ def test_logit_bias(engine):
         ),
     )
     assert list(banned.values())[0].output_token_ids[0] != g_tok
+
+
+class TestSampledWindows:
+    """Multi-step decode windows at temperature>0 (round-2: the in-graph
+    fused sampler removes the greedy-only window restriction; CPU exercises
+    the eager path of ModelRunner.multi_step_decode(sampled=True))."""
+
+    def _engine(self):
+        torch.manual_seed(0)
+        cfg = EngineConfig(
+            model=ModelConfig.tiny(vocab_size=128),
+            cache=CacheConfig(block_size=4, num_gpu_blocks=256),
+            scheduler=SchedulerConfig(
+                max_num_seqs=8,
+                max_num_batched_tokens=256,
+                max_model_len=128,
+                multi_step=4,
+            ),
+            device="cpu",
+            eos_token_id=-1,
+        )
+        return LLMEngine(cfg)
+
+    def test_sampled_batch_uses_windows(self):
+        eng = self._engine()
+        sp = SamplingParams(temperature=0.8, top_p=0.9, max_tokens=12,
+                            seed=7, ignore_eos=True)
+        for i in range(3):
+            eng.add_request([1 + i, 2, 3], sp, request_id=f"s{i}")
+        steps = 0
+        outs = {}
+        while eng.scheduler.has_unfinished() and steps < 40:
+            for o in eng.step():
+                outs.setdefault(o.request_id, []).extend(o.new_token_ids)
+            steps += 1
+        assert all(len(v) == 12 for v in outs.values())
+        # windows mean fewer engine steps than tokens: 1 prefill + ceil(12/4)
+        # window steps (vs 12 single-token steps without windows)
+        assert steps <= 1 + 3 + 2
+
+    def test_mixed_greedy_and_sampled_window(self):
+        eng = self._engine()
+        greedy = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+        samp = SamplingParams(temperature=1.0, max_tokens=8, seed=3,
+                              ignore_eos=True)
+        eng.add_request([5, 6, 7], greedy, request_id="g")
+        eng.add_request([5, 6, 7], samp, request_id="s")
+        outs = {}
+        steps = 0
+        while eng.scheduler.has_unfinished() and steps < 40:
+            for o in eng.step():
+                outs.setdefault(o.request_id, []).extend(o.new_token_ids)
+            steps += 1
+        assert len(outs["g"]) == 8 and len(outs["s"]) == 8
+        # the greedy row must match a pure-greedy run exactly
+        ref_eng = self._engine()
+        ref = ref_eng.generate([[5, 6, 7]], greedy)
+        assert outs["g"] == list(ref.values())[0].output_token_ids
+
+    def test_penalties_fall_back_to_stepwise(self):
+        """Requests with penalties can't window (host must see history);
+        they still produce max_tokens tokens via single steps."""
+        eng = self._engine()
+        sp = SamplingParams(
+            temperature=0.8, max_tokens=6, presence_penalty=0.5,
+            ignore_eos=True,
+        )
+        eng.add_request([9, 9, 9], sp, request_id="p")
+        outs = []
+        steps = 0
+        while eng.scheduler.has_unfinished() and steps < 40:
+            for o in eng.step():
+                outs.extend(o.new_token_ids)
+            steps += 1
+        assert len(outs) == 6
+        # no windows: one token per step (prefill samples the first token,
+        # then 5 single-token decode steps)
+        assert steps >= 6
