@@ -54,7 +54,8 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     // split-context: partial output when n_splits > 1
     const int n_splits,
     float* __restrict__ part_out,  // [S, H, n_splits, D]
-    float* __restrict__ part_ml    // [S, H, n_splits, 2] (m, l)
+    float* __restrict__ part_ml,   // [S, H, n_splits, 2] (m, l)
+    const int window = 0           // sliding window (0 = full attention)
 ) {
   constexpr int ACC = D / 64;
   const int kv_head = blockIdx.x;
@@ -71,8 +72,12 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 
   const int nblocks = (ctx + PAGE - 1) / PAGE;
   // split-context range
-  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
-  const int blk_lo = split * blocks_per_split;
+  // sliding window: only pages holding the last `window` tokens are read
+  const int wstart = (window > 0 && ctx > window) ? ctx - window : 0;
+  const int first_blk = wstart / PAGE;
+  const int blocks_per_split =
+      (nblocks - first_blk + n_splits - 1) / n_splits;
+  const int blk_lo = first_blk + split * blocks_per_split;
   const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
   if (blk_lo >= blk_hi) {
     // record empty partial so the reducer can skip it
@@ -143,7 +148,7 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     const short8_t* kp =
         reinterpret_cast<const short8_t*>(page + tok * D + part * QFRAG);
     const int gtok = bi * PAGE + tok;
-    const bool tok_valid = gtok < ctx;
+    const bool tok_valid = gtok < ctx && gtok >= wstart;
     const short* vpage =
         v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
 
@@ -416,7 +421,8 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
     const int* __restrict__ block_tables, const int* __restrict__ context_lens,
     const float scale, const int num_kv_heads, const int group,
     const int max_blocks, const long q_row_stride, const int n_splits,
-    float* __restrict__ part_out, float* __restrict__ part_ml) {
+    float* __restrict__ part_out, float* __restrict__ part_ml,
+    const int window = 0) {
   constexpr int ACC = D / 64;
   constexpr int QFRAG = D / 4;
   const int kv_head = blockIdx.x;
@@ -431,8 +437,12 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
   const int part = lane & 3;
 
   const int nblocks = (ctx + PAGE - 1) / PAGE;
-  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
-  const int blk_lo = split * blocks_per_split;
+  // sliding window: only pages holding the last `window` tokens are read
+  const int wstart = (window > 0 && ctx > window) ? ctx - window : 0;
+  const int first_blk = wstart / PAGE;
+  const int blocks_per_split =
+      (nblocks - first_blk + n_splits - 1) / n_splits;
+  const int blk_lo = first_blk + split * blocks_per_split;
   const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
   const int head = kv_head * group + wave;
   const bool active = wave < group;
@@ -473,7 +483,7 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
     const unsigned char* vpage =
         v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
     const int gtok = bi * PAGE + tok;
-    const bool tok_valid = gtok < ctx;
+    const bool tok_valid = gtok < ctx && gtok >= wstart;
     // burst-issue: K quarter-row (32 B) + V 16-B row chunks (2 tokens)
     uint4_t kreg[QFRAG / 16];
 #pragma unroll
@@ -571,7 +581,8 @@ __global__ __launch_bounds__(256) void paged_attention_u2_kernel(
     const int* __restrict__ block_tables, const int* __restrict__ context_lens,
     const float scale, const int num_kv_heads, const int group,
     const int max_blocks, const long q_row_stride, const int n_splits,
-    float* __restrict__ part_out, float* __restrict__ part_ml) {
+    float* __restrict__ part_out, float* __restrict__ part_ml,
+    const int window = 0) {
   constexpr int ACC = D / 64;
   constexpr int QFRAG = D / 4;
   const int kv_head = blockIdx.x;
@@ -586,8 +597,12 @@ __global__ __launch_bounds__(256) void paged_attention_u2_kernel(
   const int part = lane & 3;
 
   const int nblocks = (ctx + PAGE - 1) / PAGE;
-  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
-  const int blk_lo = split * blocks_per_split;
+  // sliding window: only pages holding the last `window` tokens are read
+  const int wstart = (window > 0 && ctx > window) ? ctx - window : 0;
+  const int first_blk = wstart / PAGE;
+  const int blocks_per_split =
+      (nblocks - first_blk + n_splits - 1) / n_splits;
+  const int blk_lo = first_blk + split * blocks_per_split;
   const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
   if (blk_lo >= blk_hi) {
     if (n_splits > 1 && threadIdx.x < (unsigned)group) {
@@ -651,7 +666,7 @@ __global__ __launch_bounds__(256) void paged_attention_u2_kernel(
     }
     s = group_reduce_sum<4>(s);
     const int gtok = bi * PAGE + tok;
-    s = (gtok < ctx) ? s * scale : NEG_INF;
+    s = (gtok < ctx && gtok >= wstart) ? s * scale : NEG_INF;
     const float tmax = wave_reduce_max(s);
     const float m_new = fmaxf(m, tmax);
     const float rescale = __expf(m - m_new);
@@ -721,7 +736,8 @@ __global__ __launch_bounds__(256) void paged_attention_ws_kernel(
     const int* __restrict__ block_tables, const int* __restrict__ context_lens,
     const float scale, const int num_kv_heads, const int max_blocks,
     const long q_row_stride, const int n_splits,
-    float* __restrict__ part_out, float* __restrict__ part_ml) {
+    float* __restrict__ part_out, float* __restrict__ part_ml,
+    const int window = 0) {
   constexpr int ACC = D / 64;
   constexpr int QFRAG = D / 4;
   const int kv_head = blockIdx.x;
@@ -736,8 +752,12 @@ __global__ __launch_bounds__(256) void paged_attention_ws_kernel(
   const int part = lane & 3;
 
   const int nblocks = (ctx + PAGE - 1) / PAGE;
-  const int blocks_per_split = (nblocks + n_splits - 1) / n_splits;
-  const int blk_lo = split * blocks_per_split;
+  // sliding window: only pages holding the last `window` tokens are read
+  const int wstart = (window > 0 && ctx > window) ? ctx - window : 0;
+  const int first_blk = wstart / PAGE;
+  const int blocks_per_split =
+      (nblocks - first_blk + n_splits - 1) / n_splits;
+  const int blk_lo = first_blk + split * blocks_per_split;
   const int blk_hi = min(nblocks, blk_lo + blocks_per_split);
 
   __shared__ float lds_ml[NWAVES][GROUP][2];
@@ -793,7 +813,7 @@ __global__ __launch_bounds__(256) void paged_attention_ws_kernel(
       }
     }
     const int gtok = bi * PAGE + tok;
-    const bool tok_valid = gtok < ctx;
+    const bool tok_valid = gtok < ctx && gtok >= wstart;
 #pragma unroll 1
     for (int g = 0; g < GROUP; ++g) {
       float s = 0.f;
@@ -949,7 +969,7 @@ extern "C" hipError_t ks_paged_attention_decode_fp8(
     const void* block_tables, const void* context_lens, float scale,
     int num_seqs, int num_heads, int num_kv_heads, int head_dim,
     int max_blocks, int block_size, long q_row_stride, int n_splits,
-    void* part_out, void* part_ml, hipStream_t stream) {
+    void* part_out, void* part_ml, int window, hipStream_t stream) {
   if (block_size != PAGE || head_dim != 128) return hipErrorInvalidValue;
   const int group = num_heads / num_kv_heads;
   if (group > NWAVES) return hipErrorInvalidValue;  // fp8: GQA group <= 4
@@ -961,7 +981,7 @@ extern "C" hipError_t ks_paged_attention_decode_fp8(
                      (const unsigned char*)v_cache,
                      (const int*)block_tables, (const int*)context_lens,
                      scale, num_kv_heads, group, max_blocks, q_row_stride,
-                     n_splits, (float*)part_out, (float*)part_ml);
+                     n_splits, (float*)part_out, (float*)part_ml, window);
   HIP_CHECK_KERNEL();
   return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
                              head_dim, n_splits, stream);
@@ -972,7 +992,7 @@ extern "C" hipError_t ks_paged_attention_decode(
     const void* block_tables, const void* context_lens, float scale,
     int num_seqs, int num_heads, int num_kv_heads, int head_dim,
     int max_blocks, int block_size, long q_row_stride, int n_splits,
-    void* part_out, void* part_ml, hipStream_t stream) {
+    void* part_out, void* part_ml, int window, hipStream_t stream) {
   if (block_size != PAGE) return hipErrorInvalidValue;
   const int group = num_heads / num_kv_heads;
   const int hpw = (group + NWAVES - 1) / NWAVES;  // q heads per wave
@@ -998,7 +1018,7 @@ extern "C" hipError_t ks_paged_attention_decode(
                      (const short*)k_cache, (const short*)v_cache,          \
                      (const int*)block_tables, (const int*)context_lens,    \
                      scale, num_kv_heads, max_blocks, q_row_stride,         \
-                     n_splits, (float*)part_out, (float*)part_ml)
+                     n_splits, (float*)part_out, (float*)part_ml, window)
     if (group == 1) LAUNCH_WS(1);
     else if (group == 2) LAUNCH_WS(2);
     else if (group == 4) LAUNCH_WS(4);
@@ -1021,7 +1041,7 @@ extern "C" hipError_t ks_paged_attention_decode(
                        (const short*)k_cache, (const short*)v_cache,
                        (const int*)block_tables, (const int*)context_lens,
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
-                       n_splits, (float*)part_out, (float*)part_ml);
+                       n_splits, (float*)part_out, (float*)part_ml, window);
     HIP_CHECK_KERNEL();
     return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
                                head_dim, n_splits, stream);
@@ -1038,7 +1058,7 @@ extern "C" hipError_t ks_paged_attention_decode(
                        (const short*)k_cache, (const short*)v_cache,
                        (const int*)block_tables, (const int*)context_lens,
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
-                       n_splits, (float*)part_out, (float*)part_ml);
+                       n_splits, (float*)part_out, (float*)part_ml, window);
     HIP_CHECK_KERNEL();
     return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
                                head_dim, n_splits, stream);
@@ -1081,7 +1101,7 @@ extern "C" hipError_t ks_paged_attention_decode(
           0, stream, (short*)out, (const short*)q, (const short*)k_cache,
           (const short*)v_cache, (const int*)block_tables,
           (const int*)context_lens, scale, num_kv_heads, group, max_blocks,
-          q_row_stride, n_splits, (float*)part_out, (float*)part_ml);
+          q_row_stride, n_splits, (float*)part_out, (float*)part_ml, window);
     } else {
       hipLaunchKernelGGL(
           (paged_attention_kernel<128, 1, 1, true, false, true>), grid,
@@ -1107,7 +1127,7 @@ extern "C" hipError_t ks_paged_attention_decode(
                        (const short*)k_cache, (const short*)v_cache,
                        (const int*)block_tables, (const int*)context_lens,
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
-                       n_splits, (float*)part_out, (float*)part_ml);
+                       n_splits, (float*)part_out, (float*)part_ml, window);
     HIP_CHECK_KERNEL();
     return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
                                head_dim, n_splits, stream);
@@ -1118,7 +1138,7 @@ extern "C" hipError_t ks_paged_attention_decode(
                        (const short*)k_cache, (const short*)v_cache,
                        (const int*)block_tables, (const int*)context_lens,
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
-                       n_splits, (float*)part_out, (float*)part_ml);
+                       n_splits, (float*)part_out, (float*)part_ml, window);
     HIP_CHECK_KERNEL();
     return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
                                head_dim, n_splits, stream);
@@ -1130,14 +1150,14 @@ extern "C" hipError_t ks_paged_attention_decode(
                          (const short*)k_cache, (const short*)v_cache,
                          (const int*)block_tables, (const int*)context_lens,
                          scale, num_kv_heads, group, max_blocks, q_row_stride,
-                         n_splits, (float*)part_out, (float*)part_ml);
+                         n_splits, (float*)part_out, (float*)part_ml, window);
     } else {
       hipLaunchKernelGGL((paged_attention_kernel<128, 1, 5>), grid, block, 0,
                          stream, (short*)out, (const short*)q,
                          (const short*)k_cache, (const short*)v_cache,
                          (const int*)block_tables, (const int*)context_lens,
                          scale, num_kv_heads, group, max_blocks, q_row_stride,
-                         n_splits, (float*)part_out, (float*)part_ml);
+                         n_splits, (float*)part_out, (float*)part_ml, window);
     }
     HIP_CHECK_KERNEL();
     return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
@@ -1149,7 +1169,7 @@ extern "C" hipError_t ks_paged_attention_decode(
                      (const short*)k_cache, (const short*)v_cache,         \
                      (const int*)block_tables, (const int*)context_lens,   \
                      scale, num_kv_heads, group, max_blocks, q_row_stride, \
-                     n_splits, (float*)part_out, (float*)part_ml)
+                     n_splits, (float*)part_out, (float*)part_ml, window)
   if (head_dim == 128) {
     if (hpw == 1) LAUNCH_PA(128, 1);
     else LAUNCH_PA(128, 2);
