@@ -55,6 +55,9 @@ class LlamaAttention(nn.Module):
         st = comm.get_state()
         self.head_dim = config.head_dim
         self.scale = 1.0 / math.sqrt(config.head_dim)
+        # Mistral-style sliding window (0/None = full attention); applied
+        # in-kernel as a lower bound on attended positions
+        self.sliding_window = int(getattr(config, "sliding_window", 0) or 0)
         self.qkv_proj = QKVParallelLinear(
             config.hidden_size,
             config.head_dim,
@@ -100,15 +103,18 @@ class LlamaAttention(nn.Module):
                 # chunks' KV + this chunk's, written to pages above)
                 out = ops.context_attention_varlen(
                     q, k_cache, v_cache, meta.block_tables, meta.cu_seqlens,
-                    meta.context_lens, meta.max_seqlen, self.scale
+                    meta.context_lens, meta.max_seqlen, self.scale,
+                    window=self.sliding_window,
                 )
             else:
                 out = ops.flash_prefill_varlen(
-                    q, k, v, meta.cu_seqlens, meta.max_seqlen, self.scale
+                    q, k, v, meta.cu_seqlens, meta.max_seqlen, self.scale,
+                    window=self.sliding_window,
                 )
         else:
             out = ops.paged_attention_decode(
-                q, k_cache, v_cache, meta.block_tables, meta.context_lens, self.scale
+                q, k_cache, v_cache, meta.block_tables, meta.context_lens,
+                self.scale, window=self.sliding_window,
             )
         o_in = out.reshape(T, -1)
         o_delta = (

@@ -98,16 +98,19 @@ def paged_attention_decode(
     context_lens: torch.Tensor,
     scale: float,
     out: Optional[torch.Tensor] = None,
+    window: int = 0,
 ) -> torch.Tensor:
     if q.is_cuda:
         if out is None:
             out = torch.empty_like(q)
         _native("paged_attention_decode").paged_attention_decode(
-            out, q, k_cache, v_cache, block_tables, context_lens, scale
+            out, q, k_cache, v_cache, block_tables, context_lens, scale,
+            window,
         )
         return out
     return torch_ref.paged_attention_decode(
-        q, k_cache, v_cache, block_tables, context_lens, scale
+        q, k_cache, v_cache, block_tables, context_lens, scale,
+        window=window,
     )
 
 
@@ -118,14 +121,18 @@ def flash_prefill_varlen(
     cu_seqlens: torch.Tensor,
     max_seqlen: int,
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     if q.is_cuda:
         out = torch.empty_like(q)
         _native("flash_prefill").flash_prefill_varlen(
-            out, q, k, v, cu_seqlens, int(max_seqlen), scale
+            out, q, k, v, cu_seqlens, int(max_seqlen), scale,
+            causal=True, window=window,
         )
         return out
-    return torch_ref.flash_prefill_varlen(q, k, v, cu_seqlens, scale, causal=True)
+    return torch_ref.flash_prefill_varlen(
+        q, k, v, cu_seqlens, scale, causal=True, window=window
+    )
 
 
 def context_attention_varlen(
@@ -137,19 +144,22 @@ def context_attention_varlen(
     context_lens: torch.Tensor,
     max_q_len: int,
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     """Prefill attention against the paged cache (chunked prefill): the
     chunk's KV must already be written to the cache; queries attend causally
-    to the whole per-seq context."""
+    to the whole per-seq context (bounded below by the sliding window when
+    window > 0)."""
     if q.is_cuda:
         out = torch.empty_like(q)
         _native("context_prefill").context_prefill_varlen(
             out, q, k_cache, v_cache, block_tables, context_lens,
-            cu_seqlens_q, int(max_q_len), scale
+            cu_seqlens_q, int(max_q_len), scale, window
         )
         return out
     return torch_ref.context_attention_varlen(
-        q, k_cache, v_cache, block_tables, cu_seqlens_q, context_lens, scale
+        q, k_cache, v_cache, block_tables, cu_seqlens_q, context_lens, scale,
+        window=window,
     )
 
 

@@ -40,7 +40,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
     const short* __restrict__ v,  // [T, Hkv, D]
     const int* __restrict__ cu_seqlens,  // [S+1]
     const int Hq, const int Hkv, const float scale, const long sq,
-    const long sk, const long sv) {
+    const long sk, const long sv, const int window) {
   const int head = blockIdx.x;
   const int tile = blockIdx.y;
   const int seq = blockIdx.z;
@@ -140,7 +140,8 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
       for (int n = 0; n < 2; ++n) {
         const int tok = t0 + n * 16 + MFMA_C_COL(lane);
         const bool valid =
-            (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len);
+            (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len) &&
+            (window <= 0 || tok > qrow - window);
         sv[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
       }
       float rowmax = group_reduce_max<16>(fmaxf(sv[0], sv[1]));
@@ -223,7 +224,8 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
     short* __restrict__ out, const short* __restrict__ q,
     const short* __restrict__ k, const short* __restrict__ v,
     const int* __restrict__ cu_seqlens, const int Hq, const int Hkv,
-    const float scale, const long sq, const long sk, const long sv) {
+    const float scale, const long sq, const long sk, const long sv,
+    const int window) {
   const int head = blockIdx.x;
   const int tile = blockIdx.y;
   const int seq = blockIdx.z;
@@ -330,7 +332,8 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
         for (int n = 0; n < KT2 / 16; ++n) {
           const int tok = t0 + n * 16 + MFMA_C_COL(lane);
           const bool valid =
-              (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len);
+              (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len) &&
+              (window <= 0 || tok > qrow - window);
           sv_[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
           rowmax = fmaxf(rowmax, sv_[n]);
         }
@@ -431,7 +434,7 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
     const int* __restrict__ ctx_lens,      // [S] total context incl. chunk
     const int* __restrict__ cu_seqlens_q,  // [S+1]
     const int Hq, const int Hkv, const int max_nb, const float scale,
-    const long sq) {
+    const long sq, const int window) {
   const int head = blockIdx.x;
   const int tile = blockIdx.y;
   const int seq = blockIdx.z;
@@ -573,7 +576,8 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
 #pragma unroll
         for (int n = 0; n < KT2 / 16; ++n) {
           const int tok = t0 + n * 16 + MFMA_C_COL(lane);
-          const bool valid = (tok <= qpos) && (qrow < q_len) && (tok < ctx);
+          const bool valid = (tok <= qpos) && (qrow < q_len) && (tok < ctx) &&
+                             (window <= 0 || tok > qpos - window);
           sv_[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
           rowmax = fmaxf(rowmax, sv_[n]);
         }
@@ -658,7 +662,7 @@ extern "C" hipError_t ks_context_prefill_varlen(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const void* block_tables, const void* ctx_lens, const void* cu_seqlens_q,
     int num_seqs, int max_q_len, int Hq, int Hkv, int head_dim, int max_nb,
-    float scale, long sq, int fp8_cache, hipStream_t stream) {
+    float scale, long sq, int fp8_cache, int window, hipStream_t stream) {
   if (head_dim != 128) return hipErrorInvalidValue;
   if (Hq % Hkv != 0) return hipErrorInvalidValue;
   const int max_tiles = (max_q_len + NWAVES * QW2 - 1) / (NWAVES * QW2);
@@ -670,13 +674,15 @@ extern "C" hipError_t ks_context_prefill_varlen(
                        (const unsigned char*)k_cache,
                        (const unsigned char*)v_cache,
                        (const int*)block_tables, (const int*)ctx_lens,
-                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq);
+                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq,
+                       window);
   } else {
     hipLaunchKernelGGL((context_prefill_kernel<128, short>), grid, dim3(256),
                        0, stream, (short*)out, (const short*)q,
                        (const short*)k_cache, (const short*)v_cache,
                        (const int*)block_tables, (const int*)ctx_lens,
-                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq);
+                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq,
+                       window);
   }
   HIP_CHECK_KERNEL();
   return hipSuccess;
@@ -686,7 +692,7 @@ extern "C" hipError_t ks_flash_prefill_varlen(
     void* out, const void* q, const void* k, const void* v,
     const void* cu_seqlens, int num_seqs, int max_seqlen, int Hq, int Hkv,
     int head_dim, float scale, long sq, long sk, long sv, int causal,
-    hipStream_t stream) {
+    int window, hipStream_t stream) {
   if (head_dim != 128 && head_dim != 64) return hipErrorInvalidValue;
   if (Hq % Hkv != 0) return hipErrorInvalidValue;
   const int max_tiles = (max_seqlen + NWAVES * QW - 1) / (NWAVES * QW);
@@ -704,12 +710,14 @@ extern "C" hipError_t ks_flash_prefill_varlen(
       hipLaunchKernelGGL((flash_prefill_v2_kernel<128, true>), grid2,
                          dim3(256), 0, stream, (short*)out, (const short*)q,
                          (const short*)k, (const short*)v,
-                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv);
+                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
+                         window);
     } else {
       hipLaunchKernelGGL((flash_prefill_v2_kernel<128, false>), grid2,
                          dim3(256), 0, stream, (short*)out, (const short*)q,
                          (const short*)k, (const short*)v,
-                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv);
+                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
+                         window);
     }
     HIP_CHECK_KERNEL();
     return hipSuccess;
@@ -718,7 +726,7 @@ extern "C" hipError_t ks_flash_prefill_varlen(
   hipLaunchKernelGGL((flash_prefill_kernel<DD, CC>), grid, dim3(256), 0,     \
                      stream, (short*)out, (const short*)q, (const short*)k,  \
                      (const short*)v, (const int*)cu_seqlens, Hq, Hkv,       \
-                     scale, sq, sk, sv)
+                     scale, sq, sk, sv, window)
   if (head_dim == 128) {
     if (causal) LAUNCH_FP(128, true);
     else LAUNCH_FP(128, false);

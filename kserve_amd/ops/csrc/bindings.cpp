@@ -21,20 +21,22 @@ hipError_t ks_reshape_and_cache(const void*, const void*, void*, void*,
 hipError_t ks_paged_attention_decode(void*, const void*, const void*,
                                      const void*, const void*, const void*,
                                      float, int, int, int, int, int, int,
-                                     long, int, void*, void*, hipStream_t);
+                                     long, int, void*, void*, int,
+                                     hipStream_t);
 hipError_t ks_flash_prefill_varlen(void*, const void*, const void*,
                                    const void*, const void*, int, int, int,
                                    int, int, float, long, long, long, int,
-                                   hipStream_t);
+                                   int, hipStream_t);
 hipError_t ks_context_prefill_varlen(void*, const void*, const void*,
                                      const void*, const void*, const void*,
                                      const void*, int, int, int, int, int,
-                                     int, float, long, int, hipStream_t);
+                                     int, float, long, int, int,
+                                     hipStream_t);
 hipError_t ks_paged_attention_decode_fp8(void*, const void*, const void*,
                                          const void*, const void*,
                                          const void*, float, int, int, int,
                                          int, int, int, long, int, void*,
-                                         void*, hipStream_t);
+                                         void*, int, hipStream_t);
 hipError_t ks_reshape_and_cache_fp8(const void*, const void*, void*, void*,
                                     const void*, int, int, int, int, long,
                                     long, hipStream_t);
@@ -173,7 +175,7 @@ void reshape_and_cache(at::Tensor& k, at::Tensor& v, at::Tensor& k_cache,
 void paged_attention_decode(at::Tensor& out, at::Tensor& q,
                             at::Tensor& k_cache, at::Tensor& v_cache,
                             at::Tensor& block_tables, at::Tensor& context_lens,
-                            double scale) {
+                            double scale, int64_t window) {
   CHECK_BF16_CONTIG(out);
   CHECK_BF16_ROWS(q);
   CHECK_KV_CACHE(k_cache);
@@ -213,7 +215,7 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
                   v_cache.data_ptr(), block_tables.data_ptr(),
                   context_lens.data_ptr(), (float)scale, S, H, Hkv, D,
                   max_blocks, block_size, (long)q.stride(0), n_splits, po,
-                  pml, current_stream()),
+                  pml, (int)window, current_stream()),
               "paged_attention_decode_fp8");
     return;
   }
@@ -222,14 +224,14 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
                 v_cache.data_ptr(), block_tables.data_ptr(),
                 context_lens.data_ptr(), (float)scale, S, H, Hkv, D,
                 max_blocks, block_size, (long)q.stride(0), n_splits, po, pml,
-                current_stream()),
+                (int)window, current_stream()),
             "paged_attention_decode");
 }
 
 void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
                           at::Tensor& v, at::Tensor& cu_seqlens,
                           int64_t max_seqlen, double scale,
-                          bool causal = true) {
+                          bool causal = true, int64_t window = 0) {
   CHECK_BF16_CONTIG(out);
   CHECK_BF16_ROWS(q);
   CHECK_BF16_ROWS(k);
@@ -244,7 +246,8 @@ void flash_prefill_varlen(at::Tensor& out, at::Tensor& q, at::Tensor& k,
                                     num_seqs, (int)max_seqlen, Hq, Hkv, D,
                                     (float)scale, (long)q.stride(0),
                                     (long)k.stride(0), (long)v.stride(0),
-                                    causal ? 1 : 0, current_stream()),
+                                    causal ? 1 : 0, (int)window,
+                                    current_stream()),
             "flash_prefill_varlen");
 }
 
@@ -252,7 +255,7 @@ void context_prefill_varlen(at::Tensor& out, at::Tensor& q,
                             at::Tensor& k_cache, at::Tensor& v_cache,
                             at::Tensor& block_tables, at::Tensor& ctx_lens,
                             at::Tensor& cu_seqlens_q, int64_t max_q_len,
-                            double scale) {
+                            double scale, int64_t window) {
   CHECK_BF16_CONTIG(out);
   CHECK_BF16_ROWS(q);
   CHECK_KV_CACHE(k_cache);
@@ -272,7 +275,8 @@ void context_prefill_varlen(at::Tensor& out, at::Tensor& q,
                 v_cache.data_ptr(), block_tables.data_ptr(),
                 ctx_lens.data_ptr(), cu_seqlens_q.data_ptr(), num_seqs,
                 (int)max_q_len, Hq, Hkv, D, (int)block_tables.size(1),
-                (float)scale, (long)q.stride(0), fp8, current_stream()),
+                (float)scale, (long)q.stride(0), fp8, (int)window,
+                current_stream()),
             "context_prefill_varlen");
 }
 
@@ -415,7 +419,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k"),
         pybind11::arg("v"), pybind11::arg("cu_seqlens"),
         pybind11::arg("max_seqlen"), pybind11::arg("scale"),
-        pybind11::arg("causal") = true);
+        pybind11::arg("causal") = true, pybind11::arg("window") = 0);
   m.def("context_prefill_varlen", &context_prefill_varlen,
         "MFMA prefill attention against the paged KV cache (chunked prefill)");
   m.def("layer_norm", &layer_norm, "LayerNorm (bf16)");

@@ -134,6 +134,7 @@ def paged_attention_decode(
     block_tables: torch.Tensor,  # [num_seqs, max_blocks] int32 (padded)
     context_lens: torch.Tensor,  # [num_seqs] int32 — includes current token
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     num_seqs, num_heads, head_dim = q.shape
     num_kv_heads = k_cache.shape[1]
@@ -155,6 +156,8 @@ def paged_attention_decode(
         scores = torch.einsum(
             "hd,htd->ht", qh, keys.float().repeat_interleave(group, dim=0)
         ) * scale
+        if window > 0 and ctx > window:
+            scores[:, : ctx - window] = float("-inf")
         probs = torch.softmax(scores, dim=-1)
         o = torch.einsum(
             "ht,htd->hd", probs, vals.float().repeat_interleave(group, dim=0)
@@ -170,6 +173,7 @@ def flash_prefill_varlen(
     cu_seqlens: torch.Tensor,  # [num_seqs+1] int32
     scale: float,
     causal: bool = True,
+    window: int = 0,
 ) -> torch.Tensor:
     num_heads = q.shape[1]
     num_kv_heads = k.shape[1]
@@ -186,6 +190,12 @@ def flash_prefill_varlen(
             mask = torch.triu(
                 torch.ones(T, T, dtype=torch.bool, device=q.device), diagonal=1
             )
+            if window > 0:
+                # sliding window: token i sees only (i-window, i]
+                mask |= torch.tril(
+                    torch.ones(T, T, dtype=torch.bool, device=q.device),
+                    diagonal=-window,
+                )
             scores.masked_fill_(mask, float("-inf"))
         probs = torch.softmax(scores, dim=-1)
         o = torch.einsum("hts,hsd->htd", probs, vs)
@@ -201,6 +211,7 @@ def context_attention_varlen(
     cu_seqlens_q: torch.Tensor,  # [num_seqs+1] new-token offsets
     context_lens: torch.Tensor,  # [num_seqs] total context (incl. new tokens)
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     """Prefill attention against the paged cache (supports chunked prefill:
     new tokens attend to all cached tokens before them, causally)."""
@@ -228,7 +239,10 @@ def context_attention_varlen(
         start = ctx - n_new
         pos_q = torch.arange(start, ctx, device=q.device).unsqueeze(1)
         pos_k = torch.arange(ctx, device=q.device).unsqueeze(0)
-        scores.masked_fill_((pos_k > pos_q).unsqueeze(0), float("-inf"))
+        bad = pos_k > pos_q
+        if window > 0:
+            bad |= pos_k <= pos_q - window
+        scores.masked_fill_(bad.unsqueeze(0), float("-inf"))
         probs = torch.softmax(scores, dim=-1)
         o = torch.einsum("hts,hsd->htd", probs, vals)
         out[a:b] = o.transpose(0, 1).to(q.dtype)

@@ -80,9 +80,9 @@ def build_model(args):
             model_cfg.expert_parallel = True
         if model_cfg.sliding_window:
             logger.info(
-                "model declares sliding_window=%d; this engine attends the "
-                "full context (a superset of windowed attention)",
-                model_cfg.sliding_window,
+                "sliding_window=%d: attention bounded in-kernel to the last "
+                "%d positions (Mistral-family numerics beyond the window)",
+                model_cfg.sliding_window, model_cfg.sliding_window,
             )
         cfg = EngineConfig(
             model=model_cfg,

@@ -497,3 +497,87 @@ class TestGemm8Experimental:
         torch.testing.assert_close(
             d.float(), ref, atol=K * 2e-3, rtol=3e-2
         )
+
+
+class TestSlidingWindowKernels:
+    """Window-bounded attention vs fp32 torch references (Mistral-family):
+    the HIP kernels take a `window` arg and must match the masked refs for
+    contexts beyond the window, across split-context and variant dispatch."""
+
+    @pytest.mark.parametrize(
+        "S,H,Hkv,ctx_max,W",
+        [
+            (4, 32, 8, 300, 64),    # llama-8b shape, window < ctx
+            (2, 8, 2, 700, 128),    # long ctx, deep split
+            (8, 4, 4, 50, 64),      # window > ctx (no-op)
+            (200, 32, 8, 400, 96),  # V4 large-batch variant
+        ],
+    )
+    def test_decode_window(self, dev, S, H, Hkv, ctx_max, W):
+        from kserve_amd import ops
+
+        D, bs = 128, 16
+        torch.manual_seed(S + W)
+        ctx = torch.randint(max(1, ctx_max // 2), ctx_max + 1, (S,),
+                            dtype=torch.int32)
+        max_blocks = int((int(ctx.max()) + bs - 1) // bs)
+        B = S * max_blocks + 1
+        kc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev)
+        vc = torch.randn(B, Hkv, bs, D, dtype=torch.bfloat16, device=dev)
+        bt = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+            S, max_blocks
+        )
+        q = torch.randn(S, H, D, dtype=torch.bfloat16, device=dev)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.paged_attention_decode(
+            q, kc, vc, bt.to(dev), ctx.to(dev), scale, window=W
+        )
+        ref = torch_ref.paged_attention_decode(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt, ctx,
+            scale, window=W,
+        )
+        bf16_close(got, ref)
+
+    @pytest.mark.parametrize("T,W", [(200, 64), (64, 64), (333, 100)])
+    def test_flash_prefill_window(self, dev, T, W):
+        from kserve_amd import ops
+
+        H, Hkv, D = 8, 2, 128
+        torch.manual_seed(T)
+        q = torch.randn(T, H, D, dtype=torch.bfloat16, device=dev) / 4
+        k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev) / 4
+        v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=dev) / 4
+        cu = torch.tensor([0, T], dtype=torch.int32)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.flash_prefill_varlen(
+            q, k, v, cu.to(dev), T, scale, window=W
+        )
+        ref = torch_ref.flash_prefill_varlen(
+            q.float().cpu(), k.float().cpu(), v.float().cpu(), cu, scale,
+            causal=True, window=W,
+        )
+        bf16_close(got, ref)
+
+    def test_context_prefill_window(self, dev):
+        from kserve_amd import ops
+
+        H, Hkv, D, bs = 8, 2, 128, 16
+        ctx, n_new, W = 150, 30, 48
+        torch.manual_seed(7)
+        nb = -(-ctx // bs)
+        kc = torch.randn(nb + 1, Hkv, bs, D, dtype=torch.bfloat16, device=dev) / 4
+        vc = torch.randn(nb + 1, Hkv, bs, D, dtype=torch.bfloat16, device=dev) / 4
+        bt = torch.arange(1, nb + 1, dtype=torch.int32).unsqueeze(0)
+        q = torch.randn(n_new, H, D, dtype=torch.bfloat16, device=dev) / 4
+        cu_q = torch.tensor([0, n_new], dtype=torch.int32)
+        cl = torch.tensor([ctx], dtype=torch.int32)
+        scale = 1.0 / math.sqrt(D)
+        got = ops.context_attention_varlen(
+            q, kc, vc, bt.to(dev), cu_q.to(dev), cl.to(dev), n_new, scale,
+            window=W,
+        )
+        ref = torch_ref.context_attention_varlen(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt, cu_q,
+            cl, scale, window=W,
+        )
+        bf16_close(got, ref)

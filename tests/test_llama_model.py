@@ -421,3 +421,108 @@ class TestLlama31RopeScaling:
         b = [o.output_token_ids
              for o in LLMEngine(cfg(True)).generate(prompts, sp).values()]
         assert a == b
+
+
+class TestSlidingWindow:
+    """Sliding-window attention (Mistral-family): beyond the window the
+    outputs must match an explicitly-masked full-attention reference, and
+    must DIFFER from unwindowed attention (round-1 recorded the window but
+    attended the full context)."""
+
+    def _refs(self, T=24, W=8, H=2, D=16):
+        torch.manual_seed(0)
+        q = torch.randn(T, H, D)
+        k = torch.randn(T, H, D)
+        v = torch.randn(T, H, D)
+        cu = torch.tensor([0, T], dtype=torch.int32)
+        return q, k, v, cu, W
+
+    def test_flash_prefill_window_masks_lower_positions(self):
+        from kserve_amd.ops import torch_ref
+
+        q, k, v, cu, W = self._refs()
+        out_w = torch_ref.flash_prefill_varlen(q, k, v, cu, 0.25, window=W)
+        out_full = torch_ref.flash_prefill_varlen(q, k, v, cu, 0.25)
+        T = q.shape[0]
+        # inside the window: identical; beyond: different
+        assert torch.allclose(out_w[: W], out_full[: W], atol=1e-6)
+        assert not torch.allclose(out_w[W + 1 :], out_full[W + 1 :])
+        # explicit reference: per-token softmax over its window only
+        for i in (W + 1, T - 1):
+            lo = i - W + 1
+            scores = torch.einsum(
+                "hd,shd->hs", q[i].float(), k[lo : i + 1].float()
+            ) * 0.25
+            probs = torch.softmax(scores, dim=-1)
+            o = torch.einsum("hs,shd->hd", probs, v[lo : i + 1].float())
+            assert torch.allclose(out_w[i].float(), o, atol=1e-5)
+
+    def test_decode_window_matches_masked_reference(self):
+        from kserve_amd.ops import torch_ref
+
+        torch.manual_seed(1)
+        H, Hkv, D, bs = 4, 2, 16, 4
+        nb = 8
+        ctx = 29
+        W = 12
+        q = torch.randn(1, H, D)
+        k_cache = torch.randn(nb + 1, Hkv, bs, D)
+        v_cache = torch.randn(nb + 1, Hkv, bs, D)
+        bt = torch.arange(1, nb + 1, dtype=torch.int32).unsqueeze(0)
+        cl = torch.tensor([ctx], dtype=torch.int32)
+        out_w = torch_ref.paged_attention_decode(
+            q, k_cache, v_cache, bt, cl, 0.25, window=W
+        )
+        out_full = torch_ref.paged_attention_decode(
+            q, k_cache, v_cache, bt, cl, 0.25
+        )
+        assert not torch.allclose(out_w, out_full)
+        # reference: gather the last W tokens only
+        group = H // Hkv
+        keys = k_cache[bt[0].long()].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[
+            :, ctx - W : ctx
+        ].float().repeat_interleave(group, dim=0)
+        vals = v_cache[bt[0].long()].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[
+            :, ctx - W : ctx
+        ].float().repeat_interleave(group, dim=0)
+        scores = torch.einsum("hd,htd->ht", q[0].float(), keys) * 0.25
+        o = torch.einsum(
+            "ht,htd->hd", torch.softmax(scores, dim=-1), vals
+        )
+        assert torch.allclose(out_w[0].float(), o, atol=1e-5)
+
+    def test_context_attention_window(self):
+        from kserve_amd.ops import torch_ref
+
+        torch.manual_seed(2)
+        H, Hkv, D, bs = 2, 1, 16, 4
+        ctx, n_new, W = 20, 4, 6
+        nb = -(-ctx // bs)
+        k_cache = torch.randn(nb + 1, Hkv, bs, D)
+        v_cache = torch.randn(nb + 1, Hkv, bs, D)
+        bt = torch.arange(1, nb + 1, dtype=torch.int32).unsqueeze(0)
+        q = torch.randn(n_new, H, D)
+        cu_q = torch.tensor([0, n_new], dtype=torch.int32)
+        cl = torch.tensor([ctx], dtype=torch.int32)
+        out_w = torch_ref.context_attention_varlen(
+            q, k_cache, v_cache, bt, cu_q, cl, 0.25, window=W
+        )
+        # reference per query row
+        group = H // Hkv
+        keys = k_cache[bt[0].long()].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[
+            :, :ctx
+        ].float().repeat_interleave(group, dim=0)
+        vals = v_cache[bt[0].long()].permute(1, 0, 2, 3).reshape(Hkv, -1, D)[
+            :, :ctx
+        ].float().repeat_interleave(group, dim=0)
+        for r in range(n_new):
+            pos = ctx - n_new + r
+            lo = max(0, pos - W + 1)
+            scores = torch.einsum(
+                "hd,htd->ht", q[r].float(), keys[:, lo : pos + 1]
+            ) * 0.25
+            o = torch.einsum(
+                "ht,htd->hd", torch.softmax(scores, dim=-1),
+                vals[:, lo : pos + 1],
+            )
+            assert torch.allclose(out_w[r].float(), o, atol=1e-5)
