@@ -24,6 +24,12 @@ from kserve_amd.logging import logger
 _STOP = object()
 
 
+def _publish_batch(items):
+    """Runs on the server's event loop: deliver one step's outputs."""
+    for q, out in items:
+        q.put_nowait(out)
+
+
 class AsyncLLMEngine:
     def __init__(self, config: EngineConfig, tokenizer=None, lora_modules=None):
         self.config = config
@@ -124,14 +130,20 @@ class AsyncLLMEngine:
                     pass
                 if self.engine.has_unfinished():
                     outputs = self.engine.step()
+                    # ONE cross-thread callback per event loop per step (a
+                    # per-output call_soon_threadsafe wakes the server loop
+                    # hundreds of times per step at high concurrency)
+                    by_loop: Dict[object, list] = {}
                     for out in outputs:
                         entry = self._streams.get(out.request_id)
                         if entry is None:
                             continue
                         q, loop = entry
-                        loop.call_soon_threadsafe(q.put_nowait, out)
+                        by_loop.setdefault(loop, []).append((q, out))
                         if out.finished:
                             self._streams.pop(out.request_id, None)
+                    for loop, items in by_loop.items():
+                        loop.call_soon_threadsafe(_publish_batch, items)
         except BaseException as e:  # engine loop must fail loudly
             logger.exception("Engine loop died")
             self._dead = e

@@ -161,6 +161,9 @@ class LLMModel(OpenAIModel):
         choices = []
         prompt_tokens = 0
         completion_tokens = 0
+        # submit every (prompt, n) sample concurrently: the engine batches
+        # them in one continuous-batching schedule instead of serial awaits
+        jobs = []
         for i, p in enumerate(prompt_list):
             ids = self._encode_prompt(p)
             prompt_tokens += len(ids)
@@ -174,43 +177,52 @@ class LLMModel(OpenAIModel):
                         n=1,
                         seed=(sp.seed + j) if sp.seed is not None else None,
                     )
-                try:
-                    out = await self.async_engine.generate_full(ids, sp_j)
-                except ValueError as e:
-                    raise InvalidInput(str(e)) from e
-                completion_tokens += len(out.output_token_ids)
-                text = (
-                    out.output_text
-                    if getattr(out, "output_text", None)
-                    else self._decode(out.output_token_ids)
+                jobs.append((i, j, ids, sp_j))
+        import asyncio as _asyncio
+
+        try:
+            outs = await _asyncio.gather(
+                *(
+                    self.async_engine.generate_full(ids, sp_j)
+                    for (_, _, ids, sp_j) in jobs
                 )
-                if request.echo:
-                    text = self._decode(ids) + text
-                lp = None
-                if out.logprobs:
-                    toks = [
-                        self._decode([t]) for t in out.output_token_ids
-                    ]
-                    lp = CompletionLogprobs(
-                        tokens=toks,
-                        token_logprobs=[
-                            step.get(t)
-                            for t, step in zip(out.output_token_ids, out.logprobs)
-                        ],
-                        top_logprobs=[
-                            {self._decode([k]): v for k, v in step.items()}
-                            for step in out.logprobs
-                        ],
-                        text_offset=[],
-                    )
-                choices.append(
-                    CompletionChoice(
-                        index=i * sp.n + j,
-                        text=text,
-                        logprobs=lp,
-                        finish_reason=out.finish_reason or "stop",
-                    )
+            )
+        except ValueError as e:
+            raise InvalidInput(str(e)) from e
+        for (i, j, ids, _), out in zip(jobs, outs):
+            completion_tokens += len(out.output_token_ids)
+            text = (
+                out.output_text
+                if getattr(out, "output_text", None)
+                else self._decode(out.output_token_ids)
+            )
+            if request.echo:
+                text = self._decode(ids) + text
+            lp = None
+            if out.logprobs:
+                toks = [
+                    self._decode([t]) for t in out.output_token_ids
+                ]
+                lp = CompletionLogprobs(
+                    tokens=toks,
+                    token_logprobs=[
+                        step.get(t)
+                        for t, step in zip(out.output_token_ids, out.logprobs)
+                    ],
+                    top_logprobs=[
+                        {self._decode([k]): v for k, v in step.items()}
+                        for step in out.logprobs
+                    ],
+                    text_offset=[],
                 )
+            choices.append(
+                CompletionChoice(
+                    index=i * sp.n + j,
+                    text=text,
+                    logprobs=lp,
+                    finish_reason=out.finish_reason or "stop",
+                )
+            )
         return Completion(
             model=self.name,
             choices=choices,
