@@ -81,6 +81,26 @@ class AsyncLLMEngine:
 
     # -- engine thread ---------------------------------------------------------
     def _run_loop(self):
+        import os
+
+        if os.environ.get("KS_ENGINE_PROFILE") == "1":
+            import cProfile
+
+            prof = cProfile.Profile()
+            try:
+                prof.runcall(self._run_loop_inner)
+            finally:
+                import pstats
+                import sys
+
+                print("==== engine thread profile ====", file=sys.stderr)
+                pstats.Stats(prof, stream=sys.stderr).sort_stats(
+                    "cumulative"
+                ).print_stats(30)
+            return
+        self._run_loop_inner()
+
+    def _run_loop_inner(self):
         try:
             while not self._stopping:
                 # drain submissions; block briefly when idle

@@ -55,7 +55,7 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     const int n_splits,
     float* __restrict__ part_out,  // [S, H, n_splits, D]
     float* __restrict__ part_ml,   // [S, H, n_splits, 2] (m, l)
-    const int window = 0           // sliding window (0 = full attention)
+    const int window               // sliding window (0 = full attention)
 ) {
   constexpr int ACC = D / 64;
   const int kv_head = blockIdx.x;
@@ -422,7 +422,7 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
     const float scale, const int num_kv_heads, const int group,
     const int max_blocks, const long q_row_stride, const int n_splits,
     float* __restrict__ part_out, float* __restrict__ part_ml,
-    const int window = 0) {
+    const int window) {
   constexpr int ACC = D / 64;
   constexpr int QFRAG = D / 4;
   const int kv_head = blockIdx.x;
@@ -582,7 +582,7 @@ __global__ __launch_bounds__(256) void paged_attention_u2_kernel(
     const float scale, const int num_kv_heads, const int group,
     const int max_blocks, const long q_row_stride, const int n_splits,
     float* __restrict__ part_out, float* __restrict__ part_ml,
-    const int window = 0) {
+    const int window) {
   constexpr int ACC = D / 64;
   constexpr int QFRAG = D / 4;
   const int kv_head = blockIdx.x;
@@ -737,7 +737,7 @@ __global__ __launch_bounds__(256) void paged_attention_ws_kernel(
     const float scale, const int num_kv_heads, const int max_blocks,
     const long q_row_stride, const int n_splits,
     float* __restrict__ part_out, float* __restrict__ part_ml,
-    const int window = 0) {
+    const int window) {
   constexpr int ACC = D / 64;
   constexpr int QFRAG = D / 4;
   const int kv_head = blockIdx.x;
@@ -1109,7 +1109,7 @@ extern "C" hipError_t ks_paged_attention_decode(
           (const short*)k_cache, (const short*)v_cache,
           (const int*)block_tables, (const int*)context_lens, scale,
           num_kv_heads, group, max_blocks, q_row_stride, n_splits,
-          (float*)part_out, (float*)part_ml);
+          (float*)part_out, (float*)part_ml, window);
     }
     HIP_CHECK_KERNEL();
     return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,

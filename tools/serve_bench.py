@@ -27,6 +27,8 @@ async def main():
     parser.add_argument("--prompt-len", type=int, default=512)
     parser.add_argument("--max-tokens", type=int, default=64)
     parser.add_argument("--stream", action="store_true")
+    parser.add_argument("--profile", action="store_true",
+                        help="cProfile the server event loop + engine thread")
     args = parser.parse_args()
 
     import torch
@@ -108,13 +110,36 @@ async def main():
         tokens_out[0] += body["usage"]["completion_tokens"]
         latencies.append(dt)
 
+    prof = None
+    if args.profile:
+        import cProfile
+
+        prof = cProfile.Profile()
+        prof.enable()
+        os.environ["KS_ENGINE_PROFILE"] = "1"
     t0 = time.perf_counter()
     await asyncio.gather(*[one_request(i) for i in range(args.requests)])
     elapsed = time.perf_counter() - t0
+    if prof is not None:
+        import pstats
+
+        prof.disable()
+        stats = pstats.Stats(prof, stream=sys.stderr)
+        print("==== server event loop profile ====", file=sys.stderr)
+        stats.sort_stats("cumulative").print_stats(30)
     lat_sorted = sorted(latencies)
+    eng = model.async_engine.engine
+    stats_extra = {}
+    try:
+        bm = eng.scheduler.block_manager
+        stats_extra["kv_usage"] = round(bm.usage, 3)
+    except Exception:
+        pass
     print(
         {
             "serving_output_tok_s": round(tokens_out[0] / elapsed, 1),
+            "prefill_tokens": args.requests * args.prompt_len,
+            **stats_extra,
             "requests": args.requests,
             "concurrency": args.concurrency,
             "elapsed_s": round(elapsed, 2),
