@@ -184,12 +184,24 @@ class LLMEngine:
         if (
             self.device.type == "cuda"
             and not config.enforce_eager
-            and config.model.num_local_experts == 0
             and comm.get_state().pp_size == 1
         ):
-            # MoE routing (data-dependent expert buckets) is not yet
-            # hipGraph-capturable; MoE decodes run eager
-            self.runner.capture_decode_graphs()
+            # MoE decode captures via the dense-bmm path (static shapes);
+            # buckets are capped at the dense/sparse crossover, larger
+            # batches run eager sparse steps
+            if config.model.num_local_experts > 0:
+                from kserve_amd.models.llama import MixtralMoE
+
+                self.runner.capture_decode_graphs(
+                    batch_sizes=[
+                        b
+                        for b in (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
+                                  192, 256, 384)
+                        if b <= MixtralMoE.DENSE_MAX_TOKENS
+                    ]
+                )
+            else:
+                self.runner.capture_decode_graphs()
         logger.info(
             "Engine ready in %.1fs: %s (%.2fB params, tp=%d, device=%s)",
             time.monotonic() - t0,

@@ -159,12 +159,20 @@ class MixtralMoE(nn.Module):
     """Sparse mixture-of-experts MLP block (Mixtral architecture).
 
     Reference parity: the Mixtral family the reference's vLLM backend
-    serves. MI355X-native v1: token-grouped expert GEMMs via hipBLASLt —
-    tokens are bucketed per expert, each expert runs the dense SwiGLU path
-    on its bucket, outputs are weighted scatter-adds. Under TP each
-    expert's weights are column/row-sharded like the dense MLP and ONE
-    all-reduce covers the whole block.
+    serves. MI355X-native v2: per-expert weights live in STACKED tensors
+    ([E_local, out, in]) so decode runs as two strided-batched GEMMs
+    (torch.bmm -> hipBLASLt) over every local expert with the routing
+    weights applied as a zero-masked mix — static shapes, no host sync,
+    hipGraph-capturable. Prefill keeps the token-bucketed sparse loop
+    (fewer FLOPs at large T). Under TP each expert is column/row-sharded
+    like the dense MLP and ONE all-reduce covers the whole block; EP
+    partitions full-width experts across the group instead.
     """
+
+    # decode batches up to this size take the dense-bmm path (weight reads
+    # dominate there, so the E/top_k FLOP overcompute is ~free); larger
+    # batches use the sparse loop
+    DENSE_MAX_TOKENS = 384
 
     def __init__(
         self,
@@ -178,6 +186,7 @@ class MixtralMoE(nn.Module):
         self.layer_idx = layer_idx
         self.num_experts = config.num_local_experts
         self.top_k = config.num_experts_per_tok
+        H, inter = config.hidden_size, config.intermediate_size
         # expert parallelism: experts are PARTITIONED across the group
         # (full-width weights per expert) instead of every expert being
         # TP-sharded; the per-block all-reduce sums the partial outputs
@@ -190,64 +199,78 @@ class MixtralMoE(nn.Module):
             )
             self.experts_per_rank = self.num_experts // st.tp_size
             self.expert_lo = st.tp_rank * self.experts_per_rank
+            self.inter_local = inter
         else:
             self.experts_per_rank = self.num_experts
             self.expert_lo = 0
+            assert inter % st.tp_size == 0
+            self.inter_local = inter // st.tp_size
         self.gate = nn.Parameter(
-            torch.empty(self.num_experts, config.hidden_size, dtype=dtype),
+            torch.empty(self.num_experts, H, dtype=dtype),
             requires_grad=False,
         )
-        if self.expert_parallel:
-            H, inter = config.hidden_size, config.intermediate_size
-            self.ep_gate_up = nn.Parameter(
-                torch.empty(self.experts_per_rank, 2 * inter, H, dtype=dtype),
-                requires_grad=False,
-            )
-            self.ep_down = nn.Parameter(
-                torch.empty(self.experts_per_rank, H, inter, dtype=dtype),
-                requires_grad=False,
-            )
-            self.gate_up = self.down = None
-        else:
-            self.gate_up = nn.ModuleList(
-                [
-                    MergedColumnParallelLinear(
-                        config.hidden_size, config.intermediate_size, dtype=dtype
-                    )
-                    for _ in range(self.num_experts)
-                ]
-            )
-            self.down = nn.ModuleList(
-                [
-                    RowParallelLinear(
-                        config.intermediate_size,
-                        config.hidden_size,
-                        dtype=dtype,
-                        reduce_output=False,  # one all-reduce for the block
-                    )
-                    for _ in range(self.num_experts)
-                ]
-            )
-
-    def _expert_fwd(self, local_e: int, xe: torch.Tensor) -> torch.Tensor:
-        if self.expert_parallel:
-            gu = torch.nn.functional.linear(xe, self.ep_gate_up[local_e])
-            act = ops.silu_and_mul(gu)
-            return torch.nn.functional.linear(act, self.ep_down[local_e])
-        return self.down[local_e](
-            ops.silu_and_mul(self.gate_up[local_e](xe))
+        # stacked per-expert weights: [E_local, 2*I_local, H] (gate|up) and
+        # [E_local, H, I_local]
+        self.w_gate_up = nn.Parameter(
+            torch.empty(
+                self.experts_per_rank, 2 * self.inter_local, H, dtype=dtype
+            ),
+            requires_grad=False,
+        )
+        self.w_down = nn.Parameter(
+            torch.empty(
+                self.experts_per_rank, H, self.inter_local, dtype=dtype
+            ),
+            requires_grad=False,
         )
 
-    def forward(self, x: torch.Tensor, meta=None) -> torch.Tensor:
+    def load_expert(
+        self, local_e: int, w1: torch.Tensor, w3: torch.Tensor,
+        w2: torch.Tensor,
+    ) -> None:
+        """Load one expert from full-width HF tensors (w1=gate [I,H],
+        w3=up [I,H], w2=down [H,I]), applying this rank's shard."""
+        dt = self.w_gate_up.dtype
+        if self.expert_parallel:
+            self.w_gate_up.data[local_e].copy_(
+                torch.cat([w1, w3], dim=0).to(dt)
+            )
+            self.w_down.data[local_e].copy_(w2.to(dt))
+            return
+        r = comm.get_state().tp_rank
+        lo, hi = r * self.inter_local, (r + 1) * self.inter_local
+        self.w_gate_up.data[local_e, : self.inter_local].copy_(w1[lo:hi].to(dt))
+        self.w_gate_up.data[local_e, self.inter_local :].copy_(w3[lo:hi].to(dt))
+        self.w_down.data[local_e].copy_(w2[:, lo:hi].to(dt))
+
+    def _forward_dense(self, x, topi, topw):
+        """Capture-safe grouped path: every local expert runs on every
+        token (two bmm), the router weights zero out non-selected pairs.
+        Exact (routing enters only through the final mix)."""
+        E = self.experts_per_rank
+        T, H = x.shape
+        w_dense = torch.zeros(
+            T, self.num_experts, dtype=torch.float32, device=x.device
+        )
+        w_dense.scatter_add_(1, topi, topw)
+        w_loc = w_dense[:, self.expert_lo : self.expert_lo + E]  # [T, E]
+        xb = x.unsqueeze(0).expand(E, T, H)
+        gu = torch.bmm(xb, self.w_gate_up.transpose(1, 2))  # [E, T, 2I]
+        act = ops.silu_and_mul(gu.reshape(E * T, -1)).reshape(E, T, -1)
+        ye = torch.bmm(act, self.w_down.transpose(1, 2))  # [E, T, H]
+        return torch.einsum(
+            "eth,te->th", ye.float(), w_loc
+        ).to(x.dtype)
+
+    def _forward_sparse(self, x, topi, topw):
+        """Token-bucketed loop (prefill): exact FLOPs, host-side gather."""
         T = x.shape[0]
-        router_logits = torch.nn.functional.linear(x.float(), self.gate.float())
-        probs = torch.softmax(router_logits, dim=-1)
-        topw, topi = probs.topk(self.top_k, dim=-1)
-        topw = topw / topw.sum(dim=-1, keepdim=True)  # mixtral renorm
         out = torch.zeros_like(x)
         flat_i = topi.flatten()
         flat_w = topw.flatten()
-        token_idx = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
+        token_idx = torch.arange(T, device=x.device).repeat_interleave(
+            self.top_k
+        )
         for le in range(self.experts_per_rank):
             e = self.expert_lo + le
             sel = (flat_i == e).nonzero(as_tuple=True)[0]
@@ -255,8 +278,27 @@ class MixtralMoE(nn.Module):
                 continue
             rows = token_idx[sel]
             xe = x.index_select(0, rows)
-            ye = self._expert_fwd(le, xe)
-            out.index_add_(0, rows, ye * flat_w[sel].unsqueeze(1).to(ye.dtype))
+            gu = torch.nn.functional.linear(xe, self.w_gate_up[le])
+            act = ops.silu_and_mul(gu)
+            ye = torch.nn.functional.linear(act, self.w_down[le])
+            out.index_add_(
+                0, rows, ye * flat_w[sel].unsqueeze(1).to(ye.dtype)
+            )
+        return out
+
+    def forward(self, x: torch.Tensor, meta=None) -> torch.Tensor:
+        T = x.shape[0]
+        router_logits = torch.nn.functional.linear(x.float(), self.gate.float())
+        probs = torch.softmax(router_logits, dim=-1)
+        topw, topi = probs.topk(self.top_k, dim=-1)
+        topw = topw / topw.sum(dim=-1, keepdim=True)  # mixtral renorm
+        capturing = (
+            x.is_cuda and torch.cuda.is_current_stream_capturing()
+        )
+        if capturing or T <= self.DENSE_MAX_TOKENS:
+            out = self._forward_dense(x, topi, topw)
+        else:
+            out = self._forward_sparse(x, topi, topw)
         if comm.get_state().tp_size > 1:
             out = comm.tp_all_reduce(out)
         return out
@@ -471,17 +513,12 @@ class LlamaForCausalLM(nn.Module):
                     )
                     for le, j in enumerate(owned):
                         ep = p + f"block_sparse_moe.experts.{j}."
-                        w1 = get(ep + "w1.weight")
-                        w3 = get(ep + "w3.weight")
-                        w2 = get(ep + "w2.weight")
-                        if moe.expert_parallel:
-                            moe.ep_gate_up.data[le].copy_(
-                                torch.cat([w1, w3], dim=0).to(moe.gate.dtype)
-                            )
-                            moe.ep_down.data[le].copy_(w2.to(moe.gate.dtype))
-                        else:
-                            moe.gate_up[le].load_shards(w1, w3)
-                            moe.down[le].load_shard(w2)
+                        moe.load_expert(
+                            le,
+                            get(ep + "w1.weight"),
+                            get(ep + "w3.weight"),
+                            get(ep + "w2.weight"),
+                        )
                 else:
                     # fused layout (transformers >= 4.5x): mlp.gate.weight,
                     # mlp.experts.gate_up_proj [E, 2I, H], .down_proj [E, H, I]
@@ -492,16 +529,9 @@ class LlamaForCausalLM(nn.Module):
                     dn = get(p + "mlp.experts.down_proj")
                     inter = gu.shape[1] // 2
                     for le, j in enumerate(owned):
-                        if moe.expert_parallel:
-                            moe.ep_gate_up.data[le].copy_(
-                                gu[j].to(moe.gate.dtype)
-                            )
-                            moe.ep_down.data[le].copy_(dn[j].to(moe.gate.dtype))
-                        else:
-                            moe.gate_up[le].load_shards(
-                                gu[j][:inter], gu[j][inter:]
-                            )
-                            moe.down[le].load_shard(dn[j])
+                        moe.load_expert(
+                            le, gu[j][:inter], gu[j][inter:], dn[j]
+                        )
             else:
                 layer.mlp.gate_up_proj.load_shards(
                     get(p + "mlp.gate_proj.weight"), get(p + "mlp.up_proj.weight")

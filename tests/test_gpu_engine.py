@@ -382,8 +382,8 @@ def test_prefix_cache_on_gpu():
 
 
 def test_mixtral_moe_on_gpu():
-    """Sparse-MoE engine path on GPU (eager: MoE routing is not yet
-    graph-captured); batched greedy equals single greedy."""
+    """MoE engine path on GPU: decode is hipGraph-captured via the
+    dense-bmm path (round 2); batched greedy equals single greedy."""
     from kserve_amd.engine.config import (
         CacheConfig,
         EngineConfig,
@@ -410,6 +410,7 @@ def test_mixtral_moe_on_gpu():
         eos_token_id=-1,
     )
     engine = LLMEngine(cfg)
+    assert engine.runner._graphs, "MoE decode graphs must be captured"
     sp = SamplingParams(temperature=0.0, max_tokens=10)
     out = engine.generate([[1, 2, 3], [9, 8, 7, 6]], sp)
     assert all(len(o.output_token_ids) == 10 for o in out.values())
@@ -418,6 +419,15 @@ def test_mixtral_moe_on_gpu():
         list(single.values())[0].output_token_ids
         == list(out.values())[0].output_token_ids
     )
+    # graph path must agree with a fully-eager engine
+    del engine
+    torch.cuda.empty_cache()
+    cfg.enforce_eager = True
+    eager = LLMEngine(cfg)
+    out_e = eager.generate([[1, 2, 3], [9, 8, 7, 6]], sp)
+    assert [o.output_token_ids for o in out_e.values()] == [
+        o.output_token_ids for o in out.values()
+    ]
 
 
 def test_sampled_window_graph_topk1_matches_greedy():

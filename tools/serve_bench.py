@@ -70,6 +70,8 @@ async def main():
     if not use_gpu:
         cfg.cache.num_gpu_blocks = 512
 
+    if args.profile:
+        os.environ["KS_ENGINE_PROFILE"] = "1"  # before the engine thread starts
     model = LLMModel("bench", cfg)
     repo = ModelRepository()
     repo.update(model)
@@ -116,7 +118,6 @@ async def main():
 
         prof = cProfile.Profile()
         prof.enable()
-        os.environ["KS_ENGINE_PROFILE"] = "1"
     t0 = time.perf_counter()
     await asyncio.gather(*[one_request(i) for i in range(args.requests)])
     elapsed = time.perf_counter() - t0
