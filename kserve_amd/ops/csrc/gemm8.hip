@@ -74,7 +74,15 @@ __device__ __forceinline__ bf16x8_t lds_frag(const short* lds_base,
       reinterpret_cast<const char*>(lds_base) + byte);
 }
 
-template <bool SW>
+// SCHED 0: stage one half-tile per phase, drain (vmcnt 0) at phase 3 —
+//   the last load has <1 phase of MFMA to hide under.
+// SCHED 1: front-load the stages (A0+A1 at phase 0, B0+B1 at phase 1) with
+//   a counted vmcnt(4) at phase 2 (A halves landed; FIFO retirement) and
+//   vmcnt(0) at phase 3 covering only the B halves, which then have 2-3
+//   phases in flight. (Step toward the guide's counted-vmcnt discipline —
+//   its full 3-half-tiles-in-flight schedule needs half-granular read
+//   ordering; this keeps the simple whole-tile flip.)
+template <bool SW, int SCHED = 0>
 __global__ __launch_bounds__(512, 1) void gemm8_kernel(
     short* __restrict__ D,        // [M, N] bf16
     const short* __restrict__ A,  // [M, K] bf16
@@ -156,15 +164,25 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
           b_frag[n][ks] =
               lds_frag<SW>(bb, row, ks * 32 + ((lane >> 4) << 3));
       }
-      // ---- stage one half-tile of the NEXT K-tile ----
+      // ---- stage NEXT K-tile half-tiles (schedule-dependent) ----
       if (have_next) {
         const int other = buf ^ 1;
-        if (ph == 0) stage_half<SW>(a_next, K, 0, A_BUF(other), wave, lane);
-        else if (ph == 1)
-          stage_half<SW>(a_next, K, 128, A_BUF(other), wave, lane);
-        else if (ph == 2)
-          stage_half<SW>(w_next, K, 0, B_BUF(other), wave, lane);
-        else stage_half<SW>(w_next, K, 128, B_BUF(other), wave, lane);
+        if constexpr (SCHED == 0) {
+          if (ph == 0) stage_half<SW>(a_next, K, 0, A_BUF(other), wave, lane);
+          else if (ph == 1)
+            stage_half<SW>(a_next, K, 128, A_BUF(other), wave, lane);
+          else if (ph == 2)
+            stage_half<SW>(w_next, K, 0, B_BUF(other), wave, lane);
+          else stage_half<SW>(w_next, K, 128, B_BUF(other), wave, lane);
+        } else {
+          if (ph == 0) {
+            stage_half<SW>(a_next, K, 0, A_BUF(other), wave, lane);
+            stage_half<SW>(a_next, K, 128, A_BUF(other), wave, lane);
+          } else if (ph == 1) {
+            stage_half<SW>(w_next, K, 0, B_BUF(other), wave, lane);
+            stage_half<SW>(w_next, K, 128, B_BUF(other), wave, lane);
+          }
+        }
       }
       __builtin_amdgcn_s_barrier();
       asm volatile("s_waitcnt lgkmcnt(0)");
@@ -179,8 +197,14 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
             acc[qm + m][qn + n] =
                 mfma16x16x32(a_frag[m][ks], b_frag[n][ks], acc[qm + m][qn + n]);
       __builtin_amdgcn_s_setprio(0);
-      // drain staged loads once per K-tile (phase 4), never vmcnt(0)
-      // mid-loop except at the boundary before the buffer flips
+      // drain staged loads before the buffer flips. SCHED 1: counted
+      // vmcnt(4) at phase 2 retires the A halves early (FIFO), leaving
+      // only the B halves for the phase-3 wait with 2 phases in flight.
+      if constexpr (SCHED == 1) {
+        if (ph == 2 && have_next) {
+          asm volatile("s_waitcnt vmcnt(4)");
+        }
+      }
       if (ph == 3 && have_next) {
         asm volatile("s_waitcnt vmcnt(0)");
       }
@@ -213,23 +237,43 @@ extern "C" hipError_t ks_gemm8(void* d, const void* a, const void* w, int M,
   const size_t lds_bytes = 4 * BM * BK * sizeof(short);  // 128 KiB
   // dynamic LDS above the 64 KiB default needs an explicit opt-in
   static bool attr_set = [] {
-    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<true>,
+    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<true, 0>,
                         hipFuncAttributeMaxDynamicSharedMemorySize,
                         4 * BM * BK * sizeof(short));
-    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<false>,
+    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<false, 0>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        4 * BM * BK * sizeof(short));
+    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<true, 1>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        4 * BM * BK * sizeof(short));
+    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<false, 1>,
                         hipFuncAttributeMaxDynamicSharedMemorySize,
                         4 * BM * BK * sizeof(short));
     return true;
   }();
   (void)attr_set;
+  static const int sched = [] {
+    const char* e = getenv("KS_GEMM8_SCHED");
+    return (e && e[0] == '1') ? 1 : 0;
+  }();
   if (use_swizzle) {
-    hipLaunchKernelGGL((gemm8_kernel<true>), grid, dim3(512), lds_bytes,
-                       stream, (short*)d, (const short*)a, (const short*)w,
-                       M, N, K);
+    if (sched == 1)
+      hipLaunchKernelGGL((gemm8_kernel<true, 1>), grid, dim3(512), lds_bytes,
+                         stream, (short*)d, (const short*)a, (const short*)w,
+                         M, N, K);
+    else
+      hipLaunchKernelGGL((gemm8_kernel<true, 0>), grid, dim3(512), lds_bytes,
+                         stream, (short*)d, (const short*)a, (const short*)w,
+                         M, N, K);
   } else {
-    hipLaunchKernelGGL((gemm8_kernel<false>), grid, dim3(512), lds_bytes,
-                       stream, (short*)d, (const short*)a, (const short*)w,
-                       M, N, K);
+    if (sched == 1)
+      hipLaunchKernelGGL((gemm8_kernel<false, 1>), grid, dim3(512),
+                         lds_bytes, stream, (short*)d, (const short*)a,
+                         (const short*)w, M, N, K);
+    else
+      hipLaunchKernelGGL((gemm8_kernel<false, 0>), grid, dim3(512),
+                         lds_bytes, stream, (short*)d, (const short*)a,
+                         (const short*)w, M, N, K);
   }
   HIP_CHECK_KERNEL();
   return hipSuccess;
