@@ -161,6 +161,18 @@ def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
+def tp_all_reduce_async(t: torch.Tensor):
+    """Launch the TP sum-all-reduce without blocking kernel issue; returns
+    a work handle (None when tp_size==1). The collective is ordered after
+    everything already queued on the current stream, so later independent
+    GEMMs overlap with it (xGMI link time hides under compute)."""
+    if _STATE.tp_size > 1:
+        return dist.all_reduce(
+            t, op=dist.ReduceOp.SUM, group=_STATE.tp_group, async_op=True
+        )
+    return None
+
+
 def tp_all_gather(t: torch.Tensor, dim: int = -1) -> torch.Tensor:
     if _STATE.tp_size == 1:
         return t

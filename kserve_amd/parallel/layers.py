@@ -141,7 +141,19 @@ class QKVParallelLinear(nn.Module):
 
 class RowParallelLinear(nn.Module):
     """Y = sum_ranks(X_shard @ W_shard^T); input features sharded; output
-    all-reduced (the per-layer RCCL call over xGMI)."""
+    all-reduced (the per-layer RCCL call over xGMI).
+
+    Comm/compute overlap (north star): with ``overlap_chunks`` > 1 the GEMM
+    is split along output features and each chunk's all-reduce launches
+    async as soon as its GEMM is queued — the next chunk's GEMM overlaps
+    the previous chunk's xGMI time (per-link-bound ring). All handles are
+    waited before the concatenated result is returned, so numerics are
+    identical to the synchronous path.
+    """
+
+    # overlap engages above this many output elements (small decode
+    # batches are latency-bound on the collective's launch, not its bytes)
+    OVERLAP_MIN_NUMEL = 1 << 20
 
     def __init__(
         self,
@@ -150,6 +162,7 @@ class RowParallelLinear(nn.Module):
         bias: bool = False,
         dtype: torch.dtype = torch.bfloat16,
         reduce_output: bool = True,
+        overlap_chunks: int = 2,
     ):
         super().__init__()
         st = comm.get_state()
@@ -167,10 +180,37 @@ class RowParallelLinear(nn.Module):
             if bias
             else None
         )
+        self.overlap_chunks = max(1, overlap_chunks)
 
     def forward(
         self, x: torch.Tensor, delta: Optional[torch.Tensor] = None
     ) -> torch.Tensor:
+        use_overlap = (
+            self.reduce_output
+            and self.tp_size > 1
+            and delta is None
+            and self.overlap_chunks > 1
+            and x.shape[0] * self.out_features >= self.OVERLAP_MIN_NUMEL
+        )
+        if use_overlap:
+            n = self.overlap_chunks
+            step = -(-self.out_features // n)
+            parts = []
+            works = []
+            for c in range(n):
+                wc = self.weight[c * step : (c + 1) * step]
+                if wc.shape[0] == 0:
+                    break
+                yc = ops.linear(x, wc)
+                works.append(comm.tp_all_reduce_async(yc))
+                parts.append(yc)
+            for w in works:
+                if w is not None:
+                    w.wait()
+            y = torch.cat(parts, dim=-1)
+            if self.bias is not None:
+                y = y + self.bias
+            return y
         y = ops.linear(x, self.weight)
         if delta is not None:
             # per-rank partial (e.g. LoRA with A column-sharded): must be

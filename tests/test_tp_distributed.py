@@ -202,3 +202,45 @@ def test_tp2_engine_generate_consistent():
         p.join(timeout=60)
     assert results[0] == results[1]
     assert all(len(t) == 6 for t in results[0])
+
+
+def _tp2_overlap_worker(rank, world, port, q):
+    try:
+        _set_env(rank, world, port)
+        from kserve_amd.parallel import comm
+        from kserve_amd.parallel.layers import RowParallelLinear
+
+        comm.init_distributed(tp_size=2, backend="gloo")
+        torch.manual_seed(1)
+        full_w = torch.randn(64, 32)
+        x = torch.randn(16, 16)  # this rank's shard input
+
+        sync = RowParallelLinear(32, 64, dtype=torch.float32,
+                                 overlap_chunks=1)
+        sync.load_shard(full_w)
+        over = RowParallelLinear(32, 64, dtype=torch.float32,
+                                 overlap_chunks=2)
+        over.load_shard(full_w)
+        over.OVERLAP_MIN_NUMEL = 1  # force the overlap path at test sizes
+        y_sync = sync(x)
+        y_over = over(x)
+        torch.testing.assert_close(y_over, y_sync, atol=1e-5, rtol=1e-5)
+        # odd chunking still exact
+        over3 = RowParallelLinear(32, 64, dtype=torch.float32,
+                                  overlap_chunks=3)
+        over3.load_shard(full_w)
+        over3.OVERLAP_MIN_NUMEL = 1
+        torch.testing.assert_close(over3(x), y_sync, atol=1e-5, rtol=1e-5)
+        comm.destroy_distributed()
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
+
+
+@pytest.mark.timeout(300)
+def test_tp2_overlapped_all_reduce_matches_sync():
+    """Chunked async all-reduce (comm/compute overlap) must be numerically
+    identical to the synchronous reduction."""
+    _run_workers(_tp2_overlap_worker, port=29617)
