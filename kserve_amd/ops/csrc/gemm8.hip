@@ -25,11 +25,15 @@ constexpr int FRAG_M = 8, FRAG_N = 4;   // 16x16 fragments per wave
 // per-quadrant fragment split: 4M x 2N, four quadrants per K-tile
 constexpr int QM = 4, QN = 2;
 
-// st_16x32 swizzle on a byte offset within each 1024-B subtile:
-// XOR byte-bit-5 with bit-9 (guide: bank-conflict 141x reduction)
+// st_16x32 swizzle on a byte offset: XOR three row bits into the 16-B
+// chunk index (guide T2 recipe: byte ^= ((row&7)<<4), row = byte>>7 for a
+// 128-B row). Spreads a 16-lane column-slice read across all 32 banks —
+// the 1-bit (bit9->bit5) variant of round 1 left ~4-8-way conflicts
+// (measured 126M SQ_LDS_BANK_CONFLICT at 4096^3; PMC 2026-09-12).
+// Involution: bits >=7 are untouched, so applying twice is identity.
 template <bool SW>
 __device__ __forceinline__ int swz(int byte_off) {
-  if constexpr (SW) return byte_off ^ (((byte_off >> 9) & 1) << 5);
+  if constexpr (SW) return byte_off ^ (((byte_off >> 7) & 7) << 4);
   return byte_off;
 }
 
