@@ -419,3 +419,98 @@ class TestStorage:
 
         with pytest.raises(Exception):
             Storage.download("carrier-pigeon://model", str(tmp_path / "o4"))
+
+
+class TestQpextMerging:
+    """Prometheus exposition merging (reference qpext sanitizeMetrics
+    :113 + scrape :198): one header per family, duplicate series
+    disambiguated by source, noisy default collectors dropped."""
+
+    def test_families_merge_with_single_header(self):
+        from kserve_amd.agent.qpext import sanitize_metrics
+
+        qp = (
+            "# HELP requests_total Total requests\n"
+            "# TYPE requests_total counter\n"
+            'requests_total{code="200"} 5\n'
+        )
+        app = (
+            "# HELP requests_total Total requests\n"
+            "# TYPE requests_total counter\n"
+            'requests_total{code="500"} 1\n'
+            "# HELP model_load_seconds Load time\n"
+            "# TYPE model_load_seconds gauge\n"
+            "model_load_seconds 1.5\n"
+        )
+        merged = sanitize_metrics([("queue-proxy", qp), ("app", app)])
+        assert merged.count("# TYPE requests_total counter") == 1
+        assert 'requests_total{code="200"} 5' in merged
+        assert 'requests_total{code="500"} 1' in merged
+        assert "model_load_seconds 1.5" in merged
+
+    def test_duplicate_series_get_source_label(self):
+        from kserve_amd.agent.qpext import sanitize_metrics
+
+        a = "# TYPE up gauge\nup 1\n"
+        b = "# TYPE up gauge\nup 0\n"
+        merged = sanitize_metrics([("queue-proxy", a), ("app", b)])
+        assert "up 1" in merged
+        assert 'up{source="app"} 0' in merged
+
+    def test_noise_prefixes_dropped_and_histograms_kept(self):
+        from kserve_amd.agent.qpext import sanitize_metrics
+
+        text = (
+            "# HELP python_gc_count GC\n# TYPE python_gc_count counter\n"
+            "python_gc_count 3\n"
+            "# HELP lat Latency\n# TYPE lat histogram\n"
+            'lat_bucket{le="0.1"} 2\nlat_sum 0.3\nlat_count 2\n'
+        )
+        merged = sanitize_metrics([("app", text)])
+        assert "python_gc_count" not in merged
+        assert 'lat_bucket{le="0.1"} 2' in merged
+        assert "lat_count 2" in merged
+
+    def test_scrape_config_from_env(self):
+        from kserve_amd.agent.qpext import ScrapeConfiguration
+
+        cfg = ScrapeConfiguration.from_env(
+            {
+                "AGGREGATE_PROMETHEUS_METRICS_PORT": "8080",
+                "APP_METRICS_PATH": "/stats",
+                "QUEUE_PROXY_METRICS_PORT": "9091",
+                "METRICS_SCRAPE_TIMEOUT_S": "2",
+            }
+        )
+        urls = [t.url for t in cfg.targets]
+        assert "http://127.0.0.1:9091/metrics" in urls
+        assert "http://127.0.0.1:8080/stats" in urls
+        assert all(t.timeout_s == 2.0 for t in cfg.targets)
+
+    def test_app_merges_and_survives_dead_target(self):
+        import httpx
+        from fastapi.testclient import TestClient
+
+        from kserve_amd.agent.qpext import ScrapeConfiguration, ScrapeTarget, create_qpext_app
+
+        async def handler(request):
+            if "good" in str(request.url):
+                return httpx.Response(
+                    200, text="# TYPE ok gauge\nok 1\n"
+                )
+            raise httpx.ConnectError("down")
+
+        app = create_qpext_app(
+            config=ScrapeConfiguration(
+                targets=[
+                    ScrapeTarget("http://good/metrics", source_label="app"),
+                    ScrapeTarget("http://dead/metrics"),
+                ]
+            ),
+            transport=httpx.MockTransport(handler),
+        )
+        client = TestClient(app)
+        r = client.get("/metrics")
+        assert r.status_code == 200
+        assert "ok 1" in r.text
+        assert client.get("/healthz").json()["status"] == "ok"
