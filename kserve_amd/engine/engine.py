@@ -260,7 +260,16 @@ class LLMEngine:
             req.guided_machine = ChoiceMachine(
                 [c.encode("utf-8") for c in sampling_params.guided_choice]
             )
-        elif sampling_params.response_format in ("json", "json_object"):
+        elif (
+            sampling_params.response_format == "json_schema"
+            and sampling_params.json_schema
+        ):
+            from kserve_amd.engine.guided import SchemaMachine
+
+            req.guided_machine = SchemaMachine(sampling_params.json_schema)
+        elif sampling_params.response_format in ("json", "json_object",
+                                                 "json_schema"):
+            # json_schema without a schema degrades to valid-JSON-object
             from kserve_amd.engine.guided import JsonMachine
 
             req.guided_machine = JsonMachine(object_only=True)

@@ -330,3 +330,369 @@ class ChoiceMachine:
             if not self.advance(b):
                 return False
         return True
+
+
+# ---------------------------------------------------------------------------
+# JSON-Schema-constrained machine (OpenAI response_format json_schema).
+#
+# Strict structured outputs over the practical schema subset: objects with
+# typed properties (emitted in schema order, compact serialization — no
+# interstitial whitespace, matching outlines/OpenAI strict mode), strings,
+# enums (any scalar constants), integers/numbers, booleans, null, arrays
+# with minItems/maxItems, nested combinations, and nullable via
+# type: [T, "null"] (byte-time branch). Same machine protocol as
+# JsonMachine (clone/signature/advance/complete), so GuidedJsonProcessor's
+# signature-memoized masks apply unchanged.
+# ---------------------------------------------------------------------------
+
+import json as _json
+
+_CONSUME = 0
+_POP_CONSUME = 1
+_POP_REDISPATCH = 2
+_ILLEGAL = 3
+
+
+class _FixedFrame:
+    __slots__ = ("data", "pos")
+
+    def __init__(self, data: bytes, pos: int = 0):
+        self.data = data
+        self.pos = pos
+
+    def clone(self):
+        return _FixedFrame(self.data, self.pos)
+
+    def sig(self):
+        return ("fx", self.data, self.pos)
+
+    def feed(self, b: int):
+        if self.data[self.pos] != b:
+            return _ILLEGAL
+        self.pos += 1
+        return _POP_CONSUME if self.pos == len(self.data) else _CONSUME
+
+
+class _StringFrame:
+    """'"' body '"' with escape handling; any UTF-8 content."""
+
+    __slots__ = ("state",)  # 0=open quote, 1=body, 2=escape
+
+    def __init__(self, state: int = 0):
+        self.state = state
+
+    def clone(self):
+        return _StringFrame(self.state)
+
+    def sig(self):
+        return ("str", self.state)
+
+    def feed(self, b: int):
+        if self.state == 0:
+            if b == 0x22:
+                self.state = 1
+                return _CONSUME
+            return _ILLEGAL
+        if self.state == 1:
+            if b == 0x22:
+                return _POP_CONSUME
+            if b == 0x5C:
+                self.state = 2
+                return _CONSUME
+            return _CONSUME if b >= 0x20 else _ILLEGAL
+        if b in STRING_ESCAPES:
+            self.state = 1
+            return _CONSUME
+        return _ILLEGAL
+
+
+class _NumberFrame:
+    __slots__ = ("ns", "integer", "started")
+
+    def __init__(self, integer: bool, ns: int = -1):
+        self.ns = ns  # -1 = nothing consumed yet
+        self.integer = integer
+        self.started = ns >= 0
+
+    def clone(self):
+        f = _NumberFrame(self.integer, self.ns)
+        f.started = self.started
+        return f
+
+    def sig(self):
+        return ("num", self.ns, self.integer)
+
+    @property
+    def terminal(self) -> bool:
+        return self.ns in _NUM_TERMINAL
+
+    def feed(self, b: int):
+        ns = self.ns
+        if ns == -1:
+            if b == 0x2D:
+                self.ns = N_SIGN
+                return _CONSUME
+            if b in DIGITS:
+                self.ns = N_ZERO if b == 0x30 else N_INT
+                return _CONSUME
+            return _ILLEGAL
+        if b in DIGITS:
+            if ns == N_ZERO:
+                return _ILLEGAL
+            self.ns = {
+                N_SIGN: N_ZERO if b == 0x30 else N_INT,
+                N_INT: N_INT,
+                N_FRAC_START: N_FRAC,
+                N_FRAC: N_FRAC,
+                N_EXP_START: N_EXP,
+                N_EXP_SIGN: N_EXP,
+                N_EXP: N_EXP,
+            }[ns]
+            return _CONSUME
+        if not self.integer:
+            if b == 0x2E and ns in (N_ZERO, N_INT):
+                self.ns = N_FRAC_START
+                return _CONSUME
+            if b in (0x65, 0x45) and ns in (N_ZERO, N_INT, N_FRAC):
+                self.ns = N_EXP_START
+                return _CONSUME
+            if b in (0x2B, 0x2D) and ns == N_EXP_START:
+                self.ns = N_EXP_SIGN
+                return _CONSUME
+        if self.terminal:
+            return _POP_REDISPATCH  # delimiter belongs to the parent
+        return _ILLEGAL
+
+
+class _EnumFrame:
+    """One of a fixed set of serialized JSON constants."""
+
+    __slots__ = ("choices", "live", "pos")
+
+    def __init__(self, choices, live=None, pos: int = 0):
+        self.choices = choices
+        self.live = live if live is not None else list(range(len(choices)))
+        self.pos = pos
+
+    def clone(self):
+        return _EnumFrame(self.choices, list(self.live), self.pos)
+
+    def sig(self):
+        return ("enum", id(self.choices), tuple(self.live), self.pos)
+
+    def feed(self, b: int):
+        nxt = [
+            i for i in self.live
+            if self.pos < len(self.choices[i]) and self.choices[i][self.pos] == b
+        ]
+        if not nxt:
+            return _ILLEGAL
+        self.live = nxt
+        self.pos += 1
+        if all(len(self.choices[i]) == self.pos for i in self.live):
+            return _POP_CONSUME
+        return _CONSUME
+
+
+class _ArrayFrame:
+    """'[' item (',' item)* ']' with [minItems, maxItems] bounds.
+    Pushes item frames; between items decides ',' vs ']'."""
+
+    __slots__ = ("item_schema", "count", "min_items", "max_items", "state")
+    # state: 0=expect '[', 1=first item or ']', 2=',' or ']'
+
+    def __init__(self, item_schema, min_items: int, max_items, count=0,
+                 state=0):
+        self.item_schema = item_schema
+        self.min_items = min_items
+        self.max_items = max_items
+        self.count = count
+        self.state = state
+
+    def clone(self):
+        return _ArrayFrame(self.item_schema, self.min_items, self.max_items,
+                           self.count, self.state)
+
+    def sig(self):
+        return ("arr", id(self.item_schema), self.count, self.state)
+
+    def feed(self, b: int):
+        if self.state == 0:
+            if b != 0x5B:
+                return _ILLEGAL
+            self.state = 1
+            return _CONSUME
+        if self.state == 1:
+            if b == 0x5D and self.min_items <= 0:
+                return _POP_CONSUME
+            if self.max_items is not None and self.count >= self.max_items:
+                return _ILLEGAL
+            # start of the first item: push its frames, redispatch b
+            self.count += 1
+            self.state = 2
+            return ("push", _frames_for(self.item_schema))
+        # state 2: after an item
+        if b == 0x2C:  # ,
+            if self.max_items is not None and self.count >= self.max_items:
+                return _ILLEGAL
+            self.count += 1
+            return ("push_consume", _frames_for(self.item_schema))
+        if b == 0x5D and self.count >= self.min_items:
+            return _POP_CONSUME
+        return _ILLEGAL
+
+
+class _BranchFrame:
+    """Byte-time union (nullable / anyOf of scalars): live sub-machines
+    advance in lockstep; legal while any survives."""
+
+    __slots__ = ("subs",)
+
+    def __init__(self, subs):
+        self.subs = subs  # list of SchemaMachine
+
+    def clone(self):
+        return _BranchFrame([m.clone() for m in self.subs])
+
+    def sig(self):
+        return ("br", tuple(m.signature() for m in self.subs))
+
+    def feed(self, b: int):
+        survivors = []
+        for m in self.subs:
+            mm = m.clone()
+            if mm.advance(b):
+                survivors.append(mm)
+        if not survivors:
+            # a completed sub-value means the byte belongs to the parent
+            if any(m.complete for m in self.subs):
+                return _POP_REDISPATCH
+            return _ILLEGAL
+        self.subs = survivors
+        if all(m.complete and not m._can_continue() for m in survivors):
+            return _POP_CONSUME
+        return _CONSUME
+
+
+class SchemaUnsupported(ValueError):
+    pass
+
+
+def _frames_for(schema: dict):
+    """Compile a schema node into a frame list (top of stack first)."""
+    if "enum" in schema:
+        choices = [
+            _json.dumps(v, separators=(",", ":")).encode() for v in schema["enum"]
+        ]
+        return [_EnumFrame(choices)]
+    if "const" in schema:
+        return [_FixedFrame(
+            _json.dumps(schema["const"], separators=(",", ":")).encode()
+        )]
+    t = schema.get("type")
+    if isinstance(t, list):
+        subs = [
+            SchemaMachine({**schema, "type": tt}) for tt in t
+        ]
+        return [_BranchFrame(subs)]
+    if t == "string":
+        return [_StringFrame()]
+    if t == "integer":
+        return [_NumberFrame(integer=True)]
+    if t == "number":
+        return [_NumberFrame(integer=False)]
+    if t == "boolean":
+        return [_EnumFrame([b"true", b"false"])]
+    if t == "null":
+        return [_FixedFrame(b"null")]
+    if t == "array":
+        return [
+            _ArrayFrame(
+                schema.get("items", {"type": "string"}),
+                schema.get("minItems", 0),
+                schema.get("maxItems"),
+            )
+        ]
+    if t == "object" or "properties" in schema:
+        props = schema.get("properties", {})
+        if not props:
+            raise SchemaUnsupported(
+                "object schemas need properties (free-form objects: use "
+                "response_format json_object)"
+            )
+        # strict serialization: every property, schema order, compact
+        frames = []
+        keys = list(props.keys())
+        prefix = b"{"
+        for i, k in enumerate(keys):
+            key_bytes = _json.dumps(k).encode() + b":"
+            frames.append(_FixedFrame(prefix + key_bytes))
+            frames.extend(_frames_for(props[k]))
+            prefix = b","
+        frames.append(_FixedFrame(b"}"))
+        return frames
+    raise SchemaUnsupported(f"unsupported schema node: {schema!r}")
+
+
+class SchemaMachine:
+    """Byte-level acceptor for one JSON document matching ``schema``."""
+
+    __slots__ = ("frames", "_schema")
+
+    def __init__(self, schema: dict, _frames=None):
+        self._schema = schema
+        self.frames = (
+            _frames if _frames is not None else list(reversed(_frames_for(schema)))
+        )  # stack: top = last element
+
+    def clone(self) -> "SchemaMachine":
+        return SchemaMachine(
+            self._schema, _frames=[f.clone() for f in self.frames]
+        )
+
+    def signature(self) -> Tuple:
+        return ("schema", tuple(f.sig() for f in self.frames[-3:]),
+                len(self.frames))
+
+    def _can_continue(self) -> bool:
+        """True when a trailing number frame could still accept digits."""
+        return bool(self.frames)
+
+    @property
+    def complete(self) -> bool:
+        if not self.frames:
+            return True
+        # a single terminal number frame at top level completes via EOS
+        return (
+            len(self.frames) == 1
+            and isinstance(self.frames[-1], _NumberFrame)
+            and self.frames[-1].terminal
+        )
+
+    def advance(self, b: int) -> bool:
+        while True:
+            if not self.frames:
+                return False  # document finished; nothing may follow
+            top = self.frames[-1]
+            r = top.feed(b)
+            if r == _CONSUME:
+                return True
+            if r == _POP_CONSUME:
+                self.frames.pop()
+                return True
+            if r == _POP_REDISPATCH:
+                self.frames.pop()
+                continue
+            if isinstance(r, tuple):
+                action, new_frames = r
+                self.frames.extend(reversed(new_frames))
+                if action == "push_consume":
+                    return True
+                continue  # "push": redispatch b to the new top
+            return False
+
+    def accepts(self, data: bytes) -> bool:
+        for b in data:
+            if not self.advance(b):
+                return False
+        return True

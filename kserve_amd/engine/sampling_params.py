@@ -20,8 +20,11 @@ class SamplingParams:
     min_p: float = 0.0
     # OpenAI logit_bias: token-id -> additive bias (-100..100)
     logit_bias: Optional[Dict[int, float]] = None
-    # guided decoding: "json_object" constrains output to valid JSON
+    # guided decoding: "json_object" constrains output to valid JSON;
+    # "json_schema" additionally constrains to json_schema below
     response_format: Optional[str] = None
+    # guided decoding: JSON Schema for response_format "json_schema"
+    json_schema: Optional[Dict] = None
     # guided decoding: constrain output to one of these strings
     guided_choice: Optional[List[str]] = None
     max_tokens: int = 16

@@ -53,6 +53,14 @@ def _to_sampling_params(
             if getattr(req, "response_format", None)
             else None
         ),
+        json_schema=(
+            # OpenAI shape: response_format.json_schema.schema
+            ((req.response_format or {}).get("json_schema") or {}).get(
+                "schema"
+            )
+            if getattr(req, "response_format", None)
+            else None
+        ),
         logit_bias=(
             {int(k): float(v) for k, v in req.logit_bias.items()}
             if getattr(req, "logit_bias", None)
