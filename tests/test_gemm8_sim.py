@@ -17,7 +17,8 @@ QM, QN = 4, 2
 
 
 def swz(byte, on):
-    return byte ^ ((((byte >> 9) & 1) << 5) if on else 0)
+    # T2 st_16x32: XOR three row bits into the 16-B chunk index
+    return byte ^ (((((byte >> 7) & 7) << 4)) if on else 0)
 
 
 def stage_tile(src_tile, swizzle):
