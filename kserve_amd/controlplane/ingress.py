@@ -260,3 +260,70 @@ def render_keda_scaled_object(
             "triggers": [trigger],
         },
     }
+
+
+# -- OTel Collector sidecar CR (reconcilers/otel/otel_reconciler.go:302) ----
+
+def render_otel_collector(
+    name: str,
+    namespace: str,
+    metric_names: Optional[List[str]] = None,
+    otlp_endpoint: str = "",
+    scrape_interval_s: int = 15,
+) -> Dict:
+    """OpenTelemetryCollector sidecar CR created per-ISVC when PodMetrics
+    autoscaling is used: scrapes the pod's Prometheus endpoint, filters to
+    the metrics the autoscaler consumes, ships OTLP (reference
+    otel_reconciler.go; wired by raw_kube_reconciler.go:80-101)."""
+    filter_cfg = {}
+    processors = ["batch"]
+    if metric_names:
+        filter_cfg = {
+            "filter/metrics": {
+                "metrics": {
+                    "include": {
+                        "match_type": "strict",
+                        "metric_names": list(metric_names),
+                    }
+                }
+            }
+        }
+        processors = ["filter/metrics", "batch"]
+    return {
+        "apiVersion": "opentelemetry.io/v1beta1",
+        "kind": "OpenTelemetryCollector",
+        "metadata": {"name": name, "namespace": namespace},
+        "spec": {
+            "mode": "sidecar",
+            "config": {
+                "receivers": {
+                    "prometheus": {
+                        "config": {
+                            "scrape_configs": [
+                                {
+                                    "job_name": "kserve-container",
+                                    "scrape_interval": f"{scrape_interval_s}s",
+                                    "static_configs": [
+                                        {"targets": ["localhost:8080"]}
+                                    ],
+                                }
+                            ]
+                        }
+                    }
+                },
+                "processors": {"batch": {}, **filter_cfg},
+                "exporters": {
+                    "otlp": {"endpoint": otlp_endpoint or "keda-otel-scaler:4317"}
+                },
+                "service": {
+                    "pipelines": {
+                        "metrics": {
+                            "receivers": ["prometheus"],
+                            "processors": processors,
+                            "exporters": ["otlp"],
+                        }
+                    }
+                },
+            },
+        },
+    }

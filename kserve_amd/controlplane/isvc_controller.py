@@ -68,6 +68,7 @@ MANAGED_GVKS = (
     "gateway.networking.k8s.io/v1/HTTPRoute",
     "networking.k8s.io/v1/Ingress",
     "serving.knative.dev/v1/Service",
+    "opentelemetry.io/v1beta1/OpenTelemetryCollector",
 )
 
 

@@ -393,3 +393,37 @@ class TestLiveISVCController:
         converge(isvc_ctrl, dep_ctrl)
         isvc = server.get(ISVC_GVK, "default", "iris")
         assert get_condition(isvc["status"], "Ready")["status"] == "True"
+
+
+def test_podmetrics_scale_metric_deploys_otel_collector(env):
+    """Custom (PodMetrics) scale metrics deploy the OTel collector sidecar
+    CR feeding the scaler (reference otel_reconciler.go wiring)."""
+    server, isvc_ctrl, dep_ctrl = env
+    obj = make_isvc()
+    obj["spec"]["predictor"]["minReplicas"] = 1
+    obj["spec"]["predictor"]["maxReplicas"] = 4
+    obj["spec"]["predictor"]["scaleMetric"] = "llm_tokens_per_second"
+    server.create(obj)
+    converge(isvc_ctrl, dep_ctrl)
+    otel = server.get(
+        "opentelemetry.io/v1beta1/OpenTelemetryCollector",
+        "default",
+        "iris-predictor",
+    )
+    cfg = otel["spec"]["config"]
+    inc = cfg["processors"]["filter/metrics"]["metrics"]["include"]
+    assert inc["metric_names"] == ["llm_tokens_per_second"]
+    assert otel["spec"]["mode"] == "sidecar"
+    # switch back to cpu -> collector pruned
+    cur = server.get(ISVC_GVK, "default", "iris")
+    cur["spec"]["predictor"]["scaleMetric"] = "cpu"
+    server.update(cur)
+    converge(isvc_ctrl, dep_ctrl)
+    assert (
+        server.try_get(
+            "opentelemetry.io/v1beta1/OpenTelemetryCollector",
+            "default",
+            "iris-predictor",
+        )
+        is None
+    )
