@@ -14,6 +14,7 @@
 // Replaces the vLLM paged-attention path the reference delegates to
 // (SURVEY.md §2.7 row 1).
 #include "common.h"
+#include "mfma_layouts.h"
 
 #define NEG_INF (-1e30f)
 
@@ -38,7 +39,7 @@ typedef __bf16 bf16x2v_t __attribute__((ext_vector_type(2)));
 // per-dim accumulator holds a 2-token partial and the cross-token sum is
 // deferred to a single epilogue reduction (ablation KS_ATTN_V4)
 template <int D, int HPW, int OCC = 1, bool PB = false, bool D2 = false,
-          bool V4 = false>
+          bool V4 = false, bool MQ = false>
 __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     short* __restrict__ out,            // [S, H, D] bf16
     const short* __restrict__ q,        // [S, H, D]
@@ -119,6 +120,18 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     }
   }
 
+  // MQ: the wave's q row as MFMA A-fragments (lane supplies k-chunk
+  // (lane>>4)*8 of every 32-dim mfma step; row index is lane&15 but all
+  // lanes load the head's own q — rows 1-15 of C are ignored)
+  bf16x8_t q_mf[MQ ? D / 32 : 1];
+  if constexpr (MQ) {
+    const short* qb = q + (long)seq * q_row_stride + (long)heads[0] * D;
+#pragma unroll
+    for (int ki = 0; ki < D / 32; ++ki)
+      q_mf[ki] = *reinterpret_cast<const bf16x8_t*>(
+          qb + ki * 32 + ((lane >> 4) << 3));
+  }
+
   float m[HPW], l[HPW];
   float acc[HPW][ACC];
   float acc8[V4 ? 8 : 1];  // V4: 8-dim 2-token partial accumulator
@@ -151,6 +164,58 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
     const bool tok_valid = gtok < ctx && gtok >= wstart;
     const short* vpage =
         v_cache + (((long)block_id * num_kv_heads + kv_head) * PAGE) * D;
+
+    if constexpr (MQ && HPW == 1 && V4) {
+      // ---- MFMA QK (round-2 experiment): the page's 16 tokens form the
+      // B columns of four 16x16x32 MFMA steps (K^T fragments are
+      // contiguous 16-B reads: lane reads dims (lane>>4)*8 of token
+      // lane&15), freeing the VALU dot + 4-lane reduce entirely. Scores
+      // land in C row 0 = acc reg 0 of lanes 0-15 (token = lane). PV
+      // stays on the V4 wide-load path. ----
+      short8_t vwide4[4];
+      const int vdim = (lane & 15) * 8;
+      const int vtok4 = lane >> 4;
+#pragma unroll
+      for (int q4 = 0; q4 < 4; ++q4)
+        vwide4[q4] = *reinterpret_cast<const short8_t*>(
+            vpage + (q4 * 4 + vtok4) * D + vdim);
+      f32x4_t cfrag = f32x4_t{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ki = 0; ki < D / 32; ++ki) {
+        const bf16x8_t bk = *reinterpret_cast<const bf16x8_t*>(
+            page + (lane & 15) * D + ki * 32 + ((lane >> 4) << 3));
+        cfrag = mfma16x16x32(q_mf[ki], bk, cfrag);
+      }
+      const int mytok = bi * PAGE + (lane & 15);
+      const bool my_valid =
+          lane < 16 && mytok < ctx && mytok >= wstart;
+      float s = my_valid ? cfrag[0] * scale : NEG_INF;
+      float tmax = s;
+#pragma unroll
+      for (int off = 1; off < 64; off <<= 1)
+        tmax = fmaxf(tmax, __shfl_xor(tmax, off, 64));
+      const float m_new = fmaxf(m[0], tmax);
+      const float rescale = __expf(m[0] - m_new);
+      const float p = (s > NEG_INF) ? __expf(s - m_new) : 0.f;
+      float psum = p;
+#pragma unroll
+      for (int off = 1; off < 64; off <<= 1)
+        psum += __shfl_xor(psum, off, 64);
+      l[0] = l[0] * rescale + psum;
+      m[0] = m_new;
+      if (lane < 16) p_bc[wave][lane] = p;
+      asm volatile("s_waitcnt lgkmcnt(0)");
+#pragma unroll
+      for (int a = 0; a < 8; ++a) acc8[a] *= rescale;
+#pragma unroll
+      for (int q4 = 0; q4 < 4; ++q4) {
+        const float pt = p_bc[wave][q4 * 4 + vtok4];
+#pragma unroll
+        for (int a = 0; a < 8; ++a)
+          acc8[a] += pt * bf16_bits_to_float(vwide4[q4][a]);
+      }
+      continue;
+    }
 
     if constexpr (HPW == 1) {
       // fast path: both the K fragments and the V row words for the page
@@ -1042,6 +1107,25 @@ extern "C" hipError_t ks_paged_attention_decode(
                        (const int*)block_tables, (const int*)context_lens,
                        scale, num_kv_heads, group, max_blocks, q_row_stride,
                        n_splits, (float*)part_out, (float*)part_ml, window);
+    HIP_CHECK_KERNEL();
+    return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
+                               head_dim, n_splits, stream);
+  }
+
+  // MFMA-QK experiment (A/B: KS_ATTN_MQ=1): QK on the matrix pipe,
+  // V4 wide-load PV
+  static const bool use_mq = [] {
+    const char* e = getenv("KS_ATTN_MQ");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_mq && hpw == 1 && head_dim == 128) {
+    hipLaunchKernelGGL(
+        (paged_attention_kernel<128, 1, 1, true, false, true, true>), grid,
+        block, 0, stream, (short*)out, (const short*)q,
+        (const short*)k_cache, (const short*)v_cache,
+        (const int*)block_tables, (const int*)context_lens, scale,
+        num_kv_heads, group, max_blocks, q_row_stride, n_splits,
+        (float*)part_out, (float*)part_ml, window);
     HIP_CHECK_KERNEL();
     return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
                                head_dim, n_splits, stream);

@@ -36,7 +36,7 @@ def run_point(engine, SamplingParams, concurrency, prompt_len, max_tokens,
     for i in range(concurrency):
         prompt = [rng.randrange(vocab) for _ in range(prompt_len)]
         engine.add_request(prompt, sp, request_id=f"c{concurrency}-{i}")
-    ttfts = []
+    ttfts = {}
     done = 0
     decode_t0 = None
     decode_tokens = 0
@@ -48,12 +48,14 @@ def run_point(engine, SamplingParams, concurrency, prompt_len, max_tokens,
                 done += 1
         if decode_t0 is None:
             for r in list(engine.scheduler.running):
-                if r.first_token_time is not None:
-                    ttfts.append(r.first_token_time - t_submit)
-            if len(ttfts) >= concurrency:
+                if r.first_token_time is not None and r.request_id not in ttfts:
+                    ttfts[r.request_id] = r.first_token_time - t_submit
+            # decode starts once the whole wave has prefilled
+            if len(ttfts) >= concurrency and engine.scheduler.num_waiting == 0:
                 decode_t0 = now
         else:
             decode_tokens += sum(len(o.new_token_ids) for o in outs)
+    ttfts = list(ttfts.values())
     total = time.perf_counter() - t_submit
     decode_time = time.perf_counter() - decode_t0 if decode_t0 else total
     out_tokens = concurrency * max_tokens
