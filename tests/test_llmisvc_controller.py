@@ -1,0 +1,151 @@
+"""Live LLMInferenceService controller on the fake API server (reference
+llmisvc/controller_int_*_test.go envtest suites)."""
+
+import base64
+
+from kserve_amd.controlplane.apiserver import FakeAPIServer
+from kserve_amd.controlplane.controller import get_condition
+from kserve_amd.controlplane.isvc_controller import FakeDeploymentController
+from kserve_amd.controlplane.llmisvc_controller import (
+    LLM_GVK,
+    LLMInferenceServiceController,
+)
+from tests.test_live_controller import converge
+
+
+def make_llm(name="llama", **spec):
+    base = {
+        "baseRefs": ["kserve-config-llm-template"],
+        "model": {"name": "meta/llama-3-8b", "uri": "hf://meta/llama-3-8b"},
+        "workload": {"replicas": 1, "parallelism": {"tensor": 1}},
+    }
+    base.update(spec)
+    return {
+        "apiVersion": "serving.kserve.io/v1alpha2",
+        "kind": "LLMInferenceService",
+        "metadata": {"name": name, "namespace": "default"},
+        "spec": base,
+    }
+
+
+def build_env():
+    server = FakeAPIServer()
+    llm = LLMInferenceServiceController(server).build()
+    dep = FakeDeploymentController(server).build()
+    return server, llm, dep
+
+
+def test_decode_workload_converges_ready():
+    server, llm, dep = build_env()
+    server.create(make_llm())
+    converge(llm, dep)
+    d = server.get("apps/v1/Deployment", "default", "llama-decode")
+    args = d["spec"]["template"]["spec"]["containers"][0]["args"]
+    assert "--tensor-parallel-size=1" in args
+    assert server.get("v1/Service", "default", "llama-decode")
+    cr = server.get(LLM_GVK, "default", "llama")
+    assert get_condition(cr["status"], "DecodeReady")["status"] == "True"
+    assert get_condition(cr["status"], "Ready")["status"] == "True"
+    assert cr["status"]["url"].startswith("http://llama-decode.default")
+
+
+def test_prefill_pool_and_scheduler():
+    server, llm, dep = build_env()
+    server.create(
+        make_llm(
+            prefill={"replicas": 2, "parallelism": {"tensor": 1}},
+            scheduler={"enabled": True},
+        )
+    )
+    converge(llm, dep)
+    assert server.get("apps/v1/Deployment", "default", "llama-prefill")
+    epp = server.get("apps/v1/Deployment", "default", "llama-epp")
+    ports = epp["spec"]["template"]["spec"]["containers"][0]["ports"]
+    assert {"containerPort": 9002, "name": "grpc"} in ports
+    assert server.get("v1/Service", "default", "llama-epp")
+    cr = server.get(LLM_GVK, "default", "llama")
+    assert get_condition(cr["status"], "PrefillReady")["status"] == "True"
+
+
+def test_multi_node_renders_lws_and_stays_pending():
+    server, llm, dep = build_env()
+    server.create(
+        make_llm(workload={"replicas": 1,
+                           "parallelism": {"tensor": 8, "pipeline": 2}})
+    )
+    converge(llm, dep)
+    lws = server.get(
+        "leaderworkerset.x-k8s.io/v1/LeaderWorkerSet", "default",
+        "llama-decode",
+    )
+    assert lws["spec"]["leaderWorkerTemplate"]["size"] == 2
+    # nothing marks LWS ready in this env -> controller keeps it pending
+    cr = server.get(LLM_GVK, "default", "llama")
+    assert get_condition(cr["status"], "Ready")["status"] == "False"
+
+
+def test_invalid_router_sets_condition():
+    server, llm, dep = build_env()
+    server.create(
+        make_llm(router={"route": {"http": {"spec": {}, "refs": [{"name": "r"}]}}})
+    )
+    converge(llm, dep)
+    cr = server.get(LLM_GVK, "default", "llama")
+    rv = get_condition(cr["status"], "RouterValid")
+    assert rv["status"] == "False"
+    assert "mutually exclusive" in rv["message"]
+    assert server.try_get("apps/v1/Deployment", "default", "llama-decode") is None
+
+
+def test_unknown_base_ref_reports():
+    server, llm, dep = build_env()
+    server.create(make_llm(baseRefs=["nope"]))
+    converge(llm, dep)
+    cr = server.get(LLM_GVK, "default", "llama")
+    ready = get_condition(cr["status"], "Ready")
+    assert ready["reason"] == "ConfigMergeError"
+
+
+def test_cluster_config_cr_extends_presets():
+    server, llm, dep = build_env()
+    server.create(
+        {
+            "apiVersion": "serving.kserve.io/v1alpha2",
+            "kind": "LLMInferenceServiceConfig",
+            "metadata": {"name": "org-defaults", "namespace": "default"},
+            "spec": {"workload": {"maxNumSeqs": 2048}},
+        }
+    )
+    server.create(make_llm(baseRefs=["org-defaults"]))
+    converge(llm, dep)
+    d = server.get("apps/v1/Deployment", "default", "llama-decode")
+    args = d["spec"]["template"]["spec"]["containers"][0]["args"]
+    assert "--max_num_seqs=2048" in args
+
+
+def test_tls_secret_created_once():
+    server, llm, dep = build_env()
+    server.create(make_llm(tls={"selfSigned": True}))
+    converge(llm, dep)
+    sec = server.get("v1/Secret", "default", "llama-tls")
+    assert sec["type"] == "kubernetes.io/tls"
+    crt1 = sec["data"]["tls.crt"]
+    assert base64.b64decode(crt1).startswith(b"-----BEGIN CERTIFICATE")
+    # a second reconcile must NOT rotate the pair
+    cr = server.get(LLM_GVK, "default", "llama")
+    cr["spec"]["workload"]["replicas"] = 2
+    server.update(cr)
+    converge(llm, dep)
+    assert server.get("v1/Secret", "default", "llama-tls")["data"][
+        "tls.crt"
+    ] == crt1
+
+
+def test_delete_cascades():
+    server, llm, dep = build_env()
+    server.create(make_llm(scheduler={"enabled": True}))
+    converge(llm, dep)
+    server.delete(LLM_GVK, "default", "llama")
+    converge(llm, dep)
+    assert server.try_get("apps/v1/Deployment", "default", "llama-decode") is None
+    assert server.try_get("apps/v1/Deployment", "default", "llama-epp") is None
