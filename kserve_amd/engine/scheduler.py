@@ -262,13 +262,28 @@ class Scheduler:
             self.block_manager.free_cpu(req)
 
     # -- multi-step decode windows -------------------------------------------
+    # window length while admission is blocked: a waiting request was
+    # already unschedulable THIS step (seats or KV), so a short window
+    # cannot starve it — it bounds the extra admission delay instead of
+    # dropping to one-token steps under steady trickle-in traffic
+    BLOCKED_ADMISSION_WINDOW = 4
+
     def reserve_decode_window(self, batch: ScheduledBatch, max_k: int) -> int:
         """Largest k such that all scheduled requests can decode k steps with
-        no scheduling events (no prefill waiting, no swaps, no preemption,
-        no per-request finish before step k except stop-token/EOS which is
-        handled post-hoc). Reserves the KV blocks for the window."""
-        if max_k <= 1 or self.waiting or self.swapped or batch.preempted:
+        no scheduling events (no swaps, no preemption, no per-request
+        finish before step k except stop-token/EOS which is handled
+        post-hoc). Reserves the KV blocks for the window.
+
+        A non-empty waiting queue here means _schedule_prefill just failed
+        to admit its head (this method is only reached from the decode
+        branch), so decoding cannot make that request MORE blocked; the
+        window is merely capped to bound the admission latency."""
+        if max_k <= 1 or self.swapped or batch.preempted:
             return 1
+        if self.waiting:
+            max_k = min(max_k, self.BLOCKED_ADMISSION_WINDOW)
+            if max_k <= 1:
+                return 1
         reqs = batch.requests
         if not reqs:
             return 1

@@ -193,3 +193,40 @@ def test_decode_not_starved_when_seats_full():
     assert not b2.is_prefill, "decode must proceed while prefills wait"
     assert len(b2.requests) == 2
     assert sched.num_waiting == 2
+
+
+def test_window_capped_not_disabled_when_admission_blocked():
+    """Round 2: a waiting request that cannot be admitted (seats full) no
+    longer forces one-token decode steps — the window is capped to
+    BLOCKED_ADMISSION_WINDOW instead (it was unschedulable this step
+    either way)."""
+    from kserve_amd.engine.config import (
+        CacheConfig,
+        EngineConfig,
+        ModelConfig,
+        SchedulerConfig,
+    )
+    from kserve_amd.engine.engine import LLMEngine
+    from kserve_amd.engine.sampling_params import SamplingParams
+
+    cfg = EngineConfig(
+        model=ModelConfig.tiny(vocab_size=64),
+        cache=CacheConfig(block_size=4, num_gpu_blocks=256),
+        scheduler=SchedulerConfig(
+            max_num_seqs=2, max_num_batched_tokens=64, max_model_len=64,
+            multi_step=8,
+        ),
+        device="cpu",
+        eos_token_id=-1,
+    )
+    eng = LLMEngine(cfg)
+    sp = SamplingParams(temperature=0.0, max_tokens=24, ignore_eos=True)
+    for i in range(3):  # 3 requests, 2 seats -> one always waits
+        eng.add_request([1 + i, 2, 3], sp, request_id=f"r{i}")
+    # prefill the two admitted requests
+    eng.step()
+    assert eng.scheduler.num_waiting == 1
+    batch = eng.scheduler.schedule()
+    assert not batch.is_prefill
+    k = eng.scheduler.reserve_decode_window(batch, 8)
+    assert k == eng.scheduler.BLOCKED_ADMISSION_WINDOW  # capped, not 1
