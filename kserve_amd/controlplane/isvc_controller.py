@@ -56,6 +56,7 @@ from kserve_amd.controlplane.v1beta1 import (
 ISVC_GVK = "serving.kserve.io/v1beta1/InferenceService"
 SR_GVK = "serving.kserve.io/v1alpha1/ServingRuntime"
 CSR_GVK = "serving.kserve.io/v1alpha1/ClusterServingRuntime"
+CSC_GVK = "serving.kserve.io/v1alpha1/ClusterStorageContainer"
 FINALIZER = "inferenceservice.finalizers"
 
 # every kind the controller may create (for pruning + owned watches)
@@ -323,11 +324,27 @@ class InferenceServiceController:
                 sa_name
             )
 
+        # custom storage containers: CSC CRs override the injected
+        # storage-initializer by URI format
+        csc_spec = None
+        uri = isvc.spec.predictor.storage_uri
+        if uri:
+            from kserve_amd.controlplane.webhook import (
+                apply_storage_container,
+                resolve_storage_container,
+            )
+
+            csc_spec = resolve_storage_container(
+                uri, self.server.list(CSC_GVK)
+            )
+
         # apply all desired manifests with owner references
         applied_keys = set()
         for m in manifests:
             if cred and m.get("kind") == "Deployment":
                 inject_credentials(m["spec"]["template"], *cred)
+            if csc_spec and m.get("kind") == "Deployment":
+                apply_storage_container(m["spec"]["template"], csc_spec)
             m["metadata"].setdefault("namespace", namespace)
             m["metadata"].setdefault("labels", {})[
                 "serving.kserve.io/inferenceservice"

@@ -246,3 +246,86 @@ def reconcile_llm(llm: LLMInferenceService) -> Dict[str, object]:
     if sched is not None:
         out["scheduler"] = sched
     return out
+
+
+# -- scaling: WVA / KEDA (reference v1alpha2 ScalingSpec :548-690) -----------
+
+def render_scaling(llm: LLMInferenceService, scaling: Dict) -> List[Dict]:
+    """ScalingSpec → manifests. ``wva`` renders a
+    WorkloadVariantAutoscaler CR (llm-d WVA: per-variant profiles with
+    SLO targets); ``keda`` renders a ScaledObject on the decode
+    Deployment; ``hpa`` (default) is handled by the caller's HPA path.
+    Reference: ScalingSpec/WVASpec/KEDAScalingSpec
+    (llm_inference_service_types.go:548-690), KEDA Fallback floor
+    (:668-672)."""
+    out: List[Dict] = []
+    name = f"{llm.name}-decode"
+    wva = scaling.get("wva")
+    if wva:
+        out.append(
+            {
+                "apiVersion": "llmd.ai/v1alpha1",
+                "kind": "WorkloadVariantAutoscaler",
+                "metadata": {"name": name, "namespace": llm.namespace},
+                "spec": {
+                    "scaleTargetRef": {
+                        "apiVersion": "apps/v1",
+                        "kind": "Deployment",
+                        "name": name,
+                    },
+                    "minReplicas": wva.get("minReplicas", 1),
+                    "maxReplicas": wva.get("maxReplicas", 8),
+                    "sloTargets": {
+                        "ttftMs": wva.get("ttftMs", 500),
+                        "tpotMs": wva.get("tpotMs", 50),
+                    },
+                    "profiles": wva.get("profiles", []),
+                },
+            }
+        )
+    keda = scaling.get("keda")
+    if keda:
+        so = {
+            "apiVersion": "keda.sh/v1alpha1",
+            "kind": "ScaledObject",
+            "metadata": {"name": name, "namespace": llm.namespace},
+            "spec": {
+                "scaleTargetRef": {
+                    "apiVersion": "apps/v1",
+                    "kind": "Deployment",
+                    "name": name,
+                },
+                "minReplicaCount": keda.get("minReplicas", 1),
+                "maxReplicaCount": keda.get("maxReplicas", 8),
+                "triggers": keda.get("triggers", []),
+            },
+        }
+        if keda.get("fallback"):
+            # replica floor during metric outages (:668-672)
+            so["spec"]["fallback"] = {
+                "failureThreshold": keda["fallback"].get("failureThreshold", 3),
+                "replicas": keda["fallback"].get("replicas", 1),
+            }
+        out.append(so)
+    return out
+
+
+def render_inference_pool(llm: LLMInferenceService) -> Dict:
+    """Gateway API Inference Extension InferencePool selecting the decode
+    pods, with the EPP as the endpoint-picker extension (reference
+    scheduler.go:221-387 InferencePool v1/v1alpha2 wiring)."""
+    s = llm.spec.scheduler or SchedulerSpec()
+    return {
+        "apiVersion": "inference.networking.x-k8s.io/v1alpha2",
+        "kind": "InferencePool",
+        "metadata": {"name": llm.name, "namespace": llm.namespace},
+        "spec": {
+            "targetPortNumber": 8080,
+            "selector": {"app": f"{llm.name}-decode"},
+            "extensionRef": {
+                "name": f"{llm.name}-epp",
+                "portNumber": s.grpc_port,
+                "failureMode": "FailClose",
+            },
+        },
+    }

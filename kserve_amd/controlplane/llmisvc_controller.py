@@ -42,6 +42,8 @@ from kserve_amd.controlplane.llmisvc import (
     SchedulerSpec,
     TracingSpec,
     WorkloadSpec,
+    render_inference_pool,
+    render_scaling,
     render_scheduler,
     render_workload,
 )
@@ -60,6 +62,9 @@ MANAGED = (
     "leaderworkerset.x-k8s.io/v1/LeaderWorkerSet",
     "v1/Service",
     "v1/Secret",
+    "keda.sh/v1alpha1/ScaledObject",
+    "llmd.ai/v1alpha1/WorkloadVariantAutoscaler",
+    "inference.networking.x-k8s.io/v1alpha2/InferencePool",
 )
 
 
@@ -207,6 +212,7 @@ class LLMInferenceServiceController:
         sched = render_scheduler(llm)
         if sched is not None:
             apply(sched)
+            apply(render_inference_pool(llm))
             apply(
                 {
                     "apiVersion": "v1",
@@ -222,6 +228,8 @@ class LLMInferenceServiceController:
                     },
                 }
             )
+        for m in render_scaling(llm, merged.get("scaling", {}) or {}):
+            apply(m)
         if (raw_spec.get("tls") or {}).get("selfSigned"):
             # one-time self-signed pair; regenerate only if absent
             if self.server.try_get("v1/Secret", namespace, f"{name}-tls") is None:

@@ -139,3 +139,45 @@ def mutate_pod(
     pod = inject_storage_initializer(pod, storage_init_image)
     pod = inject_agent(pod, agent_image)
     return pod
+
+
+# -- ClusterStorageContainer resolution (reference
+# storage_initializer_injector.go:123-199; CR types
+# storage_container_types.go:29-73) ------------------------------------------
+
+def resolve_storage_container(uri: str, containers) -> Dict:
+    """First ClusterStorageContainer CR whose supportedUriFormats matches
+    the storage URI (prefix or regex), or None. The matching CR's container
+    spec overrides the default storage-initializer image/env/resources."""
+    import re as _re
+
+    for csc in containers or []:
+        spec = csc.get("spec", {}) or {}
+        for fmt in spec.get("supportedUriFormats", []) or []:
+            prefix = fmt.get("prefix")
+            if prefix and uri.startswith(prefix):
+                return spec.get("container")
+            regex = fmt.get("regex")
+            if regex and _re.match(regex, uri):
+                return spec.get("container")
+    return None
+
+
+def apply_storage_container(pod: Dict, container_spec: Dict) -> Dict:
+    """Overlay a resolved CSC container spec onto the injected
+    storage-initializer init container (image/env/resources; args keep the
+    (uri, dest) contract)."""
+    if not container_spec:
+        return pod
+    for init in pod.get("spec", {}).get("initContainers", []):
+        if init.get("name") == "storage-initializer":
+            if container_spec.get("image"):
+                init["image"] = container_spec["image"]
+            if container_spec.get("env"):
+                have = {e["name"] for e in init.setdefault("env", [])}
+                init["env"].extend(
+                    e for e in container_spec["env"] if e["name"] not in have
+                )
+            if container_spec.get("resources"):
+                init["resources"] = container_spec["resources"]
+    return pod

@@ -427,3 +427,33 @@ def test_podmetrics_scale_metric_deploys_otel_collector(env):
         )
         is None
     )
+
+
+def test_cluster_storage_container_overrides_initializer(env):
+    """ClusterStorageContainer CRs replace the storage-initializer image
+    for matching URI formats (reference
+    storage_initializer_injector.go:123-199)."""
+    server, isvc_ctrl, dep_ctrl = env
+    server.create(
+        {
+            "apiVersion": "serving.kserve.io/v1alpha1",
+            "kind": "ClusterStorageContainer",
+            "metadata": {"name": "custom-s3"},
+            "spec": {
+                "supportedUriFormats": [{"prefix": "s3://models/"}],
+                "container": {
+                    "name": "storage-initializer",
+                    "image": "corp/custom-s3-init:2",
+                    "env": [{"name": "S3_USE_ACCEL", "value": "1"}],
+                },
+            },
+        }
+    )
+    server.create(make_isvc())  # storageUri s3://models/iris
+    converge(isvc_ctrl, dep_ctrl)
+    dep = server.get(DEP, "default", "iris-predictor")
+    init = dep["spec"]["template"]["spec"]["initContainers"][0]
+    assert init["image"] == "corp/custom-s3-init:2"
+    assert {"name": "S3_USE_ACCEL", "value": "1"} in init["env"]
+    # args keep the (uri, dest) contract
+    assert init["args"][0] == "s3://models/iris"

@@ -149,3 +149,47 @@ def test_delete_cascades():
     converge(llm, dep)
     assert server.try_get("apps/v1/Deployment", "default", "llama-decode") is None
     assert server.try_get("apps/v1/Deployment", "default", "llama-epp") is None
+
+
+def test_scaling_wva_and_keda_with_fallback():
+    server, llm, dep = build_env()
+    server.create(
+        make_llm(
+            scaling={
+                "wva": {"minReplicas": 1, "maxReplicas": 6, "ttftMs": 300},
+                "keda": {
+                    "maxReplicas": 4,
+                    "triggers": [{"type": "prometheus",
+                                  "metadata": {"query": "q"}}],
+                    "fallback": {"replicas": 2},
+                },
+            }
+        )
+    )
+    converge(llm, dep)
+    wva = server.get(
+        "llmd.ai/v1alpha1/WorkloadVariantAutoscaler", "default", "llama-decode"
+    )
+    assert wva["spec"]["sloTargets"]["ttftMs"] == 300
+    so = server.get("keda.sh/v1alpha1/ScaledObject", "default", "llama-decode")
+    assert so["spec"]["fallback"]["replicas"] == 2
+    # removing scaling prunes both
+    cr = server.get(LLM_GVK, "default", "llama")
+    del cr["spec"]["scaling"]
+    server.update(cr)
+    converge(llm, dep)
+    assert server.try_get(
+        "llmd.ai/v1alpha1/WorkloadVariantAutoscaler", "default", "llama-decode"
+    ) is None
+
+
+def test_inference_pool_wired_to_epp():
+    server, llm, dep = build_env()
+    server.create(make_llm(scheduler={"enabled": True}))
+    converge(llm, dep)
+    pool = server.get(
+        "inference.networking.x-k8s.io/v1alpha2/InferencePool", "default",
+        "llama",
+    )
+    assert pool["spec"]["selector"] == {"app": "llama-decode"}
+    assert pool["spec"]["extensionRef"]["name"] == "llama-epp"
