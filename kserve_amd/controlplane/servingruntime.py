@@ -114,11 +114,43 @@ def default_cluster_runtimes(image_prefix: str = "kserve-amd") -> List[ServingRu
         rt("kserve-amd-sklearnserver", ["sklearn"], "kserve_amd.runtimes.sklearnserver"),
         rt("kserve-amd-xgbserver", ["xgboost"], "kserve_amd.runtimes.xgbserver"),
         rt("kserve-amd-lgbserver", ["lightgbm"], "kserve_amd.runtimes.lgbserver"),
+        rt("kserve-amd-paddleserver", ["paddle"], "kserve_amd.runtimes.paddleserver"),
+        rt("kserve-amd-pmmlserver", ["pmml"], "kserve_amd.runtimes.pmmlserver"),
+        ServingRuntime(
+            name="kserve-amd-predictiveserver",
+            supported_model_formats=[
+                SupportedModelFormat(name=f, auto_select=True, priority=0)
+                for f in ("sklearn", "xgboost", "lightgbm", "onnx")
+            ],
+            protocol_versions=["v1", "v2"],
+            container={
+                "name": "kserve-container",
+                "image": f"{image_prefix}/kserve-amd-predictiveserver:latest",
+                "command": ["python", "-m",
+                            "kserve_amd.runtimes.predictiveserver"],
+                "args": [
+                    "--model_name={{.Name}}",
+                    "--model_dir=/mnt/models",
+                    "--http_port=8080",
+                    "--grpc_port=8081",
+                ],
+            },
+        ),
+        rt("kserve-amd-autogluonserver", ["autogluon"], "kserve_amd.runtimes.autogluonserver"),
         rt(
             "kserve-amd-huggingfaceserver",
             ["huggingface"],
             "kserve_amd.runtimes.huggingfaceserver",
             protocols=("v1", "v2", "openai"),
+            workers=True,
+        ),
+        # the LLM-dedicated runtime (vllmserver row of the reference
+        # catalog): same module, engine backend forced
+        rt(
+            "kserve-amd-llmserver",
+            ["huggingface"],
+            "kserve_amd.runtimes.huggingfaceserver",
+            protocols=("openai", "v2"),
             workers=True,
         ),
     ]
