@@ -153,6 +153,8 @@ class Controller:
                 self.queue.add((ns, owner))
 
     def start_watches(self) -> None:
+        if self._watches:  # idempotent: build() subscribes, start() reuses
+            return
         w = self.server.watch(self.primary_gvk)
         self._watches.append((w, True))
         for g in self.owned_gvks:

@@ -128,3 +128,52 @@ class TestTrainedModelController:
         cr = server.get(TM_GVK, "default", "m-bad")
         ready = get_condition(cr["status"], "Ready")
         assert ready["status"] == "False" and ready["reason"] == "InvalidSpec"
+
+
+def test_manager_runs_all_controllers_threaded():
+    """cmd/manager equivalent: one process, every controller on live
+    threads, converging CRs end-to-end (threaded mode, not the test
+    drive loop)."""
+    import time
+
+    from kserve_amd.controlplane.apiserver import FakeAPIServer
+    from kserve_amd.controlplane.manager import build_controllers
+    from kserve_amd.controlplane.isvc_controller import (
+        FakeDeploymentController,
+    )
+    from tests.test_live_controller import make_isvc
+
+    server = FakeAPIServer()
+    controllers = build_controllers(server)
+    dep_ctrl = FakeDeploymentController(server).build()
+    controllers.append(dep_ctrl)
+    for c in controllers:
+        c.start()
+    try:
+        server.create(make_isvc(name="mgr-iris"))
+        server.create(make_tm("mgr-tm", isvc="mgr-iris", uri="s3://b/m"))
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            isvc = server.try_get(
+                "serving.kserve.io/v1beta1/InferenceService", "default",
+                "mgr-iris",
+            )
+            cm = server.try_get(
+                "v1/ConfigMap", "default", "modelconfig-mgr-iris-0"
+            )
+            if (
+                isvc
+                and get_condition(isvc.get("status", {}), "Ready")
+                and get_condition(isvc["status"], "Ready")["status"] == "True"
+                and cm is not None
+            ):
+                break
+            time.sleep(0.05)
+        isvc = server.get(
+            "serving.kserve.io/v1beta1/InferenceService", "default", "mgr-iris"
+        )
+        assert get_condition(isvc["status"], "Ready")["status"] == "True"
+        assert server.get("v1/ConfigMap", "default", "modelconfig-mgr-iris-0")
+    finally:
+        for c in controllers:
+            c.stop()
