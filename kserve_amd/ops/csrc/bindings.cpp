@@ -189,12 +189,25 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
   int Hkv = k_cache.size(1);
   int block_size = k_cache.size(2);
   int max_blocks = block_tables.size(1);
-  // split-context heuristic: fill the 256 CUs (guide §1: need >>256 wgs)
+  // split-context heuristic. PMC (2026-09-12, S=256 ctx=576): the V4
+  // kernel is memory-LATENCY-bound with WAIT/BUSY ~17x and only ~32
+  // waves/CU over the kernel's life — occupancy starvation, not
+  // bandwidth. Splitting raises wave count (latency hiding) at the cost
+  // of the fp32 partial traffic + reduce pass; target enough waves that
+  // every CU holds its resident maximum a few times over.
   int n_splits = 1;
   long base_wgs = (long)S * Hkv;
-  if (base_wgs < 512) {
-    n_splits = (int)((512 + base_wgs - 1) / base_wgs);
+  static const int split_override = [] {
+    const char* e = getenv("KS_ATTN_SPLIT");
+    return e ? atoi(e) : 0;  // 0 = heuristic
+  }();
+  if (split_override > 0) {
+    n_splits = split_override;
+  } else if (base_wgs < 4096) {
+    n_splits = (int)((4096 + base_wgs - 1) / base_wgs);
     if (n_splits > 16) n_splits = 16;
+  }
+  if (n_splits > 1) {
     int max_split_blocks = (max_blocks + n_splits - 1) / n_splits;
     if (max_split_blocks < 1) n_splits = 1;
   }
