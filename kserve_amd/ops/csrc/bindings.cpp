@@ -189,22 +189,29 @@ void paged_attention_decode(at::Tensor& out, at::Tensor& q,
   int Hkv = k_cache.size(1);
   int block_size = k_cache.size(2);
   int max_blocks = block_tables.size(1);
-  // split-context heuristic. PMC (2026-09-12, S=256 ctx=576): the V4
-  // kernel is memory-LATENCY-bound with WAIT/BUSY ~17x and only ~32
-  // waves/CU over the kernel's life — occupancy starvation, not
-  // bandwidth. Splitting raises wave count (latency hiding) at the cost
-  // of the fp32 partial traffic + reduce pass; target enough waves that
-  // every CU holds its resident maximum a few times over.
+  // split-context heuristic, measured A/B 2026-09-12 (gpurun_out/
+  // attn_split_*.log): splitting HURTS at S>=256 (3.69->3.43 TB/s, the
+  // fp32 partials + reduce outweigh extra waves) and for the group=8
+  // HPW=2 shapes (2.05->1.81), but the 512-workgroup group<=4 regime
+  // gains (S=64: 3.75->3.96 at 4 splits) and tiny batches need deep
+  // splits (S=8: 0.71 unsplit vs 1.6-1.7 at 8-16). Rule: group<=4 AND
+  // base<=512 -> clamp(2048/base, 2, 8); otherwise split only below
+  // 512 workgroups (the round-1 rule).
   int n_splits = 1;
   long base_wgs = (long)S * Hkv;
+  const int grp = H / Hkv;
   static const int split_override = [] {
     const char* e = getenv("KS_ATTN_SPLIT");
     return e ? atoi(e) : 0;  // 0 = heuristic
   }();
   if (split_override > 0) {
     n_splits = split_override;
-  } else if (base_wgs < 4096) {
-    n_splits = (int)((4096 + base_wgs - 1) / base_wgs);
+  } else if (grp <= 4 && base_wgs <= 512) {
+    n_splits = (int)(2048 / base_wgs);
+    if (n_splits < 2) n_splits = 2;
+    if (n_splits > 8) n_splits = 8;
+  } else if (base_wgs < 512) {
+    n_splits = (int)((512 + base_wgs - 1) / base_wgs);
     if (n_splits > 16) n_splits = 16;
   }
   if (n_splits > 1) {
