@@ -202,6 +202,9 @@ class SchedulerConfig:
     speculative_ngram: int = 0
     speculative_ngram_min: int = 2
     speculative_ngram_max: int = 3
+    # draft-MODEL speculation: tokens proposed per round by the draft
+    # model configured in EngineConfig.draft_model. 0 disables.
+    speculative_k: int = 0
 
 
 @dataclass
