@@ -228,4 +228,8 @@ class EngineConfig:
     enforce_eager: bool = False  # True disables hipGraph capture
     # weights: path to model dir (safetensors) or None for random init
     model_path: Optional[str] = None
+    # draft-model speculation: the small model's config (random-init unless
+    # loaded via engine.draft.load_hf_state_dict); pairs with
+    # scheduler.speculative_k
+    draft_model: Optional[ModelConfig] = None
     eos_token_id: int = 128001

@@ -33,12 +33,14 @@ class _Shadow:
     """BlockManager keys tables by request_id; the draft tracks its own
     tables with lightweight shadows of the live requests."""
 
-    __slots__ = ("request_id", "num_tokens", "block_table")
+    __slots__ = ("request_id", "num_tokens", "block_table",
+                 "num_cached_tokens")
 
     def __init__(self, request_id: str):
         self.request_id = request_id
         self.num_tokens = 0
         self.block_table: List[int] = []
+        self.num_cached_tokens = 0
 
 
 class DraftModelWorker:

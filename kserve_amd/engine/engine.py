@@ -171,6 +171,12 @@ class LLMEngine:
             model = self._build_model()
         self.model = model
         self.runner = ModelRunner(config, model, str(self.device))
+        # draft-model speculation (vLLM draft-model mode; see draft_worker)
+        self.draft = None
+        if config.draft_model is not None and config.scheduler.speculative_k > 0:
+            from kserve_amd.engine.draft_worker import DraftModelWorker
+
+            self.draft = DraftModelWorker(config, config.draft_model)
         num_blocks = self.runner.profile_and_allocate_kv()
         self.scheduler = Scheduler(
             config.scheduler, config.cache, num_blocks,
@@ -360,6 +366,19 @@ class LLMEngine:
                     ]
                     if any(drafts):
                         return self._run_spec_decode(batch, drafts)
+            # draft-MODEL speculation: a small model proposes k tokens,
+            # the main model verifies them in one paged-context forward
+            if self.draft is not None and pure_greedy and pp == 1:
+                k_cap = (
+                    self.scheduler.reserve_decode_window(
+                        batch, sched.speculative_k + 1
+                    )
+                    - 1
+                )
+                if k_cap > 0:
+                    drafts = self.draft.propose(batch.requests, k_cap)
+                    if any(drafts):
+                        return self._run_spec_decode(batch, drafts)
             # multi-step window: decode with no pending scheduling events
             # runs as back-to-back hipGraph replays. temperature>0 batches
             # use the sampled-graph variant (in-graph fused sampler).
@@ -500,6 +519,8 @@ class LLMEngine:
         self.scheduler.finish_requests(finished)
         for req in finished:
             self.runner.release_request(req.request_id)
+            if self.draft is not None:
+                self.draft.release(req.request_id)
         LLM_KV_USAGE.set(self.scheduler.block_manager.usage)
         return outputs
 
@@ -610,6 +631,8 @@ class LLMEngine:
         self.scheduler.finish_requests(finished)
         for req in finished:
             self.runner.release_request(req.request_id)
+            if self.draft is not None:
+                self.draft.release(req.request_id)
         return outputs
 
     def _run_decode_window(
@@ -675,6 +698,8 @@ class LLMEngine:
         self.scheduler.finish_requests(finished)
         for req in finished:
             self.runner.release_request(req.request_id)
+            if self.draft is not None:
+                self.draft.release(req.request_id)
         LLM_KV_USAGE.set(self.scheduler.block_manager.usage)
         return outputs
 

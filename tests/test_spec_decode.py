@@ -149,3 +149,107 @@ def test_all_features_combined_matches_plain():
     b0 = [o.output_token_ids for o in full.generate(prompts, sp).values()]
     b1 = [o.output_token_ids for o in full.generate(prompts, sp).values()]
     assert a == b0 == b1
+
+
+class TestDraftModelSpeculation:
+    """Draft-MODEL speculation (round 2): exact greedy outputs regardless
+    of draft quality; a perfect draft (same weights) accepts everything."""
+
+    def _engine(self, draft_seed_offset=1, k=3):
+        import dataclasses
+
+        from kserve_amd.engine.config import (
+            CacheConfig,
+            EngineConfig,
+            ModelConfig,
+            SchedulerConfig,
+        )
+        from kserve_amd.engine.engine import LLMEngine
+
+        mcfg = ModelConfig.tiny(vocab_size=256)
+        cfg = EngineConfig(
+            model=mcfg,
+            cache=CacheConfig(block_size=4, num_gpu_blocks=512),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256,
+                max_model_len=128, multi_step=1, speculative_k=k,
+            ),
+            device="cpu",
+            seed=11,
+            eos_token_id=-1,
+            draft_model=dataclasses.replace(mcfg) if k > 0 else None,
+        )
+        eng = LLMEngine(cfg)
+        if eng.draft is not None and draft_seed_offset == 0:
+            # perfect draft: copy the MAIN model's weights
+            eng.draft.model.load_state_dict(
+                eng.runner.model.state_dict()
+            )
+        return eng
+
+    def _plain(self):
+        from kserve_amd.engine.config import (
+            CacheConfig,
+            EngineConfig,
+            ModelConfig,
+            SchedulerConfig,
+        )
+        from kserve_amd.engine.engine import LLMEngine
+
+        cfg = EngineConfig(
+            model=ModelConfig.tiny(vocab_size=256),
+            cache=CacheConfig(block_size=4, num_gpu_blocks=512),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256,
+                max_model_len=128, multi_step=1,
+            ),
+            device="cpu",
+            seed=11,
+            eos_token_id=-1,
+        )
+        return LLMEngine(cfg)
+
+    def _generate(self, eng, prompts, n=16):
+        from kserve_amd.engine.sampling_params import SamplingParams
+
+        sp = SamplingParams(temperature=0.0, max_tokens=n, ignore_eos=True)
+        out = eng.generate(prompts, sp)
+        steps = getattr(eng, "_dbg_steps", None)
+        return [o.output_token_ids for o in out.values()]
+
+    def test_exact_greedy_with_imperfect_draft(self):
+        prompts = [[1, 2, 3, 4], [9, 8, 7]]
+        ref = self._generate(self._plain(), prompts)
+        got = self._generate(self._engine(draft_seed_offset=1), prompts)
+        assert got == ref
+
+    def test_perfect_draft_accepts_and_matches(self):
+        import time
+
+        prompts = [[5, 6, 7, 8]]
+        ref = self._generate(self._plain(), prompts)
+        eng = self._engine(draft_seed_offset=0)
+        # count engine steps: a perfect draft emits k+1 tokens per step
+        steps = 0
+        from kserve_amd.engine.sampling_params import SamplingParams
+
+        sp = SamplingParams(temperature=0.0, max_tokens=16, ignore_eos=True)
+        eng.add_request(prompts[0], sp, request_id="p")
+        toks = []
+        while eng.scheduler.has_unfinished() and steps < 40:
+            for o in eng.step():
+                toks.extend(o.new_token_ids)
+            steps += 1
+        assert toks == ref[0]
+        # 1 prefill + ceil(15/4)=4 spec rounds (k=3 -> up to 4/step)
+        assert steps <= 6, f"perfect draft should compress steps, got {steps}"
+
+    def test_draft_state_released_on_finish(self):
+        eng = self._engine()
+        prompts = [[1, 2, 3]]
+        self._generate(eng, prompts, n=8)
+        assert eng.draft._shadows == {}
+        bm = eng.draft.block_manager
+        # all blocks back on the free list (block 0 stays reserved)
+        assert len(bm._free) == bm.num_blocks - 1
+        assert bm._tables == {}
