@@ -562,3 +562,40 @@ def test_sliding_window_engine_graph_matches_eager():
     lf = prefill_logits(m_full).float()
     lw = prefill_logits(m_win).float()
     assert not torch.allclose(lf, lw, atol=1e-2, rtol=1e-2)
+
+
+def test_draft_model_speculation_on_gpu():
+    """Draft-model speculation on the native GPU path: outputs equal the
+    plain greedy stream; a same-weights draft compresses steps."""
+    import dataclasses
+
+    from kserve_amd.engine.engine import LLMEngine
+    from kserve_amd.engine.sampling_params import SamplingParams
+
+    base = _cfg(enforce_eager=False)
+    base.scheduler.multi_step = 1
+    prompts = [[1, 2, 3, 4, 5], [7, 8, 9]]
+    sp = SamplingParams(temperature=0.0, max_tokens=20, ignore_eos=True)
+
+    plain = LLMEngine(base)
+    ref = [o.output_token_ids for o in plain.generate(prompts, sp).values()]
+    del plain
+    torch.cuda.empty_cache()
+
+    cfg = _cfg(enforce_eager=False)
+    cfg.scheduler.multi_step = 1
+    cfg.scheduler.speculative_k = 3
+    cfg.draft_model = dataclasses.replace(cfg.model)
+    eng = LLMEngine(cfg)
+    eng.draft.model.load_state_dict(eng.runner.model.state_dict())
+    steps = 0
+    outs = {}
+    for p, rid in zip(prompts, ("a", "b")):
+        eng.add_request(p, sp, request_id=rid)
+    while eng.scheduler.has_unfinished() and steps < 60:
+        for o in eng.step():
+            outs.setdefault(o.request_id, []).extend(o.new_token_ids)
+        steps += 1
+    assert [outs["a"], outs["b"]] == ref
+    # perfect draft at k=3: ~20/4 spec rounds + prefill
+    assert steps <= 9, f"expected compressed steps, got {steps}"
