@@ -229,7 +229,8 @@ class EngineConfig:
     # weights: path to model dir (safetensors) or None for random init
     model_path: Optional[str] = None
     # draft-model speculation: the small model's config (random-init unless
-    # loaded via engine.draft.load_hf_state_dict); pairs with
-    # scheduler.speculative_k
+    # draft_model_path / engine.draft.load_hf_state_dict provides weights);
+    # pairs with scheduler.speculative_k
     draft_model: Optional[ModelConfig] = None
+    draft_model_path: Optional[str] = None
     eos_token_id: int = 128001

@@ -177,6 +177,14 @@ class LLMEngine:
             from kserve_amd.engine.draft_worker import DraftModelWorker
 
             self.draft = DraftModelWorker(config, config.draft_model)
+            if config.draft_model_path:
+                from kserve_amd.engine.weights import (
+                    load_safetensors_weights,
+                )
+
+                load_safetensors_weights(
+                    self.draft.model, config.draft_model_path
+                )
         num_blocks = self.runner.profile_and_allocate_kv()
         self.scheduler = Scheduler(
             config.scheduler, config.cache, num_blocks,

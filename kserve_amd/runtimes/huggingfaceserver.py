@@ -78,6 +78,13 @@ def build_model(args):
         )
         if getattr(args, "expert_parallel", False):
             model_cfg.expert_parallel = True
+        # draft-model speculation (vLLM --speculative-config equivalent)
+        draft_cfg = None
+        draft_dir = getattr(args, "speculative_draft_model", None)
+        if draft_dir:
+            draft_cfg = ModelConfig.from_hf_config(
+                os.path.join(draft_dir, "config.json")
+            )
         if model_cfg.sliding_window:
             logger.info(
                 "sliding_window=%d: attention bounded in-kernel to the last "
@@ -93,7 +100,15 @@ def build_model(args):
             scheduler=SchedulerConfig(
                 max_num_seqs=getattr(args, "max_num_seqs", 256),
                 max_model_len=getattr(args, "max_model_len", 8192),
+                speculative_ngram=getattr(args, "speculative_ngram", 0),
+                speculative_k=(
+                    getattr(args, "num_speculative_tokens", 3)
+                    if draft_cfg is not None
+                    else 0
+                ),
             ),
+            draft_model=draft_cfg,
+            draft_model_path=draft_dir,
             parallel=ParallelConfig(
                 tensor_parallel_size=getattr(args, "tensor_parallel_size", 1),
                 pipeline_parallel_size=getattr(args, "pipeline_parallel_size", 1),
@@ -149,6 +164,15 @@ def main(argv=None):
     parser.add_argument("--kv-cache-dtype", dest="kv_cache_dtype", default="auto",
                         choices=["auto", "fp8"])
     # LoRA adapter serving (reference: --enable-lora --lora-modules name=path)
+    parser.add_argument("--speculative-draft-model",
+                        dest="speculative_draft_model", default=None,
+                        help="HF dir of a small draft model (draft-model "
+                             "speculation; weights loaded from its "
+                             "safetensors)")
+    parser.add_argument("--num-speculative-tokens",
+                        dest="num_speculative_tokens", type=int, default=3)
+    parser.add_argument("--speculative-ngram", dest="speculative_ngram",
+                        type=int, default=0)
     parser.add_argument("--enable-lora", dest="enable_lora", action="store_true")
     parser.add_argument(
         "--lora-modules", dest="lora_modules", nargs="*", default=[],
