@@ -31,9 +31,49 @@ _LOCAL_PREFIX = "file://"
 
 class Storage:
     @staticmethod
+    def _apply_storage_spec_env() -> None:
+        """storage-spec mode (reference kserve_storage.py
+        _update_with_storage_spec :425): the credentials builder mounts one
+        key of the common storage-config Secret as STORAGE_CONFIG (JSON)
+        plus optional STORAGE_OVERRIDE_CONFIG; translate its fields into
+        the provider env vars the native clients read."""
+        raw = os.environ.get("STORAGE_CONFIG")
+        if not raw:
+            return
+        try:
+            spec = json.loads(raw)
+        except json.JSONDecodeError:
+            logger.warning("STORAGE_CONFIG is not valid JSON; ignoring")
+            return
+        override = os.environ.get("STORAGE_OVERRIDE_CONFIG")
+        if override:
+            try:
+                spec.update(json.loads(override))
+            except json.JSONDecodeError:
+                pass
+        mapping = {
+            "access_key_id": "AWS_ACCESS_KEY_ID",
+            "secret_access_key": "AWS_SECRET_ACCESS_KEY",
+            "endpoint_url": "AWS_ENDPOINT_URL",
+            "region": "AWS_DEFAULT_REGION",
+            "anonymous": "AWS_ANONYMOUS_CREDENTIAL",
+            "verify_ssl": "S3_VERIFY_SSL",
+            "ca_bundle": "AWS_CA_BUNDLE",
+            "account_name": "AZURE_STORAGE_ACCOUNT",
+            "account_key": "AZURE_STORAGE_ACCESS_KEY",
+            "sas_token": "AZURE_STORAGE_SAS_TOKEN",
+            "hdfs_namenode": "HDFS_NAMENODE",
+            "hf_token": "HF_TOKEN",
+        }
+        for key, env in mapping.items():
+            if key in spec and spec[key] is not None:
+                os.environ[env] = str(spec[key])
+
+    @staticmethod
     def download(uri: str, out_dir: Optional[str] = None) -> str:
         """Download the artifact(s) at ``uri`` into ``out_dir`` (defaults to
         a temp dir; the pod contract uses /mnt/models)."""
+        Storage._apply_storage_spec_env()
         logger.info("Copying contents of %s to local", uri)
         if out_dir is None:
             out_dir = tempfile.mkdtemp()

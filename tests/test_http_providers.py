@@ -404,3 +404,40 @@ class TestOCIRegistry:
             f"oci://{host}/acme/models:v1", str(tmp_path / "out")
         )
         assert (tmp_path / "out" / "models" / "weights.bin").exists()
+
+
+def test_storage_config_env_drives_s3(http_server, tmp_path, monkeypatch):
+    """storage-spec mode: STORAGE_CONFIG (the mounted storage-config
+    Secret key) supplies endpoint + credentials without individual env
+    vars (reference _update_with_storage_spec)."""
+    endpoint = http_server(FakeS3Handler)
+    monkeypatch.delenv("AWS_ENDPOINT_URL", raising=False)
+    monkeypatch.delenv("AWS_ACCESS_KEY_ID", raising=False)
+    monkeypatch.setenv(
+        "STORAGE_CONFIG",
+        json.dumps(
+            {
+                "type": "s3",
+                "endpoint_url": endpoint,
+                "access_key_id": S3_KEY_ID,
+                "secret_access_key": S3_SECRET,
+            }
+        ),
+    )
+    out = Storage.download("s3://bkt/models/iris", str(tmp_path / "o"))
+    assert sorted(os.listdir(out)) == ["metadata.json", "model.joblib", "sub"]
+
+
+def test_storage_override_config_wins(http_server, tmp_path, monkeypatch):
+    endpoint = http_server(FakeS3Handler)
+    monkeypatch.setenv(
+        "STORAGE_CONFIG",
+        json.dumps({"endpoint_url": "http://127.0.0.1:1",
+                    "access_key_id": S3_KEY_ID,
+                    "secret_access_key": S3_SECRET}),
+    )
+    monkeypatch.setenv(
+        "STORAGE_OVERRIDE_CONFIG", json.dumps({"endpoint_url": endpoint})
+    )
+    out = Storage.download("s3://bkt/models/iris", str(tmp_path / "o2"))
+    assert (tmp_path / "o2" / "model.joblib").exists()
