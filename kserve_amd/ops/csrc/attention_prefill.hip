@@ -309,7 +309,6 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
       const int rbase = wq0 + rt * 16;
       if (CAUSAL && t0 > rbase + 15) continue;  // fully-masked row-tile
       f32x4_t s_frag[KT2 / 16];
-      __builtin_amdgcn_s_setprio(1);  // T5: favor the QK MFMA burst
 #pragma unroll
       for (int n = 0; n < KT2 / 16; ++n) {
         s_frag[n] = f32x4_t{0.f, 0.f, 0.f, 0.f};
@@ -322,7 +321,6 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
           s_frag[n] = mfma16x16x32(a_q[rt][c], bk, s_frag[n]);
         }
       }
-      __builtin_amdgcn_s_setprio(0);
       float p[KT2 / 16][4];
       float rescale[4];
 #pragma unroll
@@ -371,7 +369,6 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
     asm volatile("s_waitcnt lgkmcnt(0)");
     __builtin_amdgcn_sched_barrier(0);
     // ---- PV: O[32 q][D] += P[32 q][64 t] @ V[64 t][D] ----
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int rt = 0; rt < 2; ++rt) {
       if (CAUSAL && t0 > wq0 + rt * 16 + 15) continue;
@@ -390,7 +387,6 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
         }
       }
     }
-    __builtin_amdgcn_s_setprio(0);
     asm volatile("s_waitcnt lgkmcnt(0)");
     __builtin_amdgcn_sched_barrier(0);
   }
@@ -557,7 +553,6 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
       const int rbase = wq0 + rt * 16;
       if (t0 > start_pos + rbase + 15) continue;
       f32x4_t s_frag[KT2 / 16];
-      __builtin_amdgcn_s_setprio(1);  // T5: favor the QK MFMA burst
 #pragma unroll
       for (int n = 0; n < KT2 / 16; ++n) {
         s_frag[n] = f32x4_t{0.f, 0.f, 0.f, 0.f};
@@ -570,7 +565,6 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
           s_frag[n] = mfma16x16x32(a_q[rt][c], bk, s_frag[n]);
         }
       }
-      __builtin_amdgcn_s_setprio(0);
       float p[KT2 / 16][4];
       float rescale[4];
 #pragma unroll
@@ -618,7 +612,6 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
     }
     asm volatile("s_waitcnt lgkmcnt(0)");
     __builtin_amdgcn_sched_barrier(0);
-    __builtin_amdgcn_s_setprio(1);  // T5: PV MFMA burst
 #pragma unroll
     for (int rt = 0; rt < 2; ++rt) {
       if (t0 > start_pos + wq0 + rt * 16 + 15) continue;
@@ -637,7 +630,6 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
         }
       }
     }
-    __builtin_amdgcn_s_setprio(0);
     asm volatile("s_waitcnt lgkmcnt(0)");
     __builtin_amdgcn_sched_barrier(0);
   }
