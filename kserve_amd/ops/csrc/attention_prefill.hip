@@ -219,7 +219,7 @@ constexpr int KT2 = 64;   // kv tokens per tile
 constexpr int QW2 = 32;   // q rows per wave
 constexpr int PSTR = 72;  // p_lds row stride (64 + 8 shorts: 2-way banks)
 
-template <int D, bool CAUSAL>  // D == 128
+template <int D, bool CAUSAL, bool AS = false>  // D == 128
 __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
     short* __restrict__ out, const short* __restrict__ q,
     const short* __restrict__ k, const short* __restrict__ v,
@@ -275,31 +275,78 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
   const int kv_limit = CAUSAL ? min(len, tile_base + NWAVES * QW2) : len;
   const int ntiles = (kv_limit + KT2 - 1) / KT2;
 
-  for (int kt = 0; kt < ntiles; ++kt) {
-    const int t0 = kt * KT2;
-    __syncthreads();
-    for (int i = threadIdx.x; i < KT2 * (D / 8); i += 256) {
+  // T14 async-stage split (guide: +17% attn): the NEXT tile's K/V global
+  // loads are ISSUED into registers during this tile's compute; the
+  // ds_writes land after the barrier, so HBM latency hides under the
+  // MFMA phase instead of serializing the staging loop.
+  constexpr int CHUNKS = AS ? (KT2 * (D / 8) + 255) / 256 : 1;
+  short8_t kreg[CHUNKS], vreg[CHUNKS];
+  auto issue_tile = [&](int tt0) {
+#pragma unroll
+    for (int n = 0; n < CHUNKS; ++n) {
+      const int i = threadIdx.x + n * 256;
       const int r = i / (D / 8);
       const int c8 = (i % (D / 8)) * 8;
-      const int tok = t0 + r;
-      short8_t val, vv;
+      const int tok = tt0 + r;
       if (tok < len) {
-        val = *reinterpret_cast<const short8_t*>(
+        kreg[n] = *reinterpret_cast<const short8_t*>(
             k + (long)(q_start + tok) * sk + (long)kv_head * D + c8);
-        vv = *reinterpret_cast<const short8_t*>(
+        vreg[n] = *reinterpret_cast<const short8_t*>(
             v + (long)(q_start + tok) * sv + (long)kv_head * D + c8);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          val[j] = 0;
-          vv[j] = 0;
+          kreg[n][j] = 0;
+          vreg[n][j] = 0;
         }
       }
-      *reinterpret_cast<short8_t*>(&k_tile[k_idx<D>(r, c8)]) = val;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) v_t[(c8 + j) * KT2 + (r ^ (((c8 + j) & 3) << 3))] = vv[j];
     }
+  };
+  auto commit_tile = [&]() {
+#pragma unroll
+    for (int n = 0; n < CHUNKS; ++n) {
+      const int i = threadIdx.x + n * 256;
+      const int r = i / (D / 8);
+      const int c8 = (i % (D / 8)) * 8;
+      *reinterpret_cast<short8_t*>(&k_tile[k_idx<D>(r, c8)]) = kreg[n];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v_t[(c8 + j) * KT2 + (r ^ (((c8 + j) & 3) << 3))] = vreg[n][j];
+    }
+  };
+  if constexpr (AS) issue_tile(0);
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int t0 = kt * KT2;
     __syncthreads();
+    if constexpr (AS) {
+      commit_tile();
+      __syncthreads();
+      if (kt + 1 < ntiles) issue_tile(t0 + KT2);
+    } else {
+      for (int i = threadIdx.x; i < KT2 * (D / 8); i += 256) {
+        const int r = i / (D / 8);
+        const int c8 = (i % (D / 8)) * 8;
+        const int tok = t0 + r;
+        short8_t val, vv;
+        if (tok < len) {
+          val = *reinterpret_cast<const short8_t*>(
+              k + (long)(q_start + tok) * sk + (long)kv_head * D + c8);
+          vv = *reinterpret_cast<const short8_t*>(
+              v + (long)(q_start + tok) * sv + (long)kv_head * D + c8);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            val[j] = 0;
+            vv[j] = 0;
+          }
+        }
+        *reinterpret_cast<short8_t*>(&k_tile[k_idx<D>(r, c8)]) = val;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v_t[(c8 + j) * KT2 + (r ^ (((c8 + j) & 3) << 3))] = vv[j];
+      }
+      __syncthreads();
+    }
     if (!active) continue;
     if (CAUSAL && t0 > wq0 + QW2 - 1) continue;
 
@@ -706,8 +753,24 @@ extern "C" hipError_t ks_flash_prefill_varlen(
   if (head_dim == 128 && !use_v1) {
     const int max_tiles2 = (max_seqlen + NWAVES * QW2 - 1) / (NWAVES * QW2);
     dim3 grid2(Hq, max_tiles2, num_seqs);
-    if (causal) {
+    static const bool use_as = [] {  // T14 A/B: KS_PREFILL_AS=1
+      const char* e = getenv("KS_PREFILL_AS");
+      return e != nullptr && e[0] == '1';
+    }();
+    if (causal && use_as) {
+      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, true, true>), grid2,
+                         dim3(256), 0, stream, (short*)out, (const short*)q,
+                         (const short*)k, (const short*)v,
+                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
+                         window);
+    } else if (causal) {
       hipLaunchKernelGGL((flash_prefill_v2_kernel<128, true>), grid2,
+                         dim3(256), 0, stream, (short*)out, (const short*)q,
+                         (const short*)k, (const short*)v,
+                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
+                         window);
+    } else if (use_as) {
+      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, false, true>), grid2,
                          dim3(256), 0, stream, (short*)out, (const short*)q,
                          (const short*)k, (const short*)v,
                          (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
