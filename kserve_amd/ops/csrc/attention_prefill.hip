@@ -32,7 +32,7 @@ __device__ __forceinline__ int v_idx(int d, int tok) {
   return d * KT + (tok ^ ((d & 3) << 3));
 }
 
-template <int D, bool CAUSAL>  // D == 128
+template <int D, bool CAUSAL, bool WIN = false>  // D == 128
 __global__ __launch_bounds__(256) void flash_prefill_kernel(
     short* __restrict__ out,      // [T, Hq, D] bf16
     const short* __restrict__ q,  // [T, Hq, D]
@@ -141,7 +141,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
         const int tok = t0 + n * 16 + MFMA_C_COL(lane);
         const bool valid =
             (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len) &&
-            (window <= 0 || tok > qrow - window);
+            (!WIN || tok > qrow - window);
         sv[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
       }
       float rowmax = group_reduce_max<16>(fmaxf(sv[0], sv[1]));
@@ -219,7 +219,7 @@ constexpr int KT2 = 64;   // kv tokens per tile
 constexpr int QW2 = 32;   // q rows per wave
 constexpr int PSTR = 72;  // p_lds row stride (64 + 8 shorts: 2-way banks)
 
-template <int D, bool CAUSAL, bool AS = false>  // D == 128
+template <int D, bool CAUSAL, bool AS = false, bool WIN = false>
 __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
     short* __restrict__ out, const short* __restrict__ q,
     const short* __restrict__ k, const short* __restrict__ v,
@@ -380,7 +380,7 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
           const int tok = t0 + n * 16 + MFMA_C_COL(lane);
           const bool valid =
               (!CAUSAL || tok <= qrow) && (qrow < len) && (tok < len) &&
-              (window <= 0 || tok > qrow - window);
+              (!WIN || tok > qrow - window);
           sv_[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
           rowmax = fmaxf(rowmax, sv_[n]);
         }
@@ -471,7 +471,7 @@ __global__ __launch_bounds__(256) void flash_prefill_v2_kernel(
 // ---------------------------------------------------------------------------
 // CT: cache element type — short (bf16) or unsigned char (fp8 E4M3,
 // converted to bf16 during LDS staging; MFMA math stays bf16)
-template <int D, typename CT = short>  // D == 128
+template <int D, typename CT = short, bool WIN = false>  // D == 128
 __global__ __launch_bounds__(256) void context_prefill_kernel(
     short* __restrict__ out,        // [Tq, Hq, D] bf16
     const short* __restrict__ q,    // [Tq, Hq, D] (row stride sq)
@@ -624,7 +624,7 @@ __global__ __launch_bounds__(256) void context_prefill_kernel(
         for (int n = 0; n < KT2 / 16; ++n) {
           const int tok = t0 + n * 16 + MFMA_C_COL(lane);
           const bool valid = (tok <= qpos) && (qrow < q_len) && (tok < ctx) &&
-                             (window <= 0 || tok > qpos - window);
+                             (!WIN || tok > qpos - window);
           sv_[n] = valid ? s_frag[n][reg] * scale : NEG_INF;
           rowmax = fmaxf(rowmax, sv_[n]);
         }
@@ -715,22 +715,22 @@ extern "C" hipError_t ks_context_prefill_varlen(
   const int max_tiles = (max_q_len + NWAVES * QW2 - 1) / (NWAVES * QW2);
   if (max_tiles == 0 || num_seqs == 0) return hipSuccess;
   dim3 grid(Hq, max_tiles, num_seqs);
+#define LAUNCH_CTX(CT, WW)                                                  \
+  hipLaunchKernelGGL((context_prefill_kernel<128, CT, WW>), grid,           \
+                     dim3(256), 0, stream, (short*)out, (const short*)q,    \
+                     (const CT*)k_cache, (const CT*)v_cache,                \
+                     (const int*)block_tables, (const int*)ctx_lens,        \
+                     (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq,  \
+                     window)
+  const bool winc = window > 0;
   if (fp8_cache) {
-    hipLaunchKernelGGL((context_prefill_kernel<128, unsigned char>), grid,
-                       dim3(256), 0, stream, (short*)out, (const short*)q,
-                       (const unsigned char*)k_cache,
-                       (const unsigned char*)v_cache,
-                       (const int*)block_tables, (const int*)ctx_lens,
-                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq,
-                       window);
+    if (winc) LAUNCH_CTX(unsigned char, true);
+    else LAUNCH_CTX(unsigned char, false);
   } else {
-    hipLaunchKernelGGL((context_prefill_kernel<128, short>), grid, dim3(256),
-                       0, stream, (short*)out, (const short*)q,
-                       (const short*)k_cache, (const short*)v_cache,
-                       (const int*)block_tables, (const int*)ctx_lens,
-                       (const int*)cu_seqlens_q, Hq, Hkv, max_nb, scale, sq,
-                       window);
+    if (winc) LAUNCH_CTX(short, true);
+    else LAUNCH_CTX(short, false);
   }
+#undef LAUNCH_CTX
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }
@@ -757,45 +757,43 @@ extern "C" hipError_t ks_flash_prefill_varlen(
       const char* e = getenv("KS_PREFILL_AS");
       return e != nullptr && e[0] == '1';
     }();
+#define LAUNCH_FP2(CC, AA, WW)                                              \
+  hipLaunchKernelGGL((flash_prefill_v2_kernel<128, CC, AA, WW>), grid2,     \
+                     dim3(256), 0, stream, (short*)out, (const short*)q,    \
+                     (const short*)k, (const short*)v,                      \
+                     (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,    \
+                     window)
+    const bool win = window > 0;
     if (causal && use_as) {
-      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, true, true>), grid2,
-                         dim3(256), 0, stream, (short*)out, (const short*)q,
-                         (const short*)k, (const short*)v,
-                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
-                         window);
+      if (win) LAUNCH_FP2(true, true, true);
+      else LAUNCH_FP2(true, true, false);
     } else if (causal) {
-      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, true>), grid2,
-                         dim3(256), 0, stream, (short*)out, (const short*)q,
-                         (const short*)k, (const short*)v,
-                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
-                         window);
+      if (win) LAUNCH_FP2(true, false, true);
+      else LAUNCH_FP2(true, false, false);
     } else if (use_as) {
-      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, false, true>), grid2,
-                         dim3(256), 0, stream, (short*)out, (const short*)q,
-                         (const short*)k, (const short*)v,
-                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
-                         window);
+      if (win) LAUNCH_FP2(false, true, true);
+      else LAUNCH_FP2(false, true, false);
     } else {
-      hipLaunchKernelGGL((flash_prefill_v2_kernel<128, false>), grid2,
-                         dim3(256), 0, stream, (short*)out, (const short*)q,
-                         (const short*)k, (const short*)v,
-                         (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,
-                         window);
+      if (win) LAUNCH_FP2(false, false, true);
+      else LAUNCH_FP2(false, false, false);
     }
+#undef LAUNCH_FP2
     HIP_CHECK_KERNEL();
     return hipSuccess;
   }
-#define LAUNCH_FP(DD, CC)                                                    \
-  hipLaunchKernelGGL((flash_prefill_kernel<DD, CC>), grid, dim3(256), 0,     \
-                     stream, (short*)out, (const short*)q, (const short*)k,  \
-                     (const short*)v, (const int*)cu_seqlens, Hq, Hkv,       \
-                     scale, sq, sk, sv, window)
+#define LAUNCH_FP(DD, CC, WW)                                               \
+  hipLaunchKernelGGL((flash_prefill_kernel<DD, CC, WW>), grid, dim3(256),   \
+                     0, stream, (short*)out, (const short*)q,               \
+                     (const short*)k, (const short*)v,                      \
+                     (const int*)cu_seqlens, Hq, Hkv, scale, sq, sk, sv,    \
+                     window)
+  const bool win1 = window > 0;
   if (head_dim == 128) {
-    if (causal) LAUNCH_FP(128, true);
-    else LAUNCH_FP(128, false);
+    if (causal) { if (win1) LAUNCH_FP(128, true, true); else LAUNCH_FP(128, true, false); }
+    else { if (win1) LAUNCH_FP(128, false, true); else LAUNCH_FP(128, false, false); }
   } else {
-    if (causal) LAUNCH_FP(64, true);
-    else LAUNCH_FP(64, false);
+    if (causal) { if (win1) LAUNCH_FP(64, true, true); else LAUNCH_FP(64, true, false); }
+    else { if (win1) LAUNCH_FP(64, false, true); else LAUNCH_FP(64, false, false); }
   }
 #undef LAUNCH_FP
   HIP_CHECK_KERNEL();
