@@ -37,6 +37,12 @@ import requests
 
 DOWNLOAD_WORKERS = int(os.environ.get("KSERVE_DOWNLOAD_WORKERS", "8"))
 EMPTY_SHA256 = hashlib.sha256(b"").hexdigest()
+# (connect, read) timeouts: a wedged object store must fail the download,
+# not hang the storage-initializer forever
+REQUEST_TIMEOUT = (
+    float(os.environ.get("KSERVE_STORAGE_CONNECT_TIMEOUT_S", "10")),
+    float(os.environ.get("KSERVE_STORAGE_READ_TIMEOUT_S", "600")),
+)
 
 
 # ---------------------------------------------------------------------------
@@ -168,7 +174,7 @@ class S3Client:
         url = f"{self.endpoint}{quote(path, safe='/')}" + (f"?{qs}" if qs else "")
         r = self.http.request(
             method, url, headers=headers, data=body,
-            verify=self.verify_ssl, stream=True,
+            verify=self.verify_ssl, stream=True, timeout=REQUEST_TIMEOUT,
         )
         if r.status_code >= 400:
             raise RuntimeError(
@@ -292,7 +298,10 @@ class AzureBlobClient:
             ]
         qs = "&".join(f"{quote(k, safe='~')}={quote(v, safe='~')}" for k, v in q)
         url = f"{self.endpoint}{quote(path, safe='/')}" + (f"?{qs}" if qs else "")
-        r = self.http.request(method, url, headers=headers, stream=True)
+        r = self.http.request(
+            method, url, headers=headers, stream=True,
+            timeout=REQUEST_TIMEOUT,
+        )
         if r.status_code >= 400:
             raise RuntimeError(
                 f"Azure {method} {path} failed: {r.status_code} {r.text[:300]}"
@@ -376,6 +385,7 @@ class GCSClient:
                 f"{self.endpoint}/storage/v1/b/{bucket}/o",
                 params=params,
                 headers=self._headers(),
+                timeout=REQUEST_TIMEOUT,
             )
             if r.status_code >= 400:
                 raise RuntimeError(f"GCS list failed: {r.status_code}")
@@ -391,6 +401,7 @@ class GCSClient:
             params={"alt": "media"},
             headers=self._headers(),
             stream=True,
+            timeout=REQUEST_TIMEOUT,
         )
         if r.status_code >= 400:
             raise RuntimeError(f"GCS get {name} failed: {r.status_code}")
@@ -446,7 +457,9 @@ class WebHDFSClient:
 
     def list_status(self, path: str) -> List[Dict]:
         r = self.http.get(
-            f"{self.namenode}/webhdfs/v1{path}", params=self._params("LISTSTATUS")
+            f"{self.namenode}/webhdfs/v1{path}",
+            params=self._params("LISTSTATUS"),
+            timeout=REQUEST_TIMEOUT,
         )
         if r.status_code >= 400:
             raise RuntimeError(f"webhdfs LISTSTATUS {path}: {r.status_code}")
@@ -458,6 +471,7 @@ class WebHDFSClient:
             params=self._params("OPEN"),
             stream=True,
             allow_redirects=True,
+            timeout=REQUEST_TIMEOUT,
         )
         if r.status_code >= 400:
             raise RuntimeError(f"webhdfs OPEN {path}: {r.status_code}")
@@ -515,6 +529,7 @@ class OCIRegistryClient:
         r = self.http.get(
             f"{self.base}/v2/{name}/manifests/{reference}",
             headers=self._headers(self.MANIFEST_TYPES),
+            timeout=REQUEST_TIMEOUT,
         )
         if r.status_code >= 400:
             raise RuntimeError(
@@ -527,6 +542,7 @@ class OCIRegistryClient:
             f"{self.base}/v2/{name}/blobs/{digest}",
             headers=self._headers(),
             stream=True,
+            timeout=REQUEST_TIMEOUT,
         )
         if r.status_code >= 400:
             raise RuntimeError(f"OCI blob {digest}: {r.status_code}")

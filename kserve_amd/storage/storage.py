@@ -145,7 +145,11 @@ class Storage:
             m = re.match(r"^HEADERS_(.+)$", k)
             if m:
                 headers[m.group(1).replace("_", "-")] = v
-        with requests.get(uri, stream=True, headers=headers) as r:
+        from kserve_amd.storage.http_providers import REQUEST_TIMEOUT
+
+        with requests.get(
+            uri, stream=True, headers=headers, timeout=REQUEST_TIMEOUT
+        ) as r:
             r.raise_for_status()
             target = os.path.join(out_dir, name)
             with open(target, "wb") as f:
