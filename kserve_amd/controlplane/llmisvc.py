@@ -224,6 +224,15 @@ def render_scheduler(llm: LLMInferenceService, image="kserve-amd/endpoint-picker
                         {
                             "name": "main",
                             "image": image,
+                            "command": [
+                                "python", "-m",
+                                "kserve_amd.agent.endpoint_picker",
+                            ],
+                            "args": [
+                                "--endpoints",
+                                f"http://{llm.name}-decode.{llm.namespace}:80",
+                                "--port", str(s.grpc_port),
+                            ],
                             "ports": [
                                 {"containerPort": s.grpc_port, "name": "grpc"},
                                 {"containerPort": s.health_port, "name": "health"},
