@@ -180,7 +180,33 @@ def render_workload(
             },
         }
     # multi-node: LeaderWorkerSet (reference workload_multi_node.go:42-139)
+    # with the torchrun-rendezvous health probes (reference wires Ray
+    # health_check.py probes into the multinode runtime; ours probes the
+    # torch.distributed TCPStore instead — kserve_amd/parallel/health.py)
     group_size = p.pipeline
+    probe_cmd = ["python", "-m", "kserve_amd.parallel.health"]
+    liveness = {
+        "exec": {"command": probe_cmd + ["gpu"]},
+        "periodSeconds": 30,
+        "failureThreshold": 3,
+    }
+    readiness = {
+        "exec": {"command": probe_cmd + ["store"]},
+        "periodSeconds": 10,
+    }
+    leader_container = {**container, "livenessProbe": liveness,
+                        "readinessProbe": readiness}
+    worker_container = {
+        **container,
+        "livenessProbe": liveness,
+        "readinessProbe": readiness,
+        # workers wait for the leader's rendezvous endpoint to listen
+        "startupProbe": {
+            "exec": {"command": probe_cmd + ["master"]},
+            "periodSeconds": 5,
+            "failureThreshold": 60,
+        },
+    }
     return {
         "apiVersion": "leaderworkerset.x-k8s.io/v1",
         "kind": "LeaderWorkerSet",
@@ -191,11 +217,13 @@ def render_workload(
                 "size": group_size,
                 "leaderTemplate": {
                     "metadata": {"labels": {"app": name, "role": "leader"}},
-                    "spec": {"containers": [container], "volumes": volumes},
+                    "spec": {"containers": [leader_container],
+                             "volumes": volumes},
                 },
                 "workerTemplate": {
                     "metadata": {"labels": {"app": name, "role": "worker"}},
-                    "spec": {"containers": [container], "volumes": volumes},
+                    "spec": {"containers": [worker_container],
+                             "volumes": volumes},
                 },
             },
         },
