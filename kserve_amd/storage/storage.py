@@ -98,9 +98,11 @@ class Storage:
             return Storage._download_hdfs(uri, out_dir)
         if scheme in ("oci", "oci+native"):
             return Storage._download_oci(uri, out_dir)
+        if scheme.startswith("git+") or scheme == "git":
+            return Storage._download_git(uri, out_dir)
         raise ValueError(
             f"Cannot recognize storage type for {uri}; "
-            "supported: file, pvc, http(s), hf, s3, gs, wasb, hdfs, oci"
+            "supported: file, pvc, http(s), hf, s3, gs, wasb, hdfs, oci, git"
         )
 
     @staticmethod
@@ -282,4 +284,40 @@ class Storage:
         OCIRegistryClient(registry, insecure=insecure).pull_model(
             name, reference, out_dir
         )
+        return out_dir
+
+    @staticmethod
+    def _download_git(uri: str, out_dir: str) -> str:
+        """Clone a git repo (reference kserve_storage.py git provider):
+        ``git+https://host/repo.git[@ref][#subdir]`` — shallow-clones the
+        ref and copies (optionally only ``subdir``) into out_dir."""
+        import subprocess
+
+        raw = uri.split("git+", 1)[1] if uri.startswith("git+") else uri
+        raw, _, subdir = raw.partition("#")
+        clone_url, ref = raw, None
+        # an @ after the last / is a ref, not a userinfo separator
+        tail = raw.rsplit("/", 1)[-1]
+        if "@" in tail:
+            clone_url, ref = raw.rsplit("@", 1)
+        with tempfile.TemporaryDirectory() as tmp:
+            cmd = ["git", "clone", "--depth", "1"]
+            if ref:
+                cmd += ["--branch", ref]
+            subprocess.run(
+                cmd + [clone_url, tmp], check=True, capture_output=True,
+                timeout=600,
+            )
+            src = os.path.join(tmp, subdir) if subdir else tmp
+            if not os.path.isdir(src):
+                raise FileNotFoundError(f"{subdir!r} not in repository")
+            for entry in os.listdir(src):
+                if entry == ".git":
+                    continue
+                s = os.path.join(src, entry)
+                d = os.path.join(out_dir, entry)
+                if os.path.isdir(s):
+                    shutil.copytree(s, d, dirs_exist_ok=True)
+                else:
+                    shutil.copy2(s, d)
         return out_dir

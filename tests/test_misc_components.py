@@ -420,6 +420,37 @@ class TestStorage:
         with pytest.raises(Exception):
             Storage.download("carrier-pigeon://model", str(tmp_path / "o4"))
 
+    def test_git_clone_with_subdir(self, tmp_path):
+        import pathlib
+        import subprocess
+
+        from kserve_amd.storage.storage import Storage
+
+        repo = tmp_path / "src-repo"
+        (repo / "models" / "a").mkdir(parents=True)
+        (repo / "models" / "a" / "weights.bin").write_bytes(b"\x01\x02")
+        (repo / "README.md").write_text("top")
+        env = {
+            "GIT_AUTHOR_NAME": "t", "GIT_AUTHOR_EMAIL": "t@t",
+            "GIT_COMMITTER_NAME": "t", "GIT_COMMITTER_EMAIL": "t@t",
+            "HOME": str(tmp_path), "PATH": os.environ["PATH"],
+        }
+        for cmd in (
+            ["git", "init", "-q", "-b", "main"],
+            ["git", "add", "-A"],
+            ["git", "commit", "-q", "-m", "init"],
+        ):
+            subprocess.run(cmd, cwd=repo, check=True, env=env)
+
+        out = Storage.download(
+            f"git+file://{repo}#models/a", str(tmp_path / "git-out")
+        )
+        assert (pathlib.Path(out) / "weights.bin").read_bytes() == b"\x01\x02"
+        # whole-repo clone (no subdir) copies everything but .git
+        out2 = Storage.download(f"git+file://{repo}", str(tmp_path / "git-out2"))
+        assert (pathlib.Path(out2) / "README.md").exists()
+        assert not (pathlib.Path(out2) / ".git").exists()
+
 
 class TestQpextMerging:
     """Prometheus exposition merging (reference qpext sanitizeMetrics
