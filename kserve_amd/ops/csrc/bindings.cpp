@@ -404,7 +404,7 @@ void skinny_gemm(at::Tensor& out, at::Tensor& x, at::Tensor& w) {
             "skinny_gemm");
 }
 
-void gemm8(at::Tensor& d, at::Tensor& a, at::Tensor& w, bool swizzle) {
+void gemm8(at::Tensor& d, at::Tensor& a, at::Tensor& w, int64_t swizzle) {
   // EXPERIMENTAL (see gemm8.hip header): not used by ops.linear dispatch
   CHECK_BF16_CONTIG(d);
   CHECK_BF16_CONTIG(a);
@@ -413,7 +413,7 @@ void gemm8(at::Tensor& d, at::Tensor& a, at::Tensor& w, bool swizzle) {
   TORCH_CHECK(w.size(1) == K && d.size(0) == M && d.size(1) == N,
               "shape mismatch");
   check_hip(ks_gemm8(d.data_ptr(), a.data_ptr(), w.data_ptr(), M, N, K,
-                     swizzle ? 1 : 0, current_stream()),
+                     (int)swizzle, current_stream()),
             "gemm8");
 }
 

@@ -25,15 +25,25 @@ constexpr int FRAG_M = 8, FRAG_N = 4;   // 16x16 fragments per wave
 // per-quadrant fragment split: 4M x 2N, four quadrants per K-tile
 constexpr int QM = 4, QN = 2;
 
-// st_16x32 swizzle on a byte offset: XOR three row bits into the 16-B
-// chunk index (guide T2 recipe: byte ^= ((row&7)<<4), row = byte>>7 for a
-// 128-B row). Spreads a 16-lane column-slice read across all 32 banks —
-// the 1-bit (bit9->bit5) variant of round 1 left ~4-8-way conflicts
-// (measured 126M SQ_LDS_BANK_CONFLICT at 4096^3; PMC 2026-09-12).
-// Involution: bits >=7 are untouched, so applying twice is identity.
-template <bool SW>
+// LDS swizzle on a byte offset (applied identically at stage and read,
+// so any involution is correct; the choice only moves bank conflicts).
+//
+// SW=1 — guide T2 st_16x32: byte ^= ((row&7)<<4) (row = byte>>7).
+//   Fixed the 126M-conflict 1-bit variant (PMC 2026-09-12), but the
+//   fragment read pattern still 2-way conflicts: a quarter-wave reads
+//   rows i and i+8 at the same column, and (i&7) == ((i+8)&7) maps both
+//   to the SAME 16-B granule -> same bank pair.
+// SW=2 — conflict-free for this kernel's reads: XOR granule bits 4..7
+//   with (row>>1)&15 (byte bits 8..11, which the XOR does not touch, so
+//   it stays an involution). For a quarter-wave (16 consecutive rows,
+//   fixed column) the granule index becomes
+//     g(i) = [c ^ (i>>1)] | ((i&1)<<3)
+//   — a permutation of 0..15, i.e. the 16 lanes' 16-B reads tile all
+//   64 banks exactly once.
+template <int SW>
 __device__ __forceinline__ int swz(int byte_off) {
-  if constexpr (SW) return byte_off ^ (((byte_off >> 7) & 7) << 4);
+  if constexpr (SW == 1) return byte_off ^ (((byte_off >> 7) & 7) << 4);
+  if constexpr (SW == 2) return byte_off ^ (((byte_off >> 8) & 15) << 4);
   return byte_off;
 }
 
@@ -46,7 +56,7 @@ __device__ __forceinline__ int swz(int byte_off) {
 // swizzle is realised by permuting the per-lane global source byte while
 // the LDS write stays linear (HipKittens' pre-swizzled-source pattern):
 // LDS[lin] must hold data[swz(lin)] (swz is an XOR involution).
-template <bool SW>
+template <int SW>
 __device__ __forceinline__ void stage_half(
     const short* __restrict__ src,  // tile base (row 0, k 0 of this tile)
     long ld,                        // source leading dim (elements)
@@ -70,12 +80,45 @@ __device__ __forceinline__ void stage_half(
 }
 
 // read a 16-B bf16x8 from the (possibly swizzled) LDS tile
-template <bool SW>
+template <int SW>
 __device__ __forceinline__ bf16x8_t lds_frag(const short* lds_base,
                                              int row, int col) {
   const int byte = swz<SW>(row * 128 + col * 2);
   return *reinterpret_cast<const bf16x8_t*>(
       reinterpret_cast<const char*>(lds_base) + byte);
+}
+
+// Stage one 16 KiB unit cut to the PHASE-GRANULAR read sets (the
+// half-granular re-cut NEXT.md item 1 derives): KIND 0 is an A
+// quarter-pair {rows r..r+63, 128+r..128+r+63} (phases 0-1 read r=0,
+// phases 2-3 read r=64 of each wave's 128-row strip), KIND 1 is a B
+// qn-strip {rows r+64w..r+64w+31, w<4} (even phases read r=0, odd r=32).
+// Cutting stage units to these sets lets a unit be overwritten one or
+// two phases after its last read — the prerequisite for the counted
+// per-K-tile vmcnt (guide T3/T4) without a third LDS buffer.
+template <int SW, int KIND>
+__device__ __forceinline__ void stage_unit(
+    const short* __restrict__ src, long ld, int row_add, short* lds_base,
+    int wave, int lane) {
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int chunk = wave + i * 8;  // 16 chunks of 1 KiB (8 rows each)
+    int row0;
+    if constexpr (KIND == 0)
+      row0 = ((chunk >> 3) << 7) + ((chunk & 7) << 3);
+    else
+      row0 = ((chunk >> 2) << 6) + ((chunk & 3) << 3);
+    row0 += row_add;
+    const int lin = row0 * 128;                        // wave-uniform
+    const int sb = swz<SW>(lin + lane * 16);           // per-lane source
+    const int srow = sb >> 7;
+    const int scol = (sb & 127) >> 1;
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int*>(src + (long)srow * ld + scol),
+        reinterpret_cast<unsigned int*>(
+            reinterpret_cast<char*>(lds_base) + lin),
+        16, 0, 0);
+  }
 }
 
 // SCHED 0: stage one half-tile per phase, drain (vmcnt 0) at phase 3 —
@@ -86,7 +129,7 @@ __device__ __forceinline__ bf16x8_t lds_frag(const short* lds_base,
 //   phases in flight. (Step toward the guide's counted-vmcnt discipline —
 //   its full 3-half-tiles-in-flight schedule needs half-granular read
 //   ordering; this keeps the simple whole-tile flip.)
-template <bool SW, int SCHED = 0>
+template <int SW, int SCHED = 0>
 __global__ __launch_bounds__(512, 1) void gemm8_kernel(
     short* __restrict__ D,        // [M, N] bf16
     const short* __restrict__ A,  // [M, K] bf16
@@ -231,54 +274,192 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
   }
 }
 
+// Half-granular 8-phase schedule (guide T3+T4, the full counted-vmcnt
+// discipline; NEXT.md round-3 item 1). One 16 KiB stage unit per phase,
+// issued in future-consumption order, landing checked by ONE counted
+// s_waitcnt vmcnt(4) per K-tile (never 0 in the main loop):
+//
+//   ph0: stage B-qn1(kt+1) -> other buf   (region last read kt-1 ph3)
+//   ph1: stage A-qp1(kt+1) -> other buf   (last read kt-1 ph3)
+//   ph2: stage A-qp0(kt+2) -> THIS buf    (last read kt ph1 — in-place)
+//   ph3: stage B-qn0(kt+2) -> THIS buf    (last read kt ph2 — in-place)
+//
+// FIFO vmem retirement means vmcnt(4) at end of ph3 proves everything
+// up to A-qp1(kt+1) landed (only the two kt+2 units may remain in
+// flight), so every unit gets >=4 phases of memory latency hiding vs
+// <1 phase for the SCHED-0 drain. In-place staging into the live
+// buffer is safe because each phase ends with lgkmcnt(0)+s_barrier:
+// all waves' ds_reads of the overwritten region completed one phase
+// before the overwriting global_load_lds issues.
+template <int SW>
+__global__ __launch_bounds__(512, 1) void gemm8_hg_kernel(
+    short* __restrict__ D, const short* __restrict__ A,
+    const short* __restrict__ W, const int M, const int N, const int K) {
+  const int nwg = gridDim.x * gridDim.y;
+  const int orig = blockIdx.y * gridDim.x + blockIdx.x;
+  const int q = nwg / 8, r = nwg % 8;
+  const int xcd = orig % 8, idx = orig / 8;
+  const int wgid =
+      (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  const int tiles_n = N / BN;
+  const int tile_m = (wgid / tiles_n) * BM;
+  const int tile_n = (wgid % tiles_n) * BN;
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  const int wrow0 = wm * 128;
+  const int wcol0 = wn * 64;
+
+  extern __shared__ short lds[];
+
+  f32x4_t acc[FRAG_M][FRAG_N];
+#pragma unroll
+  for (int m = 0; m < FRAG_M; ++m)
+#pragma unroll
+    for (int n = 0; n < FRAG_N; ++n) acc[m][n] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+
+  const short* a_tile = A + (long)tile_m * K;
+  const short* w_tile = W + (long)tile_n * K;
+  const int ktiles = K / BK;
+
+  // ---- prologue: tile 0 fully (consumption order), then the two tile-1
+  // units the steady-state loop can't cover (they'd be kt-1 ph2/ph3) ----
+  stage_unit<SW, 0>(a_tile, K, 0, A_BUF(0), wave, lane);    // A-qp0(0)
+  stage_unit<SW, 1>(w_tile, K, 0, B_BUF(0), wave, lane);    // B-qn0(0)
+  stage_unit<SW, 1>(w_tile, K, 32, B_BUF(0), wave, lane);   // B-qn1(0)
+  stage_unit<SW, 0>(a_tile, K, 64, A_BUF(0), wave, lane);   // A-qp1(0)
+  if (ktiles > 1) {
+    stage_unit<SW, 0>(a_tile + BK, K, 0, A_BUF(1), wave, lane);
+    stage_unit<SW, 1>(w_tile + BK, K, 0, B_BUF(1), wave, lane);
+    asm volatile("s_waitcnt vmcnt(4)");  // tile-0 units landed (FIFO)
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int buf = kt & 1;
+    const short* ab = A_BUF(buf);
+    const short* bb = B_BUF(buf);
+    const bool have1 = kt + 1 < ktiles;
+    const bool have2 = kt + 2 < ktiles;
+    const short* a1 = a_tile + (kt + 1) * BK;
+    const short* w1 = w_tile + (kt + 1) * BK;
+    const short* a2 = a_tile + (kt + 2) * BK;
+    const short* w2 = w_tile + (kt + 2) * BK;
+
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      const int qm = (ph >> 1) * QM;
+      const int qn = (ph & 1) * QN;
+      bf16x8_t a_frag[QM][2];
+#pragma unroll
+      for (int m = 0; m < QM; ++m) {
+        const int row = wrow0 + (qm + m) * 16 + MFMA_RC_OF(lane);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          a_frag[m][ks] =
+              lds_frag<SW>(ab, row, ks * 32 + ((lane >> 4) << 3));
+      }
+      bf16x8_t b_frag[QN][2];
+#pragma unroll
+      for (int n = 0; n < QN; ++n) {
+        const int row = wcol0 + (qn + n) * 16 + MFMA_RC_OF(lane);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          b_frag[n][ks] =
+              lds_frag<SW>(bb, row, ks * 32 + ((lane >> 4) << 3));
+      }
+      // one stage unit per phase, future-consumption order
+      if (ph == 0 && have1)
+        stage_unit<SW, 1>(w1, K, 32, B_BUF(buf ^ 1), wave, lane);
+      else if (ph == 1 && have1)
+        stage_unit<SW, 0>(a1, K, 64, A_BUF(buf ^ 1), wave, lane);
+      else if (ph == 2 && have2)
+        stage_unit<SW, 0>(a2, K, 0, A_BUF(buf), wave, lane);
+      else if (ph == 3 && have2)
+        stage_unit<SW, 1>(w2, K, 0, B_BUF(buf), wave, lane);
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)");
+      __builtin_amdgcn_sched_barrier(0);  // MFMA must not hoist past (G#18)
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int m = 0; m < QM; ++m)
+#pragma unroll
+        for (int n = 0; n < QN; ++n)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[qm + m][qn + n] =
+                mfma16x16x32(a_frag[m][ks], b_frag[n][ks], acc[qm + m][qn + n]);
+      __builtin_amdgcn_s_setprio(0);
+      if (ph == 3 && have1) {
+        if (have2) {
+          // only the two kt+2 units (4 loads) may stay in flight
+          asm volatile("s_waitcnt vmcnt(4)");
+        } else {
+          // no ph2/ph3 stages were issued: vmcnt(4) would pass with the
+          // kt+1 units still in flight — drain instead (epilogue only)
+          asm volatile("s_waitcnt vmcnt(0)");
+        }
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+#pragma unroll
+  for (int m = 0; m < FRAG_M; ++m) {
+#pragma unroll
+    for (int n = 0; n < FRAG_N; ++n) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tile_m + wrow0 + m * 16 + MFMA_C_ROW(lane, reg);
+        const int col = tile_n + wcol0 + n * 16 + MFMA_C_COL(lane);
+        D[(long)row * N + col] = float_to_bf16_bits(acc[m][n][reg]);
+      }
+    }
+  }
+}
+
 }  // namespace
 
+// use_swizzle: 0 = none, 1 = T2 st_16x32 (3-bit), 2 = conflict-free 4-bit
+// (row>>1). Schedule via KS_GEMM8_SCHED: 0/1 = whole/front-loaded
+// half-tile double buffer, 2 = half-granular counted-vmcnt pipeline.
 extern "C" hipError_t ks_gemm8(void* d, const void* a, const void* w, int M,
                                int N, int K, int use_swizzle,
                                hipStream_t stream) {
   if (M % BM || N % BN || K % BK) return hipErrorInvalidValue;
+  if (use_swizzle < 0 || use_swizzle > 2) return hipErrorInvalidValue;
   dim3 grid(N / BN, M / BM);
   const size_t lds_bytes = 4 * BM * BK * sizeof(short);  // 128 KiB
+  using Kfn = void (*)(short*, const short*, const short*, int, int, int);
+  // [swizzle][sched] dispatch table; sched 2 = half-granular kernel
+  static const Kfn table[3][3] = {
+      {gemm8_kernel<0, 0>, gemm8_kernel<0, 1>, gemm8_hg_kernel<0>},
+      {gemm8_kernel<1, 0>, gemm8_kernel<1, 1>, gemm8_hg_kernel<1>},
+      {gemm8_kernel<2, 0>, gemm8_kernel<2, 1>, gemm8_hg_kernel<2>},
+  };
   // dynamic LDS above the 64 KiB default needs an explicit opt-in
   static bool attr_set = [] {
-    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<true, 0>,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        4 * BM * BK * sizeof(short));
-    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<false, 0>,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        4 * BM * BK * sizeof(short));
-    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<true, 1>,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        4 * BM * BK * sizeof(short));
-    (void)hipFuncSetAttribute((const void*)&gemm8_kernel<false, 1>,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        4 * BM * BK * sizeof(short));
+    for (int s = 0; s < 3; ++s)
+      for (int j = 0; j < 3; ++j)
+        (void)hipFuncSetAttribute((const void*)table[s][j],
+                                  hipFuncAttributeMaxDynamicSharedMemorySize,
+                                  4 * BM * BK * sizeof(short));
     return true;
   }();
   (void)attr_set;
   static const int sched = [] {
     const char* e = getenv("KS_GEMM8_SCHED");
-    return (e && e[0] == '1') ? 1 : 0;
+    const int s = e ? atoi(e) : 0;
+    return (s >= 0 && s <= 2) ? s : 0;
   }();
-  if (use_swizzle) {
-    if (sched == 1)
-      hipLaunchKernelGGL((gemm8_kernel<true, 1>), grid, dim3(512), lds_bytes,
-                         stream, (short*)d, (const short*)a, (const short*)w,
-                         M, N, K);
-    else
-      hipLaunchKernelGGL((gemm8_kernel<true, 0>), grid, dim3(512), lds_bytes,
-                         stream, (short*)d, (const short*)a, (const short*)w,
-                         M, N, K);
-  } else {
-    if (sched == 1)
-      hipLaunchKernelGGL((gemm8_kernel<false, 1>), grid, dim3(512),
-                         lds_bytes, stream, (short*)d, (const short*)a,
-                         (const short*)w, M, N, K);
-    else
-      hipLaunchKernelGGL((gemm8_kernel<false, 0>), grid, dim3(512),
-                         lds_bytes, stream, (short*)d, (const short*)a,
-                         (const short*)w, M, N, K);
-  }
+  hipLaunchKernelGGL(table[use_swizzle][sched], grid, dim3(512), lds_bytes,
+                     stream, (short*)d, (const short*)a, (const short*)w, M,
+                     N, K);
   HIP_CHECK_KERNEL();
   return hipSuccess;
 }

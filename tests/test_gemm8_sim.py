@@ -16,9 +16,14 @@ BK = 64
 QM, QN = 4, 2
 
 
-def swz(byte, on):
-    # T2 st_16x32: XOR three row bits into the 16-B chunk index
-    return byte ^ (((((byte >> 7) & 7) << 4)) if on else 0)
+def swz(byte, mode):
+    # mode 1 — T2 st_16x32: XOR three row bits into the 16-B chunk index
+    # mode 2 — conflict-free 4-bit: XOR (row>>1)&15 into granule bits 4-7
+    if mode == 1 or mode is True:
+        return byte ^ (((byte >> 7) & 7) << 4)
+    if mode == 2:
+        return byte ^ (((byte >> 8) & 15) << 4)
+    return byte
 
 
 def stage_tile(src_tile, swizzle):
@@ -60,7 +65,84 @@ def mfma_16x16x32(a_frags, b_frags, acc):
     return acc + A @ B
 
 
-@pytest.mark.parametrize("swizzle", [False, True])
+def stage_unit(src_tile, lds, kind, row_add, swizzle):
+    """Emulate gemm8.hip stage_unit: one 16 KiB phase-granular unit.
+    kind 0 = A quarter-pair (rows r..r+63 and 128+r..128+r+63),
+    kind 1 = B qn-strip (rows 64w+r..64w+r+31 for w<4)."""
+    flat = src_tile.reshape(-1)
+    for chunk in range(16):
+        if kind == 0:
+            row0 = ((chunk >> 3) << 7) + ((chunk & 7) << 3)
+        else:
+            row0 = ((chunk >> 2) << 6) + ((chunk & 3) << 3)
+        row0 += row_add
+        base = row0 * 128  # bytes
+        for lane in range(64):
+            for e in range(8):
+                dst_byte = base + lane * 16 + e * 2
+                src_byte = swz(base + lane * 16, swizzle) + e * 2
+                lds[dst_byte // 2] = flat[src_byte // 2]
+
+
+@pytest.mark.parametrize("swizzle", [0, 1, 2])
+def test_stage_unit_union_equals_full_tile(swizzle):
+    """The four half-granular units (A-qp0/A-qp1 or B-qn0/B-qn1 pairs)
+    together write exactly what stage_half writes, and each unit touches
+    only its liveness set (the rows its two phases read)."""
+    rng = np.random.default_rng(1)
+    tile = rng.standard_normal((BM, BK))
+    ref = stage_tile(tile, swizzle)
+
+    lds = np.full(BM * BK, np.nan)
+    stage_unit(tile, lds, 0, 0, swizzle)    # A-qp0
+    # rows covered so far: 0-63 and 128-191 only
+    view = lds.reshape(BM, BK)
+    assert not np.isnan(view[0:64]).any() and not np.isnan(view[128:192]).any()
+    assert np.isnan(view[64:128]).all() and np.isnan(view[192:256]).all()
+    stage_unit(tile, lds, 0, 64, swizzle)   # A-qp1
+    np.testing.assert_array_equal(lds, ref)
+
+    lds = np.full(BM * BK, np.nan)
+    stage_unit(tile, lds, 1, 0, swizzle)    # B-qn0
+    view = lds.reshape(BM, BK)
+    for w in range(4):
+        assert not np.isnan(view[w * 64 : w * 64 + 32]).any()
+        assert np.isnan(view[w * 64 + 32 : w * 64 + 64]).all()
+    stage_unit(tile, lds, 1, 32, swizzle)   # B-qn1
+    np.testing.assert_array_equal(lds, ref)
+
+
+def quarter_wave_bank_starts(swizzle, r0=0, col_bytes=0):
+    """LDS bank-group start (addr>>4 mod 16) for each lane of a hardware
+    quarter-wave reading rows r0..r0+15 at a fixed column. ds_read_b128 is
+    conflict-free iff the 16 starts are a permutation of 0..15 (each 16-B
+    read covers 4 banks; 16 x 4 = all 64 banks exactly once)."""
+    return [
+        (swz((r0 + i) * 128 + col_bytes, swizzle) >> 4) & 15
+        for i in range(16)
+    ]
+
+
+def test_swizzle2_reads_are_bank_conflict_free():
+    """Mode 2 (4-bit row>>1 XOR) makes every quarter-wave fragment read a
+    bank permutation; modes 0/1 do not (the measured residual conflicts).
+    Checked across all row offsets and fragment columns the kernel uses."""
+    for r0 in range(0, 256, 16):
+        for ks in (0, 1):
+            for g in range(4):
+                col_bytes = ks * 64 + g * 16
+                starts = quarter_wave_bank_starts(2, r0, col_bytes)
+                assert sorted(starts) == list(range(16)), (r0, col_bytes)
+    # and the involution property that makes any mode correct
+    for mode in (0, 1, 2):
+        for b in range(0, 32768, 97):
+            assert swz(swz(b, mode), mode) == b
+    # mode 1 leaves a 2-way conflict (rows i and i+8 share a granule)
+    starts1 = quarter_wave_bank_starts(1)
+    assert sorted(starts1) != list(range(16))
+
+
+@pytest.mark.parametrize("swizzle", [0, 1, 2])
 def test_gemm8_index_math_reproduces_matmul(swizzle):
     rng = np.random.default_rng(0)
     M = N = 256

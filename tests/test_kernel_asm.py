@@ -72,8 +72,8 @@ class TestDecodeAsm:
         asm = compile_to_asm(tmp_path, "gemm8")
         bodies = kernel_bodies(asm)
         g8 = {n: b for n, b in bodies.items() if "gemm8_kernel" in n}
-        # 2 swizzle x 2 stage-schedule instantiations
-        assert len(g8) == 4, f"expected 4 instantiations, got {list(g8)}"
+        # 3 swizzle x 2 stage-schedule instantiations
+        assert len(g8) == 6, f"expected 6 instantiations, got {list(g8)}"
         for name, body in g8.items():
             counts = {
                 "mfma": len(re.findall(r"v_mfma_f32_16x16x32_bf16", body)),
@@ -87,6 +87,31 @@ class TestDecodeAsm:
             assert counts["setprio"] == 8, (name, counts)
             assert counts["ds_read"] == 48, (name, counts)  # 12 x 4 phases
             assert counts["ds_write"] == 0, (name, counts)  # staging is DMA
+
+    def test_gemm8_half_granular_counted_vmcnt(self, tmp_path):
+        """SCHED=2 half-granular schedule: one 16 KiB unit staged per
+        phase, a single counted vmcnt(4) per K-tile in the main loop
+        (the guide's T3+T4 discipline — never drain to 0 mid-loop)."""
+        asm = compile_to_asm(tmp_path, "gemm8")
+        bodies = kernel_bodies(asm)
+        hg = {n: b for n, b in bodies.items()
+              if "gemm8_hg_kernel" in n and "stub" not in n
+              and "v_mfma" in b}
+        assert len(hg) == 3, f"expected 3 hg instantiations, got {list(hg)}"
+        for name, body in hg.items():
+            counts = {
+                "mfma": len(re.findall(r"v_mfma_f32_16x16x32_bf16", body)),
+                "glds": len(re.findall(r"global_load_lds", body)),
+                "vmcnt4": len(re.findall(r"s_waitcnt vmcnt\(4\)", body)),
+                "vmcnt0": len(re.findall(r"s_waitcnt vmcnt\(0\)", body)),
+                "ds_write": len(re.findall(r"ds_write", body)),
+            }
+            assert counts["mfma"] == 64, (name, counts)
+            # 12 prologue (6 units) + 8 main loop (4 units x 2 loads)
+            assert counts["glds"] == 20, (name, counts)
+            assert counts["vmcnt4"] >= 1, (name, counts)   # counted waits
+            assert counts["vmcnt0"] <= 2, (name, counts)   # epilogue only
+            assert counts["ds_write"] == 0, (name, counts)
 
 
 class TestSamplerAsm:

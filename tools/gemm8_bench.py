@@ -33,7 +33,7 @@ def time_fn(fn, iters=20):
     t1.record(); torch.cuda.synchronize()
     return t0.elapsed_time(t1) / iters * 1e3  # us
 
-print(f"{'M':>6} {'N':>6} {'K':>6} {'blaslt TF':>10} {'g8 TF':>8} {'g8+sw TF':>9}")
+print(f"{'M':>6} {'N':>6} {'K':>6} {'blaslt TF':>10} {'g8 TF':>8} {'sw1 TF':>8} {'sw2 TF':>8}")
 for M, N, K in SHAPES:
     a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
     w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
@@ -41,7 +41,7 @@ for M, N, K in SHAPES:
     fl = 2.0 * M * N * K
     t_ref = time_fn(lambda: torch.nn.functional.linear(a, w))
     row = [fl / (t_ref * 1e-6) / 1e12]
-    for sw in (False, True):
+    for sw in (0, 1, 2):
         if M % 256 or N % 256 or K % 64:
             row.append(float("nan")); continue
         try:
@@ -50,4 +50,4 @@ for M, N, K in SHAPES:
         except Exception as e:
             print("  gemm8 failed:", e)
             row.append(float("nan"))
-    print(f"{M:6d} {N:6d} {K:6d} {row[0]:10.1f} {row[1]:8.1f} {row[2]:9.1f}")
+    print(f"{M:6d} {N:6d} {K:6d} {row[0]:10.1f} {row[1]:8.1f} {row[2]:8.1f} {row[3]:8.1f}")
