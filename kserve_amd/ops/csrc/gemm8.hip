@@ -129,7 +129,11 @@ __device__ __forceinline__ void stage_unit(
 //   phases in flight. (Step toward the guide's counted-vmcnt discipline —
 //   its full 3-half-tiles-in-flight schedule needs half-granular read
 //   ordering; this keeps the simple whole-tile flip.)
-template <int SW, int SCHED = 0>
+// ABLATE (perf decomposition only — results are WRONG for !=0, guide
+// m233 methodology): 1 = no staging in the K-loop (ds_read+MFMA+barrier
+// structure on stale LDS), 2 = no per-phase ds_reads (fragments read
+// once; stage+MFMA+barriers), 3 = MFMA-only loop (matrix-pipe ceiling).
+template <int SW, int SCHED = 0, int ABLATE = 0>
 __global__ __launch_bounds__(512, 1) void gemm8_kernel(
     short* __restrict__ D,        // [M, N] bf16
     const short* __restrict__ A,  // [M, K] bf16
@@ -194,6 +198,8 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
       const int qn = (ph & 1) * QN;
       // ---- ds-read the quadrant's A/B fragments (12 x ds_read_b128) ----
       bf16x8_t a_frag[QM][2];
+      bf16x8_t b_frag[QN][2];
+      if constexpr (ABLATE < 2) {
 #pragma unroll
       for (int m = 0; m < QM; ++m) {
         const int row = wrow0 + (qm + m) * 16 + MFMA_RC_OF(lane);
@@ -202,7 +208,6 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
           a_frag[m][ks] =
               lds_frag<SW>(ab, row, ks * 32 + ((lane >> 4) << 3));
       }
-      bf16x8_t b_frag[QN][2];
 #pragma unroll
       for (int n = 0; n < QN; ++n) {
         const int row = wcol0 + (qn + n) * 16 + MFMA_RC_OF(lane);
@@ -211,10 +216,23 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
           b_frag[n][ks] =
               lds_frag<SW>(bb, row, ks * 32 + ((lane >> 4) << 3));
       }
+      } else {
+        // stale single fragment pair (timing structure only)
+#pragma unroll
+        for (int m = 0; m < QM; ++m)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            a_frag[m][ks] = lds_frag<SW>(ab, wrow0 + MFMA_RC_OF(lane), 0);
+#pragma unroll
+        for (int n = 0; n < QN; ++n)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            b_frag[n][ks] = lds_frag<SW>(bb, wcol0 + MFMA_RC_OF(lane), 0);
+      }
       // ---- stage NEXT K-tile half-tiles (schedule-dependent) ----
-      if (have_next) {
+      if (have_next && ABLATE != 1 && ABLATE != 3) {
         const int other = buf ^ 1;
-        if constexpr (SCHED == 0) {
+        if constexpr (SCHED != 1) {
           if (ph == 0) stage_half<SW>(a_next, K, 0, A_BUF(other), wave, lane);
           else if (ph == 1)
             stage_half<SW>(a_next, K, 128, A_BUF(other), wave, lane);
@@ -231,10 +249,18 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
           }
         }
       }
-      __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)");
-      __builtin_amdgcn_sched_barrier(0);  // MFMA must not hoist past (G#18)
-      __builtin_amdgcn_s_setprio(1);
+      // SCHED 3/4: no mid-phase barrier — in the SCHED-0 staging pattern
+      // the stage always targets the inactive buffer, so phase cohesion
+      // is a scheduling choice, not a correctness requirement.
+      if constexpr (SCHED < 3) __builtin_amdgcn_s_barrier();
+      // SCHED 6: no explicit wait/fence — the compiler tracks the
+      // ds_read->MFMA dependencies itself and inserts minimal counted
+      // lgkmcnt waits, software-pipelining reads into the MFMA burst.
+      if constexpr (SCHED < 6) {
+        asm volatile("s_waitcnt lgkmcnt(0)");
+        __builtin_amdgcn_sched_barrier(0);  // MFMA must not hoist past (G#18)
+      }
+      if constexpr (SCHED < 4) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int m = 0; m < QM; ++m)
 #pragma unroll
@@ -243,19 +269,28 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
           for (int ks = 0; ks < 2; ++ks)
             acc[qm + m][qn + n] =
                 mfma16x16x32(a_frag[m][ks], b_frag[n][ks], acc[qm + m][qn + n]);
-      __builtin_amdgcn_s_setprio(0);
+      if constexpr (SCHED < 4) __builtin_amdgcn_s_setprio(0);
       // drain staged loads before the buffer flips. SCHED 1: counted
       // vmcnt(4) at phase 2 retires the A halves early (FIFO), leaving
       // only the B halves for the phase-3 wait with 2 phases in flight.
       if constexpr (SCHED == 1) {
-        if (ph == 2 && have_next) {
+        if (ph == 2 && have_next && ABLATE == 0) {
           asm volatile("s_waitcnt vmcnt(4)");
         }
       }
-      if (ph == 3 && have_next) {
+      if (ph == 3 && have_next && (ABLATE == 0 || ABLATE == 2)) {
         asm volatile("s_waitcnt vmcnt(0)");
       }
-      __builtin_amdgcn_s_barrier();
+      // SCHED 5: barrier only at the K-tile boundary — within a tile
+      // every read hits the pre-staged buffer and every stage hits the
+      // other one, so the intermediate end-of-phase barriers
+      // synchronize nothing; only the post-vmcnt(0) flip point needs
+      // all waves' loads retired.
+      if constexpr (SCHED >= 5) {
+        if (ph == 3) __builtin_amdgcn_s_barrier();
+      } else {
+        __builtin_amdgcn_s_barrier();
+      }
     }
   }
 
@@ -423,6 +458,151 @@ __global__ __launch_bounds__(512, 1) void gemm8_hg_kernel(
   }
 }
 
+// Read-ahead schedule (SCHED 7): SCHED-5 sync structure (one barrier
+// per K-tile, no setprio) + one-phase ds_read readahead. Phase p's MFMA
+// runs while phase p+1's 12 ds_reads are in flight; the counted
+// s_waitcnt lgkmcnt(12) before the MFMA retires exactly phase p's reads
+// (LDS ops retire in order) and leaves p+1's outstanding. Doubles the
+// live fragment set (2 x 48 VGPRs) so the allocator moves the 128-VGPR
+// accumulator file to AGPRs — exactly what they exist for.
+template <int SW>
+__global__ __launch_bounds__(512, 1) void gemm8_ra_kernel(
+    short* __restrict__ D, const short* __restrict__ A,
+    const short* __restrict__ W, const int M, const int N, const int K) {
+  const int nwg = gridDim.x * gridDim.y;
+  const int orig = blockIdx.y * gridDim.x + blockIdx.x;
+  const int q = nwg / 8, r = nwg % 8;
+  const int xcd = orig % 8, idx = orig / 8;
+  const int wgid =
+      (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  const int tiles_n = N / BN;
+  const int tile_m = (wgid / tiles_n) * BM;
+  const int tile_n = (wgid % tiles_n) * BN;
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  const int wrow0 = wm * 128;
+  const int wcol0 = wn * 64;
+
+  extern __shared__ short lds[];
+
+  f32x4_t acc[FRAG_M][FRAG_N];
+#pragma unroll
+  for (int m = 0; m < FRAG_M; ++m)
+#pragma unroll
+    for (int n = 0; n < FRAG_N; ++n) acc[m][n] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+
+  const short* a_tile = A + (long)tile_m * K;
+  const short* w_tile = W + (long)tile_n * K;
+  const int ktiles = K / BK;
+
+  stage_half<SW>(a_tile, K, 0, A_BUF(0), wave, lane);
+  stage_half<SW>(a_tile, K, 128, A_BUF(0), wave, lane);
+  stage_half<SW>(w_tile, K, 0, B_BUF(0), wave, lane);
+  stage_half<SW>(w_tile, K, 128, B_BUF(0), wave, lane);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int buf = kt & 1;
+    const short* ab = A_BUF(buf);
+    const short* bb = B_BUF(buf);
+    const int nxt = kt + 1;
+    const short* a_next = a_tile + nxt * BK;
+    const short* w_next = w_tile + nxt * BK;
+    const bool have_next = nxt < ktiles;
+
+    // double fragment set: [ph & 1]
+    bf16x8_t a_frag[2][QM][2];
+    bf16x8_t b_frag[2][QN][2];
+
+    // reads for ph0 (buffer just validated by the tile-boundary barrier)
+#pragma unroll
+    for (int m = 0; m < QM; ++m)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        a_frag[0][m][ks] = lds_frag<SW>(
+            ab, wrow0 + m * 16 + MFMA_RC_OF(lane),
+            ks * 32 + ((lane >> 4) << 3));
+#pragma unroll
+    for (int n = 0; n < QN; ++n)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        b_frag[0][n][ks] = lds_frag<SW>(
+            bb, wcol0 + n * 16 + MFMA_RC_OF(lane),
+            ks * 32 + ((lane >> 4) << 3));
+
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      const int qm = (ph >> 1) * QM;
+      const int qn = (ph & 1) * QN;
+      const int cur = ph & 1;
+      if (ph < 3) {
+        const int nqm = ((ph + 1) >> 1) * QM;
+        const int nqn = ((ph + 1) & 1) * QN;
+#pragma unroll
+        for (int m = 0; m < QM; ++m)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            a_frag[cur ^ 1][m][ks] = lds_frag<SW>(
+                ab, wrow0 + (nqm + m) * 16 + MFMA_RC_OF(lane),
+                ks * 32 + ((lane >> 4) << 3));
+#pragma unroll
+        for (int n = 0; n < QN; ++n)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            b_frag[cur ^ 1][n][ks] = lds_frag<SW>(
+                bb, wcol0 + (nqn + n) * 16 + MFMA_RC_OF(lane),
+                ks * 32 + ((lane >> 4) << 3));
+      }
+      if (have_next) {
+        const int other = buf ^ 1;
+        if (ph == 0) stage_half<SW>(a_next, K, 0, A_BUF(other), wave, lane);
+        else if (ph == 1)
+          stage_half<SW>(a_next, K, 128, A_BUF(other), wave, lane);
+        else if (ph == 2)
+          stage_half<SW>(w_next, K, 0, B_BUF(other), wave, lane);
+        else stage_half<SW>(w_next, K, 128, B_BUF(other), wave, lane);
+      }
+      // retire exactly this phase's reads; leave the 12 readahead
+      // ds_reads in flight under the MFMA burst
+      if (ph < 3)
+        asm volatile("s_waitcnt lgkmcnt(12)");
+      else
+        asm volatile("s_waitcnt lgkmcnt(0)");
+      __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+      for (int m = 0; m < QM; ++m)
+#pragma unroll
+        for (int n = 0; n < QN; ++n)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[qm + m][qn + n] = mfma16x16x32(
+                a_frag[cur][m][ks], b_frag[cur][n][ks], acc[qm + m][qn + n]);
+      if (ph == 3) {
+        if (have_next) asm volatile("s_waitcnt vmcnt(0)");
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+  }
+
+#pragma unroll
+  for (int m = 0; m < FRAG_M; ++m) {
+#pragma unroll
+    for (int n = 0; n < FRAG_N; ++n) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tile_m + wrow0 + m * 16 + MFMA_C_ROW(lane, reg);
+        const int col = tile_n + wcol0 + n * 16 + MFMA_C_COL(lane);
+        D[(long)row * N + col] = float_to_bf16_bits(acc[m][n][reg]);
+      }
+    }
+  }
+}
+
 }  // namespace
 
 // use_swizzle: 0 = none, 1 = T2 st_16x32 (3-bit), 2 = conflict-free 4-bit
@@ -436,16 +616,25 @@ extern "C" hipError_t ks_gemm8(void* d, const void* a, const void* w, int M,
   dim3 grid(N / BN, M / BM);
   const size_t lds_bytes = 4 * BM * BK * sizeof(short);  // 128 KiB
   using Kfn = void (*)(short*, const short*, const short*, int, int, int);
-  // [swizzle][sched] dispatch table; sched 2 = half-granular kernel
-  static const Kfn table[3][3] = {
-      {gemm8_kernel<0, 0>, gemm8_kernel<0, 1>, gemm8_hg_kernel<0>},
-      {gemm8_kernel<1, 0>, gemm8_kernel<1, 1>, gemm8_hg_kernel<1>},
-      {gemm8_kernel<2, 0>, gemm8_kernel<2, 1>, gemm8_hg_kernel<2>},
+  // [swizzle][sched] dispatch; sched 2 = half-granular kernel,
+  // 3 = no mid-phase barrier, 4 = 3 + no setprio, 5 = 4 + barrier only
+  // at the K-tile boundary, 6 = 5 + compiler-scheduled waits,
+  // 7 = 5 + one-phase ds_read readahead (counted lgkmcnt, AGPR acc)
+  static const Kfn table[3][8] = {
+      {gemm8_kernel<0, 0>, gemm8_kernel<0, 1>, gemm8_hg_kernel<0>,
+       gemm8_kernel<0, 3>, gemm8_kernel<0, 4>, gemm8_kernel<0, 5>,
+       gemm8_kernel<0, 6>, gemm8_ra_kernel<0>},
+      {gemm8_kernel<1, 0>, gemm8_kernel<1, 1>, gemm8_hg_kernel<1>,
+       gemm8_kernel<1, 3>, gemm8_kernel<1, 4>, gemm8_kernel<1, 5>,
+       gemm8_kernel<1, 6>, gemm8_ra_kernel<1>},
+      {gemm8_kernel<2, 0>, gemm8_kernel<2, 1>, gemm8_hg_kernel<2>,
+       gemm8_kernel<2, 3>, gemm8_kernel<2, 4>, gemm8_kernel<2, 5>,
+       gemm8_kernel<2, 6>, gemm8_ra_kernel<2>},
   };
   // dynamic LDS above the 64 KiB default needs an explicit opt-in
   static bool attr_set = [] {
     for (int s = 0; s < 3; ++s)
-      for (int j = 0; j < 3; ++j)
+      for (int j = 0; j < 8; ++j)
         (void)hipFuncSetAttribute((const void*)table[s][j],
                                   hipFuncAttributeMaxDynamicSharedMemorySize,
                                   4 * BM * BK * sizeof(short));
@@ -454,9 +643,32 @@ extern "C" hipError_t ks_gemm8(void* d, const void* a, const void* w, int M,
   (void)attr_set;
   static const int sched = [] {
     const char* e = getenv("KS_GEMM8_SCHED");
-    const int s = e ? atoi(e) : 0;
-    return (s >= 0 && s <= 2) ? s : 0;
+    const int s = e ? atoi(e) : 5;  // 5 measured best (profiles/gemm8_ladder.md)
+    return (s >= 0 && s <= 7) ? s : 5;
   }();
+  // KS_GEMM8_ABLATE: m233-style decomposition (WRONG results; perf only)
+  static const int ablate = [] {
+    const char* e = getenv("KS_GEMM8_ABLATE");
+    const int s = e ? atoi(e) : 0;
+    return (s >= 1 && s <= 3) ? s : 0;
+  }();
+  if (ablate) {
+    // decomposition runs against the best schedule (5)
+    static const Kfn abl[4] = {nullptr, gemm8_kernel<1, 5, 1>,
+                               gemm8_kernel<1, 5, 2>, gemm8_kernel<1, 5, 3>};
+    static bool abl_attr = [] {
+      for (int i = 1; i < 4; ++i)
+        (void)hipFuncSetAttribute((const void*)abl[i],
+                                  hipFuncAttributeMaxDynamicSharedMemorySize,
+                                  4 * BM * BK * sizeof(short));
+      return true;
+    }();
+    (void)abl_attr;
+    hipLaunchKernelGGL(abl[ablate], grid, dim3(512), lds_bytes, stream,
+                       (short*)d, (const short*)a, (const short*)w, M, N, K);
+    HIP_CHECK_KERNEL();
+    return hipSuccess;
+  }
   hipLaunchKernelGGL(table[use_swizzle][sched], grid, dim3(512), lds_bytes,
                      stream, (short*)d, (const short*)a, (const short*)w, M,
                      N, K);

@@ -71,8 +71,13 @@ class TestDecodeAsm:
         around the MFMA bursts, counted (non-zero-only-at-boundary) vmcnt."""
         asm = compile_to_asm(tmp_path, "gemm8")
         bodies = kernel_bodies(asm)
-        g8 = {n: b for n, b in bodies.items() if "gemm8_kernel" in n}
-        # 3 swizzle x 2 stage-schedule instantiations
+        # canonical variants only: SCHED 0/1, ABLATE 0 (mangled
+        # gemm8_kernelILi<sw>ELi<sched>ELi0EE); sched 3/4 drop a barrier
+        # and ablate bodies intentionally drop instructions
+        g8 = {
+            n: b for n, b in bodies.items()
+            if re.search(r"gemm8_kernelILi[0-2]ELi[01]ELi0EE", n)
+        }
         assert len(g8) == 6, f"expected 6 instantiations, got {list(g8)}"
         for name, body in g8.items():
             counts = {

@@ -51,3 +51,44 @@ for M, N, K in SHAPES:
             print("  gemm8 failed:", e)
             row.append(float("nan"))
     print(f"{M:6d} {N:6d} {K:6d} {row[0]:10.1f} {row[1]:8.1f} {row[2]:8.1f} {row[3]:8.1f}")
+
+
+# m233-style structural decomposition (KS_GEMM8_ABLATE): results WRONG,
+# timing only — where do the non-MFMA cycles go?
+if os.environ.get("KS_GEMM8_DECOMP") == "1":
+    M = N = K = 4096
+    a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+    d = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
+    fl = 2.0 * M * N * K
+    import subprocess
+    names = {1: "no-stage (ds_read+MFMA+bar)",
+             2: "no-dsread (stage+MFMA+bar)",
+             3: "MFMA-only (+bar)"}
+    print("decomposition @4096^3 (sw1, sched5; separate processes):")
+    for abl, label in names.items():
+        out = subprocess.run(
+            ["python", "-c", f"""
+import os, sys
+os.environ['KS_GEMM8_ABLATE'] = '{abl}'
+sys.path.insert(0, {repr(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))})
+import torch, kserve_amd_C
+dev = 'cuda:0'
+M = N = K = 4096
+a = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+w = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+d = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
+for _ in range(3): kserve_amd_C.gemm8(d, a, w, 1)
+torch.cuda.synchronize()
+t0 = torch.cuda.Event(enable_timing=True); t1 = torch.cuda.Event(enable_timing=True)
+t0.record()
+for _ in range(20): kserve_amd_C.gemm8(d, a, w, 1)
+t1.record(); torch.cuda.synchronize()
+print(t0.elapsed_time(t1) / 20 * 1e3)
+"""],
+            capture_output=True, text=True)
+        try:
+            us = float(out.stdout.strip().splitlines()[-1])
+            print(f"  ablate {abl} {label:32s}: {fl / (us * 1e-6) / 1e12:7.1f} TF")
+        except Exception:
+            print(f"  ablate {abl} failed: {out.stderr[-200:]}")
