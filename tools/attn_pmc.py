@@ -21,7 +21,7 @@ bt = torch.arange(1, S * nb + 1, dtype=torch.int32, device=dev).reshape(S, nb)
 ct = torch.full((S,), ctx, dtype=torch.int32, device=dev)
 q = torch.randn(S, H, D, dtype=torch.bfloat16, device=dev)
 out = torch.empty_like(q)
-for _ in range(10):
+for _ in range(int(os.environ.get("KS_ATTN_PMC_ITERS", "10"))):
     ops.paged_attention_decode(q, kc, vc, bt, ct, 1.0 / math.sqrt(D), out=out)
 torch.cuda.synchronize()
 print("done")
