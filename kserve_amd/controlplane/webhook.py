@@ -25,6 +25,13 @@ ANN_BATCHER = "internal.serving.kserve.io/batcher"
 ANN_BATCHER_MAX_SIZE = "internal.serving.kserve.io/batcher-max-batchsize"
 ANN_BATCHER_MAX_LATENCY = "internal.serving.kserve.io/batcher-max-latency"
 ANN_AGENT = "internal.serving.kserve.io/agent"
+ANN_METRICS_AGG = "serving.kserve.io/enable-metric-aggregation"
+ANN_PROM_PORT = "serving.kserve.io/metrics-port"
+ANN_PROM_PATH = "serving.kserve.io/metrics-path"
+ANN_ACCELERATOR = "serving.kserve.io/accelerator"
+ANN_ISTIO_CNI = "sidecar.istio.io/interceptionMode"
+ANN_CA_BUNDLE = "serving.kserve.io/ca-bundle-configmap"
+GPU_RESOURCE = "amd.com/gpu"
 
 
 def inject_storage_initializer(pod: Dict, image: str) -> Dict:
@@ -129,6 +136,89 @@ def inject_agent(pod: Dict, image: str) -> Dict:
     return pod
 
 
+def inject_metrics_aggregator(pod: Dict) -> Dict:
+    """qpext wiring (reference metrics_aggregate_injector.go:94, env
+    :52): when aggregation is on, tell the queue-proxy extension where
+    the kserve container's Prometheus endpoint lives."""
+    ann = pod["metadata"].get("annotations", {})
+    if ann.get(ANN_METRICS_AGG) != "true":
+        return pod
+    env = [
+        {"name": "KSERVE_CONTAINER_PROMETHEUS_METRICS_PORT",
+         "value": ann.get(ANN_PROM_PORT, "8080")},
+        {"name": "KSERVE_CONTAINER_PROMETHEUS_METRICS_PATH",
+         "value": ann.get(ANN_PROM_PATH, "/metrics")},
+    ]
+    for c in pod["spec"]["containers"]:
+        if c.get("name") == "queue-proxy":
+            c.setdefault("env", []).extend(env)
+    # Prometheus scrapes one port per pod: point it at the aggregator
+    ann.setdefault("prometheus.kserve.io/port", "9088")
+    ann.setdefault("prometheus.kserve.io/path", "/metrics")
+    pod["metadata"]["annotations"] = ann
+    return pod
+
+
+def inject_accelerator_selector(pod: Dict) -> Dict:
+    """Node-pool pinning for GPU pods (reference mutator.go GKE
+    accelerator injector): a pod with amd.com/gpu limits and an
+    accelerator annotation gets the matching nodeSelector."""
+    ann = pod["metadata"].get("annotations", {})
+    acc = ann.get(ANN_ACCELERATOR)
+    if not acc:
+        return pod
+    wants_gpu = any(
+        GPU_RESOURCE in (c.get("resources", {}).get("limits") or {})
+        for c in pod["spec"].get("containers", [])
+    )
+    if wants_gpu:
+        pod["spec"].setdefault("nodeSelector", {})[
+            "kserve.amd.com/accelerator"] = acc
+    return pod
+
+
+def inject_ca_bundle(pod: Dict) -> Dict:
+    """Custom CA bundle for the storage-initializer (reference
+    storage_initializer_injector.go:918): mount the named ConfigMap and
+    point AWS_CA_BUNDLE/CA_BUNDLE_CONFIGMAP_NAME at it."""
+    ann = pod["metadata"].get("annotations", {})
+    cm = ann.get(ANN_CA_BUNDLE)
+    if not cm:
+        return pod
+    inits = pod["spec"].get("initContainers", [])
+    target = next((c for c in inits if c["name"] == "storage-initializer"),
+                  None)
+    if target is None:
+        return pod
+    pod["spec"].setdefault("volumes", []).append(
+        {"name": "cabundle-cert", "configMap": {"name": cm}}
+    )
+    target.setdefault("volumeMounts", []).append(
+        {"name": "cabundle-cert", "mountPath": "/etc/ssl/custom-certs"}
+    )
+    target.setdefault("env", []).extend([
+        {"name": "CA_BUNDLE_CONFIGMAP_NAME", "value": cm},
+        {"name": "AWS_CA_BUNDLE",
+         "value": "/etc/ssl/custom-certs/cabundle.crt"},
+    ])
+    return pod
+
+
+def inject_istio_cni_security_context(pod: Dict) -> Dict:
+    """Istio CNI compatibility (reference :802): init containers must
+    not run as UID 1337 (the proxy's UID) or traffic bypasses the mesh;
+    pin the storage-initializer to a distinct non-root UID."""
+    ann = pod["metadata"].get("annotations", {})
+    if ANN_ISTIO_CNI not in ann:
+        return pod
+    for c in pod["spec"].get("initContainers", []):
+        if c["name"] == "storage-initializer":
+            sc = c.setdefault("securityContext", {})
+            sc.setdefault("runAsUser", 1000)
+            sc.setdefault("runAsNonRoot", True)
+    return pod
+
+
 def mutate_pod(
     pod: Dict,
     storage_init_image: str = "kserve-amd/storage-initializer:latest",
@@ -138,6 +228,10 @@ def mutate_pod(
     pod = copy.deepcopy(pod)
     pod = inject_storage_initializer(pod, storage_init_image)
     pod = inject_agent(pod, agent_image)
+    pod = inject_metrics_aggregator(pod)
+    pod = inject_accelerator_selector(pod)
+    pod = inject_ca_bundle(pod)
+    pod = inject_istio_cni_security_context(pod)
     return pod
 
 

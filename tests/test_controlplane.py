@@ -208,6 +208,52 @@ class TestWebhook:
         assert len(out["spec"]["containers"]) == 1
         assert not out["spec"].get("initContainers")
 
+    def test_metrics_aggregator_env_on_queue_proxy(self):
+        pod = self._pod({
+            "serving.kserve.io/enable-metric-aggregation": "true",
+            "serving.kserve.io/metrics-port": "9099",
+        })
+        pod["spec"]["containers"].append({"name": "queue-proxy", "image": "qp"})
+        out = mutate_pod(pod)
+        qp = [c for c in out["spec"]["containers"]
+              if c["name"] == "queue-proxy"][0]
+        env = {e["name"]: e["value"] for e in qp["env"]}
+        assert env["KSERVE_CONTAINER_PROMETHEUS_METRICS_PORT"] == "9099"
+        assert env["KSERVE_CONTAINER_PROMETHEUS_METRICS_PATH"] == "/metrics"
+        assert out["metadata"]["annotations"][
+            "prometheus.kserve.io/port"] == "9088"
+
+    def test_accelerator_selector_requires_gpu_limits(self):
+        pod = self._pod({"serving.kserve.io/accelerator": "mi355x"})
+        out = mutate_pod(pod)
+        assert "nodeSelector" not in out["spec"]  # no GPU limits -> no pin
+        pod["spec"]["containers"][0]["resources"] = {
+            "limits": {"amd.com/gpu": "8"}
+        }
+        out = mutate_pod(pod)
+        assert out["spec"]["nodeSelector"][
+            "kserve.amd.com/accelerator"] == "mi355x"
+
+    def test_ca_bundle_mount_and_istio_cni_uid(self):
+        pod = self._pod({
+            "internal.serving.kserve.io/storage-initializer-sourceuri":
+                "s3://b/m",
+            "serving.kserve.io/ca-bundle-configmap": "corp-ca",
+            "sidecar.istio.io/interceptionMode": "REDIRECT",
+        })
+        out = mutate_pod(pod)
+        init = out["spec"]["initContainers"][0]
+        assert init["name"] == "storage-initializer"
+        env = {e["name"]: e["value"] for e in init["env"]}
+        assert env["CA_BUNDLE_CONFIGMAP_NAME"] == "corp-ca"
+        assert env["AWS_CA_BUNDLE"].endswith("cabundle.crt")
+        mounts = {m["name"] for m in init["volumeMounts"]}
+        assert "cabundle-cert" in mounts
+        assert {v["name"] for v in out["spec"]["volumes"]} >= {
+            "kserve-provision-location", "cabundle-cert"}
+        sc = init["securityContext"]
+        assert sc["runAsUser"] == 1000 and sc["runAsNonRoot"] is True
+
 
 class TestModelConfig:
     def test_build(self):
