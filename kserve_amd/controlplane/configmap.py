@@ -43,6 +43,9 @@ class DeployConfig:
     """configmap.go DeployConfig (:139-142)."""
 
     default_deployment_mode: str = "RawDeployment"  # or Serverless
+    # cluster supports native OCI ImageVolumes (drives the advisory
+    # condition, reference controller.go:774-831)
+    image_volume_available: bool = False
 
 
 @dataclass

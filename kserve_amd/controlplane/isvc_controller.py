@@ -391,6 +391,26 @@ class InferenceServiceController:
                 )
                 all_ready &= ready
 
+        # OCI ImageVolume advisory (reference controller.go:774-831):
+        # oci:// models served through the modelcar sidecar on clusters
+        # where the native ImageVolume feature is available get a
+        # non-blocking advisory condition nudging migration.
+        if uri and uri.startswith(("oci://", "oci+native://")):
+            iv_available = getattr(cfg.deploy, "image_volume_available",
+                                   False)
+            set_condition(
+                status, "OCIImageVolumeAdvisory",
+                "True" if iv_available else "False",
+                reason="ImageVolumeAvailable" if iv_available
+                else "ModelcarFallback",
+                message=(
+                    "cluster supports native OCI ImageVolumes; consider "
+                    "migrating off the modelcar sidecar"
+                    if iv_available else
+                    "oci:// model served via modelcar sidecar"
+                ),
+            )
+
         set_condition(status, "IngressReady", "True")
         set_condition(
             status, "Ready", "True" if all_ready else "False",

@@ -176,6 +176,37 @@ class TestLiveISVCController:
         assert isvc["status"]["url"].startswith("http://iris-default.")
         assert "inferenceservice.finalizers" in isvc["metadata"]["finalizers"]
 
+    def test_oci_image_volume_advisory_condition(self, env):
+        """oci:// models get the non-blocking advisory condition
+        (reference controller.go:774-831): ModelcarFallback by default,
+        flipping when the cluster advertises ImageVolume support."""
+        server, isvc_ctrl, dep_ctrl = env
+        obj = make_isvc(name="ocimodel")
+        obj["spec"]["predictor"]["model"]["storageUri"] = "oci://reg/m:1"
+        server.create(obj)
+        converge(isvc_ctrl, dep_ctrl)
+        isvc = server.get(ISVC_GVK, "default", "ocimodel")
+        adv = get_condition(isvc["status"], "OCIImageVolumeAdvisory")
+        assert adv["status"] == "False" and adv["reason"] == "ModelcarFallback"
+        # s3 models carry no advisory
+        server.create(make_isvc(name="plain"))
+        converge(isvc_ctrl, dep_ctrl)
+        plain = server.get(ISVC_GVK, "default", "plain")
+        assert get_condition(plain["status"], "OCIImageVolumeAdvisory") is None
+        # flip the cluster capability through the config system
+        server.create({
+            "apiVersion": "v1", "kind": "ConfigMap",
+            "metadata": {"name": "inferenceservice-config",
+                         "namespace": "kserve"},
+            "data": {"deploy": '{"imageVolumeAvailable": true}'},
+        })
+        isvc_ctrl.queue.add(("default", "ocimodel"))
+        converge(isvc_ctrl, dep_ctrl)
+        isvc = server.get(ISVC_GVK, "default", "ocimodel")
+        adv = get_condition(isvc["status"], "OCIImageVolumeAdvisory")
+        assert adv["status"] == "True"
+        assert adv["reason"] == "ImageVolumeAvailable"
+
     def test_requeues_until_pods_ready(self, env):
         server, isvc_ctrl, _ = env
         # deployment controller that stays unavailable for a few rounds
