@@ -478,7 +478,11 @@ __global__ __launch_bounds__(256, OCC) void paged_attention_kernel(
 // ---------------------------------------------------------------------------
 typedef unsigned int uint4_t __attribute__((ext_vector_type(4)));
 
-template <int D>  // D == 128
+// CVTB: QK chain via gfx950 scaled converts — fp8 pairs go straight to
+// bf16 (cvt_scalef32_pk_bf16_fp8, exact: e4m3 fits bf16's mantissa) and
+// dot against packed-bf16 q with fdot2_f32_bf16, replacing the
+// cvt_pk_f32 + 4xFMA chain (48 -> 32 VALU per lane per page on QK).
+template <int D, bool CVTB = false>  // D == 128
 __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
     short* __restrict__ out, const short* __restrict__ q,
     const unsigned char* __restrict__ k_cache,
@@ -523,12 +527,19 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
   if (!active) return;
 
   __shared__ float p_bc8[4][PAGE];
-  float q_frag[QFRAG];
+  float q_frag[CVTB ? 1 : QFRAG];
+  bf16x2v_t q_pk8[CVTB ? QFRAG / 2 : 1];
   {
     const short* qp =
         q + (long)seq * q_row_stride + (long)head * D + part * QFRAG;
+    if constexpr (CVTB) {
+      const bf16x2v_t* qp2 = reinterpret_cast<const bf16x2v_t*>(qp);
 #pragma unroll
-    for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
+      for (int j = 0; j < QFRAG / 2; ++j) q_pk8[j] = qp2[j];
+    } else {
+#pragma unroll
+      for (int j = 0; j < QFRAG; ++j) q_frag[j] = bf16_bits_to_float(qp[j]);
+    }
   }
   float m = NEG_INF, l = 0.f;
   // wide-V scheme (same as the bf16 V4 path): 8 lanes span a 128-B fp8
@@ -566,11 +577,21 @@ __global__ __launch_bounds__(256) void paged_attention_fp8_kernel(
 #pragma unroll
       for (int w = 0; w < 4; ++w) {
         const int ui = (int)kreg[c][w];
-        const float2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(ui, false);
-        const float2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(ui, true);
         const int j = c * 16 + w * 4;
-        s += q_frag[j] * lo[0] + q_frag[j + 1] * lo[1] +
-             q_frag[j + 2] * hi[0] + q_frag[j + 3] * hi[1];
+        if constexpr (CVTB) {
+          const bf16x2v_t lo =
+              __builtin_amdgcn_cvt_scalef32_pk_bf16_fp8(ui, 1.0f, false);
+          const bf16x2v_t hi =
+              __builtin_amdgcn_cvt_scalef32_pk_bf16_fp8(ui, 1.0f, true);
+          s = __builtin_amdgcn_fdot2_f32_bf16(lo, q_pk8[j / 2], s, false);
+          s = __builtin_amdgcn_fdot2_f32_bf16(hi, q_pk8[j / 2 + 1], s,
+                                              false);
+        } else {
+          const float2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(ui, false);
+          const float2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(ui, true);
+          s += q_frag[j] * lo[0] + q_frag[j + 1] * lo[1] +
+               q_frag[j + 2] * hi[0] + q_frag[j + 3] * hi[1];
+        }
       }
     }
     s = group_reduce_sum<4>(s);
@@ -1040,13 +1061,29 @@ extern "C" hipError_t ks_paged_attention_decode_fp8(
   if (group > NWAVES) return hipErrorInvalidValue;  // fp8: GQA group <= 4
   if (n_splits < 1) n_splits = 1;
   dim3 grid(num_kv_heads, num_seqs, n_splits);
-  hipLaunchKernelGGL((paged_attention_fp8_kernel<128>), grid, dim3(256), 0,
-                     stream, (short*)out, (const short*)q,
-                     (const unsigned char*)k_cache,
-                     (const unsigned char*)v_cache,
-                     (const int*)block_tables, (const int*)context_lens,
-                     scale, num_kv_heads, group, max_blocks, q_row_stride,
-                     n_splits, (float*)part_out, (float*)part_ml, window);
+  // scaled-convert QK chain (fp8->bf16 + fdot2) is the default;
+  // KS_FP8_CVTB=0 reverts to the cvt_pk_f32 + FMA chain for A/B.
+  static const bool cvtb = [] {
+    const char* e = getenv("KS_FP8_CVTB");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (cvtb) {
+    hipLaunchKernelGGL((paged_attention_fp8_kernel<128, true>), grid,
+                       dim3(256), 0, stream, (short*)out, (const short*)q,
+                       (const unsigned char*)k_cache,
+                       (const unsigned char*)v_cache,
+                       (const int*)block_tables, (const int*)context_lens,
+                       scale, num_kv_heads, group, max_blocks, q_row_stride,
+                       n_splits, (float*)part_out, (float*)part_ml, window);
+  } else {
+    hipLaunchKernelGGL((paged_attention_fp8_kernel<128, false>), grid,
+                       dim3(256), 0, stream, (short*)out, (const short*)q,
+                       (const unsigned char*)k_cache,
+                       (const unsigned char*)v_cache,
+                       (const int*)block_tables, (const int*)context_lens,
+                       scale, num_kv_heads, group, max_blocks, q_row_stride,
+                       n_splits, (float*)part_out, (float*)part_ml, window);
+  }
   HIP_CHECK_KERNEL();
   return launch_split_reduce(out, part_out, part_ml, num_seqs, num_heads,
                              head_dim, n_splits, stream);
