@@ -205,6 +205,9 @@ def main():
                 + (f"xtp{args.tp}" if args.tp > 1 else ""),
                 "kv_block_size": 16,
                 "hipgraph": not cfg.enforce_eager,
+                # headline runs use the bf16 KV default; fp8 KV is an
+                # explicit opt-in and must be visible in the record
+                "kv_cache_dtype": args.kv_cache_dtype,
             },
         }
         print(json.dumps(result), flush=True)
