@@ -93,6 +93,22 @@ class ModelConfig:
         )
 
     @classmethod
+    def mixtral_8x7b(cls) -> "ModelConfig":
+        return cls(
+            vocab_size=32000,
+            hidden_size=4096,
+            intermediate_size=14336,
+            num_layers=32,
+            num_heads=32,
+            num_kv_heads=8,
+            rope_theta=1e6,
+            max_position_embeddings=8192,
+            num_local_experts=8,
+            num_experts_per_tok=2,
+            model_name="mixtral-8x7b",
+        )
+
+    @classmethod
     def tiny(cls, vocab_size: int = 256) -> "ModelConfig":
         """Small config for CPU tests."""
         return cls(
