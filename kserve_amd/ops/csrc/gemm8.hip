@@ -121,6 +121,18 @@ __device__ __forceinline__ void stage_unit(
   }
 }
 
+// SW==1 fast fragment reads: the st_16x32 XOR key is (row&7)<<4 and the
+// fragment rows differ only by 16-multiples, so the key reduces to
+// (lane&7)<<4 — invariant across phases and K-tiles. The whole swizzled
+// per-lane byte offset hoists out of the loop (one value per ks), and
+// each fragment read becomes a single ds_read at base + a compile-time
+// immediate ((qm+m)*2048 / (qn+n)*2048). Verified exhaustively against
+// swz<1>() for every (lane, wave, quadrant, ks) by test_gemm8_sim.py.
+__device__ __forceinline__ int swz1_lane_low(int lane, int ks) {
+  const int low = ks * 64 + ((lane >> 4) << 4) + (lane & 15) * 128;
+  return low ^ ((lane & 7) << 4);
+}
+
 // SCHED 0: stage one half-tile per phase, drain (vmcnt 0) at phase 3 —
 //   the last load has <1 phase of MFMA to hide under.
 // SCHED 1: front-load the stages (A0+A1 at phase 0, B0+B1 at phase 1) with
@@ -200,6 +212,23 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
       bf16x8_t a_frag[QM][2];
       bf16x8_t b_frag[QN][2];
       if constexpr (ABLATE < 2) {
+      if constexpr (SW == 1) {
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          const char* abase = reinterpret_cast<const char*>(ab) +
+                              wm * 16384 + swz1_lane_low(lane, ks);
+          const char* bbase = reinterpret_cast<const char*>(bb) +
+                              wn * 8192 + swz1_lane_low(lane, ks);
+#pragma unroll
+          for (int m = 0; m < QM; ++m)
+            a_frag[m][ks] = *reinterpret_cast<const bf16x8_t*>(
+                abase + (qm + m) * 2048);
+#pragma unroll
+          for (int n = 0; n < QN; ++n)
+            b_frag[n][ks] = *reinterpret_cast<const bf16x8_t*>(
+                bbase + (qn + n) * 2048);
+        }
+      } else {
 #pragma unroll
       for (int m = 0; m < QM; ++m) {
         const int row = wrow0 + (qm + m) * 16 + MFMA_RC_OF(lane);
@@ -215,6 +244,7 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
         for (int ks = 0; ks < 2; ++ks)
           b_frag[n][ks] =
               lds_frag<SW>(bb, row, ks * 32 + ((lane >> 4) << 3));
+      }
       }
       } else {
         // stale single fragment pair (timing structure only)

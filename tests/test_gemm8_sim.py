@@ -189,3 +189,29 @@ def test_gemm8_index_math_reproduces_matmul(swizzle):
                             acc = mfma_16x16x32(a_frags, b_frags, acc)
                         D[fr:fr + 16, fc:fc + 16] = acc
     np.testing.assert_allclose(D, ref, rtol=1e-10, atol=1e-10)
+
+
+def test_swz1_lane_hoist_equivalence():
+    """gemm8.hip swz1_lane_low: for SW=1 the XOR key reduces to
+    (lane&7)<<4 (fragment rows differ by 16-multiples), so
+    swz(row*128+col*2) == wave_base + lane_low(ks) + quadrant*2048.
+    Exhaustive over every lane/wave/quadrant/ks the kernel uses."""
+    def lane_low(lane, ks):
+        low = ks * 64 + ((lane >> 4) << 4) + (lane & 15) * 128
+        return low ^ ((lane & 7) << 4)
+
+    for lane in range(64):
+        rc = lane & 15
+        col = lambda ks: ks * 32 + ((lane >> 4) << 3)
+        for wm in range(2):             # A side: wrow0 = wm*128
+            for qmm in range(8):
+                for ks in range(2):
+                    row = wm * 128 + qmm * 16 + rc
+                    assert swz(row * 128 + col(ks) * 2, 1) == (
+                        wm * 16384 + lane_low(lane, ks) + qmm * 2048)
+        for wn in range(4):             # B side: wcol0 = wn*64
+            for qnn in range(4):
+                for ks in range(2):
+                    row = wn * 64 + qnn * 16 + rc
+                    assert swz(row * 128 + col(ks) * 2, 1) == (
+                        wn * 8192 + lane_low(lane, ks) + qnn * 2048)
