@@ -38,6 +38,44 @@ def create_error_response(message: str, status_code: int = 400) -> JSONResponse:
     )
 
 
+async def with_cancellation(handler_coro, raw_request: Request):
+    """Race the handler against client disconnect (reference: vLLM's
+    with_cancellation decorator applied to every OpenAI route,
+    openai/config.py route registration). If the client goes away while
+    the request is still generating, the handler task is cancelled;
+    engine streams abort themselves from their finally blocks, so the
+    scheduler stops spending GPU on the dead request. Returns the
+    handler result, or a 499 response on disconnect."""
+    import asyncio
+    import contextlib
+
+    handler = asyncio.ensure_future(handler_coro)
+
+    async def watch_disconnect():
+        # the body is already consumed, so the only message left on the
+        # ASGI receive channel is http.disconnect
+        while True:
+            message = await raw_request.receive()
+            if message["type"] == "http.disconnect":
+                return
+
+    watcher = asyncio.ensure_future(watch_disconnect())
+    try:
+        done, _ = await asyncio.wait(
+            {handler, watcher}, return_when=asyncio.FIRST_COMPLETED
+        )
+        if handler in done:
+            return handler.result()
+        handler.cancel()
+        with contextlib.suppress(asyncio.CancelledError):
+            await handler
+        return Response(status_code=499)  # client closed request
+    finally:
+        watcher.cancel()
+        with contextlib.suppress(asyncio.CancelledError):
+            await watcher
+
+
 class OpenAIEndpoints:
     def __init__(self, dataplane: DataPlane, models: List[OpenAIModel]):
         self.dataplane = dataplane
@@ -64,7 +102,11 @@ class OpenAIEndpoints:
             model = self._get_model(request.model)
         except ModelNotFound as e:
             return create_error_response(str(e), 404)
-        result = await model.create_completion(request, raw_request)
+        result = await with_cancellation(
+            model.create_completion(request, raw_request), raw_request
+        )
+        if isinstance(result, Response) and result.status_code == 499:
+            return result
         if hasattr(result, "__anext__"):
             return StreamingResponse(
                 _sse(result), media_type="text/event-stream"
@@ -83,7 +125,11 @@ class OpenAIEndpoints:
             model = self._get_model(request.model)
         except ModelNotFound as e:
             return create_error_response(str(e), 404)
-        result = await model.create_chat_completion(request, raw_request)
+        result = await with_cancellation(
+            model.create_chat_completion(request, raw_request), raw_request
+        )
+        if isinstance(result, Response) and result.status_code == 499:
+            return result
         if hasattr(result, "__anext__"):
             return StreamingResponse(_sse(result), media_type="text/event-stream")
         if isinstance(result, JSONResponse):
@@ -100,7 +146,11 @@ class OpenAIEndpoints:
             model = self._get_model(request.model)
         except ModelNotFound as e:
             return create_error_response(str(e), 404)
-        result = await model.create_embedding(request, raw_request)
+        result = await with_cancellation(
+            model.create_embedding(request, raw_request), raw_request
+        )
+        if isinstance(result, Response) and result.status_code == 499:
+            return result
         if isinstance(result, JSONResponse):
             return result
         return JSONResponse(content=result.model_dump(exclude_none=True))
@@ -115,7 +165,11 @@ class OpenAIEndpoints:
             model = self._get_model(request.model)
         except ModelNotFound as e:
             return create_error_response(str(e), 404)
-        result = await model.create_rerank(request, raw_request)
+        result = await with_cancellation(
+            model.create_rerank(request, raw_request), raw_request
+        )
+        if isinstance(result, Response) and result.status_code == 499:
+            return result
         if isinstance(result, JSONResponse):
             return result
         return JSONResponse(content=result.model_dump(exclude_none=True))

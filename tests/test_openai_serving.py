@@ -442,3 +442,100 @@ def test_response_format_json_object(client):
 
     out = asyncio.new_event_loop().run_until_complete(run())
     assert JsonMachine().accepts(bytes(out.output_token_ids))
+
+
+class TestWithCancellation:
+    """Client-disconnect cancellation (reference: vLLM's with_cancellation
+    on every OpenAI route). The watcher listens on the ASGI receive
+    channel; http.disconnect cancels the in-flight handler."""
+
+    def _request(self, messages):
+        import asyncio
+
+        from fastapi import Request
+
+        q = asyncio.Queue()
+        for m in messages:
+            q.put_nowait(m)
+
+        async def receive():
+            return await q.get()
+
+        return Request(
+            {"type": "http", "method": "POST", "headers": [],
+             "path": "/", "query_string": b""},
+            receive=receive,
+        )
+
+    def test_handler_completes_normally(self):
+        import asyncio
+
+        from kserve_amd.protocol.rest.openai.endpoints import (
+            with_cancellation,
+        )
+
+        async def run():
+            req = self._request([])  # no disconnect ever arrives
+
+            async def handler():
+                await asyncio.sleep(0.01)
+                return "done"
+
+            return await with_cancellation(handler(), req)
+
+        assert asyncio.new_event_loop().run_until_complete(run()) == "done"
+
+    def test_disconnect_cancels_handler(self):
+        import asyncio
+
+        from kserve_amd.protocol.rest.openai.endpoints import (
+            with_cancellation,
+        )
+
+        cancelled = {}
+
+        async def run():
+            req = self._request([{"type": "http.disconnect"}])
+
+            async def slow_handler():
+                try:
+                    await asyncio.sleep(30)
+                except asyncio.CancelledError:
+                    cancelled["yes"] = True
+                    raise
+                return "never"
+
+            return await with_cancellation(slow_handler(), req)
+
+        resp = asyncio.new_event_loop().run_until_complete(run())
+        assert cancelled.get("yes") is True
+        assert resp.status_code == 499
+
+    def test_engine_abort_on_cancel(self, client):
+        """Cancelling the handler mid-generation aborts the engine
+        request: the scheduler drains back to empty."""
+        import asyncio
+
+        model = client.app.state.llm_model
+
+        async def run():
+            from kserve_amd.engine.sampling_params import SamplingParams
+
+            gen = model.async_engine.generate(
+                [1, 2, 3], SamplingParams(max_tokens=512, ignore_eos=True),
+                request_id="cancel-me",
+            )
+            task = asyncio.ensure_future(gen.__anext__())
+            await task  # first token arrives
+            closer = asyncio.ensure_future(gen.aclose())
+            await closer
+            # abort sentinel drains through the engine thread
+            for _ in range(100):
+                eng = model.async_engine.engine
+                if (eng.scheduler.num_waiting == 0
+                        and not eng.scheduler.running):
+                    return True
+                await asyncio.sleep(0.05)
+            return False
+
+        assert asyncio.new_event_loop().run_until_complete(run())
