@@ -178,20 +178,40 @@ def create_epp_app(picker: EndpointPicker) -> FastAPI:
 
 
 def main(argv=None):
+    """EPP process: Envoy ext-proc gRPC on --port (the :9002 contract the
+    llmisvc scheduler Deployment exposes) + the HTTP picker/health app on
+    --http-port (:9003)."""
     import argparse
+    import asyncio
 
     import uvicorn
 
     ap = argparse.ArgumentParser(description="kserve-amd endpoint picker")
     ap.add_argument("--endpoints", required=True,
                     help="comma-separated pool member base URLs")
-    ap.add_argument("--port", type=int, default=9002)
+    ap.add_argument("--port", type=int, default=9002,
+                    help="ext-proc gRPC port")
+    ap.add_argument("--http-port", type=int, default=9003,
+                    help="HTTP /pick /endpoints /healthz port")
     ap.add_argument("--scrape-interval", type=float, default=1.0)
     args = ap.parse_args(argv)
     picker = EndpointPicker(
         args.endpoints.split(","), scrape_interval_s=args.scrape_interval
     )
-    uvicorn.run(create_epp_app(picker), host="0.0.0.0", port=args.port)
+
+    async def serve():
+        from kserve_amd.agent.ext_proc import create_ext_proc_server
+
+        grpc_server = create_ext_proc_server(picker, args.port)
+        await grpc_server.start()
+        config = uvicorn.Config(
+            create_epp_app(picker), host="0.0.0.0", port=args.http_port,
+            log_level="warning",
+        )
+        await uvicorn.Server(config).serve()
+        await grpc_server.stop(grace=2.0)
+
+    asyncio.run(serve())
 
 
 if __name__ == "__main__":

@@ -260,6 +260,7 @@ def render_scheduler(llm: LLMInferenceService, image="kserve-amd/endpoint-picker
                                 "--endpoints",
                                 f"http://{llm.name}-decode.{llm.namespace}:80",
                                 "--port", str(s.grpc_port),
+                                "--http-port", str(s.health_port),
                             ],
                             "ports": [
                                 {"containerPort": s.grpc_port, "name": "grpc"},

@@ -165,3 +165,102 @@ class TestHTTPSurface:
         app = create_epp_app(picker)
         with TestClient(app) as client:
             assert client.post("/pick", json={}).status_code == 503
+
+
+class TestExtProc:
+    """Envoy ext-proc v3 face of the EPP (the GIE contract the reference
+    gateway speaks to llm-d's picker): request_headers in ->
+    header-mutation response naming the pool member."""
+
+    def _picker(self, states, down=()):
+        p = EndpointPicker(
+            [f"http://{h}" for h in states],
+            transport=member_transport(states, down),
+        )
+        asyncio.run(p.scrape_once())
+        return p
+
+    def test_pick_response_sets_destination_header(self):
+        from kserve_amd.agent.ext_proc import (
+            DESTINATION_HEADER,
+            pick_response,
+        )
+
+        p = self._picker({
+            "a": {"waiting": 5, "running": 0, "kv": 0.1},
+            "b": {"waiting": 0, "running": 0, "kv": 0.1},
+        })
+        resp = pick_response(p, {})
+        assert resp.WhichOneof("response") == "request_headers"
+        common = resp.request_headers.response
+        assert common.status == 0  # CONTINUE
+        muts = {o.header.key: o.header.raw_value
+                for o in common.header_mutation.set_headers}
+        assert muts[DESTINATION_HEADER] == b"http://b"
+
+    def test_pick_response_empty_pool_continues_unmutated(self):
+        from kserve_amd.agent.ext_proc import pick_response
+
+        p = EndpointPicker([], transport=httpx.MockTransport(
+            lambda r: httpx.Response(500)))
+        resp = pick_response(p, {})
+        assert resp.request_headers.response.status == 0
+        assert not resp.request_headers.response.header_mutation.set_headers
+
+    def test_grpc_stream_roundtrip(self):
+        """Full grpc.aio bidi stream: headers (with session id) then a
+        body chunk; both phases answered, destination sticky."""
+        import grpc
+
+        from kserve_amd.agent.ext_proc import (
+            DESTINATION_HEADER,
+            SERVICE_NAME,
+            HeaderValue,
+            ProcessingRequest,
+            ProcessingResponse,
+            create_ext_proc_server,
+        )
+
+        states = {
+            "a": {"waiting": 0, "running": 0, "kv": 0.0},
+            "b": {"waiting": 0, "running": 0, "kv": 0.0},
+        }
+        picker = self._picker(states)
+
+        async def run():
+            server = create_ext_proc_server(picker, 0)
+            port = server.add_insecure_port("127.0.0.1:0")
+            await server.start()
+            try:
+                async with grpc.aio.insecure_channel(
+                    f"127.0.0.1:{port}"
+                ) as ch:
+                    stream = ch.stream_stream(
+                        f"/{SERVICE_NAME}/Process",
+                        request_serializer=lambda m: m.SerializeToString(),
+                        response_deserializer=ProcessingResponse.FromString,
+                    )
+                    call = stream()
+                    req = ProcessingRequest()
+                    req.request_headers.headers.headers.append(
+                        HeaderValue(key="x-session-id", value="u1"))
+                    await call.write(req)
+                    first = await call.read()
+                    body = ProcessingRequest()
+                    body.request_body.body = b"{}"
+                    body.request_body.end_of_stream = True
+                    await call.write(body)
+                    second = await call.read()
+                    await call.done_writing()
+                    return first, second
+            finally:
+                await server.stop(grace=None)
+
+        first, second = asyncio.run(run())
+        muts = {o.header.key: o.header.raw_value for o in
+                first.request_headers.response.header_mutation.set_headers}
+        dest = muts[DESTINATION_HEADER]
+        assert dest in (b"http://a", b"http://b")
+        assert second.WhichOneof("response") == "request_body"
+        # sticky: same session hashes to the same member
+        assert picker.pick(session_id="u1") == dest.decode()
