@@ -63,16 +63,28 @@ __device__ __forceinline__ void stage_half(
     int row0,                       // first row of the half-tile (0 or 128)
     short* lds_base,                // LDS base of the FULL 256x64 tile
     int wave, int lane) {
+  // SW==1 source-address hoist: every chunk base is an 8-row multiple,
+  // so the XOR key is ((lane>>3)&7)<<4 and the per-lane source offset
+  // (row delta * ld + column) is loop- and chunk-invariant.
+  long lane_src = 0;
+  if constexpr (SW == 1) {
+    const int lo = (lane * 16) ^ (((lane >> 3) & 7) << 4);
+    lane_src = (long)(lo >> 7) * ld + ((lo & 127) >> 1);
+  }
   // 16 KiB = 16 chunks of 1 KiB; 8 waves stage 2 chunks each
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
     const int chunk = wave + i * 8;
     const int lin = row0 * 128 + chunk * 1024;        // wave-uniform
-    const int sb = swz<SW>(lin + lane * 16);          // per-lane source
-    const int srow = sb >> 7;                         // /128 bytes per row
-    const int scol = (sb & 127) >> 1;                 // byte -> element
+    const short* sp;
+    if constexpr (SW == 1) {
+      sp = src + (long)(lin >> 7) * ld + lane_src;
+    } else {
+      const int sb = swz<SW>(lin + lane * 16);        // per-lane source
+      sp = src + (long)(sb >> 7) * ld + ((sb & 127) >> 1);
+    }
     __builtin_amdgcn_global_load_lds(
-        reinterpret_cast<const unsigned int*>(src + (long)srow * ld + scol),
+        reinterpret_cast<const unsigned int*>(sp),
         reinterpret_cast<unsigned int*>(
             reinterpret_cast<char*>(lds_base) + lin),
         16, 0, 0);
@@ -110,11 +122,17 @@ __device__ __forceinline__ void stage_unit(
       row0 = ((chunk >> 2) << 6) + ((chunk & 3) << 3);
     row0 += row_add;
     const int lin = row0 * 128;                        // wave-uniform
-    const int sb = swz<SW>(lin + lane * 16);           // per-lane source
-    const int srow = sb >> 7;
-    const int scol = (sb & 127) >> 1;
+    const short* sp;
+    if constexpr (SW == 1) {
+      // same hoist as stage_half: chunk bases are 8-row multiples
+      const int lo = (lane * 16) ^ (((lane >> 3) & 7) << 4);
+      sp = src + (long)(row0 + (lo >> 7)) * ld + ((lo & 127) >> 1);
+    } else {
+      const int sb = swz<SW>(lin + lane * 16);         // per-lane source
+      sp = src + (long)(sb >> 7) * ld + ((sb & 127) >> 1);
+    }
     __builtin_amdgcn_global_load_lds(
-        reinterpret_cast<const unsigned int*>(src + (long)srow * ld + scol),
+        reinterpret_cast<const unsigned int*>(sp),
         reinterpret_cast<unsigned int*>(
             reinterpret_cast<char*>(lds_base) + lin),
         16, 0, 0);

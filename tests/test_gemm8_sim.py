@@ -215,3 +215,14 @@ def test_swz1_lane_hoist_equivalence():
                     row = wn * 64 + qnn * 16 + rc
                     assert swz(row * 128 + col(ks) * 2, 1) == (
                         wn * 8192 + lane_low(lane, ks) + qnn * 2048)
+
+
+def test_swz1_stage_hoist_equivalence():
+    """gemm8.hip stage_half/stage_unit SW=1 source hoist: chunk bases
+    are 8-row multiples, so swz(lin + lane*16) decomposes into
+    lin + ((lane*16) ^ (((lane>>3)&7)<<4)) for every chunk base."""
+    for lane in range(64):
+        lo = (lane * 16) ^ (((lane >> 3) & 7) << 4)
+        for row0 in range(0, 256, 8):
+            lin = row0 * 128
+            assert swz(lin + lane * 16, 1) == lin + lo
