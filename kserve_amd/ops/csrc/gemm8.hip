@@ -628,8 +628,8 @@ __global__ __launch_bounds__(512, 1) void gemm8_ra_kernel(
         for (int n = 0; n < QN; ++n)
 #pragma unroll
           for (int ks = 0; ks < 2; ++ks)
-            acc[qm + m][qn + n] = mfma16x16x32(
-                a_frag[cur][m][ks], b_frag[cur][n][ks], acc[qm + m][qn + n]);
+            mfma16x16x32_agpr(a_frag[cur][m][ks], b_frag[cur][n][ks],
+                              acc[qm + m][qn + n]);
       if (ph == 3) {
         if (have_next) asm volatile("s_waitcnt vmcnt(0)");
         __builtin_amdgcn_s_barrier();

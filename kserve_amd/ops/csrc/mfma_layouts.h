@@ -26,3 +26,15 @@ __device__ __forceinline__ f32x4_t mfma16x16x32(bf16x8_t a, bf16x8_t b,
                                                 f32x4_t c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
+
+// AGPR-accumulator variant: the "a" constraint pins the accumulator to
+// the accumulation register file, freeing ~4 VGPRs per fragment for
+// deep-pipelined kernels the allocator otherwise cannot fit (the
+// builtin's heuristic keeps accumulators in VGPRs even at full
+// pressure). acc is read back to VGPRs only at the epilogue.
+__device__ __forceinline__ void mfma16x16x32_agpr(bf16x8_t a, bf16x8_t b,
+                                                  f32x4_t& acc) {
+  __asm__ volatile("v_mfma_f32_16x16x32_bf16 %0, %1, %2, %0"
+                   : "+a"(acc)
+                   : "v"(a), "v"(b));
+}
