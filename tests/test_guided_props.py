@@ -96,3 +96,60 @@ def test_schema_machine_walks_conform(schema_idx, rng_bytes):
                 assert isinstance(doc[key], list)
             if "enum" in spec:
                 assert doc[key] in spec["enum"]
+
+
+# -- router condition fuzz (same property-testing batch) ------------------
+
+json_scalars = st.one_of(
+    st.none(), st.booleans(),
+    st.integers(min_value=-1000, max_value=1000),
+    st.floats(-1e3, 1e3, allow_nan=False),
+    st.text(st.characters(blacklist_categories=("Cs",),
+                          blacklist_characters=".#"), max_size=6),
+)
+json_docs = st.recursive(
+    json_scalars,
+    lambda inner: st.one_of(
+        st.lists(inner, max_size=4),
+        st.dictionaries(
+            st.text(st.characters(whitelist_categories=("Ll",)),
+                    min_size=1, max_size=5),
+            inner, max_size=4),
+    ),
+    max_leaves=12,
+)
+
+
+@settings(max_examples=150, deadline=None)
+@given(json_docs, st.text(max_size=20))
+def test_router_condition_never_crashes(doc, condition):
+    """InferenceGraph Switch/step conditions come from user manifests:
+    evaluation over ARBITRARY bodies and condition strings must return a
+    bool, never raise (the reference treats invalid conditions as
+    non-matching)."""
+    from kserve_amd.graph.router import condition_matches, gjson_get
+
+    assert condition_matches(doc, condition) in (True, False)
+    ok, _ = gjson_get(doc, condition)
+    assert ok in (True, False)
+
+
+@settings(max_examples=100, deadline=None)
+@given(json_docs)
+def test_gjson_roundtrip_paths(doc):
+    """Every reachable leaf of a document is retrievable by its own
+    dotted path with exists=True."""
+    from kserve_amd.graph.router import gjson_get
+
+    def walk(node, path):
+        ok, val = gjson_get(doc, path)
+        assert ok, path
+        assert val == node or (val != val and node != node)
+        if isinstance(node, dict):
+            for k, v in node.items():
+                walk(v, f"{path}.{k}" if path else k)
+        elif isinstance(node, list):
+            for i, v in enumerate(node):
+                walk(v, f"{path}.{i}" if path else str(i))
+
+    walk(doc, "")
