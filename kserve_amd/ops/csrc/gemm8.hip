@@ -212,6 +212,8 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
   __builtin_amdgcn_s_barrier();
 
   const int ktiles = K / BK;
+  // (2-K-tile unroll measured neutral-to-slightly-negative — ladder
+  // addendum 4; the simple loop stays.)
   for (int kt = 0; kt < ktiles; ++kt) {
     const int buf = kt & 1;
     const short* ab = A_BUF(buf);
