@@ -48,6 +48,8 @@ def main():
     parser.add_argument("--eager", action="store_true",
                         help="disable hipGraph capture")
     parser.add_argument("--temperature", type=float, default=0.0)
+    parser.add_argument("--multi-step", dest="multi_step", type=int, default=8,
+                        help="decode iterations per hipGraph window")
     parser.add_argument("--kv-cache-dtype", type=str, default="auto",
                         choices=["auto", "fp8"],
                         help="opt-in fp8 E4M3 KV cache (NOT the headline "
@@ -88,7 +90,7 @@ def main():
 
     # each timed "step" is one scheduling iteration = up to `multi_step`
     # decode iterations (hipGraph window); budget the KV/model length for it
-    window = 8
+    window = args.multi_step
     max_model_len = args.prompt_len + (args.steps + args.warmup) * window + 64
     cfg = EngineConfig(
         model=mcfg,
@@ -102,6 +104,7 @@ def main():
             max_num_seqs=args.concurrency,
             max_num_batched_tokens=16384,
             max_model_len=max_model_len,
+            multi_step=args.multi_step,
         ),
         device=device,
         seed=1234 + state.dp_rank,
