@@ -18,6 +18,20 @@ import httpx
 from kserve_amd import constants
 from kserve_amd.errors import InvalidInput
 from kserve_amd.logging import logger, trace_logger
+
+# per-request latency trace lines (reference ModelServer
+# --enable_latency_logging); module-level so ModelServer's flag reaches
+# every Model instance
+_LATENCY_LOGGING = True
+
+
+def set_latency_logging(enabled: bool) -> None:
+    global _LATENCY_LOGGING
+    _LATENCY_LOGGING = enabled
+
+
+def enable_latency_logging() -> bool:
+    return _LATENCY_LOGGING
 from kserve_amd.metrics import get_labeled_histograms
 from kserve_amd.protocol.infer_type import InferRequest, InferResponse
 
@@ -128,15 +142,16 @@ class Model(InferenceModel):
         response = await self._maybe_await(self.postprocess(result, headers))
         t3 = time.perf_counter()
         self._histograms["postprocess"].observe(t3 - t2)
-        trace_logger.info(
-            "requestId: %s, preprocess_ms: %.3f, explain_ms: %.3f, "
-            "predict_ms: %.3f, postprocess_ms: %.3f",
-            request_id,
-            (t1 - t0) * 1000,
-            (t2 - t1) * 1000 if verb == "explain" else 0,
-            (t2 - t1) * 1000 if verb != "explain" else 0,
-            (t3 - t2) * 1000,
-        )
+        if enable_latency_logging():
+            trace_logger.info(
+                "requestId: %s, preprocess_ms: %.3f, explain_ms: %.3f, "
+                "predict_ms: %.3f, postprocess_ms: %.3f",
+                request_id,
+                (t1 - t0) * 1000,
+                (t2 - t1) * 1000 if verb == "explain" else 0,
+                (t2 - t1) * 1000 if verb != "explain" else 0,
+                (t3 - t2) * 1000,
+            )
         return response
 
     @staticmethod

@@ -33,6 +33,8 @@ def build_arg_parser() -> argparse.ArgumentParser:
     parser.add_argument("--enable_grpc", default=True, type=lambda x: str(x).lower() == "true")
     parser.add_argument("--enable_docs_url", default=False, type=lambda x: str(x).lower() == "true")
     parser.add_argument("--access_log", default=False, type=lambda x: str(x).lower() == "true")
+    parser.add_argument("--enable_latency_logging", default=True,
+                        type=lambda x: str(x).lower() == "true")
     # transformer mode
     parser.add_argument("--predictor_host", default=None, type=str)
     parser.add_argument("--predictor_protocol", default="v1", type=str)
@@ -53,8 +55,12 @@ class ModelServer:
         workers: int = 1,
         max_threads: int = 4,
         access_log: bool = False,
+        enable_latency_logging: bool = True,
     ):
         configure_logging()
+        from kserve_amd.model import set_latency_logging
+
+        set_latency_logging(enable_latency_logging)
         self.http_port = http_port
         self.grpc_port = grpc_port
         self.workers = workers

@@ -545,3 +545,36 @@ class TestQpextMerging:
         assert r.status_code == 200
         assert "ok 1" in r.text
         assert client.get("/healthz").json()["status"] == "ok"
+
+
+class TestLatencyLogging:
+    def test_flag_gates_trace_lines(self, caplog):
+        """--enable_latency_logging parity (reference ModelServer): the
+        per-request latency trace line is emitted by default and
+        suppressed when disabled."""
+        import logging
+
+        from kserve_amd import model as model_mod
+        from kserve_amd.logging import TRACE_LOGGER_NAME
+
+        class Echo(Model):
+            def predict(self, payload, headers=None):
+                return {"predictions": payload["instances"]}
+
+        m = Echo("echo")
+        m.ready = True
+        loop = asyncio.new_event_loop()
+        try:
+            model_mod.set_latency_logging(True)
+            with caplog.at_level(logging.INFO, logger=TRACE_LOGGER_NAME):
+                loop.run_until_complete(m({"instances": [1]}))
+            assert any("preprocess_ms" in r.message for r in caplog.records)
+            caplog.clear()
+            model_mod.set_latency_logging(False)
+            with caplog.at_level(logging.INFO, logger=TRACE_LOGGER_NAME):
+                loop.run_until_complete(m({"instances": [1]}))
+            assert not any("preprocess_ms" in r.message
+                           for r in caplog.records)
+        finally:
+            model_mod.set_latency_logging(True)
+            loop.close()
