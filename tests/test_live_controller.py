@@ -216,6 +216,27 @@ class TestLiveISVCController:
         isvc = server.get(ISVC_GVK, "default", "iris")
         assert get_condition(isvc["status"], "Ready")["status"] == "True"
 
+    def test_canary_rollout_and_promotion(self, env):
+        """Reference canary e2e (test_canary*.py): setting
+        canaryTrafficPercent creates the -canary Deployment; promotion
+        (removing the percent) prunes it again."""
+        server, isvc_ctrl, dep_ctrl = env
+        obj = make_isvc(name="cnr")
+        obj["spec"]["predictor"]["canaryTrafficPercent"] = 10
+        server.create(obj)
+        converge(isvc_ctrl, dep_ctrl)
+        assert server.try_get(DEP, "default", "cnr-predictor")
+        assert server.try_get(DEP, "default", "cnr-predictor-canary")
+        # promotion: drop the canary percent
+        cur = server.get(ISVC_GVK, "default", "cnr")
+        del cur["spec"]["predictor"]["canaryTrafficPercent"]
+        server.update(cur)
+        converge(isvc_ctrl, dep_ctrl)
+        assert server.try_get(DEP, "default", "cnr-predictor")
+        assert server.try_get(DEP, "default", "cnr-predictor-canary") is None
+        isvc = server.get(ISVC_GVK, "default", "cnr")
+        assert get_condition(isvc["status"], "Ready")["status"] == "True"
+
     def test_drift_is_repaired(self, env):
         server, isvc_ctrl, dep_ctrl = env
         server.create(make_isvc())
