@@ -1,0 +1,174 @@
+#!/usr/bin/env python3
+"""BASELINE config 2 at the SERVING level: BERT-base fill-mask through
+the full REST stack — V2 `infer` requests (input_ids INT64 [B, S]) over
+real localhost HTTP into the native varlen encoder on cuda:0.
+
+The server runs in its own process (model + kernels on the GPU); an
+aiohttp open-loop client drives concurrent V2 requests. Reports
+sequences/s and request latency percentiles. Synthetic token batches,
+random-init weights (BASELINE "synthetic data / random-init")."""
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+VOCAB = 30522
+
+
+def start_server(port: int, seq_len: int):
+    import numpy as np
+    import torch
+    import uvicorn
+
+    from kserve_amd.model import Model
+    from kserve_amd.model_repository import ModelRepository
+    from kserve_amd.protocol.dataplane import DataPlane
+    from kserve_amd.protocol.infer_type import (
+        InferOutput,
+        InferRequest,
+        InferResponse,
+    )
+    from kserve_amd.protocol.rest.server import create_app
+    from kserve_amd.models.bert import BertConfig, BertForMaskedLM
+
+    use_gpu = torch.cuda.is_available()
+    dev = "cuda:0" if use_gpu else "cpu"
+    cfg = BertConfig()  # bert-base
+    torch.manual_seed(0)
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+    model = BertForMaskedLM(cfg, dtype=dtype, device=dev)
+    with torch.no_grad():
+        for p in model.parameters():
+            if p.dim() >= 2:
+                p.normal_(0.0, 0.02)
+            else:
+                p.fill_(0.01)
+
+    class BertV2(Model):
+        def __init__(self):
+            super().__init__("bert")
+            self.ready = True
+
+        def predict(self, payload, headers=None):
+            assert isinstance(payload, InferRequest)
+            ids = payload.inputs[0].as_numpy().astype(np.int64)
+            b, s = ids.shape
+            t = torch.from_numpy(ids).to(dev)
+            flat = t.reshape(-1)
+            cu = torch.arange(0, (b + 1) * s, s, dtype=torch.int32,
+                              device=dev)
+            with torch.no_grad():
+                logits = model(flat, cu)
+            pred = logits.argmax(-1).to(torch.int64)
+            if use_gpu:
+                torch.cuda.synchronize()
+            out_np = pred.reshape(b, s).cpu().numpy()
+            out = InferOutput("predictions", [b, s], "INT64")
+            out.set_data_from_numpy(out_np, binary_data=False)
+            return InferResponse(payload.id, self.name, [out])
+
+    repo = ModelRepository()
+    repo.update(BertV2())
+    app = create_app(DataPlane(repo))
+    uvicorn.run(app, host="127.0.0.1", port=port, log_level="error")
+
+
+async def drive(port, batch, seq_len, requests, concurrency):
+    import random
+
+    import aiohttp
+
+    rng = random.Random(0)
+    body = {
+        "inputs": [{
+            "name": "input_ids", "shape": [batch, seq_len],
+            "datatype": "INT64",
+            "data": [[rng.randrange(VOCAB) for _ in range(seq_len)]
+                     for _ in range(batch)],
+        }]
+    }
+    url = f"http://127.0.0.1:{port}/v2/models/bert/infer"
+    lat = []
+    conn = aiohttp.TCPConnector(limit=concurrency)
+    timeout = aiohttp.ClientTimeout(total=120)
+    async with aiohttp.ClientSession(connector=conn,
+                                     timeout=timeout) as client:
+        # warmup
+        for _ in range(4):
+            async with client.post(url, json=body) as r:
+                assert r.status == 200, await r.text()
+
+        sem = asyncio.Semaphore(concurrency)
+
+        async def one():
+            async with sem:
+                t0 = time.perf_counter()
+                async with client.post(url, json=body) as r:
+                    await r.read()
+                    assert r.status == 200
+                lat.append(time.perf_counter() - t0)
+
+        t0 = time.perf_counter()
+        await asyncio.gather(*[one() for _ in range(requests)])
+        elapsed = time.perf_counter() - t0
+    lat.sort()
+    return {
+        "metric": "bert-base fill-mask seq/s through V2 HTTP",
+        "value": round(requests * batch / elapsed, 1),
+        "batch": batch, "seq_len": seq_len, "requests": requests,
+        "concurrency": concurrency,
+        "latency_p50_ms": round(lat[len(lat) // 2] * 1000, 1),
+        "latency_p99_ms": round(lat[int(0.99 * len(lat)) - 1] * 1000, 1),
+        "elapsed_s": round(elapsed, 2),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=18150)
+    ap.add_argument("--batch", type=int, default=32)
+    ap.add_argument("--seq-len", type=int, default=128)
+    ap.add_argument("--requests", type=int, default=200)
+    ap.add_argument("--concurrency", type=int, default=8)
+    ap.add_argument("--serve", action="store_true")
+    args = ap.parse_args()
+    if args.serve:
+        start_server(args.port, args.seq_len)
+        return
+
+    import subprocess
+
+    import requests as rq
+
+    proc = subprocess.Popen(
+        [sys.executable, os.path.abspath(__file__), "--serve",
+         "--port", str(args.port), "--seq-len", str(args.seq_len)],
+    )
+    try:
+        for _ in range(600):  # model init + first import can take a while
+            try:
+                if rq.get(f"http://127.0.0.1:{args.port}/",
+                          timeout=1).status_code == 200:
+                    break
+            except Exception:
+                time.sleep(0.5)
+        else:
+            raise RuntimeError("bert server did not come up")
+        res = asyncio.new_event_loop().run_until_complete(
+            drive(args.port, args.batch, args.seq_len, args.requests,
+                  args.concurrency)
+        )
+        print(json.dumps(res), flush=True)
+    finally:
+        proc.terminate()
+        proc.wait(timeout=15)
+
+
+if __name__ == "__main__":
+    main()
