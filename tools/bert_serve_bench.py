@@ -79,21 +79,42 @@ def start_server(port: int, seq_len: int):
     uvicorn.run(app, host="127.0.0.1", port=port, log_level="error")
 
 
-async def drive(port, batch, seq_len, requests, concurrency):
+async def drive(port, batch, seq_len, requests, concurrency,
+                binary=False):
     import random
 
     import aiohttp
+    import numpy as np
 
     rng = random.Random(0)
-    body = {
-        "inputs": [{
-            "name": "input_ids", "shape": [batch, seq_len],
-            "datatype": "INT64",
-            "data": [[rng.randrange(VOCAB) for _ in range(seq_len)]
-                     for _ in range(batch)],
-        }]
-    }
     url = f"http://127.0.0.1:{port}/v2/models/bert/infer"
+    headers = {}
+    if binary:
+        # V2 binary tensor extension: JSON prefix + raw INT64 bytes,
+        # prefix length in the inference-content-length header
+        from kserve_amd.protocol.infer_type import InferInput, InferRequest
+
+        arr = np.array(
+            [[rng.randrange(VOCAB) for _ in range(seq_len)]
+             for _ in range(batch)], dtype=np.int64)
+        inp = InferInput("input_ids", [batch, seq_len], "INT64")
+        inp.set_data_from_numpy(arr, binary_data=True)
+        body_bytes, json_len = InferRequest("bert", [inp]).to_rest()
+        headers = {
+            "inference-content-length": str(json_len),
+            "content-type": "application/octet-stream",
+        }
+        post_kwargs = {"data": body_bytes, "headers": headers}
+    else:
+        body = {
+            "inputs": [{
+                "name": "input_ids", "shape": [batch, seq_len],
+                "datatype": "INT64",
+                "data": [[rng.randrange(VOCAB) for _ in range(seq_len)]
+                         for _ in range(batch)],
+            }]
+        }
+        post_kwargs = {"json": body}
     lat = []
     conn = aiohttp.TCPConnector(limit=concurrency)
     timeout = aiohttp.ClientTimeout(total=120)
@@ -101,7 +122,7 @@ async def drive(port, batch, seq_len, requests, concurrency):
                                      timeout=timeout) as client:
         # warmup
         for _ in range(4):
-            async with client.post(url, json=body) as r:
+            async with client.post(url, **post_kwargs) as r:
                 assert r.status == 200, await r.text()
 
         sem = asyncio.Semaphore(concurrency)
@@ -109,7 +130,7 @@ async def drive(port, batch, seq_len, requests, concurrency):
         async def one():
             async with sem:
                 t0 = time.perf_counter()
-                async with client.post(url, json=body) as r:
+                async with client.post(url, **post_kwargs) as r:
                     await r.read()
                     assert r.status == 200
                 lat.append(time.perf_counter() - t0)
@@ -119,7 +140,8 @@ async def drive(port, batch, seq_len, requests, concurrency):
         elapsed = time.perf_counter() - t0
     lat.sort()
     return {
-        "metric": "bert-base fill-mask seq/s through V2 HTTP",
+        "metric": "bert-base fill-mask seq/s through V2 HTTP"
+                  + (" (binary tensors)" if binary else " (JSON)"),
         "value": round(requests * batch / elapsed, 1),
         "batch": batch, "seq_len": seq_len, "requests": requests,
         "concurrency": concurrency,
@@ -137,6 +159,8 @@ def main():
     ap.add_argument("--requests", type=int, default=200)
     ap.add_argument("--concurrency", type=int, default=8)
     ap.add_argument("--serve", action="store_true")
+    ap.add_argument("--binary", action="store_true",
+                    help="V2 binary tensor extension request bodies")
     args = ap.parse_args()
     if args.serve:
         start_server(args.port, args.seq_len)
@@ -162,7 +186,7 @@ def main():
             raise RuntimeError("bert server did not come up")
         res = asyncio.new_event_loop().run_until_complete(
             drive(args.port, args.batch, args.seq_len, args.requests,
-                  args.concurrency)
+                  args.concurrency, binary=args.binary)
         )
         print(json.dumps(res), flush=True)
     finally:
