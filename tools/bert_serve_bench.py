@@ -56,6 +56,19 @@ def start_server(port: int, seq_len: int):
             self.ready = True
 
         def predict(self, payload, headers=None):
+            if isinstance(payload, dict):  # V1 :predict (batcher path)
+                arr = np.asarray(payload["instances"], dtype=np.int64)
+                b, s = arr.shape
+                t = torch.from_numpy(arr).to(dev)
+                cu = torch.arange(0, (b + 1) * s, s, dtype=torch.int32,
+                                  device=dev)
+                with torch.no_grad():
+                    logits = model(t.reshape(-1), cu)
+                pred = logits.argmax(-1).to(torch.int64)
+                if use_gpu:
+                    torch.cuda.synchronize()
+                return {"predictions":
+                        pred.reshape(b, s).cpu().tolist()}
             assert isinstance(payload, InferRequest)
             ids = payload.inputs[0].as_numpy().astype(np.int64)
             b, s = ids.shape
@@ -76,6 +89,20 @@ def start_server(port: int, seq_len: int):
     repo = ModelRepository()
     repo.update(BertV2())
     app = create_app(DataPlane(repo))
+    if os.environ.get("KS_BERT_BATCHER") == "1":
+        import threading
+
+        from kserve_amd.agent.batcher import create_batcher_proxy_app
+
+        proxy = create_batcher_proxy_app(
+            f"http://127.0.0.1:{port}", "bert",
+            max_batch_size=int(os.environ.get("KS_BATCH_MAX", "64")),
+            max_latency_ms=int(os.environ.get("KS_BATCH_LAT_MS", "20")),
+        )
+        pcfg = uvicorn.Config(proxy, host="127.0.0.1", port=port + 1,
+                              log_level="error")
+        threading.Thread(target=uvicorn.Server(pcfg).run,
+                         daemon=True).start()
     uvicorn.run(app, host="127.0.0.1", port=port, log_level="error")
 
 
@@ -151,6 +178,54 @@ async def drive(port, batch, seq_len, requests, concurrency,
     }
 
 
+async def drive_v1(port, seq_len, requests, concurrency, batcher):
+    """1-sequence V1 :predict requests — the shape the in-pod batcher
+    sidecar exists to coalesce (reference pkg/batcher)."""
+    import random
+
+    import aiohttp
+
+    rng = random.Random(0)
+    url = f"http://127.0.0.1:{port}/v1/models/bert:predict"
+    lat = []
+    conn = aiohttp.TCPConnector(limit=concurrency)
+    timeout = aiohttp.ClientTimeout(total=120)
+    async with aiohttp.ClientSession(connector=conn,
+                                     timeout=timeout) as client:
+        def body():
+            return {"instances":
+                    [[rng.randrange(VOCAB) for _ in range(seq_len)]]}
+
+        for _ in range(4):
+            async with client.post(url, json=body()) as r:
+                assert r.status == 200, await r.text()
+
+        sem = asyncio.Semaphore(concurrency)
+
+        async def one():
+            async with sem:
+                t0 = time.perf_counter()
+                async with client.post(url, json=body()) as r:
+                    await r.read()
+                    assert r.status == 200
+                lat.append(time.perf_counter() - t0)
+
+        t0 = time.perf_counter()
+        await asyncio.gather(*[one() for _ in range(requests)])
+        elapsed = time.perf_counter() - t0
+    lat.sort()
+    return {
+        "metric": "bert 1-seq V1 requests/s "
+                  + ("via batcher sidecar" if batcher else "direct"),
+        "value": round(requests / elapsed, 1),
+        "seq_len": seq_len, "requests": requests,
+        "concurrency": concurrency,
+        "latency_p50_ms": round(lat[len(lat) // 2] * 1000, 1),
+        "latency_p99_ms": round(lat[int(0.99 * len(lat)) - 1] * 1000, 1),
+        "elapsed_s": round(elapsed, 2),
+    }
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--port", type=int, default=18150)
@@ -161,6 +236,11 @@ def main():
     ap.add_argument("--serve", action="store_true")
     ap.add_argument("--binary", action="store_true",
                     help="V2 binary tensor extension request bodies")
+    ap.add_argument("--v1-singles", action="store_true",
+                    help="drive 1-sequence V1 :predict requests")
+    ap.add_argument("--batcher", action="store_true",
+                    help="with --v1-singles: route through the batcher "
+                         "proxy sidecar (port+1)")
     args = ap.parse_args()
     if args.serve:
         start_server(args.port, args.seq_len)
@@ -170,9 +250,13 @@ def main():
 
     import requests as rq
 
+    env = dict(os.environ)
+    if args.batcher:
+        env["KS_BERT_BATCHER"] = "1"
     proc = subprocess.Popen(
         [sys.executable, os.path.abspath(__file__), "--serve",
          "--port", str(args.port), "--seq-len", str(args.seq_len)],
+        env=env,
     )
     try:
         for _ in range(600):  # model init + first import can take a while
@@ -184,10 +268,17 @@ def main():
                 time.sleep(0.5)
         else:
             raise RuntimeError("bert server did not come up")
-        res = asyncio.new_event_loop().run_until_complete(
-            drive(args.port, args.batch, args.seq_len, args.requests,
-                  args.concurrency, binary=args.binary)
-        )
+        if args.v1_singles:
+            res = asyncio.new_event_loop().run_until_complete(
+                drive_v1(args.port + (1 if args.batcher else 0),
+                         args.seq_len, args.requests, args.concurrency,
+                         batcher=args.batcher)
+            )
+        else:
+            res = asyncio.new_event_loop().run_until_complete(
+                drive(args.port, args.batch, args.seq_len, args.requests,
+                      args.concurrency, binary=args.binary)
+            )
         print(json.dumps(res), flush=True)
     finally:
         proc.terminate()
