@@ -12,12 +12,14 @@ from __future__ import annotations
 import asyncio
 import json
 import time
+import uuid as _uuid
 from dataclasses import dataclass, field
 from typing import Any, Awaitable, Callable, Dict, List, Optional
 
 import httpx
 from fastapi import FastAPI, Request, Response
 
+from kserve_amd.agent.payload_logger import LogEntry as _LogEntry
 from kserve_amd.constants import DEFAULT_MAX_BATCH_SIZE, DEFAULT_MAX_LATENCY_MS
 from kserve_amd.logging import logger
 
@@ -118,10 +120,18 @@ def create_batcher_proxy_app(
     max_batch_size: int = DEFAULT_MAX_BATCH_SIZE,
     max_latency_ms: int = DEFAULT_MAX_LATENCY_MS,
     transport=None,
+    payload_logger=None,
 ):
     """Agent-style reverse proxy: batches ``POST /v1/models/{m}:predict``,
-    passes everything else through (reference agent chain, main.go:429-449)."""
+    passes everything else through, and (when a PayloadLogger is given)
+    captures request/response payloads as CloudEvents exactly like the
+    reference agent chain (cmd/agent/main.go:429-449: Drainer -> logger
+    -> batcher -> reverse proxy)."""
     app = FastAPI()
+    if payload_logger is not None:
+        @app.on_event("startup")
+        async def _start_logger():
+            await payload_logger.start()
     client = httpx.AsyncClient(base_url=backend_url, transport=transport, timeout=60)
 
     async def call_backend(instances: List[Any]) -> Dict:
@@ -164,18 +174,40 @@ def create_batcher_proxy_app(
 
     @app.post("/v1/models/{name}:predict")
     async def predict(name: str, request: Request):
-        body = await request.json()
+        raw = await request.body()
+        try:
+            body = json.loads(raw)
+        except json.JSONDecodeError:
+            return Response(
+                content=json.dumps({"error": "invalid JSON"}),
+                status_code=400,
+            )
         instances = body.get("instances")
         if not isinstance(instances, list):
             return Response(
                 content=json.dumps({"error": "instances must be a list"}),
                 status_code=400,
             )
+        rid = request.headers.get("x-request-id") or str(_uuid.uuid4())
+        if payload_logger is not None:
+            await payload_logger.log(_LogEntry(
+                request_id=rid,
+                event_type="org.kubeflow.serving.inference.request",
+                model_name=name,
+                payload=raw,
+            ))
         app.state.inflight += 1
         try:
             result = await batcher.predict(instances)
         finally:
             app.state.inflight -= 1
+        if payload_logger is not None:
+            await payload_logger.log(_LogEntry(
+                request_id=rid,
+                event_type="org.kubeflow.serving.inference.response",
+                model_name=name,
+                payload=json.dumps(result).encode(),
+            ))
         return result
 
     @app.api_route(

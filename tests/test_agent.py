@@ -348,3 +348,43 @@ class TestLoggerStoreAndBatching:
 
         total = sum(pq.read_table(f).num_rows for f in files)
         assert total == 7
+
+
+class TestAgentChainLogging:
+    def test_proxy_logs_request_and_response_cloudevents(self, tmp_path):
+        """Agent chain parity (cmd/agent/main.go:429-449): the proxy
+        captures request AND response payloads through the PayloadLogger
+        when one is attached."""
+        from kserve_amd.agent.payload_logger import PayloadLogger
+
+        def backend(request: httpx.Request) -> httpx.Response:
+            body = json.loads(request.content)
+            return httpx.Response(
+                200, json={"predictions": [x * 2 for x in body["instances"]]}
+            )
+
+        pl = PayloadLogger(store_path=str(tmp_path), num_workers=1)
+        app = create_batcher_proxy_app(
+            "http://backend", "m", max_batch_size=4, max_latency_ms=10,
+            transport=httpx.MockTransport(backend), payload_logger=pl,
+        )
+        with TestClient(app) as c:
+            r = c.post(
+                "/v1/models/m:predict",
+                json={"instances": [3]},
+                headers={"x-request-id": "req-7"},
+            )
+            assert r.status_code == 200
+            # drain the async logger inside the app's loop
+            c.portal.call(pl.stop)
+        files = sorted(p.name for p in tmp_path.iterdir())
+        assert any("request" in f for f in files), files
+        assert any("response" in f for f in files), files
+        # json marshaller stores the raw payload; file name carries the id
+        req_file = [p for p in tmp_path.iterdir() if "request" in p.name][0]
+        assert req_file.name.startswith("req-7-")
+        logged = json.loads(req_file.read_text())
+        assert logged["instances"] == [3]
+        resp_file = [p for p in tmp_path.iterdir()
+                     if "response" in p.name][0]
+        assert json.loads(resp_file.read_text())["predictions"] == [6]
