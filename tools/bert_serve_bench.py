@@ -94,10 +94,17 @@ def start_server(port: int, seq_len: int):
 
         from kserve_amd.agent.batcher import create_batcher_proxy_app
 
+        payload_logger = None
+        if os.environ.get("KS_BERT_LOGGER") == "1":
+            from kserve_amd.agent.payload_logger import PayloadLogger
+
+            payload_logger = PayloadLogger(
+                store_path="/tmp/bert-payload-logs", num_workers=2)
         proxy = create_batcher_proxy_app(
             f"http://127.0.0.1:{port}", "bert",
             max_batch_size=int(os.environ.get("KS_BATCH_MAX", "64")),
             max_latency_ms=int(os.environ.get("KS_BATCH_LAT_MS", "20")),
+            payload_logger=payload_logger,
         )
         pcfg = uvicorn.Config(proxy, host="127.0.0.1", port=port + 1,
                               log_level="error")
@@ -216,7 +223,9 @@ async def drive_v1(port, seq_len, requests, concurrency, batcher):
     lat.sort()
     return {
         "metric": "bert 1-seq V1 requests/s "
-                  + ("via batcher sidecar" if batcher else "direct"),
+                  + ("via batcher sidecar" if batcher else "direct")
+                  + (" + payload logger"
+                     if os.environ.get("KS_BERT_LOGGER") == "1" else ""),
         "value": round(requests / elapsed, 1),
         "seq_len": seq_len, "requests": requests,
         "concurrency": concurrency,
@@ -241,6 +250,9 @@ def main():
     ap.add_argument("--batcher", action="store_true",
                     help="with --v1-singles: route through the batcher "
                          "proxy sidecar (port+1)")
+    ap.add_argument("--logger", action="store_true",
+                    help="with --batcher: attach the payload logger "
+                         "(CloudEvents to a local file sink)")
     args = ap.parse_args()
     if args.serve:
         start_server(args.port, args.seq_len)
@@ -253,6 +265,9 @@ def main():
     env = dict(os.environ)
     if args.batcher:
         env["KS_BERT_BATCHER"] = "1"
+    if args.logger:
+        env["KS_BERT_LOGGER"] = "1"
+        os.environ["KS_BERT_LOGGER"] = "1"  # label only
     proc = subprocess.Popen(
         [sys.executable, os.path.abspath(__file__), "--serve",
          "--port", str(args.port), "--seq-len", str(args.seq_len)],
