@@ -40,7 +40,8 @@ def main():
                         help="concurrent sequences per GPU")
     parser.add_argument("--prompt-len", type=int, default=512)
     parser.add_argument("--model", type=str, default="llama-3-8b",
-                        choices=["llama-3-8b", "llama-3-70b", "mixtral-8x7b", "tiny"])
+                        choices=["llama-3-8b", "llama-3-70b", "mixtral-8x7b",
+                                 "mistral-7b", "tiny"])
     parser.add_argument("--tp", type=int, default=1,
                         help="tensor-parallel degree (ranks per engine)")
     parser.add_argument("--kv-blocks", type=int, default=0,
@@ -81,6 +82,8 @@ def main():
         mcfg = ModelConfig.llama3_70b()
     elif args.model == "mixtral-8x7b":
         mcfg = ModelConfig.mixtral_8x7b()
+    elif args.model == "mistral-7b":
+        mcfg = ModelConfig.mistral_7b()
     else:
         mcfg = ModelConfig(
             vocab_size=1024, hidden_size=512, intermediate_size=1024,

@@ -49,10 +49,9 @@ class ModelConfig:
     expert_parallel: bool = False
     # HF rope_scaling dict (Llama-3.1 'llama3' type supported)
     rope_scaling: Optional[dict] = None
-    # Mistral-style sliding-window size from the HF config. This engine
-    # runs FULL attention regardless (strictly more context than the
-    # windowed reference behavior); recorded so servers can warn when a
-    # model declares one.
+    # Mistral-style sliding-window size from the HF config; enforced in
+    # all three attention kernels (decode/flash/paged-context) as an
+    # absolute-position bound since round 2.
     sliding_window: int = None
 
     def __post_init__(self):
@@ -90,6 +89,21 @@ class ModelConfig:
             rope_theta=500000.0,
             max_position_embeddings=8192,
             model_name="llama-3-70b",
+        )
+
+    @classmethod
+    def mistral_7b(cls) -> "ModelConfig":
+        return cls(
+            vocab_size=32000,
+            hidden_size=4096,
+            intermediate_size=14336,
+            num_layers=32,
+            num_heads=32,
+            num_kv_heads=8,
+            rope_theta=10000.0,
+            max_position_embeddings=8192,
+            sliding_window=4096,
+            model_name="mistral-7b",
         )
 
     @classmethod
